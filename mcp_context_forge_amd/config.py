@@ -1,0 +1,166 @@
+"""Gateway settings.
+
+MI355X-native analog of the reference's pydantic-settings ``Settings``
+(reference: mcpgateway/config.py:175, ~1,150 fields). We keep the same
+env-var contract for the fields that matter operationally and add the GPU
+pipeline knobs that replace the reference's Redis/cache configuration
+(reference: mcpgateway/config.py:2891 cache_type — here the caches are
+HBM-resident tensors per BASELINE.json).
+
+pydantic-settings is not in the image, so env loading is implemented
+directly: every field of :class:`Settings` can be overridden with the
+``FORGE_`` prefix (e.g. ``FORGE_PORT=8080``), falling back to the
+reference's unprefixed names where they exist (``HOST``, ``PORT``,
+``DATABASE_URL``, ``JWT_SECRET_KEY`` ...).
+"""
+
+from __future__ import annotations
+
+import json
+import os
+from typing import Any, List, Optional
+
+from pydantic import BaseModel, Field
+
+
+class Settings(BaseModel):
+    # --- app / network ---
+    app_name: str = "MCP Context Forge AMD"
+    host: str = "0.0.0.0"
+    port: int = 4444
+    app_root_path: str = ""
+    environment: str = "development"
+
+    # --- auth (reference: mcpgateway/auth.py, config basic_auth_* / jwt_*) ---
+    basic_auth_user: str = "admin"
+    basic_auth_password: str = "changeme"
+    auth_required: bool = True
+    jwt_secret_key: str = "my-test-key"
+    jwt_algorithm: str = "HS256"
+    jwt_audience: str = "mcpgateway-api"
+    jwt_issuer: str = "mcpgateway"
+    token_expiry: int = 10080  # minutes
+    platform_admin_email: str = "admin@example.com"
+    platform_admin_password: str = "changeme"
+
+    # --- database (reference: db.py:122 build_engine) ---
+    database_url: str = "sqlite:///./mcp.db"
+    db_pool_size: int = 16
+    db_max_overflow: int = 8
+
+    # --- protocol ---
+    protocol_version: str = "2025-11-25"
+
+    # --- federation (reference: config federation_*, health_check_interval :2766) ---
+    federation_enabled: bool = True
+    federation_timeout: int = 30
+    federation_sync_timeout: int = 60
+    health_check_interval: int = 60
+    health_check_timeout: int = 10
+    unhealthy_threshold: int = 3
+    max_tool_retries: int = 3
+    retry_base_delay_ms: int = 100
+    retry_max_delay_ms: int = 5000
+    retry_jitter: float = 0.25
+
+    # --- transports / sessions ---
+    sse_keepalive_interval: int = 30
+    session_ttl: int = 3600
+    message_ttl: int = 600
+    event_store_max_events: int = 512
+    websocket_ping_interval: int = 30
+
+    # --- rate limiting (reference: middleware/rate_limit_middleware.py) ---
+    rate_limit_enabled: bool = False
+    rate_limit_requests_per_minute: int = 6000
+    rate_limit_burst: int = 200
+
+    # --- security / validation (reference: common/validators.py SecurityValidator) ---
+    max_request_body_bytes: int = 4 * 1024 * 1024
+    max_json_depth: int = 64
+    max_string_length: int = 1 * 1024 * 1024
+    max_header_bytes: int = 16 * 1024
+    security_headers_enabled: bool = True
+    cors_allow_origins: List[str] = Field(default_factory=lambda: ["*"])
+    skip_ssl_verify: bool = False
+
+    # --- plugins (reference: plugins/config.yaml + PLUGINS_ENABLED) ---
+    plugins_enabled: bool = True
+    plugin_config_file: str = "plugins/config.yaml"
+
+    # --- metrics / observability ---
+    metrics_buffer_flush_interval: float = 60.0
+    metrics_buffer_max_size: int = 1000
+    otel_enable_observability: bool = False
+    log_level: str = "INFO"
+
+    # --- admin UI / APIs ---
+    admin_ui_enabled: bool = True
+    admin_api_enabled: bool = True
+
+    # --- GPU pipeline (MI355X-native; replaces cache_type=redis in the reference) ---
+    gpu_enabled: bool = True          # auto-falls back to CPU reference path when no HIP device
+    gpu_batch_max_requests: int = 8192
+    gpu_batch_window_us: int = 500    # adaptive micro-batch window
+    gpu_feature_dim: int = 4096       # hashed count-vector dim for classifiers/semantic cache
+    gpu_classifier_hidden: int = 1024
+    gpu_classifier_classes: int = 8
+    gpu_semcache_capacity: int = 65536
+    gpu_semcache_threshold: float = 0.92
+    gpu_streams: int = 2              # compute + copy overlap
+    gpu_dtype: str = "bf16"
+
+    # --- multi-GPU scale-out (reference analog: session_affinity over Redis; here RCCL) ---
+    world_size: int = 1
+    rank: int = 0
+    upstream_shards: int = 64         # upstreams per GPU rank
+
+    # --- well-known / misc ---
+    docs_enabled: bool = True
+    version: str = "0.1.0"
+
+    @classmethod
+    def from_env(cls, env: Optional[dict] = None) -> "Settings":
+        env = dict(os.environ if env is None else env)
+        values: dict[str, Any] = {}
+        for name, field_info in cls.model_fields.items():
+            raw = env.get("FORGE_" + name.upper())
+            if raw is None:
+                raw = env.get(name.upper())
+            if raw is None:
+                continue
+            ann = field_info.annotation
+            try:
+                if ann is bool:
+                    values[name] = raw.strip().lower() in ("1", "true", "yes", "on")
+                elif ann is int:
+                    values[name] = int(raw)
+                elif ann is float:
+                    values[name] = float(raw)
+                elif ann == List[str]:
+                    values[name] = json.loads(raw) if raw.startswith("[") else [s.strip() for s in raw.split(",")]
+                else:
+                    values[name] = raw
+            except (ValueError, json.JSONDecodeError):
+                continue
+        return cls(**values)
+
+
+_settings: Optional[Settings] = None
+
+
+def get_settings() -> Settings:
+    global _settings
+    if _settings is None:
+        _settings = Settings.from_env()
+    return _settings
+
+
+def set_settings(s: Settings) -> None:
+    global _settings
+    _settings = s
+
+
+def reset_settings() -> None:
+    global _settings
+    _settings = None

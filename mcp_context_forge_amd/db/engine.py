@@ -1,0 +1,111 @@
+"""Engine construction + native migration chain.
+
+Reference analog: mcpgateway/db.py:122 `build_engine` and
+mcpgateway/bootstrap_db.py (alembic `upgrade head` at startup). Alembic is
+not in this image, so the migration chain is implemented natively: a
+``forge_schema_version`` table records applied revisions; `run_migrations`
+applies the linear chain in order, exactly like `alembic upgrade head`.
+Revision 0001 materializes the full current schema (models.Base.metadata);
+later schema changes MUST be appended as new revisions, never by editing
+0001 — same discipline as the reference's 117-migration chain.
+"""
+
+from __future__ import annotations
+
+import threading
+from contextlib import contextmanager
+from typing import Callable, Iterator, List, Tuple
+
+from sqlalchemy import Connection, create_engine, text
+from sqlalchemy.engine import Engine
+from sqlalchemy.orm import Session, sessionmaker
+from sqlalchemy.pool import StaticPool
+
+from .models import Base
+
+
+def build_engine(database_url: str, pool_size: int = 16, max_overflow: int = 8) -> Engine:
+    """Build a SQLAlchemy engine with pool-error resilience (reference: db.py:617-716)."""
+    kwargs: dict = {"pool_pre_ping": True, "future": True}
+    if database_url.startswith("sqlite"):
+        kwargs["connect_args"] = {"check_same_thread": False, "timeout": 30}
+        if ":memory:" in database_url or database_url == "sqlite://":
+            kwargs["poolclass"] = StaticPool
+    else:
+        kwargs["pool_size"] = pool_size
+        kwargs["max_overflow"] = max_overflow
+    engine = create_engine(database_url, **kwargs)
+    if database_url.startswith("sqlite"):
+        from sqlalchemy import event
+
+        @event.listens_for(engine, "connect")
+        def _set_sqlite_pragma(dbapi_conn, _rec):  # pragma: no cover - driver hook
+            cur = dbapi_conn.cursor()
+            cur.execute("PRAGMA journal_mode=WAL")
+            cur.execute("PRAGMA synchronous=NORMAL")
+            cur.execute("PRAGMA foreign_keys=ON")
+            cur.close()
+
+    return engine
+
+
+# ---------------------------------------------------------------------------
+# Migration chain (alembic-equivalent)
+# ---------------------------------------------------------------------------
+
+def _rev_0001_initial(conn: Connection) -> None:
+    Base.metadata.create_all(conn)
+
+
+# Linear chain: (revision_id, apply_fn). Append-only.
+MIGRATIONS: List[Tuple[str, Callable[[Connection], None]]] = [
+    ("0001_initial_registry", _rev_0001_initial),
+]
+
+
+def run_migrations(engine: Engine) -> List[str]:
+    """Apply pending revisions; returns the list applied (bootstrap_db analog)."""
+    applied: List[str] = []
+    with engine.begin() as conn:
+        conn.execute(
+            text(
+                "CREATE TABLE IF NOT EXISTS forge_schema_version ("
+                "revision VARCHAR(64) PRIMARY KEY, applied_at TIMESTAMP DEFAULT CURRENT_TIMESTAMP)"
+            )
+        )
+        done = {r[0] for r in conn.execute(text("SELECT revision FROM forge_schema_version"))}
+        for rev, fn in MIGRATIONS:
+            if rev in done:
+                continue
+            fn(conn)
+            conn.execute(text("INSERT INTO forge_schema_version (revision) VALUES (:r)"), {"r": rev})
+            applied.append(rev)
+    return applied
+
+
+class Database:
+    """Session factory + lifecycle wrapper (reference: db.py SessionLocal + ResilientSession :422)."""
+
+    def __init__(self, database_url: str, pool_size: int = 16, max_overflow: int = 8):
+        self.url = database_url
+        self.engine = build_engine(database_url, pool_size, max_overflow)
+        self._sessionmaker = sessionmaker(bind=self.engine, expire_on_commit=False, future=True)
+        self._lock = threading.Lock()
+
+    def migrate(self) -> List[str]:
+        return run_migrations(self.engine)
+
+    @contextmanager
+    def session(self) -> Iterator[Session]:
+        s = self._sessionmaker()
+        try:
+            yield s
+            s.commit()
+        except Exception:
+            s.rollback()
+            raise
+        finally:
+            s.close()
+
+    def close(self) -> None:
+        self.engine.dispose()

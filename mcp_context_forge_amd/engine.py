@@ -1,0 +1,232 @@
+"""GatewayEngine — composition root + JSON-RPC method dispatch.
+
+Reference analogs: mcpgateway/main.py lifespan (:1443) wires ~60 service
+singletons; `_handle_rpc_authenticated` (:11197) is the method dispatch
+matrix (initialize / ping / tools/list / tools/call / resources/* /
+prompts/* / completion/complete / logging/setLevel / notifications/*).
+
+The engine is transport-agnostic: HTTP (/rpc, /mcp), SSE, WS, stdio and the
+bench harness all feed `handle_rpc_bytes` / `handle_rpc`. When a GPU is
+present the batched entry point `process_rpc_batch` stages the whole batch
+through the HIP plugin pipeline (gpu/pipeline.py) before fan-out.
+"""
+
+from __future__ import annotations
+
+import asyncio
+import json
+import logging
+import time
+from typing import Any, Dict, List, Optional
+
+from .config import Settings, get_settings
+from .db.engine import Database
+from .plugins.framework import PluginManager
+from .plugins.loader import default_chain_specs, load_plugin_manager
+from .protocol import jsonrpc
+from .protocol.mcp import PROTOCOL_VERSION, initialize_result
+from .registry.registry import NotFoundError, Registry
+from .services.content import CompletionService, PromptService, ResourceService, RootService
+from .services.gateway_service import GatewayService
+from .services.metrics import MetricsBuffer
+from .services.sessions import SessionRegistry
+from .services.tool_service import ToolInvocationError, ToolNotFoundError, ToolService
+
+logger = logging.getLogger(__name__)
+
+
+class GatewayEngine:
+    def __init__(self, settings: Optional[Settings] = None, plugin_manager: Optional[PluginManager] = None,
+                 rank: int = 0, world_size: int = 1):
+        self.settings = settings or get_settings()
+        self.rank = rank
+        self.world_size = world_size
+        self.db = Database(self.settings.database_url, self.settings.db_pool_size, self.settings.db_max_overflow)
+        self.db.migrate()
+        self.registry = Registry(self.db)
+        if plugin_manager is not None:
+            self.plugins = plugin_manager
+        elif self.settings.plugins_enabled:
+            import os
+
+            if os.path.exists(self.settings.plugin_config_file):
+                self.plugins = load_plugin_manager(self.settings.plugin_config_file)
+            else:
+                self.plugins = load_plugin_manager(specs=default_chain_specs())
+        else:
+            self.plugins = PluginManager([], enabled=False)
+        self.metrics = MetricsBuffer(self.db, self.settings.metrics_buffer_flush_interval,
+                                     self.settings.metrics_buffer_max_size)
+        self.tool_service = ToolService(self.registry, self.plugins, self.metrics,
+                                        max_retries=self.settings.max_tool_retries)
+        self.gateway_service = GatewayService(self.registry, self.tool_service, self.settings, rank, world_size)
+        self.prompt_service = PromptService(self.registry, self.plugins)
+        self.resource_service = ResourceService(self.registry, self.plugins)
+        self.completion_service = CompletionService(self.registry)
+        self.root_service = RootService()
+        self.sessions = SessionRegistry(self.settings.session_ttl, rank)
+        self.started_at = time.time()
+        self.gpu_pipeline = None  # attached lazily by gpu.pipeline when available
+        self._log_level = "info"
+
+    # -- lifecycle ---------------------------------------------------------------
+    async def startup(self) -> None:
+        if self.settings.federation_enabled and self.settings.health_check_interval > 0:
+            await self.gateway_service.start_health_loop()
+
+    async def shutdown(self) -> None:
+        await self.gateway_service.stop()
+        await self.tool_service.aclose()
+        await self.plugins.shutdown()
+        self.metrics.flush()
+        self.db.close()
+
+    def enable_gpu(self) -> bool:
+        """Attach the HIP batch pipeline if a device is present. Returns success."""
+        if self.gpu_pipeline is not None:
+            return True
+        if not self.settings.gpu_enabled:
+            return False
+        try:
+            import torch
+
+            if not torch.cuda.is_available():
+                return False
+            from .gpu.pipeline import GpuPluginPipeline
+
+            self.gpu_pipeline = GpuPluginPipeline(self)
+            return True
+        except Exception as exc:  # pragma: no cover - GPU-only path
+            logger.warning("GPU pipeline unavailable: %s", exc)
+            return False
+
+    # -- RPC dispatch (reference: main.py:11330-11537 method chain) ----------------
+    async def handle_rpc(self, req: jsonrpc.JSONRPCRequest, user: Optional[str] = None,
+                         server_id: Optional[str] = None, session: Optional[Any] = None,
+                         headers: Optional[Dict[str, str]] = None) -> Optional[jsonrpc.JSONRPCResponse]:
+        method = req.method
+        params = req.params if isinstance(req.params, dict) else {}
+        rid = req.id
+        try:
+            if method == "initialize":
+                if session is not None:
+                    session.initialized = True
+                    session.protocol_version = params.get("protocolVersion")
+                result = initialize_result(params.get("protocolVersion"))
+            elif method in ("ping", "notifications/initialized", "notifications/cancelled", "notifications/roots/list_changed"):
+                if method != "ping":
+                    return None  # notifications get no response
+                result = {}
+            elif method == "tools/list":
+                tools = await self.tool_service.list_tools(server_id=server_id)
+                result = {
+                    "tools": [
+                        {
+                            "name": t["name"],
+                            "description": t.get("description") or "",
+                            "inputSchema": t.get("input_schema") or {"type": "object"},
+                            **({"outputSchema": t["output_schema"]} if t.get("output_schema") else {}),
+                            **({"annotations": t["annotations"]} if t.get("annotations") else {}),
+                        }
+                        for t in tools
+                    ]
+                }
+            elif method == "tools/call":
+                name = params.get("name")
+                if not isinstance(name, str) or not name:
+                    raise jsonrpc.JSONRPCError(jsonrpc.INVALID_PARAMS, "missing tool name")
+                result = await self.tool_service.invoke_tool(
+                    name, params.get("arguments") or {}, user=user, server_id=server_id, headers=headers
+                )
+            elif method == "resources/list":
+                result = {"resources": self.resource_service.list_resources()}
+            elif method == "resources/templates/list":
+                result = {"resourceTemplates": self.resource_service.list_templates()}
+            elif method == "resources/read":
+                uri = params.get("uri")
+                if not uri:
+                    raise jsonrpc.JSONRPCError(jsonrpc.INVALID_PARAMS, "missing uri")
+                result = await self.resource_service.read_resource(uri, user=user)
+            elif method == "resources/subscribe":
+                if session is not None and params.get("uri"):
+                    self.resource_service.subscribe(session.session_id, params["uri"])
+                result = {}
+            elif method == "resources/unsubscribe":
+                if session is not None and params.get("uri"):
+                    self.resource_service.unsubscribe(session.session_id, params["uri"])
+                result = {}
+            elif method == "prompts/list":
+                result = {"prompts": self.prompt_service.list_prompts()}
+            elif method == "prompts/get":
+                name = params.get("name")
+                if not name:
+                    raise jsonrpc.JSONRPCError(jsonrpc.INVALID_PARAMS, "missing prompt name")
+                result = await self.prompt_service.get_prompt(name, params.get("arguments") or {}, user=user)
+            elif method == "completion/complete":
+                result = await self.completion_service.complete(params.get("ref") or {}, params.get("argument") or {})
+            elif method == "roots/list":
+                result = {"roots": self.root_service.list_roots()}
+            elif method == "logging/setLevel":
+                self._log_level = params.get("level", "info")
+                result = {}
+            elif method == "sampling/createMessage":
+                # reference: handlers/sampling.py — gateway has no model client here;
+                # sampling is forwarded to the LLM proxy when configured.
+                raise jsonrpc.JSONRPCError(jsonrpc.METHOD_NOT_FOUND, "sampling requires a configured LLM provider")
+            else:
+                raise jsonrpc.JSONRPCError(jsonrpc.METHOD_NOT_FOUND, f"Method not found: {method}")
+            if req.is_notification:
+                return None
+            return jsonrpc.result_response(rid, result)
+        except jsonrpc.JSONRPCError as exc:
+            if req.is_notification:
+                return None
+            return jsonrpc.JSONRPCResponse(id=rid, error=exc)
+        except ToolNotFoundError as exc:
+            return jsonrpc.error_response(rid, jsonrpc.INVALID_PARAMS, str(exc))
+        except ToolInvocationError as exc:
+            return jsonrpc.error_response(rid, exc.code, str(exc))
+        except Exception as exc:  # pragma: no cover - defensive
+            logger.exception("rpc internal error")
+            return jsonrpc.error_response(rid, jsonrpc.INTERNAL_ERROR, str(exc))
+
+    async def handle_rpc_bytes(self, raw: bytes, user: Optional[str] = None,
+                               server_id: Optional[str] = None, session: Optional[Any] = None,
+                               headers: Optional[Dict[str, str]] = None) -> Optional[bytes]:
+        """Bytes-in/bytes-out single request (transport fast path)."""
+        try:
+            req = jsonrpc.parse_request_bytes(raw)
+        except jsonrpc.JSONRPCError as exc:
+            return jsonrpc.JSONRPCResponse(id=None, error=exc).to_bytes()
+        resp = await self.handle_rpc(req, user=user, server_id=server_id, session=session, headers=headers)
+        return resp.to_bytes() if resp is not None else None
+
+    # -- batched entry point (GPU hot path; bench + micro-batching transports) ----
+    async def process_rpc_batch(self, raws: List[bytes], user: Optional[str] = None,
+                                server_id: Optional[str] = None) -> List[Optional[bytes]]:
+        """Process a batch of raw JSON-RPC requests.
+
+        With a GPU pipeline attached, tools/call requests take the staged
+        HIP plugin chain (parse → scan/mask/validate/classify on device →
+        fan-out → post chain); everything else falls through per-request.
+        """
+        if self.gpu_pipeline is not None:
+            return await self.gpu_pipeline.process_batch(raws, user=user, server_id=server_id)
+        return list(await asyncio.gather(*(self.handle_rpc_bytes(r, user=user, server_id=server_id) for r in raws)))
+
+    # -- health/version ------------------------------------------------------------
+    def health(self) -> Dict[str, Any]:
+        return {"status": "healthy", "uptime_s": round(time.time() - self.started_at, 1)}
+
+    def version_info(self) -> Dict[str, Any]:
+        import torch
+
+        return {
+            "name": "mcp-context-forge-amd",
+            "version": self.settings.version,
+            "protocol_version": PROTOCOL_VERSION,
+            "rank": self.rank,
+            "world_size": self.world_size,
+            "gpu": bool(self.gpu_pipeline),
+            "torch": torch.__version__,
+        }

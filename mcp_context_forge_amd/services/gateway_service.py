@@ -1,0 +1,220 @@
+"""Federation: peer gateway / upstream MCP server lifecycle.
+
+Reference analog: services/gateway_service.py — register_gateway (:1636),
+_initialize_gateway (:5008) → connect + initialize + tools/resources/prompts
+sync (:5648/:5776/:5889), health loop check_health_of_gateways (:4412) with
+bounded concurrency (:4477 limited_check), failure handling (:4362) and
+reactivation (:4529). The reference elects a health-check leader via Redis
+SET NX (:1254); here rank 0 of the RCCL world is the leader by construction
+(parallel/ world), and single-rank deployments just run the loop.
+"""
+
+from __future__ import annotations
+
+import asyncio
+import logging
+import time
+from typing import Any, Dict, List, Optional
+
+from ..config import Settings
+from ..registry.registry import ConflictError, NotFoundError, Registry
+from ..utils import slugify
+from .tool_service import ToolService
+from .upstream import HttpUpstreamClient, InProcUpstream, UpstreamClient, UpstreamError
+
+logger = logging.getLogger(__name__)
+
+
+class GatewayConnectionError(Exception):
+    pass
+
+
+class GatewayService:
+    def __init__(self, registry: Registry, tool_service: ToolService, settings: Optional[Settings] = None,
+                 rank: int = 0, world_size: int = 1):
+        self.registry = registry
+        self.tools = tool_service
+        self.settings = settings or Settings()
+        self.rank = rank
+        self.world_size = world_size
+        self._health_task: Optional[asyncio.Task] = None
+        self._stop = asyncio.Event()
+
+    # -- client construction ----------------------------------------------------
+    def _make_client(self, gateway: Dict[str, Any]) -> UpstreamClient:
+        headers: Dict[str, str] = {}
+        if gateway.get("auth_type") == "bearer" and gateway.get("auth_value"):
+            headers["authorization"] = f"Bearer {gateway['auth_value']}"
+        elif gateway.get("auth_type") == "headers" and gateway.get("auth_value"):
+            import json as _json
+
+            try:
+                headers.update(_json.loads(gateway["auth_value"]))
+            except Exception:
+                pass
+        return HttpUpstreamClient(gateway["url"], headers=headers, timeout=self.settings.federation_timeout)
+
+    # -- registration (reference: register_gateway :1636) -----------------------
+    async def register_gateway(self, name: str, url: str, transport: str = "streamablehttp",
+                               description: str = "", auth_type: Optional[str] = None,
+                               auth_value: Optional[str] = None, tags: Optional[List[str]] = None,
+                               client: Optional[UpstreamClient] = None,
+                               owner_rank: Optional[int] = None) -> Dict[str, Any]:
+        """Synchronous-path registration: connect, initialize, sync capabilities.
+
+        `client` lets tests/bench inject an InProcUpstream (fake upstream
+        harness) — the sync + health machinery is identical either way.
+        """
+        gateway = self.registry.create(
+            "gateway",
+            name=name,
+            url=url,
+            transport=transport,
+            description=description,
+            auth_type=auth_type,
+            auth_value=auth_value,
+            tags=tags or [],
+            status="pending",
+            owner_rank=owner_rank if owner_rank is not None else (hash(name) % self.world_size),
+        )
+        try:
+            await self._initialize_gateway(gateway, client)
+        except Exception as exc:
+            self.registry.update("gateway", gateway["id"], status="unreachable", reachable=False)
+            raise GatewayConnectionError(f"failed to initialize gateway {name}: {exc}") from exc
+        return self.registry.get("gateway", gateway["id"])
+
+    async def _initialize_gateway(self, gateway: Dict[str, Any], client: Optional[UpstreamClient] = None) -> None:
+        """Connect + initialize + sync registry (reference: _initialize_gateway :5008)."""
+        client = client or self._make_client(gateway)
+        init = await asyncio.wait_for(client.initialize(), timeout=self.settings.federation_sync_timeout)
+        tools = await client.list_tools()
+        resources = await client.list_resources()
+        prompts = await client.list_prompts()
+        self.tools.attach_upstream(gateway["id"], client)
+        self.registry.sync_gateway_tools(gateway, tools)
+        self._sync_resources(gateway, resources)
+        self._sync_prompts(gateway, prompts)
+        self.registry.update(
+            "gateway",
+            gateway["id"],
+            status="active",
+            reachable=True,
+            consecutive_failures=0,
+            capabilities=(init or {}).get("capabilities", {}),
+        )
+
+    def _sync_resources(self, gateway: Dict[str, Any], resources: List[Dict[str, Any]]) -> None:
+        for r in resources:
+            uri = r.get("uri")
+            if not uri:
+                continue
+            existing = self.registry.find("resource", uri)
+            fields = dict(uri=uri, name=r.get("name", uri), description=r.get("description", ""),
+                          mime_type=r.get("mimeType", "text/plain"), gateway_id=gateway["id"])
+            if existing:
+                self.registry.update("resource", existing["id"], **fields)
+            else:
+                self.registry.create("resource", **fields)
+
+    def _sync_prompts(self, gateway: Dict[str, Any], prompts: List[Dict[str, Any]]) -> None:
+        for p in prompts:
+            name = p.get("name")
+            if not name:
+                continue
+            qname = f"{slugify(gateway['name'])}-{name}"
+            existing = self.registry.find("prompt", qname)
+            fields = dict(name=qname, description=p.get("description", ""),
+                          argument_schema={"arguments": p.get("arguments", [])}, gateway_id=gateway["id"])
+            if existing:
+                self.registry.update("prompt", existing["id"], **fields)
+            else:
+                self.registry.create("prompt", **fields)
+
+    # -- health loop (reference: check_health_of_gateways :4412) ----------------
+    async def check_health_once(self, concurrency: int = 16) -> Dict[str, bool]:
+        sem = asyncio.Semaphore(concurrency)
+        results: Dict[str, bool] = {}
+
+        async def limited_check(gw: Dict[str, Any]) -> None:
+            async with sem:
+                ok = await self._check_gateway(gw)
+                results[gw["id"]] = ok
+
+        gws = [g for g in self.registry.list("gateway") if g.get("enabled", True)
+               and g.get("owner_rank", 0) % self.world_size == self.rank]
+        await asyncio.gather(*(limited_check(g) for g in gws))
+        return results
+
+    async def _check_gateway(self, gw: Dict[str, Any]) -> bool:
+        client = self.tools.upstream_for(gw["id"])
+        try:
+            if client is None:
+                await self._initialize_gateway(gw)
+                return True
+            ok = await asyncio.wait_for(client.ping(), timeout=self.settings.health_check_timeout)
+        except Exception:
+            ok = False
+        if ok:
+            if not gw.get("reachable", True):
+                # reactivation (reference: _mark_gateway_reachable :4529)
+                try:
+                    await self._initialize_gateway(gw, client)
+                except Exception:
+                    return False
+            self.registry.update("gateway", gw["id"], reachable=True, status="active", consecutive_failures=0)
+            return True
+        fails = int(gw.get("consecutive_failures", 0)) + 1
+        fields: Dict[str, Any] = {"consecutive_failures": fails}
+        if fails >= self.settings.unhealthy_threshold:
+            # failure handling (reference: _handle_gateway_failure :4362)
+            fields.update(reachable=False, status="unreachable")
+            for t in self.registry.tools_for_gateway(gw["id"]):
+                self.registry.update("tool", t["id"], reachable=False)
+        self.registry.update("gateway", gw["id"], **fields)
+        return False
+
+    async def start_health_loop(self) -> None:
+        if self._health_task is not None:
+            return
+        self._stop.clear()
+
+        async def loop() -> None:
+            while not self._stop.is_set():
+                try:
+                    await self.check_health_once()
+                except Exception as exc:  # pragma: no cover
+                    logger.warning("health loop error: %s", exc)
+                try:
+                    await asyncio.wait_for(self._stop.wait(), timeout=self.settings.health_check_interval)
+                except asyncio.TimeoutError:
+                    pass
+
+        self._health_task = asyncio.create_task(loop())
+
+    async def stop(self) -> None:
+        self._stop.set()
+        if self._health_task:
+            self._health_task.cancel()
+            try:
+                await self._health_task
+            except (asyncio.CancelledError, Exception):
+                pass
+            self._health_task = None
+
+    async def refresh_gateway(self, gateway_id: str) -> Dict[str, Any]:
+        """Manual refresh (reference: refresh_gateway_manually :6548)."""
+        gw = self.registry.get("gateway", gateway_id)
+        await self._initialize_gateway(gw, self.tools.upstream_for(gateway_id))
+        return self.registry.get("gateway", gateway_id)
+
+    async def delete_gateway(self, gateway_id: str) -> None:
+        gw = self.registry.get("gateway", gateway_id)
+        self.registry.update("gateway", gateway_id, status="deleting")
+        for t in self.registry.tools_for_gateway(gateway_id):
+            self.registry.delete("tool", t["id"])
+        client = self.tools.upstream_for(gateway_id)
+        if client:
+            await client.aclose()
+            self.tools._upstreams.pop(gateway_id, None)
+        self.registry.delete("gateway", gateway_id)

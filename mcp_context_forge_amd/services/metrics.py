@@ -1,0 +1,129 @@
+"""Metrics buffering + rollups.
+
+Reference analogs: services/metrics_buffer_service.py:84-131 (60 s flush /
+1000-entry force-flush), services/metrics_rollup_service.py (hourly rollups),
+services/metrics_query_service.py. On multi-GPU runs the per-rank buffers
+are aggregated with a RCCL all-reduce instead of per-row DB writes from
+every worker — see parallel/collectives.py.
+"""
+
+from __future__ import annotations
+
+import datetime
+import threading
+import time
+from collections import defaultdict
+from typing import Any, Dict, List, Optional
+
+from ..db.engine import Database
+from ..db.models import DbMetricRollup, DbToolMetric
+
+
+class MetricsBuffer:
+    def __init__(self, db: Optional[Database] = None, flush_interval: float = 60.0, max_size: int = 1000):
+        self.db = db
+        self.flush_interval = flush_interval
+        self.max_size = max_size
+        self._rows: List[Dict[str, Any]] = []
+        self._lock = threading.Lock()
+        self._last_flush = time.monotonic()
+        # live counters for /metrics + admin dashboards
+        self.counters: Dict[str, float] = defaultdict(float)
+        self.latency_sum_ms: Dict[str, float] = defaultdict(float)
+        self.latency_count: Dict[str, int] = defaultdict(int)
+
+    def record_tool_metric(self, tool_id: str, response_time_ms: float, success: bool, error: Optional[str] = None) -> None:
+        with self._lock:
+            self._rows.append(
+                {"tool_id": tool_id, "response_time_ms": response_time_ms, "is_success": success, "error_message": error}
+            )
+            self.counters["tool_invocations_total"] += 1
+            if not success:
+                self.counters["tool_errors_total"] += 1
+            self.latency_sum_ms[tool_id] += response_time_ms
+            self.latency_count[tool_id] += 1
+            need_flush = len(self._rows) >= self.max_size
+        if need_flush:
+            self.flush()
+
+    def record_batch(self, tool_ids: List[str], response_time_ms: float, successes: List[bool]) -> None:
+        """Batched record from the GPU pipeline: one lock, N rows."""
+        with self._lock:
+            for tid, ok in zip(tool_ids, successes):
+                self._rows.append({"tool_id": tid, "response_time_ms": response_time_ms, "is_success": ok, "error_message": None})
+                self.latency_sum_ms[tid] += response_time_ms
+                self.latency_count[tid] += 1
+            self.counters["tool_invocations_total"] += len(tool_ids)
+            self.counters["tool_errors_total"] += sum(1 for s in successes if not s)
+            need_flush = len(self._rows) >= self.max_size
+        if need_flush:
+            self.flush()
+
+    def maybe_flush(self) -> None:
+        if time.monotonic() - self._last_flush >= self.flush_interval:
+            self.flush()
+
+    def flush(self) -> int:
+        with self._lock:
+            rows, self._rows = self._rows, []
+            self._last_flush = time.monotonic()
+        if not rows or self.db is None:
+            return len(rows)
+        with self.db.session() as s:
+            s.bulk_insert_mappings(DbToolMetric, rows)
+        return len(rows)
+
+    def snapshot(self) -> Dict[str, Any]:
+        with self._lock:
+            top = sorted(self.latency_count.items(), key=lambda kv: -kv[1])[:25]
+            return {
+                "counters": dict(self.counters),
+                "pending_rows": len(self._rows),
+                "top_tools": [
+                    {
+                        "tool_id": t,
+                        "count": c,
+                        "avg_ms": self.latency_sum_ms[t] / max(c, 1),
+                    }
+                    for t, c in top
+                ],
+            }
+
+
+def rollup_hourly(db: Database, entity_type: str = "tool") -> int:
+    """Raw rows → hourly rollups, then prune raw (reference: metrics_rollup_service)."""
+    from sqlalchemy import delete, func, select
+
+    created = 0
+    with db.session() as s:
+        rows = s.execute(
+            select(
+                DbToolMetric.tool_id,
+                func.strftime("%Y-%m-%d %H:00:00", DbToolMetric.timestamp).label("hour")
+                if db.url.startswith("sqlite")
+                else func.date_trunc("hour", DbToolMetric.timestamp).label("hour"),
+                func.count(),
+                func.sum(1 - func.cast(DbToolMetric.is_success, __import__("sqlalchemy").Integer)),
+                func.sum(DbToolMetric.response_time_ms),
+                func.min(DbToolMetric.response_time_ms),
+                func.max(DbToolMetric.response_time_ms),
+            ).group_by(DbToolMetric.tool_id, "hour")
+        ).all()
+        for tool_id, hour, count, errs, total, mn, mx in rows:
+            if isinstance(hour, str):
+                hour = datetime.datetime.fromisoformat(hour)
+            s.add(
+                DbMetricRollup(
+                    entity_type=entity_type,
+                    entity_id=tool_id,
+                    hour=hour,
+                    count=count,
+                    error_count=int(errs or 0),
+                    total_ms=float(total or 0.0),
+                    min_ms=float(mn or 0.0),
+                    max_ms=float(mx or 0.0),
+                )
+            )
+            created += 1
+        s.execute(delete(DbToolMetric))
+    return created

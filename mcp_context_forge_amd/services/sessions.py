@@ -1,0 +1,140 @@
+"""Transport session registry + resumable event store.
+
+Reference analogs: cache/session_registry.py (memory/redis/database backends,
+broadcast :1271, respond loop :1499), transports/streamablehttp_transport.py
+InMemoryEventStore (:467, replay :615), transports/redis_event_store.py.
+
+MI355X mapping: the Redis backend's cross-worker role is
+replaced by rank-ownership — each session is owned by exactly one GPU rank
+and cross-rank messages ride the RCCL control channel (parallel/). The
+`database` backend is kept for durability/multi-node parity; `memory` is the
+single-rank fast path.
+"""
+
+from __future__ import annotations
+
+import asyncio
+import itertools
+import time
+import uuid
+from collections import deque
+from dataclasses import dataclass, field
+from typing import Any, AsyncIterator, Deque, Dict, List, Optional, Tuple
+
+
+@dataclass
+class SessionEvent:
+    event_id: str
+    message: dict
+
+
+@dataclass
+class Session:
+    session_id: str
+    transport: str = "streamablehttp"
+    server_id: Optional[str] = None
+    user: Optional[str] = None
+    owner_rank: int = 0
+    created_at: float = field(default_factory=time.monotonic)
+    last_accessed: float = field(default_factory=time.monotonic)
+    queue: "asyncio.Queue[dict]" = field(default_factory=asyncio.Queue)
+    initialized: bool = False
+    protocol_version: Optional[str] = None
+
+
+class EventStore:
+    """Per-session resumable event log with Last-Event-ID replay
+    (reference: InMemoryEventStore.replay_events_after, streamablehttp_transport.py:615)."""
+
+    def __init__(self, max_events_per_session: int = 512):
+        self.max_events = max_events_per_session
+        self._events: Dict[str, Deque[SessionEvent]] = {}
+        self._counter = itertools.count(1)
+
+    def store(self, session_id: str, message: dict) -> str:
+        eid = f"{session_id}-{next(self._counter)}"
+        dq = self._events.setdefault(session_id, deque(maxlen=self.max_events))
+        dq.append(SessionEvent(eid, message))
+        return eid
+
+    def replay_after(self, session_id: str, last_event_id: Optional[str]) -> List[SessionEvent]:
+        dq = self._events.get(session_id)
+        if not dq:
+            return []
+        if last_event_id is None:
+            return list(dq)
+        out: List[SessionEvent] = []
+        seen = False
+        for ev in dq:
+            if seen:
+                out.append(ev)
+            elif ev.event_id == last_event_id:
+                seen = True
+        return out if seen else list(dq)
+
+    def drop(self, session_id: str) -> None:
+        self._events.pop(session_id, None)
+
+
+class SessionRegistry:
+    """Memory-backend session registry (reference: cache/session_registry.py:105-215)."""
+
+    def __init__(self, ttl_s: float = 3600.0, rank: int = 0, event_store: Optional[EventStore] = None):
+        self.ttl_s = ttl_s
+        self.rank = rank
+        self._sessions: Dict[str, Session] = {}
+        self.event_store = event_store or EventStore()
+
+    def create(self, transport: str = "streamablehttp", server_id: Optional[str] = None,
+               user: Optional[str] = None, session_id: Optional[str] = None) -> Session:
+        sid = session_id or uuid.uuid4().hex
+        sess = Session(session_id=sid, transport=transport, server_id=server_id, user=user, owner_rank=self.rank)
+        self._sessions[sid] = sess
+        return sess
+
+    def get(self, session_id: str) -> Optional[Session]:
+        sess = self._sessions.get(session_id)
+        if sess:
+            sess.last_accessed = time.monotonic()
+        return sess
+
+    def remove(self, session_id: str) -> None:
+        self._sessions.pop(session_id, None)
+        self.event_store.drop(session_id)
+
+    def count(self) -> int:
+        return len(self._sessions)
+
+    async def broadcast(self, session_id: str, message: dict) -> bool:
+        """Deliver a message to the transport holding `session_id`
+        (reference: session_registry.broadcast :1271)."""
+        sess = self._sessions.get(session_id)
+        if sess is None:
+            return False
+        self.event_store.store(session_id, message)
+        await sess.queue.put(message)
+        return True
+
+    async def respond_stream(self, session_id: str, keepalive_s: float = 30.0) -> AsyncIterator[Tuple[Optional[str], dict]]:
+        """Yield (event_id, message) for an SSE/streamable GET consumer
+        (reference: respond loop session_registry.py:1499)."""
+        sess = self._sessions.get(session_id)
+        if sess is None:
+            return
+        while True:
+            try:
+                message = await asyncio.wait_for(sess.queue.get(), timeout=keepalive_s)
+            except asyncio.TimeoutError:
+                yield None, {"type": "keepalive"}
+                continue
+            if message.get("__close__"):
+                return
+            yield message.get("__event_id__"), message
+
+    def cleanup_expired(self) -> int:
+        """Stale-session reaper (reference: session_registry.py:1879-2034)."""
+        now = time.monotonic()
+        stale = [sid for sid, s in self._sessions.items() if now - s.last_accessed > self.ttl_s]
+        for sid in stale:
+            self.remove(sid)
+        return len(stale)

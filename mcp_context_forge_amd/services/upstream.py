@@ -1,0 +1,267 @@
+"""Upstream MCP clients — how the gateway talks to federated servers.
+
+Reference analogs: gateway_service.connect_to_streamablehttp_server (:7070)
+and connect_to_sse_server (:6900); tool_service MCP dispatch (:5849);
+upstream_session_registry (session reuse). Implemented directly over httpx —
+no MCP SDK dependency — because the wire format is plain JSON-RPC.
+
+`InProcUpstream` is the fake-upstream test/bench harness (reference analog:
+the containerized fast_time_server auto-registered in docker-compose.yml:1485):
+it round-trips the request through real JSON bytes so the serialization cost
+is honest, without a socket.
+"""
+
+from __future__ import annotations
+
+import asyncio
+import itertools
+import json
+import time
+from typing import Any, Awaitable, Callable, Dict, List, Optional
+
+import httpx
+
+from ..protocol import jsonrpc
+from ..protocol.mcp import PROTOCOL_VERSION
+
+
+class UpstreamError(Exception):
+    def __init__(self, message: str, code: int = jsonrpc.SERVER_UNAVAILABLE):
+        self.code = code
+        super().__init__(message)
+
+
+class UpstreamClient:
+    """Interface for one upstream MCP server."""
+
+    async def initialize(self) -> Dict[str, Any]:
+        raise NotImplementedError
+
+    async def list_tools(self) -> List[Dict[str, Any]]:
+        raise NotImplementedError
+
+    async def list_resources(self) -> List[Dict[str, Any]]:
+        return []
+
+    async def list_prompts(self) -> List[Dict[str, Any]]:
+        return []
+
+    async def call_tool(self, name: str, arguments: Dict[str, Any], headers: Optional[Dict[str, str]] = None) -> Dict[str, Any]:
+        raise NotImplementedError
+
+    async def ping(self) -> bool:
+        raise NotImplementedError
+
+    async def aclose(self) -> None:
+        pass
+
+
+class HttpUpstreamClient(UpstreamClient):
+    """Streamable-HTTP JSON-RPC client with session reuse.
+
+    One persistent httpx.AsyncClient per upstream (reference:
+    services/http_client_service.py:57 SharedHttpClient pooling).
+    """
+
+    _ids = itertools.count(1)
+
+    def __init__(self, url: str, headers: Optional[Dict[str, str]] = None, timeout: float = 30.0,
+                 client: Optional[httpx.AsyncClient] = None):
+        self.url = url
+        self.base_headers = dict(headers or {})
+        self.timeout = timeout
+        self._client = client or httpx.AsyncClient(timeout=timeout)
+        self._owned = client is None
+        self.session_id: Optional[str] = None
+        self._init_lock = asyncio.Lock()
+        self.initialized = False
+
+    async def _rpc(self, method: str, params: Any = None, notification: bool = False,
+                   extra_headers: Optional[Dict[str, str]] = None) -> Any:
+        body: Dict[str, Any] = {"jsonrpc": "2.0", "method": method}
+        if params is not None:
+            body["params"] = params
+        if not notification:
+            body["id"] = next(self._ids)
+        headers = {"content-type": "application/json", "accept": "application/json, text/event-stream"}
+        headers.update(self.base_headers)
+        if extra_headers:
+            headers.update(extra_headers)
+        if self.session_id:
+            headers["mcp-session-id"] = self.session_id
+        try:
+            resp = await self._client.post(self.url, content=json.dumps(body).encode(), headers=headers)
+        except httpx.HTTPError as exc:
+            raise UpstreamError(f"upstream {self.url} unreachable: {exc}") from exc
+        sid = resp.headers.get("mcp-session-id")
+        if sid:
+            self.session_id = sid
+        if notification:
+            return None
+        if resp.status_code >= 400:
+            raise UpstreamError(f"upstream {self.url} HTTP {resp.status_code}")
+        ctype = resp.headers.get("content-type", "")
+        data: Optional[dict] = None
+        if ctype.startswith("text/event-stream"):
+            # single JSON-RPC response delivered over SSE framing
+            for line in resp.text.splitlines():
+                if line.startswith("data:"):
+                    data = json.loads(line[5:].strip())
+        else:
+            data = resp.json()
+        if data is None:
+            raise UpstreamError(f"upstream {self.url}: empty response")
+        if "error" in data:
+            err = data["error"]
+            raise UpstreamError(f"upstream error {err.get('code')}: {err.get('message')}", code=err.get("code", jsonrpc.SERVER_ERROR))
+        return data.get("result")
+
+    async def initialize(self) -> Dict[str, Any]:
+        async with self._init_lock:
+            if self.initialized:
+                return {}
+            result = await self._rpc(
+                "initialize",
+                {
+                    "protocolVersion": PROTOCOL_VERSION,
+                    "capabilities": {},
+                    "clientInfo": {"name": "mcp-context-forge-amd", "version": "0.1.0"},
+                },
+            )
+            await self._rpc("notifications/initialized", notification=True)
+            self.initialized = True
+            return result or {}
+
+    async def list_tools(self) -> List[Dict[str, Any]]:
+        await self.initialize()
+        result = await self._rpc("tools/list", {})
+        return (result or {}).get("tools", [])
+
+    async def list_resources(self) -> List[Dict[str, Any]]:
+        await self.initialize()
+        try:
+            result = await self._rpc("resources/list", {})
+            return (result or {}).get("resources", [])
+        except UpstreamError:
+            return []
+
+    async def list_prompts(self) -> List[Dict[str, Any]]:
+        await self.initialize()
+        try:
+            result = await self._rpc("prompts/list", {})
+            return (result or {}).get("prompts", [])
+        except UpstreamError:
+            return []
+
+    async def call_tool(self, name: str, arguments: Dict[str, Any], headers: Optional[Dict[str, str]] = None) -> Dict[str, Any]:
+        await self.initialize()
+        result = await self._rpc("tools/call", {"name": name, "arguments": arguments}, extra_headers=headers)
+        return result or {}
+
+    async def ping(self) -> bool:
+        try:
+            await self._rpc("ping")
+            return True
+        except UpstreamError:
+            return False
+
+    async def aclose(self) -> None:
+        if self._owned:
+            await self._client.aclose()
+
+
+ToolHandler = Callable[[Dict[str, Any]], Awaitable[Any]]
+
+
+class InProcUpstream(UpstreamClient):
+    """In-process fake MCP upstream with honest bytes-level round-trip.
+
+    Used by unit tests and by bench.py's 64-upstream federation config
+    (BASELINE.json config 2) — the serialization boundary is real (request
+    and response pass through JSON bytes), only the socket is elided.
+    """
+
+    def __init__(self, name: str = "fake", latency_s: float = 0.0):
+        self.name = name
+        self.latency_s = latency_s
+        self._tools: Dict[str, tuple] = {}
+        self.calls = 0
+
+    def add_tool(self, name: str, handler: ToolHandler, description: str = "",
+                 input_schema: Optional[dict] = None, output_schema: Optional[dict] = None) -> None:
+        self._tools[name] = (handler, description, input_schema or {"type": "object"}, output_schema)
+
+    async def initialize(self) -> Dict[str, Any]:
+        return {"protocolVersion": PROTOCOL_VERSION, "serverInfo": {"name": self.name, "version": "0"}, "capabilities": {"tools": {}}}
+
+    async def list_tools(self) -> List[Dict[str, Any]]:
+        out = []
+        for name, (_h, desc, ischema, oschema) in self._tools.items():
+            td = {"name": name, "description": desc, "inputSchema": ischema}
+            if oschema:
+                td["outputSchema"] = oschema
+            out.append(td)
+        return out
+
+    async def call_tool(self, name: str, arguments: Dict[str, Any], headers: Optional[Dict[str, str]] = None) -> Dict[str, Any]:
+        # bytes boundary in
+        wire = json.dumps({"name": name, "arguments": arguments}).encode()
+        req = json.loads(wire)
+        self.calls += 1
+        if self.latency_s:
+            await asyncio.sleep(self.latency_s)
+        ent = self._tools.get(req["name"])
+        if ent is None:
+            raise UpstreamError(f"tool {name} not found on upstream {self.name}", code=jsonrpc.METHOD_NOT_FOUND)
+        handler = ent[0]
+        value = await handler(req["arguments"])
+        if isinstance(value, dict) and "content" in value:
+            result = value
+        else:
+            result = {
+                "content": [{"type": "text", "text": value if isinstance(value, str) else json.dumps(value)}],
+                "structuredContent": value if isinstance(value, (dict, list)) else None,
+                "isError": False,
+            }
+        # bytes boundary out
+        return json.loads(json.dumps(result))
+
+    async def ping(self) -> bool:
+        return True
+
+
+def make_fake_time_upstream(name: str = "fast_time", latency_s: float = 0.0) -> InProcUpstream:
+    """fast_time_server analog (reference benchmark upstream; payload2.json
+    targets its `convert_time` tool)."""
+    up = InProcUpstream(name, latency_s)
+
+    async def get_system_time(args: Dict[str, Any]) -> Any:
+        return {"time": time.strftime("%Y-%m-%dT%H:%M:%SZ", time.gmtime()), "timezone": args.get("timezone", "UTC")}
+
+    async def convert_time(args: Dict[str, Any]) -> Any:
+        return {
+            "time": args.get("time", "2026-01-01T00:00:00Z"),
+            "source_timezone": args.get("source_timezone", "UTC"),
+            "target_timezone": args.get("target_timezone", "UTC"),
+            "converted": True,
+        }
+
+    async def echo(args: Dict[str, Any]) -> Any:
+        return args
+
+    up.add_tool("get_system_time", get_system_time, "Get current system time",
+                {"type": "object", "properties": {"timezone": {"type": "string"}}})
+    up.add_tool(
+        "convert_time", convert_time, "Convert time between timezones",
+        {
+            "type": "object",
+            "properties": {
+                "time": {"type": "string"},
+                "source_timezone": {"type": "string"},
+                "target_timezone": {"type": "string"},
+            },
+            "required": ["time", "source_timezone", "target_timezone"],
+        },
+    )
+    up.add_tool("echo", echo, "Echo arguments back")
+    return up
