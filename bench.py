@@ -1,0 +1,234 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: JSON-RPC tools/call throughput through the full
+gateway pipeline (BASELINE.json metric: tool-call req/s + p50 latency,
+1/2/4/8 MI355X).
+
+Per rank and per step, R synthetic JSON-RPC `tools/call` requests (the
+reference hey rig's convert_time payload shape, tests/hey/payload2.json,
+with varying arguments) are processed through the complete hot path:
+parse → GPU plugin chain (json_guard, deny/pii/regex/harm DFA scans,
+hashed featurize, bf16-MFMA moderation classifier, HBM semantic-cache
+sweep) → federation fan-out to 64 in-proc MCP upstreams per rank (bytes
+round-trip, no socket) → post chain (toon, guards, cache insert) →
+serialize. With N>1 ranks, requests whose target tool is owned by another
+GPU ride an RCCL all-to-all over xGMI (BASELINE config 3), responses ride
+it back.
+
+Contract: W untimed warmup steps, then EXACTLY K timed steps bracketed by
+barrier + torch.cuda.synchronize on both sides; elapsed = MAX over ranks;
+rank 0 prints one JSON line. value = whole-job req/s across all N GPUs.
+CPU fallback (--no-gpu or no device) runs the same semantics on the
+per-request reference path with a smaller default R.
+"""
+
+from __future__ import annotations
+
+import argparse
+import asyncio
+import json
+import os
+import random
+import statistics
+import sys
+import time
+from typing import List
+
+REFERENCE_BEST_RPS = 10454.16  # MCP tools-only 60s/1000u, Rust full mode
+
+
+def make_request(rng: random.Random, tool_names: List[str], rid: int, flagged_frac: float) -> bytes:
+    name = tool_names[rng.randrange(len(tool_names))]
+    r = rng.random()
+    if name.endswith("convert_time"):
+        args = {
+            "time": f"2026-01-{rng.randrange(1,29):02d}T{rng.randrange(24):02d}:{rng.randrange(60):02d}:{rng.randrange(60):02d}Z",
+            "source_timezone": rng.choice(["UTC", "America/New_York", "Europe/London", "Asia/Tokyo"]),
+            "target_timezone": rng.choice(["UTC", "America/Chicago", "Europe/Berlin", "Asia/Kolkata"]),
+        }
+    elif name.endswith("get_system_time"):
+        args = {"timezone": rng.choice(["UTC", "America/New_York", "Europe/Paris"])}
+    else:
+        args = {"msg": f"payload {rng.randrange(1 << 30)} lorem ipsum dolor sit amet", "n": rng.randrange(100)}
+    if r < flagged_frac * 0.5:
+        args["note"] = f"contact me at user{rng.randrange(1000)}@example.com"  # PII slow path
+    elif r < flagged_frac:
+        args["note"] = "this is   spaced\ttext"  # normalizer slow path
+    return json.dumps(
+        {"jsonrpc": "2.0", "id": rid, "method": "tools/call", "params": {"name": name, "arguments": args}},
+        separators=(",", ":"),
+    ).encode()
+
+
+async def build_engine(rank: int, world: int, upstreams: int, use_gpu: bool):
+    from mcp_context_forge_amd.config import Settings
+    from mcp_context_forge_amd.engine import GatewayEngine
+    from mcp_context_forge_amd.services.upstream import make_fake_time_upstream
+
+    settings = Settings(
+        database_url="sqlite://",
+        federation_enabled=False,  # no background health loop during timing
+        auth_required=False,
+        gpu_enabled=use_gpu,
+        rank=rank,
+        world_size=world,
+    )
+    engine = GatewayEngine(settings, rank=rank, world_size=world)
+    # 64 federated upstreams per rank (BASELINE config 2); each exposes 3 tools.
+    for u in range(upstreams):
+        up = make_fake_time_upstream(name=f"up{rank}-{u}")
+        await engine.gateway_service.register_gateway(
+            name=f"up{rank}-{u}", url=f"inproc://up{rank}-{u}", client=up, owner_rank=rank)
+    if use_gpu:
+        ok = engine.enable_gpu()
+        if not ok:
+            raise RuntimeError("GPU requested but pipeline unavailable")
+    return engine
+
+
+def local_tool_names(rank: int, upstreams: int) -> List[str]:
+    names = []
+    for u in range(upstreams):
+        for t in ("convert_time", "get_system_time", "echo"):
+            names.append(f"up{rank}-{u}-{t}")
+    return names
+
+
+async def run_step(engine, world: int, rank: int, raws: List[bytes], dest: List[int]) -> int:
+    """Process one step's traffic; returns number of origin-counted requests."""
+    if world == 1:
+        out = await engine.process_rpc_batch(raws)
+        assert len(out) == len(raws)
+        return len(raws)
+    from mcp_context_forge_amd.parallel import collectives
+
+    buckets: List[List[bytes]] = [[] for _ in range(world)]
+    for r, d in zip(raws, dest):
+        buckets[d].append(r)
+    arrivals = collectives.all_to_all_objects(buckets)  # [src] -> list of raw
+    flat: List[bytes] = []
+    spans = []
+    for src, lst in enumerate(arrivals):
+        spans.append((src, len(lst)))
+        flat.extend(lst)
+    responses = await engine.process_rpc_batch(flat)
+    # return responses to origins
+    back: List[List[bytes]] = [[] for _ in range(world)]
+    off = 0
+    for src, n in spans:
+        back[src] = [r or b"" for r in responses[off:off + n]]
+        off += n
+    returned = collectives.all_to_all_objects(back)
+    got = sum(len(lst) for lst in returned)
+    assert got == len(raws), (got, len(raws))
+    return len(raws)
+
+
+async def main() -> None:
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=20)
+    ap.add_argument("--warmup", type=int, default=5)
+    ap.add_argument("--requests-per-step", type=int, default=0, help="per rank; 0 = auto")
+    ap.add_argument("--upstreams", type=int, default=64)
+    ap.add_argument("--flagged-frac", type=float, default=0.02)
+    ap.add_argument("--no-gpu", action="store_true")
+    args = ap.parse_args()
+
+    import torch
+
+    from mcp_context_forge_amd.parallel import collectives
+
+    rank, world = collectives.init_from_env()
+    n_gpus = max(world, args.gpus if world == 1 else world)
+    use_gpu = torch.cuda.is_available() and not args.no_gpu
+    R = args.requests_per_step or (8192 if use_gpu else 256)
+
+    engine = await build_engine(rank, world, args.upstreams, use_gpu)
+
+    # traffic targets tools across ALL ranks (uniform) — drives the all_to_all
+    all_names: List[List[str]] = [local_tool_names(r, args.upstreams) for r in range(world)]
+
+    def gen_step(step: int):
+        rng = random.Random(1000003 * (rank + 1) + step)
+        raws, dest = [], []
+        for i in range(R):
+            d = rng.randrange(world)
+            raws.append(make_request(rng, all_names[d], step * R + i, args.flagged_frac))
+            dest.append(d)
+        return raws, dest
+
+    def sync():
+        collectives.barrier()
+        if use_gpu:
+            torch.cuda.synchronize()
+
+    for s in range(args.warmup):
+        raws, dest = gen_step(s)
+        await run_step(engine, world, rank, raws, dest)
+    sync()
+
+    step_times: List[float] = []
+    total = 0
+    t_start = time.monotonic()
+    for s in range(args.steps):
+        t0 = time.monotonic()
+        raws, dest = gen_step(10_000 + s)
+        total += await run_step(engine, world, rank, raws, dest)
+        if use_gpu:
+            torch.cuda.synchronize()
+        step_times.append(time.monotonic() - t0)
+    sync()
+    elapsed = time.monotonic() - t_start
+
+    # MAX elapsed over ranks
+    if world > 1:
+        import torch.distributed as dist
+
+        t = torch.tensor([elapsed], dtype=torch.float64,
+                         device="cuda" if (use_gpu and dist.get_backend() == "nccl") else "cpu")
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    if rank == 0:
+        whole_job_requests = R * args.steps * world
+        value = whole_job_requests / elapsed
+        ms_per_step = elapsed / args.steps * 1000.0
+        p50_ms = statistics.median(step_times) * 1000.0
+        stats = engine.gpu_pipeline.stats() if engine.gpu_pipeline else {}
+        out = {
+            "metric": "tool_call_req_per_s",
+            "value": round(value, 2),
+            "unit": "req/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": round(value / REFERENCE_BEST_RPS, 3),
+            "dtype": "bf16",
+            "data": "synthetic",
+            "config": {
+                "model": "gateway-pipeline(deny+regex+pii+schema+moderation-mlp-4096x1024x8+semcache-65536x4096+toon)",
+                "global_batch": R * world,
+                "seq_len": 0,
+                "parallelism": f"shard{world}" if world > 1 else "single",
+                "upstreams": args.upstreams * world,
+                "requests_per_step_per_rank": R,
+                "p50_batch_ms": round(p50_ms, 3),
+                "flagged_frac": args.flagged_frac,
+                "gpu_path": bool(engine.gpu_pipeline),
+                "pipeline_stats": {k: v for k, v in stats.items() if k != "banks"},
+                "reference_metric": "MCP tools/call RPS (BASELINE.md: 10454.16 on x86 CPU compose)",
+            },
+        }
+        print(json.dumps(out))
+    await engine.shutdown()
+    if world > 1:
+        import torch.distributed as dist
+
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    sys.exit(asyncio.run(main()))

@@ -1,0 +1,77 @@
+"""Request micro-batch staging: host parse → packed device buffers.
+
+The MI355X analog of the reference's per-request asyncio hot loop
+(main.py:11197 _handle_rpc_authenticated): N requests are parsed once,
+their canonical argument texts packed into ONE contiguous uint8 buffer +
+offsets, and shipped to HBM in a single H2D copy for the kernel chain.
+"""
+
+from __future__ import annotations
+
+import json
+from dataclasses import dataclass, field
+from typing import Any, Dict, List, Optional, Tuple
+
+import numpy as np
+import torch
+
+from ..protocol import jsonrpc
+
+
+@dataclass
+class ParsedRequest:
+    index: int                      # position in the input batch
+    raw: bytes
+    error: Optional[jsonrpc.JSONRPCError] = None
+    req: Optional[jsonrpc.JSONRPCRequest] = None
+    tool_name: Optional[str] = None
+    arguments: Optional[dict] = None
+    arg_text: bytes = b""           # canonical sorted-compact JSON of arguments
+    tool: Optional[dict] = None     # resolved registry entity
+
+
+def canonical_text(arguments: Any) -> bytes:
+    return json.dumps(arguments or {}, separators=(",", ":"), sort_keys=True, default=str).encode()
+
+
+def parse_batch(raws: List[bytes]) -> List[ParsedRequest]:
+    out: List[ParsedRequest] = []
+    for i, raw in enumerate(raws):
+        pr = ParsedRequest(index=i, raw=raw)
+        try:
+            pr.req = jsonrpc.parse_request_bytes(raw)
+            if pr.req.method == "tools/call":
+                params = pr.req.params if isinstance(pr.req.params, dict) else {}
+                name = params.get("name")
+                if not isinstance(name, str) or not name:
+                    pr.error = jsonrpc.JSONRPCError(jsonrpc.INVALID_PARAMS, "missing tool name")
+                else:
+                    pr.tool_name = name
+                    pr.arguments = params.get("arguments") or {}
+                    pr.arg_text = canonical_text(pr.arguments)
+        except jsonrpc.JSONRPCError as exc:
+            pr.error = exc
+        out.append(pr)
+    return out
+
+
+def pack_texts(texts: List[bytes], device: str = "cuda") -> Tuple[torch.Tensor, torch.Tensor]:
+    """Pack byte strings into (data u8 [total], offsets i32 [B+1]) on device."""
+    offsets = np.zeros(len(texts) + 1, dtype=np.int32)
+    for i, t in enumerate(texts):
+        offsets[i + 1] = offsets[i] + len(t)
+    blob = b"".join(texts)
+    data_np = np.frombuffer(blob, dtype=np.uint8) if blob else np.zeros(1, dtype=np.uint8)
+    data = torch.from_numpy(data_np.copy()).to(device, non_blocking=True)
+    offs = torch.from_numpy(offsets).to(device, non_blocking=True)
+    return data, offs
+
+
+def pad_rows(t: torch.Tensor, multiple: int = 128) -> torch.Tensor:
+    """Zero-pad dim0 to a multiple (MFMA tile requirement); returns possibly-larger view."""
+    b = t.shape[0]
+    rem = b % multiple
+    if rem == 0:
+        return t
+    pad = torch.zeros((multiple - rem,) + tuple(t.shape[1:]), dtype=t.dtype, device=t.device)
+    return torch.cat([t, pad], dim=0)

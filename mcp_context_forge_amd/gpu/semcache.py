@@ -1,0 +1,110 @@
+"""HBM-resident semantic response cache.
+
+Reference analog: plugins/response_cache_by_prompt (count-vector cosine in
+Python dicts) + the Redis registry caches. Per BASELINE.json the cache keys
+live in device HBM as a bf16 matrix [capacity, dim]; lookup is a chunked
+MFMA GEMM (feats @ keys^T) folded by the rows_argmax_merge kernel — no
+Redis, no host round-trip per entry. Results (host objects) stay on host
+indexed by slot.
+"""
+
+from __future__ import annotations
+
+import time
+from typing import Any, List, Optional, Tuple
+
+import numpy as np
+import torch
+
+from ..ops import hip
+
+
+def tool_hash(name: str) -> int:
+    h = 1469598103934665603
+    for b in name.encode():
+        h = ((h ^ b) * 1099511628211) & ((1 << 63) - 1)
+    return h
+
+
+class GpuSemanticCache:
+    def __init__(self, capacity: int = 65536, dim: int = 4096, threshold: float = 0.92,
+                 ttl_s: float = 600.0, device: str = "cuda", chunk: int = 4096):
+        assert capacity % 128 == 0 and chunk % 128 == 0 and dim % 64 == 0
+        self.capacity = capacity
+        self.dim = dim
+        self.threshold = threshold
+        self.ttl_s = ttl_s
+        self.chunk = min(chunk, capacity)
+        self.device = device
+        self.keys = torch.zeros((capacity, dim), dtype=torch.bfloat16, device=device)
+        self.valid = torch.zeros(capacity, dtype=torch.uint8, device=device)
+        # host-side metadata mirrors (slot-indexed)
+        self.tool_hashes = np.zeros(capacity, dtype=np.int64)
+        self.timestamps = np.zeros(capacity, dtype=np.float64)
+        self.results: List[Any] = [None] * capacity
+        self.write_ptr = 0
+        self.size = 0
+        self.hits = 0
+        self.misses = 0
+
+    def lookup(self, feats_bf16: torch.Tensor) -> Tuple[torch.Tensor, torch.Tensor]:
+        """feats [Bpad, dim] bf16 → (best_val fp32 [Bpad], best_idx i32 [Bpad]) on device.
+
+        Chunked sweep over the key matrix; scores for padded/invalid slots are
+        suppressed by the validity mask inside the merge kernel.
+        """
+        bpad = feats_bf16.shape[0]
+        best_val = torch.full((bpad,), -1e30, dtype=torch.float32, device=self.device)
+        best_idx = torch.full((bpad,), -1, dtype=torch.int32, device=self.device)
+        if self.size == 0:
+            return best_val, best_idx
+        active = min(self.size, self.capacity)
+        active_pad = ((active + self.chunk - 1) // self.chunk) * self.chunk
+        for c0 in range(0, active_pad, self.chunk):
+            nc = min(self.chunk, self.capacity - c0)
+            scores = hip.gemm_bt(feats_bf16, self.keys[c0:c0 + nc])
+            hip.rows_argmax_merge(scores, best_val, best_idx, idx_base=c0, valid=self.valid[c0:c0 + nc])
+        return best_val, best_idx
+
+    def resolve_hits(self, best_val: np.ndarray, best_idx: np.ndarray,
+                     tool_hashes: np.ndarray) -> List[Optional[Any]]:
+        """Host-side hit confirmation: threshold + tool identity + TTL
+        (reference semantics: _find_best threshold lookup :163)."""
+        now = time.monotonic()
+        out: List[Optional[Any]] = []
+        for val, idx, th in zip(best_val, best_idx, tool_hashes):
+            if idx < 0 or val < self.threshold:
+                out.append(None)
+                self.misses += 1
+                continue
+            slot = int(idx)
+            if self.tool_hashes[slot] != th or (now - self.timestamps[slot]) > self.ttl_s:
+                out.append(None)
+                self.misses += 1
+                continue
+            self.hits += 1
+            out.append(self.results[slot])
+        return out
+
+    def insert_batch(self, feats_bf16: torch.Tensor, rows: List[int],
+                     tool_hashes: List[int], results: List[Any]) -> None:
+        """Scatter feature rows into cache slots (ring eviction)."""
+        if not rows:
+            return
+        n = len(rows)
+        slots = [(self.write_ptr + i) % self.capacity for i in range(n)]
+        self.write_ptr = (self.write_ptr + n) % self.capacity
+        self.size = min(self.size + n, self.capacity)
+        src = feats_bf16[torch.tensor(rows, dtype=torch.long, device=self.device)].contiguous()
+        slot_t = torch.tensor(slots, dtype=torch.int32, device=self.device)
+        hip.rows_scatter_bf16(src, slot_t, self.keys)
+        self.valid[slot_t.long()] = 1
+        now = time.monotonic()
+        for s, th, res in zip(slots, tool_hashes, results):
+            self.tool_hashes[s] = th
+            self.timestamps[s] = now
+            self.results[s] = res
+
+    def stats(self) -> dict:
+        return {"size": self.size, "capacity": self.capacity, "hits": self.hits, "misses": self.misses,
+                "hbm_bytes": self.keys.numel() * 2}

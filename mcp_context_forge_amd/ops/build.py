@@ -1,0 +1,55 @@
+"""Build the in-tree HIP kernel library for gfx950.
+
+Direct hipcc (no torch-extension ABI): kernels take raw pointers + a
+hipStream_t, Python binds via ctypes on torch tensors' data_ptr(), launching
+on torch's current HIP stream — zero-copy, no sync. The .so is built
+IN-TREE so the gpurun snapshot ships it.
+"""
+
+from __future__ import annotations
+
+import os
+import subprocess
+import sys
+from pathlib import Path
+
+CSRC = Path(__file__).parent / "csrc"
+LIB = Path(__file__).parent / "libforge_hip.so"
+ARCH = os.environ.get("PYTORCH_ROCM_ARCH", "gfx950")
+HIPCC = os.environ.get("HIPCC", "/opt/rocm/bin/hipcc")
+
+SOURCES = ["scan.hip", "featurize.hip", "json_guard.hip", "gemm_bf16.hip", "runtime.hip"]
+
+
+def needs_build() -> bool:
+    if not LIB.exists():
+        return True
+    lib_mtime = LIB.stat().st_mtime
+    for f in SOURCES + ["common.h"]:
+        if (CSRC / f).stat().st_mtime > lib_mtime:
+            return True
+    return False
+
+
+def build(force: bool = False, verbose: bool = True) -> Path:
+    if not force and not needs_build():
+        return LIB
+    cmd = [
+        HIPCC,
+        f"--offload-arch={ARCH}",
+        "-O3",
+        "-std=c++17",
+        "-fPIC",
+        "-shared",
+        "-o",
+        str(LIB),
+    ] + [str(CSRC / f) for f in SOURCES]
+    if verbose:
+        print("[forge-hip]", " ".join(cmd), file=sys.stderr)
+    subprocess.run(cmd, check=True)
+    return LIB
+
+
+if __name__ == "__main__":
+    build(force="--force" in sys.argv)
+    print(LIB)

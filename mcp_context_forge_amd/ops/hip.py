@@ -1,0 +1,153 @@
+"""ctypes bindings for the in-tree HIP kernel library (libforge_hip.so).
+
+All entry points take torch tensors already resident on the device and
+launch on torch's *current* HIP stream — no extra synchronization, composes
+with torch ops and HIP graphs. On a GPU box the library is REQUIRED: any
+failure raises, never a silent eager fallback (per the build contract).
+"""
+
+from __future__ import annotations
+
+import ctypes
+from functools import lru_cache
+from typing import Optional, Tuple
+
+import numpy as np
+import torch
+
+from .dfa import ScanTables
+
+
+class ForgeHipError(RuntimeError):
+    pass
+
+
+@lru_cache(maxsize=1)
+def _load() -> ctypes.CDLL:
+    from .build import LIB, build
+
+    if not LIB.exists():
+        build()
+    lib = ctypes.CDLL(str(LIB))
+    protos = {
+        "forge_scan": [ctypes.c_void_p] * 2 + [ctypes.c_int] + [ctypes.c_void_p] * 3 + [ctypes.c_int] * 2 + [ctypes.c_void_p] * 3,
+        "forge_featurize": [ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int, ctypes.c_int, ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p],
+        "forge_json_guard": [ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int, ctypes.c_int, ctypes.c_int, ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p],
+        "forge_gemm_bt": [ctypes.c_void_p] * 4 + [ctypes.c_int] * 5 + [ctypes.c_void_p],
+        "forge_gemv_head": [ctypes.c_void_p] * 4 + [ctypes.c_int] * 4 + [ctypes.c_void_p],
+        "forge_rows_argmax_merge": [ctypes.c_void_p, ctypes.c_int, ctypes.c_int, ctypes.c_int, ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p],
+        "forge_rows_scatter_bf16": [ctypes.c_void_p] * 3 + [ctypes.c_int] * 2 + [ctypes.c_void_p],
+        "forge_synchronize": [ctypes.c_void_p],
+    }
+    for name, argtypes in protos.items():
+        fn = getattr(lib, name)
+        fn.argtypes = argtypes
+        fn.restype = ctypes.c_int
+    return lib
+
+
+def available() -> bool:
+    try:
+        return torch.cuda.is_available() and _load() is not None
+    except Exception:
+        return False
+
+
+def _stream() -> ctypes.c_void_p:
+    return ctypes.c_void_p(torch.cuda.current_stream().cuda_stream)
+
+
+def _check(name: str, rc: int) -> None:
+    if rc != 0:
+        raise ForgeHipError(f"{name} failed with code {rc}")
+
+
+def _ptr(t: Optional[torch.Tensor]) -> ctypes.c_void_p:
+    return ctypes.c_void_p(t.data_ptr()) if t is not None else ctypes.c_void_p(0)
+
+
+class DeviceScanTables:
+    """ScanTables uploaded to device memory."""
+
+    def __init__(self, tables: ScanTables, device: str = "cuda"):
+        self.meta = tables
+        # torch lacks uint16/uint32 storage in places — bit-identical views as int16/int32
+        self.next = torch.from_numpy(tables.next.view(np.int16).copy()).to(device)
+        self.klass = torch.from_numpy(tables.klass.copy()).to(device)
+        self.accept = torch.from_numpy(tables.accept.view(np.int32).copy()).to(device)
+        self.n_states = tables.n_states
+        self.n_classes = tables.n_classes
+
+
+def scan(data: torch.Tensor, offsets: torch.Tensor, tables: DeviceScanTables,
+         first_end: bool = False) -> Tuple[torch.Tensor, Optional[torch.Tensor]]:
+    """data uint8 [total], offsets int32 [B+1] on device → mask int32 [B]."""
+    batch = offsets.numel() - 1
+    out_mask = torch.zeros(batch, dtype=torch.int32, device=data.device)
+    out_end = torch.full((batch,), -1, dtype=torch.int32, device=data.device) if first_end else None
+    _check("forge_scan", _load().forge_scan(
+        _ptr(data), _ptr(offsets), batch,
+        _ptr(tables.next), _ptr(tables.klass), _ptr(tables.accept),
+        tables.n_states, tables.n_classes,
+        _ptr(out_mask), _ptr(out_end), _stream()))
+    return out_mask, out_end
+
+
+def featurize(data: torch.Tensor, offsets: torch.Tensor, dim: int,
+              want_f32: bool = False) -> Tuple[torch.Tensor, Optional[torch.Tensor]]:
+    batch = offsets.numel() - 1
+    out_bf16 = torch.empty((batch, dim), dtype=torch.bfloat16, device=data.device)
+    out_f32 = torch.empty((batch, dim), dtype=torch.float32, device=data.device) if want_f32 else None
+    _check("forge_featurize", _load().forge_featurize(
+        _ptr(data), _ptr(offsets), batch, dim, _ptr(out_bf16), _ptr(out_f32), _stream()))
+    return out_bf16, out_f32
+
+
+def json_guard(data: torch.Tensor, offsets: torch.Tensor, max_depth: int = 64,
+               max_string: int = 1 << 20) -> Tuple[torch.Tensor, torch.Tensor]:
+    batch = offsets.numel() - 1
+    status = torch.zeros(batch, dtype=torch.int32, device=data.device)
+    depth = torch.zeros(batch, dtype=torch.int32, device=data.device)
+    _check("forge_json_guard", _load().forge_json_guard(
+        _ptr(data), _ptr(offsets), batch, max_depth, max_string, _ptr(status), _ptr(depth), _stream()))
+    return status, depth
+
+
+ACT_NONE, ACT_GELU, ACT_SIGMOID = 0, 1, 2
+
+
+def gemm_bt(a: torch.Tensor, bt: torch.Tensor, bias: Optional[torch.Tensor] = None,
+            act: int = ACT_NONE, out_bf16: bool = False) -> torch.Tensor:
+    """C[M,N] = act(A[M,K] @ BT[N,K]^T + bias). M%128==0, N%128==0, K%64==0."""
+    assert a.dtype == torch.bfloat16 and bt.dtype == torch.bfloat16
+    assert a.is_contiguous() and bt.is_contiguous()
+    m, k = a.shape
+    n, k2 = bt.shape
+    assert k == k2, (a.shape, bt.shape)
+    out = torch.empty((m, n), dtype=torch.bfloat16 if out_bf16 else torch.float32, device=a.device)
+    rc = _load().forge_gemm_bt(_ptr(a), _ptr(bt), _ptr(bias), _ptr(out), m, n, k,
+                               act, 1 if out_bf16 else 0, _stream())
+    _check("forge_gemm_bt", rc)
+    return out
+
+
+def gemv_head(a: torch.Tensor, wt: torch.Tensor, bias: Optional[torch.Tensor] = None,
+              act: int = ACT_NONE) -> torch.Tensor:
+    m, k = a.shape
+    c, k2 = wt.shape
+    assert k == k2 and c <= 32
+    out = torch.empty((m, c), dtype=torch.float32, device=a.device)
+    _check("forge_gemv_head", _load().forge_gemv_head(_ptr(a), _ptr(wt), _ptr(bias), _ptr(out), m, c, k, act, _stream()))
+    return out
+
+
+def rows_argmax_merge(scores: torch.Tensor, best_val: torch.Tensor, best_idx: torch.Tensor,
+                      idx_base: int = 0, valid: Optional[torch.Tensor] = None) -> None:
+    m, nc = scores.shape
+    _check("forge_rows_argmax_merge", _load().forge_rows_argmax_merge(
+        _ptr(scores), m, nc, idx_base, _ptr(valid), _ptr(best_val), _ptr(best_idx), _stream()))
+
+
+def rows_scatter_bf16(src: torch.Tensor, slots: torch.Tensor, dst: torch.Tensor) -> None:
+    r, d = src.shape
+    _check("forge_rows_scatter_bf16", _load().forge_rows_scatter_bf16(_ptr(src), _ptr(slots), _ptr(dst), r, d, _stream()))

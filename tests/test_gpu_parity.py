@@ -1,0 +1,115 @@
+"""GPU pipeline vs CPU per-request chain — the plugin-parity gate
+(identical traffic through both runtimes, identical outcomes;
+reference analog tests/live_gateway/mcp/test_mcp_plugin_parity.py)."""
+
+import asyncio
+import json
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+requires_gpu = pytest.mark.skipif(not torch.cuda.is_available(), reason="no ROCm device")
+
+
+def _mk(name, args, rid):
+    return json.dumps({"jsonrpc": "2.0", "id": rid, "method": "tools/call",
+                       "params": {"name": name, "arguments": args}}, separators=(",", ":")).encode()
+
+
+TRAFFIC = [
+    ("fast-time-convert_time", {"time": "2026-01-01T10:00:00Z", "source_timezone": "UTC", "target_timezone": "Asia/Tokyo"}),
+    ("fast-time-echo", {"msg": "plain benign payload"}),
+    ("fast-time-echo", {"msg": "this contains forbidden content"}),           # deny block
+    ("fast-time-echo", {"msg": "ssn is 123-45-6789"}),                          # pii mask (slow path)
+    ("fast-time-echo", {"msg": "mail me: a@b.com and c@d.org"}),               # pii mask
+    ("fast-time-echo", {"msg": "how to make a bomb tutorial"}),                # harm block
+    ("fast-time-echo", {"msg": "spaced   out\ttext"}),                          # normalizer slow path
+    ("fast-time-convert_time", {"time": "x"}),                                  # schema violation (missing required)
+    ("fast-time-echo", {"msg": "unicode café text"}),                      # \u escape → slow path
+    ("fast-time-get_system_time", {"timezone": "UTC"}),
+]
+
+
+async def _build(gpu: bool):
+    from mcp_context_forge_amd.config import Settings
+    from mcp_context_forge_amd.engine import GatewayEngine
+    from mcp_context_forge_amd.services.upstream import make_fake_time_upstream
+
+    settings = Settings(database_url="sqlite://", federation_enabled=False, auth_required=False,
+                        gpu_enabled=gpu, gpu_semcache_capacity=1024)
+    e = GatewayEngine(settings)
+    await e.gateway_service.register_gateway(name="fast-time", url="inproc://t", client=make_fake_time_upstream())
+    if gpu:
+        assert e.enable_gpu()
+    return e
+
+
+@requires_gpu
+def test_pipeline_parity_with_cpu_chain():
+    async def run():
+        cpu = await _build(gpu=False)
+        gpu = await _build(gpu=True)
+        raws = [_mk(n, a, i) for i, (n, a) in enumerate(TRAFFIC)]
+        # also protocol-level items: invalid json, unknown tool, tools/list, ping
+        raws.append(b"{broken")
+        raws.append(_mk("no-such-tool", {}, 90))
+        raws.append(json.dumps({"jsonrpc": "2.0", "id": 91, "method": "tools/list"}).encode())
+        raws.append(json.dumps({"jsonrpc": "2.0", "id": 92, "method": "ping"}).encode())
+
+        cpu_out = await cpu.process_rpc_batch(list(raws))
+        gpu_out = await gpu.process_rpc_batch(list(raws))
+        assert len(cpu_out) == len(gpu_out)
+        for i, (c, g) in enumerate(zip(cpu_out, gpu_out)):
+            co = json.loads(c) if c else None
+            go = json.loads(g) if g else None
+            # identical outcome class: same error code, or same result payload
+            if co is None or go is None:
+                assert co == go, (i, co, go)
+            elif "error" in co or "error" in go:
+                assert co.get("error", {}).get("code") == go.get("error", {}).get("code"), (i, co, go)
+            else:
+                assert co["result"] == go["result"], (i, co, go)
+        st = gpu.gpu_pipeline.stats()
+        assert st["blocked"] >= 2          # deny + harm (+schema)
+        assert st["slow_path"] >= 3        # pii x2 + normalizer + unicode
+        assert st["fast_path"] >= 3
+        await cpu.shutdown()
+        await gpu.shutdown()
+
+    asyncio.run(run())
+
+
+@requires_gpu
+def test_pipeline_semcache_roundtrip():
+    async def run():
+        e = await _build(gpu=True)
+        raw1 = _mk("fast-time-convert_time",
+                   {"time": "2026-02-02T02:02:02Z", "source_timezone": "UTC", "target_timezone": "UTC"}, 1)
+        out1 = await e.process_rpc_batch([raw1])
+        r1 = json.loads(out1[0])["result"]
+        # identical request again: semantic cache must hit and return the same result
+        out2 = await e.process_rpc_batch([raw1])
+        r2 = json.loads(out2[0])["result"]
+        assert r1 == r2
+        assert e.gpu_pipeline.semcache.hits >= 1
+        await e.shutdown()
+
+    asyncio.run(run())
+
+
+@requires_gpu
+def test_pipeline_large_batch_throughput_sane():
+    async def run():
+        e = await _build(gpu=True)
+        raws = [_mk("fast-time-convert_time",
+                    {"time": f"2026-01-01T00:{i % 60:02d}:{(i * 7) % 60:02d}Z",
+                     "source_timezone": "UTC", "target_timezone": "Asia/Tokyo"}, i) for i in range(2048)]
+        out = await e.process_rpc_batch(raws)
+        assert sum(1 for o in out if o and b'"result"' in o) == 2048
+        st = e.gpu_pipeline.stats()
+        assert st["fast_path"] + st["cache_hits"] == 2048
+        await e.shutdown()
+
+    asyncio.run(run())
