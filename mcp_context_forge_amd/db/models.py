@@ -274,6 +274,7 @@ class DbToolMetric(Base):
     response_time_ms: Mapped[float] = mapped_column(Float)
     is_success: Mapped[bool] = mapped_column(Boolean, default=True)
     error_message: Mapped[Optional[str]] = mapped_column(Text, nullable=True)
+    count: Mapped[int] = mapped_column(Integer, default=1)  # batched rows (revision 0002)
 
 
 class DbMetricRollup(Base):

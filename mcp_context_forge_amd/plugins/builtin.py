@@ -249,11 +249,16 @@ class ToonEncoderPlugin(Plugin):
         sc = result.get("structuredContent")
         if sc is None:
             return None
-        j, t, frac = toon_codec.savings(sc)
-        if j < self.min_size or frac < self.min_savings:
+        j = len(json.dumps(sc, separators=(",", ":")))
+        if j < self.min_size:  # cheap reject before paying for the TOON encode
+            return None
+        encoded = toon_codec.encode(sc)
+        t = len(encoded)
+        frac = 1.0 - t / j if j else 0.0
+        if frac < self.min_savings:
             return None
         new = dict(result)
-        new["content"] = [{"type": "text", "text": toon_codec.encode(sc)}]
+        new["content"] = [{"type": "text", "text": encoded}]
         new.setdefault("_meta", {})["toon"] = {"json_bytes": j, "toon_bytes": t, "savings": round(frac, 4)}
         return new
 
@@ -369,33 +374,37 @@ class ResponseCacheByPromptPlugin(Plugin):
         self.threshold = float(self.config.get("threshold", 0.92))
         self.capacity = int(self.config.get("capacity", 4096))
         self.ttl = float(self.config.get("ttl", 600.0))
-        self.vectors = np.zeros((0, self.dim), dtype=np.float32)
-        self.entries: List[Tuple[str, float, Any]] = []  # (tool, ts, result)
+        # preallocated ring (matrix never reallocates — mirrors the GPU cache)
+        self.vectors = np.zeros((self.capacity, self.dim), dtype=np.float32)
+        self.entries: List[Optional[Tuple[str, float, Any]]] = [None] * self.capacity
+        self.size = 0
+        self.write_ptr = 0
         self.hits = 0
         self.misses = 0
 
     def lookup(self, tool: str, text: str) -> Optional[Any]:
-        if not self.entries:
+        if self.size == 0:
             self.misses += 1
             return None
         v = featurize(text, self.dim)
-        sims = self.vectors @ v
+        sims = self.vectors[: self.size] @ v
         best = int(np.argmax(sims))
         now = time.monotonic()
-        ent_tool, ts, result = self.entries[best]
-        if sims[best] >= self.threshold and ent_tool == tool and now - ts <= self.ttl:
-            self.hits += 1
-            return result
+        ent = self.entries[best]
+        if ent is not None:
+            ent_tool, ts, result = ent
+            if sims[best] >= self.threshold and ent_tool == tool and now - ts <= self.ttl:
+                self.hits += 1
+                return result
         self.misses += 1
         return None
 
     def insert(self, tool: str, text: str, result: Any) -> None:
-        v = featurize(text, self.dim).reshape(1, -1)
-        if len(self.entries) >= self.capacity:
-            self.vectors = self.vectors[1:]
-            self.entries = self.entries[1:]
-        self.vectors = np.concatenate([self.vectors, v], axis=0) if self.vectors.size else v
-        self.entries.append((tool, time.monotonic(), result))
+        slot = self.write_ptr
+        self.vectors[slot] = featurize(text, self.dim)
+        self.entries[slot] = (tool, time.monotonic(), result)
+        self.write_ptr = (self.write_ptr + 1) % self.capacity
+        self.size = min(self.size + 1, self.capacity)
 
     async def tool_pre_invoke(self, ctx: PluginContext) -> PluginResult:
         hit = self.lookup(ctx.name, _text_of(ctx.args))

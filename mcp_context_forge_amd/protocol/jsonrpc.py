@@ -141,7 +141,8 @@ def parse_request_bytes(raw: bytes) -> JSONRPCRequest:
     reference's ordering at mcpgateway/main.py:11225-11270).
     """
     try:
-        obj = json.loads(raw)
+        # decode first: json.loads(bytes) pays a detect_encoding pass per call
+        obj = json.loads(raw.decode("utf-8") if isinstance(raw, (bytes, bytearray)) else raw)
     except Exception as exc:
         raise JSONRPCError(PARSE_ERROR, data=str(exc)) from exc
     return JSONRPCRequest.from_dict(obj)

@@ -47,7 +47,7 @@ class MetricsBuffer:
             self.flush()
 
     def record_batch(self, tool_ids: List[str], response_time_ms: float, successes: List[bool]) -> None:
-        """Batched record from the GPU pipeline: one lock, N rows."""
+        """Batched record: one lock, N rows."""
         with self._lock:
             for tid, ok in zip(tool_ids, successes):
                 self._rows.append({"tool_id": tid, "response_time_ms": response_time_ms, "is_success": ok, "error_message": None})
@@ -55,6 +55,23 @@ class MetricsBuffer:
                 self.latency_count[tid] += 1
             self.counters["tool_invocations_total"] += len(tool_ids)
             self.counters["tool_errors_total"] += sum(1 for s in successes if not s)
+            need_flush = len(self._rows) >= self.max_size
+        if need_flush:
+            self.flush()
+
+    def record_aggregate(self, tool_id: str, count: int, errors: int, response_time_ms: float) -> None:
+        """Aggregated record from the GPU pipeline: one row per (tool, batch)
+        carrying `count` (schema revision 0002) instead of N rows — the
+        batched analog of the reference's per-row metric buffering."""
+        if count <= 0:
+            return
+        with self._lock:
+            self._rows.append({"tool_id": tool_id, "response_time_ms": response_time_ms,
+                               "is_success": errors == 0, "error_message": None, "count": count})
+            self.counters["tool_invocations_total"] += count
+            self.counters["tool_errors_total"] += errors
+            self.latency_sum_ms[tool_id] += response_time_ms * count
+            self.latency_count[tool_id] += count
             need_flush = len(self._rows) >= self.max_size
         if need_flush:
             self.flush()
