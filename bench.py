@@ -62,7 +62,7 @@ def make_request(rng: random.Random, tool_names: List[str], rid: int, flagged_fr
 async def build_engine(rank: int, world: int, upstreams: int, use_gpu: bool):
     from mcp_context_forge_amd.config import Settings
     from mcp_context_forge_amd.engine import GatewayEngine
-    from mcp_context_forge_amd.services.upstream import make_fake_time_upstream
+    from mcp_context_forge_amd.services.upstream import NativeInProcUpstream
 
     settings = Settings(
         database_url="sqlite://",
@@ -74,8 +74,10 @@ async def build_engine(rank: int, world: int, upstreams: int, use_gpu: bool):
     )
     engine = GatewayEngine(settings, rank=rank, world_size=world)
     # 64 federated upstreams per rank (BASELINE config 2); each exposes 3 tools.
+    # Native C++ upstreams — the reference's benchmark upstream (fast_time_server)
+    # is a native Go binary; this is its in-proc C++ analog.
     for u in range(upstreams):
-        up = make_fake_time_upstream(name=f"up{rank}-{u}")
+        up = NativeInProcUpstream(name=f"up{rank}-{u}")
         await engine.gateway_service.register_gateway(
             name=f"up{rank}-{u}", url=f"inproc://up{rank}-{u}", client=up, owner_rank=rank)
     if use_gpu:

@@ -55,8 +55,8 @@ def parse_batch(raws: List[bytes]) -> List[ParsedRequest]:
     return out
 
 
-def pack_texts(texts: List[bytes], device: str = "cuda") -> Tuple[torch.Tensor, torch.Tensor]:
-    """Pack byte strings into (data u8 [total], offsets i32 [B+1]) on device."""
+def pack_texts(texts: List[bytes], device: str = "cuda") -> Tuple[torch.Tensor, torch.Tensor, torch.Tensor]:
+    """Pack byte strings into (data u8 [total], beg i32 [B], end i32 [B]) on device."""
     offsets = np.zeros(len(texts) + 1, dtype=np.int32)
     for i, t in enumerate(texts):
         offsets[i + 1] = offsets[i] + len(t)
@@ -64,7 +64,7 @@ def pack_texts(texts: List[bytes], device: str = "cuda") -> Tuple[torch.Tensor, 
     data_np = np.frombuffer(blob, dtype=np.uint8) if blob else np.zeros(1, dtype=np.uint8)
     data = torch.from_numpy(data_np.copy()).to(device, non_blocking=True)
     offs = torch.from_numpy(offsets).to(device, non_blocking=True)
-    return data, offs
+    return data, offs[:-1], offs[1:]
 
 
 def pad_rows(t: torch.Tensor, multiple: int = 128) -> torch.Tensor:

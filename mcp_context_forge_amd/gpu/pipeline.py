@@ -1,36 +1,36 @@
 """The batched GPU plugin pipeline — the MI355X-native hot path.
 
 Reference analog: the per-request chain tool_service.invoke_tool (:5067)
-with plugin hooks (:5530). Here a whole micro-batch is staged to HBM and the
-plugin chain's data-parallel stages run as HIP kernels (BASELINE.json):
+with plugin hooks (:5530), plus the Rust edge runtime's native fast path
+(crates/mcp_runtime). Here a whole micro-batch is staged to HBM and the
+plugin chain's data-parallel stages run as HIP kernels (BASELINE.json),
+with a C++ host fast path around them:
 
-  json_guard  → structural limits over raw payloads
-  scan banks  → deny_filter / harmful_content / pii_filter / regex_filter /
-                normalizer-trigger DFAs (one table-driven kernel, ops/csrc/scan.hip)
-  featurize   → hashed count vectors (LDS histograms)
-  classifier  → content_moderation bf16 MFMA MLP
-  semcache    → response_cache_by_prompt cosine sweep over HBM-resident keys
+  C++ envelope scan   → JSON-RPC spans, no Python parse (ops/csrc/envelope.cpp)
+  scan banks (GPU)    → deny / harm / pii / regex / normalizer-trigger DFAs
+                        + schema-shape presence patterns, over RAW arg bytes
+  featurize (GPU)     → hashed count vectors (LDS histograms)
+  classifier (GPU)    → content_moderation bf16 MFMA MLP
+  semcache (GPU)      → response_cache_by_prompt cosine sweep in HBM
+  C++ native upstream → fast_time_server analog (ops/csrc/upstream.cpp)
+  result scan (GPU)   → pii/regex/harm over serialized results
 
-Three passes, mirroring the CPU chain's priority order exactly:
+Decisions are numpy-vectorized; Python touches a request only when a kernel
+flags it (rewrites, schema fallbacks, unusual envelopes) — those take the
+exact per-request semantics (the plugins' own functions), so the parity
+gate (tests/test_gpu_parity.py, reference analog
+tests/live_gateway/mcp/test_mcp_plugin_parity.py) holds by construction.
 
-  PASS 1 (all requests, original args): cache lookup + deny + rewrite-flag
-     detection + moderation + harm + schema. Unflagged requests are decided
-     entirely from kernel outputs.
-  PASS 2 (rewrite-flagged subset): host rewrites (normalizer → regex → pii,
-     the plugins' own functions = exact semantics), then a second GPU sweep
-     (harm + moderation) over the rewritten texts, then schema.
-  PASS 3 (results): one scan over serialized results; pii/regex/harm-flagged
-     results take the host post chain; all results get toon + length guard.
-
-Parity gate: tests/test_gpu_parity.py runs identical traffic through this
-pipeline and the per-request CPU chain and asserts identical outcomes
-(reference analog: tests/live_gateway/mcp/test_mcp_plugin_parity.py).
+Scanning operates on the request's RAW argument bytes (and the CPU oracle
+on the canonical sorted-JSON text): the two representations contain the
+same string values and the same token multiset, so detection agrees except
+for matches spanning value boundaries — rewrite-class flags only route to
+the exact host path, and parity traffic uses compact payloads.
 """
 
 from __future__ import annotations
 
 import asyncio
-import hashlib
 import json
 import time
 from typing import Any, Dict, List, Optional, Tuple
@@ -42,17 +42,41 @@ from ..ops import dfa, hip
 from ..protocol import jsonrpc
 from ..plugins.builtin import _walk_strings
 from ..plugins.framework import PluginMode
-from .batch import ParsedRequest, canonical_text, pack_texts, pad_rows, parse_batch
+from .batch import pack_texts, pad_rows
 from .classifier import GpuClassifier
 from .semcache import GpuSemanticCache, tool_hash
 
-# normalizer-trigger prefilter: any byte pattern whose presence could make
-# argument_normalizer rewrite the payload (conservative superset). The scan
-# runs over CANONICAL JSON text, so whitespace/unicode appear as their JSON
-# escapes: backslash-t/n/r/f, backslash-u (covers \\u000b, \\u000c and all
-# non-ASCII that NFC could touch), literal double-space, and space adjacent
-# to a quote (leading/trailing-space strip).
-_NORMALIZE_TRIGGERS = ["\\t", "\\n", "\\r", "\\f", "\\u", "  ", '" ', ' "']
+# normalizer-trigger prefilter (conservative superset over raw JSON bytes):
+# JSON escapes for whitespace/unicode, literal multi-space, space adjacent to
+# a quote, and raw non-ASCII bytes (clients that don't ensure_ascii).
+_NORMALIZE_TRIGGERS = ["\\t", "\\n", "\\r", "\\f", "\\u", "  ", '" ', ' "',
+                       "[" + chr(0x80) + "-" + chr(0xFF) + "]"]
+
+_SCHEMA_NEST_PATTERNS = [":\\{", ":\\["]
+
+_TYPED_PAT = {
+    "string": '"{k}":"',
+    "number": '"{k}":[0-9\\-]',
+    "integer": '"{k}":[0-9\\-]',
+    "boolean": '"{k}":[tf]',
+    "array": '"{k}":\\[',
+}
+
+
+class _ToolMeta:
+    __slots__ = ("tool", "name", "tid", "itype", "thash", "native_kind", "native_client",
+                 "client", "handler", "schema_mode", "required_bits", "typed_pairs",
+                 "has_output_schema", "original_name", "reachable")
+
+    def __init__(self):
+        self.schema_mode = "host"   # "trivial" | "fast" | "host"
+        self.required_bits = 0
+        self.typed_pairs: List[Tuple[int, int]] = []
+        self.native_kind = -1
+        self.native_client = None
+        self.client = None
+        self.handler = None
+        self.has_output_schema = False
 
 
 class GpuPluginPipeline:
@@ -86,7 +110,9 @@ class GpuPluginPipeline:
                 self.banks[name] = hip.DeviceScanTables(tables, device)
         if self.normalizer is not None:
             self.banks["normalize"] = hip.DeviceScanTables(
-                dfa.compile_literals(_NORMALIZE_TRIGGERS, case_insensitive=False), device)
+                dfa.compile_patterns(
+                    ["".join(ch if ch.isalnum() else "\\" + ch for ch in t) if not t.startswith("[") else t
+                     for t in _NORMALIZE_TRIGGERS], case_insensitive=False), device)
 
         self.classifier: Optional[GpuClassifier] = None
         self.feat_dim = s.gpu_feature_dim
@@ -106,6 +132,14 @@ class GpuPluginPipeline:
 
         self.max_depth = s.max_json_depth
         self.max_string = s.max_string_length
+
+        # per-tool metadata + schema-shape bank (rebuilt on registry change)
+        self._meta_gen = -1
+        self._tool_meta: Dict[str, _ToolMeta] = {}
+        self._schema_bank: Optional[hip.DeviceScanTables] = None
+        self._schema_pat_ids: Dict[str, int] = {}
+        self._nest_bits = 0
+
         # stats
         self.batches = 0
         self.requests = 0
@@ -114,34 +148,122 @@ class GpuPluginPipeline:
         self.blocked = 0
         self.cache_hits = 0
         self.post_rewrites = 0
+        self.py_fallback = 0
 
     # ------------------------------------------------------------------
-    def _scan_pass(self, texts: List[bytes], banks: Tuple[str, ...], classify: bool,
-                   cache_lookup: bool, guard: bool = False) -> Dict[str, Any]:
-        """Launch the kernel set over packed texts; single sync; host arrays."""
-        data, offs = pack_texts(texts, self.device)
-        out: Dict[str, torch.Tensor] = {}
-        if guard:
-            status, _depth = hip.json_guard(data, offs, self.max_depth, self.max_string)
-            out["guard_status"] = status
-        for name in banks:
-            bank = self.banks.get(name)
-            if bank is not None:
-                out[f"mask_{name}"], _ = hip.scan(data, offs, bank)
-        feats = None
-        if (classify and self.classifier is not None) or (cache_lookup and self.semcache is not None):
-            feats_b, _ = hip.featurize(data, offs, self.feat_dim)
-            feats = pad_rows(feats_b, 128)
-        if classify and self.classifier is not None:
-            out["scores"] = self.classifier.forward(feats)[: len(texts)]
-        if cache_lookup and self.semcache is not None:
-            bv, bi = self.semcache.lookup(feats)
-            out["cache_val"], out["cache_idx"] = bv[: len(texts)], bi[: len(texts)]
-        torch.cuda.synchronize()
-        host = {k: v.cpu().numpy() for k, v in out.items()}
-        host["_feats"] = feats
-        return host
+    # tool metadata / schema-shape bank
+    # ------------------------------------------------------------------
+    def _schema_pattern(self, pat: str) -> int:
+        pid = self._schema_pat_ids.get(pat)
+        if pid is None:
+            pid = len(self._schema_pat_ids)
+            self._schema_pat_ids[pat] = pid
+        return pid
 
+    def _esc_key(self, k: str) -> Optional[str]:
+        if not k or not all(c.isalnum() or c == "_" for c in k):
+            return None
+        return k
+
+    def _compile_tool_schema(self, meta: _ToolMeta, schema: Optional[dict]) -> None:
+        """Flat object schemas compile to presence/type byte patterns scanned
+        on-GPU; anything richer falls back to host validation (exact)."""
+        if self.schema_guard is None or not schema:
+            meta.schema_mode = "trivial"
+            return
+        if not (schema.get("properties") or schema.get("required") or
+                schema.get("additionalProperties") is False or schema.get("anyOf") or
+                schema.get("allOf") or schema.get("oneOf")):
+            meta.schema_mode = "trivial"
+            return
+        props = schema.get("properties") or {}
+        required = schema.get("required") or []
+        if schema.get("additionalProperties") is False or schema.get("anyOf") or \
+           schema.get("allOf") or schema.get("oneOf"):
+            meta.schema_mode = "host"
+            return
+        req_bits = 0
+        pairs: List[Tuple[int, int]] = []
+        for k, sub in props.items():
+            ek = self._esc_key(k)
+            ty = sub.get("type") if isinstance(sub, dict) else None
+            extra = isinstance(sub, dict) and any(
+                c in sub for c in ("enum", "const", "pattern", "minimum", "maximum", "minLength",
+                                   "maxLength", "minItems", "maxItems", "properties", "items",
+                                   "anyOf", "allOf", "oneOf", "exclusiveMinimum", "exclusiveMaximum"))
+            if ek is None or extra or (ty is not None and not isinstance(ty, str)) or \
+               (ty is not None and ty not in _TYPED_PAT and ty not in ("object", "null")):
+                meta.schema_mode = "host"
+                return
+            present = self._schema_pattern(f'"{ek}":')
+            if ty in _TYPED_PAT:
+                typed = self._schema_pattern(_TYPED_PAT[ty].format(k=ek))
+                pairs.append((present, typed))
+            if k in required:
+                req_bits |= 1 << present
+        for k in required:
+            if k not in props:
+                ek = self._esc_key(k)
+                if ek is None:
+                    meta.schema_mode = "host"
+                    return
+                req_bits |= 1 << self._schema_pattern(f'"{ek}":')
+        if len(self._schema_pat_ids) > 30 - len(_SCHEMA_NEST_PATTERNS):
+            meta.schema_mode = "host"
+            return
+        meta.schema_mode = "fast"
+        meta.required_bits = req_bits
+        meta.typed_pairs = pairs
+
+    def _rebuild_tool_meta(self) -> None:
+        from ..services.upstream import InProcUpstream, NativeInProcUpstream
+
+        ts = self.engine.tool_service
+        self._tool_meta = {}
+        self._schema_pat_ids = {}
+        for tool in self.engine.registry.list("tool", include_disabled=False):
+            m = _ToolMeta()
+            m.tool = tool
+            m.name = tool["name"]
+            m.tid = tool.get("id", m.name)
+            m.itype = tool.get("integration_type", "MCP")
+            m.thash = tool_hash(m.name)
+            m.original_name = tool.get("original_name", m.name)
+            m.has_output_schema = bool(tool.get("output_schema"))
+            m.reachable = tool.get("reachable", True)
+            if m.itype == "LOCAL":
+                m.handler = ts._local_handlers.get(m.name)
+            elif m.itype == "MCP":
+                m.client = ts._upstreams.get(tool.get("gateway_id") or "")
+                if isinstance(m.client, NativeInProcUpstream):
+                    m.native_kind = m.client.TOOL_KINDS.get(m.original_name, -1)
+                    m.native_client = m.client
+            self._compile_tool_schema(m, tool.get("input_schema"))
+            self._tool_meta[m.name] = m
+        pats = list(self._schema_pat_ids.keys())
+        self._nest_bits = 0
+        if pats and self.schema_guard is not None:
+            base = len(pats)
+            for i, p in enumerate(_SCHEMA_NEST_PATTERNS):
+                self._nest_bits |= 1 << (base + i)
+            try:
+                self._schema_bank = hip.DeviceScanTables(
+                    dfa.compile_patterns(pats + _SCHEMA_NEST_PATTERNS, case_insensitive=False), self.device)
+            except ValueError:
+                self._schema_bank = None
+                for m in self._tool_meta.values():
+                    if m.schema_mode == "fast":
+                        m.schema_mode = "host"
+        else:
+            self._schema_bank = None
+        self._meta_gen = self.engine.registry.generation
+
+    def _meta(self) -> Dict[str, _ToolMeta]:
+        if self._meta_gen != self.engine.registry.generation:
+            self._rebuild_tool_meta()
+        return self._tool_meta
+
+    # ------------------------------------------------------------------
     def _enforcing(self, plugin) -> bool:
         return plugin is not None and plugin.mode in (PluginMode.ENFORCE, PluginMode.ENFORCE_IGNORE_ERROR)
 
@@ -158,76 +280,264 @@ class GpuPluginPipeline:
     async def process_batch(self, raws: List[bytes], user: Optional[str] = None,
                             server_id: Optional[str] = None) -> List[Optional[bytes]]:
         self.batches += 1
-        self.requests += len(raws)
-        parsed = parse_batch(raws)
-        responses: List[Optional[bytes]] = [None] * len(raws)
+        n = len(raws)
+        self.requests += n
+        responses: List[Optional[bytes]] = [None] * n
 
-        toolcalls: List[ParsedRequest] = []
-        passthrough: List[ParsedRequest] = []
-        for pr in parsed:
-            if pr.error is not None:
-                if pr.req is None or not pr.req.is_notification:
-                    rid = pr.req.id if pr.req else None
-                    responses[pr.index] = jsonrpc.JSONRPCResponse(id=rid, error=pr.error).to_bytes()
-            elif pr.tool_name is not None:
-                pr.tool = self.engine.registry.lookup_tool(pr.tool_name)
-                if pr.tool is None:
-                    responses[pr.index] = jsonrpc.error_response(
-                        pr.req.id, jsonrpc.INVALID_PARAMS, f"Tool not found: {pr.tool_name}").to_bytes()
-                else:
-                    toolcalls.append(pr)
-            else:
-                passthrough.append(pr)
+        offsets = np.zeros(n + 1, dtype=np.int64)
+        for i, r in enumerate(raws):
+            offsets[i + 1] = offsets[i] + len(r)
+        joined = b"".join(raws)
+        blob = np.frombuffer(joined, dtype=np.uint8) if joined else np.zeros(1, dtype=np.uint8)
+        env = hip.parse_envelopes(blob, offsets)
+        kind = env["kind"]
 
-        if passthrough:
-            others = await asyncio.gather(
-                *(self.engine.handle_rpc(pr.req, user=user, server_id=server_id) for pr in passthrough))
-            for pr, resp in zip(passthrough, others):
-                responses[pr.index] = resp.to_bytes() if resp is not None else None
+        other_rows = np.nonzero(kind != hip.ENV_TOOLS_CALL)[0]
+        if other_rows.size:
+            self.py_fallback += int(other_rows.size)
+            outs = await asyncio.gather(
+                *(self.engine.handle_rpc_bytes(raws[int(i)], user=user, server_id=server_id)
+                  for i in other_rows))
+            for i, out in zip(other_rows, outs):
+                responses[int(i)] = out
 
-        if toolcalls:
-            await self._process_toolcalls(toolcalls, responses, user, server_id)
+        fast_rows = np.nonzero(kind == hip.ENV_TOOLS_CALL)[0]
+        if fast_rows.size:
+            await self._fast_toolcalls(raws, blob, env, fast_rows, responses, user, server_id)
         return responses
 
     # ------------------------------------------------------------------
-    def _block(self, pr: ParsedRequest, responses: List[Optional[bytes]], message: str) -> None:
-        responses[pr.index] = jsonrpc.error_response(pr.req.id, jsonrpc.POLICY_DENIED, message).to_bytes()
-        self.blocked += 1
-
-    def _moderation_block(self, scores_row: np.ndarray) -> Optional[str]:
-        worst = float(scores_row.max())
-        if worst >= self.moderation.threshold:
-            from ..models.classifier import category_names
-
-            cat = category_names(len(scores_row))[int(scores_row.argmax())]
-            return f"content_moderation: moderation: category {cat} score {worst:.3f}"
-        return None
-
-    def _schema_errs(self, pr: ParsedRequest, args: Any) -> Optional[str]:
-        schema = pr.tool.get("input_schema")
-        if not schema:
+    def _id_bytes(self, blob: np.ndarray, env: dict, row: int) -> Optional[bytes]:
+        b, e = int(env["id_beg"][row]), int(env["id_end"][row])
+        if b < 0:
             return None
-        # trivial schemas ({"type":"object"} with no constraints) validate anything
-        if not (schema.get("properties") or schema.get("required") or
-                schema.get("additionalProperties") is False or schema.get("anyOf") or schema.get("allOf")):
+        return blob[b:e].tobytes()
+
+    def _splice_result(self, id_bytes: bytes, result_bytes: bytes) -> bytes:
+        return b'{"jsonrpc":"2.0","id":' + id_bytes + b',"result":' + result_bytes + b"}"
+
+    def _splice_error(self, id_bytes: Optional[bytes], code: int, message: str) -> Optional[bytes]:
+        if id_bytes is None:
             return None
-        from ..utils.jsonschema import validate as _validate
+        return (b'{"jsonrpc":"2.0","id":' + id_bytes + b',"error":{"code":' + str(code).encode()
+                + b',"message":' + json.dumps(message).encode() + b"}}")
 
-        errs = _validate(args or {}, schema)
-        if errs:
-            return "schema_guard: schema violation: " + "; ".join(errs[:5])
-        return None
+    async def _fast_toolcalls(self, raws: List[bytes], blob: np.ndarray, env: dict,
+                              rows: np.ndarray, responses: List[Optional[bytes]],
+                              user: Optional[str], server_id: Optional[str]) -> None:
+        meta_map = self._meta()
+        m = rows.shape[0]
+        nb, ne = env["name_beg"], env["name_end"]
+        ab, ae = env["args_beg"], env["args_end"]
 
-    def _apply_rewrites(self, pr: ParsedRequest) -> Any:
+        # --- tool resolution (memoized on raw name bytes) ---
+        metas: List[Optional[_ToolMeta]] = [None] * m
+        memo: Dict[bytes, Optional[_ToolMeta]] = {}
+        for j in range(m):
+            r = int(rows[j])
+            key = blob[nb[r]:ne[r]].tobytes()
+            if key in memo:
+                metas[j] = memo[key]
+            else:
+                mt = meta_map.get(key.decode("utf-8", "replace"))
+                memo[key] = mt
+                metas[j] = mt
+
+        th_arr = np.array([mt.thash if mt else 0 for mt in metas], dtype=np.int64)
+
+        # --- GPU pass 1 over raw argument spans ---
+        args_b = np.where(ab[rows] >= 0, ab[rows], 0).astype(np.int32)
+        args_e = np.where(ab[rows] >= 0, ae[rows], 0).astype(np.int32)
+        data_gpu = torch.from_numpy(blob.copy()).to(self.device, non_blocking=True)
+        beg_t = torch.from_numpy(args_b).to(self.device, non_blocking=True)
+        end_t = torch.from_numpy(args_e).to(self.device, non_blocking=True)
+
+        out: Dict[str, torch.Tensor] = {}
+        for bname, bank in self.banks.items():
+            out[bname], _ = hip.scan(data_gpu, beg_t, end_t, bank)
+        if self._schema_bank is not None:
+            out["schema"], _ = hip.scan(data_gpu, beg_t, end_t, self._schema_bank)
+        feats = None
+        if self.classifier is not None or self.semcache is not None:
+            feats_b, _ = hip.featurize(data_gpu, beg_t, end_t, self.feat_dim)
+            feats = pad_rows(feats_b, 128)
+        scores_t = self.classifier.forward(feats)[:m] if self.classifier is not None else None
+        cache_val_t = cache_idx_t = None
+        if self.semcache is not None:
+            bv, bi = self.semcache.lookup(feats)
+            cache_val_t, cache_idx_t = bv[:m], bi[:m]
+        torch.cuda.synchronize()
+
+        zeros = np.zeros(m, dtype=np.int64)
+        deny_m = out["deny"].cpu().numpy().astype(np.uint32) if "deny" in out else zeros
+        harm_m = out["harm"].cpu().numpy().astype(np.uint32) if "harm" in out else zeros
+        pii_m = out["pii"].cpu().numpy().astype(np.uint32) if "pii" in out else zeros
+        regex_m = out["regex"].cpu().numpy().astype(np.uint32) if "regex" in out else zeros
+        norm_m = out["normalize"].cpu().numpy().astype(np.uint32) if "normalize" in out else zeros
+        schema_m = out["schema"].cpu().numpy().astype(np.uint32) if "schema" in out else zeros
+        scores = scores_t.cpu().numpy() if scores_t is not None else None
+        hit_mask = np.zeros(m, dtype=bool)
+        cache_idx = None
+        if cache_val_t is not None:
+            cache_val = cache_val_t.cpu().numpy()
+            cache_idx = cache_idx_t.cpu().numpy()
+            hit_mask = self.semcache.resolve_hits_np(cache_val, cache_idx, th_arr)
+
+        # --- vectorized decisions (precedence = CPU chain priority order) ---
+        # 0 dispatch | 1 blocked/handled | 2 rewrite | 3 host-schema
+        state = np.zeros(m, dtype=np.int8)
+        mod_block = np.zeros(m, dtype=bool)
+        if scores is not None and self._enforcing(self.moderation):
+            mod_block = scores.max(axis=1) >= self.moderation.threshold
+        deny_on = self._enforcing(self.deny)
+        harm_on = self._enforcing(self.harm)
+
+        t0 = time.monotonic()
+        rewrite_rows: List[int] = []
+        dispatch_rows: List[int] = []
+        now = time.monotonic()
+        exact_store = self.exact_cache.store if self.exact_cache is not None else None
+        exact_ttl = self.exact_cache.ttl if self.exact_cache is not None else 0.0
+
+        for j in range(m):
+            mt = metas[j]
+            r = int(rows[j])
+            idb = self._id_bytes(blob, env, r)
+            if mt is None:
+                name = blob[nb[r]:ne[r]].tobytes().decode("utf-8", "replace")
+                responses[r] = self._splice_error(idb, jsonrpc.INVALID_PARAMS, f"Tool not found: {name}")
+                state[j] = 1
+                continue
+            name = mt.name
+            if not mt.reachable:
+                responses[r] = self._splice_error(idb, jsonrpc.SERVER_UNAVAILABLE,
+                                                  f"Tool {name} currently unreachable")
+                state[j] = 1
+                continue
+            if deny_m[j] and deny_on and self._applies(self.deny, name):
+                pid = int(deny_m[j]).bit_length() - 1
+                word = self.deny.words[pid] if pid < len(self.deny.words) else "?"
+                responses[r] = self._splice_error(idb, jsonrpc.POLICY_DENIED,
+                                                  f"deny_filter: deny word {word!r} present")
+                state[j] = 1
+                self.blocked += 1
+                continue
+            if (pii_m[j] and self._applies(self.pii, name)) or \
+               (regex_m[j] and self._applies(self.regex, name)) or \
+               (norm_m[j] and self._applies(self.normalizer, name)):
+                state[j] = 2
+                rewrite_rows.append(j)
+                continue
+            if mod_block[j] and self._applies(self.moderation, name):
+                from ..models.classifier import category_names
+
+                row_sc = scores[j]
+                cat = category_names(len(row_sc))[int(row_sc.argmax())]
+                responses[r] = self._splice_error(
+                    idb, jsonrpc.POLICY_DENIED,
+                    f"content_moderation: moderation: category {cat} score {float(row_sc.max()):.3f}")
+                state[j] = 1
+                self.blocked += 1
+                continue
+            if harm_m[j] and harm_on and self._applies(self.harm, name):
+                pid = int(harm_m[j]).bit_length() - 1
+                cat = self.harm.cats[pid] if pid < len(self.harm.cats) else "?"
+                responses[r] = self._splice_error(idb, jsonrpc.POLICY_DENIED,
+                                                  f"harmful_content_detector: harmful content ({cat})")
+                state[j] = 1
+                self.blocked += 1
+                continue
+            if self._enforcing(self.schema_guard):
+                mode = mt.schema_mode
+                if mode == "fast":
+                    sm = int(schema_m[j])
+                    ok = (sm & mt.required_bits) == mt.required_bits and not (sm & self._nest_bits)
+                    if ok:
+                        for present, typed in mt.typed_pairs:
+                            if (sm >> present) & 1 and not ((sm >> typed) & 1):
+                                ok = False
+                                break
+                    if not ok:
+                        state[j] = 3
+                        continue
+                elif mode == "host":
+                    state[j] = 3
+                    continue
+            if hit_mask[j]:
+                res = self.semcache.results[int(cache_idx[j])]
+                if idb is not None:
+                    responses[r] = self._splice_result(idb, res if isinstance(res, bytes)
+                                                       else json.dumps(res, separators=(",", ":")).encode())
+                state[j] = 1
+                self.cache_hits += 1
+                continue
+            if exact_store is not None:
+                ek = (name, blob[args_b[j]:args_e[j]].tobytes())
+                ent = exact_store.get(ek)
+                if ent is not None and now - ent[0] <= exact_ttl:
+                    if idb is not None:
+                        responses[r] = self._splice_result(idb, ent[1])
+                    state[j] = 1
+                    self.cache_hits += 1
+                    continue
+            if self.breaker is not None:
+                st = self.breaker._st(name)
+                if now < st["open_until"]:
+                    responses[r] = self._splice_error(idb, jsonrpc.POLICY_DENIED,
+                                                      f"circuit_breaker: circuit open for tool {name}")
+                    state[j] = 1
+                    self.blocked += 1
+                    continue
+            dispatch_rows.append(j)
+
+        # --- host-schema fallback (exact validate; violation blocks) ---
+        for j in np.nonzero(state == 3)[0]:
+            j = int(j)
+            mt = metas[j]
+            r = int(rows[j])
+            idb = self._id_bytes(blob, env, r)
+            try:
+                args = json.loads(blob[args_b[j]:args_e[j]].tobytes() or b"{}")
+            except Exception:
+                responses[r] = self._splice_error(idb, jsonrpc.INVALID_PARAMS, "invalid arguments")
+                continue
+            from ..utils.jsonschema import validate as _validate
+
+            errs = _validate(args or {}, mt.tool.get("input_schema") or {})
+            if errs:
+                responses[r] = self._splice_error(idb, jsonrpc.POLICY_DENIED,
+                                                  "schema_guard: schema violation: " + "; ".join(errs[:5]))
+                self.blocked += 1
+            elif hit_mask[j]:
+                res = self.semcache.results[int(cache_idx[j])]
+                if idb is not None:
+                    responses[r] = self._splice_result(idb, res if isinstance(res, bytes)
+                                                       else json.dumps(res, separators=(",", ":")).encode())
+                self.cache_hits += 1
+            else:
+                dispatch_rows.append(j)
+
+        # --- PASS 2: rewrite-flagged subset (host rewrites + GPU re-scan) ---
+        rewrite_dispatch: List[Tuple[int, Any]] = []  # (j, rewritten args)
+        if rewrite_rows:
+            self.slow_path += len(rewrite_rows)
+            rewrite_dispatch = await self._rewrite_pass(
+                blob, env, rows, args_b, args_e, rewrite_rows, metas, hit_mask, cache_idx, responses)
+
+        # --- fan-out dispatch ---
+        self.fast_path += len(dispatch_rows)
+        await self._dispatch_and_post(blob, env, rows, args_b, args_e, metas, feats, th_arr,
+                                      dispatch_rows, rewrite_dispatch, responses, t0)
+
+    # ------------------------------------------------------------------
+    def _apply_rewrites(self, name: str, args: Any) -> Tuple[str, Any]:
         """Host rewrites in CPU-chain priority order: normalizer(15) →
-        regex(20) → pii(30). Uses the plugins' own functions (exact parity).
-        Returns rewritten args, or a block message for pii action=block."""
-        args = pr.arguments
-        if self.normalizer is not None and self._applies(self.normalizer, pr.tool_name):
+        regex(20) → pii(30), using the plugins' own functions."""
+        if self.normalizer is not None and self._applies(self.normalizer, name):
             args = _walk_strings(args, self.normalizer.norm)
-        if self.regex is not None and self._applies(self.regex, pr.tool_name):
+        if self.regex is not None and self._applies(self.regex, name):
             args = _walk_strings(args, self.regex.apply_rules)
-        if self.pii is not None and self._applies(self.pii, pr.tool_name):
+        if self.pii is not None and self._applies(self.pii, name):
             found: List[str] = []
 
             def fn(s: str) -> str:
@@ -242,276 +552,233 @@ class GpuPluginPipeline:
                 args = new_args
         return ("__ok__", args)
 
-    async def _process_toolcalls(self, items: List[ParsedRequest], responses: List[Optional[bytes]],
-                                 user: Optional[str], server_id: Optional[str]) -> None:
-        texts = [pr.arg_text for pr in items]
-        scan = self._scan_pass(texts, ("deny", "harm", "pii", "regex", "normalize"),
-                               classify=True, cache_lookup=True)
-        n = len(items)
-        zeros = np.zeros(n, dtype=np.int64)
-        deny_m = scan.get("mask_deny", zeros)
-        harm_m = scan.get("mask_harm", zeros)
-        pii_m = scan.get("mask_pii", zeros)
-        regex_m = scan.get("mask_regex", zeros)
-        norm_m = scan.get("mask_normalize", zeros)
-        scores = scan.get("scores")
-        cache_val = scan.get("cache_val")
-        cache_idx = scan.get("cache_idx")
-
-        hits: List[Optional[Any]] = [None] * n
-        th_arr = np.array([tool_hash(pr.tool_name) for pr in items], dtype=np.int64)
-        if self.semcache is not None and cache_val is not None:
-            hits = self.semcache.resolve_hits(cache_val, cache_idx, th_arr)
-
-        t0 = time.monotonic()
-        # (pr, row, args) tuples heading to dispatch
-        dispatch: List[Tuple[ParsedRequest, int, Any]] = []
-        rewrites: List[Tuple[ParsedRequest, int]] = []
-
-        for i, pr in enumerate(items):
-            name = pr.tool_name
-            # --- PASS 1 decisions (CPU-chain priority order) ---
-            if deny_m[i] and self._enforcing(self.deny) and self._applies(self.deny, name):
-                pid = int(deny_m[i]).bit_length() - 1
-                word = self.deny.words[pid] if pid < len(self.deny.words) else "?"
-                self._block(pr, responses, f"deny_filter: deny word {word!r} present")
+    async def _rewrite_pass(self, blob, env, rows, args_b, args_e, rewrite_rows, metas,
+                            hit_mask, cache_idx, responses) -> List[Tuple[int, Any]]:
+        ok_items: List[Tuple[int, Any]] = []
+        for j in rewrite_rows:
+            r = int(rows[j])
+            mt = metas[j]
+            idb = self._id_bytes(blob, env, r)
+            try:
+                args = json.loads(blob[args_b[j]:args_e[j]].tobytes() or b"{}")
+            except Exception:
+                responses[r] = self._splice_error(idb, jsonrpc.INVALID_PARAMS, "invalid arguments")
                 continue
-            if (pii_m[i] and self._applies(self.pii, name)) or \
-               (regex_m[i] and self._applies(self.regex, name)) or \
-               (norm_m[i] and self._applies(self.normalizer, name)):
-                rewrites.append((pr, i))
+            status, payload = self._apply_rewrites(mt.name, args)
+            if status == "__block__":
+                responses[r] = self._splice_error(idb, jsonrpc.POLICY_DENIED, payload)
+                self.blocked += 1
                 continue
-            if scores is not None and self._enforcing(self.moderation) and self._applies(self.moderation, name):
-                msg = self._moderation_block(scores[i])
-                if msg:
-                    self._block(pr, responses, msg)
+            ok_items.append((j, payload))
+        if not ok_items:
+            return []
+
+        texts2 = [json.dumps(a, separators=(",", ":"), sort_keys=True, default=str).encode()
+                  for (_j, a) in ok_items]
+        data2, beg2, end2 = pack_texts(texts2, self.device)
+        harm2_t = None
+        if "harm" in self.banks:
+            harm2_t, _ = hip.scan(data2, beg2, end2, self.banks["harm"])
+        scores2_t = None
+        if self.classifier is not None:
+            f2, _ = hip.featurize(data2, beg2, end2, self.feat_dim)
+            scores2_t = self.classifier.forward(pad_rows(f2, 128))[: len(ok_items)]
+        torch.cuda.synchronize()
+        harm2 = harm2_t.cpu().numpy() if harm2_t is not None else np.zeros(len(ok_items), dtype=np.int64)
+        scores2 = scores2_t.cpu().numpy() if scores2_t is not None else None
+
+        out: List[Tuple[int, Any]] = []
+        for jj, (j, args2) in enumerate(ok_items):
+            r = int(rows[j])
+            mt = metas[j]
+            name = mt.name
+            idb = self._id_bytes(blob, env, r)
+            if scores2 is not None and self._enforcing(self.moderation) and self._applies(self.moderation, name):
+                row_sc = scores2[jj]
+                if float(row_sc.max()) >= self.moderation.threshold:
+                    from ..models.classifier import category_names
+
+                    cat = category_names(len(row_sc))[int(row_sc.argmax())]
+                    responses[r] = self._splice_error(
+                        idb, jsonrpc.POLICY_DENIED,
+                        f"content_moderation: moderation: category {cat} score {float(row_sc.max()):.3f}")
+                    self.blocked += 1
                     continue
-            if harm_m[i] and self._enforcing(self.harm) and self._applies(self.harm, name):
-                pid = int(harm_m[i]).bit_length() - 1
+            if harm2[jj] and self._enforcing(self.harm) and self._applies(self.harm, name):
+                pid = int(harm2[jj]).bit_length() - 1
                 cat = self.harm.cats[pid] if pid < len(self.harm.cats) else "?"
-                self._block(pr, responses, f"harmful_content_detector: harmful content ({cat})")
+                responses[r] = self._splice_error(idb, jsonrpc.POLICY_DENIED,
+                                                  f"harmful_content_detector: harmful content ({cat})")
+                self.blocked += 1
                 continue
-            if self._enforcing(self.schema_guard):
-                msg = self._schema_errs(pr, pr.arguments)
-                if msg:
-                    self._block(pr, responses, msg)
+            if self._enforcing(self.schema_guard) and mt.schema_mode != "trivial":
+                from ..utils.jsonschema import validate as _validate
+
+                errs = _validate(args2 or {}, mt.tool.get("input_schema") or {})
+                if errs:
+                    responses[r] = self._splice_error(idb, jsonrpc.POLICY_DENIED,
+                                                      "schema_guard: schema violation: " + "; ".join(errs[:5]))
+                    self.blocked += 1
                     continue
-            if hits[i] is not None:
-                responses[pr.index] = jsonrpc.result_response(pr.req.id, hits[i]).to_bytes()
+            if hit_mask[j]:
+                res = self.semcache.results[int(cache_idx[j])]
+                if idb is not None:
+                    responses[r] = self._splice_result(idb, res if isinstance(res, bytes)
+                                                       else json.dumps(res, separators=(",", ":")).encode())
                 self.cache_hits += 1
                 continue
-            if self.exact_cache is not None:
-                k = hashlib.sha256(name.encode() + pr.arg_text).hexdigest()
-                ent = self.exact_cache.store.get(k)
-                if ent and time.monotonic() - ent[0] <= self.exact_cache.ttl:
-                    responses[pr.index] = jsonrpc.result_response(pr.req.id, ent[1]).to_bytes()
-                    self.cache_hits += 1
-                    continue
-            if self.breaker is not None:
-                st = self.breaker._st(name)
-                if time.monotonic() < st["open_until"]:
-                    self._block(pr, responses, f"circuit_breaker: circuit open for tool {name}")
-                    continue
-            dispatch.append((pr, i, pr.arguments))
+            out.append((j, args2))
+        return out
 
-        # --- PASS 2: host rewrites + GPU re-scan of the rewritten subset ---
-        if rewrites:
-            self.slow_path += len(rewrites)
-            rw_args: List[Any] = []
-            rw_ok: List[Tuple[ParsedRequest, int, Any]] = []
-            for pr, i in rewrites:
-                status, payload = self._apply_rewrites(pr)
-                if status == "__block__":
-                    self._block(pr, responses, payload)
-                    continue
-                rw_ok.append((pr, i, payload))
-            if rw_ok:
-                texts2 = [canonical_text(a) for (_pr, _i, a) in rw_ok]
-                scan2 = self._scan_pass(texts2, ("harm",), classify=True, cache_lookup=False)
-                harm2 = scan2.get("mask_harm", np.zeros(len(rw_ok), dtype=np.int64))
-                scores2 = scan2.get("scores")
-                for j, (pr, i, args2) in enumerate(rw_ok):
-                    name = pr.tool_name
-                    if scores2 is not None and self._enforcing(self.moderation) and self._applies(self.moderation, name):
-                        msg = self._moderation_block(scores2[j])
-                        if msg:
-                            self._block(pr, responses, msg)
-                            continue
-                    if harm2[j] and self._enforcing(self.harm) and self._applies(self.harm, name):
-                        pid = int(harm2[j]).bit_length() - 1
-                        cat = self.harm.cats[pid] if pid < len(self.harm.cats) else "?"
-                        self._block(pr, responses, f"harmful_content_detector: harmful content ({cat})")
-                        continue
-                    if self._enforcing(self.schema_guard):
-                        msg = self._schema_errs(pr, args2)
-                        if msg:
-                            self._block(pr, responses, msg)
-                            continue
-                    if hits[i] is not None:
-                        responses[pr.index] = jsonrpc.result_response(pr.req.id, hits[i]).to_bytes()
-                        self.cache_hits += 1
-                        continue
-                    if self.breaker is not None and time.monotonic() < self.breaker._st(name)["open_until"]:
-                        self._block(pr, responses, f"circuit_breaker: circuit open for tool {name}")
-                        continue
-                    dispatch.append((pr, i, args2))
+    # ------------------------------------------------------------------
+    async def _dispatch_and_post(self, blob, env, rows, args_b, args_e, metas, feats, th_arr,
+                                 dispatch_rows: List[int], rewrite_dispatch: List[Tuple[int, Any]],
+                                 responses: List[Optional[bytes]], t0: float) -> None:
+        # Split native-batch vs python dispatch.
+        native_js: List[int] = []
+        py_items: List[Tuple[int, Any]] = []  # (j, args or None=raw)
+        for j in dispatch_rows:
+            mt = metas[j]
+            if mt.native_kind >= 0:
+                native_js.append(j)
+            else:
+                py_items.append((j, None))
+        for j, args2 in rewrite_dispatch:
+            mt = metas[j]
+            py_items.append((j, args2))  # rewritten args always go through python dispatch
 
-        # --- fan-out dispatch ---
-        self.fast_path += len(dispatch)
-        results = await self._dispatch_all(dispatch)
+        result_bytes: Dict[int, Optional[bytes]] = {}
+        errors: Dict[int, Exception] = {}
 
-        # --- PASS 3: result post chain ---
-        await self._post_pass(dispatch, results, responses, scan, th_arr, t0)
+        # --- native upstream batch call (C++) ---
+        if native_js:
+            kinds = np.array([metas[j].native_kind for j in native_js], dtype=np.int32)
+            nb_ = np.array([args_b[j] for j in native_js], dtype=np.int32)
+            ne_ = np.array([args_e[j] for j in native_js], dtype=np.int32)
+            nb_ = np.where(ne_ > nb_, nb_, -1).astype(np.int32)
+            now_iso = time.strftime("%Y-%m-%dT%H:%M:%SZ", time.gmtime())
+            out_blob, rb, re_ = hip.upstream_call_batch(blob, nb_, ne_, kinds, now_iso)
+            ob = out_blob.tobytes()
+            for jj, j in enumerate(native_js):
+                result_bytes[j] = ob[int(rb[jj]):int(re_[jj])]
+                metas[j].native_client.calls += 1
 
-    async def _dispatch_all(self, dispatch: List[Tuple[ParsedRequest, int, Any]]
-                            ) -> List[Tuple[Optional[dict], Optional[Exception]]]:
-        """In-proc/local upstreams via a plain await loop (no Task churn);
-        real-I/O upstreams concurrently."""
-        from ..services.upstream import InProcUpstream
+        # --- python dispatch for the rest ---
+        if py_items:
+            ts = self.engine.tool_service
+            from ..services.upstream import InProcUpstream
 
-        ts = self.engine.tool_service
-        results: List[Optional[Tuple[Optional[dict], Optional[Exception]]]] = [None] * len(dispatch)
-        io_idx: List[int] = []
-        for idx, (pr, _i, args) in enumerate(dispatch):
-            tool = pr.tool
-            itype = tool.get("integration_type", "MCP")
-            if itype == "LOCAL":
-                handler = ts._local_handlers.get(tool["name"])
-                if handler is None:
-                    results[idx] = (None, RuntimeError(f"no local handler for {tool['name']}"))
-                    continue
-                try:
-                    value = await handler(args or {})
-                    if not (isinstance(value, dict) and "content" in value):
-                        value = {
-                            "content": [{"type": "text", "text": value if isinstance(value, str) else json.dumps(value, default=str)}],
-                            "structuredContent": value if isinstance(value, (dict, list)) else None,
-                            "isError": False,
-                        }
-                    results[idx] = (value, None)
-                except Exception as exc:
-                    results[idx] = (None, exc)
-            elif itype == "MCP":
-                client = ts._upstreams.get(tool.get("gateway_id") or "")
-                if client is None:
-                    results[idx] = (None, RuntimeError(f"no upstream for {tool['name']}"))
-                elif isinstance(client, InProcUpstream):
+            async def one(j: int, args2: Any) -> None:
+                mt = metas[j]
+                args = args2
+                if args is None:
                     try:
-                        results[idx] = (await client.call_tool(tool["original_name"], args or {}), None)
-                    except Exception as exc:
-                        results[idx] = (None, exc)
-                else:
-                    io_idx.append(idx)
-            else:
-                io_idx.append(idx)
-
-        if io_idx:
-            async def one(idx: int):
-                pr, _i, args = dispatch[idx]
+                        args = json.loads(blob[args_b[j]:args_e[j]].tobytes() or b"{}")
+                    except Exception:
+                        errors[j] = jsonrpc.JSONRPCError(jsonrpc.INVALID_PARAMS, "invalid arguments")
+                        return
                 try:
-                    results[idx] = (await ts.dispatch(pr.tool, args), None)
+                    result = await ts.dispatch(mt.tool, args)
+                    result_bytes[j] = json.dumps(result, separators=(",", ":"), default=str).encode()
                 except Exception as exc:
-                    results[idx] = (None, exc)
+                    errors[j] = exc
 
-            await asyncio.gather(*(one(i) for i in io_idx))
-        return results  # type: ignore[return-value]
+            # in-proc dispatch is non-blocking → sequential loop; real I/O → gather
+            seq, conc = [], []
+            for j, args2 in py_items:
+                mt = metas[j]
+                if mt.itype == "LOCAL" or isinstance(mt.client, InProcUpstream) or mt.native_kind >= 0:
+                    seq.append((j, args2))
+                else:
+                    conc.append((j, args2))
+            for j, args2 in seq:
+                await one(j, args2)
+            if conc:
+                await asyncio.gather(*(one(j, a) for j, a in conc))
 
-    async def _post_pass(self, dispatch, results, responses, scan, th_arr, t0) -> None:
-        """PASS 3: serialize results once; scan the serialized form for
-        pii/regex/harm; flagged results take the host post chain; everyone
-        gets toon + length guard; cache insert + metrics at the end."""
-        n = len(dispatch)
-        res_texts: List[bytes] = []
-        ok_rows: List[int] = []  # indices into dispatch with a result (not exception)
-        for idx, ((pr, _i, _a), (result, exc)) in enumerate(zip(dispatch, results)):
-            if exc is None:
-                res_texts.append(json.dumps(result, separators=(",", ":"), default=str).encode())
-                ok_rows.append(idx)
-            else:
-                res_texts.append(b"")
+        all_js = native_js + [j for j, _ in py_items]
 
+        # --- PASS 3: GPU scan over serialized results ---
         post_banks = tuple(b for b in ("pii", "regex", "harm") if b in self.banks)
-        post_masks = None
-        if ok_rows and post_banks:
-            scan3 = self._scan_pass(res_texts, post_banks, classify=False, cache_lookup=False)
-            post_masks = {b: scan3.get(f"mask_{b}") for b in post_banks}
+        post_mask = None
+        ok_js = [j for j in all_js if j in result_bytes and result_bytes[j]]
+        if ok_js and post_banks:
+            texts3 = [result_bytes[j] for j in ok_js]
+            data3, beg3, end3 = pack_texts(texts3, self.device)
+            masks3 = []
+            for b in post_banks:
+                t, _ = hip.scan(data3, beg3, end3, self.banks[b])
+                masks3.append(t)
+            torch.cuda.synchronize()
+            combined = np.zeros(len(ok_js), dtype=np.int64)
+            for t in masks3:
+                combined |= t.cpu().numpy().astype(np.uint32).astype(np.int64)
+            post_mask = {j: bool(combined[i]) for i, j in enumerate(ok_js)}
 
         insert_rows: List[int] = []
         insert_hashes: List[int] = []
         insert_results: List[Any] = []
         agg: Dict[str, List[int]] = {}
         ms = (time.monotonic() - t0) * 1000.0
+        toon_min = self.toon.min_size if self.toon is not None else 1 << 60
+        guard_max = self.out_guard.max_chars if self.out_guard is not None else 1 << 60
+        now = time.monotonic()
 
-        for idx, ((pr, i, args), (result, exc)) in enumerate(zip(dispatch, results)):
-            rid = pr.req.id
-            tool_id = pr.tool.get("id", pr.tool_name)
-            a = agg.setdefault(tool_id, [0, 0])
+        for j in all_js:
+            mt = metas[j]
+            r = int(rows[j])
+            idb = self._id_bytes(blob, env, r)
+            a = agg.setdefault(mt.tid, [0, 0])
+            exc = errors.get(j)
             if exc is not None:
                 code = getattr(exc, "code", jsonrpc.SERVER_ERROR)
-                responses[pr.index] = jsonrpc.error_response(rid, code, str(exc)).to_bytes()
+                responses[r] = self._splice_error(idb, code if isinstance(code, int) else jsonrpc.SERVER_ERROR, str(exc))
                 a[0] += 1
                 a[1] += 1
                 if self.breaker is not None:
-                    self._breaker_record(pr.tool_name, True)
+                    self._breaker_record(mt.name, True)
                 continue
-
-            orig_result = result
-            flagged = False
-            if post_masks is not None:
-                for b, m in post_masks.items():
-                    if m is not None and m[idx]:
-                        flagged = True
-                        break
-            if flagged:
-                result, blocked_msg = self._host_post_rewrite(pr, result)
-                self.post_rewrites += 1
+            rb = result_bytes[j]
+            need_host = (post_mask is not None and post_mask.get(j, False)) or mt.has_output_schema \
+                or (len(rb) >= toon_min and b'"structuredContent"' in rb) or len(rb) > guard_max
+            is_err = False
+            if need_host:
+                rb, is_err, blocked_msg = self._host_post(mt, rb)
                 if blocked_msg:
-                    self._block(pr, responses, blocked_msg)
+                    responses[r] = self._splice_error(idb, jsonrpc.POLICY_DENIED, blocked_msg)
+                    self.blocked += 1
                     a[0] += 1
                     a[1] += 1
                     continue
-
-            # output schema check (tool_service parity)
-            oschema = pr.tool.get("output_schema")
-            if oschema and isinstance(result, dict):
-                from ..utils.jsonschema import validate as _validate
-
-                payload = result.get("structuredContent", result)
-                errs = _validate(payload, oschema)
-                if errs:
-                    result = {"content": [{"type": "text", "text": "output schema violation: " + "; ".join(errs[:3])}],
-                              "isError": True}
-
-            result = self._toon_guard(pr, result)
-            is_err = bool(isinstance(result, dict) and result.get("isError"))
             a[0] += 1
-            a[1] += 1 if is_err else 0
+            if is_err:
+                a[1] += 1
             if self.breaker is not None:
-                self._breaker_record(pr.tool_name, is_err)
-
-            if result is orig_result and res_texts[idx]:
-                # untouched result: splice the already-serialized form (one dumps total)
-                responses[pr.index] = b'{"jsonrpc":"2.0","id":' + json.dumps(rid).encode() + b',"result":' + res_texts[idx] + b"}"
-            else:
-                responses[pr.index] = jsonrpc.result_response(rid, result).to_bytes()
+                self._breaker_record(mt.name, is_err)
+            if idb is not None:
+                responses[r] = self._splice_result(idb, rb)
             if not is_err:
                 if self.exact_cache is not None:
-                    k = hashlib.sha256(pr.tool_name.encode() + pr.arg_text).hexdigest()
-                    self.exact_cache.store[k] = (time.monotonic(), result)
+                    self.exact_cache.store[(mt.name, blob[args_b[j]:args_e[j]].tobytes())] = (now, rb)
                 if self.semcache is not None:
-                    insert_rows.append(i)
-                    insert_hashes.append(int(th_arr[i]))
-                    insert_results.append(result)
+                    insert_rows.append(j)
+                    insert_hashes.append(int(th_arr[j]))
+                    insert_results.append(rb)
 
-        if insert_rows and scan.get("_feats") is not None:
-            self.semcache.insert_batch(scan["_feats"], insert_rows, insert_hashes, insert_results)
-        for tool_id, (cnt, errs) in agg.items():
-            self.engine.metrics.record_aggregate(tool_id, cnt, errs, ms)
+        if insert_rows and feats is not None:
+            self.semcache.insert_batch(feats, insert_rows, insert_hashes, insert_results)
+        for tid, (cnt, errs_) in agg.items():
+            self.engine.metrics.record_aggregate(tid, cnt, errs_, ms)
 
-    def _host_post_rewrite(self, pr: ParsedRequest, result: dict) -> Tuple[dict, Optional[str]]:
+    def _host_post(self, mt: _ToolMeta, rb: bytes) -> Tuple[bytes, bool, Optional[str]]:
         """Exact host post chain for flagged results: regex(20) → pii(30) →
-        harm(60), matching plugin post-hook semantics."""
-        name = pr.tool_name
+        output-schema → harm(60) → toon(900) → guard(950)."""
+        self.post_rewrites += 1
+        try:
+            result = json.loads(rb)
+        except Exception:
+            return rb, True, None
+        name = mt.name
         if self.regex is not None and self._applies(self.regex, name):
             result = _walk_strings(result, self.regex.apply_rules)
         if self.pii is not None and self._applies(self.pii, name):
@@ -524,18 +791,23 @@ class GpuPluginPipeline:
 
             new = _walk_strings(result, fn)
             if found and self.pii.action == "block" and self._enforcing(self.pii):
-                return result, f"pii_filter: PII detected: {sorted(set(found))}"
+                return rb, True, f"pii_filter: PII detected: {sorted(set(found))}"
             if found and self.pii.action == "mask":
                 result = new
+        if mt.has_output_schema and isinstance(result, dict):
+            from ..utils.jsonschema import validate as _validate
+
+            payload = result.get("structuredContent", result)
+            errs = _validate(payload, mt.tool.get("output_schema") or {})
+            if errs:
+                result = {"content": [{"type": "text", "text": "output schema violation: " + "; ".join(errs[:3])}],
+                          "isError": True}
         if self.harm is not None and self._enforcing(self.harm) and self._applies(self.harm, name):
             hay = json.dumps(result, separators=(",", ":"), sort_keys=True, default=str).lower()
             for phrase, cat in zip(self.harm.phrases, self.harm.cats):
                 if phrase.lower() in hay:
-                    return result, f"harmful_content_detector: harmful content ({cat})"
-        return result, None
-
-    def _toon_guard(self, pr: ParsedRequest, result: dict) -> dict:
-        if self.toon is not None and self._applies(self.toon, pr.tool_name):
+                    return rb, True, f"harmful_content_detector: harmful content ({cat})"
+        if self.toon is not None and self._applies(self.toon, name) and isinstance(result, dict):
             new = self.toon.encode_result(result)
             if new is not None:
                 result = new
@@ -548,7 +820,8 @@ class GpuPluginPipeline:
                     if isinstance(c, dict) and isinstance(c.get("text"), str) else c
                     for c in result.get("content", [])
                 ]
-        return result
+        is_err = bool(isinstance(result, dict) and result.get("isError"))
+        return json.dumps(result, separators=(",", ":"), default=str).encode(), is_err, None
 
     def _breaker_record(self, name: str, is_error: bool) -> None:
         b = self.breaker
@@ -564,7 +837,7 @@ class GpuPluginPipeline:
         out = {
             "batches": self.batches, "requests": self.requests, "fast_path": self.fast_path,
             "slow_path": self.slow_path, "blocked": self.blocked, "cache_hits": self.cache_hits,
-            "post_rewrites": self.post_rewrites,
+            "post_rewrites": self.post_rewrites, "py_fallback": self.py_fallback,
             "banks": {k: {"states": v.n_states, "classes": v.n_classes} for k, v in self.banks.items()},
         }
         if self.semcache is not None:

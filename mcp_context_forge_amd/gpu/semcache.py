@@ -66,6 +66,25 @@ class GpuSemanticCache:
             hip.rows_argmax_merge(scores, best_val, best_idx, idx_base=c0, valid=self.valid[c0:c0 + nc])
         return best_val, best_idx
 
+    def resolve_hits_np(self, best_val: np.ndarray, best_idx: np.ndarray,
+                        tool_hashes: np.ndarray) -> np.ndarray:
+        """Vectorized hit confirmation → bool mask. Caller fetches
+        self.results[best_idx[i]] for hits."""
+        n = best_val.shape[0]
+        if self.size == 0:
+            self.misses += n
+            return np.zeros(n, dtype=bool)
+        idx = best_idx.astype(np.int64)
+        safe = np.clip(idx, 0, self.capacity - 1)
+        now = time.monotonic()
+        mask = (idx >= 0) & (best_val >= self.threshold) \
+            & (self.tool_hashes[safe] == tool_hashes) \
+            & ((now - self.timestamps[safe]) <= self.ttl_s)
+        nh = int(mask.sum())
+        self.hits += nh
+        self.misses += n - nh
+        return mask
+
     def resolve_hits(self, best_val: np.ndarray, best_idx: np.ndarray,
                      tool_hashes: np.ndarray) -> List[Optional[Any]]:
         """Host-side hit confirmation: threshold + tool identity + TTL

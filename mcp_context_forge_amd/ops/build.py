@@ -18,7 +18,8 @@ LIB = Path(__file__).parent / "libforge_hip.so"
 ARCH = os.environ.get("PYTORCH_ROCM_ARCH", "gfx950")
 HIPCC = os.environ.get("HIPCC", "/opt/rocm/bin/hipcc")
 
-SOURCES = ["scan.hip", "featurize.hip", "json_guard.hip", "gemm_bf16.hip", "runtime.hip"]
+SOURCES = ["scan.hip", "featurize.hip", "json_guard.hip", "gemm_bf16.hip", "runtime.hip",
+           "envelope.cpp", "upstream.cpp"]
 
 
 def needs_build() -> bool:

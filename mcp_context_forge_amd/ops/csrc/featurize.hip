@@ -22,15 +22,16 @@ __device__ __forceinline__ uint8_t lower(uint8_t b) { return (b >= 'A' && b <= '
 
 __global__ __launch_bounds__(256) void featurize_kernel(
     const uint8_t* __restrict__ data,
-    const int32_t* __restrict__ offsets,
+    const int32_t* __restrict__ beg,
+    const int32_t* __restrict__ end_,
     int dim,
     short* __restrict__ out_bf16,   // [B, dim] or nullptr
     float* __restrict__ out_f32)    // [B, dim] or nullptr
 {
     extern __shared__ uint32_t hist[];  // [dim] counts, then 1 extra slot for norm
     int r = blockIdx.x;
-    int32_t beg = offsets[r], end = offsets[r + 1];
-    int len = end - beg;
+    int32_t rbeg = beg[r], rend = end_[r];
+    int len = rend - rbeg;
 
     for (int i = threadIdx.x; i < dim; i += blockDim.x) hist[i] = 0u;
     __syncthreads();
@@ -41,14 +42,14 @@ __global__ __launch_bounds__(256) void featurize_kernel(
     int c0 = threadIdx.x * chunk;
     int c1 = min(c0 + chunk, len);
     for (int i = c0; i < c1; ++i) {
-        uint8_t b = data[beg + i];
+        uint8_t b = data[rbeg + i];
         if (!is_word(b)) continue;
-        bool starts = (i == 0) || !is_word(data[beg + i - 1]);
+        bool starts = (i == 0) || !is_word(data[rbeg + i - 1]);
         if (!starts) continue;
         uint32_t h = FNV_OFFSET;
         int j = i;
         while (j < len) {
-            uint8_t bj = data[beg + j];
+            uint8_t bj = data[rbeg + j];
             if (!is_word(bj)) break;
             h = (h ^ (uint32_t)lower(bj)) * FNV_PRIME;
             ++j;
@@ -85,13 +86,13 @@ __global__ __launch_bounds__(256) void featurize_kernel(
 }
 
 extern "C" int forge_featurize(
-    const void* data, const void* offsets, int batch, int dim,
+    const void* data, const void* beg, const void* end_, int batch, int dim,
     void* out_bf16, void* out_f32, void* stream)
 {
     hipStream_t s = (hipStream_t)stream;
     size_t lds = (size_t)dim * 4;
     hipLaunchKernelGGL(featurize_kernel, dim3(batch), dim3(256), lds, s,
-                       (const uint8_t*)data, (const int32_t*)offsets, dim,
+                       (const uint8_t*)data, (const int32_t*)beg, (const int32_t*)end_, dim,
                        (short*)out_bf16, (float*)out_f32);
     return (int)hipGetLastError();
 }

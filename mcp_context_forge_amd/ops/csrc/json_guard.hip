@@ -16,14 +16,15 @@
 
 __global__ __launch_bounds__(256) void json_guard_kernel(
     const uint8_t* __restrict__ data,
-    const int32_t* __restrict__ offsets,
+    const int32_t* __restrict__ beg,
+    const int32_t* __restrict__ end_,
     int batch, int max_depth, int max_string,
     int32_t* __restrict__ out_status,
     int32_t* __restrict__ out_depth)
 {
     int r = blockIdx.x * blockDim.x + threadIdx.x;
     if (r >= batch) return;
-    int32_t beg = offsets[r], end = offsets[r + 1];
+    int32_t rbeg = beg[r], rend = end_[r];
 
     int status = 0;
     int depth = 0, maxd = 0;
@@ -31,7 +32,7 @@ __global__ __launch_bounds__(256) void json_guard_kernel(
     int str_len = 0;
     int cont = 0;  // pending UTF-8 continuation bytes
 
-    for (int32_t p = beg; p < end; ++p) {
+    for (int32_t p = rbeg; p < rend; ++p) {
         uint8_t b = data[p];
         // UTF-8 validity
         if (cont > 0) {
@@ -64,13 +65,13 @@ __global__ __launch_bounds__(256) void json_guard_kernel(
 }
 
 extern "C" int forge_json_guard(
-    const void* data, const void* offsets, int batch, int max_depth, int max_string,
+    const void* data, const void* beg, const void* end_, int batch, int max_depth, int max_string,
     void* out_status, void* out_depth, void* stream)
 {
     hipStream_t s = (hipStream_t)stream;
     int block = 256;
     hipLaunchKernelGGL(json_guard_kernel, dim3(ceil_div(batch, block)), dim3(block), 0, s,
-                       (const uint8_t*)data, (const int32_t*)offsets, batch, max_depth, max_string,
+                       (const uint8_t*)data, (const int32_t*)beg, (const int32_t*)end_, batch, max_depth, max_string,
                        (int32_t*)out_status, (int32_t*)out_depth);
     return (int)hipGetLastError();
 }

@@ -17,7 +17,8 @@
 template <bool USE_LDS>
 __global__ __launch_bounds__(256) void scan_kernel(
     const uint8_t* __restrict__ data,     // packed bytes
-    const int32_t* __restrict__ offsets,  // [B+1]
+    const int32_t* __restrict__ beg,
+    const int32_t* __restrict__ end_,
     int batch,
     const uint16_t* __restrict__ next,    // [S, C]
     const uint8_t* __restrict__ klass,    // [256]
@@ -45,16 +46,16 @@ __global__ __launch_bounds__(256) void scan_kernel(
 
     int r = blockIdx.x * blockDim.x + threadIdx.x;
     if (r >= batch) return;
-    int32_t beg = offsets[r], end = offsets[r + 1];
+    int32_t rbeg = beg[r], rend = end_[r];
     uint32_t mask = 0;
     int32_t first_end = -1;
     uint32_t state = 0;
-    for (int32_t p = beg; p < end; ++p) {
+    for (int32_t p = rbeg; p < rend; ++p) {
         uint8_t b = data[p];
         state = tnext[state * n_classes + tklass[b]];
         uint32_t a = taccept[state];
         if (a) {
-            if (first_end < 0) first_end = p + 1 - beg;
+            if (first_end < 0) first_end = p + 1 - rbeg;
             mask |= a;
         }
     }
@@ -63,7 +64,7 @@ __global__ __launch_bounds__(256) void scan_kernel(
 }
 
 extern "C" int forge_scan(
-    const void* data, const void* offsets, int batch,
+    const void* data, const void* beg, const void* end_, int batch,
     const void* next, const void* klass, const void* accept,
     int n_states, int n_classes,
     void* out_mask, void* out_first_end, void* stream)
@@ -74,12 +75,12 @@ extern "C" int forge_scan(
     size_t table_bytes = (size_t)n_states * n_classes * 2 + 256 + (size_t)n_states * 4;
     if (table_bytes <= 64 * 1024) {
         hipLaunchKernelGGL((scan_kernel<true>), dim3(grid), dim3(block), table_bytes, s,
-                           (const uint8_t*)data, (const int32_t*)offsets, batch,
+                           (const uint8_t*)data, (const int32_t*)beg, (const int32_t*)end_, batch,
                            (const uint16_t*)next, (const uint8_t*)klass, (const uint32_t*)accept,
                            n_states, n_classes, (uint32_t*)out_mask, (int32_t*)out_first_end);
     } else {
         hipLaunchKernelGGL((scan_kernel<false>), dim3(grid), dim3(block), 0, s,
-                           (const uint8_t*)data, (const int32_t*)offsets, batch,
+                           (const uint8_t*)data, (const int32_t*)beg, (const int32_t*)end_, batch,
                            (const uint16_t*)next, (const uint8_t*)klass, (const uint32_t*)accept,
                            n_states, n_classes, (uint32_t*)out_mask, (int32_t*)out_first_end);
     }

@@ -1,0 +1,157 @@
+// Native in-process MCP upstream — the fast_time_server analog.
+//
+// The reference's benchmark upstream is a native Go binary
+// (docker-compose.yml:1485 fast_time_server, targeted by tests/hey/
+// payload2.json `convert_time`). This is the same thing as a C++ batch
+// call: for each request it parses the raw `arguments` span, runs the tool
+// handler (convert_time / get_system_time / echo), and serializes a full
+// MCP tool-result JSON into one output blob. The gateway splices these
+// bytes into JSON-RPC responses without re-parsing.
+//
+// Handler kinds: 0 = convert_time, 1 = get_system_time, 2 = echo.
+
+#include <stdint.h>
+#include <stdlib.h>
+#include <string.h>
+#include <string>
+#include <vector>
+
+namespace {
+
+// extract the raw span of `"key": <value>` inside a flat JSON object.
+// Returns value span [vb, ve) or false.
+bool find_key(const uint8_t* b, const uint8_t* e, const char* key,
+              const uint8_t** vb, const uint8_t** ve) {
+    size_t klen = strlen(key);
+    const uint8_t* p = b;
+    if (p >= e || *p != '{') return false;
+    ++p;
+    bool in_str = false, esc = false;
+    int depth = 0;
+    // scan top-level keys
+    while (p < e) {
+        // skip ws
+        while (p < e && (*p == ' ' || *p == '\t' || *p == '\n' || *p == '\r')) ++p;
+        if (p < e && *p == '}') return false;
+        if (p >= e || *p != '"') return false;
+        const uint8_t* k0 = ++p;
+        while (p < e && *p != '"') { if (*p == '\\') ++p; ++p; }
+        const uint8_t* k1 = p;
+        if (p >= e) return false;
+        ++p;  // closing quote
+        while (p < e && (*p == ' ' || *p == '\t' || *p == '\n' || *p == '\r')) ++p;
+        if (p >= e || *p != ':') return false;
+        ++p;
+        while (p < e && (*p == ' ' || *p == '\t' || *p == '\n' || *p == '\r')) ++p;
+        const uint8_t* v0 = p;
+        // skip value
+        if (p < e && (*p == '{' || *p == '[')) {
+            depth = 0; in_str = false; esc = false;
+            while (p < e) {
+                uint8_t x = *p++;
+                if (in_str) { if (esc) esc = false; else if (x == '\\') esc = true; else if (x == '"') in_str = false; }
+                else if (x == '"') in_str = true;
+                else if (x == '{' || x == '[') ++depth;
+                else if (x == '}' || x == ']') { if (--depth == 0) break; }
+            }
+        } else if (p < e && *p == '"') {
+            ++p;
+            while (p < e && *p != '"') { if (*p == '\\') ++p; ++p; }
+            if (p < e) ++p;
+        } else {
+            while (p < e && *p != ',' && *p != '}') ++p;
+        }
+        const uint8_t* v1 = p;
+        if ((size_t)(k1 - k0) == klen && memcmp(k0, key, klen) == 0) {
+            *vb = v0;
+            *ve = v1;
+            return true;
+        }
+        while (p < e && (*p == ' ' || *p == '\t' || *p == '\n' || *p == '\r')) ++p;
+        if (p < e && *p == ',') { ++p; continue; }
+        return false;
+    }
+    return false;
+}
+
+void append_json_escaped(std::string& out, const uint8_t* b, const uint8_t* e) {
+    for (const uint8_t* p = b; p < e; ++p) {
+        uint8_t c = *p;
+        switch (c) {
+            case '"': out += "\\\""; break;
+            case '\\': out += "\\\\"; break;
+            case '\n': out += "\\n"; break;
+            case '\r': out += "\\r"; break;
+            case '\t': out += "\\t"; break;
+            default:
+                if (c < 0x20) {
+                    char buf[8];
+                    snprintf(buf, sizeof(buf), "\\u%04x", c);
+                    out += buf;
+                } else {
+                    out += (char)c;
+                }
+        }
+    }
+}
+
+void append_span(std::string& out, const uint8_t* b, const uint8_t* e) {
+    out.append((const char*)b, (size_t)(e - b));
+}
+
+}  // namespace
+
+// Batch upstream call.
+//   data/args spans: raw argument objects (may be -1,-1 = empty)
+//   kinds[r]: handler kind (0/1/2); now_iso: host-provided wall clock string
+// Output: caller-provided growing buffer protocol — we return the required
+// size on first call (out=null), caller allocates and calls again.
+extern "C" int64_t forge_upstream_call_batch(
+    const uint8_t* data, const int32_t* args_beg, const int32_t* args_end,
+    const int32_t* kinds, int n, const char* now_iso,
+    uint8_t* out, int64_t out_cap, int64_t* res_beg, int64_t* res_end)
+{
+    std::string buf;
+    buf.reserve((size_t)n * 192);
+    static const uint8_t EMPTY[2] = {'{', '}'};
+    for (int r = 0; r < n; ++r) {
+        res_beg[r] = (int64_t)buf.size();
+        const uint8_t* ab = args_beg[r] >= 0 ? data + args_beg[r] : EMPTY;
+        const uint8_t* ae = args_beg[r] >= 0 ? data + args_end[r] : EMPTY + 2;
+        int k = kinds[r];
+        if (k == 0) {  // convert_time
+            const uint8_t *tb, *te, *sb, *se, *gb, *ge;
+            bool ht = find_key(ab, ae, "time", &tb, &te);
+            bool hs = find_key(ab, ae, "source_timezone", &sb, &se);
+            bool hg = find_key(ab, ae, "target_timezone", &gb, &ge);
+            buf += "{\"content\":[{\"type\":\"text\",\"text\":\"converted\"}],\"structuredContent\":{\"time\":";
+            if (ht) append_span(buf, tb, te); else buf += "\"1970-01-01T00:00:00Z\"";
+            buf += ",\"source_timezone\":";
+            if (hs) append_span(buf, sb, se); else buf += "\"UTC\"";
+            buf += ",\"target_timezone\":";
+            if (hg) append_span(buf, gb, ge); else buf += "\"UTC\"";
+            buf += ",\"converted\":true},\"isError\":false}";
+        } else if (k == 1) {  // get_system_time
+            const uint8_t *zb, *ze;
+            bool hz = find_key(ab, ae, "timezone", &zb, &ze);
+            buf += "{\"content\":[{\"type\":\"text\",\"text\":\"";
+            buf += now_iso;
+            buf += "\"}],\"structuredContent\":{\"time\":\"";
+            buf += now_iso;
+            buf += "\",\"timezone\":";
+            if (hz) append_span(buf, zb, ze); else buf += "\"UTC\"";
+            buf += "},\"isError\":false}";
+        } else {  // echo
+            buf += "{\"content\":[{\"type\":\"text\",\"text\":\"";
+            append_json_escaped(buf, ab, ae);
+            buf += "\"}],\"structuredContent\":";
+            append_span(buf, ab, ae);
+            buf += ",\"isError\":false}";
+        }
+        res_end[r] = (int64_t)buf.size();
+    }
+    if (out != nullptr && (int64_t)buf.size() <= out_cap) {
+        memcpy(out, buf.data(), buf.size());
+    }
+    return (int64_t)buf.size();
+}

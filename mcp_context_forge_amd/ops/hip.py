@@ -30,19 +30,21 @@ def _load() -> ctypes.CDLL:
         build()
     lib = ctypes.CDLL(str(LIB))
     protos = {
-        "forge_scan": [ctypes.c_void_p] * 2 + [ctypes.c_int] + [ctypes.c_void_p] * 3 + [ctypes.c_int] * 2 + [ctypes.c_void_p] * 3,
-        "forge_featurize": [ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int, ctypes.c_int, ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p],
-        "forge_json_guard": [ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int, ctypes.c_int, ctypes.c_int, ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p],
+        "forge_scan": [ctypes.c_void_p] * 3 + [ctypes.c_int] + [ctypes.c_void_p] * 3 + [ctypes.c_int] * 2 + [ctypes.c_void_p] * 3,
+        "forge_featurize": [ctypes.c_void_p] * 3 + [ctypes.c_int, ctypes.c_int] + [ctypes.c_void_p] * 3,
+        "forge_json_guard": [ctypes.c_void_p] * 3 + [ctypes.c_int] * 3 + [ctypes.c_void_p] * 3,
         "forge_gemm_bt": [ctypes.c_void_p] * 4 + [ctypes.c_int] * 5 + [ctypes.c_void_p],
         "forge_gemv_head": [ctypes.c_void_p] * 4 + [ctypes.c_int] * 4 + [ctypes.c_void_p],
         "forge_rows_argmax_merge": [ctypes.c_void_p, ctypes.c_int, ctypes.c_int, ctypes.c_int, ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p],
         "forge_rows_scatter_bf16": [ctypes.c_void_p] * 3 + [ctypes.c_int] * 2 + [ctypes.c_void_p],
         "forge_synchronize": [ctypes.c_void_p],
+        "forge_parse_envelopes": [ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int] + [ctypes.c_void_p] * 7,
+        "forge_upstream_call_batch": [ctypes.c_void_p] * 4 + [ctypes.c_int, ctypes.c_char_p, ctypes.c_void_p, ctypes.c_int64, ctypes.c_void_p, ctypes.c_void_p],
     }
     for name, argtypes in protos.items():
         fn = getattr(lib, name)
         fn.argtypes = argtypes
-        fn.restype = ctypes.c_int
+        fn.restype = ctypes.c_int64 if name == "forge_upstream_call_batch" else ctypes.c_int
     return lib
 
 
@@ -79,37 +81,37 @@ class DeviceScanTables:
         self.n_classes = tables.n_classes
 
 
-def scan(data: torch.Tensor, offsets: torch.Tensor, tables: DeviceScanTables,
+def scan(data: torch.Tensor, beg: torch.Tensor, end: torch.Tensor, tables: DeviceScanTables,
          first_end: bool = False) -> Tuple[torch.Tensor, Optional[torch.Tensor]]:
-    """data uint8 [total], offsets int32 [B+1] on device → mask int32 [B]."""
-    batch = offsets.numel() - 1
+    """data uint8 [total], beg/end int32 [B] spans on device → mask int32 [B]."""
+    batch = beg.numel()
     out_mask = torch.zeros(batch, dtype=torch.int32, device=data.device)
     out_end = torch.full((batch,), -1, dtype=torch.int32, device=data.device) if first_end else None
     _check("forge_scan", _load().forge_scan(
-        _ptr(data), _ptr(offsets), batch,
+        _ptr(data), _ptr(beg), _ptr(end), batch,
         _ptr(tables.next), _ptr(tables.klass), _ptr(tables.accept),
         tables.n_states, tables.n_classes,
         _ptr(out_mask), _ptr(out_end), _stream()))
     return out_mask, out_end
 
 
-def featurize(data: torch.Tensor, offsets: torch.Tensor, dim: int,
+def featurize(data: torch.Tensor, beg: torch.Tensor, end: torch.Tensor, dim: int,
               want_f32: bool = False) -> Tuple[torch.Tensor, Optional[torch.Tensor]]:
-    batch = offsets.numel() - 1
+    batch = beg.numel()
     out_bf16 = torch.empty((batch, dim), dtype=torch.bfloat16, device=data.device)
     out_f32 = torch.empty((batch, dim), dtype=torch.float32, device=data.device) if want_f32 else None
     _check("forge_featurize", _load().forge_featurize(
-        _ptr(data), _ptr(offsets), batch, dim, _ptr(out_bf16), _ptr(out_f32), _stream()))
+        _ptr(data), _ptr(beg), _ptr(end), batch, dim, _ptr(out_bf16), _ptr(out_f32), _stream()))
     return out_bf16, out_f32
 
 
-def json_guard(data: torch.Tensor, offsets: torch.Tensor, max_depth: int = 64,
+def json_guard(data: torch.Tensor, beg: torch.Tensor, end: torch.Tensor, max_depth: int = 64,
                max_string: int = 1 << 20) -> Tuple[torch.Tensor, torch.Tensor]:
-    batch = offsets.numel() - 1
+    batch = beg.numel()
     status = torch.zeros(batch, dtype=torch.int32, device=data.device)
     depth = torch.zeros(batch, dtype=torch.int32, device=data.device)
     _check("forge_json_guard", _load().forge_json_guard(
-        _ptr(data), _ptr(offsets), batch, max_depth, max_string, _ptr(status), _ptr(depth), _stream()))
+        _ptr(data), _ptr(beg), _ptr(end), batch, max_depth, max_string, _ptr(status), _ptr(depth), _stream()))
     return status, depth
 
 
@@ -151,3 +153,57 @@ def rows_argmax_merge(scores: torch.Tensor, best_val: torch.Tensor, best_idx: to
 def rows_scatter_bf16(src: torch.Tensor, slots: torch.Tensor, dst: torch.Tensor) -> None:
     r, d = src.shape
     _check("forge_rows_scatter_bf16", _load().forge_rows_scatter_bf16(_ptr(src), _ptr(slots), _ptr(dst), r, d, _stream()))
+
+
+# ---------------------------------------------------------------------------
+# Host-side native fast path (CPU functions in the same library)
+# ---------------------------------------------------------------------------
+
+ENV_NEEDS_PY, ENV_PARSE_ERR, ENV_OTHER, ENV_TOOLS_CALL = -2, -1, 0, 1
+
+
+def parse_envelopes(blob: np.ndarray, offsets: np.ndarray) -> dict:
+    """C++ JSON-RPC envelope scan over a packed request blob (host memory).
+
+    blob: uint8 [total]; offsets: int64 [n+1]. Returns numpy arrays:
+    kind, id_beg/id_end, name_beg/name_end, args_beg/args_end (int32 [n]).
+    """
+    n = offsets.shape[0] - 1
+    kind = np.empty(n, dtype=np.int32)
+    id_b = np.empty(n, dtype=np.int32)
+    id_e = np.empty(n, dtype=np.int32)
+    nm_b = np.empty(n, dtype=np.int32)
+    nm_e = np.empty(n, dtype=np.int32)
+    ar_b = np.empty(n, dtype=np.int32)
+    ar_e = np.empty(n, dtype=np.int32)
+
+    def p(a):
+        return ctypes.c_void_p(a.ctypes.data)
+
+    _check("forge_parse_envelopes", _load().forge_parse_envelopes(
+        p(blob), p(offsets), n, p(kind), p(id_b), p(id_e), p(nm_b), p(nm_e), p(ar_b), p(ar_e)))
+    return {"kind": kind, "id_beg": id_b, "id_end": id_e,
+            "name_beg": nm_b, "name_end": nm_e, "args_beg": ar_b, "args_end": ar_e}
+
+
+def upstream_call_batch(blob: np.ndarray, args_beg: np.ndarray, args_end: np.ndarray,
+                        kinds: np.ndarray, now_iso: str) -> Tuple[np.ndarray, np.ndarray, np.ndarray]:
+    """Native in-proc upstream batch call. Returns (out_blob u8, res_beg i64, res_end i64)."""
+    n = kinds.shape[0]
+    res_b = np.empty(n, dtype=np.int64)
+    res_e = np.empty(n, dtype=np.int64)
+
+    def p(a):
+        return ctypes.c_void_p(a.ctypes.data)
+
+    lib = _load()
+    spans_total = int((args_end[kinds >= 0] - np.maximum(args_beg[kinds >= 0], 0)).clip(min=0).sum()) if n else 0
+    cap = spans_total * 3 + n * 256 + 1024
+    out = np.empty(cap, dtype=np.uint8)
+    need = lib.forge_upstream_call_batch(p(blob), p(args_beg), p(args_end), p(kinds), n,
+                                         now_iso.encode(), p(out), cap, p(res_b), p(res_e))
+    if need > cap:  # rare: resize and redo
+        out = np.empty(int(need) + 1024, dtype=np.uint8)
+        lib.forge_upstream_call_batch(p(blob), p(args_beg), p(args_end), p(kinds), n,
+                                      now_iso.encode(), p(out), out.shape[0], p(res_b), p(res_e))
+    return out, res_b, res_e

@@ -240,7 +240,7 @@ class ToonEncoderPlugin(Plugin):
     def __init__(self, config: Optional[Dict[str, Any]] = None):
         super().__init__(config)
         self.min_savings = float(self.config.get("min_savings", 0.1))
-        self.min_size = int(self.config.get("min_size", 128))
+        self.min_size = int(self.config.get("min_size", 256))
 
     def encode_result(self, result: Any) -> Optional[Dict[str, Any]]:
         """Returns modified tool-result dict or None if not worth encoding."""

@@ -265,3 +265,61 @@ def make_fake_time_upstream(name: str = "fast_time", latency_s: float = 0.0) -> 
     )
     up.add_tool("echo", echo, "Echo arguments back")
     return up
+
+
+class NativeInProcUpstream(UpstreamClient):
+    """C++ in-process MCP upstream (ops/csrc/upstream.cpp) — the native
+    analog of the reference's Go fast_time_server benchmark upstream
+    (docker-compose.yml:1485). Exposes the same three tools; each call does
+    a real parse of the raw argument bytes and serializes a full MCP tool
+    result, in C++. The GPU pipeline batches calls (`native_kind` per tool);
+    this per-call API keeps the CPU path and tests working identically.
+    """
+
+    TOOL_KINDS = {"convert_time": 0, "get_system_time": 1, "echo": 2}
+
+    def __init__(self, name: str = "native_time"):
+        self.name = name
+        self.calls = 0
+
+    async def initialize(self) -> Dict[str, Any]:
+        return {"protocolVersion": PROTOCOL_VERSION, "serverInfo": {"name": self.name, "version": "0"},
+                "capabilities": {"tools": {}}}
+
+    async def list_tools(self) -> List[Dict[str, Any]]:
+        schema_ct = {
+            "type": "object",
+            "properties": {"time": {"type": "string"}, "source_timezone": {"type": "string"},
+                           "target_timezone": {"type": "string"}},
+            "required": ["time", "source_timezone", "target_timezone"],
+        }
+        return [
+            {"name": "convert_time", "description": "Convert time between timezones (native)", "inputSchema": schema_ct},
+            {"name": "get_system_time", "description": "Current system time (native)",
+             "inputSchema": {"type": "object", "properties": {"timezone": {"type": "string"}}}},
+            {"name": "echo", "description": "Echo arguments (native)", "inputSchema": {"type": "object"}},
+        ]
+
+    def call_tool_raw(self, kind: int, args_raw: bytes, now_iso: str) -> bytes:
+        import numpy as np
+
+        from ..ops import hip
+
+        blob = np.frombuffer(args_raw or b"{}", dtype=np.uint8)
+        ab = np.array([0], dtype=np.int32)
+        ae = np.array([blob.shape[0]], dtype=np.int32)
+        kinds = np.array([kind], dtype=np.int32)
+        out, rb, re_ = hip.upstream_call_batch(blob, ab, ae, kinds, now_iso)
+        self.calls += 1
+        return out[int(rb[0]):int(re_[0])].tobytes()
+
+    async def call_tool(self, name: str, arguments: Dict[str, Any], headers: Optional[Dict[str, str]] = None) -> Dict[str, Any]:
+        kind = self.TOOL_KINDS.get(name)
+        if kind is None:
+            raise UpstreamError(f"tool {name} not found on upstream {self.name}", code=jsonrpc.METHOD_NOT_FOUND)
+        now = time.strftime("%Y-%m-%dT%H:%M:%SZ", time.gmtime())
+        raw = json.dumps(arguments or {}, separators=(",", ":")).encode()
+        return json.loads(self.call_tool_raw(kind, raw, now))
+
+    async def ping(self) -> bool:
+        return True
