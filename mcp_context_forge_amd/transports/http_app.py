@@ -1,0 +1,519 @@
+"""HTTP edge: FastAPI application.
+
+Reference analogs: mcpgateway/main.py app assembly (:2096), middleware stack
+(:3282-3474, ordering per docs/docs/architecture/middleware-ordering.md),
+REST CRUD routers (:3557), JSON-RPC POST /rpc (:7896), protocol endpoints
+(:3962-4082), per-server SSE pair (:4478/:4597), streamable-HTTP /mcp
+(transports/streamablehttp_transport.py), WebSocket /ws (:11866),
+well-known endpoints (routers/well_known.py), admin API (admin.py).
+
+The /rpc and /mcp POST paths feed the GPU micro-batch collector when the
+HIP pipeline is attached; everything else is standard control-plane work.
+"""
+
+from __future__ import annotations
+
+import asyncio
+import json
+import time
+import uuid
+from typing import Any, Dict, List, Optional
+
+from fastapi import Depends, FastAPI, HTTPException, Query, Request, Response, WebSocket, WebSocketDisconnect
+from fastapi.responses import HTMLResponse, JSONResponse, PlainTextResponse, StreamingResponse
+from starlette.middleware.base import BaseHTTPMiddleware
+from starlette.middleware.cors import CORSMiddleware
+
+from ..auth.service import AuthContext, AuthError, AuthService, PermissionError_
+from ..config import Settings
+from ..engine import GatewayEngine
+from ..gpu.collector import BatchCollector
+from ..protocol import jsonrpc
+from ..protocol.mcp import PROTOCOL_VERSION, initialize_result
+from ..registry.registry import ConflictError, NotFoundError
+from ..services.gateway_service import GatewayConnectionError
+from ..utils import TokenBucket
+
+# ---------------------------------------------------------------------------
+# middleware (reference execution order, trimmed to the load-bearing set)
+# ---------------------------------------------------------------------------
+
+
+class SecurityHeadersMiddleware(BaseHTTPMiddleware):
+    """reference: middleware/security_headers.py"""
+
+    async def dispatch(self, request, call_next):
+        resp = await call_next(request)
+        resp.headers.setdefault("X-Content-Type-Options", "nosniff")
+        resp.headers.setdefault("X-Frame-Options", "DENY")
+        resp.headers.setdefault("Referrer-Policy", "no-referrer")
+        resp.headers.setdefault("Content-Security-Policy", "default-src 'self'; script-src 'self' 'unsafe-inline'; style-src 'self' 'unsafe-inline'")
+        return resp
+
+
+class HeaderSizeMiddleware(BaseHTTPMiddleware):
+    """reference: middleware HeaderSizeMiddleware"""
+
+    def __init__(self, app, max_bytes: int = 16384):
+        super().__init__(app)
+        self.max_bytes = max_bytes
+
+    async def dispatch(self, request, call_next):
+        total = sum(len(k) + len(v) for k, v in request.headers.raw)
+        if total > self.max_bytes:
+            return JSONResponse({"detail": "headers too large"}, status_code=431)
+        return await call_next(request)
+
+
+class RateLimitMiddleware(BaseHTTPMiddleware):
+    """Token-bucket per client IP (reference: middleware/rate_limit_middleware.py)."""
+
+    def __init__(self, app, rpm: int, burst: int):
+        super().__init__(app)
+        self.rpm = rpm
+        self.burst = burst
+        self.buckets: Dict[str, TokenBucket] = {}
+
+    async def dispatch(self, request, call_next):
+        client = request.client.host if request.client else "?"
+        bucket = self.buckets.get(client)
+        if bucket is None:
+            bucket = self.buckets.setdefault(client, TokenBucket(self.rpm, self.burst))
+        if not bucket.allow():
+            return JSONResponse({"detail": "rate limit exceeded"}, status_code=429)
+        return await call_next(request)
+
+
+class CorrelationIDMiddleware(BaseHTTPMiddleware):
+    """reference: CorrelationIDMiddleware — propagate/emit X-Correlation-ID."""
+
+    async def dispatch(self, request, call_next):
+        cid = request.headers.get("x-correlation-id") or uuid.uuid4().hex[:16]
+        resp = await call_next(request)
+        resp.headers["X-Correlation-ID"] = cid
+        return resp
+
+
+class BodyLimitMiddleware(BaseHTTPMiddleware):
+    """reference: ValidationMiddleware body-size guard."""
+
+    def __init__(self, app, max_bytes: int):
+        super().__init__(app)
+        self.max_bytes = max_bytes
+
+    async def dispatch(self, request, call_next):
+        cl = request.headers.get("content-length")
+        if cl and cl.isdigit() and int(cl) > self.max_bytes:
+            return JSONResponse({"detail": "request body too large"}, status_code=413)
+        return await call_next(request)
+
+
+# ---------------------------------------------------------------------------
+
+
+def build_app(engine: GatewayEngine, auth: Optional[AuthService] = None) -> FastAPI:
+    settings = engine.settings
+    auth = auth or AuthService(engine.db, settings)
+    auth.bootstrap_admin()
+
+    from contextlib import asynccontextmanager
+
+    @asynccontextmanager
+    async def lifespan(app_: FastAPI):
+        # reference: main.py lifespan (:1443) — service startup, GPU attach
+        await engine.startup()
+        if engine.enable_gpu():
+            app_.state.collector = BatchCollector(engine.process_rpc_batch,
+                                                  max_batch=settings.gpu_batch_max_requests,
+                                                  window_us=settings.gpu_batch_window_us)
+        yield
+        await engine.shutdown()
+
+    app = FastAPI(title=settings.app_name, version=settings.version, lifespan=lifespan,
+                  docs_url="/docs" if settings.docs_enabled else None, redoc_url=None)
+    app.state.engine = engine
+    app.state.auth = auth
+    app.state.collector = None
+    app.state.started = time.time()
+
+    # middleware registration — innermost added first (starlette wraps outward);
+    # effective order matches reference §2.4 (CORS outermost → ... → handlers)
+    app.add_middleware(CorrelationIDMiddleware)
+    app.add_middleware(BodyLimitMiddleware, max_bytes=settings.max_request_body_bytes)
+    if settings.rate_limit_enabled:
+        app.add_middleware(RateLimitMiddleware, rpm=settings.rate_limit_requests_per_minute,
+                           burst=settings.rate_limit_burst)
+    app.add_middleware(HeaderSizeMiddleware, max_bytes=settings.max_header_bytes)
+    if settings.security_headers_enabled:
+        app.add_middleware(SecurityHeadersMiddleware)
+    app.add_middleware(CORSMiddleware, allow_origins=settings.cors_allow_origins,
+                       allow_methods=["*"], allow_headers=["*"])
+
+    # -- auth dependency -------------------------------------------------------
+    async def get_auth(request: Request) -> AuthContext:
+        try:
+            return auth.authenticate(request.headers.get("authorization"))
+        except AuthError as exc:
+            raise HTTPException(status_code=exc.status, detail=str(exc),
+                                headers={"WWW-Authenticate": "Bearer"}) from exc
+
+    def require(permission: str):
+        async def dep(request: Request, ctx: AuthContext = Depends(get_auth)) -> AuthContext:
+            try:
+                auth.require_permission(ctx, permission)
+            except PermissionError_ as exc:
+                raise HTTPException(status_code=403, detail=str(exc)) from exc
+            return ctx
+
+        return dep
+
+    async def rpc_bytes(raw: bytes, ctx: AuthContext, server_id: Optional[str] = None) -> Optional[bytes]:
+        sid = server_id or ctx.server_id  # token-scoped server (reference: token_scoping.py)
+        if app.state.collector is not None and server_id is None and ctx.server_id is None:
+            return await app.state.collector.submit(raw)
+        return await engine.handle_rpc_bytes(raw, user=ctx.user, server_id=sid)
+
+    # -- health / version / metrics (reference: main.py /health /version, prometheus) --
+    @app.get("/health")
+    async def health():
+        return engine.health()
+
+    @app.get("/healthz")
+    async def healthz():
+        return {"status": "ok"}
+
+    @app.get("/ready")
+    async def ready():
+        return {"ready": True}
+
+    @app.get("/version")
+    async def version(ctx: AuthContext = Depends(get_auth)):
+        return engine.version_info()
+
+    @app.get("/metrics")
+    async def metrics():
+        snap = engine.metrics.snapshot()
+        lines = ["# TYPE mcpgateway_tool_invocations_total counter"]
+        for k, v in snap["counters"].items():
+            lines.append(f"mcpgateway_{k} {v}")
+        if engine.gpu_pipeline is not None:
+            st = engine.gpu_pipeline.stats()
+            for k in ("requests", "fast_path", "slow_path", "blocked", "cache_hits"):
+                lines.append(f"mcpgateway_gpu_{k} {st[k]}")
+        return PlainTextResponse("\n".join(lines) + "\n")
+
+    # -- protocol endpoints (reference: main.py:3962-4082) -----------------------
+    @app.post("/protocol/initialize")
+    async def protocol_initialize(request: Request, ctx: AuthContext = Depends(get_auth)):
+        body = await request.json()
+        return initialize_result(body.get("protocolVersion") if isinstance(body, dict) else None)
+
+    @app.post("/protocol/ping")
+    async def protocol_ping(request: Request, ctx: AuthContext = Depends(get_auth)):
+        body = await request.json()
+        rid = body.get("id") if isinstance(body, dict) else None
+        return {"jsonrpc": "2.0", "id": rid, "result": {}}
+
+    # -- JSON-RPC endpoint (THE hot path; reference: main.py:7896) ----------------
+    @app.post("/rpc")
+    async def handle_rpc(request: Request, ctx: AuthContext = Depends(get_auth)):
+        raw = await request.body()
+        out = await rpc_bytes(raw, ctx)
+        if out is None:
+            return Response(status_code=202)
+        return Response(content=out, media_type="application/json")
+
+    # -- streamable HTTP /mcp (reference: transports/streamablehttp_transport.py) --
+    async def mcp_post(request: Request, ctx: AuthContext, server_id: Optional[str] = None):
+        raw = await request.body()
+        session_id = request.headers.get("mcp-session-id")
+        sess = engine.sessions.get(session_id) if session_id else None
+        headers = {}
+        try:
+            req = jsonrpc.parse_request_bytes(raw)
+        except jsonrpc.JSONRPCError as exc:
+            return Response(content=jsonrpc.JSONRPCResponse(id=None, error=exc).to_bytes(),
+                            media_type="application/json")
+        if req.method == "initialize":
+            sess = engine.sessions.create(transport="streamablehttp", server_id=server_id, user=ctx.user)
+            headers["mcp-session-id"] = sess.session_id
+        elif sess is None and session_id:
+            return JSONResponse({"detail": "session not found"}, status_code=404)
+        resp = await engine.handle_rpc(req, user=ctx.user, server_id=server_id or ctx.server_id, session=sess)
+        if resp is None:
+            return Response(status_code=202, headers=headers)
+        return Response(content=resp.to_bytes(), media_type="application/json", headers=headers)
+
+    @app.post("/mcp")
+    async def mcp_endpoint(request: Request, ctx: AuthContext = Depends(get_auth)):
+        return await mcp_post(request, ctx)
+
+    @app.post("/servers/{server_id}/mcp")
+    async def mcp_server_endpoint(server_id: str, request: Request, ctx: AuthContext = Depends(get_auth)):
+        return await mcp_post(request, ctx, server_id=server_id)
+
+    @app.get("/mcp")
+    async def mcp_get(request: Request, ctx: AuthContext = Depends(get_auth)):
+        session_id = request.headers.get("mcp-session-id")
+        sess = engine.sessions.get(session_id) if session_id else None
+        if sess is None:
+            return JSONResponse({"detail": "session not found"}, status_code=404)
+        last_event_id = request.headers.get("last-event-id")
+
+        async def stream():
+            # Last-Event-ID replay (reference: InMemoryEventStore.replay_events_after :615)
+            for ev in engine.sessions.event_store.replay_after(sess.session_id, last_event_id):
+                yield f"id: {ev.event_id}\nevent: message\ndata: {json.dumps(ev.message)}\n\n"
+            async for eid, message in engine.sessions.respond_stream(sess.session_id,
+                                                                     keepalive_s=engine.settings.sse_keepalive_interval):
+                if message.get("type") == "keepalive":
+                    yield ": keepalive\n\n"
+                else:
+                    prefix = f"id: {eid}\n" if eid else ""
+                    yield f"{prefix}event: message\ndata: {json.dumps(message)}\n\n"
+
+        return StreamingResponse(stream(), media_type="text/event-stream")
+
+    @app.delete("/mcp")
+    async def mcp_delete(request: Request, ctx: AuthContext = Depends(get_auth)):
+        session_id = request.headers.get("mcp-session-id")
+        if session_id:
+            engine.sessions.remove(session_id)
+        return Response(status_code=204)
+
+    # -- legacy SSE pair (reference: main.py:4478 sse_endpoint / :4597 message_endpoint) --
+    @app.get("/servers/{server_id}/sse")
+    async def sse_endpoint(server_id: str, request: Request, ctx: AuthContext = Depends(get_auth)):
+        sess = engine.sessions.create(transport="sse", server_id=server_id, user=ctx.user)
+        endpoint_url = f"/servers/{server_id}/message?session_id={sess.session_id}"
+
+        async def stream():
+            yield f"event: endpoint\ndata: {endpoint_url}\n\n"
+            async for eid, message in engine.sessions.respond_stream(sess.session_id,
+                                                                     keepalive_s=engine.settings.sse_keepalive_interval):
+                if message.get("type") == "keepalive":
+                    yield ": keepalive\n\n"
+                else:
+                    yield f"event: message\ndata: {json.dumps(message)}\n\n"
+
+        return StreamingResponse(stream(), media_type="text/event-stream")
+
+    @app.post("/servers/{server_id}/message")
+    async def message_endpoint(server_id: str, request: Request, session_id: str = Query(...),
+                               ctx: AuthContext = Depends(get_auth)):
+        raw = await request.body()
+        sess = engine.sessions.get(session_id)
+        if sess is None:
+            return JSONResponse({"detail": "session not found"}, status_code=404)
+        out = await engine.handle_rpc_bytes(raw, user=ctx.user, server_id=server_id, session=sess)
+        if out is not None:
+            await engine.sessions.broadcast(session_id, json.loads(out))
+        return JSONResponse({"status": "accepted"}, status_code=202)
+
+    # -- WebSocket /ws (reference: main.py:11866) ---------------------------------
+    @app.websocket("/ws")
+    async def websocket_endpoint(ws: WebSocket):
+        try:
+            ctx = auth.authenticate(ws.headers.get("authorization"))
+        except AuthError:
+            await ws.close(code=1008)
+            return
+        await ws.accept()
+        sess = engine.sessions.create(transport="websocket", user=ctx.user)
+        try:
+            while True:
+                raw = await ws.receive_text()
+                out = await engine.handle_rpc_bytes(raw.encode(), user=ctx.user, session=sess)
+                if out is not None:
+                    await ws.send_text(out.decode())
+        except WebSocketDisconnect:
+            pass
+        finally:
+            engine.sessions.remove(sess.session_id)
+
+    # -- entity CRUD (reference: main.py:3557 routers) ----------------------------
+    ENTITY_ROUTES = {
+        "tools": "tool", "gateways": "gateway", "servers": "server",
+        "resources": "resource", "prompts": "prompt", "a2a": "a2a_agent",
+    }
+
+    def _register_crud(plural: str, kind: str) -> None:
+        perm = {"a2a": "tools"}.get(plural, plural)
+
+        @app.get(f"/{plural}", name=f"list_{plural}")
+        async def list_entities(include_inactive: bool = False,
+                                ctx: AuthContext = Depends(require(f"{perm}.read"))):
+            return engine.registry.list(kind, include_disabled=include_inactive)
+
+        @app.get(f"/{plural}/{{entity_id}}", name=f"get_{plural}")
+        async def get_entity(entity_id: str, ctx: AuthContext = Depends(require(f"{perm}.read"))):
+            try:
+                return engine.registry.get(kind, entity_id)
+            except NotFoundError as exc:
+                raise HTTPException(404, str(exc)) from exc
+
+        @app.post(f"/{plural}", status_code=201, name=f"create_{plural}")
+        async def create_entity(request: Request, ctx: AuthContext = Depends(require(f"{perm}.create"))):
+            body = await request.json()
+            try:
+                if kind == "gateway":
+                    return await engine.gateway_service.register_gateway(
+                        name=body["name"], url=body.get("url", ""),
+                        transport=body.get("transport", "streamablehttp"),
+                        description=body.get("description", ""),
+                        auth_type=body.get("auth_type"), auth_value=body.get("auth_value"),
+                        tags=body.get("tags"))
+                if kind == "tool":
+                    body.setdefault("original_name", body.get("name"))
+                    body.setdefault("integration_type", "REST" if body.get("url") else "LOCAL")
+                return engine.registry.create(kind, **body)
+            except ConflictError as exc:
+                raise HTTPException(409, str(exc)) from exc
+            except GatewayConnectionError as exc:
+                raise HTTPException(502, str(exc)) from exc
+            except (TypeError, KeyError) as exc:
+                raise HTTPException(422, f"invalid fields: {exc}") from exc
+
+        @app.put(f"/{plural}/{{entity_id}}", name=f"update_{plural}")
+        async def update_entity(entity_id: str, request: Request,
+                                ctx: AuthContext = Depends(require(f"{perm}.update"))):
+            body = await request.json()
+            try:
+                return engine.registry.update(kind, entity_id, **body)
+            except NotFoundError as exc:
+                raise HTTPException(404, str(exc)) from exc
+            except TypeError as exc:
+                raise HTTPException(422, str(exc)) from exc
+
+        @app.delete(f"/{plural}/{{entity_id}}", status_code=204, name=f"delete_{plural}")
+        async def delete_entity(entity_id: str, ctx: AuthContext = Depends(require(f"{perm}.delete"))):
+            try:
+                if kind == "gateway":
+                    await engine.gateway_service.delete_gateway(entity_id)
+                else:
+                    engine.registry.delete(kind, entity_id)
+            except NotFoundError as exc:
+                raise HTTPException(404, str(exc)) from exc
+            return Response(status_code=204)
+
+        @app.post(f"/{plural}/{{entity_id}}/toggle", name=f"toggle_{plural}")
+        async def toggle_entity(entity_id: str, activate: bool = True,
+                                ctx: AuthContext = Depends(require(f"{perm}.update"))):
+            try:
+                return engine.registry.set_enabled(kind, entity_id, activate)
+            except NotFoundError as exc:
+                raise HTTPException(404, str(exc)) from exc
+
+    for plural, kind in ENTITY_ROUTES.items():
+        _register_crud(plural, kind)
+
+    @app.post("/gateways/{gateway_id}/refresh")
+    async def refresh_gateway(gateway_id: str, ctx: AuthContext = Depends(require("gateways.update"))):
+        try:
+            return await engine.gateway_service.refresh_gateway(gateway_id)
+        except NotFoundError as exc:
+            raise HTTPException(404, str(exc)) from exc
+        except Exception as exc:
+            raise HTTPException(502, str(exc)) from exc
+
+    # -- export / import (reference: services/export_service.py:268) --------------
+    @app.get("/export")
+    async def export_config(ctx: AuthContext = Depends(require("admin.export"))):
+        return engine.registry.export_configuration()
+
+    @app.post("/import")
+    async def import_config(request: Request, conflict_strategy: str = "update",
+                            ctx: AuthContext = Depends(require("admin.import"))):
+        body = await request.json()
+        return engine.registry.import_configuration(body, conflict_strategy)
+
+    # -- auth endpoints (reference: routers/auth.py, routers/tokens.py) ------------
+    @app.post("/auth/login")
+    async def login(request: Request):
+        body = await request.json()
+        ctx = auth.verify_user(body.get("email", ""), body.get("password", ""))
+        if ctx is None:
+            raise HTTPException(401, "invalid credentials")
+        from ..auth import jwt as jwt_mod
+
+        token = jwt_mod.create_token({"sub": ctx.user, "admin": ctx.is_admin}, settings.jwt_secret_key,
+                                     expires_minutes=settings.token_expiry,
+                                     audience=settings.jwt_audience, issuer=settings.jwt_issuer)
+        return {"access_token": token, "token_type": "bearer"}
+
+    @app.post("/tokens", status_code=201)
+    async def create_token_ep(request: Request, ctx: AuthContext = Depends(get_auth)):
+        body = await request.json()
+        raw = auth.create_api_token(ctx.user, body.get("name", "token"), scopes=body.get("scopes"),
+                                    server_id=body.get("server_id"),
+                                    expires_minutes=body.get("expires_minutes"))
+        return {"token": raw}
+
+    @app.get("/tokens")
+    async def list_tokens_ep(ctx: AuthContext = Depends(get_auth)):
+        return auth.list_api_tokens(ctx.user)
+
+    @app.delete("/tokens/{token_id}", status_code=204)
+    async def revoke_token_ep(token_id: str, ctx: AuthContext = Depends(get_auth)):
+        auth.revoke_api_token(token_id)
+        return Response(status_code=204)
+
+    # -- well-known (reference: routers/well_known.py RFC 9728) --------------------
+    @app.get("/.well-known/oauth-protected-resource")
+    async def oauth_protected_resource(request: Request):
+        base = str(request.base_url).rstrip("/")
+        return {"resource": f"{base}/mcp", "authorization_servers": [],
+                "bearer_methods_supported": ["header"], "resource_name": settings.app_name}
+
+    @app.get("/.well-known/security.txt")
+    async def security_txt():
+        return PlainTextResponse("Contact: mailto:security@example.com\n")
+
+    # -- admin API (reference: mcpgateway/admin.py, trimmed) -----------------------
+    @app.get("/admin/stats")
+    async def admin_stats(ctx: AuthContext = Depends(require("admin.read"))):
+        out = {
+            "uptime_s": round(time.time() - app.state.started, 1),
+            "entities": {k: len(engine.registry.list(k)) for k in
+                         ("tool", "gateway", "server", "resource", "prompt", "a2a_agent")},
+            "sessions": engine.sessions.count(),
+            "metrics": engine.metrics.snapshot(),
+            "plugins": engine.plugins.stats(),
+        }
+        if engine.gpu_pipeline is not None:
+            out["gpu"] = engine.gpu_pipeline.stats()
+        return out
+
+    @app.get("/admin/plugins")
+    async def admin_plugins(ctx: AuthContext = Depends(require("admin.read"))):
+        return [{"name": p.name, "mode": p.mode.value, "priority": p.priority,
+                 "hooks": [h.value for h in p.hooks], "gpu_capable": p.gpu_capable}
+                for p in engine.plugins.plugins]
+
+    @app.post("/admin/plugins/{name}/mode")
+    async def admin_plugin_mode(name: str, mode: str, ctx: AuthContext = Depends(require("admin.update"))):
+        from ..plugins.framework import PluginMode
+
+        p = engine.plugins.get(name)
+        if p is None:
+            raise HTTPException(404, f"plugin {name} not found")
+        p.mode = PluginMode(mode)
+        return {"name": name, "mode": p.mode.value}
+
+    @app.get("/admin", response_class=HTMLResponse)
+    async def admin_ui(ctx: AuthContext = Depends(require("admin.read"))):
+        from ..admin.ui import render_admin_page
+
+        return render_admin_page(engine)
+
+    return app
+
+
+def serve(engine: Optional[GatewayEngine] = None, **uvicorn_kwargs) -> None:
+    """CLI entrypoint (reference: mcpgateway/cli.py uvicorn wrapper)."""
+    import uvicorn
+
+    engine = engine or GatewayEngine()
+    app = build_app(engine)
+    settings = engine.settings
+    uvicorn.run(app, host=settings.host, port=settings.port, **uvicorn_kwargs)

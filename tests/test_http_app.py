@@ -1,0 +1,327 @@
+"""HTTP edge tests (reference analog: tests/e2e/test_main_apis.py, unit
+transports tests) — in-process via httpx ASGITransport, no sockets."""
+
+import asyncio
+import base64
+import json
+
+import httpx
+import pytest
+
+from mcp_context_forge_amd.auth.service import AuthService
+from mcp_context_forge_amd.config import Settings
+from mcp_context_forge_amd.engine import GatewayEngine
+from mcp_context_forge_amd.transports.http_app import build_app
+
+ADMIN = {"Authorization": "Basic " + base64.b64encode(b"admin:changeme").decode()}
+
+
+@pytest.fixture()
+def client_engine():
+    from contextlib import asynccontextmanager
+
+    s = Settings(database_url="sqlite://", federation_enabled=False, auth_required=True,
+                 plugins_enabled=True, gpu_enabled=False)
+    engine = GatewayEngine(s)
+
+    async def echo(args):
+        return args
+
+    engine.tool_service.register_local_tool("echo", echo, "Echo tool")
+    app = build_app(engine)
+    transport = httpx.ASGITransport(app=app)
+
+    @asynccontextmanager
+    async def client_factory():
+        async with app.router.lifespan_context(app):
+            async with httpx.AsyncClient(transport=transport, base_url="http://gw") as c:
+                yield c
+
+    yield client_factory, engine, app
+
+
+def test_health_and_version(client_engine, run):
+    client_factory, engine, app = client_engine
+
+    async def go():
+        async with client_factory() as c:
+            r = await c.get("/health")
+            assert r.status_code == 200 and r.json()["status"] == "healthy"
+            r = await c.get("/version")
+            assert r.status_code == 401  # auth required
+            r = await c.get("/version", headers=ADMIN)
+            assert r.json()["name"] == "mcp-context-forge-amd"
+
+    run(go())
+
+
+def test_rpc_roundtrip_and_auth(client_engine, run):
+    client_factory, engine, app = client_engine
+
+    async def go():
+        async with client_factory() as c:
+            body = {"jsonrpc": "2.0", "id": 1, "method": "tools/call",
+                    "params": {"name": "echo", "arguments": {"x": 1}}}
+            r = await c.post("/rpc", json=body)
+            assert r.status_code == 401
+            r = await c.post("/rpc", json=body, headers=ADMIN)
+            assert r.status_code == 200
+            assert r.json()["result"]["structuredContent"] == {"x": 1}
+            # parse error taxonomy through HTTP
+            r = await c.post("/rpc", content=b"{bad", headers=ADMIN)
+            assert r.json()["error"]["code"] == -32700
+            # notification → 202
+            r = await c.post("/rpc", json={"jsonrpc": "2.0", "method": "notifications/initialized"}, headers=ADMIN)
+            assert r.status_code == 202
+
+    run(go())
+
+
+def test_jwt_login_flow(client_engine, run):
+    client_factory, engine, app = client_engine
+
+    async def go():
+        async with client_factory() as c:
+            r = await c.post("/auth/login", json={"email": "admin@example.com", "password": "changeme"})
+            assert r.status_code == 200
+            token = r.json()["access_token"]
+            r = await c.get("/version", headers={"Authorization": f"Bearer {token}"})
+            assert r.status_code == 200
+            r = await c.get("/version", headers={"Authorization": "Bearer bogus.token.sig"})
+            assert r.status_code == 401
+
+    run(go())
+
+
+def test_api_token_lifecycle(client_engine, run):
+    client_factory, engine, app = client_engine
+
+    async def go():
+        async with client_factory() as c:
+            r = await c.post("/tokens", json={"name": "ci"}, headers=ADMIN)
+            tok = r.json()["token"]
+            assert tok.startswith("mcpg_")
+            r = await c.get("/tools", headers={"Authorization": f"Bearer {tok}"})
+            assert r.status_code == 200
+            r = await c.get("/tokens", headers=ADMIN)
+            tid = r.json()[0]["id"]
+            r = await c.delete(f"/tokens/{tid}", headers=ADMIN)
+            assert r.status_code == 204
+            r = await c.get("/tools", headers={"Authorization": f"Bearer {tok}"})
+            assert r.status_code == 401
+
+    run(go())
+
+
+def test_crud_tools_and_servers(client_engine, run):
+    client_factory, engine, app = client_engine
+
+    async def go():
+        async with client_factory() as c:
+            r = await c.post("/tools", headers=ADMIN,
+                             json={"name": "weather", "url": "http://up/api", "request_type": "GET",
+                                   "description": "REST tool"})
+            assert r.status_code == 201, r.text
+            tool = r.json()
+            assert tool["integration_type"] == "REST"
+            r = await c.post("/tools", headers=ADMIN, json={"name": "weather"})
+            assert r.status_code == 409
+            r = await c.get(f"/tools/{tool['id']}", headers=ADMIN)
+            assert r.json()["name"] == "weather"
+            r = await c.post(f"/tools/{tool['id']}/toggle?activate=false", headers=ADMIN)
+            assert r.json()["enabled"] is False
+            r = await c.post("/servers", headers=ADMIN,
+                             json={"name": "virt1", "associated_tools": [tool["id"]]})
+            assert r.status_code == 201
+            r = await c.delete(f"/tools/{tool['id']}", headers=ADMIN)
+            assert r.status_code == 204
+            r = await c.get(f"/tools/{tool['id']}", headers=ADMIN)
+            assert r.status_code == 404
+
+    run(go())
+
+
+def test_rbac_denies_nonadmin(client_engine, run):
+    client_factory, engine, app = client_engine
+
+    async def go():
+        auth: AuthService = app.state.auth
+        auth.create_user("dev@x.com", "pw12345")
+        auth.assign_role("dev@x.com", "viewer")
+        async with client_factory() as c:
+            r = await c.post("/auth/login", json={"email": "dev@x.com", "password": "pw12345"})
+            token = r.json()["access_token"]
+            hdr = {"Authorization": f"Bearer {token}"}
+            r = await c.get("/tools", headers=hdr)
+            assert r.status_code == 200
+            r = await c.post("/tools", headers=hdr, json={"name": "nope"})
+            assert r.status_code == 403
+            r = await c.get("/admin/stats", headers=hdr)
+            assert r.status_code == 403
+
+    run(go())
+
+
+def test_mcp_streamable_session_flow(client_engine, run):
+    client_factory, engine, app = client_engine
+
+    async def go():
+        async with client_factory() as c:
+            init = {"jsonrpc": "2.0", "id": 1, "method": "initialize",
+                    "params": {"protocolVersion": "2025-11-25", "capabilities": {}}}
+            r = await c.post("/mcp", json=init, headers=ADMIN)
+            assert r.status_code == 200
+            sid = r.headers["mcp-session-id"]
+            assert r.json()["result"]["protocolVersion"] == "2025-11-25"
+            hdr = {**ADMIN, "mcp-session-id": sid}
+            r = await c.post("/mcp", headers=hdr,
+                             json={"jsonrpc": "2.0", "id": 2, "method": "tools/list"})
+            assert any(t["name"] == "echo" for t in r.json()["result"]["tools"])
+            r = await c.post("/mcp", headers=hdr,
+                             json={"jsonrpc": "2.0", "id": 3, "method": "tools/call",
+                                   "params": {"name": "echo", "arguments": {"v": 7}}})
+            assert r.json()["result"]["structuredContent"] == {"v": 7}
+            # unknown session
+            r = await c.post("/mcp", headers={**ADMIN, "mcp-session-id": "nope"},
+                             json={"jsonrpc": "2.0", "id": 4, "method": "ping"})
+            assert r.status_code == 404
+            r = await c.request("DELETE", "/mcp", headers=hdr)
+            assert r.status_code == 204
+            assert engine.sessions.get(sid) is None
+
+    run(go())
+
+
+def test_event_store_replay(client_engine, run):
+    client_factory, engine, app = client_engine
+
+    async def go():
+        sess = engine.sessions.create(transport="streamablehttp")
+        eid1 = engine.sessions.event_store.store(sess.session_id, {"n": 1})
+        eid2 = engine.sessions.event_store.store(sess.session_id, {"n": 2})
+        engine.sessions.event_store.store(sess.session_id, {"n": 3})
+        evs = engine.sessions.event_store.replay_after(sess.session_id, eid2)
+        assert [e.message["n"] for e in evs] == [3]
+        evs = engine.sessions.event_store.replay_after(sess.session_id, None)
+        assert [e.message["n"] for e in evs] == [1, 2, 3]
+
+    run(go())
+
+
+def test_sse_pair_broadcast(client_engine, run):
+    """POST /servers/{id}/message routes through the session registry broadcast
+    (the streaming socket side is covered in tests/test_live_http.py — httpx's
+    ASGITransport buffers responses so infinite SSE can't be consumed in-proc)."""
+    client_factory, engine, app = client_engine
+
+    async def go():
+        srv = engine.registry.create("server", name="s-sse")
+        sess = engine.sessions.create(transport="sse", server_id=srv["id"])
+        async with client_factory() as c:
+            r = await c.post(f"/servers/{srv['id']}/message?session_id={sess.session_id}",
+                             headers=ADMIN, json={"jsonrpc": "2.0", "id": 9, "method": "ping"})
+            assert r.status_code == 202
+            msg = sess.queue.get_nowait()
+            assert msg["id"] == 9 and msg["result"] == {}
+            # event store captured it for replay
+            evs = engine.sessions.event_store.replay_after(sess.session_id, None)
+            assert evs and evs[-1].message["id"] == 9
+            r = await c.post(f"/servers/{srv['id']}/message?session_id=missing",
+                             headers=ADMIN, json={"jsonrpc": "2.0", "id": 1, "method": "ping"})
+            assert r.status_code == 404
+
+    run(go())
+
+
+def test_export_import_endpoints(client_engine, run):
+    client_factory, engine, app = client_engine
+
+    async def go():
+        async with client_factory() as c:
+            r = await c.get("/export", headers=ADMIN)
+            payload = r.json()
+            assert "entities" in payload
+            r = await c.post("/import", headers=ADMIN, json=payload)
+            assert r.json()["skipped"] >= 0
+
+    run(go())
+
+
+def test_metrics_and_wellknown(client_engine, run):
+    client_factory, engine, app = client_engine
+
+    async def go():
+        async with client_factory() as c:
+            await c.post("/rpc", headers=ADMIN,
+                         json={"jsonrpc": "2.0", "id": 1, "method": "tools/call",
+                               "params": {"name": "echo", "arguments": {}}})
+            r = await c.get("/metrics")
+            assert "mcpgateway_tool_invocations_total" in r.text
+            r = await c.get("/.well-known/oauth-protected-resource")
+            assert r.json()["resource"].endswith("/mcp")
+
+    run(go())
+
+
+def test_admin_stats_and_ui(client_engine, run):
+    client_factory, engine, app = client_engine
+
+    async def go():
+        async with client_factory() as c:
+            r = await c.get("/admin/stats", headers=ADMIN)
+            assert r.json()["entities"]["tool"] >= 1
+            r = await c.get("/admin", headers=ADMIN)
+            assert "MCP Context Forge AMD" in r.text
+            r = await c.get("/admin/plugins", headers=ADMIN)
+            names = [p["name"] for p in r.json()]
+            assert "deny_filter" in names
+            r = await c.post("/admin/plugins/deny_filter/mode?mode=permissive", headers=ADMIN)
+            assert r.json()["mode"] == "permissive"
+
+    run(go())
+
+
+def test_protocol_endpoints(client_engine, run):
+    client_factory, engine, app = client_engine
+
+    async def go():
+        async with client_factory() as c:
+            r = await c.post("/protocol/initialize", headers=ADMIN, json={"protocolVersion": "2025-11-25"})
+            assert r.json()["protocolVersion"] == "2025-11-25"
+            r = await c.post("/protocol/ping", headers=ADMIN, json={"id": 5})
+            assert r.json() == {"jsonrpc": "2.0", "id": 5, "result": {}}
+
+    run(go())
+
+
+def test_body_limit(client_engine, run):
+    client_factory, engine, app = client_engine
+
+    async def go():
+        async with client_factory() as c:
+            r = await c.post("/rpc", headers={**ADMIN, "content-length": str(100 << 20)},
+                             content=b"")
+            assert r.status_code == 413
+
+    run(go())
+
+
+def test_token_scoped_server(client_engine, run):
+    client_factory, engine, app = client_engine
+
+    async def go():
+        # a token scoped to a virtual server restricts tools/list (reference: token_scoping)
+        async def h(args):
+            return args
+
+        t1 = engine.tool_service.register_local_tool("scoped_tool", h)
+        srv = engine.registry.create("server", name="scoped", associated_tools=[t1["id"]])
+        auth: AuthService = app.state.auth
+        raw = auth.create_api_token("admin@example.com", "scoped", server_id=srv["id"])
+        async with client_factory() as c:
+            r = await c.post("/rpc", headers={"Authorization": f"Bearer {raw}"},
+                             json={"jsonrpc": "2.0", "id": 1, "method": "tools/list"})
+            tools = r.json()["result"]["tools"]
+            assert [t["name"] for t in tools] == ["scoped_tool"]
+
+    run(go())
