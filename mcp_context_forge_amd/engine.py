@@ -26,8 +26,10 @@ from .plugins.loader import default_chain_specs, load_plugin_manager
 from .protocol import jsonrpc
 from .protocol.mcp import PROTOCOL_VERSION, initialize_result
 from .registry.registry import NotFoundError, Registry
+from .services.a2a_service import A2AService
 from .services.content import CompletionService, PromptService, ResourceService, RootService
 from .services.gateway_service import GatewayService
+from .services.llm_proxy import LLMProxyError, LLMProxyService
 from .services.metrics import MetricsBuffer
 from .services.sessions import SessionRegistry
 from .services.tool_service import ToolInvocationError, ToolNotFoundError, ToolService
@@ -60,10 +62,13 @@ class GatewayEngine:
         self.tool_service = ToolService(self.registry, self.plugins, self.metrics,
                                         max_retries=self.settings.max_tool_retries)
         self.gateway_service = GatewayService(self.registry, self.tool_service, self.settings, rank, world_size)
+        self.a2a_service = A2AService(self.registry, self.plugins)
+        self.tool_service.a2a_service = self.a2a_service
         self.prompt_service = PromptService(self.registry, self.plugins)
         self.resource_service = ResourceService(self.registry, self.plugins)
         self.completion_service = CompletionService(self.registry)
         self.root_service = RootService()
+        self.llm_proxy = LLMProxyService()
         self.sessions = SessionRegistry(self.settings.session_ttl, rank)
         self.started_at = time.time()
         self.gpu_pipeline = None  # attached lazily by gpu.pipeline when available
@@ -76,6 +81,8 @@ class GatewayEngine:
 
     async def shutdown(self) -> None:
         await self.gateway_service.stop()
+        await self.a2a_service.aclose()
+        await self.llm_proxy.aclose()
         await self.tool_service.aclose()
         await self.plugins.shutdown()
         self.metrics.flush()
@@ -170,9 +177,31 @@ class GatewayEngine:
                 self._log_level = params.get("level", "info")
                 result = {}
             elif method == "sampling/createMessage":
-                # reference: handlers/sampling.py — gateway has no model client here;
-                # sampling is forwarded to the LLM proxy when configured.
-                raise jsonrpc.JSONRPCError(jsonrpc.METHOD_NOT_FOUND, "sampling requires a configured LLM provider")
+                # reference: handlers/sampling.py — forwarded to the LLM proxy
+                if not self.llm_proxy.registry.providers:
+                    raise jsonrpc.JSONRPCError(jsonrpc.METHOD_NOT_FOUND, "sampling requires a configured LLM provider")
+                messages = [
+                    {"role": m.get("role", "user"),
+                     "content": (m.get("content") or {}).get("text", "") if isinstance(m.get("content"), dict) else str(m.get("content", ""))}
+                    for m in params.get("messages", [])
+                ]
+                if params.get("systemPrompt"):
+                    messages.insert(0, {"role": "system", "content": params["systemPrompt"]})
+                try:
+                    out = await self.llm_proxy.chat_completions({
+                        "messages": messages,
+                        "max_tokens": params.get("maxTokens", 256),
+                        "temperature": params.get("temperature", 1.0),
+                    })
+                except LLMProxyError as exc:
+                    raise jsonrpc.JSONRPCError(jsonrpc.SERVER_ERROR, str(exc)) from exc
+                choice = (out.get("choices") or [{}])[0]
+                result = {
+                    "role": "assistant",
+                    "content": {"type": "text", "text": (choice.get("message") or {}).get("content", "")},
+                    "model": out.get("model", ""),
+                    "stopReason": choice.get("finish_reason", "endTurn"),
+                }
             else:
                 raise jsonrpc.JSONRPCError(jsonrpc.METHOD_NOT_FOUND, f"Method not found: {method}")
             if req.is_notification:

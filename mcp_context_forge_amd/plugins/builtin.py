@@ -70,7 +70,7 @@ class DenyFilterPlugin(Plugin):
     """Block requests containing deny-listed words (reference: deny.py:118)."""
 
     name = "deny_filter"
-    hooks = (HookType.TOOL_PRE_INVOKE, HookType.PROMPT_PRE_FETCH)
+    hooks = (HookType.TOOL_PRE_INVOKE, HookType.PROMPT_PRE_FETCH, HookType.AGENT_PRE_INVOKE)
     priority = 10
     gpu_capable = True
 
@@ -93,6 +93,7 @@ class DenyFilterPlugin(Plugin):
         return PluginResult.ok()
 
     prompt_pre_fetch = tool_pre_invoke
+    agent_pre_invoke = tool_pre_invoke
 
 
 class RegexFilterPlugin(Plugin):
@@ -152,7 +153,8 @@ class PIIFilterPlugin(Plugin):
     """Detect and mask PII in args and results (reference: cpex-pii-filter)."""
 
     name = "pii_filter"
-    hooks = (HookType.TOOL_PRE_INVOKE, HookType.TOOL_POST_INVOKE, HookType.PROMPT_PRE_FETCH)
+    hooks = (HookType.TOOL_PRE_INVOKE, HookType.TOOL_POST_INVOKE, HookType.PROMPT_PRE_FETCH,
+             HookType.AGENT_PRE_INVOKE, HookType.AGENT_POST_INVOKE)
     priority = 30
     gpu_capable = True
 
@@ -194,6 +196,8 @@ class PIIFilterPlugin(Plugin):
     tool_pre_invoke = _apply
     tool_post_invoke = _apply
     prompt_pre_fetch = _apply
+    agent_pre_invoke = _apply
+    agent_post_invoke = _apply
 
 
 class SchemaGuardPlugin(Plugin):
@@ -321,7 +325,7 @@ class HarmfulContentPlugin(Plugin):
     """Keyword-bank harm detector (reference: harmful_content_detector.py:207)."""
 
     name = "harmful_content_detector"
-    hooks = (HookType.TOOL_PRE_INVOKE, HookType.TOOL_POST_INVOKE)
+    hooks = (HookType.TOOL_PRE_INVOKE, HookType.TOOL_POST_INVOKE, HookType.AGENT_PRE_INVOKE)
     priority = 60
     gpu_capable = True
 
@@ -353,6 +357,7 @@ class HarmfulContentPlugin(Plugin):
         return PluginResult.ok()
 
     tool_post_invoke = tool_pre_invoke
+    agent_pre_invoke = tool_pre_invoke
 
 
 class ResponseCacheByPromptPlugin(Plugin):

@@ -55,6 +55,7 @@ class ToolService:
         self._local_handlers: Dict[str, ToolHandler] = {}
         self._upstreams: Dict[str, UpstreamClient] = {}  # gateway_id -> client
         self._rest_client = None  # lazy httpx.AsyncClient
+        self.a2a_service = None  # wired by the engine (A2A tool dispatch)
 
     # -- wiring ---------------------------------------------------------------
     def register_local_tool(self, name: str, handler: ToolHandler, description: str = "",
@@ -116,7 +117,23 @@ class ToolService:
         if itype == "REST":
             return await self._dispatch_rest(tool, arguments, headers)
         if itype == "A2A":
-            raise ToolInvocationError("A2A tools dispatch via a2a_service", code=jsonrpc.INVALID_REQUEST)
+            # A2A-integrated tool: arguments carry the message (reference: tool_service :6526)
+            if self.a2a_service is None:
+                raise ToolInvocationError("A2A service not wired", code=jsonrpc.SERVER_UNAVAILABLE)
+            message = arguments.get("message") if isinstance(arguments, dict) else None
+            if not isinstance(message, str):
+                message = json.dumps(arguments or {}, default=str)
+            agent_name = tool.get("original_name") or tool["name"]
+            try:
+                res = await self.a2a_service.invoke_agent(agent_name, message)
+            except Exception as exc:
+                code = getattr(exc, "code", jsonrpc.SERVER_ERROR)
+                raise ToolInvocationError(str(exc), code=code) from exc
+            return {
+                "content": [{"type": "text", "text": res.get("response", "")}],
+                "structuredContent": {k: v for k, v in res.items() if k != "raw"},
+                "isError": False,
+            }
         raise ToolInvocationError(f"unsupported integration_type {itype}")
 
     async def _dispatch_rest(self, tool: Dict[str, Any], arguments: Dict[str, Any],

@@ -458,6 +458,49 @@ def build_app(engine: GatewayEngine, auth: Optional[AuthService] = None) -> Fast
         auth.revoke_api_token(token_id)
         return Response(status_code=204)
 
+    # -- A2A invoke (reference: routers a2a + a2a_service.invoke_agent :1997) ------
+    @app.post("/a2a/{agent_name}/invoke")
+    async def a2a_invoke(agent_name: str, request: Request,
+                         ctx: AuthContext = Depends(require("tools.invoke"))):
+        from ..services.a2a_service import A2AError
+
+        body = await request.json()
+        message = body.get("message") if isinstance(body, dict) else None
+        if not isinstance(message, str):
+            message = json.dumps(body, default=str)
+        try:
+            return await engine.a2a_service.invoke_agent(
+                agent_name, message, user=ctx.user, context=body.get("context"),
+                hop_count=int(request.headers.get("x-a2a-hop-count", "0")))
+        except A2AError as exc:
+            status = {-32602: 404, -32003: 403, -32002: 502}.get(exc.code, 502)
+            raise HTTPException(status, str(exc)) from exc
+
+    # -- LLM proxy (reference: llm_proxy_service :103,442; routers/llm_*) ---------
+    @app.post("/v1/chat/completions")
+    async def chat_completions(request: Request, ctx: AuthContext = Depends(get_auth)):
+        from ..services.llm_proxy import LLMProxyError
+
+        body = await request.json()
+        try:
+            if body.get("stream"):
+                stream = engine.llm_proxy.chat_completions_stream(body)
+                return StreamingResponse(stream, media_type="text/event-stream")
+            return await engine.llm_proxy.chat_completions(body)
+        except LLMProxyError as exc:
+            raise HTTPException(exc.status, str(exc)) from exc
+
+    @app.get("/llm/providers")
+    async def list_llm_providers(ctx: AuthContext = Depends(require("admin.read"))):
+        return engine.llm_proxy.registry.list()
+
+    @app.post("/llm/providers", status_code=201)
+    async def add_llm_provider(request: Request, ctx: AuthContext = Depends(require("admin.update"))):
+        body = await request.json()
+        return engine.llm_proxy.registry.register(
+            name=body["name"], base_url=body["base_url"], api_key=body.get("api_key"),
+            models=body.get("models"), default_model=body.get("default_model"))
+
     # -- well-known (reference: routers/well_known.py RFC 9728) --------------------
     @app.get("/.well-known/oauth-protected-resource")
     async def oauth_protected_resource(request: Request):
