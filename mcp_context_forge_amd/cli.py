@@ -1,0 +1,146 @@
+"""Command-line interface.
+
+Reference analogs: mcpgateway/cli.py (uvicorn wrapper), wrapper.py (stdio
+bridge), translate.py (protocol bridge), tools/cli.py (`cforge`).
+
+  python -m mcp_context_forge_amd serve                  # HTTP gateway
+  python -m mcp_context_forge_amd stdio                  # engine as stdio MCP server
+  python -m mcp_context_forge_amd wrapper --url URL      # stdio ↔ remote gateway
+  python -m mcp_context_forge_amd translate --stdio CMD --port P  # stdio server → HTTP
+  python -m mcp_context_forge_amd token --user EMAIL     # mint a JWT
+  python -m mcp_context_forge_amd export / import FILE   # config dump/load
+  python -m mcp_context_forge_amd bench ...              # flagship benchmark
+"""
+
+from __future__ import annotations
+
+import argparse
+import asyncio
+import json
+import shlex
+import sys
+
+
+def main(argv=None) -> int:
+    ap = argparse.ArgumentParser(prog="mcp-context-forge-amd")
+    sub = ap.add_subparsers(dest="cmd", required=True)
+
+    p = sub.add_parser("serve", help="run the HTTP gateway")
+    p.add_argument("--host", default=None)
+    p.add_argument("--port", type=int, default=None)
+    p.add_argument("--workers", type=int, default=1)
+
+    sub.add_parser("stdio", help="run the gateway engine as a stdio MCP server")
+
+    p = sub.add_parser("wrapper", help="stdio bridge to a remote gateway (reference: wrapper.py)")
+    p.add_argument("--url", required=True)
+    p.add_argument("--token", default=None)
+    p.add_argument("--rpc", action="store_true", help="use /rpc instead of /mcp")
+
+    p = sub.add_parser("translate", help="expose a stdio MCP server over HTTP (reference: translate.py)")
+    p.add_argument("--stdio", required=True, help="command to launch, e.g. 'python server.py'")
+    p.add_argument("--host", default="127.0.0.1")
+    p.add_argument("--port", type=int, default=9000)
+
+    p = sub.add_parser("token", help="mint an HS256 JWT for the gateway")
+    p.add_argument("--user", default="admin@example.com")
+    p.add_argument("--admin", action="store_true")
+    p.add_argument("--expires-minutes", type=int, default=10080)
+
+    p = sub.add_parser("export", help="dump registry configuration as JSON")
+    p.add_argument("file", nargs="?", default="-")
+
+    p = sub.add_parser("import", help="load registry configuration from JSON")
+    p.add_argument("file")
+
+    args = ap.parse_args(argv)
+
+    if args.cmd == "serve":
+        from .config import get_settings
+        from .engine import GatewayEngine
+        from .transports.http_app import build_app
+
+        settings = get_settings()
+        if args.host:
+            settings.host = args.host
+        if args.port:
+            settings.port = args.port
+        import uvicorn
+
+        uvicorn.run(build_app(GatewayEngine(settings)), host=settings.host, port=settings.port)
+        return 0
+
+    if args.cmd == "stdio":
+        from .config import get_settings
+        from .engine import GatewayEngine
+        from .transports.stdio import StdioServer
+
+        engine = GatewayEngine(get_settings())
+
+        async def run():
+            await engine.startup()
+            try:
+                await StdioServer(engine).serve()
+            finally:
+                await engine.shutdown()
+
+        asyncio.run(run())
+        return 0
+
+    if args.cmd == "wrapper":
+        from .transports.stdio import GatewayWrapper
+
+        asyncio.run(GatewayWrapper(args.url, token=args.token, use_mcp=not args.rpc).serve())
+        return 0
+
+    if args.cmd == "translate":
+        from .transports.stdio import StdioSubprocess, build_translate_app
+
+        async def run():
+            sp = StdioSubprocess(shlex.split(args.stdio))
+            await sp.start()
+            import uvicorn
+
+            app = build_translate_app(sp)
+            config = uvicorn.Config(app, host=args.host, port=args.port, log_level="info")
+            try:
+                await uvicorn.Server(config).serve()
+            finally:
+                await sp.stop()
+
+        asyncio.run(run())
+        return 0
+
+    if args.cmd == "token":
+        from .auth import jwt as jwt_mod
+        from .config import get_settings
+
+        s = get_settings()
+        tok = jwt_mod.create_token({"sub": args.user, "admin": args.admin}, s.jwt_secret_key,
+                                   expires_minutes=args.expires_minutes,
+                                   audience=s.jwt_audience, issuer=s.jwt_issuer)
+        print(tok)
+        return 0
+
+    if args.cmd in ("export", "import"):
+        from .config import get_settings
+        from .engine import GatewayEngine
+
+        engine = GatewayEngine(get_settings())
+        if args.cmd == "export":
+            payload = json.dumps(engine.registry.export_configuration(), indent=2, default=str)
+            if args.file == "-":
+                print(payload)
+            else:
+                open(args.file, "w").write(payload)
+        else:
+            counts = engine.registry.import_configuration(json.load(open(args.file)))
+            print(json.dumps(counts))
+        asyncio.run(engine.shutdown())
+        return 0
+
+    return 1
+
+
+if __name__ == "__main__":
+    sys.exit(main())

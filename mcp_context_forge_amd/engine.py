@@ -31,6 +31,7 @@ from .services.content import CompletionService, PromptService, ResourceService,
 from .services.gateway_service import GatewayService
 from .services.llm_proxy import LLMProxyError, LLMProxyService
 from .services.metrics import MetricsBuffer
+from .services.observability import AuditTrail, ObservabilityService
 from .services.sessions import SessionRegistry
 from .services.tool_service import ToolInvocationError, ToolNotFoundError, ToolService
 
@@ -69,6 +70,8 @@ class GatewayEngine:
         self.completion_service = CompletionService(self.registry)
         self.root_service = RootService()
         self.llm_proxy = LLMProxyService()
+        self.observability = ObservabilityService(self.db)
+        self.audit = AuditTrail(self.db)
         self.sessions = SessionRegistry(self.settings.session_ttl, rank)
         self.started_at = time.time()
         self.gpu_pipeline = None  # attached lazily by gpu.pipeline when available
@@ -85,6 +88,7 @@ class GatewayEngine:
         await self.llm_proxy.aclose()
         await self.tool_service.aclose()
         await self.plugins.shutdown()
+        self.observability.flush()
         self.metrics.flush()
         self.db.close()
 
@@ -142,9 +146,10 @@ class GatewayEngine:
                 name = params.get("name")
                 if not isinstance(name, str) or not name:
                     raise jsonrpc.JSONRPCError(jsonrpc.INVALID_PARAMS, "missing tool name")
-                result = await self.tool_service.invoke_tool(
-                    name, params.get("arguments") or {}, user=user, server_id=server_id, headers=headers
-                )
+                with self.observability.span("tools/call", tool=name, user=user or ""):
+                    result = await self.tool_service.invoke_tool(
+                        name, params.get("arguments") or {}, user=user, server_id=server_id, headers=headers
+                    )
             elif method == "resources/list":
                 result = {"resources": self.resource_service.list_resources()}
             elif method == "resources/templates/list":

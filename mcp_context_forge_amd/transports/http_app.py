@@ -366,7 +366,9 @@ def build_app(engine: GatewayEngine, auth: Optional[AuthService] = None) -> Fast
                 if kind == "tool":
                     body.setdefault("original_name", body.get("name"))
                     body.setdefault("integration_type", "REST" if body.get("url") else "LOCAL")
-                return engine.registry.create(kind, **body)
+                ent = engine.registry.create(kind, **body)
+                engine.audit.record(ctx.user, "create", kind, ent.get("id"))
+                return ent
             except ConflictError as exc:
                 raise HTTPException(409, str(exc)) from exc
             except GatewayConnectionError as exc:
@@ -392,6 +394,7 @@ def build_app(engine: GatewayEngine, auth: Optional[AuthService] = None) -> Fast
                     await engine.gateway_service.delete_gateway(entity_id)
                 else:
                     engine.registry.delete(kind, entity_id)
+                engine.audit.record(ctx.user, "delete", kind, entity_id)
             except NotFoundError as exc:
                 raise HTTPException(404, str(exc)) from exc
             return Response(status_code=204)
@@ -526,6 +529,15 @@ def build_app(engine: GatewayEngine, auth: Optional[AuthService] = None) -> Fast
         if engine.gpu_pipeline is not None:
             out["gpu"] = engine.gpu_pipeline.stats()
         return out
+
+    @app.get("/admin/traces")
+    async def admin_traces(limit: int = 100, ctx: AuthContext = Depends(require("admin.read"))):
+        engine.observability.flush()
+        return engine.observability.query_traces(limit)
+
+    @app.get("/admin/audit")
+    async def admin_audit(limit: int = 100, ctx: AuthContext = Depends(require("admin.read"))):
+        return engine.audit.query(limit)
 
     @app.get("/admin/plugins")
     async def admin_plugins(ctx: AuthContext = Depends(require("admin.read"))):
