@@ -133,8 +133,32 @@ class GpuPluginPipeline:
         self.max_depth = s.max_json_depth
         self.max_string = s.max_string_length
 
+        # post-pass metadata bank: structuredContent presence → toon candidate
+        self.banks["postmeta"] = hip.DeviceScanTables(
+            dfa.compile_literals(['"structuredContent"'], case_insensitive=False), device)
+
+        # native decision-plane stores + string tables (fastpath.cpp)
+        self._slot_store = hip.store_new(self.semcache.capacity) if self.semcache is not None else 0
+        self._exact_native = hip.cache_new(self.exact_cache.ttl) if self.exact_cache is not None else 0
+
+        def _strtable(items):
+            blob = b"".join(x.encode() for x in items)
+            off = np.zeros(len(items) + 1, dtype=np.int32)
+            for i, x in enumerate(items):
+                off[i + 1] = off[i] + len(x.encode())
+            return np.frombuffer(blob, dtype=np.uint8).copy() if blob else np.zeros(1, dtype=np.uint8), off
+
+        from ..models.classifier import category_names
+
+        self._deny_tab = _strtable(self.deny.words if self.deny else [])
+        self._harm_tab = _strtable(self.harm.cats if self.harm else [])
+        self._mod_tab = _strtable(category_names(self.moderation.model.classes) if self.moderation else [])
+
         # per-tool metadata + schema-shape bank (rebuilt on registry change)
         self._meta_gen = -1
+        self._plugins_ver = -1
+        self._toolmap = 0
+        self._meta_list: List[_ToolMeta] = []
         self._tool_meta: Dict[str, _ToolMeta] = {}
         self._schema_bank: Optional[hip.DeviceScanTables] = None
         self._schema_pat_ids: Dict[str, int] = {}
@@ -256,10 +280,72 @@ class GpuPluginPipeline:
                         m.schema_mode = "host"
         else:
             self._schema_bank = None
+
+        # ---- flat arrays for the native decision plane (fastpath.cpp) ----
+        metas = list(self._tool_meta.values())
+        self._meta_list = metas
+        nt = len(metas)
+        name_bytes = [m.name.encode() for m in metas]
+        blob = b"".join(name_bytes)
+        self._t_name_blob = np.frombuffer(blob, dtype=np.uint8).copy() if blob else np.zeros(1, dtype=np.uint8)
+        self._t_name_beg = np.zeros(nt, dtype=np.int32)
+        self._t_name_end = np.zeros(nt, dtype=np.int32)
+        off = 0
+        for i, nb in enumerate(name_bytes):
+            self._t_name_beg[i] = off
+            off += len(nb)
+            self._t_name_end[i] = off
+        if getattr(self, "_toolmap", 0):
+            hip.toolmap_free(self._toolmap)
+        self._toolmap = hip.toolmap_new(self._t_name_blob, self._t_name_beg, self._t_name_end) if nt else 0
+        self._t_required = np.array([m.required_bits for m in metas], dtype=np.uint32) if nt else np.zeros(1, dtype=np.uint32)
+        typed = np.full(max(nt, 1), 0xFFFFFFFFFFFFFFFF, dtype=np.uint64)
+        for i, m in enumerate(metas):
+            packed = 0
+            for k in range(4):
+                pk = 0xFFFF if k >= len(m.typed_pairs) else ((m.typed_pairs[k][0] << 8) | m.typed_pairs[k][1])
+                packed |= pk << (k * 16)
+            typed[i] = packed
+        self._t_typed = typed
+        self._t_native_kind = np.array([m.native_kind for m in metas], dtype=np.int8) if nt else np.zeros(1, dtype=np.int8)
+        self._t_outschema = np.array([m.has_output_schema for m in metas], dtype=bool) if nt else np.zeros(1, dtype=bool)
+        self._t_thash = np.array([m.thash for m in metas], dtype=np.int64) if nt else np.zeros(1, dtype=np.int64)
+        self._t_index = {m.name: i for i, m in enumerate(metas)}
+
+        flags = np.zeros(max(nt, 1), dtype=np.uint32)
+        for i, m in enumerate(metas):
+            f = 0
+            if m.reachable:
+                f |= hip.TF_REACHABLE
+            if self._enforcing(self.deny) and self._applies(self.deny, m.name):
+                f |= hip.TF_DENY
+            if self.pii is not None and self._applies(self.pii, m.name):
+                f |= hip.TF_PII
+            if self.regex is not None and self._applies(self.regex, m.name):
+                f |= hip.TF_REGEX
+            if self.normalizer is not None and self._applies(self.normalizer, m.name):
+                f |= hip.TF_NORM
+            if self._enforcing(self.moderation) and self._applies(self.moderation, m.name):
+                f |= hip.TF_MOD
+            if self._enforcing(self.harm) and self._applies(self.harm, m.name):
+                f |= hip.TF_HARM
+            if self._enforcing(self.schema_guard):
+                if m.schema_mode == "fast":
+                    f |= hip.TF_SCHEMA_FAST
+                elif m.schema_mode == "host":
+                    f |= hip.TF_SCHEMA_HOST
+            if self.semcache is not None:
+                f |= hip.TF_CACHE
+            if self.exact_cache is not None:
+                f |= hip.TF_EXACT
+            flags[i] = f
+        self._t_flags = flags
         self._meta_gen = self.engine.registry.generation
+        self._plugins_ver = getattr(self.engine.plugins, "version", 0)
 
     def _meta(self) -> Dict[str, _ToolMeta]:
-        if self._meta_gen != self.engine.registry.generation:
+        if self._meta_gen != self.engine.registry.generation or \
+           self._plugins_ver != getattr(self.engine.plugins, "version", 0):
             self._rebuild_tool_meta()
         return self._tool_meta
 
@@ -325,36 +411,28 @@ class GpuPluginPipeline:
     async def _fast_toolcalls(self, raws: List[bytes], blob: np.ndarray, env: dict,
                               rows: np.ndarray, responses: List[Optional[bytes]],
                               user: Optional[str], server_id: Optional[str]) -> None:
-        meta_map = self._meta()
+        self._meta()  # refresh tool tables on registry/plugin change
         m = rows.shape[0]
-        nb, ne = env["name_beg"], env["name_end"]
+        nb = np.ascontiguousarray(env["name_beg"][rows])
+        ne = np.ascontiguousarray(env["name_end"][rows])
+        id_b = np.ascontiguousarray(env["id_beg"][rows])
+        id_e = np.ascontiguousarray(env["id_end"][rows])
         ab, ae = env["args_beg"], env["args_end"]
+        args_b = np.ascontiguousarray(np.where(ab[rows] >= 0, ab[rows], 0).astype(np.int32))
+        args_e = np.ascontiguousarray(np.where(ab[rows] >= 0, ae[rows], 0).astype(np.int32))
 
-        # --- tool resolution (memoized on raw name bytes) ---
-        metas: List[Optional[_ToolMeta]] = [None] * m
-        memo: Dict[bytes, Optional[_ToolMeta]] = {}
-        for j in range(m):
-            r = int(rows[j])
-            key = blob[nb[r]:ne[r]].tobytes()
-            if key in memo:
-                metas[j] = memo[key]
-            else:
-                mt = meta_map.get(key.decode("utf-8", "replace"))
-                memo[key] = mt
-                metas[j] = mt
-
-        th_arr = np.array([mt.thash if mt else 0 for mt in metas], dtype=np.int64)
+        tool_idx = hip.toolmap_resolve(self._toolmap, blob, nb, ne) if self._toolmap \
+            else np.full(m, -1, dtype=np.int32)
 
         # --- GPU pass 1 over raw argument spans ---
-        args_b = np.where(ab[rows] >= 0, ab[rows], 0).astype(np.int32)
-        args_e = np.where(ab[rows] >= 0, ae[rows], 0).astype(np.int32)
         data_gpu = torch.from_numpy(blob.copy()).to(self.device, non_blocking=True)
         beg_t = torch.from_numpy(args_b).to(self.device, non_blocking=True)
         end_t = torch.from_numpy(args_e).to(self.device, non_blocking=True)
-
         out: Dict[str, torch.Tensor] = {}
-        for bname, bank in self.banks.items():
-            out[bname], _ = hip.scan(data_gpu, beg_t, end_t, bank)
+        for bname in ("deny", "harm", "pii", "regex", "normalize"):
+            bank = self.banks.get(bname)
+            if bank is not None:
+                out[bname], _ = hip.scan(data_gpu, beg_t, end_t, bank)
         if self._schema_bank is not None:
             out["schema"], _ = hip.scan(data_gpu, beg_t, end_t, self._schema_bank)
         feats = None
@@ -368,132 +446,68 @@ class GpuPluginPipeline:
             cache_val_t, cache_idx_t = bv[:m], bi[:m]
         torch.cuda.synchronize()
 
-        zeros = np.zeros(m, dtype=np.int64)
-        deny_m = out["deny"].cpu().numpy().astype(np.uint32) if "deny" in out else zeros
-        harm_m = out["harm"].cpu().numpy().astype(np.uint32) if "harm" in out else zeros
-        pii_m = out["pii"].cpu().numpy().astype(np.uint32) if "pii" in out else zeros
-        regex_m = out["regex"].cpu().numpy().astype(np.uint32) if "regex" in out else zeros
-        norm_m = out["normalize"].cpu().numpy().astype(np.uint32) if "normalize" in out else zeros
-        schema_m = out["schema"].cpu().numpy().astype(np.uint32) if "schema" in out else zeros
-        scores = scores_t.cpu().numpy() if scores_t is not None else None
-        hit_mask = np.zeros(m, dtype=bool)
-        cache_idx = None
+        def mask(name):
+            t = out.get(name)
+            return t.cpu().numpy().view(np.uint32) if t is not None else np.zeros(m, dtype=np.uint32)
+
+        deny_m, harm_m, pii_m = mask("deny"), mask("harm"), mask("pii")
+        regex_m, norm_m, schema_m = mask("regex"), mask("normalize"), mask("schema")
+
+        mod_block = np.zeros(m, dtype=np.uint8)
+        mod_cat = np.zeros(m, dtype=np.int32)
+        mod_score = np.zeros(m, dtype=np.float32)
+        if scores_t is not None:
+            sc = scores_t.cpu().numpy()
+            mod_score = np.ascontiguousarray(sc.max(axis=1).astype(np.float32))
+            mod_cat = np.ascontiguousarray(sc.argmax(axis=1).astype(np.int32))
+            mod_block = (mod_score >= self.moderation.threshold).astype(np.uint8)
+
+        th_arr = np.where(tool_idx >= 0, self._t_thash[np.clip(tool_idx, 0, None)], 0)
+        hit = np.zeros(m, dtype=np.uint8)
+        hit_slot = np.full(m, -1, dtype=np.int32)
         if cache_val_t is not None:
             cache_val = cache_val_t.cpu().numpy()
-            cache_idx = cache_idx_t.cpu().numpy()
-            hit_mask = self.semcache.resolve_hits_np(cache_val, cache_idx, th_arr)
+            hit_slot = np.ascontiguousarray(cache_idx_t.cpu().numpy().astype(np.int32))
+            hit = self.semcache.resolve_hits_np(cache_val, hit_slot, th_arr).astype(np.uint8)
 
-        # --- vectorized decisions (precedence = CPU chain priority order) ---
-        # 0 dispatch | 1 blocked/handled | 2 rewrite | 3 host-schema
-        state = np.zeros(m, dtype=np.int8)
-        mod_block = np.zeros(m, dtype=bool)
-        if scores is not None and self._enforcing(self.moderation):
-            mod_block = scores.max(axis=1) >= self.moderation.threshold
-        deny_on = self._enforcing(self.deny)
-        harm_on = self._enforcing(self.harm)
+        # breaker overlay: flip TF_BREAKER_OPEN for tools with open circuits
+        flags = self._t_flags
+        if self.breaker is not None and self.breaker.state:
+            now0 = time.monotonic()
+            open_tis = [self._t_index[nm] for nm, st in self.breaker.state.items()
+                        if now0 < st["open_until"] and nm in self._t_index]
+            if open_tis:
+                flags = flags.copy()
+                flags[open_tis] |= hip.TF_BREAKER_OPEN
 
         t0 = time.monotonic()
-        rewrite_rows: List[int] = []
-        dispatch_rows: List[int] = []
-        now = time.monotonic()
-        exact_store = self.exact_cache.store if self.exact_cache is not None else None
-        exact_ttl = self.exact_cache.ttl if self.exact_cache is not None else 0.0
+        state, nk, reason, arena, rb, re_, n_arena = hip.decide(
+            blob, id_b, id_e, args_b, args_e, tool_idx, nb, ne,
+            deny_m, harm_m, pii_m, regex_m, norm_m, schema_m,
+            mod_block, mod_cat, mod_score, hit, hit_slot,
+            flags, self._t_required, self._t_typed,
+            self._t_name_beg, self._t_name_end, self._t_name_blob, self._t_native_kind,
+            self._nest_bits,
+            self._deny_tab[0], self._deny_tab[1], self._harm_tab[0], self._harm_tab[1],
+            self._mod_tab[0], self._mod_tab[1],
+            self._slot_store, self._exact_native, time.monotonic())
 
-        for j in range(m):
-            mt = metas[j]
-            r = int(rows[j])
-            idb = self._id_bytes(blob, env, r)
-            if mt is None:
-                name = blob[nb[r]:ne[r]].tobytes().decode("utf-8", "replace")
-                responses[r] = self._splice_error(idb, jsonrpc.INVALID_PARAMS, f"Tool not found: {name}")
-                state[j] = 1
-                continue
-            name = mt.name
-            if not mt.reachable:
-                responses[r] = self._splice_error(idb, jsonrpc.SERVER_UNAVAILABLE,
-                                                  f"Tool {name} currently unreachable")
-                state[j] = 1
-                continue
-            if deny_m[j] and deny_on and self._applies(self.deny, name):
-                pid = int(deny_m[j]).bit_length() - 1
-                word = self.deny.words[pid] if pid < len(self.deny.words) else "?"
-                responses[r] = self._splice_error(idb, jsonrpc.POLICY_DENIED,
-                                                  f"deny_filter: deny word {word!r} present")
-                state[j] = 1
-                self.blocked += 1
-                continue
-            if (pii_m[j] and self._applies(self.pii, name)) or \
-               (regex_m[j] and self._applies(self.regex, name)) or \
-               (norm_m[j] and self._applies(self.normalizer, name)):
-                state[j] = 2
-                rewrite_rows.append(j)
-                continue
-            if mod_block[j] and self._applies(self.moderation, name):
-                from ..models.classifier import category_names
-
-                row_sc = scores[j]
-                cat = category_names(len(row_sc))[int(row_sc.argmax())]
-                responses[r] = self._splice_error(
-                    idb, jsonrpc.POLICY_DENIED,
-                    f"content_moderation: moderation: category {cat} score {float(row_sc.max()):.3f}")
-                state[j] = 1
-                self.blocked += 1
-                continue
-            if harm_m[j] and harm_on and self._applies(self.harm, name):
-                pid = int(harm_m[j]).bit_length() - 1
-                cat = self.harm.cats[pid] if pid < len(self.harm.cats) else "?"
-                responses[r] = self._splice_error(idb, jsonrpc.POLICY_DENIED,
-                                                  f"harmful_content_detector: harmful content ({cat})")
-                state[j] = 1
-                self.blocked += 1
-                continue
-            if self._enforcing(self.schema_guard):
-                mode = mt.schema_mode
-                if mode == "fast":
-                    sm = int(schema_m[j])
-                    ok = (sm & mt.required_bits) == mt.required_bits and not (sm & self._nest_bits)
-                    if ok:
-                        for present, typed in mt.typed_pairs:
-                            if (sm >> present) & 1 and not ((sm >> typed) & 1):
-                                ok = False
-                                break
-                    if not ok:
-                        state[j] = 3
-                        continue
-                elif mode == "host":
-                    state[j] = 3
-                    continue
-            if hit_mask[j]:
-                res = self.semcache.results[int(cache_idx[j])]
-                if idb is not None:
-                    responses[r] = self._splice_result(idb, res if isinstance(res, bytes)
-                                                       else json.dumps(res, separators=(",", ":")).encode())
-                state[j] = 1
-                self.cache_hits += 1
-                continue
-            if exact_store is not None:
-                ek = (name, blob[args_b[j]:args_e[j]].tobytes())
-                ent = exact_store.get(ek)
-                if ent is not None and now - ent[0] <= exact_ttl:
-                    if idb is not None:
-                        responses[r] = self._splice_result(idb, ent[1])
-                    state[j] = 1
-                    self.cache_hits += 1
-                    continue
-            if self.breaker is not None:
-                st = self.breaker._st(name)
-                if now < st["open_until"]:
-                    responses[r] = self._splice_error(idb, jsonrpc.POLICY_DENIED,
-                                                      f"circuit_breaker: circuit open for tool {name}")
-                    state[j] = 1
-                    self.blocked += 1
-                    continue
-            dispatch_rows.append(j)
-
-        # --- host-schema fallback (exact validate; violation blocks) ---
-        for j in np.nonzero(state == 3)[0]:
+        arena_b = arena.tobytes()
+        answered = np.nonzero(state == hip.ST_ANSWERED)[0]
+        for j in answered:
             j = int(j)
-            mt = metas[j]
+            if rb[j] >= 0:
+                responses[int(rows[j])] = arena_b[rb[j]:re_[j]]
+        rc = np.bincount(reason[answered].astype(np.int64), minlength=9) if answered.size else np.zeros(9, int)
+        self.blocked += int(rc[3] + rc[4] + rc[5] + rc[8])
+        self.cache_hits += int(rc[6] + rc[7])
+
+        # --- host-schema fallback rows (exact validation) ---
+        extra_native: List[int] = []
+        extra_py: List[Tuple[int, Any]] = []
+        for j in np.nonzero(state == hip.ST_HOST_SCHEMA)[0]:
+            j = int(j)
+            mt = self._meta_list[tool_idx[j]]
             r = int(rows[j])
             idb = self._id_bytes(blob, env, r)
             try:
@@ -508,26 +522,43 @@ class GpuPluginPipeline:
                 responses[r] = self._splice_error(idb, jsonrpc.POLICY_DENIED,
                                                   "schema_guard: schema violation: " + "; ".join(errs[:5]))
                 self.blocked += 1
-            elif hit_mask[j]:
-                res = self.semcache.results[int(cache_idx[j])]
-                if idb is not None:
-                    responses[r] = self._splice_result(idb, res if isinstance(res, bytes)
-                                                       else json.dumps(res, separators=(",", ":")).encode())
+            elif hit[j]:
+                res = self._cache_result_bytes(int(hit_slot[j]))
+                if idb is not None and res is not None:
+                    responses[r] = self._splice_result(idb, res)
                 self.cache_hits += 1
+            elif mt.native_kind >= 0:
+                nk[j] = mt.native_kind
+                extra_native.append(j)
             else:
-                dispatch_rows.append(j)
+                extra_py.append((j, args))
 
-        # --- PASS 2: rewrite-flagged subset (host rewrites + GPU re-scan) ---
-        rewrite_dispatch: List[Tuple[int, Any]] = []  # (j, rewritten args)
-        if rewrite_rows:
-            self.slow_path += len(rewrite_rows)
+        # --- PASS 2: rewrite-flagged subset ---
+        rewrite_js = [int(j) for j in np.nonzero(state == hip.ST_REWRITE)[0]]
+        rewrite_dispatch: List[Tuple[int, Any]] = []
+        if rewrite_js:
+            self.slow_path += len(rewrite_js)
             rewrite_dispatch = await self._rewrite_pass(
-                blob, env, rows, args_b, args_e, rewrite_rows, metas, hit_mask, cache_idx, responses)
+                blob, env, rows, args_b, args_e, rewrite_js, tool_idx, hit, hit_slot, responses)
 
-        # --- fan-out dispatch ---
-        self.fast_path += len(dispatch_rows)
-        await self._dispatch_and_post(blob, env, rows, args_b, args_e, metas, feats, th_arr,
-                                      dispatch_rows, rewrite_dispatch, responses, t0)
+        native_js = [int(j) for j in np.nonzero(state == hip.ST_DISPATCH_NATIVE)[0]] + extra_native
+        py_items = [(int(j), None) for j in np.nonzero(state == hip.ST_DISPATCH_PY)[0]] + \
+            extra_py + rewrite_dispatch
+        self.fast_path += len(native_js) + len(py_items)
+        if native_js or py_items:
+            await self._dispatch_and_post(blob, env, rows, id_b, id_e, args_b, args_e,
+                                          tool_idx, nk, feats, th_arr, native_js, py_items,
+                                          responses, t0)
+
+    def _cache_result_bytes(self, slot: int) -> Optional[bytes]:
+        if self._slot_store:
+            res = hip.store_get(self._slot_store, slot)
+            if res:
+                return res
+        res = self.semcache.results[slot] if self.semcache is not None else None
+        if res is None:
+            return None
+        return res if isinstance(res, bytes) else json.dumps(res, separators=(",", ":")).encode()
 
     # ------------------------------------------------------------------
     def _apply_rewrites(self, name: str, args: Any) -> Tuple[str, Any]:
@@ -552,12 +583,13 @@ class GpuPluginPipeline:
                 args = new_args
         return ("__ok__", args)
 
-    async def _rewrite_pass(self, blob, env, rows, args_b, args_e, rewrite_rows, metas,
-                            hit_mask, cache_idx, responses) -> List[Tuple[int, Any]]:
+
+    async def _rewrite_pass(self, blob, env, rows, args_b, args_e, rewrite_js, tool_idx,
+                            hit, hit_slot, responses) -> List[Tuple[int, Any]]:
         ok_items: List[Tuple[int, Any]] = []
-        for j in rewrite_rows:
+        for j in rewrite_js:
+            mt = self._meta_list[tool_idx[j]]
             r = int(rows[j])
-            mt = metas[j]
             idb = self._id_bytes(blob, env, r)
             try:
                 args = json.loads(blob[args_b[j]:args_e[j]].tobytes() or b"{}")
@@ -589,9 +621,9 @@ class GpuPluginPipeline:
 
         out: List[Tuple[int, Any]] = []
         for jj, (j, args2) in enumerate(ok_items):
-            r = int(rows[j])
-            mt = metas[j]
+            mt = self._meta_list[tool_idx[j]]
             name = mt.name
+            r = int(rows[j])
             idb = self._id_bytes(blob, env, r)
             if scores2 is not None and self._enforcing(self.moderation) and self._applies(self.moderation, name):
                 row_sc = scores2[jj]
@@ -620,155 +652,201 @@ class GpuPluginPipeline:
                                                       "schema_guard: schema violation: " + "; ".join(errs[:5]))
                     self.blocked += 1
                     continue
-            if hit_mask[j]:
-                res = self.semcache.results[int(cache_idx[j])]
-                if idb is not None:
-                    responses[r] = self._splice_result(idb, res if isinstance(res, bytes)
-                                                       else json.dumps(res, separators=(",", ":")).encode())
+            if hit[j]:
+                res = self._cache_result_bytes(int(hit_slot[j]))
+                if idb is not None and res is not None:
+                    responses[r] = self._splice_result(idb, res)
                 self.cache_hits += 1
                 continue
             out.append((j, args2))
         return out
 
     # ------------------------------------------------------------------
-    async def _dispatch_and_post(self, blob, env, rows, args_b, args_e, metas, feats, th_arr,
-                                 dispatch_rows: List[int], rewrite_dispatch: List[Tuple[int, Any]],
+    async def _dispatch_and_post(self, blob, env, rows, id_b, id_e, args_b, args_e,
+                                 tool_idx, nk, feats, th_arr,
+                                 native_js: List[int], py_items: List[Tuple[int, Any]],
                                  responses: List[Optional[bytes]], t0: float) -> None:
-        # Split native-batch vs python dispatch.
-        native_js: List[int] = []
-        py_items: List[Tuple[int, Any]] = []  # (j, args or None=raw)
-        for j in dispatch_rows:
-            mt = metas[j]
-            if mt.native_kind >= 0:
-                native_js.append(j)
-            else:
-                py_items.append((j, None))
-        for j, args2 in rewrite_dispatch:
-            mt = metas[j]
-            py_items.append((j, args2))  # rewritten args always go through python dispatch
-
-        result_bytes: Dict[int, Optional[bytes]] = {}
-        errors: Dict[int, Exception] = {}
-
-        # --- native upstream batch call (C++) ---
+        # --- native upstream batch (C++) ---
+        nat_blob = np.zeros(0, dtype=np.uint8)
+        nat_beg = nat_end = np.zeros(0, dtype=np.int64)
         if native_js:
-            kinds = np.array([metas[j].native_kind for j in native_js], dtype=np.int32)
-            nb_ = np.array([args_b[j] for j in native_js], dtype=np.int32)
-            ne_ = np.array([args_e[j] for j in native_js], dtype=np.int32)
+            njs = np.asarray(native_js, dtype=np.int64)
+            kinds = np.ascontiguousarray(nk[njs].astype(np.int32))
+            nb_ = np.ascontiguousarray(args_b[njs])
+            ne_ = np.ascontiguousarray(args_e[njs])
             nb_ = np.where(ne_ > nb_, nb_, -1).astype(np.int32)
             now_iso = time.strftime("%Y-%m-%dT%H:%M:%SZ", time.gmtime())
-            out_blob, rb, re_ = hip.upstream_call_batch(blob, nb_, ne_, kinds, now_iso)
-            ob = out_blob.tobytes()
-            for jj, j in enumerate(native_js):
-                result_bytes[j] = ob[int(rb[jj]):int(re_[jj])]
-                metas[j].native_client.calls += 1
+            nat_blob, nat_beg, nat_end = hip.upstream_call_batch(blob, nb_, ne_, kinds, now_iso)
 
-        # --- python dispatch for the rest ---
+        # --- python dispatch (non-native upstreams / rewritten args) ---
+        py_results: List[Optional[bytes]] = []
+        py_errors: Dict[int, Exception] = {}
         if py_items:
             ts = self.engine.tool_service
             from ..services.upstream import InProcUpstream
 
-            async def one(j: int, args2: Any) -> None:
-                mt = metas[j]
+            async def one(idx: int, j: int, args2: Any) -> None:
+                mt = self._meta_list[tool_idx[j]]
                 args = args2
                 if args is None:
                     try:
                         args = json.loads(blob[args_b[j]:args_e[j]].tobytes() or b"{}")
                     except Exception:
-                        errors[j] = jsonrpc.JSONRPCError(jsonrpc.INVALID_PARAMS, "invalid arguments")
+                        py_errors[idx] = jsonrpc.JSONRPCError(jsonrpc.INVALID_PARAMS, "invalid arguments")
                         return
                 try:
                     result = await ts.dispatch(mt.tool, args)
-                    result_bytes[j] = json.dumps(result, separators=(",", ":"), default=str).encode()
+                    py_results[idx] = json.dumps(result, separators=(",", ":"), default=str).encode()
                 except Exception as exc:
-                    errors[j] = exc
+                    py_errors[idx] = exc
 
-            # in-proc dispatch is non-blocking → sequential loop; real I/O → gather
+            py_results = [None] * len(py_items)
             seq, conc = [], []
-            for j, args2 in py_items:
-                mt = metas[j]
+            for idx, (j, args2) in enumerate(py_items):
+                mt = self._meta_list[tool_idx[j]]
                 if mt.itype == "LOCAL" or isinstance(mt.client, InProcUpstream) or mt.native_kind >= 0:
-                    seq.append((j, args2))
+                    seq.append((idx, j, args2))
                 else:
-                    conc.append((j, args2))
-            for j, args2 in seq:
-                await one(j, args2)
+                    conc.append((idx, j, args2))
+            for idx, j, args2 in seq:
+                await one(idx, j, args2)
             if conc:
-                await asyncio.gather(*(one(j, a) for j, a in conc))
+                await asyncio.gather(*(one(idx, j, a) for idx, j, a in conc))
 
-        all_js = native_js + [j for j, _ in py_items]
+        # --- assemble one result blob (native results + python results) ---
+        all_js = list(native_js) + [j for j, _ in py_items]
+        n_all = len(all_js)
+        nat_used = int(nat_end[-1]) if native_js else 0
+        py_blob = b"".join(r or b"" for r in py_results)
+        res_blob = np.concatenate([nat_blob[:nat_used],
+                                   np.frombuffer(py_blob, dtype=np.uint8)]) if py_blob else \
+            np.ascontiguousarray(nat_blob[:nat_used])
+        if res_blob.size == 0:
+            res_blob = np.zeros(1, dtype=np.uint8)
+        res_beg = np.zeros(n_all, dtype=np.int64)
+        res_end = np.zeros(n_all, dtype=np.int64)
+        res_beg[: len(native_js)] = nat_beg
+        res_end[: len(native_js)] = nat_end
+        off = nat_used
+        err_rows = np.zeros(n_all, dtype=np.uint8)
+        for idx in range(len(py_items)):
+            k = len(native_js) + idx
+            if py_results[idx] is not None:
+                res_beg[k] = off
+                off += len(py_results[idx])
+                res_end[k] = off
+            else:
+                res_beg[k] = res_end[k] = 0
+                err_rows[k] = 1
 
-        # --- PASS 3: GPU scan over serialized results ---
-        post_banks = tuple(b for b in ("pii", "regex", "harm") if b in self.banks)
-        post_mask = None
-        ok_js = [j for j in all_js if j in result_bytes and result_bytes[j]]
-        if ok_js and post_banks:
-            texts3 = [result_bytes[j] for j in ok_js]
-            data3, beg3, end3 = pack_texts(texts3, self.device)
-            masks3 = []
-            for b in post_banks:
-                t, _ = hip.scan(data3, beg3, end3, self.banks[b])
-                masks3.append(t)
+        # --- PASS 3: GPU scans over serialized results ---
+        all_js_np = np.asarray(all_js, dtype=np.int64)
+        post_flag = np.zeros(n_all, dtype=bool)
+        toon_meta = np.zeros(n_all, dtype=bool)
+        if n_all:
+            data3 = torch.from_numpy(res_blob.copy()).to(self.device, non_blocking=True)
+            b3 = torch.from_numpy(res_beg.astype(np.int32)).to(self.device, non_blocking=True)
+            e3 = torch.from_numpy(res_end.astype(np.int32)).to(self.device, non_blocking=True)
+            masks3 = {}
+            for b in ("pii", "regex", "harm", "postmeta"):
+                if b in self.banks:
+                    masks3[b], _ = hip.scan(data3, b3, e3, self.banks[b])
             torch.cuda.synchronize()
-            combined = np.zeros(len(ok_js), dtype=np.int64)
-            for t in masks3:
-                combined |= t.cpu().numpy().astype(np.uint32).astype(np.int64)
-            post_mask = {j: bool(combined[i]) for i, j in enumerate(ok_js)}
+            for b, t in masks3.items():
+                h = t.cpu().numpy().view(np.uint32) != 0
+                if b == "postmeta":
+                    toon_meta = h
+                else:
+                    post_flag |= h
 
-        insert_rows: List[int] = []
-        insert_hashes: List[int] = []
-        insert_results: List[Any] = []
-        agg: Dict[str, List[int]] = {}
-        ms = (time.monotonic() - t0) * 1000.0
+        res_len = res_end - res_beg
         toon_min = self.toon.min_size if self.toon is not None else 1 << 60
         guard_max = self.out_guard.max_chars if self.out_guard is not None else 1 << 60
-        now = time.monotonic()
+        outschema = self._t_outschema[np.clip(tool_idx[all_js_np], 0, None)] if n_all else np.zeros(0, bool)
+        needs_host = post_flag | outschema | (toon_meta & (res_len >= toon_min)) | \
+            (res_len > guard_max) | err_rows.astype(bool)
 
-        for j in all_js:
-            mt = metas[j]
+        now = time.monotonic()
+        arena2, rb2, re2, is_err, cacheable = hip.finalize(
+            blob, id_b, id_e, args_b, args_e, tool_idx,
+            np.ascontiguousarray(all_js_np.astype(np.int32)), res_blob, res_beg, res_end,
+            np.ascontiguousarray(needs_host.astype(np.uint8)),
+            self._t_name_beg, self._t_name_end, self._t_name_blob, self._t_flags,
+            self._exact_native, now, self.exact_cache.ttl if self.exact_cache else 0.0)
+        arena2_b = arena2.tobytes()
+        for k in range(n_all):
+            if rb2[k] >= 0:
+                responses[int(rows[all_js[k]])] = arena2_b[rb2[k]:re2[k]]
+
+        # --- host post chain for flagged rows ---
+        host_ks = np.nonzero(needs_host)[0]
+        host_cached: List[Tuple[int, bytes]] = []  # (k, result bytes) for cache insert
+        for k in host_ks:
+            k = int(k)
+            j = all_js[k]
+            mt = self._meta_list[tool_idx[j]]
             r = int(rows[j])
             idb = self._id_bytes(blob, env, r)
-            a = agg.setdefault(mt.tid, [0, 0])
-            exc = errors.get(j)
-            if exc is not None:
+            if err_rows[k]:
+                exc = py_errors.get(k - len(native_js))
                 code = getattr(exc, "code", jsonrpc.SERVER_ERROR)
-                responses[r] = self._splice_error(idb, code if isinstance(code, int) else jsonrpc.SERVER_ERROR, str(exc))
-                a[0] += 1
-                a[1] += 1
-                if self.breaker is not None:
-                    self._breaker_record(mt.name, True)
+                responses[r] = self._splice_error(idb, code if isinstance(code, int) else jsonrpc.SERVER_ERROR,
+                                                  str(exc))
+                is_err[k] = 1
                 continue
-            rb = result_bytes[j]
-            need_host = (post_mask is not None and post_mask.get(j, False)) or mt.has_output_schema \
-                or (len(rb) >= toon_min and b'"structuredContent"' in rb) or len(rb) > guard_max
-            is_err = False
-            if need_host:
-                rb, is_err, blocked_msg = self._host_post(mt, rb)
-                if blocked_msg:
-                    responses[r] = self._splice_error(idb, jsonrpc.POLICY_DENIED, blocked_msg)
-                    self.blocked += 1
-                    a[0] += 1
-                    a[1] += 1
-                    continue
-            a[0] += 1
-            if is_err:
-                a[1] += 1
-            if self.breaker is not None:
-                self._breaker_record(mt.name, is_err)
+            rbytes = res_blob[res_beg[k]:res_end[k]].tobytes()
+            rbytes, host_err, blocked_msg = self._host_post(mt, rbytes)
+            if blocked_msg:
+                responses[r] = self._splice_error(idb, jsonrpc.POLICY_DENIED, blocked_msg)
+                self.blocked += 1
+                is_err[k] = 1
+                continue
+            is_err[k] = 1 if host_err else 0
             if idb is not None:
-                responses[r] = self._splice_result(idb, rb)
-            if not is_err:
-                if self.exact_cache is not None:
-                    self.exact_cache.store[(mt.name, blob[args_b[j]:args_e[j]].tobytes())] = (now, rb)
-                if self.semcache is not None:
-                    insert_rows.append(j)
-                    insert_hashes.append(int(th_arr[j]))
-                    insert_results.append(rb)
+                responses[r] = self._splice_result(idb, rbytes)
+            if not host_err:
+                cacheable[k] = 1
+                host_cached.append((k, rbytes))
 
-        if insert_rows and feats is not None:
-            self.semcache.insert_batch(feats, insert_rows, insert_hashes, insert_results)
-        for tid, (cnt, errs_) in agg.items():
-            self.engine.metrics.record_aggregate(tid, cnt, errs_, ms)
+        # --- bookkeeping: semcache insert, breaker, metrics ---
+        if self.semcache is not None and n_all:
+            ins = np.nonzero((cacheable == 1))[0]
+            if ins.size:
+                ins_js = all_js_np[ins]
+                slots = self.semcache.assign_slots(int(ins.size))
+                self.semcache.insert_features(feats, ins_js, slots, th_arr[ins_js])
+                if self._slot_store:
+                    hip.store_put_batch(self._slot_store, slots, res_blob,
+                                        np.ascontiguousarray(res_beg[ins]),
+                                        np.ascontiguousarray(res_end[ins]))
+                    for (k, rbytes) in host_cached:  # host-modified results overwrite
+                        pos = np.nonzero(ins == k)[0]
+                        if pos.size:
+                            hip.store_put(self._slot_store, int(slots[pos[0]]), rbytes)
+
+        if self.breaker is not None and n_all:
+            tis = tool_idx[all_js_np]
+            for ti in np.unique(tis):
+                sel = tis == ti
+                self._breaker_record_bulk(self._meta_list[int(ti)].name, int(sel.sum()),
+                                          int(is_err[sel].sum()))
+        if n_all:
+            ms = (time.monotonic() - t0) * 1000.0
+            tis = tool_idx[all_js_np]
+            for ti in np.unique(tis):
+                sel = tis == ti
+                self.engine.metrics.record_aggregate(self._meta_list[int(ti)].tid,
+                                                     int(sel.sum()), int(is_err[sel].sum()), ms)
+
+    def _breaker_record_bulk(self, name: str, n: int, n_err: int) -> None:
+        b = self.breaker
+        st = b._st(name)
+        st["results"].extend([True] * n_err + [False] * (n - n_err))
+        if len(st["results"]) > b.window:
+            st["results"] = st["results"][-b.window:]
+        if len(st["results"]) >= b.window and (sum(st["results"]) / len(st["results"])) >= b.error_threshold:
+            st["open_until"] = time.monotonic() + b.cooldown
+            st["results"] = []
 
     def _host_post(self, mt: _ToolMeta, rb: bytes) -> Tuple[bytes, bool, Optional[str]]:
         """Exact host post chain for flagged results: regex(20) → pii(30) →

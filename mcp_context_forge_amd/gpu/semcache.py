@@ -105,24 +105,35 @@ class GpuSemanticCache:
             out.append(self.results[slot])
         return out
 
+    def assign_slots(self, n: int) -> np.ndarray:
+        """Ring-allocate n slots (eviction = overwrite)."""
+        slots = (np.arange(n, dtype=np.int64) + self.write_ptr) % self.capacity
+        self.write_ptr = int((self.write_ptr + n) % self.capacity)
+        self.size = min(self.size + n, self.capacity)
+        return slots.astype(np.int32)
+
+    def insert_features(self, feats_bf16: torch.Tensor, rows: np.ndarray, slots: np.ndarray,
+                        tool_hashes: np.ndarray) -> None:
+        """Scatter feature rows + metadata for pre-assigned slots (results are
+        stored separately — C++ slot store on the native path)."""
+        src = feats_bf16[torch.from_numpy(rows.astype(np.int64)).to(self.device)].contiguous()
+        slot_t = torch.from_numpy(slots.astype(np.int32)).to(self.device)
+        hip.rows_scatter_bf16(src, slot_t, self.keys)
+        self.valid[slot_t.long()] = 1
+        now = time.monotonic()
+        self.tool_hashes[slots] = tool_hashes
+        self.timestamps[slots] = now
+
     def insert_batch(self, feats_bf16: torch.Tensor, rows: List[int],
                      tool_hashes: List[int], results: List[Any]) -> None:
         """Scatter feature rows into cache slots (ring eviction)."""
         if not rows:
             return
         n = len(rows)
-        slots = [(self.write_ptr + i) % self.capacity for i in range(n)]
-        self.write_ptr = (self.write_ptr + n) % self.capacity
-        self.size = min(self.size + n, self.capacity)
-        src = feats_bf16[torch.tensor(rows, dtype=torch.long, device=self.device)].contiguous()
-        slot_t = torch.tensor(slots, dtype=torch.int32, device=self.device)
-        hip.rows_scatter_bf16(src, slot_t, self.keys)
-        self.valid[slot_t.long()] = 1
-        now = time.monotonic()
-        for s, th, res in zip(slots, tool_hashes, results):
-            self.tool_hashes[s] = th
-            self.timestamps[s] = now
-            self.results[s] = res
+        slots = self.assign_slots(n)
+        self.insert_features(feats_bf16, np.asarray(rows), slots, np.asarray(tool_hashes, dtype=np.int64))
+        for s, res in zip(slots, results):
+            self.results[int(s)] = res
 
     def stats(self) -> dict:
         return {"size": self.size, "capacity": self.capacity, "hits": self.hits, "misses": self.misses,

@@ -207,3 +207,170 @@ def upstream_call_batch(blob: np.ndarray, args_beg: np.ndarray, args_end: np.nda
         lib.forge_upstream_call_batch(p(blob), p(args_beg), p(args_end), p(kinds), n,
                                       now_iso.encode(), p(out), out.shape[0], p(res_b), p(res_e))
     return out, res_b, res_e
+
+
+# ---------------------------------------------------------------------------
+# Native decision plane (fastpath.cpp)
+# ---------------------------------------------------------------------------
+
+def _np_ptr(a: Optional[np.ndarray]) -> ctypes.c_void_p:
+    return ctypes.c_void_p(a.ctypes.data) if a is not None else ctypes.c_void_p(0)
+
+
+@lru_cache(maxsize=1)
+def _fastpath_protos() -> ctypes.CDLL:
+    lib = _load()
+    lib.forge_store_new.argtypes = [ctypes.c_int]
+    lib.forge_store_new.restype = ctypes.c_void_p
+    lib.forge_store_put.argtypes = [ctypes.c_void_p, ctypes.c_int, ctypes.c_void_p, ctypes.c_int64]
+    lib.forge_store_put.restype = None
+    lib.forge_store_put_batch.argtypes = [ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int,
+                                          ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p]
+    lib.forge_store_put_batch.restype = None
+    lib.forge_store_free.argtypes = [ctypes.c_void_p]
+    lib.forge_store_free.restype = None
+    lib.forge_cache_new.argtypes = [ctypes.c_double]
+    lib.forge_cache_new.restype = ctypes.c_void_p
+    lib.forge_cache_free.argtypes = [ctypes.c_void_p]
+    lib.forge_cache_free.restype = None
+    lib.forge_toolmap_new.argtypes = [ctypes.c_void_p] * 3 + [ctypes.c_int]
+    lib.forge_toolmap_new.restype = ctypes.c_void_p
+    lib.forge_toolmap_free.argtypes = [ctypes.c_void_p]
+    lib.forge_toolmap_free.restype = None
+    lib.forge_toolmap_resolve.argtypes = [ctypes.c_void_p] * 4 + [ctypes.c_int, ctypes.c_void_p]
+    lib.forge_toolmap_resolve.restype = None
+    lib.forge_decide.argtypes = [ctypes.c_void_p] * 19 + [ctypes.c_int] + [ctypes.c_void_p] * 7 + \
+        [ctypes.c_uint32] + \
+        [ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int] * 3 + \
+        [ctypes.c_void_p, ctypes.c_void_p, ctypes.c_double] + \
+        [ctypes.c_void_p] * 4 + [ctypes.c_int64] + [ctypes.c_void_p] * 2
+    lib.forge_store_get.argtypes = [ctypes.c_void_p, ctypes.c_int, ctypes.c_void_p]
+    lib.forge_store_get.restype = ctypes.c_int64
+    lib.forge_decide.restype = ctypes.c_int64
+    lib.forge_finalize.argtypes = [ctypes.c_void_p] * 6 + [ctypes.c_int] + [ctypes.c_void_p, ctypes.c_int] + \
+        [ctypes.c_void_p] * 4 + [ctypes.c_void_p] * 4 + \
+        [ctypes.c_void_p, ctypes.c_double, ctypes.c_double] + \
+        [ctypes.c_void_p, ctypes.c_int64] + [ctypes.c_void_p] * 4
+    lib.forge_finalize.restype = ctypes.c_int64
+    return lib
+
+
+def store_new(capacity: int) -> int:
+    return _fastpath_protos().forge_store_new(capacity)
+
+
+def store_put(store: int, slot: int, data: bytes) -> None:
+    _fastpath_protos().forge_store_put(ctypes.c_void_p(store), slot, data, len(data))
+
+
+def store_put_batch(store: int, slots: np.ndarray, blob: np.ndarray,
+                    beg: np.ndarray, end: np.ndarray) -> None:
+    _fastpath_protos().forge_store_put_batch(ctypes.c_void_p(store), _np_ptr(slots), slots.shape[0],
+                                             _np_ptr(blob), _np_ptr(beg), _np_ptr(end))
+
+
+def store_free(store: int) -> None:
+    _fastpath_protos().forge_store_free(ctypes.c_void_p(store))
+
+
+def cache_new(ttl: float) -> int:
+    return _fastpath_protos().forge_cache_new(ttl)
+
+
+def cache_free(cache: int) -> None:
+    _fastpath_protos().forge_cache_free(ctypes.c_void_p(cache))
+
+
+def toolmap_new(blob: np.ndarray, beg: np.ndarray, end: np.ndarray) -> int:
+    return _fastpath_protos().forge_toolmap_new(_np_ptr(blob), _np_ptr(beg), _np_ptr(end), beg.shape[0])
+
+
+def toolmap_free(tm: int) -> None:
+    _fastpath_protos().forge_toolmap_free(ctypes.c_void_p(tm))
+
+
+def toolmap_resolve(tm: int, blob: np.ndarray, name_beg: np.ndarray, name_end: np.ndarray) -> np.ndarray:
+    m = name_beg.shape[0]
+    out = np.empty(m, dtype=np.int32)
+    _fastpath_protos().forge_toolmap_resolve(ctypes.c_void_p(tm), _np_ptr(blob),
+                                             _np_ptr(name_beg), _np_ptr(name_end), m, _np_ptr(out))
+    return out
+
+
+def decide(blob, id_beg, id_end, args_beg, args_end, tool_idx, name_beg, name_end,
+           deny_m, harm_m, pii_m, regex_m, norm_m, schema_m,
+           mod_block, mod_cat, mod_score, hit, hit_slot,
+           tool_flags, tool_required_bits, tool_typed_pairs,
+           tname_beg, tname_end, tname_blob, tool_native_kind, nest_bits,
+           deny_words, deny_off, harm_cats, harm_off, mod_cats, mod_off,
+           slot_store: int, exact_cache: int, now: float):
+    """→ (state i32[m], native_kind i32[m], arena bytes np, resp_beg i64, resp_end i64)."""
+    m = id_beg.shape[0]
+    lib = _fastpath_protos()
+    state = np.empty(m, dtype=np.int32)
+    nk = np.empty(m, dtype=np.int32)
+    reason = np.empty(m, dtype=np.int8)
+    resp_beg = np.empty(m, dtype=np.int64)
+    resp_end = np.empty(m, dtype=np.int64)
+    cap = m * 96 + 4096
+    while True:
+        arena = np.empty(cap, dtype=np.uint8)
+        n = lib.forge_decide(
+            _np_ptr(blob), _np_ptr(id_beg), _np_ptr(id_end), _np_ptr(args_beg), _np_ptr(args_end),
+            _np_ptr(tool_idx), _np_ptr(name_beg), _np_ptr(name_end),
+            _np_ptr(deny_m), _np_ptr(harm_m), _np_ptr(pii_m), _np_ptr(regex_m), _np_ptr(norm_m), _np_ptr(schema_m),
+            _np_ptr(mod_block), _np_ptr(mod_cat), _np_ptr(mod_score), _np_ptr(hit), _np_ptr(hit_slot),
+            m,
+            _np_ptr(tool_flags), _np_ptr(tool_required_bits), _np_ptr(tool_typed_pairs),
+            _np_ptr(tname_beg), _np_ptr(tname_end), _np_ptr(tname_blob), _np_ptr(tool_native_kind),
+            ctypes.c_uint32(nest_bits),
+            _np_ptr(deny_words), _np_ptr(deny_off), deny_off.shape[0] - 1,
+            _np_ptr(harm_cats), _np_ptr(harm_off), harm_off.shape[0] - 1,
+            _np_ptr(mod_cats), _np_ptr(mod_off), mod_off.shape[0] - 1,
+            ctypes.c_void_p(slot_store), ctypes.c_void_p(exact_cache), now,
+            _np_ptr(state), _np_ptr(nk), _np_ptr(reason), _np_ptr(arena), cap, _np_ptr(resp_beg), _np_ptr(resp_end))
+        if n >= 0:
+            return state, nk, reason, arena, resp_beg, resp_end, int(n)
+        cap = -int(n) + 4096
+
+
+def finalize(blob, id_beg, id_end, args_beg, args_end, tool_idx,
+             rows, res_blob, res_beg, res_end, needs_host,
+             tname_beg, tname_end, tname_blob, tool_flags,
+             exact_cache: int, now: float, exact_ttl: float):
+    """→ (arena np, resp_beg i64, resp_end i64, is_error u8, cacheable u8)."""
+    lib = _fastpath_protos()
+    n_rows = rows.shape[0]
+    resp_beg2 = np.empty(n_rows, dtype=np.int64)
+    resp_end2 = np.empty(n_rows, dtype=np.int64)
+    is_err = np.empty(n_rows, dtype=np.uint8)
+    cacheable = np.empty(n_rows, dtype=np.uint8)
+    cap = int(res_blob.shape[0]) + n_rows * 64 + 4096
+    while True:
+        arena = np.empty(cap, dtype=np.uint8)
+        n = lib.forge_finalize(
+            _np_ptr(blob), _np_ptr(id_beg), _np_ptr(id_end), _np_ptr(args_beg), _np_ptr(args_end),
+            _np_ptr(tool_idx), id_beg.shape[0],
+            _np_ptr(rows), n_rows,
+            _np_ptr(res_blob), _np_ptr(res_beg), _np_ptr(res_end), _np_ptr(needs_host),
+            _np_ptr(tname_beg), _np_ptr(tname_end), _np_ptr(tname_blob), _np_ptr(tool_flags),
+            ctypes.c_void_p(exact_cache), now, exact_ttl,
+            _np_ptr(arena), cap, _np_ptr(resp_beg2), _np_ptr(resp_end2),
+            _np_ptr(is_err), _np_ptr(cacheable))
+        if n >= 0:
+            return arena, resp_beg2, resp_end2, is_err, cacheable
+        cap = -int(n) + 4096
+
+
+TF_REACHABLE, TF_DENY, TF_PII, TF_REGEX, TF_NORM, TF_MOD, TF_HARM = (1 << i for i in range(7))
+TF_SCHEMA_FAST, TF_SCHEMA_HOST, TF_CACHE, TF_EXACT, TF_BREAKER_OPEN = (1 << i for i in range(7, 12))
+ST_DISPATCH_NATIVE, ST_ANSWERED, ST_REWRITE, ST_HOST_SCHEMA, ST_DISPATCH_PY = 0, 1, 2, 3, 4
+
+
+def store_get(store: int, slot: int) -> Optional[bytes]:
+    lib = _fastpath_protos()
+    ptr = ctypes.c_void_p(0)
+    n = lib.forge_store_get(ctypes.c_void_p(store), slot, ctypes.byref(ptr))
+    if n < 0 or not ptr.value:
+        return None
+    return ctypes.string_at(ptr.value, int(n))

@@ -173,12 +173,18 @@ class PluginManager:
         self.plugins: List[Plugin] = sorted(plugins or [], key=lambda p: p.priority)
         self.enabled = enabled
         self.timeout_s = timeout_s
+        self.version = 0  # bumped on runtime plugin changes (GPU pipeline re-syncs)
         self.global_state: Dict[str, Any] = {}
         self._stats: Dict[str, Dict[str, Any]] = {}
 
     def add(self, plugin: Plugin) -> None:
         self.plugins.append(plugin)
         self.plugins.sort(key=lambda p: p.priority)
+        self.version += 1
+
+    def bump(self) -> None:
+        """Signal runtime plugin-config change (mode flips etc.)."""
+        self.version += 1
 
     def get(self, name: str) -> Optional[Plugin]:
         for p in self.plugins:

@@ -553,6 +553,7 @@ def build_app(engine: GatewayEngine, auth: Optional[AuthService] = None) -> Fast
         if p is None:
             raise HTTPException(404, f"plugin {name} not found")
         p.mode = PluginMode(mode)
+        engine.plugins.bump()
         return {"name": name, "mode": p.mode.value}
 
     @app.get("/admin", response_class=HTMLResponse)
