@@ -132,6 +132,9 @@ class GpuPluginPipeline:
 
         self.max_depth = s.max_json_depth
         self.max_string = s.max_string_length
+        # pinned-host staging arena (bump-allocated per pass; reset after each sync)
+        self._pin = torch.empty(16 << 20, dtype=torch.uint8, pin_memory=True)
+        self._pin_off = 0
 
         # post-pass metadata bank: structuredContent presence → toon candidate
         self.banks["postmeta"] = hip.DeviceScanTables(
@@ -425,9 +428,10 @@ class GpuPluginPipeline:
             else np.full(m, -1, dtype=np.int32)
 
         # --- GPU pass 1 over raw argument spans ---
-        data_gpu = torch.from_numpy(blob.copy()).to(self.device, non_blocking=True)
-        beg_t = torch.from_numpy(args_b).to(self.device, non_blocking=True)
-        end_t = torch.from_numpy(args_e).to(self.device, non_blocking=True)
+        self._pin_reset()
+        data_gpu = self._upload(blob)
+        beg_t = self._upload(args_b)
+        end_t = self._upload(args_e)
         out: Dict[str, torch.Tensor] = {}
         for bname in ("deny", "harm", "pii", "regex", "normalize"):
             bank = self.banks.get(bname)
@@ -549,6 +553,27 @@ class GpuPluginPipeline:
             await self._dispatch_and_post(blob, env, rows, id_b, id_e, args_b, args_e,
                                           tool_idx, nk, feats, th_arr, native_js, py_items,
                                           responses, t0)
+
+    def _pin_reset(self) -> None:
+        self._pin_off = 0
+
+    def _upload(self, arr: np.ndarray) -> torch.Tensor:
+        """Stage a host array through the pinned arena → async H2D. Safe to
+        call repeatedly within one pass; reset after the pass's sync."""
+        raw = arr.view(np.uint8).reshape(-1)
+        n = raw.nbytes
+        off = (self._pin_off + 255) & ~255
+        if off + n > self._pin.numel():
+            grow = max(self._pin.numel() * 2, off + n)
+            self._pin = torch.empty(grow, dtype=torch.uint8, pin_memory=True)
+            off = 0
+        self._pin_off = off + n
+        view = self._pin[off:off + n]
+        view.numpy()[:] = raw
+        dev = view.to(self.device, non_blocking=True)
+        tdt = {np.dtype(np.uint8): torch.uint8, np.dtype(np.int32): torch.int32,
+               np.dtype(np.int64): torch.int64, np.dtype(np.float32): torch.float32}[np.dtype(arr.dtype)]
+        return dev.view(tdt).reshape(arr.shape)
 
     def _cache_result_bytes(self, slot: int) -> Optional[bytes]:
         if self._slot_store:
@@ -744,9 +769,10 @@ class GpuPluginPipeline:
         post_flag = np.zeros(n_all, dtype=bool)
         toon_meta = np.zeros(n_all, dtype=bool)
         if n_all:
-            data3 = torch.from_numpy(res_blob.copy()).to(self.device, non_blocking=True)
-            b3 = torch.from_numpy(res_beg.astype(np.int32)).to(self.device, non_blocking=True)
-            e3 = torch.from_numpy(res_end.astype(np.int32)).to(self.device, non_blocking=True)
+            self._pin_reset()
+            data3 = self._upload(res_blob)
+            b3 = self._upload(res_beg.astype(np.int32))
+            e3 = self._upload(res_end.astype(np.int32))
             masks3 = {}
             for b in ("pii", "regex", "harm", "postmeta"):
                 if b in self.banks:
