@@ -74,9 +74,10 @@ DEFAULT_PERMISSIONS = {
 
 
 class AuthService:
-    def __init__(self, db: Database, settings: Settings):
+    def __init__(self, db: Database, settings: Settings, token_blocklist=None):
         self.db = db
         self.settings = settings
+        self.token_blocklist = token_blocklist  # reference: services/token_blocklist.py
         self._perm_cache: dict = {}
 
     # -- bootstrap -------------------------------------------------------------
@@ -173,6 +174,8 @@ class AuthService:
                                               issuer=self.settings.jwt_issuer)
             except jwt_mod.JWTError as exc:
                 raise AuthError(f"Invalid token: {exc}") from exc
+            if self.token_blocklist is not None and self.token_blocklist.is_blocked(claims.get("jti")):
+                raise AuthError("Token revoked")
             user = claims.get("sub") or claims.get("username") or claims.get("email") or "unknown"
             return AuthContext(user=user, is_admin=bool(claims.get("admin") or user == self.settings.platform_admin_email
                                                         or user == self.settings.basic_auth_user),

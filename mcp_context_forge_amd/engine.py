@@ -29,6 +29,7 @@ from .registry.registry import NotFoundError, Registry
 from .services.a2a_service import A2AService
 from .services.content import CompletionService, PromptService, ResourceService, RootService
 from .services.gateway_service import GatewayService
+from .services.governance import CatalogService, ContentSecurity, PasswordPolicy, TokenBlocklist
 from .services.llm_proxy import LLMProxyError, LLMProxyService
 from .services.metrics import MetricsBuffer
 from .services.observability import AuditTrail, ObservabilityService
@@ -71,6 +72,10 @@ class GatewayEngine:
         self.root_service = RootService()
         self.llm_proxy = LLMProxyService()
         self.observability = ObservabilityService(self.db)
+        self.catalog = CatalogService()
+        self.password_policy = PasswordPolicy()
+        self.token_blocklist = TokenBlocklist()
+        self.content_security = ContentSecurity()
         self.audit = AuditTrail(self.db)
         self.sessions = SessionRegistry(self.settings.session_ttl, rank)
         self.started_at = time.time()

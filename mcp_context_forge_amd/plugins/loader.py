@@ -24,9 +24,17 @@ from .builtin import BUILTIN_PLUGINS
 from .framework import HookType, Plugin, PluginManager
 
 
+def _all_builtin():
+    from .content import EXTRA_PLUGINS
+
+    out = dict(BUILTIN_PLUGINS)
+    out.update(EXTRA_PLUGINS)
+    return out
+
+
 def _resolve_class(name: str, kind: str):
     if kind in ("builtin", "native", ""):
-        cls = BUILTIN_PLUGINS.get(name)
+        cls = _all_builtin().get(name)
         if cls is None:
             raise KeyError(f"unknown builtin plugin {name!r}")
         return cls

@@ -113,7 +113,7 @@ class BodyLimitMiddleware(BaseHTTPMiddleware):
 
 def build_app(engine: GatewayEngine, auth: Optional[AuthService] = None) -> FastAPI:
     settings = engine.settings
-    auth = auth or AuthService(engine.db, settings)
+    auth = auth or AuthService(engine.db, settings, token_blocklist=engine.token_blocklist)
     auth.bootstrap_admin()
 
     from contextlib import asynccontextmanager
@@ -503,6 +503,32 @@ def build_app(engine: GatewayEngine, auth: Optional[AuthService] = None) -> Fast
         return engine.llm_proxy.registry.register(
             name=body["name"], base_url=body["base_url"], api_key=body.get("api_key"),
             models=body.get("models"), default_model=body.get("default_model"))
+
+    # -- catalog (reference: services/catalog_service.py) --------------------------
+    @app.get("/catalog")
+    async def catalog_list(category: Optional[str] = None, ctx: AuthContext = Depends(require("tools.read"))):
+        return engine.catalog.list(category)
+
+    @app.post("/catalog/{catalog_id}/register", status_code=201)
+    async def catalog_register(catalog_id: str, ctx: AuthContext = Depends(require("gateways.create"))):
+        try:
+            return await engine.catalog.register_from_catalog(catalog_id, engine.gateway_service)
+        except KeyError as exc:
+            raise HTTPException(404, str(exc)) from exc
+        except GatewayConnectionError as exc:
+            raise HTTPException(502, str(exc)) from exc
+
+    # -- user management with password policy (reference: email_auth + password_policy)
+    @app.post("/auth/register", status_code=201)
+    async def register_user(request: Request, ctx: AuthContext = Depends(require("admin.update"))):
+        body = await request.json()
+        errs = engine.password_policy.validate(body.get("password", ""))
+        if errs:
+            raise HTTPException(422, "; ".join(errs))
+        auth.create_user(body["email"], body["password"], body.get("full_name", ""),
+                         bool(body.get("is_admin", False)))
+        engine.audit.record(ctx.user, "create", "user", body["email"])
+        return {"email": body["email"]}
 
     # -- well-known (reference: routers/well_known.py RFC 9728) --------------------
     @app.get("/.well-known/oauth-protected-resource")

@@ -1,0 +1,132 @@
+"""Governance services: catalog, password policy, token blocklist, content
+security (reference analogs: catalog_service, password_policy,
+token_blocklist, content_security)."""
+
+import time
+
+from mcp_context_forge_amd.services.governance import (
+    CatalogService,
+    ContentSecurity,
+    PasswordPolicy,
+    TokenBlocklist,
+)
+
+
+def test_catalog_default_and_register(run):
+    from mcp_context_forge_amd.config import Settings
+    from mcp_context_forge_amd.engine import GatewayEngine
+    from mcp_context_forge_amd.services.upstream import make_fake_time_upstream
+
+    cat = CatalogService()
+    assert cat.get("fast-time") is not None
+    assert cat.list("development")[0]["id"] == "github"
+
+    engine = GatewayEngine(Settings(database_url="sqlite://", federation_enabled=False, auth_required=False))
+
+    async def go():
+        # register via catalog with an injected in-proc client
+        entry = engine.catalog.get("fast-time")
+        gw = await engine.gateway_service.register_gateway(
+            name=entry["name"], url=entry["url"], client=make_fake_time_upstream())
+        assert gw["status"] == "active"
+        assert engine.registry.find("tool", "fast-time-server-convert_time") is not None
+        await engine.shutdown()
+
+    run(go())
+
+
+def test_catalog_yaml_load(tmp_path):
+    f = tmp_path / "cat.yml"
+    f.write_text("catalog_servers:\n  - id: x\n    name: X\n    url: http://x/mcp\n    category: misc\n")
+    cat = CatalogService(str(f))
+    assert cat.get("x")["category"] == "misc"
+
+
+def test_password_policy():
+    pp = PasswordPolicy()
+    assert pp.validate("Str0ngEnough!") == []
+    errs = pp.validate("weak")
+    assert any("10 characters" in e for e in errs)
+    assert pp.validate("alllowercase1") != []
+    assert any("common" in e for e in pp.validate("ChangeMe12")) is False  # not in list verbatim
+    assert any("common" in e for e in pp.validate("changeme"))
+
+
+def test_token_blocklist():
+    bl = TokenBlocklist()
+    bl.block("jti-1")
+    bl.block("jti-2", expires_at=time.time() - 1)
+    assert bl.is_blocked("jti-1")
+    assert not bl.is_blocked("jti-2")  # expired entry auto-clears
+    assert not bl.is_blocked(None)
+    bl2 = TokenBlocklist()
+    bl2.merge(bl.snapshot())
+    assert bl2.is_blocked("jti-1")
+
+
+def test_blocked_jwt_rejected(run):
+    import asyncio
+
+    import httpx
+
+    from mcp_context_forge_amd.auth import jwt as jwt_mod
+    from mcp_context_forge_amd.config import Settings
+    from mcp_context_forge_amd.engine import GatewayEngine
+    from mcp_context_forge_amd.transports.http_app import build_app
+
+    s = Settings(database_url="sqlite://", federation_enabled=False, auth_required=True)
+    engine = GatewayEngine(s)
+    app = build_app(engine)
+    tok = jwt_mod.create_token({"sub": "x@y.com", "jti": "revoked-1"}, s.jwt_secret_key,
+                               audience=s.jwt_audience, issuer=s.jwt_issuer, expires_minutes=60)
+    engine.token_blocklist.block("revoked-1")
+    transport = httpx.ASGITransport(app=app)
+
+    async def go():
+        async with app.router.lifespan_context(app):
+            async with httpx.AsyncClient(transport=transport, base_url="http://gw") as c:
+                r = await c.get("/tools", headers={"Authorization": f"Bearer {tok}"})
+                assert r.status_code == 401 and "revoked" in r.text.lower()
+
+    run(go())
+
+
+def test_content_security():
+    cs = ContentSecurity(max_result_bytes=100, max_content_items=2)
+    ok = {"content": [{"type": "text", "text": "small"}]}
+    assert cs.check_result(ok) == []
+    big = {"content": [{"type": "text", "text": "x" * 200}]}
+    assert any("too large" in e for e in cs.check_result(big))
+    many = {"content": [{"type": "text", "text": "x"}] * 3}
+    assert any("too many" in e for e in cs.check_result(many))
+    weird = {"content": [{"type": "wasm", "text": ""}]}
+    assert any("not allowed" in e for e in cs.check_result(weird))
+
+
+def test_password_policy_endpoint(run):
+    import httpx
+
+    from mcp_context_forge_amd.config import Settings
+    from mcp_context_forge_amd.engine import GatewayEngine
+    from mcp_context_forge_amd.transports.http_app import build_app
+
+    import base64
+
+    ADMIN = {"Authorization": "Basic " + base64.b64encode(b"admin:changeme").decode()}
+    engine = GatewayEngine(Settings(database_url="sqlite://", federation_enabled=False, auth_required=True))
+    app = build_app(engine)
+    transport = httpx.ASGITransport(app=app)
+
+    async def go():
+        async with app.router.lifespan_context(app):
+            async with httpx.AsyncClient(transport=transport, base_url="http://gw") as c:
+                r = await c.post("/auth/register", headers=ADMIN,
+                                 json={"email": "n@x.com", "password": "weak"})
+                assert r.status_code == 422
+                r = await c.post("/auth/register", headers=ADMIN,
+                                 json={"email": "n@x.com", "password": "Str0ngEnough!"})
+                assert r.status_code == 201
+                r = await c.post("/auth/login", json={"email": "n@x.com", "password": "Str0ngEnough!"})
+                assert r.status_code == 200
+
+    run(go())
