@@ -19,7 +19,7 @@ ARCH = os.environ.get("PYTORCH_ROCM_ARCH", "gfx950")
 HIPCC = os.environ.get("HIPCC", "/opt/rocm/bin/hipcc")
 
 SOURCES = ["scan.hip", "featurize.hip", "json_guard.hip", "gemm_bf16.hip", "runtime.hip",
-           "envelope.cpp", "upstream.cpp", "fastpath.cpp"]
+           "gemm_v2.hip", "envelope.cpp", "upstream.cpp", "fastpath.cpp"]
 
 
 def needs_build() -> bool:

@@ -34,6 +34,7 @@ def _load() -> ctypes.CDLL:
         "forge_featurize": [ctypes.c_void_p] * 3 + [ctypes.c_int, ctypes.c_int] + [ctypes.c_void_p] * 3,
         "forge_json_guard": [ctypes.c_void_p] * 3 + [ctypes.c_int] * 3 + [ctypes.c_void_p] * 3,
         "forge_gemm_bt": [ctypes.c_void_p] * 4 + [ctypes.c_int] * 5 + [ctypes.c_void_p],
+        "forge_gemm_bt_v2": [ctypes.c_void_p] * 4 + [ctypes.c_int] * 5 + [ctypes.c_void_p],
         "forge_gemv_head": [ctypes.c_void_p] * 4 + [ctypes.c_int] * 4 + [ctypes.c_void_p],
         "forge_rows_argmax_merge": [ctypes.c_void_p, ctypes.c_int, ctypes.c_int, ctypes.c_int, ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p],
         "forge_rows_scatter_bf16": [ctypes.c_void_p] * 3 + [ctypes.c_int] * 2 + [ctypes.c_void_p],
@@ -127,6 +128,14 @@ def gemm_bt(a: torch.Tensor, bt: torch.Tensor, bias: Optional[torch.Tensor] = No
     n, k2 = bt.shape
     assert k == k2, (a.shape, bt.shape)
     out = torch.empty((m, n), dtype=torch.bfloat16 if out_bf16 else torch.float32, device=a.device)
+    # v2 (256² tile, 4-phase ring, counted vmcnt) when shapes allow; v1 fallback
+    import os
+
+    if m % 256 == 0 and n % 256 == 0 and k % 32 == 0 and not os.environ.get("FORGE_GEMM_V1"):
+        rc = _load().forge_gemm_bt_v2(_ptr(a), _ptr(bt), _ptr(bias), _ptr(out), m, n, k,
+                                      act, 1 if out_bf16 else 0, _stream())
+        _check("forge_gemm_bt_v2", rc)
+        return out
     rc = _load().forge_gemm_bt(_ptr(a), _ptr(bt), _ptr(bias), _ptr(out), m, n, k,
                                act, 1 if out_bf16 else 0, _stream())
     _check("forge_gemm_bt", rc)

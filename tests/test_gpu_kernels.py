@@ -216,3 +216,50 @@ def test_semcache_hit_miss_ttl(hip):
     # wrong tool: miss
     hits2 = cache.resolve_hits(bv[:1].cpu().numpy(), bi[:1].cpu().numpy(), np.array([tool_hash("other")]))
     assert hits2[0] is None
+
+
+@requires_gpu
+@pytest.mark.parametrize("m,n,k", [(256, 256, 32), (256, 256, 64), (256, 256, 96),
+                                   (512, 256, 4096), (256, 512, 128)])
+def test_gemm_v2_matches_torch(hip, m, n, k):
+    """256²-tile 4-phase ring GEMM (gemm_v2.hip) vs fp32 torch — asymmetric
+    random inputs (guide G9: transpose-detecting refcheck)."""
+    torch.manual_seed(11 + m + n + k)
+    a = (torch.randn(m, k) * 0.5).bfloat16().cuda()
+    bt = (torch.randn(n, k) * 0.5).bfloat16().cuda()
+    ref = a.float().cpu() @ bt.float().cpu().t()
+    got = hip.gemm_bt(a, bt).cpu()  # dispatcher picks v2 for these shapes
+    torch.cuda.synchronize()
+    err = (got - ref).abs().max().item()
+    scale = ref.abs().max().item() + 1e-6
+    assert err / scale < 2e-2, (err, scale)
+
+
+@requires_gpu
+def test_gemm_v2_fused_epilogue(hip):
+    torch.manual_seed(13)
+    m, n, k = 256, 256, 128
+    a = (torch.randn(m, k) * 0.3).bfloat16().cuda()
+    bt = (torch.randn(n, k) * 0.3).bfloat16().cuda()
+    bias = torch.randn(n).cuda()
+    ref = torch.sigmoid(a.float().cpu() @ bt.float().cpu().t() + bias.cpu())
+    got = hip.gemm_bt(a, bt, bias, act=hip.ACT_SIGMOID).cpu()
+    assert (got - ref).abs().max().item() < 2e-2
+
+
+@requires_gpu
+def test_gemm_v2_vs_v1_same_result(hip):
+    import os
+
+    torch.manual_seed(17)
+    m, n, k = 512, 512, 512
+    a = (torch.randn(m, k) * 0.5).bfloat16().cuda()
+    bt = (torch.randn(n, k) * 0.5).bfloat16().cuda()
+    v2 = hip.gemm_bt(a, bt).cpu()
+    os.environ["FORGE_GEMM_V1"] = "1"
+    try:
+        v1 = hip.gemm_bt(a, bt).cpu()
+    finally:
+        del os.environ["FORGE_GEMM_V1"]
+    torch.cuda.synchronize()
+    assert torch.equal(v1, v2) or (v1 - v2).abs().max().item() < 1e-3
