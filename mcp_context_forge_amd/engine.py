@@ -253,6 +253,18 @@ class GatewayEngine:
             return await self.gpu_pipeline.process_batch(raws, user=user, server_id=server_id)
         return list(await asyncio.gather(*(self.handle_rpc_bytes(r, user=user, server_id=server_id) for r in raws)))
 
+    async def notify_list_changed(self, kind: str) -> None:
+        """Fan out listChanged notifications to live sessions (reference:
+        services/notification_service.py - upstream listChanged to listeners)."""
+        method = {"tool": "notifications/tools/list_changed",
+                  "resource": "notifications/resources/list_changed",
+                  "prompt": "notifications/prompts/list_changed"}.get(kind)
+        if method is None:
+            return
+        msg = {"jsonrpc": "2.0", "method": method}
+        for sid in list(self.sessions._sessions.keys()):
+            await self.sessions.broadcast(sid, msg)
+
     # -- health/version ------------------------------------------------------------
     def health(self) -> Dict[str, Any]:
         return {"status": "healthy", "uptime_s": round(time.time() - self.started_at, 1)}

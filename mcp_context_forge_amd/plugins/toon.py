@@ -151,6 +151,13 @@ _HEAD = re.compile(r"^(?P<key>\"(?:[^\"\\]|\\.)*\"|[^:\[{]+)?(?:\[(?P<n>\d+)\](?
 
 def decode(text: str) -> Any:
     lines = [ln for ln in text.splitlines() if ln.strip()]
+    # bare top-level scalar
+    if len(lines) == 1 and not _HEAD.match(lines[0].strip()):
+        return _parse_scalar(lines[0].strip())
+    if len(lines) == 1:
+        m = _HEAD.match(lines[0].strip())
+        if m and m.group("n") is None and m.group("key") is not None and ":" not in lines[0]:
+            return _parse_scalar(lines[0].strip())
     pos = 0
 
     def indent_of(ln: str) -> int:

@@ -368,6 +368,7 @@ def build_app(engine: GatewayEngine, auth: Optional[AuthService] = None) -> Fast
                     body.setdefault("integration_type", "REST" if body.get("url") else "LOCAL")
                 ent = engine.registry.create(kind, **body)
                 engine.audit.record(ctx.user, "create", kind, ent.get("id"))
+                await engine.notify_list_changed(kind)
                 return ent
             except ConflictError as exc:
                 raise HTTPException(409, str(exc)) from exc
@@ -395,6 +396,7 @@ def build_app(engine: GatewayEngine, auth: Optional[AuthService] = None) -> Fast
                 else:
                     engine.registry.delete(kind, entity_id)
                 engine.audit.record(ctx.user, "delete", kind, entity_id)
+                await engine.notify_list_changed(kind)
             except NotFoundError as exc:
                 raise HTTPException(404, str(exc)) from exc
             return Response(status_code=204)
