@@ -43,6 +43,17 @@ def main():
                                               m, n, k, 0, 0, hip._stream()))
         return out
 
+    lib.forge_gemm_bt_v2_var.argtypes = [__import__("ctypes").c_void_p] * 3 + [__import__("ctypes").c_int] * 4 + [__import__("ctypes").c_void_p]
+    lib.forge_gemm_bt_v2_var.restype = __import__("ctypes").c_int
+
+    def mkvar(var):
+        def f():
+            out = torch.empty((m, n), dtype=torch.float32, device="cuda")
+            hip._check("v2var", lib.forge_gemm_bt_v2_var(hip._ptr(a), hip._ptr(bt), hip._ptr(out),
+                                                         m, n, k, var, hip._stream()))
+            return out
+        return f
+
     # refcheck first
     r1, r2 = v1(), v2()
     torch.cuda.synchronize()
@@ -52,12 +63,19 @@ def main():
     t2 = bench(v2)
     t1b = bench(v1)
     t2b = bench(v2)
-    print(json.dumps({
+    res = {
         "shape": [m, n, k], "v1_v2_absdiff": dmax,
         "v1_tf": round(flop / min(t1, t1b) / 1e12, 1),
         "v2_tf": round(flop / min(t2, t2b) / 1e12, 1),
-        "speedup": round(min(t1, t1b) / min(t2, t2b), 3),
-    }))
+    }
+    for var in (1, 2, 3):
+        fv = mkvar(var)
+        rv = fv()
+        torch.cuda.synchronize()
+        res[f"var{var}_absdiff"] = (r1 - rv).abs().max().item()
+        tv = min(bench(fv), bench(fv))
+        res[f"var{var}_tf"] = round(flop / tv / 1e12, 1)
+    print(json.dumps(res))
 
 
 if __name__ == "__main__":

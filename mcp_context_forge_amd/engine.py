@@ -81,13 +81,44 @@ class GatewayEngine:
         self.started_at = time.time()
         self.gpu_pipeline = None  # attached lazily by gpu.pipeline when available
         self._log_level = "info"
+        self._maintenance_task = None
+        self._maintenance_stop = None
 
     # -- lifecycle ---------------------------------------------------------------
     async def startup(self) -> None:
         if self.settings.federation_enabled and self.settings.health_check_interval > 0:
             await self.gateway_service.start_health_loop()
+        # periodic maintenance (reference: lifespan background tasks — metrics
+        # buffer flush :60s, session reaper, span persistence)
+        if self._maintenance_task is None:
+            self._maintenance_stop = asyncio.Event()
+            self._maintenance_task = asyncio.create_task(self._maintenance_loop())
+
+    async def _maintenance_loop(self) -> None:
+        interval = max(5.0, min(60.0, self.settings.metrics_buffer_flush_interval))
+        while not self._maintenance_stop.is_set():
+            try:
+                await asyncio.wait_for(self._maintenance_stop.wait(), timeout=interval)
+                return
+            except asyncio.TimeoutError:
+                pass
+            try:
+                self.metrics.maybe_flush()
+                self.sessions.cleanup_expired()
+                self.observability.flush()
+                self.token_blocklist.purge_expired()
+            except Exception:  # pragma: no cover - defensive
+                logger.exception("maintenance loop error")
 
     async def shutdown(self) -> None:
+        if self._maintenance_task is not None:
+            self._maintenance_stop.set()
+            self._maintenance_task.cancel()
+            try:
+                await self._maintenance_task
+            except (asyncio.CancelledError, Exception):
+                pass
+            self._maintenance_task = None
         await self.gateway_service.stop()
         await self.a2a_service.aclose()
         await self.llm_proxy.aclose()

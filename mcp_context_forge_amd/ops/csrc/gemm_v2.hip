@@ -32,7 +32,7 @@ __device__ __forceinline__ uint32_t swz2(uint32_t L) {
     return L ^ (((L >> 6) & 3u) << 4);
 }
 
-template <int ACT, bool OUT_BF16, bool HAS_BIAS>
+template <int ACT, bool OUT_BF16, bool HAS_BIAS, int VAR = 0>
 __global__ __launch_bounds__(512, 1) void gemm_bt_v2_kernel(
     const short* __restrict__ A,   // [M,K] bf16
     const short* __restrict__ BT,  // [N,K] bf16
@@ -50,8 +50,21 @@ __global__ __launch_bounds__(512, 1) void gemm_bt_v2_kernel(
     const int wave = tid >> 6;       // 0..7
     const int wm = wave >> 2;        // 0..1  (M half)
     const int wn = wave & 3;         // 0..3  (N quarter)
-    const int tile_m = blockIdx.x * V2_BM;
-    const int tile_n = blockIdx.y * V2_BN;
+    int bm = blockIdx.x, bn = blockIdx.y;
+    if constexpr (VAR == 3) {
+        // T1 XCD-aware bijective swizzle (guide m204): 1D grid; consecutive
+        // ids on one XCD share an N-panel (bn) whose 2 MB B slab L2-fits.
+        int gx = M / V2_BM;
+        int nwg = gx * (N / V2_BN);
+        int id = blockIdx.x;
+        int q = nwg / 8, r = nwg % 8;
+        int xcd = id % 8, pos = id / 8;
+        int swz_id = (xcd < r) ? xcd * (q + 1) + pos : r * (q + 1) + (xcd - r) * q + pos;
+        bm = swz_id % gx;
+        bn = swz_id / gx;
+    }
+    const int tile_m = bm * V2_BM;
+    const int tile_n = bn * V2_BN;
     const int n_tiles = K / V2_BK;
 
     f32x4_v2 acc[8][4];
@@ -130,14 +143,14 @@ __global__ __launch_bounds__(512, 1) void gemm_bt_v2_kernel(
                 a0 = *(const bf16x8_v2*)(slot_a + swz2(r0 * V2_ROWB + a_col));
                 a1 = *(const bf16x8_v2*)(slot_a + swz2(r1 * V2_ROWB + a_col));
             }
-            __builtin_amdgcn_s_setprio(1);
+            if constexpr (VAR < 2) __builtin_amdgcn_s_setprio(1);
             #pragma unroll
             for (int fc = 0; fc < 4; ++fc) {
                 acc[2 * q][fc] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b_frag[fc], acc[2 * q][fc], 0, 0, 0);
                 acc[2 * q + 1][fc] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, b_frag[fc], acc[2 * q + 1][fc], 0, 0, 0);
             }
-            __builtin_amdgcn_s_setprio(0);
-            __builtin_amdgcn_s_barrier();
+            if constexpr (VAR < 2) __builtin_amdgcn_s_setprio(0);
+            if constexpr (VAR == 0) __builtin_amdgcn_s_barrier();
         }
         // tile boundary: retire tile kt+1's loads (counted — tile kt+2 stays in flight)
         if (kt + 1 < n_tiles) {
@@ -170,6 +183,23 @@ __global__ __launch_bounds__(512, 1) void gemm_bt_v2_kernel(
             }
         }
     }
+}
+
+extern "C" int forge_gemm_bt_v2_var(
+    const void* A, const void* BT, void* C, int M, int N, int K, int var, void* stream)
+{
+    if ((M % V2_BM) || (N % V2_BN) || (K % V2_BK)) return 9001;
+    hipStream_t s = (hipStream_t)stream;
+    dim3 grid(M / V2_BM, N / V2_BN);
+    dim3 block(512);
+    switch (var) {
+        case 0: hipLaunchKernelGGL((gemm_bt_v2_kernel<0, false, false, 0>), grid, block, 0, s, (const short*)A, (const short*)BT, nullptr, C, M, N, K); break;
+        case 1: hipLaunchKernelGGL((gemm_bt_v2_kernel<0, false, false, 1>), grid, block, 0, s, (const short*)A, (const short*)BT, nullptr, C, M, N, K); break;
+        case 2: hipLaunchKernelGGL((gemm_bt_v2_kernel<0, false, false, 2>), grid, block, 0, s, (const short*)A, (const short*)BT, nullptr, C, M, N, K); break;
+        case 3: hipLaunchKernelGGL((gemm_bt_v2_kernel<0, false, false, 3>), dim3((M / V2_BM) * (N / V2_BN)), block, 0, s, (const short*)A, (const short*)BT, nullptr, C, M, N, K); break;
+        default: return 9002;
+    }
+    return (int)hipGetLastError();
 }
 
 extern "C" int forge_gemm_bt_v2(

@@ -341,9 +341,19 @@ def build_app(engine: GatewayEngine, auth: Optional[AuthService] = None) -> Fast
         perm = {"a2a": "tools"}.get(plural, plural)
 
         @app.get(f"/{plural}", name=f"list_{plural}")
-        async def list_entities(include_inactive: bool = False,
+        async def list_entities(include_inactive: bool = False, limit: int = 0, cursor: int = 0,
                                 ctx: AuthContext = Depends(require(f"{perm}.read"))):
-            return engine.registry.list(kind, include_disabled=include_inactive)
+            # cursor pagination (reference: utils/pagination.py nextCursor)
+            items = engine.registry.list(kind, include_disabled=include_inactive)
+            if limit <= 0:
+                return items
+            page = items[cursor:cursor + limit]
+            from fastapi.encoders import jsonable_encoder
+
+            next_cursor = cursor + limit if cursor + limit < len(items) else None
+            return JSONResponse(jsonable_encoder({"items": page, "nextCursor": next_cursor,
+                                                  "total": len(items)}),
+                                headers={"X-Total-Count": str(len(items))})
 
         @app.get(f"/{plural}/{{entity_id}}", name=f"get_{plural}")
         async def get_entity(entity_id: str, ctx: AuthContext = Depends(require(f"{perm}.read"))):

@@ -325,3 +325,22 @@ def test_token_scoped_server(client_engine, run):
             assert [t["name"] for t in tools] == ["scoped_tool"]
 
     run(go())
+
+
+def test_list_pagination(client_engine, run):
+    client_factory, engine, app = client_engine
+
+    async def go():
+        for i in range(5):
+            engine.registry.create("server", name=f"pg-{i}")
+        async with client_factory() as c:
+            r = await c.get("/servers?limit=2", headers=ADMIN)
+            body = r.json()
+            assert len(body["items"]) == 2 and body["nextCursor"] == 2 and body["total"] == 5
+            r = await c.get("/servers?limit=2&cursor=4", headers=ADMIN)
+            body = r.json()
+            assert len(body["items"]) == 1 and body["nextCursor"] is None
+            r = await c.get("/servers", headers=ADMIN)
+            assert isinstance(r.json(), list)  # unpaginated default
+
+    run(go())
