@@ -1,5 +1,4 @@
-// bf16 MFMA GEMM v2 — 256×256 tile, 4-phase interleave, 4-slot LDS ring,
-// counted vmcnt, s_setprio around MFMA clusters.
+// bf16 MFMA GEMM v2 — 256×256 tile, 4-slot LDS ring, counted vmcnt.
 //
 // Applies the CDNA4 guide's verified technique stack (§5.5): T1-less variant
 // of the 256² schedule — T3+T4 (phase interleave with counted `s_waitcnt
@@ -211,8 +210,11 @@ extern "C" int forge_gemm_bt_v2(
     dim3 grid(M / V2_BM, N / V2_BN);
     dim3 block(512);
     bool hb = bias != nullptr;
+    // production configuration = VAR 2 (free-running ring + counted vmcnt;
+    // phase barriers/setprio measured NEGATIVE at this occupancy — see
+    // profiles/README.md ablation table)
     #define DISPATCH2(A_, O_, B_)                                                                \
-        hipLaunchKernelGGL((gemm_bt_v2_kernel<A_, O_, B_>), grid, block, 0, s,                   \
+        hipLaunchKernelGGL((gemm_bt_v2_kernel<A_, O_, B_, 2>), grid, block, 0, s,                \
                            (const short*)A, (const short*)BT, (const float*)bias, C, M, N, K)
     switch (act * 4 + (out_bf16 ? 2 : 0) + (hb ? 1 : 0)) {
         case 0: DISPATCH2(0, false, false); break;
