@@ -45,7 +45,15 @@ def _device() -> torch.device:
     return torch.device("cpu")
 
 
-def all_to_all_bytes(buckets: List[bytes]) -> List[bytes]:
+def new_bus_group():
+    """Dedicated communicator for the affinity bus so its rounds never
+    interleave with application collectives (ordering rule)."""
+    if not dist.is_initialized():
+        return None
+    return dist.new_group(backend=dist.get_backend())
+
+
+def all_to_all_bytes(buckets: List[bytes], group=None) -> List[bytes]:
     """Exchange byte buffers: buckets[d] goes to rank d; returns what each
     rank sent to us (index = source rank).
 
@@ -63,11 +71,11 @@ def all_to_all_bytes(buckets: List[bytes]) -> List[bytes]:
     send_sizes = torch.tensor([len(b) for b in buckets], dtype=torch.int64, device=dev)
     recv_sizes = torch.empty(world, dtype=torch.int64, device=dev)
     try:
-        dist.all_to_all_single(recv_sizes, send_sizes)
+        dist.all_to_all_single(recv_sizes, send_sizes, group=group)
     except RuntimeError:
         # gloo without alltoall support: all_gather the full matrix
         gathered: List[Any] = [None] * world
-        dist.all_gather_object(gathered, buckets)
+        dist.all_gather_object(gathered, buckets, group=group)
         me = dist.get_rank()
         return [gathered[src][me] for src in range(world)]
 
@@ -78,7 +86,7 @@ def all_to_all_bytes(buckets: List[bytes]) -> List[bytes]:
     recv_buf = torch.empty(recv_total, dtype=torch.uint8, device=dev)
     in_splits = [len(b) for b in buckets]
     out_splits = [int(x) for x in recv_sizes.tolist()]
-    dist.all_to_all_single(recv_buf, send_buf, out_splits, in_splits)
+    dist.all_to_all_single(recv_buf, send_buf, out_splits, in_splits, group=group)
     flat = recv_buf.cpu().numpy().tobytes()
     out: List[bytes] = []
     off = 0
@@ -88,10 +96,10 @@ def all_to_all_bytes(buckets: List[bytes]) -> List[bytes]:
     return out
 
 
-def all_to_all_objects(buckets: List[List[Any]]) -> List[List[Any]]:
+def all_to_all_objects(buckets: List[List[Any]], group=None) -> List[List[Any]]:
     """Object-level fan-out: buckets[d] (a list) is delivered to rank d."""
     payloads = [pickle.dumps(b) for b in buckets]
-    received = all_to_all_bytes(payloads)
+    received = all_to_all_bytes(payloads, group=group)
     return [pickle.loads(p) if p else [] for p in received]
 
 
