@@ -42,6 +42,12 @@ def main(argv=None) -> int:
     p.add_argument("--host", default="127.0.0.1")
     p.add_argument("--port", type=int, default=9000)
 
+    p = sub.add_parser("reverse-proxy", help="tunnel a local stdio MCP server out to a remote gateway")
+    p.add_argument("--url", required=True, help="remote gateway base URL")
+    p.add_argument("--stdio", required=True, help="local server command, e.g. 'python server.py'")
+    p.add_argument("--name", default=None)
+    p.add_argument("--token", default=None)
+
     p = sub.add_parser("token", help="mint an HS256 JWT for the gateway")
     p.add_argument("--user", default="admin@example.com")
     p.add_argument("--admin", action="store_true")
@@ -106,6 +112,32 @@ def main(argv=None) -> int:
             try:
                 await uvicorn.Server(config).serve()
             finally:
+                await sp.stop()
+
+        asyncio.run(run())
+        return 0
+
+    if args.cmd == "reverse-proxy":
+        from .transports.reverse_proxy import ReverseProxyClient
+        from .transports.stdio import StdioSubprocess
+
+        async def run():
+            sp = StdioSubprocess(shlex.split(args.stdio))
+            await sp.start()
+
+            async def forward(raw: bytes):
+                try:
+                    want_id = json.loads(raw).get("id")
+                except ValueError:
+                    want_id = None
+                return await sp.send_raw(raw, want_id)
+
+            rp = ReverseProxyClient(args.url, args.name or "reverse-proxy", forward, token=args.token)
+            try:
+                await rp.register()
+                await rp.serve()
+            finally:
+                await rp.aclose()
                 await sp.stop()
 
         asyncio.run(run())

@@ -42,17 +42,26 @@ class GatewayService:
 
     # -- client construction ----------------------------------------------------
     def _make_client(self, gateway: Dict[str, Any]) -> UpstreamClient:
+        import json as _json
+
         headers: Dict[str, str] = {}
+        token_provider = None
         if gateway.get("auth_type") == "bearer" and gateway.get("auth_value"):
             headers["authorization"] = f"Bearer {gateway['auth_value']}"
         elif gateway.get("auth_type") == "headers" and gateway.get("auth_value"):
-            import json as _json
-
             try:
                 headers.update(_json.loads(gateway["auth_value"]))
             except Exception:
                 pass
-        return HttpUpstreamClient(gateway["url"], headers=headers, timeout=self.settings.federation_timeout)
+        elif gateway.get("auth_type") == "oauth" and gateway.get("auth_value"):
+            # client-credentials upstream auth (reference: oauth_manager.py)
+            from ..auth.oauth import provider_from_auth_value
+
+            cfg = gateway["auth_value"]
+            token_provider = provider_from_auth_value(_json.loads(cfg) if isinstance(cfg, str) else cfg)
+        return HttpUpstreamClient(gateway["url"], headers=headers,
+                                  timeout=self.settings.federation_timeout,
+                                  token_provider=token_provider)
 
     # -- registration (reference: register_gateway :1636) -----------------------
     async def register_gateway(self, name: str, url: str, transport: str = "streamablehttp",

@@ -66,7 +66,7 @@ class HttpUpstreamClient(UpstreamClient):
     _ids = itertools.count(1)
 
     def __init__(self, url: str, headers: Optional[Dict[str, str]] = None, timeout: float = 30.0,
-                 client: Optional[httpx.AsyncClient] = None):
+                 client: Optional[httpx.AsyncClient] = None, token_provider=None):
         self.url = url
         self.base_headers = dict(headers or {})
         self.timeout = timeout
@@ -75,6 +75,8 @@ class HttpUpstreamClient(UpstreamClient):
         self.session_id: Optional[str] = None
         self._init_lock = asyncio.Lock()
         self.initialized = False
+        # OAuth client-credentials provider (reference: oauth_manager flow)
+        self.token_provider = token_provider
 
     async def _rpc(self, method: str, params: Any = None, notification: bool = False,
                    extra_headers: Optional[Dict[str, str]] = None) -> Any:
@@ -89,8 +91,15 @@ class HttpUpstreamClient(UpstreamClient):
             headers.update(extra_headers)
         if self.session_id:
             headers["mcp-session-id"] = self.session_id
+        if self.token_provider is not None:
+            headers["authorization"] = f"Bearer {await self.token_provider.get_token(self._client)}"
         try:
             resp = await self._client.post(self.url, content=json.dumps(body).encode(), headers=headers)
+            if resp.status_code == 401 and self.token_provider is not None:
+                # token re-exchange retry (reference: tool_service :5742)
+                self.token_provider.invalidate()
+                headers["authorization"] = f"Bearer {await self.token_provider.get_token(self._client)}"
+                resp = await self._client.post(self.url, content=json.dumps(body).encode(), headers=headers)
         except httpx.HTTPError as exc:
             raise UpstreamError(f"upstream {self.url} unreachable: {exc}") from exc
         sid = resp.headers.get("mcp-session-id")

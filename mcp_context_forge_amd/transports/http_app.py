@@ -516,6 +516,72 @@ def build_app(engine: GatewayEngine, auth: Optional[AuthService] = None) -> Fast
             name=body["name"], base_url=body["base_url"], api_key=body.get("api_key"),
             models=body.get("models"), default_model=body.get("default_model"))
 
+    # -- reverse proxy tunnel (reference: mcpgateway/reverse_proxy.py) -------------
+    from .reverse_proxy import TunnelRegistry, TunnelUpstream
+
+    app.state.tunnels = TunnelRegistry()
+
+    @app.post("/reverse-proxy/register", status_code=201)
+    async def rp_register(request: Request, ctx: AuthContext = Depends(require("gateways.create"))):
+        body = await request.json()
+        name = body.get("name") or f"tunnel-{uuid.uuid4().hex[:6]}"
+        tunnel = app.state.tunnels.create(name)
+        return {"tunnel_id": tunnel.tunnel_id, "name": name}
+
+    @app.get("/reverse-proxy/stream")
+    async def rp_stream(tunnel_id: str, ctx: AuthContext = Depends(require("gateways.create"))):
+        tunnel = app.state.tunnels.get(tunnel_id)
+        if tunnel is None:
+            raise HTTPException(404, "tunnel not found")
+        tunnel.connected = True
+
+        async def gen():
+            # once the client is pumping, register the tunneled upstream as a
+            # federated gateway (sync happens over the tunnel itself)
+            async def do_register():
+                try:
+                    await engine.gateway_service.register_gateway(
+                        name=tunnel.name, url=f"tunnel://{tunnel.tunnel_id}",
+                        client=TunnelUpstream(tunnel))
+                except Exception:
+                    pass
+
+            reg_task = asyncio.create_task(do_register())
+            try:
+                while True:
+                    try:
+                        body = await asyncio.wait_for(tunnel.requests.get(),
+                                                      timeout=engine.settings.sse_keepalive_interval)
+                    except asyncio.TimeoutError:
+                        yield ": keepalive\n\n"
+                        continue
+                    yield f"data: {json.dumps(body, separators=(',', ':'))}\n\n"
+            finally:
+                reg_task.cancel()
+                tunnel.connected = False
+
+        return StreamingResponse(gen(), media_type="text/event-stream")
+
+    @app.post("/reverse-proxy/respond")
+    async def rp_respond(tunnel_id: str, request: Request,
+                         ctx: AuthContext = Depends(require("gateways.create"))):
+        tunnel = app.state.tunnels.get(tunnel_id)
+        if tunnel is None:
+            raise HTTPException(404, "tunnel not found")
+        body = await request.json()
+        ok = tunnel.resolve(body.get("id"), body)
+        return {"resolved": ok}
+
+    @app.delete("/reverse-proxy/{tunnel_id}", status_code=204)
+    async def rp_delete(tunnel_id: str, ctx: AuthContext = Depends(require("gateways.create"))):
+        tunnel = app.state.tunnels.get(tunnel_id)
+        if tunnel is not None:
+            gw = engine.registry.find("gateway", tunnel.name)
+            if gw is not None:
+                await engine.gateway_service.delete_gateway(gw["id"])
+            app.state.tunnels.remove(tunnel_id)
+        return Response(status_code=204)
+
     # -- catalog (reference: services/catalog_service.py) --------------------------
     @app.get("/catalog")
     async def catalog_list(category: Optional[str] = None, ctx: AuthContext = Depends(require("tools.read"))):
