@@ -77,24 +77,39 @@ MIGRATIONS: List[Tuple[str, Callable[[Connection], None]]] = [
 ]
 
 
-def run_migrations(engine: Engine) -> List[str]:
-    """Apply pending revisions; returns the list applied (bootstrap_db analog)."""
-    applied: List[str] = []
-    with engine.begin() as conn:
-        conn.execute(
-            text(
-                "CREATE TABLE IF NOT EXISTS forge_schema_version ("
-                "revision VARCHAR(64) PRIMARY KEY, applied_at TIMESTAMP DEFAULT CURRENT_TIMESTAMP)"
-            )
-        )
-        done = {r[0] for r in conn.execute(text("SELECT revision FROM forge_schema_version"))}
-        for rev, fn in MIGRATIONS:
-            if rev in done:
-                continue
-            fn(conn)
-            conn.execute(text("INSERT INTO forge_schema_version (revision) VALUES (:r)"), {"r": rev})
-            applied.append(rev)
-    return applied
+def run_migrations(engine: Engine, retries: int = 10) -> List[str]:
+    """Apply pending revisions; returns the list applied (bootstrap_db analog).
+
+    Concurrency-safe for multi-rank startup against one shared database:
+    several processes may race `upgrade head` (exactly like N gunicorn
+    workers racing alembic in the reference) — losers retry and observe the
+    winner's revisions.
+    """
+    import time as _time
+
+    last_exc: Exception = RuntimeError("unreachable")
+    for attempt in range(retries):
+        applied: List[str] = []
+        try:
+            with engine.begin() as conn:
+                conn.execute(
+                    text(
+                        "CREATE TABLE IF NOT EXISTS forge_schema_version ("
+                        "revision VARCHAR(64) PRIMARY KEY, applied_at TIMESTAMP DEFAULT CURRENT_TIMESTAMP)"
+                    )
+                )
+                done = {r[0] for r in conn.execute(text("SELECT revision FROM forge_schema_version"))}
+                for rev, fn in MIGRATIONS:
+                    if rev in done:
+                        continue
+                    fn(conn)
+                    conn.execute(text("INSERT INTO forge_schema_version (revision) VALUES (:r)"), {"r": rev})
+                    applied.append(rev)
+            return applied
+        except Exception as exc:  # racing peer won — back off and re-read
+            last_exc = exc
+            _time.sleep(0.1 * (attempt + 1))
+    raise last_exc
 
 
 class Database:

@@ -80,6 +80,7 @@ class GatewayEngine:
         self.sessions = SessionRegistry(self.settings.session_ttl, rank)
         self.started_at = time.time()
         self.gpu_pipeline = None  # attached lazily by gpu.pipeline when available
+        self.bus = None           # RcclBus when running multi-rank (parallel/runtime.py)
         self._log_level = "info"
         self._maintenance_task = None
         self._maintenance_stop = None

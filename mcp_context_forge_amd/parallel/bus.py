@@ -60,10 +60,19 @@ class RcclBus:
         self.rounds = 0
         self.forwarded = 0
         self.handled = 0
+        self._publish_q: List[Any] = []
+        # broadcast subscriber (registry/plugin invalidation channel —
+        # reference: signed Redis pub/sub, plugins/__init__.py:46-100)
+        self.on_publish: Optional[Callable[[int, Any], None]] = None
 
     # -- public API (asyncio side) ------------------------------------------
     def owner_of(self, session_id: str) -> int:
         return stable_hash(session_id) % self.world
+
+    def publish(self, payload: Any) -> None:
+        """Deliver `payload` to every OTHER rank's on_publish next round."""
+        with self._lock:
+            self._publish_q.append(payload)
 
     async def submit(self, dest: int, payload: Any) -> Any:
         """Forward `payload` to `dest`; resolves with the handler's response."""
@@ -107,9 +116,14 @@ class RcclBus:
         with self._lock:
             outgoing, self._pending = self._pending, []
         stopping = self._stop.is_set()
+        with self._lock:
+            pubs, self._publish_q = self._publish_q, []
         buckets: List[List[Tuple[str, int, int, Any]]] = [[] for _ in range(self.world)]
         for d in range(self.world):
             buckets[d].append(("ctl", self.rank, 0, {"stopping": stopping}))
+            if d != self.rank:
+                for pub in pubs:
+                    buckets[d].append(("pub", self.rank, 0, pub))
         for dest, msg_id, payload in outgoing:
             buckets[dest].append(("req", self.rank, msg_id, payload))
         for dest, msg_id, payload in self._take_responses():
@@ -123,6 +137,9 @@ class RcclBus:
                 if kind == "ctl":
                     if payload.get("stopping"):
                         n_stopping += 1
+                elif kind == "pub":
+                    if self.on_publish is not None and self.loop is not None:
+                        self.loop.call_soon_threadsafe(self.on_publish, src, payload)
                 elif kind == "req":
                     self.forwarded += 1
                     self._execute(src, msg_id, payload)

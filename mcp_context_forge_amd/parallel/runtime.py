@@ -1,0 +1,88 @@
+"""Distributed gateway runtime: one engine per GPU rank, wired together.
+
+Reference analog: the reference's multi-worker deployment =
+gunicorn workers + Redis (session-affinity RPC, invalidation pub/sub, leader
+election). Here: torchrun launches one process per GPU; this module connects
+the GatewayEngine to the RcclBus so that
+
+  * cross-rank request forwarding works (session/tool ownership),
+  * registry mutations broadcast an invalidation that makes peer ranks
+    reload from the shared database,
+  * rank 0 is the leader for background singletons (health loop).
+"""
+
+from __future__ import annotations
+
+import asyncio
+import json
+from typing import Any, Optional
+
+from ..engine import GatewayEngine
+from . import collectives
+from .bus import RcclBus
+
+
+class DistributedGateway:
+    def __init__(self, engine: GatewayEngine, cadence_us: int = 1000):
+        self.engine = engine
+        self.rank, self.world = collectives.rank_world()
+        self.bus: Optional[RcclBus] = None
+        if self.world > 1:
+            self.bus = RcclBus(self._handle_forward, cadence_us=cadence_us)
+            self.bus.on_publish = self._on_publish
+            engine.bus = self.bus
+
+    async def start(self) -> None:
+        await self.engine.startup()
+        if self.bus is not None:
+            self.bus.loop = asyncio.get_running_loop()
+            self.bus.start()
+
+    async def stop(self) -> None:
+        if self.bus is not None:
+            self.bus.stop()
+        await self.engine.shutdown()
+
+    # -- forwarded execution (reference: _execute_forwarded_request :950) ----
+    async def _handle_forward(self, payload: Any) -> Any:
+        if isinstance(payload, dict) and payload.get("kind") == "rpc":
+            out = await self.engine.handle_rpc_bytes(payload["raw"].encode()
+                                                     if isinstance(payload["raw"], str) else payload["raw"],
+                                                     user=payload.get("user"),
+                                                     server_id=payload.get("server_id"))
+            return {"raw": out}
+        return {"error": "unknown payload kind"}
+
+    async def forward_rpc(self, dest_rank: int, raw: bytes, user: Optional[str] = None,
+                          server_id: Optional[str] = None) -> Optional[bytes]:
+        """Execute a raw JSON-RPC request on `dest_rank` and return its bytes."""
+        assert self.bus is not None
+        out = await self.bus.submit(dest_rank, {"kind": "rpc", "raw": raw, "user": user,
+                                                "server_id": server_id})
+        return out.get("raw")
+
+    def owner_of_tool(self, tool_name: str) -> int:
+        tool = self.engine.registry.lookup_tool(tool_name)
+        if tool is None:
+            return self.rank
+        gw_id = tool.get("gateway_id")
+        if gw_id:
+            try:
+                return int(self.engine.registry.get("gateway", gw_id).get("owner_rank", 0)) % self.world
+            except Exception:
+                return self.rank
+        return self.rank
+
+    # -- invalidation broadcast (reference: Redis pub/sub invalidation) ------
+    def broadcast_invalidation(self, kind: str = "registry") -> None:
+        if self.bus is not None:
+            self.bus.publish({"kind": "invalidate", "what": kind})
+
+    def _on_publish(self, src: int, payload: Any) -> None:
+        if isinstance(payload, dict) and payload.get("kind") == "invalidate":
+            # peer mutated the shared DB — refresh the hot caches
+            self.engine.registry.load_all()
+
+    @property
+    def is_leader(self) -> bool:
+        return self.rank == 0
