@@ -9,6 +9,11 @@ the GatewayEngine to the RcclBus so that
   * registry mutations broadcast an invalidation that makes peer ranks
     reload from the shared database,
   * rank 0 is the leader for background singletons (health loop).
+
+Pitfall: torch.distributed collectives BLOCK the calling thread. Never call
+`collectives.barrier()` (or any default-group collective) directly on the
+asyncio event loop of a rank that may concurrently serve forwarded bus
+requests — wrap it in `asyncio.to_thread(...)` so the loop keeps running.
 """
 
 from __future__ import annotations

@@ -26,7 +26,7 @@ async def main():
     await dg.start()
     try:
         assert dg.is_leader == (rank == 0)
-        collectives.barrier()
+        await asyncio.to_thread(collectives.barrier)
         if rank == 0:
             # leader registers a LOCAL tool (handler exists only on rank 0)
             async def only_here(args):
@@ -48,7 +48,10 @@ async def main():
             out = await dg.forward_rpc(0, raw)
             res = json.loads(out)
             assert res["result"]["structuredContent"] == {"served_by_rank": 0, "q": 9}, res
-        collectives.barrier()
+        # NB: dist.barrier() would block the event loop and starve the bus
+        # handler on the serving rank — run it in a thread (documented in
+        # parallel/runtime.py).
+        await asyncio.to_thread(collectives.barrier)
     finally:
         await dg.stop()
     if rank == 0:
