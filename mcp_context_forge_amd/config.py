@@ -84,6 +84,9 @@ class Settings(BaseModel):
     cors_allow_origins: List[str] = Field(default_factory=lambda: ["*"])
     skip_ssl_verify: bool = False
 
+    # --- passthrough headers (reference: utils/passthrough_headers.py) ---
+    passthrough_headers: List[str] = Field(default_factory=lambda: ["x-tenant-id", "x-request-id"])
+
     # --- plugins (reference: plugins/config.yaml + PLUGINS_ENABLED) ---
     plugins_enabled: bool = True
     plugin_config_file: str = "plugins/config.yaml"

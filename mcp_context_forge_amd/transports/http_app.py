@@ -215,10 +215,17 @@ def build_app(engine: GatewayEngine, auth: Optional[AuthService] = None) -> Fast
         return {"jsonrpc": "2.0", "id": rid, "result": {}}
 
     # -- JSON-RPC endpoint (THE hot path; reference: main.py:7896) ----------------
+    passthrough = [h.lower() for h in settings.passthrough_headers]
+
     @app.post("/rpc")
     async def handle_rpc(request: Request, ctx: AuthContext = Depends(get_auth)):
         raw = await request.body()
-        out = await rpc_bytes(raw, ctx)
+        # passthrough headers ride to upstreams (reference: passthrough_headers.py)
+        fwd = {h: request.headers[h] for h in passthrough if h in request.headers}
+        if fwd and (app.state.collector is None or ctx.server_id is not None):
+            out = await engine.handle_rpc_bytes(raw, user=ctx.user, server_id=ctx.server_id, headers=fwd)
+        else:
+            out = await rpc_bytes(raw, ctx)
         if out is None:
             return Response(status_code=202)
         return Response(content=out, media_type="application/json")
@@ -642,6 +649,22 @@ def build_app(engine: GatewayEngine, auth: Optional[AuthService] = None) -> Fast
     @app.get("/admin/audit")
     async def admin_audit(limit: int = 100, ctx: AuthContext = Depends(require("admin.read"))):
         return engine.audit.query(limit)
+
+    @app.get("/admin/metrics/rollups")
+    async def admin_rollups(ctx: AuthContext = Depends(require("admin.read"))):
+        # roll raw rows into hourly aggregates and return them
+        from sqlalchemy import select
+
+        from ..db.models import DbMetricRollup
+        from ..services.metrics import rollup_hourly
+
+        engine.metrics.flush()
+        rollup_hourly(engine.db)
+        with engine.db.session() as s_:
+            rows = s_.execute(select(DbMetricRollup).order_by(DbMetricRollup.hour.desc()).limit(200)).scalars().all()
+            return [{"entity_id": r.entity_id, "hour": str(r.hour), "count": r.count,
+                     "error_count": r.error_count, "avg_ms": (r.total_ms / r.count) if r.count else 0}
+                    for r in rows]
 
     @app.get("/admin/plugins")
     async def admin_plugins(ctx: AuthContext = Depends(require("admin.read"))):

@@ -344,3 +344,56 @@ def test_list_pagination(client_engine, run):
             assert isinstance(r.json(), list)  # unpaginated default
 
     run(go())
+
+
+def test_passthrough_headers_reach_upstream(client_engine, run):
+    """reference: utils/passthrough_headers.py — selected inbound headers ride
+    to the upstream dispatch."""
+    client_factory, engine, app = client_engine
+    seen = {}
+
+    async def header_echo(args):
+        return args
+
+    # header capture via a LOCAL tool is not enough (locals don't get headers);
+    # use the plugin hook: header_injector runs http_pre, but dispatch headers
+    # surface in tool_service.dispatch for MCP/REST. Easiest observable: the
+    # engine passes headers into invoke_tool → plugin ctx.headers.
+    from mcp_context_forge_amd.plugins.framework import HookType, Plugin, PluginResult
+
+    class Capture(Plugin):
+        name = "capture"
+        hooks = (HookType.TOOL_PRE_INVOKE,)
+        priority = 1
+
+        async def tool_pre_invoke(self, ctx):
+            seen.update(ctx.headers)
+            return PluginResult.ok()
+
+    engine.plugins.add(Capture())
+
+    async def go():
+        async with client_factory() as c:
+            r = await c.post("/rpc", headers={**ADMIN, "X-Tenant-Id": "acme", "X-Secret": "no"},
+                             json={"jsonrpc": "2.0", "id": 1, "method": "tools/call",
+                                   "params": {"name": "echo", "arguments": {}}})
+            assert r.status_code == 200
+            assert seen.get("x-tenant-id") == "acme"
+            assert "x-secret" not in seen
+
+    run(go())
+
+
+def test_admin_rollups_endpoint(client_engine, run):
+    client_factory, engine, app = client_engine
+
+    async def go():
+        async with client_factory() as c:
+            await c.post("/rpc", headers=ADMIN,
+                         json={"jsonrpc": "2.0", "id": 1, "method": "tools/call",
+                               "params": {"name": "echo", "arguments": {}}})
+            r = await c.get("/admin/metrics/rollups", headers=ADMIN)
+            rows = r.json()
+            assert rows and rows[0]["count"] >= 1
+
+    run(go())
