@@ -850,19 +850,21 @@ class GpuPluginPipeline:
                         if pos.size:
                             hip.store_put(self._slot_store, int(slots[pos[0]]), rbytes)
 
-        if self.breaker is not None and n_all:
-            tis = tool_idx[all_js_np]
-            for ti in np.unique(tis):
-                sel = tis == ti
-                self._breaker_record_bulk(self._meta_list[int(ti)].name, int(sel.sum()),
-                                          int(is_err[sel].sum()))
         if n_all:
             ms = (time.monotonic() - t0) * 1000.0
             tis = tool_idx[all_js_np]
-            for ti in np.unique(tis):
-                sel = tis == ti
-                self.engine.metrics.record_aggregate(self._meta_list[int(ti)].tid,
-                                                     int(sel.sum()), int(is_err[sel].sum()), ms)
+            uniq, inv = np.unique(tis, return_inverse=True)
+            counts = np.bincount(inv, minlength=uniq.size)
+            errs = np.bincount(inv, weights=is_err.astype(np.float64), minlength=uniq.size).astype(np.int64)
+            names = [self._meta_list[int(ti)].name for ti in uniq]
+            if self.breaker is not None:
+                # breaker bookkeeping only where it can change state: tools with
+                # errors this batch, or tools that already have a window/state
+                for k, name in enumerate(names):
+                    if errs[k] or name in self.breaker.state:
+                        self._breaker_record_bulk(name, int(counts[k]), int(errs[k]))
+            self.engine.metrics.record_aggregate_many(
+                [self._meta_list[int(ti)].tid for ti in uniq], counts, errs, ms)
 
     def _breaker_record_bulk(self, name: str, n: int, n_err: int) -> None:
         b = self.breaker

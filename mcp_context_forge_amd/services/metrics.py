@@ -76,6 +76,28 @@ class MetricsBuffer:
         if need_flush:
             self.flush()
 
+    def record_aggregate_many(self, tool_ids: List[str], counts, errors, response_time_ms: float) -> None:
+        """One-lock batched aggregate (GPU pipeline: one row per tool per batch)."""
+        with self._lock:
+            total = 0
+            total_err = 0
+            for tid, cnt, err in zip(tool_ids, counts, errors):
+                cnt = int(cnt)
+                err = int(err)
+                if cnt <= 0:
+                    continue
+                self._rows.append({"tool_id": tid, "response_time_ms": response_time_ms,
+                                   "is_success": err == 0, "error_message": None, "count": cnt})
+                self.latency_sum_ms[tid] += response_time_ms * cnt
+                self.latency_count[tid] += cnt
+                total += cnt
+                total_err += err
+            self.counters["tool_invocations_total"] += total
+            self.counters["tool_errors_total"] += total_err
+            need_flush = len(self._rows) >= self.max_size
+        if need_flush:
+            self.flush()
+
     def maybe_flush(self) -> None:
         if time.monotonic() - self._last_flush >= self.flush_interval:
             self.flush()
