@@ -114,12 +114,11 @@ class GpuSemanticCache:
 
     def insert_features(self, feats_bf16: torch.Tensor, rows: np.ndarray, slots: np.ndarray,
                         tool_hashes: np.ndarray) -> None:
-        """Scatter feature rows + metadata for pre-assigned slots (results are
-        stored separately — C++ slot store on the native path)."""
-        src = feats_bf16[torch.from_numpy(rows.astype(np.int64)).to(self.device)].contiguous()
-        slot_t = torch.from_numpy(slots.astype(np.int32)).to(self.device)
-        hip.rows_scatter_bf16(src, slot_t, self.keys)
-        self.valid[slot_t.long()] = 1
+        """Fused gather+scatter of feature rows into pre-assigned slots
+        (one kernel; results are stored separately — C++ slot store)."""
+        rows_t = torch.from_numpy(rows.astype(np.int32)).to(self.device, non_blocking=True)
+        slot_t = torch.from_numpy(slots.astype(np.int32)).to(self.device, non_blocking=True)
+        hip.rows_gather_scatter_bf16(feats_bf16, rows_t, slot_t, self.keys, self.valid)
         now = time.monotonic()
         self.tool_hashes[slots] = tool_hashes
         self.timestamps[slots] = now

@@ -38,6 +38,7 @@ def _load() -> ctypes.CDLL:
         "forge_gemv_head": [ctypes.c_void_p] * 4 + [ctypes.c_int] * 4 + [ctypes.c_void_p],
         "forge_rows_argmax_merge": [ctypes.c_void_p, ctypes.c_int, ctypes.c_int, ctypes.c_int, ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p],
         "forge_rows_scatter_bf16": [ctypes.c_void_p] * 3 + [ctypes.c_int] * 2 + [ctypes.c_void_p],
+        "forge_rows_gather_scatter_bf16": [ctypes.c_void_p] * 5 + [ctypes.c_int] * 2 + [ctypes.c_void_p],
         "forge_synchronize": [ctypes.c_void_p],
         "forge_parse_envelopes": [ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int] + [ctypes.c_void_p] * 7,
         "forge_upstream_call_batch": [ctypes.c_void_p] * 4 + [ctypes.c_int, ctypes.c_char_p, ctypes.c_void_p, ctypes.c_int64, ctypes.c_void_p, ctypes.c_void_p],
@@ -157,6 +158,15 @@ def rows_argmax_merge(scores: torch.Tensor, best_val: torch.Tensor, best_idx: to
     m, nc = scores.shape
     _check("forge_rows_argmax_merge", _load().forge_rows_argmax_merge(
         _ptr(scores), m, nc, idx_base, _ptr(valid), _ptr(best_val), _ptr(best_idx), _stream()))
+
+
+def rows_gather_scatter_bf16(src: torch.Tensor, src_rows: torch.Tensor, dst_slots: torch.Tensor,
+                             dst: torch.Tensor, valid: Optional[torch.Tensor] = None) -> None:
+    """dst[dst_slots[i]] = src[src_rows[i]]; valid[dst_slots[i]] = 1."""
+    r = src_rows.numel()
+    d = src.shape[1]
+    _check("forge_rows_gather_scatter_bf16", _load().forge_rows_gather_scatter_bf16(
+        _ptr(src), _ptr(src_rows), _ptr(dst_slots), _ptr(dst), _ptr(valid), r, d, _stream()))
 
 
 def rows_scatter_bf16(src: torch.Tensor, slots: torch.Tensor, dst: torch.Tensor) -> None:

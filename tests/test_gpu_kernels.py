@@ -263,3 +263,18 @@ def test_gemm_v2_vs_v1_same_result(hip):
         del os.environ["FORGE_GEMM_V1"]
     torch.cuda.synchronize()
     assert torch.equal(v1, v2) or (v1 - v2).abs().max().item() < 1e-3
+
+
+@requires_gpu
+def test_rows_gather_scatter(hip):
+    src = torch.randn(16, 64).bfloat16().cuda()
+    dst = torch.zeros(32, 64).bfloat16().cuda()
+    valid = torch.zeros(32, dtype=torch.uint8).cuda()
+    rows = torch.tensor([3, 7, 0], dtype=torch.int32).cuda()
+    slots = torch.tensor([10, 2, 31], dtype=torch.int32).cuda()
+    hip.rows_gather_scatter_bf16(src, rows, slots, dst, valid)
+    torch.cuda.synchronize()
+    for r, s in ((3, 10), (7, 2), (0, 31)):
+        assert torch.equal(dst[s], src[r])
+        assert valid[s].item() == 1
+    assert valid.sum().item() == 3
