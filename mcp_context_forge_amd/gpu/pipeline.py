@@ -448,7 +448,9 @@ class GpuPluginPipeline:
         if self.semcache is not None:
             bv, bi = self.semcache.lookup(feats)
             cache_val_t, cache_idx_t = bv[:m], bi[:m]
-        torch.cuda.synchronize()
+        # off-loop sync: a blocking synchronize would stall the event loop and
+        # starve the micro-batch collector under live HTTP load
+        await asyncio.to_thread(torch.cuda.synchronize)
 
         def mask(name):
             t = out.get(name)
@@ -640,7 +642,7 @@ class GpuPluginPipeline:
         if self.classifier is not None:
             f2, _ = hip.featurize(data2, beg2, end2, self.feat_dim)
             scores2_t = self.classifier.forward(pad_rows(f2, 128))[: len(ok_items)]
-        torch.cuda.synchronize()
+        await asyncio.to_thread(torch.cuda.synchronize)
         harm2 = harm2_t.cpu().numpy() if harm2_t is not None else np.zeros(len(ok_items), dtype=np.int64)
         scores2 = scores2_t.cpu().numpy() if scores2_t is not None else None
 
@@ -777,7 +779,7 @@ class GpuPluginPipeline:
             for b in ("pii", "regex", "harm", "postmeta"):
                 if b in self.banks:
                     masks3[b], _ = hip.scan(data3, b3, e3, self.banks[b])
-            torch.cuda.synchronize()
+            await asyncio.to_thread(torch.cuda.synchronize)
             for b, t in masks3.items():
                 h = t.cpu().numpy().view(np.uint32) != 0
                 if b == "postmeta":
