@@ -1,10 +1,11 @@
 """Adaptive micro-batch collector for interactive transports.
 
 SURVEY.md §7 "hard parts": batching an interactive RPC path without killing
-p50. Requests submitted from concurrent HTTP handlers are coalesced into one
-GPU batch when either `max_batch` requests are pending or `window_us` has
-elapsed since the first pending request — under load the window never waits;
-at low QPS a lone request pays at most the window.
+p50. Serial-consumer design: one worker loop waits for the first pending
+request, lingers `window_us`, then drains EVERYTHING pending into one batch
+and processes it; requests arriving while a batch is in flight accumulate
+for the next batch (natural backpressure → batches grow with load, latency
+stays ≈ window + one batch time at low load).
 """
 
 from __future__ import annotations
@@ -20,42 +21,45 @@ class BatchCollector:
         self.max_batch = max_batch
         self.window_s = window_us / 1e6
         self._pending: List[tuple[bytes, asyncio.Future]] = []
-        self._flush_task: Optional[asyncio.Task] = None
-        self._lock = asyncio.Lock()
+        self._wakeup: Optional[asyncio.Event] = None
+        self._worker: Optional[asyncio.Task] = None
+        self.batches = 0
+        self.max_seen = 0
+
+    def _ensure_worker(self) -> None:
+        if self._worker is None or self._worker.done():
+            self._wakeup = asyncio.Event()
+            self._worker = asyncio.create_task(self._run_loop())
 
     async def submit(self, raw: bytes) -> Optional[bytes]:
         loop = asyncio.get_running_loop()
         fut: asyncio.Future = loop.create_future()
-        async with self._lock:
-            self._pending.append((raw, fut))
-            if len(self._pending) >= self.max_batch:
-                batch = self._take()
-                asyncio.create_task(self._run(batch))
-            elif self._flush_task is None or self._flush_task.done():
-                self._flush_task = asyncio.create_task(self._delayed_flush())
+        self._ensure_worker()
+        self._pending.append((raw, fut))
+        self._wakeup.set()
         return await fut
 
-    def _take(self):
-        batch = self._pending
-        self._pending = []
-        return batch
-
-    async def _delayed_flush(self) -> None:
-        await asyncio.sleep(self.window_s)
-        async with self._lock:
-            batch = self._take()
-        if batch:
-            await self._run(batch)
-
-    async def _run(self, batch) -> None:
-        raws = [raw for raw, _ in batch]
-        try:
-            outs = await self.process(raws)
-        except Exception as exc:
-            for _, fut in batch:
+    async def _run_loop(self) -> None:
+        while True:
+            if not self._pending:
+                self._wakeup.clear()
+                await self._wakeup.wait()
+            # linger so concurrent submitters can pile on
+            if self.window_s > 0:
+                await asyncio.sleep(self.window_s)
+            batch, self._pending = self._pending[: self.max_batch], self._pending[self.max_batch:]
+            if not batch:
+                continue
+            self.batches += 1
+            self.max_seen = max(self.max_seen, len(batch))
+            raws = [raw for raw, _ in batch]
+            try:
+                outs = await self.process(raws)
+            except Exception as exc:
+                for _, fut in batch:
+                    if not fut.done():
+                        fut.set_exception(exc)
+                continue
+            for (_, fut), out in zip(batch, outs):
                 if not fut.done():
-                    fut.set_exception(exc)
-            return
-        for (_, fut), out in zip(batch, outs):
-            if not fut.done():
-                fut.set_result(out)
+                    fut.set_result(out)

@@ -108,6 +108,75 @@ class BodyLimitMiddleware(BaseHTTPMiddleware):
         return await call_next(request)
 
 
+class RpcFastPath:
+    """Pure-ASGI fast lane for POST /rpc — bypasses FastAPI routing/DI and
+    the BaseHTTPMiddleware stack for the hot endpoint, exactly the role the
+    reference gives its Rust edge runtime for /mcp (crates/mcp_runtime:
+    nginx routes the hot path to the sidecar, Python keeps the rest).
+    Auth contexts are memoized (reference: cache/auth_cache.py)."""
+
+    def __init__(self, app, fastapi_app):
+        self.app = app
+        self.fastapi = fastapi_app
+        self._auth_cache: Dict[bytes, tuple] = {}
+
+    def _auth(self, authz: Optional[bytes]):
+        import time as _t
+
+        ent = self._auth_cache.get(authz)
+        now = _t.monotonic()
+        if ent is not None and ent[1] > now:
+            return ent[0]
+        try:
+            ctx = self.fastapi.state.auth.authenticate(authz.decode() if authz else None)
+        except AuthError:
+            return None
+        if len(self._auth_cache) > 4096:
+            self._auth_cache.clear()
+        self._auth_cache[authz] = (ctx, now + 30.0)  # 30 s TTL (reference auth_cache)
+        return ctx
+
+    async def __call__(self, scope, receive, send):
+        if scope["type"] != "http" or scope["method"] != "POST" or scope["path"] != "/rpc":
+            await self.app(scope, receive, send)
+            return
+        headers = {k: v for k, v in scope["headers"]}
+        st = self.fastapi.state
+        if st.fastpath_passthrough and any(h in headers for h in st.fastpath_passthrough):
+            await self.app(scope, receive, send)  # passthrough headers need the full path
+            return
+        cl = headers.get(b"content-length")
+        if cl and cl.isdigit() and int(cl) > st.engine.settings.max_request_body_bytes:
+            await self._respond(send, 413, b'{"detail":"request body too large"}')
+            return
+        ctx = self._auth(headers.get(b"authorization"))
+        if ctx is None:
+            await self._respond(send, 401, b'{"detail":"Not authenticated"}')
+            return
+        body = b""
+        while True:
+            msg = await receive()
+            body += msg.get("body", b"")
+            if not msg.get("more_body", False):
+                break
+        collector = st.collector
+        if collector is not None and ctx.server_id is None:
+            out = await collector.submit(body)
+        else:
+            out = await st.engine.handle_rpc_bytes(body, user=ctx.user, server_id=ctx.server_id)
+        if out is None:
+            await self._respond(send, 202, b"")
+        else:
+            await self._respond(send, 200, out)
+
+    @staticmethod
+    async def _respond(send, status: int, body: bytes):
+        await send({"type": "http.response.start", "status": status,
+                    "headers": [(b"content-type", b"application/json"),
+                                (b"content-length", str(len(body)).encode())]})
+        await send({"type": "http.response.body", "body": body})
+
+
 # ---------------------------------------------------------------------------
 
 
@@ -148,6 +217,10 @@ def build_app(engine: GatewayEngine, auth: Optional[AuthService] = None) -> Fast
         app.add_middleware(SecurityHeadersMiddleware)
     app.add_middleware(CORSMiddleware, allow_origins=settings.cors_allow_origins,
                        allow_methods=["*"], allow_headers=["*"])
+    app.state.fastpath_passthrough = [h.lower().encode() for h in settings.passthrough_headers]
+    if not settings.rate_limit_enabled:
+        # hot-lane /rpc (outermost; skipped when rate limiting must apply)
+        app.add_middleware(RpcFastPath, fastapi_app=app)
 
     # -- auth dependency -------------------------------------------------------
     async def get_auth(request: Request) -> AuthContext:
