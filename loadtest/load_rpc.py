@@ -1,6 +1,7 @@
 #!/usr/bin/env python3
 """HTTP load rig for POST /rpc (reference analog: tests/hey/hey.sh —
-10k requests / 200 concurrency / tools/call payload).
+10k requests / 200 concurrency / tools/call payload). aiohttp-based: httpx
+caps out near ~100 RPS at high concurrency and would measure the client.
 
     python loadtest/load_rpc.py --url http://localhost:4444 --n 10000 --c 200
 """
@@ -11,7 +12,7 @@ import json
 import statistics
 import time
 
-import httpx
+import aiohttp
 
 
 async def main():
@@ -31,28 +32,32 @@ async def main():
                    "arguments": {"time": "2026-01-01T10:00:00Z",
                                  "source_timezone": "UTC", "target_timezone": "Asia/Tokyo"}},
     }).encode()
+    headers = {"Authorization": auth, "content-type": "application/json"}
     lat: list = []
     errors = 0
-    sem = asyncio.Semaphore(args.c)
+    per_worker = [args.n // args.c] * args.c
+    for i in range(args.n % args.c):
+        per_worker[i] += 1
 
-    async with httpx.AsyncClient(base_url=args.url, timeout=30.0,
-                                 limits=httpx.Limits(max_connections=args.c)) as client:
-        async def one():
+    connector = aiohttp.TCPConnector(limit=args.c, limit_per_host=args.c)
+    timeout = aiohttp.ClientTimeout(total=60)
+    async with aiohttp.ClientSession(connector=connector, timeout=timeout) as session:
+
+        async def worker(count: int):
             nonlocal errors
-            async with sem:
+            for _ in range(count):
                 t0 = time.monotonic()
                 try:
-                    r = await client.post("/rpc", content=payload,
-                                          headers={"Authorization": auth,
-                                                   "content-type": "application/json"})
-                    if r.status_code != 200 or b"error" in r.content[:60]:
-                        errors += 1
-                except httpx.HTTPError:
+                    async with session.post(args.url + "/rpc", data=payload, headers=headers) as r:
+                        body = await r.read()
+                        if r.status != 200 or b'"error"' in body[:60]:
+                            errors += 1
+                except Exception:
                     errors += 1
                 lat.append(time.monotonic() - t0)
 
         t_start = time.monotonic()
-        await asyncio.gather(*(one() for _ in range(args.n)))
+        await asyncio.gather(*(worker(c) for c in per_worker))
         elapsed = time.monotonic() - t_start
 
     lat.sort()
