@@ -71,6 +71,11 @@ class GatewayEngine:
         self.completion_service = CompletionService(self.registry)
         self.root_service = RootService()
         self.llm_proxy = LLMProxyService()
+        from .services.diagnostics import PerformanceService, SupportBundle, ToolOps
+
+        self.support_bundle = SupportBundle(self)
+        self.performance = PerformanceService(self)
+        self.toolops = ToolOps(self)
         self.observability = ObservabilityService(self.db)
         self.catalog = CatalogService()
         self.password_policy = PasswordPolicy()

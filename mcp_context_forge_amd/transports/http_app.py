@@ -739,6 +739,23 @@ def build_app(engine: GatewayEngine, auth: Optional[AuthService] = None) -> Fast
                      "error_count": r.error_count, "avg_ms": (r.total_ms / r.count) if r.count else 0}
                     for r in rows]
 
+    @app.get("/admin/support-bundle")
+    async def admin_support_bundle(ctx: AuthContext = Depends(require("admin.read"))):
+        return engine.support_bundle.collect()
+
+    @app.get("/admin/performance")
+    async def admin_performance(ctx: AuthContext = Depends(require("admin.read"))):
+        engine.performance.snapshot()
+        return engine.performance.history()
+
+    @app.post("/toolops/{tool_name}/test")
+    async def toolops_test(tool_name: str, count: int = 3,
+                           ctx: AuthContext = Depends(require("tools.invoke"))):
+        try:
+            return await engine.toolops.run_tests(tool_name, count)
+        except KeyError as exc:
+            raise HTTPException(404, str(exc)) from exc
+
     @app.get("/admin/plugins")
     async def admin_plugins(ctx: AuthContext = Depends(require("admin.read"))):
         return [{"name": p.name, "mode": p.mode.value, "priority": p.priority,

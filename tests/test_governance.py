@@ -130,3 +130,27 @@ def test_password_policy_endpoint(run):
                 assert r.status_code == 200
 
     run(go())
+
+
+def test_support_bundle_and_performance(run, bare_engine):
+    bundle = bare_engine.support_bundle.collect()
+    assert bundle["settings"]["jwt_secret_key"] == "***"
+    assert "entities" in bundle and "platform" in bundle
+    snap = bare_engine.performance.snapshot()
+    assert "tool_rps" in snap
+    assert bare_engine.performance.history()
+
+
+def test_toolops_schema_synthesis(run, bare_engine):
+    async def add(args):
+        return {"sum": args.get("a", 0) + args.get("b", 0)}
+
+    bare_engine.tool_service.register_local_tool(
+        "add", add, input_schema={"type": "object",
+                                  "properties": {"a": {"type": "integer"}, "b": {"type": "integer"}}})
+
+    async def go():
+        report = await bare_engine.toolops.run_tests("add", count=2)
+        assert report["total"] == 2 and report["passed"] == 2
+
+    run(go())
