@@ -688,6 +688,47 @@ def build_app(engine: GatewayEngine, auth: Optional[AuthService] = None) -> Fast
         engine.audit.record(ctx.user, "create", "user", body["email"])
         return {"email": body["email"]}
 
+    # -- SSO (reference: services/sso_service.py; routers/sso.py) ------------------
+    from ..auth.sso import SSOError, SSOService
+
+    sso = SSOService(auth, settings)
+    app.state.sso = sso
+
+    @app.post("/auth/sso/providers", status_code=201)
+    async def sso_add_provider(request: Request, ctx: AuthContext = Depends(require("admin.update"))):
+        body = await request.json()
+        try:
+            return sso.register_provider(body["name"], body["client_id"], body["client_secret"],
+                                         preset=body.get("preset", "oidc"),
+                                         **{k: v for k, v in body.items()
+                                            if k in ("authorize_url", "token_url", "userinfo_url",
+                                                     "email_field", "scopes")})
+        except SSOError as exc:
+            raise HTTPException(exc.status, str(exc)) from exc
+
+    @app.get("/auth/sso/providers")
+    async def sso_list_providers():
+        return [{"name": n} for n in sso.providers]
+
+    @app.get("/auth/sso/{provider}/login")
+    async def sso_login(provider: str, request: Request):
+        redirect_uri = str(request.base_url).rstrip("/") + f"/auth/sso/{provider}/callback"
+        try:
+            url = sso.login_url(provider, redirect_uri)
+        except SSOError as exc:
+            raise HTTPException(exc.status, str(exc)) from exc
+        from fastapi.responses import RedirectResponse
+
+        return RedirectResponse(url, status_code=302)
+
+    @app.get("/auth/sso/{provider}/callback")
+    async def sso_callback(provider: str, code: str, state: str, request: Request):
+        redirect_uri = str(request.base_url).rstrip("/") + f"/auth/sso/{provider}/callback"
+        try:
+            return await sso.handle_callback(provider, code, state, redirect_uri)
+        except SSOError as exc:
+            raise HTTPException(exc.status, str(exc)) from exc
+
     # -- well-known (reference: routers/well_known.py RFC 9728) --------------------
     @app.get("/.well-known/oauth-protected-resource")
     async def oauth_protected_resource(request: Request):

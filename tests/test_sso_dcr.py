@@ -1,0 +1,128 @@
+"""SSO OIDC flow + RFC 7591 DCR against a fake IdP (reference analogs:
+sso_service, dcr_service)."""
+
+import asyncio
+import base64
+import socket
+
+import httpx
+import pytest
+import uvicorn
+from fastapi import FastAPI, Request
+from fastapi.responses import JSONResponse
+
+from mcp_context_forge_amd.auth.sso import SSOError, dcr_register
+from mcp_context_forge_amd.config import Settings
+from mcp_context_forge_amd.engine import GatewayEngine
+from mcp_context_forge_amd.transports.http_app import build_app
+
+ADMIN = {"Authorization": "Basic " + base64.b64encode(b"admin:changeme").decode()}
+
+
+def _free_port():
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        return s.getsockname()[1]
+
+
+async def _serve(app, port):
+    config = uvicorn.Config(app, host="127.0.0.1", port=port, log_level="error")
+    server = uvicorn.Server(config)
+    task = asyncio.create_task(server.serve())
+    for _ in range(100):
+        if server.started:
+            break
+        await asyncio.sleep(0.05)
+    return server, task
+
+
+def _fake_idp():
+    idp = FastAPI()
+
+    @idp.post("/token")
+    async def token(request: Request):
+        from urllib.parse import parse_qs
+
+        form = {k: v[0] for k, v in parse_qs((await request.body()).decode()).items()}
+        if form.get("code") != "good-code" or form.get("client_id") != "gw-client":
+            return JSONResponse({"error": "invalid_grant"}, status_code=400)
+        return {"access_token": "idp-token", "token_type": "Bearer"}
+
+    @idp.get("/userinfo")
+    async def userinfo(request: Request):
+        if request.headers.get("authorization") != "Bearer idp-token":
+            return JSONResponse({"error": "unauthorized"}, status_code=401)
+        return {"email": "sso-user@corp.com", "name": "SSO User"}
+
+    @idp.post("/register")
+    async def register(request: Request):
+        body = await request.json()
+        assert body["client_name"]
+        return JSONResponse({"client_id": "dcr-123", "client_secret": "dcr-secret",
+                             "client_name": body["client_name"]}, status_code=201)
+
+    return idp
+
+
+def test_sso_flow_end_to_end():
+    async def go():
+        idp_port = _free_port()
+        idp_server, idp_task = await _serve(_fake_idp(), idp_port)
+        engine = GatewayEngine(Settings(database_url="sqlite://", federation_enabled=False,
+                                        auth_required=True))
+        app = build_app(engine)
+        transport = httpx.ASGITransport(app=app)
+        try:
+            async with app.router.lifespan_context(app):
+                async with httpx.AsyncClient(transport=transport, base_url="http://gw") as c:
+                    r = await c.post("/auth/sso/providers", headers=ADMIN, json={
+                        "name": "corp", "client_id": "gw-client", "client_secret": "gw-secret",
+                        "preset": "oidc",
+                        "authorize_url": f"http://127.0.0.1:{idp_port}/authorize",
+                        "token_url": f"http://127.0.0.1:{idp_port}/token",
+                        "userinfo_url": f"http://127.0.0.1:{idp_port}/userinfo"})
+                    assert r.status_code == 201 and "client_secret" not in r.text
+
+                    r = await c.get("/auth/sso/corp/login", follow_redirects=False)
+                    assert r.status_code == 302
+                    loc = r.headers["location"]
+                    assert loc.startswith(f"http://127.0.0.1:{idp_port}/authorize?")
+                    from urllib.parse import parse_qs, urlparse
+
+                    state = parse_qs(urlparse(loc).query)["state"][0]
+
+                    # IdP redirects back with a code; gateway exchanges + mints a JWT
+                    r = await c.get(f"/auth/sso/corp/callback?code=good-code&state={state}")
+                    assert r.status_code == 200, r.text
+                    tok = r.json()["access_token"]
+                    assert r.json()["email"] == "sso-user@corp.com"
+                    r = await c.get("/tools", headers={"Authorization": f"Bearer {tok}"})
+                    assert r.status_code == 200
+
+                    # tampered state rejected
+                    r = await c.get(f"/auth/sso/corp/callback?code=good-code&state={state[:-4]}xxxx")
+                    assert r.status_code == 403
+                    # bad code rejected
+                    r = await c.get(f"/auth/sso/corp/callback?code=bad&state={state}")
+                    assert r.status_code == 502
+        finally:
+            idp_server.should_exit = True
+            await asyncio.wait_for(idp_task, timeout=10)
+
+    asyncio.run(go())
+
+
+def test_dcr_registration():
+    async def go():
+        idp_port = _free_port()
+        idp_server, idp_task = await _serve(_fake_idp(), idp_port)
+        try:
+            creds = await dcr_register(f"http://127.0.0.1:{idp_port}/register", "mcp-gateway")
+            assert creds["client_id"] == "dcr-123" and creds["client_secret"] == "dcr-secret"
+            with pytest.raises(SSOError):
+                await dcr_register(f"http://127.0.0.1:{idp_port}/userinfo", "x")
+        finally:
+            idp_server.should_exit = True
+            await asyncio.wait_for(idp_task, timeout=10)
+
+    asyncio.run(go())
