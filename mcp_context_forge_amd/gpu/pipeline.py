@@ -176,6 +176,11 @@ class GpuPluginPipeline:
         self.cache_hits = 0
         self.post_rewrites = 0
         self.py_fallback = 0
+        # optional per-stage wall-clock accounting (FORGE_PIPELINE_TIMING=1)
+        import os as _os
+
+        self.timing_enabled = bool(_os.environ.get("FORGE_PIPELINE_TIMING"))
+        self.timing: Dict[str, float] = {}
 
     # ------------------------------------------------------------------
     # tool metadata / schema-shape bank
@@ -366,6 +371,13 @@ class GpuPluginPipeline:
         return plugin.applies_to(PluginContext(hook=HookType.TOOL_PRE_INVOKE, name=name))
 
     # ------------------------------------------------------------------
+    def _tic(self):
+        return time.monotonic() if self.timing_enabled else 0.0
+
+    def _toc(self, key: str, t0: float) -> None:
+        if self.timing_enabled:
+            self.timing[key] = self.timing.get(key, 0.0) + (time.monotonic() - t0)
+
     async def process_batch(self, raws: List[bytes], user: Optional[str] = None,
                             server_id: Optional[str] = None) -> List[Optional[bytes]]:
         self.batches += 1
@@ -373,12 +385,14 @@ class GpuPluginPipeline:
         self.requests += n
         responses: List[Optional[bytes]] = [None] * n
 
+        t0 = self._tic()
         offsets = np.zeros(n + 1, dtype=np.int64)
         for i, r in enumerate(raws):
             offsets[i + 1] = offsets[i] + len(r)
         joined = b"".join(raws)
         blob = np.frombuffer(joined, dtype=np.uint8) if joined else np.zeros(1, dtype=np.uint8)
         env = hip.parse_envelopes(blob, offsets)
+        self._toc("pack_envelope", t0)
         kind = env["kind"]
 
         other_rows = np.nonzero(kind != hip.ENV_TOOLS_CALL)[0]
@@ -424,10 +438,13 @@ class GpuPluginPipeline:
         args_b = np.ascontiguousarray(np.where(ab[rows] >= 0, ab[rows], 0).astype(np.int32))
         args_e = np.ascontiguousarray(np.where(ab[rows] >= 0, ae[rows], 0).astype(np.int32))
 
+        t_r = self._tic()
         tool_idx = hip.toolmap_resolve(self._toolmap, blob, nb, ne) if self._toolmap \
             else np.full(m, -1, dtype=np.int32)
+        self._toc("toolmap", t_r)
 
         # --- GPU pass 1 over raw argument spans ---
+        t_g = self._tic()
         self._pin_reset()
         data_gpu = self._upload(blob)
         beg_t = self._upload(args_b)
@@ -451,6 +468,8 @@ class GpuPluginPipeline:
         # off-loop sync: a blocking synchronize would stall the event loop and
         # starve the micro-batch collector under live HTTP load
         await asyncio.to_thread(torch.cuda.synchronize)
+        self._toc("gpu_pass1", t_g)
+        t_d = self._tic()
 
         def mask(name):
             t = out.get(name)
@@ -498,6 +517,8 @@ class GpuPluginPipeline:
             self._mod_tab[0], self._mod_tab[1],
             self._slot_store, self._exact_native, time.monotonic())
 
+        self._toc("decide", t_d)
+        t_a = self._tic()
         arena_b = arena.tobytes()
         answered = np.nonzero(state == hip.ST_ANSWERED)[0]
         for j in answered:
@@ -547,6 +568,7 @@ class GpuPluginPipeline:
             rewrite_dispatch = await self._rewrite_pass(
                 blob, env, rows, args_b, args_e, rewrite_js, tool_idx, hit, hit_slot, responses)
 
+        self._toc("answer_assign", t_a)
         native_js = [int(j) for j in np.nonzero(state == hip.ST_DISPATCH_NATIVE)[0]] + extra_native
         py_items = [(int(j), None) for j in np.nonzero(state == hip.ST_DISPATCH_PY)[0]] + \
             extra_py + rewrite_dispatch
@@ -694,6 +716,7 @@ class GpuPluginPipeline:
                                  native_js: List[int], py_items: List[Tuple[int, Any]],
                                  responses: List[Optional[bytes]], t0: float) -> None:
         # --- native upstream batch (C++) ---
+        t_u = self._tic()
         nat_blob = np.zeros(0, dtype=np.uint8)
         nat_beg = nat_end = np.zeros(0, dtype=np.int64)
         if native_js:
@@ -740,6 +763,8 @@ class GpuPluginPipeline:
             if conc:
                 await asyncio.gather(*(one(idx, j, a) for idx, j, a in conc))
 
+        self._toc("dispatch", t_u)
+        t_p = self._tic()
         # --- assemble one result blob (native results + python results) ---
         all_js = list(native_js) + [j for j, _ in py_items]
         n_all = len(all_js)
@@ -794,6 +819,8 @@ class GpuPluginPipeline:
         needs_host = post_flag | outschema | (toon_meta & (res_len >= toon_min)) | \
             (res_len > guard_max) | err_rows.astype(bool)
 
+        self._toc("post_scan", t_p)
+        t_f = self._tic()
         now = time.monotonic()
         arena2, rb2, re2, is_err, cacheable = hip.finalize(
             blob, id_b, id_e, args_b, args_e, tool_idx,
@@ -867,6 +894,7 @@ class GpuPluginPipeline:
                         self._breaker_record_bulk(name, int(counts[k]), int(errs[k]))
             self.engine.metrics.record_aggregate_many(
                 [self._meta_list[int(ti)].tid for ti in uniq], counts, errs, ms)
+        self._toc("finalize", t_f)
 
     def _breaker_record_bulk(self, name: str, n: int, n_err: int) -> None:
         b = self.breaker
@@ -943,6 +971,7 @@ class GpuPluginPipeline:
 
     def stats(self) -> Dict[str, Any]:
         out = {
+            "timing_s": {k: round(v, 4) for k, v in self.timing.items()} if self.timing_enabled else None,
             "batches": self.batches, "requests": self.requests, "fast_path": self.fast_path,
             "slow_path": self.slow_path, "blocked": self.blocked, "cache_hits": self.cache_hits,
             "post_rewrites": self.post_rewrites, "py_fallback": self.py_fallback,
