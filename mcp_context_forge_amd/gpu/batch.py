@@ -58,8 +58,7 @@ def parse_batch(raws: List[bytes]) -> List[ParsedRequest]:
 def pack_texts(texts: List[bytes], device: str = "cuda") -> Tuple[torch.Tensor, torch.Tensor, torch.Tensor]:
     """Pack byte strings into (data u8 [total], beg i32 [B], end i32 [B]) on device."""
     offsets = np.zeros(len(texts) + 1, dtype=np.int32)
-    for i, t in enumerate(texts):
-        offsets[i + 1] = offsets[i] + len(t)
+    np.cumsum(np.fromiter(map(len, texts), dtype=np.int32, count=len(texts)), out=offsets[1:])
     blob = b"".join(texts)
     data_np = np.frombuffer(blob, dtype=np.uint8) if blob else np.zeros(1, dtype=np.uint8)
     data = torch.from_numpy(data_np.copy()).to(device, non_blocking=True)

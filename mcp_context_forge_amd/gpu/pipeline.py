@@ -387,8 +387,7 @@ class GpuPluginPipeline:
 
         t0 = self._tic()
         offsets = np.zeros(n + 1, dtype=np.int64)
-        for i, r in enumerate(raws):
-            offsets[i + 1] = offsets[i] + len(r)
+        np.cumsum(np.fromiter(map(len, raws), dtype=np.int64, count=n), out=offsets[1:])
         joined = b"".join(raws)
         blob = np.frombuffer(joined, dtype=np.uint8) if joined else np.zeros(1, dtype=np.uint8)
         env = hip.parse_envelopes(blob, offsets)
