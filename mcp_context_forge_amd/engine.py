@@ -82,10 +82,11 @@ class GatewayEngine:
         self.token_blocklist = TokenBlocklist()
         self.content_security = ContentSecurity()
         self.audit = AuditTrail(self.db)
-        self.sessions = SessionRegistry(self.settings.session_ttl, rank)
+        self.sessions = SessionRegistry(self.settings.session_ttl, rank, world_size=world_size)
         self.started_at = time.time()
         self.gpu_pipeline = None  # attached lazily by gpu.pipeline when available
         self.bus = None           # RcclBus when running multi-rank (parallel/runtime.py)
+        self.forward_rpc = None   # set by DistributedGateway: (dest, raw, ...) -> bytes
         self._log_level = "info"
         self._maintenance_task = None
         self._maintenance_stop = None

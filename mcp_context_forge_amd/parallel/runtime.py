@@ -36,6 +36,7 @@ class DistributedGateway:
             self.bus = RcclBus(self._handle_forward, cadence_us=cadence_us)
             self.bus.on_publish = self._on_publish
             engine.bus = self.bus
+            engine.forward_rpc = self.forward_rpc
 
     async def start(self) -> None:
         await self.engine.startup()

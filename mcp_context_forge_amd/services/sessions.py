@@ -79,15 +79,27 @@ class EventStore:
 class SessionRegistry:
     """Memory-backend session registry (reference: cache/session_registry.py:105-215)."""
 
-    def __init__(self, ttl_s: float = 3600.0, rank: int = 0, event_store: Optional[EventStore] = None):
+    def __init__(self, ttl_s: float = 3600.0, rank: int = 0, event_store: Optional[EventStore] = None,
+                 world_size: int = 1):
         self.ttl_s = ttl_s
         self.rank = rank
+        self.world_size = max(1, world_size)
         self._sessions: Dict[str, Session] = {}
         self.event_store = event_store or EventStore()
 
     def create(self, transport: str = "streamablehttp", server_id: Optional[str] = None,
                user: Optional[str] = None, session_id: Optional[str] = None) -> Session:
-        sid = session_id or uuid.uuid4().hex
+        sid = session_id
+        if sid is None:
+            # multi-rank: pick an id whose stable hash maps to THIS rank, so
+            # any rank can route a session to its owner without shared state
+            # (reference analog: session_affinity register_session_owner :664)
+            from ..parallel.bus import stable_hash
+
+            while True:
+                sid = uuid.uuid4().hex
+                if self.world_size == 1 or stable_hash(sid) % self.world_size == self.rank:
+                    break
         sess = Session(session_id=sid, transport=transport, server_id=server_id, user=user, owner_rank=self.rank)
         self._sessions[sid] = sess
         return sess

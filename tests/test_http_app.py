@@ -397,3 +397,13 @@ def test_admin_rollups_endpoint(client_engine, run):
             assert rows and rows[0]["count"] >= 1
 
     run(go())
+
+
+def test_session_ids_hash_to_owner_rank():
+    from mcp_context_forge_amd.parallel.bus import stable_hash
+    from mcp_context_forge_amd.services.sessions import SessionRegistry
+
+    reg = SessionRegistry(rank=2, world_size=4)
+    for _ in range(16):
+        s = reg.create()
+        assert stable_hash(s.session_id) % 4 == 2
