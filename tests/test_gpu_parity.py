@@ -29,18 +29,29 @@ TRAFFIC = [
     ("fast-time-convert_time", {"time": "x"}),                                  # schema violation (missing required)
     ("fast-time-echo", {"msg": "unicode café text"}),                      # \u escape → slow path
     ("fast-time-get_system_time", {"timezone": "UTC"}),
+    ("native-time-convert_time", {"time": "2026-03-03T03:03:03Z",              # native C++ upstream
+                                  "source_timezone": "UTC", "target_timezone": "UTC"}),
+    ("native-time-echo", {"nested": {"deep": [1, 2, {"x": "y"}]}}),            # nested args (schema nest trigger)
+    ("fast-time-convert_time", {"time": "2026-01-01T10:00:00Z",                # wrong type for required field
+                                "source_timezone": 42, "target_timezone": "UTC"}),
+    ("fast-time-echo", {"msg": "x" * 3000}),                                    # large payload
+    ("gone-tool", {}),                                                          # unreachable tool
 ]
 
 
 async def _build(gpu: bool):
     from mcp_context_forge_amd.config import Settings
     from mcp_context_forge_amd.engine import GatewayEngine
-    from mcp_context_forge_amd.services.upstream import make_fake_time_upstream
+    from mcp_context_forge_amd.services.upstream import NativeInProcUpstream, make_fake_time_upstream
 
     settings = Settings(database_url="sqlite://", federation_enabled=False, auth_required=False,
                         gpu_enabled=gpu, gpu_semcache_capacity=1024)
     e = GatewayEngine(settings)
     await e.gateway_service.register_gateway(name="fast-time", url="inproc://t", client=make_fake_time_upstream())
+    await e.gateway_service.register_gateway(name="native-time", url="inproc://n", client=NativeInProcUpstream())
+    # an unreachable tool (reference: reachable=False → -32002)
+    e.registry.create("tool", name="gone-tool", original_name="gone-tool",
+                      integration_type="LOCAL", reachable=False)
     if gpu:
         assert e.enable_gpu()
     return e
