@@ -32,6 +32,12 @@ def main(argv=None) -> int:
 
     sub.add_parser("stdio", help="run the gateway engine as a stdio MCP server")
 
+    p = sub.add_parser("edge-worker", help="internal: HTTP worker forwarding to a GPU owner")
+    p.add_argument("--host", default="0.0.0.0")
+    p.add_argument("--port", type=int, required=True)
+    p.add_argument("--owner-sock", required=True)
+    p.add_argument("--owner-http", required=True)
+
     p = sub.add_parser("wrapper", help="stdio bridge to a remote gateway (reference: wrapper.py)")
     p.add_argument("--url", required=True)
     p.add_argument("--token", default=None)
@@ -73,7 +79,42 @@ def main(argv=None) -> int:
             settings.port = args.port
         import uvicorn
 
+        if args.workers and args.workers > 1:
+            # multi-worker edge: owner on a private port + N public workers
+            import os as _os
+            import subprocess as _sp
+
+            public_port = settings.port
+            private_port = public_port + 1
+            settings.edge_socket = f"/tmp/forge-edge-{_os.getpid()}.sock"
+            procs = []
+            env = dict(_os.environ)
+            for _ in range(args.workers):
+                procs.append(_sp.Popen([sys.executable, "-m", "mcp_context_forge_amd", "edge-worker",
+                                        "--host", settings.host, "--port", str(public_port),
+                                        "--owner-sock", settings.edge_socket,
+                                        "--owner-http", f"http://127.0.0.1:{private_port}"], env=env))
+            try:
+                uvicorn.run(build_app(GatewayEngine(settings)), host="127.0.0.1", port=private_port)
+            finally:
+                for p_ in procs:
+                    p_.terminate()
+            return 0
         uvicorn.run(build_app(GatewayEngine(settings)), host=settings.host, port=settings.port)
+        return 0
+
+    if args.cmd == "edge-worker":
+        from .transports.edge import run_worker
+
+        import time as _time
+
+        for _ in range(120):  # wait for the owner socket
+            import os as _os
+
+            if _os.path.exists(args.owner_sock):
+                break
+            _time.sleep(0.5)
+        run_worker(args.host, args.port, args.owner_sock, args.owner_http)
         return 0
 
     if args.cmd == "stdio":

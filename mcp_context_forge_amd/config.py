@@ -118,6 +118,10 @@ class Settings(BaseModel):
     rank: int = 0
     upstream_shards: int = 64         # upstreams per GPU rank
 
+    # --- multi-worker HTTP edge (transports/edge.py) ---
+    edge_socket: str = ""             # owner-side unix socket path ("" = disabled)
+    edge_workers: int = 0             # worker processes to spawn from `serve --workers`
+
     # --- well-known / misc ---
     docs_enabled: bool = True
     version: str = "0.1.0"

@@ -195,7 +195,17 @@ def build_app(engine: GatewayEngine, auth: Optional[AuthService] = None) -> Fast
             app_.state.collector = BatchCollector(engine.process_rpc_batch,
                                                   max_batch=settings.gpu_batch_max_requests,
                                                   window_us=settings.gpu_batch_window_us)
+        owner_server = None
+        if settings.edge_socket:
+            # owner side of the multi-worker edge (transports/edge.py)
+            from .edge import GpuOwnerServer
+
+            owner_server = GpuOwnerServer(engine, app_.state.collector, settings.edge_socket)
+            await owner_server.start()
+            app_.state.owner_server = owner_server
         yield
+        if owner_server is not None:
+            await owner_server.stop()
         await engine.shutdown()
 
     app = FastAPI(title=settings.app_name, version=settings.version, lifespan=lifespan,
