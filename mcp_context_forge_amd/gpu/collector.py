@@ -39,6 +39,18 @@ class BatchCollector:
         self._wakeup.set()
         return await fut
 
+    async def submit_many(self, raws: List[bytes]) -> List[Optional[bytes]]:
+        """Submit a pre-formed group (e.g. one edge frame) with ONE future for
+        the whole group — the owner loop pays O(frames), not O(requests)."""
+        if not raws:
+            return []
+        loop = asyncio.get_running_loop()
+        fut: asyncio.Future = loop.create_future()
+        self._ensure_worker()
+        self._pending.append((raws, fut))
+        self._wakeup.set()
+        return await fut
+
     async def _run_loop(self) -> None:
         while True:
             if not self._pending:
@@ -51,15 +63,24 @@ class BatchCollector:
             if not batch:
                 continue
             self.batches += 1
-            self.max_seen = max(self.max_seen, len(batch))
-            raws = [raw for raw, _ in batch]
+            # flatten: an entry is a single bytes or a list (one frame)
+            raws: List[bytes] = []
+            spans: List[tuple] = []
+            for item, fut in batch:
+                if isinstance(item, list):
+                    spans.append((len(raws), len(item), fut, True))
+                    raws.extend(item)
+                else:
+                    spans.append((len(raws), 1, fut, False))
+                    raws.append(item)
+            self.max_seen = max(self.max_seen, len(raws))
             try:
                 outs = await self.process(raws)
             except Exception as exc:
-                for _, fut in batch:
+                for _, _n, fut, _f in spans:
                     if not fut.done():
                         fut.set_exception(exc)
                 continue
-            for (_, fut), out in zip(batch, outs):
+            for start, n, fut, is_frame in spans:
                 if not fut.done():
-                    fut.set_result(out)
+                    fut.set_result(outs[start:start + n] if is_frame else outs[start])

@@ -63,10 +63,12 @@ class GpuOwnerServer:
             self._server.close()
             await self._server.wait_closed()
 
-    async def _process_one(self, raw: bytes) -> Optional[bytes]:
+    async def _process_frame(self, raws: List[bytes]) -> List[Optional[bytes]]:
         if self.collector is not None:
-            return await self.collector.submit(raw)
-        return await self.engine.handle_rpc_bytes(raw)
+            # one future per FRAME (not per request): the owner loop pays
+            # O(frames) bookkeeping while still coalescing across workers
+            return await self.collector.submit_many(raws)
+        return await self.engine.process_rpc_batch(raws)
 
     async def _handle(self, reader: asyncio.StreamReader, writer: asyncio.StreamWriter) -> None:
         write_lock = asyncio.Lock()
@@ -86,7 +88,7 @@ class GpuOwnerServer:
                 self.requests += len(items)
 
                 async def run_frame(items=items):
-                    outs = await asyncio.gather(*(self._process_one(b) for (_i, b) in items))
+                    outs = await self._process_frame([b for (_i, b) in items])
                     parts = [_U32.pack(len(items))]
                     for (req_id, _b), out in zip(items, outs):
                         parts.append(_U64.pack(req_id))
