@@ -9,9 +9,10 @@ WORKERS=${1:-6}
 CLIENTS=${2:-2}
 CONC=${3:-400}
 NREQ=${4:-40000}
-PRIV=9444
-PUB=8444
-SOCK=/tmp/forge-edge.sock
+TAG=${5:-0}   # distinct ports per invocation: stale SO_REUSEPORT listeners
+PRIV=$((9444 + TAG))   # from a previous run would otherwise steal connections
+PUB=$((8444 + TAG))
+SOCK=/tmp/forge-edge-$TAG.sock
 export PYTHONPATH=/root/repo
 mkdir -p gpurun_out
 PIDS=()
