@@ -277,8 +277,13 @@ def run_worker(public_host: str, public_port: int, owner_sock: str, owner_http: 
     import uvicorn
 
     app = build_worker_app(owner_sock, owner_http)
-    sock = socket.socket(socket.AF_INET, socket.SOCK_STREAM)
+    # proto MUST be IPPROTO_TCP: asyncio only sets TCP_NODELAY on accepted
+    # transports when sock.proto == IPPROTO_TCP, and accepted sockets inherit
+    # the listener's proto. With the default proto=0, Nagle stays on and the
+    # h11 two-write response stalls ~40 ms on the client's delayed ACK.
+    sock = socket.socket(socket.AF_INET, socket.SOCK_STREAM, socket.IPPROTO_TCP)
     sock.setsockopt(socket.SOL_SOCKET, socket.SO_REUSEADDR, 1)
+    sock.setsockopt(socket.IPPROTO_TCP, socket.TCP_NODELAY, 1)
     if hasattr(socket, "SO_REUSEPORT"):
         sock.setsockopt(socket.SOL_SOCKET, socket.SO_REUSEPORT, 1)
     sock.bind((public_host, public_port))
