@@ -70,10 +70,17 @@ def _rev_0002_tool_metrics_count(conn: Connection) -> None:
         conn.exec_driver_sql("ALTER TABLE tool_metrics ADD COLUMN IF NOT EXISTS count INTEGER DEFAULT 1")
 
 
+def _rev_0003_plugin_bindings(conn: Connection) -> None:
+    """Per-tool plugin bindings table. No-op for fresh DBs (0001 create_all
+    already built it from the current model)."""
+    Base.metadata.tables["plugin_bindings"].create(conn, checkfirst=True)
+
+
 # Linear chain: (revision_id, apply_fn). Append-only.
 MIGRATIONS: List[Tuple[str, Callable[[Connection], None]]] = [
     ("0001_initial_registry", _rev_0001_initial),
     ("0002_tool_metrics_count", _rev_0002_tool_metrics_count),
+    ("0003_plugin_bindings", _rev_0003_plugin_bindings),
 ]
 
 

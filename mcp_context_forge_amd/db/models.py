@@ -336,6 +336,25 @@ class DbStructuredLog(Base):
     context: Mapped[dict] = mapped_column(JSON, default=dict)
 
 
+class DbPluginBinding(Base):
+    """Per-tool plugin binding: attach/override a plugin for one tool
+    (reference: db.py:6875 tool_plugin_bindings + routers/tool_plugin_bindings.py).
+
+    `name` is the composite unique key "<tool_name>::<plugin_name>"."""
+
+    __tablename__ = "plugin_bindings"
+
+    id: Mapped[str] = mapped_column(String(36), primary_key=True, default=_uuid)
+    name: Mapped[str] = mapped_column(String(512), unique=True, index=True)
+    tool_name: Mapped[str] = mapped_column(String(255), index=True)
+    plugin_name: Mapped[str] = mapped_column(String(255))
+    mode: Mapped[Optional[str]] = mapped_column(String(32), nullable=True)  # enforce|permissive|disabled|None=keep
+    config: Mapped[Optional[dict]] = mapped_column(JSON, nullable=True)     # per-tool config override
+    enabled: Mapped[bool] = mapped_column(Boolean, default=True)
+    created_at: Mapped[datetime.datetime] = mapped_column(DateTime, default=utcnow)
+    updated_at: Mapped[datetime.datetime] = mapped_column(DateTime, default=utcnow, onupdate=utcnow)
+
+
 class DbGlobalConfig(Base):
     """Runtime-mutable global config row (reference: db.py:2550 GlobalConfig)."""
 

@@ -83,6 +83,7 @@ class GatewayEngine:
         self.content_security = ContentSecurity()
         self.audit = AuditTrail(self.db)
         self.sessions = SessionRegistry(self.settings.session_ttl, rank, world_size=world_size)
+        self.sync_plugin_bindings()
         self.started_at = time.time()
         self.gpu_pipeline = None  # attached lazily by gpu.pipeline when available
         self.bus = None           # RcclBus when running multi-rank (parallel/runtime.py)
@@ -290,6 +291,41 @@ class GatewayEngine:
         if self.gpu_pipeline is not None:
             return await self.gpu_pipeline.process_batch(raws, user=user, server_id=server_id)
         return list(await asyncio.gather(*(self.handle_rpc_bytes(r, user=user, server_id=server_id) for r in raws)))
+
+    # -- per-tool plugin bindings (reference: routers/tool_plugin_bindings.py) --
+    def sync_plugin_bindings(self) -> None:
+        """Load registry plugin_binding rows into the plugin manager
+        (tool -> plugin -> {mode, config}); bumps plugins.version so the
+        GPU pipeline recompiles its per-tool flag/host-chain tables."""
+        bmap: Dict[str, Dict[str, Dict[str, Any]]] = {}
+        for row in self.registry.list("plugin_binding"):
+            if not row.get("enabled", True):
+                continue
+            bmap.setdefault(row["tool_name"], {})[row["plugin_name"]] = {
+                "mode": row.get("mode"), "config": row.get("config")}
+        self.plugins.set_bindings(bmap)
+
+    def set_plugin_binding(self, tool_name: str, plugin_name: str,
+                           mode: Optional[str] = None, config: Optional[dict] = None) -> Dict[str, Any]:
+        key = f"{tool_name}::{plugin_name}"
+        existing = self.registry.find("plugin_binding", key)
+        if existing is not None:
+            ent = self.registry.update("plugin_binding", existing["id"], mode=mode, config=config, enabled=True)
+        else:
+            ent = self.registry.create("plugin_binding", name=key, tool_name=tool_name,
+                                       plugin_name=plugin_name, mode=mode, config=config)
+        self.sync_plugin_bindings()
+        return ent
+
+    def delete_plugin_binding(self, tool_name: str, plugin_name: str) -> None:
+        key = f"{tool_name}::{plugin_name}"
+        existing = self.registry.find("plugin_binding", key)
+        if existing is None:
+            from .registry.registry import NotFoundError
+
+            raise NotFoundError(f"binding {key!r} not found")
+        self.registry.delete("plugin_binding", existing["id"])
+        self.sync_plugin_bindings()
 
     async def notify_list_changed(self, kind: str) -> None:
         """Fan out listChanged notifications to live sessions (reference:

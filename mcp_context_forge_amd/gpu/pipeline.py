@@ -66,7 +66,7 @@ _TYPED_PAT = {
 class _ToolMeta:
     __slots__ = ("tool", "name", "tid", "itype", "thash", "native_kind", "native_client",
                  "client", "handler", "schema_mode", "required_bits", "typed_pairs",
-                 "has_output_schema", "original_name", "reachable")
+                 "has_output_schema", "original_name", "reachable", "host_chain")
 
     def __init__(self):
         self.schema_mode = "host"   # "trivial" | "fast" | "host"
@@ -77,6 +77,7 @@ class _ToolMeta:
         self.client = None
         self.handler = None
         self.has_output_schema = False
+        self.host_chain = False     # per-tool plugin binding forces the CPU chain
 
 
 class GpuPluginPipeline:
@@ -176,6 +177,7 @@ class GpuPluginPipeline:
         self.cache_hits = 0
         self.post_rewrites = 0
         self.py_fallback = 0
+        self.host_bound = 0   # requests routed to the CPU chain by plugin bindings
         # optional per-stage wall-clock accounting (FORGE_PIPELINE_TIMING=1)
         import os as _os
 
@@ -320,24 +322,35 @@ class GpuPluginPipeline:
         self._t_thash = np.array([m.thash for m in metas], dtype=np.int64) if nt else np.zeros(1, dtype=np.int64)
         self._t_index = {m.name: i for i, m in enumerate(metas)}
 
+        # per-tool plugin bindings: mode flips for the bank plugins map onto
+        # the flag bits; config overrides or bindings on non-bank plugins
+        # force the full CPU chain for that tool (correctness over speed)
+        mgr = self.engine.plugins
+        bank_names = {p.name for p in (self.deny, self.pii, self.regex, self.normalizer,
+                                       self.moderation, self.harm, self.schema_guard) if p is not None}
         flags = np.zeros(max(nt, 1), dtype=np.uint32)
+        hostbound = np.zeros(max(nt, 1), dtype=bool)
         for i, m in enumerate(metas):
+            bmap = mgr.bindings_for_tool(m.name)
+            m.host_chain = any(b.get("config") or (pname not in bank_names)
+                               for pname, b in bmap.items())
+            hostbound[i] = m.host_chain
             f = 0
             if m.reachable:
                 f |= hip.TF_REACHABLE
-            if self._enforcing(self.deny) and self._applies(self.deny, m.name):
+            if self._active(self.deny, m.name, block_class=True):
                 f |= hip.TF_DENY
-            if self.pii is not None and self._applies(self.pii, m.name):
+            if self._active(self.pii, m.name, block_class=False):
                 f |= hip.TF_PII
-            if self.regex is not None and self._applies(self.regex, m.name):
+            if self._active(self.regex, m.name, block_class=False):
                 f |= hip.TF_REGEX
-            if self.normalizer is not None and self._applies(self.normalizer, m.name):
+            if self._active(self.normalizer, m.name, block_class=False):
                 f |= hip.TF_NORM
-            if self._enforcing(self.moderation) and self._applies(self.moderation, m.name):
+            if self._active(self.moderation, m.name, block_class=True):
                 f |= hip.TF_MOD
-            if self._enforcing(self.harm) and self._applies(self.harm, m.name):
+            if self._active(self.harm, m.name, block_class=True):
                 f |= hip.TF_HARM
-            if self._enforcing(self.schema_guard):
+            if self._active(self.schema_guard, m.name, block_class=True):
                 if m.schema_mode == "fast":
                     f |= hip.TF_SCHEMA_FAST
                 elif m.schema_mode == "host":
@@ -348,6 +361,7 @@ class GpuPluginPipeline:
                 f |= hip.TF_EXACT
             flags[i] = f
         self._t_flags = flags
+        self._t_hostbound = hostbound if hostbound.any() else None
         self._meta_gen = self.engine.registry.generation
         self._plugins_ver = getattr(self.engine.plugins, "version", 0)
 
@@ -360,6 +374,20 @@ class GpuPluginPipeline:
     # ------------------------------------------------------------------
     def _enforcing(self, plugin) -> bool:
         return plugin is not None and plugin.mode in (PluginMode.ENFORCE, PluginMode.ENFORCE_IGNORE_ERROR)
+
+    def _active(self, plugin, tool_name: str, block_class: bool) -> bool:
+        """Per-tool effective activation of a GPU bank: a plugin binding may
+        flip the mode for this tool (reference: tool_plugin_bindings).
+        block_class banks (deny/harm/moderation/schema) only fire in enforce
+        modes; rewrite banks (pii/regex/normalize) fire unless disabled."""
+        if plugin is None:
+            return False
+        mode = self.engine.plugins.effective_mode(plugin, tool_name)
+        if mode == PluginMode.DISABLED:
+            return False
+        if block_class and mode == PluginMode.PERMISSIVE:
+            return False
+        return self._applies(plugin, tool_name)
 
     def _applies(self, plugin, name: str) -> bool:
         if plugin is None:
@@ -441,6 +469,35 @@ class GpuPluginPipeline:
         tool_idx = hip.toolmap_resolve(self._toolmap, blob, nb, ne) if self._toolmap \
             else np.full(m, -1, dtype=np.int32)
         self._toc("toolmap", t_r)
+
+        # tools with semantics-altering plugin bindings run the full CPU
+        # chain (the binding-aware invoke_hook path); everything else stays
+        # on the GPU fast path
+        hostbound = getattr(self, "_t_hostbound", None)
+        if hostbound is not None:
+            hb_mask = np.zeros(m, dtype=bool)
+            valid = tool_idx >= 0
+            hb_mask[valid] = hostbound[tool_idx[valid]]
+            if hb_mask.any():
+                hb_rows = rows[hb_mask]
+                self.host_bound += int(hb_rows.size)
+                outs = await asyncio.gather(
+                    *(self.engine.handle_rpc_bytes(raws[int(i)], user=user, server_id=server_id)
+                      for i in hb_rows))
+                for i, out in zip(hb_rows, outs):
+                    responses[int(i)] = out
+                keep = ~hb_mask
+                rows = rows[keep]
+                m = rows.shape[0]
+                if m == 0:
+                    return
+                tool_idx = np.ascontiguousarray(tool_idx[keep])
+                nb = np.ascontiguousarray(nb[keep])
+                ne = np.ascontiguousarray(ne[keep])
+                id_b = np.ascontiguousarray(id_b[keep])
+                id_e = np.ascontiguousarray(id_e[keep])
+                args_b = np.ascontiguousarray(args_b[keep])
+                args_e = np.ascontiguousarray(args_e[keep])
 
         # --- GPU pass 1 over raw argument spans ---
         t_g = self._tic()
@@ -973,6 +1030,7 @@ class GpuPluginPipeline:
             "timing_s": {k: round(v, 4) for k, v in self.timing.items()} if self.timing_enabled else None,
             "batches": self.batches, "requests": self.requests, "fast_path": self.fast_path,
             "slow_path": self.slow_path, "blocked": self.blocked, "cache_hits": self.cache_hits,
+            "host_bound": self.host_bound,
             "post_rewrites": self.post_rewrites, "py_fallback": self.py_fallback,
             "banks": {k: {"states": v.n_states, "classes": v.n_classes} for k, v in self.banks.items()},
         }

@@ -176,6 +176,10 @@ class PluginManager:
         self.version = 0  # bumped on runtime plugin changes (GPU pipeline re-syncs)
         self.global_state: Dict[str, Any] = {}
         self._stats: Dict[str, Dict[str, Any]] = {}
+        # per-tool bindings (reference: tool_plugin_bindings db.py:6875):
+        # tool name -> plugin name -> {"mode": str|None, "config": dict|None}
+        self.bindings: Dict[str, Dict[str, Dict[str, Any]]] = {}
+        self._derived: Dict[tuple, Plugin] = {}  # (plugin, tool) -> config-override instance
 
     def add(self, plugin: Plugin) -> None:
         self.plugins.append(plugin)
@@ -194,6 +198,40 @@ class PluginManager:
 
     def for_hook(self, hook: HookType) -> List[Plugin]:
         return [p for p in self.plugins if hook in p.hooks and p.mode != PluginMode.DISABLED]
+
+    # -- per-tool bindings -------------------------------------------------
+    def set_bindings(self, bindings: Dict[str, Dict[str, Dict[str, Any]]]) -> None:
+        """Replace the whole binding map (loaded from the registry)."""
+        self.bindings = bindings
+        self._derived.clear()
+        self.version += 1
+
+    def bindings_for_tool(self, tool: Optional[str]) -> Dict[str, Dict[str, Any]]:
+        return self.bindings.get(tool, {}) if tool else {}
+
+    def effective_mode(self, plugin: Plugin, tool: Optional[str]) -> PluginMode:
+        b = self.bindings_for_tool(tool).get(plugin.name)
+        if b and b.get("mode"):
+            return PluginMode(b["mode"])
+        return plugin.mode
+
+    def _effective_plugin(self, plugin: Plugin, tool: Optional[str]) -> tuple:
+        """(instance, mode) for this tool: a binding may flip the mode or
+        derive a config-override instance (cached per tool)."""
+        b = self.bindings_for_tool(tool).get(plugin.name)
+        if not b:
+            return plugin, plugin.mode
+        mode = PluginMode(b["mode"]) if b.get("mode") else plugin.mode
+        if b.get("config"):
+            key = (plugin.name, tool)
+            inst = self._derived.get(key)
+            if inst is None:
+                cfg = {**plugin.config, **b["config"], "mode": mode.value, "priority": plugin.priority}
+                inst = type(plugin)(cfg)
+                inst.name = plugin.name
+                self._derived[key] = inst
+            return inst, mode
+        return plugin, mode
 
     def _record(self, plugin: Plugin, hook: HookType, ok: bool, blocked: bool, ms: float) -> None:
         st = self._stats.setdefault(plugin.name, {"calls": 0, "errors": 0, "blocked": 0, "total_ms": 0.0})
@@ -219,18 +257,24 @@ class PluginManager:
         ctx.global_state = self.global_state
         import time as _t
 
-        for plugin in self.for_hook(hook):
-            if not plugin.applies_to(ctx):
+        tool = ctx.name or None
+        for plugin in self.plugins:
+            if hook not in plugin.hooks:
+                continue
+            eff, mode = self._effective_plugin(plugin, tool) if self.bindings else (plugin, plugin.mode)
+            if mode == PluginMode.DISABLED:
+                continue
+            if not eff.applies_to(ctx):
                 continue
             t0 = _t.monotonic()
             try:
-                method = getattr(plugin, _HOOK_METHOD[hook])
+                method = getattr(eff, _HOOK_METHOD[hook])
                 result: PluginResult = await asyncio.wait_for(method(ctx), timeout=self.timeout_s)
                 ms = (_t.monotonic() - t0) * 1000.0
                 blocked = not result.continue_processing
                 self._record(plugin, hook, True, blocked, ms)
                 if blocked:
-                    if plugin.mode == PluginMode.PERMISSIVE:
+                    if mode == PluginMode.PERMISSIVE:
                         logger.warning("plugin %s would block (%s) - permissive", plugin.name, result.violation)
                         continue
                     raise PluginViolationError(plugin.name, result.violation or "blocked", result.violation_code, result.metadata)
@@ -241,8 +285,8 @@ class PluginManager:
             except Exception as exc:
                 ms = (_t.monotonic() - t0) * 1000.0
                 self._record(plugin, hook, False, False, ms)
-                if plugin.mode in (PluginMode.ENFORCE_IGNORE_ERROR, PluginMode.PERMISSIVE):
-                    logger.warning("plugin %s error ignored (%s mode): %s", plugin.name, plugin.mode.value, exc)
+                if mode in (PluginMode.ENFORCE_IGNORE_ERROR, PluginMode.PERMISSIVE):
+                    logger.warning("plugin %s error ignored (%s mode): %s", plugin.name, mode.value, exc)
                     continue
                 raise
         return ctx

@@ -22,6 +22,7 @@ from ..db.engine import Database
 from ..db.models import (
     DbA2AAgent,
     DbGateway,
+    DbPluginBinding,
     DbPrompt,
     DbResource,
     DbServer,
@@ -53,6 +54,7 @@ _MODEL: Dict[str, Type] = {
     "prompt": DbPrompt,
     "server": DbServer,
     "a2a_agent": DbA2AAgent,
+    "plugin_binding": DbPluginBinding,
 }
 
 _UNIQUE_FIELD = {
@@ -62,6 +64,7 @@ _UNIQUE_FIELD = {
     "prompt": "name",
     "server": "name",
     "a2a_agent": "name",
+    "plugin_binding": "name",
 }
 
 

@@ -521,6 +521,33 @@ def build_app(engine: GatewayEngine, auth: Optional[AuthService] = None) -> Fast
     for plural, kind in ENTITY_ROUTES.items():
         _register_crud(plural, kind)
 
+    # -- per-tool plugin bindings (reference: routers/tool_plugin_bindings.py) --
+    @app.get("/tools/{tool_name}/plugin-bindings")
+    async def list_plugin_bindings(tool_name: str, ctx: AuthContext = Depends(require("tools.read"))):
+        return [b for b in engine.registry.list("plugin_binding") if b["tool_name"] == tool_name]
+
+    @app.put("/tools/{tool_name}/plugin-bindings/{plugin_name}")
+    async def set_plugin_binding(tool_name: str, plugin_name: str, request: Request,
+                                 ctx: AuthContext = Depends(require("tools.update"))):
+        body = await request.json() if (await request.body()) else {}
+        mode = body.get("mode")
+        if mode is not None and mode not in ("enforce", "enforce_ignore_error", "permissive", "disabled"):
+            raise HTTPException(422, f"invalid mode {mode!r}")
+        if engine.plugins.get(plugin_name) is None:
+            raise HTTPException(404, f"plugin {plugin_name!r} not registered")
+        ent = engine.set_plugin_binding(tool_name, plugin_name, mode=mode, config=body.get("config"))
+        engine.audit.record(ctx.user, "bind", "plugin_binding", ent.get("id"))
+        return ent
+
+    @app.delete("/tools/{tool_name}/plugin-bindings/{plugin_name}", status_code=204)
+    async def delete_plugin_binding(tool_name: str, plugin_name: str,
+                                    ctx: AuthContext = Depends(require("tools.update"))):
+        try:
+            engine.delete_plugin_binding(tool_name, plugin_name)
+        except NotFoundError as exc:
+            raise HTTPException(404, str(exc)) from exc
+        return Response(status_code=204)
+
     @app.post("/gateways/{gateway_id}/refresh")
     async def refresh_gateway(gateway_id: str, ctx: AuthContext = Depends(require("gateways.update"))):
         try:

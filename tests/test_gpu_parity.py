@@ -36,6 +36,8 @@ TRAFFIC = [
                                 "source_timezone": 42, "target_timezone": "UTC"}),
     ("fast-time-echo", {"msg": "x" * 3000}),                                    # large payload
     ("gone-tool", {}),                                                          # unreachable tool
+    ("native-time-echo", {"msg": "has forbidden content but deny is unbound"}),  # binding: deny disabled
+    ("fast-time-get_system_time", {"timezone": "banana-time"}),                  # binding: config-override deny → host chain block
 ]
 
 
@@ -52,6 +54,10 @@ async def _build(gpu: bool):
     # an unreachable tool (reference: reachable=False → -32002)
     e.registry.create("tool", name="gone-tool", original_name="gone-tool",
                       integration_type="LOCAL", reachable=False)
+    # per-tool plugin bindings: a mode flip (stays on the GPU fast path via
+    # the per-tool flag table) and a config override (forces the host chain)
+    e.set_plugin_binding("native-time-echo", "deny_filter", mode="disabled")
+    e.set_plugin_binding("fast-time-get_system_time", "deny_filter", config={"words": ["banana"]})
     if gpu:
         assert e.enable_gpu()
     return e
@@ -86,6 +92,7 @@ def test_pipeline_parity_with_cpu_chain():
         assert st["blocked"] >= 2          # deny + harm (+schema)
         assert st["slow_path"] >= 3        # pii x2 + normalizer + unicode
         assert st["fast_path"] >= 3
+        assert st["host_bound"] >= 1       # config-override binding → CPU chain
         await cpu.shutdown()
         await gpu.shutdown()
 

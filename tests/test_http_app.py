@@ -407,3 +407,52 @@ def test_session_ids_hash_to_owner_rank():
     for _ in range(16):
         s = reg.create()
         assert stable_hash(s.session_id) % 4 == 2
+
+
+def test_plugin_binding_endpoints(client_engine, run):
+    """Per-tool plugin bindings over HTTP (reference: routers/tool_plugin_bindings.py):
+    disable deny_filter for one tool, see the block disappear for that tool
+    only, then delete the binding and see it return."""
+    client_factory, engine, app = client_engine
+
+    async def go():
+        async with client_factory() as c:
+            payload = {"jsonrpc": "2.0", "id": 1, "method": "tools/call",
+                       "params": {"name": "echo", "arguments": {"q": "this is forbidden text"}}}
+            # deny_filter blocks by default
+            r = await c.post("/rpc", headers=ADMIN, json=payload)
+            assert r.json()["error"]["code"] == -32003
+            # bind: disabled for echo
+            r = await c.put("/tools/echo/plugin-bindings/deny_filter", headers=ADMIN,
+                            json={"mode": "disabled"})
+            assert r.status_code == 200, r.text
+            assert r.json()["tool_name"] == "echo"
+            r = await c.get("/tools/echo/plugin-bindings", headers=ADMIN)
+            assert len(r.json()) == 1
+            r = await c.post("/rpc", headers=ADMIN, json=payload)
+            assert r.json()["result"]["structuredContent"] == {"q": "this is forbidden text"}
+            # other tools unaffected
+            engine.tool_service.register_local_tool("other", lambda a: a)
+            r = await c.post("/rpc", headers=ADMIN,
+                             json={**payload, "params": {"name": "other",
+                                                         "arguments": {"q": "forbidden"}}})
+            assert "error" in r.json()
+            # unknown plugin name -> 404; bad mode -> 422
+            r = await c.put("/tools/echo/plugin-bindings/nope", headers=ADMIN, json={})
+            assert r.status_code == 404
+            r = await c.put("/tools/echo/plugin-bindings/deny_filter", headers=ADMIN,
+                            json={"mode": "sideways"})
+            assert r.status_code == 422
+            # delete -> block returns
+            r = await c.delete("/tools/echo/plugin-bindings/deny_filter", headers=ADMIN)
+            assert r.status_code == 204
+            r = await c.post("/rpc", headers=ADMIN, json=payload)
+            assert r.json()["error"]["code"] == -32003
+            # bindings persist in the DB (survive a registry reload)
+            engine.set_plugin_binding("echo", "deny_filter", mode="permissive")
+            engine.registry.load_all()
+            engine.sync_plugin_bindings()
+            r = await c.post("/rpc", headers=ADMIN, json=payload)
+            assert "result" in r.json()
+
+    run(go())

@@ -237,3 +237,50 @@ def test_loader_yaml(tmp_path):
     mgr = load_plugin_manager(str(cfg))
     p = mgr.get("deny_filter")
     assert p is not None and p.mode == PluginMode.PERMISSIVE and p.words == ["zap"]
+
+
+# -- per-tool plugin bindings (reference: tool_plugin_bindings) ------------
+
+def test_binding_disables_plugin_for_one_tool(run):
+    p = DenyFilterPlugin({"words": ["classified"]})
+    mgr = PluginManager([p])
+    mgr.set_bindings({"open-tool": {"deny_filter": {"mode": "disabled"}}})
+    # bound tool: deny skipped entirely
+    out = run(mgr.invoke_hook(HookType.TOOL_PRE_INVOKE, ctx({"q": "classified"}, name="open-tool")))
+    assert out.args == {"q": "classified"}
+    # other tools: still enforced
+    with pytest.raises(PluginViolationError):
+        run(mgr.invoke_hook(HookType.TOOL_PRE_INVOKE, ctx({"q": "classified"}, name="other")))
+
+
+def test_binding_mode_flip_permissive(run):
+    p = DenyFilterPlugin({"words": ["classified"]})
+    mgr = PluginManager([p])
+    mgr.set_bindings({"soft-tool": {"deny_filter": {"mode": "permissive"}}})
+    out = run(mgr.invoke_hook(HookType.TOOL_PRE_INVOKE, ctx({"q": "classified"}, name="soft-tool")))
+    assert out.args == {"q": "classified"}  # logged, not blocked
+
+
+def test_binding_config_override(run):
+    p = DenyFilterPlugin({"words": ["classified"]})
+    mgr = PluginManager([p])
+    mgr.set_bindings({"strict-tool": {"deny_filter": {"config": {"words": ["banana"]}}}})
+    # override word list applies only on the bound tool
+    with pytest.raises(PluginViolationError):
+        run(mgr.invoke_hook(HookType.TOOL_PRE_INVOKE, ctx({"q": "banana"}, name="strict-tool")))
+    out = run(mgr.invoke_hook(HookType.TOOL_PRE_INVOKE, ctx({"q": "banana"}, name="other")))
+    assert out.args == {"q": "banana"}
+    # version bump on set_bindings (GPU pipeline resync trigger)
+    v = mgr.version
+    mgr.set_bindings({})
+    assert mgr.version == v + 1
+
+
+def test_binding_reenables_globally_disabled_plugin(run):
+    p = DenyFilterPlugin({"words": ["classified"], "mode": "disabled"})
+    mgr = PluginManager([p])
+    out = run(mgr.invoke_hook(HookType.TOOL_PRE_INVOKE, ctx({"q": "classified"}, name="t")))
+    assert out.args == {"q": "classified"}
+    mgr.set_bindings({"t": {"deny_filter": {"mode": "enforce"}}})
+    with pytest.raises(PluginViolationError):
+        run(mgr.invoke_hook(HookType.TOOL_PRE_INVOKE, ctx({"q": "classified"}, name="t")))
