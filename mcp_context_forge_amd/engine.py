@@ -78,6 +78,9 @@ class GatewayEngine:
         self.toolops = ToolOps(self)
         self.observability = ObservabilityService(self.db)
         self.catalog = CatalogService()
+        from .services.governance import TagService
+
+        self.tags = TagService(self.registry)
         self.password_policy = PasswordPolicy()
         self.token_blocklist = TokenBlocklist()
         self.content_security = ContentSecurity()

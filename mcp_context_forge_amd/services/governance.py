@@ -154,3 +154,39 @@ class ContentSecurity:
             if total > self.max_result_bytes:
                 errs.append(f"result too large ({total} bytes)")
         return errs
+
+
+class TagService:
+    """Cross-entity tag aggregation (reference: services/tag_service.py —
+    GET /tags with per-kind counts and optional entity listings)."""
+
+    KINDS = ("tool", "gateway", "server", "resource", "prompt", "a2a_agent")
+
+    def __init__(self, registry):
+        self.registry = registry
+
+    def list_tags(self, kinds: Optional[List[str]] = None,
+                  include_entities: bool = False) -> List[Dict[str, Any]]:
+        kinds = [k for k in (kinds or self.KINDS) if k in self.KINDS]
+        agg: Dict[str, Dict[str, Any]] = {}
+        for kind in kinds:
+            for ent in self.registry.list(kind):
+                for tag in ent.get("tags") or []:
+                    rec = agg.setdefault(tag, {"name": tag, "count": 0,
+                                               "by_kind": {}, "entities": []})
+                    rec["count"] += 1
+                    rec["by_kind"][kind] = rec["by_kind"].get(kind, 0) + 1
+                    if include_entities:
+                        key = ent.get("name") or ent.get("uri") or ent.get("id")
+                        rec["entities"].append({"kind": kind, "id": ent.get("id"), "name": key})
+        out = sorted(agg.values(), key=lambda r: (-r["count"], r["name"]))
+        if not include_entities:
+            for r in out:
+                r.pop("entities", None)
+        return out
+
+    def entities_for_tag(self, tag: str, kinds: Optional[List[str]] = None) -> List[Dict[str, Any]]:
+        for rec in self.list_tags(kinds, include_entities=True):
+            if rec["name"] == tag:
+                return rec["entities"]
+        return []

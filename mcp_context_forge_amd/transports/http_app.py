@@ -709,6 +709,19 @@ def build_app(engine: GatewayEngine, auth: Optional[AuthService] = None) -> Fast
         return Response(status_code=204)
 
     # -- catalog (reference: services/catalog_service.py) --------------------------
+    @app.get("/tags")
+    async def list_tags(kinds: str = "", include_entities: bool = False,
+                        ctx: AuthContext = Depends(require("tools.read"))):
+        """Cross-entity tag aggregation (reference: services/tag_service.py)."""
+        klist = [k.strip() for k in kinds.split(",") if k.strip()] or None
+        return engine.tags.list_tags(klist, include_entities=include_entities)
+
+    @app.get("/tags/{tag}/entities")
+    async def tag_entities(tag: str, kinds: str = "",
+                           ctx: AuthContext = Depends(require("tools.read"))):
+        klist = [k.strip() for k in kinds.split(",") if k.strip()] or None
+        return engine.tags.entities_for_tag(tag, klist)
+
     @app.get("/catalog")
     async def catalog_list(category: Optional[str] = None, ctx: AuthContext = Depends(require("tools.read"))):
         return engine.catalog.list(category)

@@ -456,3 +456,27 @@ def test_plugin_binding_endpoints(client_engine, run):
             assert "result" in r.json()
 
     run(go())
+
+
+def test_tag_service_endpoints(client_engine, run):
+    """Cross-entity tag aggregation (reference: services/tag_service.py)."""
+    client_factory, engine, app = client_engine
+
+    async def go():
+        engine.registry.create("tool", name="tagged-a", original_name="tagged-a",
+                               integration_type="LOCAL", tags=["analytics", "beta"])
+        engine.registry.create("server", name="srv-1", tags=["analytics"])
+        async with client_factory() as c:
+            r = await c.get("/tags", headers=ADMIN)
+            assert r.status_code == 200
+            tags = {t["name"]: t for t in r.json()}
+            assert tags["analytics"]["count"] == 2
+            assert tags["analytics"]["by_kind"] == {"tool": 1, "server": 1}
+            assert tags["beta"]["count"] == 1
+            r = await c.get("/tags", headers=ADMIN, params={"kinds": "server"})
+            assert {t["name"] for t in r.json()} == {"analytics"}
+            r = await c.get("/tags/analytics/entities", headers=ADMIN)
+            names = {e["name"] for e in r.json()}
+            assert names == {"tagged-a", "srv-1"}
+
+    run(go())
