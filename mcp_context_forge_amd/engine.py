@@ -295,6 +295,15 @@ class GatewayEngine:
             return await self.gpu_pipeline.process_batch(raws, user=user, server_id=server_id)
         return list(await asyncio.gather(*(self.handle_rpc_bytes(r, user=user, server_id=server_id) for r in raws)))
 
+    def invalidate_peers(self, what: str = "registry") -> None:
+        """Publish an invalidation to peer ranks over the RCCL bus (the
+        reference's Redis pub/sub invalidation). No-op single-rank."""
+        if self.bus is not None:
+            try:
+                self.bus.publish({"kind": "invalidate", "what": what})
+            except Exception:  # pragma: no cover - bus teardown race
+                logger.warning("peer invalidation publish failed", exc_info=True)
+
     # -- per-tool plugin bindings (reference: routers/tool_plugin_bindings.py) --
     def sync_plugin_bindings(self) -> None:
         """Load registry plugin_binding rows into the plugin manager
@@ -318,6 +327,7 @@ class GatewayEngine:
             ent = self.registry.create("plugin_binding", name=key, tool_name=tool_name,
                                        plugin_name=plugin_name, mode=mode, config=config)
         self.sync_plugin_bindings()
+        self.invalidate_peers("plugin_binding")
         return ent
 
     def delete_plugin_binding(self, tool_name: str, plugin_name: str) -> None:
@@ -329,6 +339,7 @@ class GatewayEngine:
             raise NotFoundError(f"binding {key!r} not found")
         self.registry.delete("plugin_binding", existing["id"])
         self.sync_plugin_bindings()
+        self.invalidate_peers("plugin_binding")
 
     async def notify_list_changed(self, kind: str) -> None:
         """Fan out listChanged notifications to live sessions (reference:

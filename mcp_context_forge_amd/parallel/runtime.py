@@ -86,8 +86,11 @@ class DistributedGateway:
 
     def _on_publish(self, src: int, payload: Any) -> None:
         if isinstance(payload, dict) and payload.get("kind") == "invalidate":
-            # peer mutated the shared DB — refresh the hot caches
+            # peer mutated the shared DB — refresh the hot caches; bindings
+            # feed the plugin manager (and bump its version, so an attached
+            # GPU pipeline recompiles its per-tool flag tables)
             self.engine.registry.load_all()
+            self.engine.sync_plugin_bindings()
 
     @property
     def is_leader(self) -> bool:

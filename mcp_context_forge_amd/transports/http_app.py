@@ -478,6 +478,7 @@ def build_app(engine: GatewayEngine, auth: Optional[AuthService] = None) -> Fast
                 ent = engine.registry.create(kind, **body)
                 engine.audit.record(ctx.user, "create", kind, ent.get("id"))
                 await engine.notify_list_changed(kind)
+                engine.invalidate_peers(kind)
                 return ent
             except ConflictError as exc:
                 raise HTTPException(409, str(exc)) from exc
@@ -506,6 +507,7 @@ def build_app(engine: GatewayEngine, auth: Optional[AuthService] = None) -> Fast
                     engine.registry.delete(kind, entity_id)
                 engine.audit.record(ctx.user, "delete", kind, entity_id)
                 await engine.notify_list_changed(kind)
+                engine.invalidate_peers(kind)
             except NotFoundError as exc:
                 raise HTTPException(404, str(exc)) from exc
             return Response(status_code=204)
