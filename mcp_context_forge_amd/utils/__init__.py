@@ -53,39 +53,7 @@ class RetryManager:
         raise last
 
 
-def jsonpath_filter(value: Any, expr: Optional[str]) -> Any:
-    """Minimal JSONPath-ish filter (reference: main.py:1281 jsonpath_modifier).
-
-    Supports the dotted/bracket subset the reference's tool rows actually use:
-    ``$.a.b[0].c``, ``$[*].name``. Full jsonpath-ng parity is a later round.
-    """
-    if not expr or expr == "$":
-        return value
-    if not expr.startswith("$"):
-        return value
-    tokens = re.findall(r"\.([A-Za-z_][\w\-]*)|\[(\d+|\*)\]", expr[1:])
-    cur: Any = value
-
-    def walk(node: Any, toks: list) -> Any:
-        for i, (key, idx) in enumerate(toks):
-            if key:
-                if isinstance(node, dict) and key in node:
-                    node = node[key]
-                else:
-                    return None
-            elif idx == "*":
-                if isinstance(node, list):
-                    return [walk(item, toks[i + 1:]) for item in node]
-                return None
-            else:
-                j = int(idx)
-                if isinstance(node, list) and j < len(node):
-                    node = node[j]
-                else:
-                    return None
-        return node
-
-    return walk(cur, tokens)
+from .jsonpath import jsonpath_filter  # noqa: F401  (full evaluator in utils/jsonpath.py)
 
 
 class TokenBucket:
