@@ -133,6 +133,9 @@ class GpuPluginPipeline:
 
         self.max_depth = s.max_json_depth
         self.max_string = s.max_string_length
+        from ..ops.pybridge import get as _pb_get
+
+        self._pb = _pb_get()  # C response-assembly loops (fails loudly if missing)
         # pinned-host staging arena (bump-allocated per pass; reset after each sync)
         self._pin = torch.empty(16 << 20, dtype=torch.uint8, pin_memory=True)
         self._pin_off = 0
@@ -575,12 +578,14 @@ class GpuPluginPipeline:
 
         self._toc("decide", t_d)
         t_a = self._tic()
-        arena_b = arena.tobytes()
         answered = np.nonzero(state == hip.ST_ANSWERED)[0]
-        for j in answered:
-            j = int(j)
-            if rb[j] >= 0:
-                responses[int(rows[j])] = arena_b[rb[j]:re_[j]]
+        if answered.size:
+            # C-loop scatter (ops/csrc/pybridge.c): no arena copy, no
+            # per-row Python slicing
+            self._pb.scatter_slices(arena, np.ascontiguousarray(rb[answered]),
+                                    np.ascontiguousarray(re_[answered]),
+                                    np.ascontiguousarray(rows[answered].astype(np.int64)),
+                                    responses)
         rc = np.bincount(reason[answered].astype(np.int64), minlength=9) if answered.size else np.zeros(9, int)
         self.blocked += int(rc[3] + rc[4] + rc[5] + rc[8])
         self.cache_hits += int(rc[6] + rc[7])
@@ -884,10 +889,10 @@ class GpuPluginPipeline:
             np.ascontiguousarray(needs_host.astype(np.uint8)),
             self._t_name_beg, self._t_name_end, self._t_name_blob, self._t_flags,
             self._exact_native, now, self.exact_cache.ttl if self.exact_cache else 0.0)
-        arena2_b = arena2.tobytes()
-        for k in range(n_all):
-            if rb2[k] >= 0:
-                responses[int(rows[all_js[k]])] = arena2_b[rb2[k]:re2[k]]
+        if n_all:
+            self._pb.scatter_slices(arena2, rb2, re2,
+                                    np.ascontiguousarray(rows[all_js_np].astype(np.int64)),
+                                    responses)
 
         # --- host post chain for flagged rows ---
         host_ks = np.nonzero(needs_host)[0]

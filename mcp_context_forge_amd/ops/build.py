@@ -22,6 +22,9 @@ SOURCES = ["scan.hip", "featurize.hip", "json_guard.hip", "gemm_bf16.hip", "runt
            "gemm_v2.hip", "envelope.cpp", "upstream.cpp", "fastpath.cpp"]
 
 
+PYBRIDGE = Path(__file__).parent / "forge_pybridge.so"
+
+
 def needs_build() -> bool:
     if not LIB.exists():
         return True
@@ -32,7 +35,25 @@ def needs_build() -> bool:
     return False
 
 
+def build_pybridge(force: bool = False, verbose: bool = True) -> Path:
+    """CPython C extension for the response-assembly hot loops (plain gcc,
+    no hip): forge_pybridge.so in-tree (ships with the gpurun snapshot)."""
+    src = CSRC / "pybridge.c"
+    if not force and PYBRIDGE.exists() and PYBRIDGE.stat().st_mtime > src.stat().st_mtime:
+        return PYBRIDGE
+    import sysconfig
+
+    cc = os.environ.get("CC", "gcc")
+    cmd = [cc, "-O2", "-fPIC", "-shared", f"-I{sysconfig.get_paths()['include']}",
+           "-o", str(PYBRIDGE), str(src)]
+    if verbose:
+        print("[forge-pybridge]", " ".join(cmd), file=sys.stderr)
+    subprocess.run(cmd, check=True)
+    return PYBRIDGE
+
+
 def build(force: bool = False, verbose: bool = True) -> Path:
+    build_pybridge(force=force, verbose=verbose)
     if not force and not needs_build():
         return LIB
     cmd = [

@@ -1,0 +1,121 @@
+/* CPython bridge for the response assembly hot loops.
+ *
+ * The GPU pipeline's last host step turns (arena bytes, per-row spans) into
+ * per-request PyBytes in the responses list. In Python that loop costs
+ * ~0.4 us/row (≈3 ms per 8k batch — the single largest host stage, see
+ * profiles/README.md). Here it is one C loop: PyBytes_FromStringAndSize
+ * straight out of the numpy arena buffer, PyList_SetItem (steals the ref,
+ * frees any prior entry).
+ *
+ * Plain CPython C API extension (no hipcc, no torch ABI): compiled by
+ * ops/build.py with the system compiler into forge_pybridge.so in-tree.
+ */
+#define PY_SSIZE_T_CLEAN
+#include <Python.h>
+#include <stdint.h>
+
+/* scatter_slices(arena, rb, re, dest, out_list [, prefix, suffix]) -> count
+ *
+ * arena: buffer (numpy uint8 or bytes) with the spliced responses
+ * rb/re: int64 buffers, span of row k (rb[k] < 0 -> skip row)
+ * dest:  int64 buffer, index into out_list for row k
+ * out_list: Python list to scatter into
+ * prefix/suffix: optional bytes wrapped around each slice (0-copy concat)
+ */
+static PyObject *scatter_slices(PyObject *self, PyObject *args) {
+    Py_buffer arena, rb, re, dest;
+    PyObject *out_list;
+    const char *prefix = NULL, *suffix = NULL;
+    Py_ssize_t prefix_len = 0, suffix_len = 0;
+
+    if (!PyArg_ParseTuple(args, "y*y*y*y*O!|y#y#", &arena, &rb, &re, &dest,
+                          &PyList_Type, &out_list,
+                          &prefix, &prefix_len, &suffix, &suffix_len))
+        return NULL;
+
+    const char *base = (const char *)arena.buf;
+    const int64_t *rbp = (const int64_t *)rb.buf;
+    const int64_t *rep = (const int64_t *)re.buf;
+    const int64_t *dp = (const int64_t *)dest.buf;
+    Py_ssize_t n = (Py_ssize_t)(rb.len / (Py_ssize_t)sizeof(int64_t));
+    Py_ssize_t list_len = PyList_GET_SIZE(out_list);
+    Py_ssize_t arena_len = arena.len;
+    Py_ssize_t count = 0;
+
+    for (Py_ssize_t k = 0; k < n; k++) {
+        int64_t b = rbp[k], e = rep[k], d = dp[k];
+        if (b < 0 || e < b || d < 0 || d >= list_len || e > arena_len)
+            continue;
+        PyObject *obj;
+        if (prefix_len || suffix_len) {
+            Py_ssize_t total = prefix_len + (Py_ssize_t)(e - b) + suffix_len;
+            obj = PyBytes_FromStringAndSize(NULL, total);
+            if (obj) {
+                char *w = PyBytes_AS_STRING(obj);
+                memcpy(w, prefix, (size_t)prefix_len);
+                memcpy(w + prefix_len, base + b, (size_t)(e - b));
+                memcpy(w + prefix_len + (e - b), suffix, (size_t)suffix_len);
+            }
+        } else {
+            obj = PyBytes_FromStringAndSize(base + b, (Py_ssize_t)(e - b));
+        }
+        if (!obj) {
+            PyBuffer_Release(&arena); PyBuffer_Release(&rb);
+            PyBuffer_Release(&re); PyBuffer_Release(&dest);
+            return NULL;
+        }
+        PyList_SetItem(out_list, d, obj); /* steals ref, frees old entry */
+        count++;
+    }
+    PyBuffer_Release(&arena); PyBuffer_Release(&rb);
+    PyBuffer_Release(&re); PyBuffer_Release(&dest);
+    return PyLong_FromSsize_t(count);
+}
+
+/* slices_list(arena, beg, end) -> list[bytes|None]
+ * Bulk-extract spans into a fresh list (None where beg<0). */
+static PyObject *slices_list(PyObject *self, PyObject *args) {
+    Py_buffer arena, beg, end;
+    if (!PyArg_ParseTuple(args, "y*y*y*", &arena, &beg, &end))
+        return NULL;
+    const char *base = (const char *)arena.buf;
+    const int64_t *bp = (const int64_t *)beg.buf;
+    const int64_t *ep = (const int64_t *)end.buf;
+    Py_ssize_t n = (Py_ssize_t)(beg.len / (Py_ssize_t)sizeof(int64_t));
+    Py_ssize_t arena_len = arena.len;
+    PyObject *out = PyList_New(n);
+    if (!out) goto fail;
+    for (Py_ssize_t k = 0; k < n; k++) {
+        int64_t b = bp[k], e = ep[k];
+        PyObject *obj;
+        if (b < 0 || e < b || e > arena_len) {
+            Py_INCREF(Py_None);
+            obj = Py_None;
+        } else {
+            obj = PyBytes_FromStringAndSize(base + b, (Py_ssize_t)(e - b));
+            if (!obj) { Py_DECREF(out); goto fail; }
+        }
+        PyList_SET_ITEM(out, k, obj);
+    }
+    PyBuffer_Release(&arena); PyBuffer_Release(&beg); PyBuffer_Release(&end);
+    return out;
+fail:
+    PyBuffer_Release(&arena); PyBuffer_Release(&beg); PyBuffer_Release(&end);
+    return NULL;
+}
+
+static PyMethodDef Methods[] = {
+    {"scatter_slices", scatter_slices, METH_VARARGS,
+     "scatter arena spans into a responses list as bytes"},
+    {"slices_list", slices_list, METH_VARARGS,
+     "extract arena spans into a new list of bytes/None"},
+    {NULL, NULL, 0, NULL},
+};
+
+static struct PyModuleDef moduledef = {
+    PyModuleDef_HEAD_INIT, "forge_pybridge", NULL, -1, Methods,
+};
+
+PyMODINIT_FUNC PyInit_forge_pybridge(void) {
+    return PyModule_Create(&moduledef);
+}

@@ -14,6 +14,7 @@
 #include <stdlib.h>
 #include <string.h>
 #include <string>
+#include <thread>
 #include <vector>
 
 namespace {
@@ -106,15 +107,14 @@ void append_span(std::string& out, const uint8_t* b, const uint8_t* e) {
 //   kinds[r]: handler kind (0/1/2); now_iso: host-provided wall clock string
 // Output: caller-provided growing buffer protocol — we return the required
 // size on first call (out=null), caller allocates and calls again.
-extern "C" int64_t forge_upstream_call_batch(
+static void upstream_rows(
     const uint8_t* data, const int32_t* args_beg, const int32_t* args_end,
-    const int32_t* kinds, int n, const char* now_iso,
-    uint8_t* out, int64_t out_cap, int64_t* res_beg, int64_t* res_end)
+    const int32_t* kinds, int r0, int r1, const char* now_iso,
+    std::string& buf, int64_t* res_beg, int64_t* res_end)
 {
-    std::string buf;
-    buf.reserve((size_t)n * 192);
+    buf.reserve((size_t)(r1 - r0) * 192);
     static const uint8_t EMPTY[2] = {'{', '}'};
-    for (int r = 0; r < n; ++r) {
+    for (int r = r0; r < r1; ++r) {
         res_beg[r] = (int64_t)buf.size();
         const uint8_t* ab = args_beg[r] >= 0 ? data + args_beg[r] : EMPTY;
         const uint8_t* ae = args_beg[r] >= 0 ? data + args_end[r] : EMPTY + 2;
@@ -150,8 +150,45 @@ extern "C" int64_t forge_upstream_call_batch(
         }
         res_end[r] = (int64_t)buf.size();
     }
-    if (out != nullptr && (int64_t)buf.size() <= out_cap) {
-        memcpy(out, buf.data(), buf.size());
+}
+
+// Batch entry: rows are independent, so chunk them across threads (each
+// with a private buffer), then prefix-sum the chunk sizes, rebase the
+// per-row spans, and copy chunks into `out`. Same grow-retry contract as
+// before: returns the needed size; copies only when it fits out_cap.
+extern "C" int64_t forge_upstream_call_batch(
+    const uint8_t* data, const int32_t* args_beg, const int32_t* args_end,
+    const int32_t* kinds, int n, const char* now_iso,
+    uint8_t* out, int64_t out_cap, int64_t* res_beg, int64_t* res_end)
+{
+    int nthreads = n >= 2048 ? 8 : (n >= 256 ? 4 : 1);
+    if (nthreads == 1) {
+        std::string buf;
+        upstream_rows(data, args_beg, args_end, kinds, 0, n, now_iso, buf, res_beg, res_end);
+        if (out != nullptr && (int64_t)buf.size() <= out_cap)
+            memcpy(out, buf.data(), buf.size());
+        return (int64_t)buf.size();
     }
-    return (int64_t)buf.size();
+    std::vector<std::string> bufs(nthreads);
+    std::vector<std::thread> threads;
+    int chunk = (n + nthreads - 1) / nthreads;
+    for (int t = 0; t < nthreads; ++t) {
+        int r0 = t * chunk, r1 = r0 + chunk < n ? r0 + chunk : n;
+        if (r0 >= r1) break;
+        threads.emplace_back([&, t, r0, r1] {
+            upstream_rows(data, args_beg, args_end, kinds, r0, r1, now_iso,
+                          bufs[t], res_beg, res_end);
+        });
+    }
+    for (auto& th : threads) th.join();
+    int64_t total = 0;
+    std::vector<int64_t> base(nthreads, 0);
+    for (int t = 0; t < nthreads; ++t) { base[t] = total; total += (int64_t)bufs[t].size(); }
+    if (out == nullptr || total > out_cap) return total;
+    for (int t = 0; t < nthreads; ++t) {
+        if (!bufs[t].empty()) memcpy(out + base[t], bufs[t].data(), bufs[t].size());
+        int r0 = t * chunk, r1 = r0 + chunk < n ? r0 + chunk : n;
+        for (int r = r0; r < r1; ++r) { res_beg[r] += base[t]; res_end[r] += base[t]; }
+    }
+    return total;
 }
