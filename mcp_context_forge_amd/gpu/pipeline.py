@@ -812,13 +812,32 @@ class GpuPluginPipeline:
                     py_errors[idx] = exc
 
             py_results = [None] * len(py_items)
-            seq, conc = [], []
+            seq, conc, nat2 = [], [], []
             for idx, (j, args2) in enumerate(py_items):
                 mt = self._meta_list[tool_idx[j]]
-                if mt.itype == "LOCAL" or isinstance(mt.client, InProcUpstream) or mt.native_kind >= 0:
+                if mt.native_kind >= 0 and args2 is not None:
+                    # rewritten args for a native tool: batch through the C++
+                    # upstream instead of per-request Python dispatch
+                    nat2.append((idx, json.dumps(args2, separators=(",", ":"), default=str).encode(),
+                                 mt.native_kind))
+                elif mt.itype == "LOCAL" or isinstance(mt.client, InProcUpstream) or mt.native_kind >= 0:
                     seq.append((idx, j, args2))
                 else:
                     conc.append((idx, j, args2))
+            if nat2:
+                blob2 = b"".join(a for _, a, _ in nat2)
+                offs2 = np.zeros(len(nat2) + 1, dtype=np.int32)
+                np.cumsum(np.fromiter((len(a) for _, a, _ in nat2), dtype=np.int32,
+                                      count=len(nat2)), out=offs2[1:])
+                kinds2 = np.fromiter((k for _, _, k in nat2), dtype=np.int32, count=len(nat2))
+                now_iso2 = time.strftime("%Y-%m-%dT%H:%M:%SZ", time.gmtime())
+                ob2, ob2b, ob2e = hip.upstream_call_batch(
+                    np.frombuffer(blob2, dtype=np.uint8) if blob2 else np.zeros(1, dtype=np.uint8),
+                    np.ascontiguousarray(offs2[:-1]), np.ascontiguousarray(offs2[1:]),
+                    kinds2, now_iso2)
+                outs2 = self._pb.slices_list(ob2, ob2b, ob2e)
+                for (idx, _a, _k), rbytes in zip(nat2, outs2):
+                    py_results[idx] = rbytes
             for idx, j, args2 in seq:
                 await one(idx, j, args2)
             if conc:

@@ -37,6 +37,7 @@ TRAFFIC = [
     ("fast-time-echo", {"msg": "x" * 3000}),                                    # large payload
     ("gone-tool", {}),                                                          # unreachable tool
     ("native-time-echo", {"msg": "has forbidden content but deny is unbound"}),  # binding: deny disabled
+    ("native-time-echo", {"msg": "mail bob@x.io, ssn 123-45-6789"}),             # pii rewrite → native C++ batch redispatch
     ("fast-time-get_system_time", {"timezone": "banana-time"}),                  # binding: config-override deny → host chain block
 ]
 
