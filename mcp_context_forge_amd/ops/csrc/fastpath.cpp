@@ -25,6 +25,7 @@
 #include <string.h>
 #include <stdio.h>
 #include <string>
+#include <thread>
 #include <unordered_map>
 #include <vector>
 
@@ -43,8 +44,14 @@ struct ExactEntry {
 };
 
 struct ExactCache {
-    std::unordered_map<uint64_t, ExactEntry> map;
+    // sharded by hash so finalize's insert phase can run one thread per
+    // shard without locks (decide's lookups are single-threaded)
+    static constexpr int SHARDS = 8;
+    std::unordered_map<uint64_t, ExactEntry> shards[SHARDS];
     double ttl = 300.0;
+    std::unordered_map<uint64_t, ExactEntry>& map_for(uint64_t h) {
+        return shards[h & (SHARDS - 1)];
+    }
 };
 
 uint64_t fnv64(const uint8_t* p, size_t n, uint64_t h = 1469598103934665603ull) {
@@ -280,8 +287,9 @@ int64_t forge_decide(
             uint64_t h = fnv64(tname_blob + tname_beg[ti], (size_t)(tname_end[ti] - tname_beg[ti]));
             h = fnv64((const uint8_t*)"\x00", 1, h);
             h = fnv64(blob + args_beg[i], (size_t)(args_end[i] - args_beg[i]), h);
-            auto it = ec->map.find(h);
-            if (it != ec->map.end() && it->second.expires > now) {
+            auto& shard = ec->map_for(h);
+            auto it = shard.find(h);
+            if (it != shard.end() && it->second.expires > now) {
                 emit_result(a, i, blob, idb, ide, it->second.value.data(), it->second.value.size());
                 state[i] = ST_ANSWERED;
                 reason_out[i] = 7;
@@ -324,42 +332,92 @@ int64_t forge_finalize(
     uint8_t* arena, int64_t arena_cap, int64_t* resp_beg, int64_t* resp_end,  // [n_rows]
     uint8_t* is_error_out, uint8_t* cacheable_out)                            // [n_rows]
 {
-    Arena a;
-    a.buf.reserve((size_t)n_rows * 64);
-    a.beg = resp_beg;
-    a.end = resp_end;
     auto* ec = (ExactCache*)exact_cache;
     static const char ERRMARK[] = "\"isError\":true";
 
-    for (int j = 0; j < n_rows; ++j) {
-        resp_beg[j] = -1;
-        resp_end[j] = -1;
-        is_error_out[j] = 0;
-        cacheable_out[j] = 0;
-        if (needs_host[j]) continue;
-        int i = rows[j];
-        const char* res = (const char*)res_blob + res_beg[j];
-        size_t rn = (size_t)(res_end[j] - res_beg[j]);
-        bool is_err = rn >= sizeof(ERRMARK) - 1 &&
-                      memmem(res, rn, ERRMARK, sizeof(ERRMARK) - 1) != nullptr;
-        is_error_out[j] = is_err ? 1 : 0;
-        emit_result(a, j, blob, id_beg[i], id_end[i], res, rn);
-        if (!is_err) {
-            cacheable_out[j] = 1;
-            int32_t ti = tool_idx[i];
-            if (ec != nullptr && ti >= 0 && (tool_flags[ti] & TF_EXACT)) {
-                uint64_t h = fnv64(tname_blob + tname_beg[ti], (size_t)(tname_end[ti] - tname_beg[ti]));
-                h = fnv64((const uint8_t*)"\x00", 1, h);
-                h = fnv64(blob + args_beg[i], (size_t)(args_end[i] - args_beg[i]), h);
-                ExactEntry& e = ec->map[h];
-                e.value.assign(res, rn);
-                e.expires = now + ec->ttl;
+    // phase A (parallel): splice responses into per-chunk arenas, compute
+    // is_error/cacheable, collect (hash, span) pairs for the exact cache
+    int nthreads = n_rows >= 2048 ? 8 : (n_rows >= 256 ? 4 : 1);
+    int chunk = (n_rows + nthreads - 1) / nthreads;
+    std::vector<std::string> bufs((size_t)nthreads);
+    std::vector<std::vector<std::pair<uint64_t, std::pair<int64_t, int64_t>>>> inserts((size_t)nthreads);
+
+    auto run_chunk = [&](int t) {
+        int j0 = t * chunk, j1 = j0 + chunk < n_rows ? j0 + chunk : n_rows;
+        Arena a;
+        a.buf.reserve((size_t)(j1 - j0) * 64);
+        a.beg = resp_beg;
+        a.end = resp_end;
+        auto& ins = inserts[(size_t)t];
+        for (int j = j0; j < j1; ++j) {
+            resp_beg[j] = -1;
+            resp_end[j] = -1;
+            is_error_out[j] = 0;
+            cacheable_out[j] = 0;
+            if (needs_host[j]) continue;
+            int i = rows[j];
+            const char* res = (const char*)res_blob + res_beg[j];
+            size_t rn = (size_t)(res_end[j] - res_beg[j]);
+            bool is_err = rn >= sizeof(ERRMARK) - 1 &&
+                          memmem(res, rn, ERRMARK, sizeof(ERRMARK) - 1) != nullptr;
+            is_error_out[j] = is_err ? 1 : 0;
+            emit_result(a, j, blob, id_beg[i], id_end[i], res, rn);
+            if (!is_err) {
+                cacheable_out[j] = 1;
+                int32_t ti = tool_idx[i];
+                if (ec != nullptr && ti >= 0 && (tool_flags[ti] & TF_EXACT)) {
+                    uint64_t h = fnv64(tname_blob + tname_beg[ti], (size_t)(tname_end[ti] - tname_beg[ti]));
+                    h = fnv64((const uint8_t*)"\x00", 1, h);
+                    h = fnv64(blob + args_beg[i], (size_t)(args_end[i] - args_beg[i]), h);
+                    ins.emplace_back(h, std::make_pair(res_beg[j], res_end[j]));
+                }
             }
         }
+        bufs[(size_t)t] = std::move(a.buf);
+    };
+    if (nthreads == 1) {
+        run_chunk(0);
+    } else {
+        std::vector<std::thread> threads;
+        for (int t = 0; t < nthreads; ++t) threads.emplace_back(run_chunk, t);
+        for (auto& th : threads) th.join();
     }
-    if ((int64_t)a.buf.size() > arena_cap) return -(int64_t)a.buf.size();
-    memcpy(arena, a.buf.data(), a.buf.size());
-    return (int64_t)a.buf.size();
+    int64_t total = 0;
+    std::vector<int64_t> base((size_t)nthreads, 0);
+    for (int t = 0; t < nthreads; ++t) { base[(size_t)t] = total; total += (int64_t)bufs[(size_t)t].size(); }
+    if (total > arena_cap) return -total;
+    // phase B (parallel by shard): exact-cache inserts, one thread per shard
+    if (ec != nullptr) {
+        auto insert_shard = [&](int s) {
+            double exp = now + ec->ttl;
+            for (auto& vec : inserts)
+                for (auto& kv : vec) {
+                    if ((int)(kv.first & (ExactCache::SHARDS - 1)) != s) continue;
+                    ExactEntry& e = ec->shards[s][kv.first];
+                    e.value.assign((const char*)res_blob + kv.second.first,
+                                   (size_t)(kv.second.second - kv.second.first));
+                    e.expires = exp;
+                }
+        };
+        bool any = false;
+        for (auto& vec : inserts) if (!vec.empty()) { any = true; break; }
+        if (any && nthreads > 1) {
+            std::vector<std::thread> threads;
+            for (int s = 0; s < ExactCache::SHARDS; ++s) threads.emplace_back(insert_shard, s);
+            for (auto& th : threads) th.join();
+        } else if (any) {
+            for (int s = 0; s < ExactCache::SHARDS; ++s) insert_shard(s);
+        }
+    }
+    // stitch chunk arenas + rebase spans
+    for (int t = 0; t < nthreads; ++t) {
+        if (!bufs[(size_t)t].empty())
+            memcpy(arena + base[(size_t)t], bufs[(size_t)t].data(), bufs[(size_t)t].size());
+        int j0 = t * chunk, j1 = j0 + chunk < n_rows ? j0 + chunk : n_rows;
+        for (int j = j0; j < j1; ++j)
+            if (resp_beg[j] >= 0) { resp_beg[j] += base[(size_t)t]; resp_end[j] += base[(size_t)t]; }
+    }
+    return total;
 }
 
 }  // extern "C"
