@@ -901,6 +901,7 @@ class GpuPluginPipeline:
 
         self._toc("post_scan", t_p)
         t_f = self._tic()
+        t_fc = self._tic()
         now = time.monotonic()
         arena2, rb2, re2, is_err, cacheable = hip.finalize(
             blob, id_b, id_e, args_b, args_e, tool_idx,
@@ -912,6 +913,8 @@ class GpuPluginPipeline:
             self._pb.scatter_slices(arena2, rb2, re2,
                                     np.ascontiguousarray(rows[all_js_np].astype(np.int64)),
                                     responses)
+        self._toc("fin_splice", t_fc)
+        t_fh = self._tic()
 
         # --- host post chain for flagged rows ---
         host_ks = np.nonzero(needs_host)[0]
@@ -943,6 +946,8 @@ class GpuPluginPipeline:
                 cacheable[k] = 1
                 host_cached.append((k, rbytes))
 
+        self._toc("fin_hostpost", t_fh)
+        t_fs = self._tic()
         # --- bookkeeping: semcache insert, breaker, metrics ---
         if self.semcache is not None and n_all:
             ins = np.nonzero((cacheable == 1))[0]
@@ -959,6 +964,8 @@ class GpuPluginPipeline:
                         if pos.size:
                             hip.store_put(self._slot_store, int(slots[pos[0]]), rbytes)
 
+        self._toc("fin_seminsert", t_fs)
+        t_fm = self._tic()
         if n_all:
             ms = (time.monotonic() - t0) * 1000.0
             tis = tool_idx[all_js_np]
@@ -974,6 +981,7 @@ class GpuPluginPipeline:
                         self._breaker_record_bulk(name, int(counts[k]), int(errs[k]))
             self.engine.metrics.record_aggregate_many(
                 [self._meta_list[int(ti)].tid for ti in uniq], counts, errs, ms)
+        self._toc("fin_metrics", t_fm)
         self._toc("finalize", t_f)
 
     def _breaker_record_bulk(self, name: str, n: int, n_err: int) -> None:

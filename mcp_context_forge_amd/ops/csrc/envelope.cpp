@@ -13,6 +13,8 @@
 
 #include <stdint.h>
 #include <string.h>
+#include <thread>
+#include <vector>
 
 namespace {
 
@@ -96,14 +98,14 @@ enum : int32_t {
     ENV_TOOLS_CALL = 1,  // canonical tools/call with name + arguments object
 };
 
-extern "C" int forge_parse_envelopes(
-    const uint8_t* data, const int64_t* offsets, int n,
+static void parse_rows(
+    const uint8_t* data, const int64_t* offsets, int r0, int r1,
     int32_t* kind,
-    int32_t* id_beg, int32_t* id_end,        // -1,-1 = absent (notification)
-    int32_t* name_beg, int32_t* name_end,    // tool name string contents
-    int32_t* args_beg, int32_t* args_end)    // raw arguments object span
+    int32_t* id_beg, int32_t* id_end,
+    int32_t* name_beg, int32_t* name_end,
+    int32_t* args_beg, int32_t* args_end)
 {
-    for (int r = 0; r < n; ++r) {
+    for (int r = r0; r < r1; ++r) {
         const uint8_t* base = data;
         Cur c{data + offsets[r], data + offsets[r + 1]};
         kind[r] = ENV_PARSE_ERR;
@@ -227,5 +229,29 @@ extern "C" int forge_parse_envelopes(
             kind[r] = ENV_OTHER;
         }
     }
+}
+
+// Rows are independent — chunk across threads for large batches.
+extern "C" int forge_parse_envelopes(
+    const uint8_t* data, const int64_t* offsets, int n,
+    int32_t* kind,
+    int32_t* id_beg, int32_t* id_end,        // -1,-1 = absent (notification)
+    int32_t* name_beg, int32_t* name_end,    // tool name string contents
+    int32_t* args_beg, int32_t* args_end)    // raw arguments object span
+{
+    int nthreads = n >= 2048 ? 8 : (n >= 512 ? 4 : 1);
+    if (nthreads == 1) {
+        parse_rows(data, offsets, 0, n, kind, id_beg, id_end, name_beg, name_end, args_beg, args_end);
+        return 0;
+    }
+    int chunk = (n + nthreads - 1) / nthreads;
+    std::vector<std::thread> threads;
+    for (int t = 0; t < nthreads; ++t) {
+        int r0 = t * chunk, r1 = r0 + chunk < n ? r0 + chunk : n;
+        if (r0 >= r1) break;
+        threads.emplace_back(parse_rows, data, offsets, r0, r1,
+                             kind, id_beg, id_end, name_beg, name_end, args_beg, args_end);
+    }
+    for (auto& th : threads) th.join();
     return 0;
 }
