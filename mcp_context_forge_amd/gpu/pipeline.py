@@ -591,7 +591,6 @@ class GpuPluginPipeline:
         self.cache_hits += int(rc[6] + rc[7])
 
         # --- host-schema fallback rows (exact validation) ---
-        extra_native: List[int] = []
         extra_py: List[Tuple[int, Any]] = []
         for j in np.nonzero(state == hip.ST_HOST_SCHEMA)[0]:
             j = int(j)
@@ -615,11 +614,24 @@ class GpuPluginPipeline:
                 if idb is not None and res is not None:
                     responses[r] = self._splice_result(idb, res)
                 self.cache_hits += 1
-            elif mt.native_kind >= 0:
-                nk[j] = mt.native_kind
-                extra_native.append(j)
             else:
+                # native tools land in the nat2 C++ batch inside dispatch
                 extra_py.append((j, args))
+
+        # start the main native upstream batch NOW, in a worker thread — the
+        # C call releases the GIL, so it overlaps with the Python rewrite
+        # pass below (the two row sets are disjoint)
+        native_js = [int(j) for j in np.nonzero(state == hip.ST_DISPATCH_NATIVE)[0]]
+        nat_fut = None
+        if native_js:
+            njs = np.asarray(native_js, dtype=np.int64)
+            kinds = np.ascontiguousarray(nk[njs].astype(np.int32))
+            nb_ = np.ascontiguousarray(args_b[njs])
+            ne_ = np.ascontiguousarray(args_e[njs])
+            nb_ = np.where(ne_ > nb_, nb_, -1).astype(np.int32)
+            now_iso = time.strftime("%Y-%m-%dT%H:%M:%SZ", time.gmtime())
+            nat_fut = asyncio.get_running_loop().run_in_executor(
+                None, hip.upstream_call_batch, blob, nb_, ne_, kinds, now_iso)
 
         # --- PASS 2: rewrite-flagged subset ---
         rewrite_js = [int(j) for j in np.nonzero(state == hip.ST_REWRITE)[0]]
@@ -630,14 +642,13 @@ class GpuPluginPipeline:
                 blob, env, rows, args_b, args_e, rewrite_js, tool_idx, hit, hit_slot, responses)
 
         self._toc("answer_assign", t_a)
-        native_js = [int(j) for j in np.nonzero(state == hip.ST_DISPATCH_NATIVE)[0]] + extra_native
         py_items = [(int(j), None) for j in np.nonzero(state == hip.ST_DISPATCH_PY)[0]] + \
             extra_py + rewrite_dispatch
         self.fast_path += len(native_js) + len(py_items)
         if native_js or py_items:
             await self._dispatch_and_post(blob, env, rows, id_b, id_e, args_b, args_e,
                                           tool_idx, nk, feats, th_arr, native_js, py_items,
-                                          responses, t0)
+                                          responses, t0, nat_fut)
 
     def _pin_reset(self) -> None:
         self._pin_off = 0
@@ -775,19 +786,14 @@ class GpuPluginPipeline:
     async def _dispatch_and_post(self, blob, env, rows, id_b, id_e, args_b, args_e,
                                  tool_idx, nk, feats, th_arr,
                                  native_js: List[int], py_items: List[Tuple[int, Any]],
-                                 responses: List[Optional[bytes]], t0: float) -> None:
-        # --- native upstream batch (C++) ---
+                                 responses: List[Optional[bytes]], t0: float,
+                                 nat_fut=None) -> None:
+        # --- native upstream batch (C++, started before the rewrite pass) ---
         t_u = self._tic()
         nat_blob = np.zeros(0, dtype=np.uint8)
         nat_beg = nat_end = np.zeros(0, dtype=np.int64)
-        if native_js:
-            njs = np.asarray(native_js, dtype=np.int64)
-            kinds = np.ascontiguousarray(nk[njs].astype(np.int32))
-            nb_ = np.ascontiguousarray(args_b[njs])
-            ne_ = np.ascontiguousarray(args_e[njs])
-            nb_ = np.where(ne_ > nb_, nb_, -1).astype(np.int32)
-            now_iso = time.strftime("%Y-%m-%dT%H:%M:%SZ", time.gmtime())
-            nat_blob, nat_beg, nat_end = hip.upstream_call_batch(blob, nb_, ne_, kinds, now_iso)
+        if nat_fut is not None:
+            nat_blob, nat_beg, nat_end = await nat_fut
 
         # --- python dispatch (non-native upstreams / rewritten args) ---
         py_results: List[Optional[bytes]] = []
