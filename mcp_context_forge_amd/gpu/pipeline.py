@@ -618,20 +618,10 @@ class GpuPluginPipeline:
                 # native tools land in the nat2 C++ batch inside dispatch
                 extra_py.append((j, args))
 
-        # start the main native upstream batch NOW, in a worker thread — the
-        # C call releases the GIL, so it overlaps with the Python rewrite
-        # pass below (the two row sets are disjoint)
+        # (measured: running the native batch in an executor thread to
+        # overlap with the rewrite pass LOSES ~10% — GIL handoff churn across
+        # every host stage — so it stays synchronous inside dispatch)
         native_js = [int(j) for j in np.nonzero(state == hip.ST_DISPATCH_NATIVE)[0]]
-        nat_fut = None
-        if native_js:
-            njs = np.asarray(native_js, dtype=np.int64)
-            kinds = np.ascontiguousarray(nk[njs].astype(np.int32))
-            nb_ = np.ascontiguousarray(args_b[njs])
-            ne_ = np.ascontiguousarray(args_e[njs])
-            nb_ = np.where(ne_ > nb_, nb_, -1).astype(np.int32)
-            now_iso = time.strftime("%Y-%m-%dT%H:%M:%SZ", time.gmtime())
-            nat_fut = asyncio.get_running_loop().run_in_executor(
-                None, hip.upstream_call_batch, blob, nb_, ne_, kinds, now_iso)
 
         # --- PASS 2: rewrite-flagged subset ---
         rewrite_js = [int(j) for j in np.nonzero(state == hip.ST_REWRITE)[0]]
@@ -648,7 +638,7 @@ class GpuPluginPipeline:
         if native_js or py_items:
             await self._dispatch_and_post(blob, env, rows, id_b, id_e, args_b, args_e,
                                           tool_idx, nk, feats, th_arr, native_js, py_items,
-                                          responses, t0, nat_fut)
+                                          responses, t0)
 
     def _pin_reset(self) -> None:
         self._pin_off = 0
@@ -786,14 +776,19 @@ class GpuPluginPipeline:
     async def _dispatch_and_post(self, blob, env, rows, id_b, id_e, args_b, args_e,
                                  tool_idx, nk, feats, th_arr,
                                  native_js: List[int], py_items: List[Tuple[int, Any]],
-                                 responses: List[Optional[bytes]], t0: float,
-                                 nat_fut=None) -> None:
-        # --- native upstream batch (C++, started before the rewrite pass) ---
+                                 responses: List[Optional[bytes]], t0: float) -> None:
+        # --- native upstream batch (C++) ---
         t_u = self._tic()
         nat_blob = np.zeros(0, dtype=np.uint8)
         nat_beg = nat_end = np.zeros(0, dtype=np.int64)
-        if nat_fut is not None:
-            nat_blob, nat_beg, nat_end = await nat_fut
+        if native_js:
+            njs = np.asarray(native_js, dtype=np.int64)
+            kinds = np.ascontiguousarray(nk[njs].astype(np.int32))
+            nb_ = np.ascontiguousarray(args_b[njs])
+            ne_ = np.ascontiguousarray(args_e[njs])
+            nb_ = np.where(ne_ > nb_, nb_, -1).astype(np.int32)
+            now_iso = time.strftime("%Y-%m-%dT%H:%M:%SZ", time.gmtime())
+            nat_blob, nat_beg, nat_end = hip.upstream_call_batch(blob, nb_, ne_, kinds, now_iso)
 
         # --- python dispatch (non-native upstreams / rewritten args) ---
         py_results: List[Optional[bytes]] = []
