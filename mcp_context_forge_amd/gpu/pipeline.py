@@ -508,6 +508,8 @@ class GpuPluginPipeline:
         data_gpu = self._upload(blob)
         beg_t = self._upload(args_b)
         end_t = self._upload(args_e)
+        self._toc("gp1_upload", t_g)
+        t_l = self._tic()
         out: Dict[str, torch.Tensor] = {}
         for bname in ("deny", "harm", "pii", "regex", "normalize"):
             bank = self.banks.get(bname)
@@ -524,9 +526,12 @@ class GpuPluginPipeline:
         if self.semcache is not None:
             bv, bi = self.semcache.lookup(feats)
             cache_val_t, cache_idx_t = bv[:m], bi[:m]
+        self._toc("gp1_launch", t_l)
+        t_s = self._tic()
         # off-loop sync: a blocking synchronize would stall the event loop and
         # starve the micro-batch collector under live HTTP load
         await asyncio.to_thread(torch.cuda.synchronize)
+        self._toc("gp1_sync", t_s)
         self._toc("gpu_pass1", t_g)
         t_d = self._tic()
 
