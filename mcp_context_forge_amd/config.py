@@ -110,6 +110,8 @@ class Settings(BaseModel):
     gpu_classifier_classes: int = 8
     gpu_semcache_capacity: int = 65536
     gpu_semcache_threshold: float = 0.92
+    # two-stage lookup: random-projection sketch dim (0 = full sweep)
+    gpu_semcache_sketch_dim: int = 256
     gpu_streams: int = 2              # compute + copy overlap
     gpu_dtype: str = "bf16"
 

@@ -129,6 +129,7 @@ class GpuPluginPipeline:
                 threshold=self.semcache_plugin.threshold,
                 ttl_s=self.semcache_plugin.ttl,
                 device=device,
+                sketch_dim=s.gpu_semcache_sketch_dim,
             )
 
         self.max_depth = s.max_json_depth

@@ -28,7 +28,8 @@ def tool_hash(name: str) -> int:
 
 class GpuSemanticCache:
     def __init__(self, capacity: int = 65536, dim: int = 4096, threshold: float = 0.92,
-                 ttl_s: float = 600.0, device: str = "cuda", chunk: int = 4096):
+                 ttl_s: float = 600.0, device: str = "cuda", chunk: int = 4096,
+                 sketch_dim: int = 256):
         assert capacity % 128 == 0 and chunk % 128 == 0 and dim % 64 == 0
         self.capacity = capacity
         self.dim = dim
@@ -38,6 +39,22 @@ class GpuSemanticCache:
         self.device = device
         self.keys = torch.zeros((capacity, dim), dtype=torch.bfloat16, device=device)
         self.valid = torch.zeros(capacity, dtype=torch.uint8, device=device)
+        # two-stage search: rank with a random-projection sketch sweep
+        # (16x less FLOP than the full sweep at dim=4096/sketch=256), then
+        # verify the top candidate with an exact full-dim dot (verify_dot
+        # kernel) — the threshold compare always sees the TRUE cosine.
+        self.sketch_dim = sketch_dim if (sketch_dim and sketch_dim % 64 == 0
+                                         and sketch_dim < dim) else 0
+        if self.sketch_dim:
+            g = torch.Generator().manual_seed(0x5EED)
+            proj = (torch.randint(0, 2, (self.sketch_dim, dim), generator=g).float() * 2 - 1)
+            self.proj_t = (proj / dim ** 0.5).to(torch.bfloat16).to(device)  # [sk, dim]
+            self.keys_sk = torch.zeros((capacity, self.sketch_dim),
+                                       dtype=torch.bfloat16, device=device)
+        else:
+            self.proj_t = None
+            self.keys_sk = None
+        self._last_sk: Optional[torch.Tensor] = None  # sketches of the last lookup batch
         # host-side metadata mirrors (slot-indexed)
         self.tool_hashes = np.zeros(capacity, dtype=np.int64)
         self.timestamps = np.zeros(capacity, dtype=np.float64)
@@ -56,10 +73,23 @@ class GpuSemanticCache:
         bpad = feats_bf16.shape[0]
         best_val = torch.full((bpad,), -1e30, dtype=torch.float32, device=self.device)
         best_idx = torch.full((bpad,), -1, dtype=torch.int32, device=self.device)
+        if self.sketch_dim:
+            # always compute sketches (reused by insert_features this batch)
+            self._last_sk = hip.gemm_bt(feats_bf16, self.proj_t).to(torch.bfloat16)
         if self.size == 0:
             return best_val, best_idx
         active = min(self.size, self.capacity)
         active_pad = ((active + self.chunk - 1) // self.chunk) * self.chunk
+        if self.sketch_dim:
+            # stage 1: approximate ranking over the sketch matrix
+            for c0 in range(0, active_pad, self.chunk):
+                nc = min(self.chunk, self.capacity - c0)
+                scores = hip.gemm_bt(self._last_sk, self.keys_sk[c0:c0 + nc])
+                hip.rows_argmax_merge(scores, best_val, best_idx, idx_base=c0,
+                                      valid=self.valid[c0:c0 + nc])
+            # stage 2: exact full-dim dot of each row's top candidate
+            best_val = hip.verify_dot(feats_bf16, self.keys, best_idx)
+            return best_val, best_idx
         for c0 in range(0, active_pad, self.chunk):
             nc = min(self.chunk, self.capacity - c0)
             scores = hip.gemm_bt(feats_bf16, self.keys[c0:c0 + nc])
@@ -119,6 +149,11 @@ class GpuSemanticCache:
         rows_t = torch.from_numpy(rows.astype(np.int32)).to(self.device, non_blocking=True)
         slot_t = torch.from_numpy(slots.astype(np.int32)).to(self.device, non_blocking=True)
         hip.rows_gather_scatter_bf16(feats_bf16, rows_t, slot_t, self.keys, self.valid)
+        if self.sketch_dim:
+            sk = self._last_sk
+            if sk is None or sk.shape[0] != feats_bf16.shape[0]:
+                sk = hip.gemm_bt(feats_bf16, self.proj_t).to(torch.bfloat16)
+            hip.rows_gather_scatter_bf16(sk, rows_t, slot_t, self.keys_sk, self.valid)
         now = time.monotonic()
         self.tool_hashes[slots] = tool_hashes
         self.timestamps[slots] = now

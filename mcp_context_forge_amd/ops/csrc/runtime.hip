@@ -83,3 +83,52 @@ extern "C" int forge_synchronize(void* stream) {
     HIP_CHECK(hipStreamSynchronize((hipStream_t)stream));
     return 0;
 }
+
+// Exact verification for the two-stage semcache search: one wave per row
+// computes the full-dim bf16 dot product between the query row and its
+// sketch-selected candidate key. 4 rows per block (4 waves), coalesced
+// 16-byte loads; wave-level reduction via shfl.
+__global__ __launch_bounds__(256) void verify_dot_kernel(
+    const short* __restrict__ feats,   // [M, D] bf16
+    const short* __restrict__ keys,    // [cap, D] bf16
+    const int32_t* __restrict__ idx,   // [M] candidate slot (-1 = none)
+    float* __restrict__ out,           // [M] exact dot (or -1e30)
+    int M, int D)
+{
+    int wave = threadIdx.x / 64;
+    int lane = threadIdx.x % 64;
+    int r = blockIdx.x * 4 + wave;
+    if (r >= M) return;
+    int32_t slot = idx[r];
+    if (slot < 0) { if (lane == 0) out[r] = -1e30f; return; }
+    const short* f = feats + (size_t)r * D;
+    const short* k = keys + (size_t)slot * D;
+    float acc = 0.f;
+    for (int i = lane * 8; i < D; i += 64 * 8) {
+        short8 fv = *(const short8*)(f + i);
+        short8 kv = *(const short8*)(k + i);
+        #pragma unroll
+        for (int e = 0; e < 8; ++e) {
+            short fs = fv[e], ks = kv[e];
+            bf16_t fb = *(const bf16_t*)&fs;
+            bf16_t kb = *(const bf16_t*)&ks;
+            acc += __bfloat162float(fb) * __bfloat162float(kb);
+        }
+    }
+    #pragma unroll
+    for (int off = 32; off > 0; off >>= 1)
+        acc += __shfl_down(acc, off, 64);
+    if (lane == 0) out[r] = acc;
+}
+
+extern "C" int forge_verify_dot(
+    const void* feats, const void* keys, const void* idx, void* out,
+    int M, int D, void* stream)
+{
+    if (D % 8) return 9004;
+    hipStream_t s = (hipStream_t)stream;
+    hipLaunchKernelGGL(verify_dot_kernel, dim3((M + 3) / 4), dim3(256), 0, s,
+                       (const short*)feats, (const short*)keys,
+                       (const int32_t*)idx, (float*)out, M, D);
+    return (int)hipGetLastError();
+}

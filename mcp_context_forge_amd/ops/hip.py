@@ -169,6 +169,15 @@ def rows_gather_scatter_bf16(src: torch.Tensor, src_rows: torch.Tensor, dst_slot
         _ptr(src), _ptr(src_rows), _ptr(dst_slots), _ptr(dst), _ptr(valid), r, d, _stream()))
 
 
+def verify_dot(feats: torch.Tensor, keys: torch.Tensor, idx: torch.Tensor) -> torch.Tensor:
+    """Exact bf16 dot of feats[r] with keys[idx[r]] (-1 → -1e30). fp32 [M]."""
+    m, d = feats.shape
+    out = torch.empty(m, dtype=torch.float32, device=feats.device)
+    _check("forge_verify_dot", _load().forge_verify_dot(
+        _ptr(feats), _ptr(keys), _ptr(idx), _ptr(out), m, d, _stream()))
+    return out
+
+
 def rows_scatter_bf16(src: torch.Tensor, slots: torch.Tensor, dst: torch.Tensor) -> None:
     r, d = src.shape
     _check("forge_rows_scatter_bf16", _load().forge_rows_scatter_bf16(_ptr(src), _ptr(slots), _ptr(dst), r, d, _stream()))

@@ -278,3 +278,50 @@ def test_rows_gather_scatter(hip):
         assert torch.equal(dst[s], src[r])
         assert valid[s].item() == 1
     assert valid.sum().item() == 3
+
+
+@requires_gpu
+def test_verify_dot_matches_torch(hip):
+    torch.manual_seed(3)
+    m, cap, d = 77, 64, 512
+    feats = (torch.randn(m, d) * 0.3).bfloat16().cuda()
+    keys = (torch.randn(cap, d) * 0.3).bfloat16().cuda()
+    idx = torch.randint(-1, cap, (m,), dtype=torch.int32).cuda()
+    out = hip.verify_dot(feats, keys, idx)
+    torch.cuda.synchronize()
+    f32, k32 = feats.float().cpu(), keys.float().cpu()
+    idx_h = idx.cpu().numpy()
+    out_h = out.cpu().numpy()
+    for i in range(m):
+        if idx_h[i] < 0:
+            assert out_h[i] <= -1e29
+        else:
+            ref = float(f32[i] @ k32[idx_h[i]])
+            assert abs(out_h[i] - ref) < 0.05 + 0.02 * abs(ref), (i, out_h[i], ref)
+
+
+@requires_gpu
+def test_semcache_sketch_two_stage(hip):
+    """Two-stage sketch+verify lookup finds the right slot and reports the
+    TRUE full-dim cosine (not the sketch approximation)."""
+    from mcp_context_forge_amd.gpu.semcache import GpuSemanticCache, tool_hash
+    from mcp_context_forge_amd.gpu.batch import pad_rows
+
+    cache = GpuSemanticCache(capacity=512, dim=1024, threshold=0.9, ttl_s=600, sketch_dim=128)
+    assert cache.sketch_dim == 128
+    torch.manual_seed(7)
+    keys = torch.nn.functional.normalize(torch.randn(100, 1024), dim=1)
+    kp = pad_rows(keys.bfloat16().cuda().contiguous(), 128)
+    cache.insert_batch(kp, list(range(100)), [tool_hash("t")] * 100, [{"slot": i} for i in range(100)])
+    # queries: exact duplicates of keys 5 and 50, plus one random miss
+    q = torch.nn.functional.normalize(
+        torch.cat([keys[5:6], keys[50:51], torch.randn(1, 1024)]), dim=1)
+    qp = pad_rows(q.bfloat16().cuda().contiguous(), 128)
+    bv, bi = cache.lookup(qp)
+    torch.cuda.synchronize()
+    bi_h, bv_h = bi[:3].cpu().numpy(), bv[:3].cpu().numpy()
+    assert bi_h[0] == 5 and bi_h[1] == 50
+    assert bv_h[0] > 0.98 and bv_h[1] > 0.98  # verified TRUE cosine of a duplicate
+    hits = cache.resolve_hits(bv_h, bi_h, np.array([tool_hash("t")] * 3))
+    assert hits[0] == {"slot": 5} and hits[1] == {"slot": 50}
+    assert hits[2] is None  # random vector: verified cosine below threshold
