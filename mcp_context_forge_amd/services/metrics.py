@@ -25,6 +25,7 @@ class MetricsBuffer:
         self.flush_interval = flush_interval
         self.max_size = max_size
         self._rows: List[Dict[str, Any]] = []
+        self._agg: Dict[str, list] = {}   # tool_id -> [count, errors, latency_sum_ms]
         self._lock = threading.Lock()
         self._last_flush = time.monotonic()
         # live counters for /metrics + admin dashboards
@@ -77,26 +78,32 @@ class MetricsBuffer:
             self.flush()
 
     def record_aggregate_many(self, tool_ids: List[str], counts, errors, response_time_ms: float) -> None:
-        """One-lock batched aggregate (GPU pipeline: one row per tool per batch)."""
+        """One-lock batched aggregate (GPU pipeline: one row per tool per
+        batch). Accumulates in-memory per tool; rows materialize at flush
+        time, so the hot path never touches the DB."""
         with self._lock:
             total = 0
             total_err = 0
+            agg = self._agg
             for tid, cnt, err in zip(tool_ids, counts, errors):
                 cnt = int(cnt)
                 err = int(err)
                 if cnt <= 0:
                     continue
-                self._rows.append({"tool_id": tid, "response_time_ms": response_time_ms,
-                                   "is_success": err == 0, "error_message": None, "count": cnt})
-                self.latency_sum_ms[tid] += response_time_ms * cnt
+                lat = response_time_ms * cnt
+                a = agg.get(tid)
+                if a is None:
+                    agg[tid] = [cnt, err, lat]
+                else:
+                    a[0] += cnt
+                    a[1] += err
+                    a[2] += lat
+                self.latency_sum_ms[tid] += lat
                 self.latency_count[tid] += cnt
                 total += cnt
                 total_err += err
             self.counters["tool_invocations_total"] += total
             self.counters["tool_errors_total"] += total_err
-            need_flush = len(self._rows) >= self.max_size
-        if need_flush:
-            self.flush()
 
     def maybe_flush(self) -> None:
         if time.monotonic() - self._last_flush >= self.flush_interval:
@@ -105,6 +112,15 @@ class MetricsBuffer:
     def flush(self) -> int:
         with self._lock:
             rows, self._rows = self._rows, []
+            for tid, (cnt, err, lat) in self._agg.items():
+                avg = lat / max(cnt, 1)
+                if err:
+                    rows.append({"tool_id": tid, "response_time_ms": avg,
+                                 "is_success": False, "error_message": None, "count": err})
+                if cnt - err:
+                    rows.append({"tool_id": tid, "response_time_ms": avg,
+                                 "is_success": True, "error_message": None, "count": cnt - err})
+            self._agg.clear()
             self._last_flush = time.monotonic()
         if not rows or self.db is None:
             return len(rows)
@@ -117,7 +133,7 @@ class MetricsBuffer:
             top = sorted(self.latency_count.items(), key=lambda kv: -kv[1])[:25]
             return {
                 "counters": dict(self.counters),
-                "pending_rows": len(self._rows),
+                "pending_rows": len(self._rows) + len(self._agg),
                 "top_tools": [
                     {
                         "tool_id": t,
