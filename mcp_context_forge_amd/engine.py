@@ -33,7 +33,7 @@ from .services.governance import CatalogService, ContentSecurity, PasswordPolicy
 from .services.llm_proxy import LLMProxyError, LLMProxyService
 from .services.metrics import MetricsBuffer
 from .services.observability import AuditTrail, ObservabilityService
-from .services.sessions import SessionRegistry
+from .services.sessions import CancellationService, ElicitationService, SessionRegistry
 from .services.tool_service import ToolInvocationError, ToolNotFoundError, ToolService
 
 logger = logging.getLogger(__name__)
@@ -86,6 +86,8 @@ class GatewayEngine:
         self.content_security = ContentSecurity()
         self.audit = AuditTrail(self.db)
         self.sessions = SessionRegistry(self.settings.session_ttl, rank, world_size=world_size)
+        self.cancellations = CancellationService()
+        self.elicitation = ElicitationService(self.sessions)
         self.sync_plugin_bindings()
         self.started_at = time.time()
         self.gpu_pipeline = None  # attached lazily by gpu.pipeline when available
@@ -171,7 +173,12 @@ class GatewayEngine:
                     session.initialized = True
                     session.protocol_version = params.get("protocolVersion")
                 result = initialize_result(params.get("protocolVersion"))
-            elif method in ("ping", "notifications/initialized", "notifications/cancelled", "notifications/roots/list_changed"):
+            elif method == "notifications/cancelled":
+                # reference: cancellation_router — cancel the in-flight task
+                if session is not None and params.get("requestId") is not None:
+                    self.cancellations.cancel(session.session_id, params["requestId"])
+                return None
+            elif method in ("ping", "notifications/initialized", "notifications/roots/list_changed"):
                 if method != "ping":
                     return None  # notifications get no response
                 result = {}
@@ -194,9 +201,22 @@ class GatewayEngine:
                 if not isinstance(name, str) or not name:
                     raise jsonrpc.JSONRPCError(jsonrpc.INVALID_PARAMS, "missing tool name")
                 with self.observability.span("tools/call", tool=name, user=user or ""):
-                    result = await self.tool_service.invoke_tool(
+                    coro = self.tool_service.invoke_tool(
                         name, params.get("arguments") or {}, user=user, server_id=server_id, headers=headers
                     )
+                    if session is not None and rid is not None:
+                        # cancellable (MCP notifications/cancelled): run as a
+                        # registered task; a cancelled request gets NO response
+                        task = asyncio.ensure_future(coro)
+                        self.cancellations.register(session.session_id, rid, task)
+                        try:
+                            result = await task
+                        except asyncio.CancelledError:
+                            return None
+                        finally:
+                            self.cancellations.unregister(session.session_id, rid)
+                    else:
+                        result = await coro
             elif method == "resources/list":
                 result = {"resources": self.resource_service.list_resources()}
             elif method == "resources/templates/list":
@@ -278,6 +298,17 @@ class GatewayEngine:
         try:
             req = jsonrpc.parse_request_bytes(raw)
         except jsonrpc.JSONRPCError as exc:
+            # a JSON-RPC *response* object (no method) is a client reply to a
+            # server-initiated request — elicitation (reference: elicitation
+            # responses ride the same POST channel)
+            try:
+                obj = json.loads(raw)
+            except Exception:
+                obj = None
+            if isinstance(obj, dict) and "method" not in obj and obj.get("id") is not None \
+                    and ("result" in obj or "error" in obj):
+                self.elicitation.resolve(obj["id"], obj.get("result"), obj.get("error"))
+                return None
             return jsonrpc.JSONRPCResponse(id=None, error=exc).to_bytes()
         resp = await self.handle_rpc(req, user=user, server_id=server_id, session=session, headers=headers)
         return resp.to_bytes() if resp is not None else None

@@ -150,3 +150,79 @@ class SessionRegistry:
         for sid in stale:
             self.remove(sid)
         return len(stale)
+
+
+class CancellationService:
+    """In-flight request registry + MCP `notifications/cancelled` handling
+    (reference: routers/cancellation_router.py + session cancel service).
+
+    tools/call requests arriving on a session are wrapped in a task and
+    registered under (session_id, request_id); a cancellation notification
+    cancels the task and the request produces no response (MCP semantics).
+    """
+
+    def __init__(self):
+        self._inflight: Dict[Tuple[str, str], "asyncio.Task"] = {}
+        self.cancelled = 0
+
+    @staticmethod
+    def _key(session_id: str, rid: Any) -> Tuple[str, str]:
+        return (session_id, str(rid))
+
+    def register(self, session_id: str, rid: Any, task: "asyncio.Task") -> None:
+        self._inflight[self._key(session_id, rid)] = task
+
+    def unregister(self, session_id: str, rid: Any) -> None:
+        self._inflight.pop(self._key(session_id, rid), None)
+
+    def cancel(self, session_id: str, rid: Any) -> bool:
+        task = self._inflight.pop(self._key(session_id, rid), None)
+        if task is None or task.done():
+            return False
+        task.cancel()
+        self.cancelled += 1
+        return True
+
+    def inflight_count(self) -> int:
+        return len(self._inflight)
+
+
+class ElicitationService:
+    """Server→client mid-call input requests (reference: session elicitation
+    service; MCP `elicitation/create`). The request rides the session's
+    message stream; the client's JSON-RPC *response* resolves the future."""
+
+    def __init__(self, sessions: "SessionRegistry"):
+        self.sessions = sessions
+        self._pending: Dict[str, "asyncio.Future"] = {}
+        self._counter = itertools.count(1)
+
+    async def elicit(self, session_id: str, message: str, requested_schema: Optional[dict] = None,
+                     timeout_s: float = 60.0) -> Any:
+        eid = f"elicit-{next(self._counter)}"
+        fut: "asyncio.Future" = asyncio.get_running_loop().create_future()
+        self._pending[eid] = fut
+        ok = await self.sessions.broadcast(session_id, {
+            "jsonrpc": "2.0", "id": eid, "method": "elicitation/create",
+            "params": {"message": message,
+                       "requestedSchema": requested_schema or {"type": "object"}}})
+        if not ok:
+            self._pending.pop(eid, None)
+            raise RuntimeError(f"no live session {session_id!r} for elicitation")
+        try:
+            return await asyncio.wait_for(fut, timeout_s)
+        finally:
+            self._pending.pop(eid, None)
+
+    def resolve(self, eid: Any, result: Any, error: Any = None) -> bool:
+        fut = self._pending.get(str(eid))
+        if fut is None or fut.done():
+            return False
+        if error is not None:
+            fut.set_exception(RuntimeError(str(error)))
+        else:
+            fut.set_result(result)
+        return True
+
+    def pending_count(self) -> int:
+        return len(self._pending)
