@@ -32,7 +32,7 @@ from .services.gateway_service import GatewayService
 from .services.governance import CatalogService, ContentSecurity, PasswordPolicy, TokenBlocklist
 from .services.llm_proxy import LLMProxyError, LLMProxyService
 from .services.metrics import MetricsBuffer
-from .services.observability import AuditTrail, ObservabilityService
+from .services.observability import AuditTrail, ComplianceService, ObservabilityService, SiemExporter
 from .services.sessions import CancellationService, ElicitationService, SessionRegistry
 from .services.tool_service import ToolInvocationError, ToolNotFoundError, ToolService
 
@@ -85,6 +85,8 @@ class GatewayEngine:
         self.token_blocklist = TokenBlocklist()
         self.content_security = ContentSecurity()
         self.audit = AuditTrail(self.db)
+        self.siem = SiemExporter(self.db)
+        self.compliance = ComplianceService(self)
         self.sessions = SessionRegistry(self.settings.session_ttl, rank, world_size=world_size)
         self.cancellations = CancellationService()
         self.elicitation = ElicitationService(self.sessions)

@@ -168,3 +168,85 @@ class AuditTrail:
             rows = s.execute(select(DbAuditLog).order_by(DbAuditLog.id.desc()).limit(limit)).scalars().all()
             return [{"actor": r.actor, "action": r.action, "entity_type": r.entity_type,
                      "entity_id": r.entity_id, "detail": r.detail, "timestamp": str(r.timestamp)} for r in rows]
+
+
+class SiemExporter:
+    """Security-event export (reference: services/siem_export_service.py +
+    security_logger.py — JSON events shipped to OpenSearch/SIEM). Here:
+    audit + structured security events rendered as JSONL for pull-based
+    export (/admin/siem/export) or pushed to a webhook sink."""
+
+    def __init__(self, db: Database):
+        self.db = db
+        self.exported = 0
+
+    def export_jsonl(self, since: Optional[str] = None, limit: int = 1000) -> str:
+        import json as _json
+
+        from sqlalchemy import select
+
+        from ..db.models import DbAuditLog, DbStructuredLog
+
+        lines: List[str] = []
+        with self.db.session() as s:
+            q = select(DbAuditLog).order_by(DbAuditLog.id.desc()).limit(limit)
+            for r in s.execute(q).scalars():
+                lines.append(_json.dumps({
+                    "type": "audit", "timestamp": r.timestamp.isoformat(),
+                    "actor": r.actor, "action": r.action,
+                    "entity_type": r.entity_type, "entity_id": r.entity_id,
+                    "detail": r.detail}))
+            q = select(DbStructuredLog).where(DbStructuredLog.level.in_(("WARNING", "ERROR", "CRITICAL"))) \
+                .order_by(DbStructuredLog.id.desc()).limit(limit)
+            for r in s.execute(q).scalars():
+                lines.append(_json.dumps({
+                    "type": "log", "timestamp": r.timestamp.isoformat(), "level": r.level,
+                    "logger": r.logger, "message": r.message, "context": r.context}))
+        self.exported += len(lines)
+        return "\n".join(lines)
+
+    async def push_webhook(self, url: str, since: Optional[str] = None, limit: int = 1000) -> int:
+        import httpx
+
+        payload = self.export_jsonl(since, limit)
+        n = payload.count("\n") + 1 if payload else 0
+        async with httpx.AsyncClient(timeout=15.0) as c:
+            r = await c.post(url, content=payload.encode(),
+                             headers={"content-type": "application/x-ndjson"})
+            r.raise_for_status()
+        return n
+
+
+class ComplianceService:
+    """Compliance posture report (reference: services/compliance.py):
+    aggregate security-relevant configuration + usage into one snapshot."""
+
+    def __init__(self, engine):
+        self.engine = engine
+
+    def report(self) -> Dict[str, Any]:
+        e = self.engine
+        s = e.settings
+        plugins = [{"name": p.name, "mode": p.mode.value, "priority": p.priority}
+                   for p in e.plugins.plugins]
+        users = e.registry.db  # db handle; user counts via ORM
+        from sqlalchemy import func, select
+
+        from ..db.models import DbApiToken, DbAuditLog, DbUser
+
+        with e.db.session() as sess:
+            n_users = sess.execute(select(func.count()).select_from(DbUser)).scalar() or 0
+            n_tokens = sess.execute(select(func.count()).select_from(DbApiToken)).scalar() or 0
+            n_audit = sess.execute(select(func.count()).select_from(DbAuditLog)).scalar() or 0
+        return {
+            "auth": {"required": s.auth_required, "sso_enabled": bool(getattr(s, "sso_providers", None)),
+                     "users": n_users, "api_tokens": n_tokens},
+            "plugins": {"enabled": e.plugins.enabled, "count": len(plugins), "chain": plugins,
+                        "bindings": sum(len(v) for v in e.plugins.bindings.values())},
+            "transport": {"federation_enabled": s.federation_enabled,
+                          "rate_limit_rpm": getattr(s, "rate_limit_rpm", None),
+                          "max_body_bytes": getattr(s, "max_request_body_bytes", None)},
+            "audit": {"events": n_audit},
+            "gpu": {"pipeline_attached": e.gpu_pipeline is not None},
+            "uptime_s": round(__import__("time").time() - e.started_at, 1),
+        }

@@ -821,6 +821,25 @@ def build_app(engine: GatewayEngine, auth: Optional[AuthService] = None) -> Fast
         engine.observability.flush()
         return engine.observability.query_traces(limit)
 
+    @app.get("/admin/siem/export")
+    async def siem_export(limit: int = 1000, ctx: AuthContext = Depends(require("admin.read"))):
+        """Security events as NDJSON (reference: siem_export_service)."""
+        return PlainTextResponse(engine.siem.export_jsonl(limit=limit),
+                                 media_type="application/x-ndjson")
+
+    @app.post("/admin/siem/push")
+    async def siem_push(request: Request, ctx: AuthContext = Depends(require("admin.write"))):
+        body = await request.json()
+        url = body.get("url")
+        if not url:
+            raise HTTPException(422, "missing url")
+        n = await engine.siem.push_webhook(url, limit=int(body.get("limit", 1000)))
+        return {"pushed": n}
+
+    @app.get("/admin/compliance/report")
+    async def compliance_report(ctx: AuthContext = Depends(require("admin.read"))):
+        return engine.compliance.report()
+
     @app.get("/admin/audit")
     async def admin_audit(limit: int = 100, ctx: AuthContext = Depends(require("admin.read"))):
         return engine.audit.query(limit)

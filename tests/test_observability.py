@@ -89,3 +89,25 @@ def test_metric_rollup(run, bare_engine):
         assert s.execute(select(DbToolMetric)).scalars().all() == []  # raw pruned
         rolls = {r.entity_id: r for r in s.execute(select(DbMetricRollup)).scalars().all()}
         assert rolls["tool-a"].count == 5
+
+
+def test_siem_export_and_compliance(run):
+    from mcp_context_forge_amd.config import Settings
+    from mcp_context_forge_amd.engine import GatewayEngine
+
+    async def go():
+        e = GatewayEngine(Settings(database_url="sqlite://", federation_enabled=False,
+                                   auth_required=False))
+        e.audit.record("admin", "create", "tool", "t1")
+        e.audit.record("admin", "delete", "tool", "t1")
+        out = e.siem.export_jsonl()
+        lines = [json.loads(l) for l in out.splitlines()]
+        assert len(lines) >= 2
+        assert {l["action"] for l in lines if l["type"] == "audit"} == {"create", "delete"}
+        rep = e.compliance.report()
+        assert rep["plugins"]["count"] > 0
+        assert rep["audit"]["events"] >= 2
+        assert rep["gpu"]["pipeline_attached"] is False
+        await e.shutdown()
+
+    run(go())
