@@ -26,9 +26,11 @@ from .framework import HookType, Plugin, PluginManager
 
 def _all_builtin():
     from .content import EXTRA_PLUGINS
+    from .integrations import INTEGRATION_PLUGINS
 
     out = dict(BUILTIN_PLUGINS)
     out.update(EXTRA_PLUGINS)
+    out.update(INTEGRATION_PLUGINS)
     return out
 
 

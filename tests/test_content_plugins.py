@@ -21,7 +21,8 @@ from mcp_context_forge_amd.plugins.content import (
     WatchdogPlugin,
     WebhookNotificationPlugin,
 )
-from mcp_context_forge_amd.plugins.framework import HookType, PluginContext
+from mcp_context_forge_amd.plugins.framework import (HookType, PluginContext, PluginManager,
+                                                     PluginViolationError)
 
 
 def ctx(args, name="t", **kw):
@@ -142,3 +143,92 @@ def test_loader_resolves_extras():
 
     p = build_plugin({"name": "json_repair"})
     assert p.name == "json_repair"
+
+
+# -- integration plugins (reference: summarizer / virus_total / vault / unified_pdp)
+
+def test_virus_total_blocks_malicious_url(run):
+    from mcp_context_forge_amd.plugins.integrations import VirusTotalCheckerPlugin
+
+    async def fake_client(kind, value):
+        return {"malicious": 5 if "evil" in value else 0}
+
+    p = VirusTotalCheckerPlugin({"client": fake_client})
+    mgr = PluginManager([p])
+    with pytest.raises(PluginViolationError):
+        run(mgr.invoke_hook(HookType.TOOL_PRE_INVOKE,
+                            ctx({"url": "visit http://evil.example/payload now"})))
+    out = run(mgr.invoke_hook(HookType.TOOL_PRE_INVOKE, ctx({"url": "see http://good.example/x"})))
+    assert out.args["url"].startswith("see")
+    # verdicts cached: same URL → no second client call
+    n = p.checked
+    run(mgr.invoke_hook(HookType.TOOL_PRE_INVOKE, ctx({"url": "see http://good.example/x"})))
+    assert p.checked == n
+
+
+def test_vault_injects_token_without_clobbering(run):
+    from mcp_context_forge_amd.plugins.integrations import VaultPlugin
+
+    p = VaultPlugin({"rules": [{"match": "github-*", "header": "Authorization", "token": "tok123"}]})
+    c = ctx({"q": 1}, name="github-search")
+    run(PluginManager([p]).invoke_hook(HookType.TOOL_PRE_INVOKE, c))
+    assert c.headers["Authorization"] == "Bearer tok123"
+    # caller-supplied credentials are never replaced
+    c2 = ctx({"q": 1}, name="github-search")
+    c2.headers["Authorization"] = "Bearer mine"
+    run(PluginManager([p]).invoke_hook(HookType.TOOL_PRE_INVOKE, c2))
+    assert c2.headers["Authorization"] == "Bearer mine"
+    # non-matching tool untouched
+    c3 = ctx({}, name="other")
+    run(PluginManager([p]).invoke_hook(HookType.TOOL_PRE_INVOKE, c3))
+    assert "Authorization" not in c3.headers
+
+
+def test_unified_pdp_rules_and_default(run):
+    from mcp_context_forge_amd.plugins.integrations import UnifiedPdpPlugin
+
+    p = UnifiedPdpPlugin({"rules": [
+        {"id": "no-prod-writes", "effect": "deny", "tools": ["prod-*"], "contains": "delete"},
+        {"id": "allow-admin", "effect": "allow", "users": ["admin"]},
+    ], "default": "deny"})
+    mgr = PluginManager([p])
+    with pytest.raises(PluginViolationError):
+        run(mgr.invoke_hook(HookType.TOOL_PRE_INVOKE,
+                            ctx({"op": "delete row 5"}, name="prod-db")))
+    out = run(mgr.invoke_hook(HookType.TOOL_PRE_INVOKE, ctx({"q": 1}, name="x", user="admin")))
+    assert out is not None
+    with pytest.raises(PluginViolationError):  # default deny for unmatched
+        run(mgr.invoke_hook(HookType.TOOL_PRE_INVOKE, ctx({"q": 1}, name="x", user="guest")))
+
+
+def test_unified_pdp_external_engine(run):
+    from mcp_context_forge_amd.plugins.integrations import UnifiedPdpPlugin
+
+    async def engine(inp):
+        return {"allow": inp["tool"] != "blocked-tool", "reason": "policy says no"}
+
+    p = UnifiedPdpPlugin({"engine": engine})
+    mgr = PluginManager([p])
+    with pytest.raises(PluginViolationError):
+        run(mgr.invoke_hook(HookType.TOOL_PRE_INVOKE, ctx({}, name="blocked-tool")))
+    run(mgr.invoke_hook(HookType.TOOL_PRE_INVOKE, ctx({}, name="fine-tool")))
+
+
+def test_summarizer_uses_llm(run):
+    from mcp_context_forge_amd.plugins.integrations import SummarizerPlugin
+
+    class FakeLLM:
+        async def chat_completions(self, body):
+            return {"choices": [{"message": {"content": "short summary"}}]}
+
+    p = SummarizerPlugin({"llm": FakeLLM(), "threshold_chars": 100})
+    c = PluginContext(hook=HookType.TOOL_POST_INVOKE, name="t",
+                      args={"content": [{"type": "text", "text": "x" * 500}]})
+    out = run(PluginManager([p]).invoke_hook(HookType.TOOL_POST_INVOKE, c))
+    assert out.args["content"][0]["text"] == "short summary"
+    assert p.summarized == 1
+    # below threshold: untouched
+    c2 = PluginContext(hook=HookType.TOOL_POST_INVOKE, name="t",
+                       args={"content": [{"type": "text", "text": "tiny"}]})
+    out2 = run(PluginManager([p]).invoke_hook(HookType.TOOL_POST_INVOKE, c2))
+    assert out2.args["content"][0]["text"] == "tiny"
