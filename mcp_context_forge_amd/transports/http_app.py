@@ -869,6 +869,37 @@ def build_app(engine: GatewayEngine, auth: Optional[AuthService] = None) -> Fast
         engine.performance.snapshot()
         return engine.performance.history()
 
+    @app.get("/admin/runtime")
+    async def admin_runtime_get(ctx: AuthContext = Depends(require("admin.read"))):
+        """Runtime data-plane state (reference: runtime_state.py shadow↔edge +
+        runtime_admin_router PATCH): which path serves /rpc right now."""
+        col = app.state.collector
+        return {"mode": "gpu-batched" if (engine.gpu_pipeline is not None and col is not None)
+                else ("gpu-direct" if engine.gpu_pipeline is not None else "cpu"),
+                "gpu_pipeline": engine.gpu_pipeline is not None,
+                "collector": None if col is None else
+                {"window_us": int(col.window_s * 1e6), "max_batch": col.max_batch,
+                 "batches": col.batches, "max_seen": col.max_seen}}
+
+    @app.patch("/admin/runtime")
+    async def admin_runtime_patch(request: Request, ctx: AuthContext = Depends(require("admin.write"))):
+        """Runtime-flippable serving mode: batching window/size, or detach
+        the collector entirely (per-request mode) without a restart."""
+        body = await request.json()
+        col = app.state.collector
+        if "window_us" in body and col is not None:
+            col.window_s = max(0, int(body["window_us"])) / 1e6
+        if "max_batch" in body and col is not None:
+            col.max_batch = max(1, int(body["max_batch"]))
+        if body.get("mode") == "cpu" and engine.gpu_pipeline is not None:
+            app.state.collector = None     # per-request CPU chain
+        elif body.get("mode") == "gpu-batched" and engine.gpu_pipeline is not None and col is None:
+            app.state.collector = BatchCollector(engine.process_rpc_batch,
+                                                 settings.gpu_batch_max_requests,
+                                                 settings.gpu_batch_window_us)
+        engine.audit.record(ctx.user, "update", "runtime", None, detail=body)
+        return await admin_runtime_get(ctx)
+
     @app.post("/toolops/{tool_name}/test")
     async def toolops_test(tool_name: str, count: int = 3,
                            ctx: AuthContext = Depends(require("tools.invoke"))):

@@ -480,3 +480,19 @@ def test_tag_service_endpoints(client_engine, run):
             assert names == {"tagged-a", "srv-1"}
 
     run(go())
+
+
+def test_admin_runtime_state(client_engine, run):
+    """Runtime-flippable serving mode (reference: runtime_state shadow/edge)."""
+    client_factory, engine, app = client_engine
+
+    async def go():
+        async with client_factory() as c:
+            r = await c.get("/admin/runtime", headers=ADMIN)
+            assert r.status_code == 200
+            assert r.json()["mode"] == "cpu"  # no GPU in CI
+            assert r.json()["gpu_pipeline"] is False
+            r = await c.patch("/admin/runtime", headers=ADMIN, json={"window_us": 900})
+            assert r.status_code == 200
+
+    run(go())
