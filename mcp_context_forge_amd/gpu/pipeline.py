@@ -511,13 +511,20 @@ class GpuPluginPipeline:
         end_t = self._upload(args_e)
         self._toc("gp1_upload", t_g)
         t_l = self._tic()
+        # the scan banks are latency-bound and underfill the chip (one lane
+        # per request ⇒ ~128 waves over 256 CUs) — launch each bank on its
+        # own HIP stream so the six DFA sweeps and the MFMA chain overlap
         out: Dict[str, torch.Tensor] = {}
-        for bname in ("deny", "harm", "pii", "regex", "normalize"):
-            bank = self.banks.get(bname)
-            if bank is not None:
-                out[bname], _ = hip.scan(data_gpu, beg_t, end_t, bank)
+        scan_banks = [(n, self.banks[n]) for n in ("deny", "harm", "pii", "regex", "normalize")
+                      if n in self.banks]
         if self._schema_bank is not None:
-            out["schema"], _ = hip.scan(data_gpu, beg_t, end_t, self._schema_bank)
+            scan_banks.append(("schema", self._schema_bank))
+        main = torch.cuda.current_stream()
+        for i, (bname, bank) in enumerate(scan_banks):
+            s_ = self._side_stream(i)
+            s_.wait_stream(main)
+            with torch.cuda.stream(s_):
+                out[bname], _ = hip.scan(data_gpu, beg_t, end_t, bank)
         feats = None
         if self.classifier is not None or self.semcache is not None:
             feats_b, _ = hip.featurize(data_gpu, beg_t, end_t, self.feat_dim)
@@ -645,6 +652,12 @@ class GpuPluginPipeline:
             await self._dispatch_and_post(blob, env, rows, id_b, id_e, args_b, args_e,
                                           tool_idx, nk, feats, th_arr, native_js, py_items,
                                           responses, t0)
+
+    def _side_stream(self, i: int) -> "torch.cuda.Stream":
+        ss = getattr(self, "_streams", None)
+        if ss is None:
+            ss = self._streams = [torch.cuda.Stream() for _ in range(8)]
+        return ss[i % len(ss)]
 
     def _pin_reset(self) -> None:
         self._pin_off = 0
@@ -888,9 +901,13 @@ class GpuPluginPipeline:
             b3 = self._upload(res_beg.astype(np.int32))
             e3 = self._upload(res_end.astype(np.int32))
             masks3 = {}
-            for b in ("pii", "regex", "harm", "postmeta"):
+            main3 = torch.cuda.current_stream()
+            for i, b in enumerate(("pii", "regex", "harm", "postmeta")):
                 if b in self.banks:
-                    masks3[b], _ = hip.scan(data3, b3, e3, self.banks[b])
+                    s_ = self._side_stream(i)
+                    s_.wait_stream(main3)
+                    with torch.cuda.stream(s_):
+                        masks3[b], _ = hip.scan(data3, b3, e3, self.banks[b])
             await asyncio.to_thread(torch.cuda.synchronize)
             for b, t in masks3.items():
                 h = t.cpu().numpy().view(np.uint32) != 0
