@@ -137,6 +137,14 @@ class GpuPluginPipeline:
         from ..ops.pybridge import get as _pb_get
 
         self._pb = _pb_get()  # C response-assembly loops (fails loudly if missing)
+        # two batches may be in flight (engine splits big batches): GPU
+        # sections (shared pinned arena + stream) serialize on this lock,
+        # host/C++ sections interleave. A short GIL switch interval keeps
+        # to_thread C calls from stalling behind long Python stretches.
+        self._gpu_lock = asyncio.Lock()
+        import sys as _sys
+
+        _sys.setswitchinterval(0.001)
         # pinned-host staging arena (bump-allocated per pass; reset after each sync)
         self._pin = torch.empty(16 << 20, dtype=torch.uint8, pin_memory=True)
         self._pin_off = 0
@@ -310,7 +318,14 @@ class GpuPluginPipeline:
             off += len(nb)
             self._t_name_end[i] = off
         if getattr(self, "_toolmap", 0):
-            hip.toolmap_free(self._toolmap)
+            # defer freeing: a concurrent in-flight batch may still resolve
+            # against the old map (freed after the next few rebuilds)
+            g = getattr(self, "_toolmap_graveyard", None)
+            if g is None:
+                g = self._toolmap_graveyard = []
+            g.append(self._toolmap)
+            while len(g) > 8:
+                hip.toolmap_free(g.pop(0))
         self._toolmap = hip.toolmap_new(self._t_name_blob, self._t_name_beg, self._t_name_end) if nt else 0
         self._t_required = np.array([m.required_bits for m in metas], dtype=np.uint32) if nt else np.zeros(1, dtype=np.uint32)
         typed = np.full(max(nt, 1), 0xFFFFFFFFFFFFFFFF, dtype=np.uint64)
@@ -422,7 +437,7 @@ class GpuPluginPipeline:
         np.cumsum(np.fromiter(map(len, raws), dtype=np.int64, count=n), out=offsets[1:])
         joined = b"".join(raws)
         blob = np.frombuffer(joined, dtype=np.uint8) if joined else np.zeros(1, dtype=np.uint8)
-        env = hip.parse_envelopes(blob, offsets)
+        env = await asyncio.to_thread(hip.parse_envelopes, blob, offsets)
         self._toc("pack_envelope", t0)
         kind = env["kind"]
 
@@ -505,6 +520,7 @@ class GpuPluginPipeline:
 
         # --- GPU pass 1 over raw argument spans ---
         t_g = self._tic()
+        await self._gpu_lock.acquire()
         self._pin_reset()
         data_gpu = self._upload(blob)
         beg_t = self._upload(args_b)
@@ -531,14 +547,18 @@ class GpuPluginPipeline:
             feats = pad_rows(feats_b, 128)
         scores_t = self.classifier.forward(feats)[:m] if self.classifier is not None else None
         cache_val_t = cache_idx_t = None
+        feats_sk = None
         if self.semcache is not None:
-            bv, bi = self.semcache.lookup(feats)
+            bv, bi, feats_sk = self.semcache.lookup(feats)
             cache_val_t, cache_idx_t = bv[:m], bi[:m]
         self._toc("gp1_launch", t_l)
         t_s = self._tic()
         # off-loop sync: a blocking synchronize would stall the event loop and
         # starve the micro-batch collector under live HTTP load
-        await asyncio.to_thread(torch.cuda.synchronize)
+        try:
+            await asyncio.to_thread(torch.cuda.synchronize)
+        finally:
+            self._gpu_lock.release()
         self._toc("gp1_sync", t_s)
         self._toc("gpu_pass1", t_g)
         t_d = self._tic()
@@ -578,7 +598,7 @@ class GpuPluginPipeline:
                 flags[open_tis] |= hip.TF_BREAKER_OPEN
 
         t0 = time.monotonic()
-        state, nk, reason, arena, rb, re_, n_arena = hip.decide(
+        state, nk, reason, arena, rb, re_, n_arena = await asyncio.to_thread(hip.decide,
             blob, id_b, id_e, args_b, args_e, tool_idx, nb, ne,
             deny_m, harm_m, pii_m, regex_m, norm_m, schema_m,
             mod_block, mod_cat, mod_score, hit, hit_slot,
@@ -651,7 +671,7 @@ class GpuPluginPipeline:
         if native_js or py_items:
             await self._dispatch_and_post(blob, env, rows, id_b, id_e, args_b, args_e,
                                           tool_idx, nk, feats, th_arr, native_js, py_items,
-                                          responses, t0)
+                                          responses, t0, feats_sk)
 
     def _side_stream(self, i: int) -> "torch.cuda.Stream":
         ss = getattr(self, "_streams", None)
@@ -795,7 +815,8 @@ class GpuPluginPipeline:
     async def _dispatch_and_post(self, blob, env, rows, id_b, id_e, args_b, args_e,
                                  tool_idx, nk, feats, th_arr,
                                  native_js: List[int], py_items: List[Tuple[int, Any]],
-                                 responses: List[Optional[bytes]], t0: float) -> None:
+                                 responses: List[Optional[bytes]], t0: float,
+                                 feats_sk=None) -> None:
         # --- native upstream batch (C++) ---
         t_u = self._tic()
         nat_blob = np.zeros(0, dtype=np.uint8)
@@ -807,7 +828,8 @@ class GpuPluginPipeline:
             ne_ = np.ascontiguousarray(args_e[njs])
             nb_ = np.where(ne_ > nb_, nb_, -1).astype(np.int32)
             now_iso = time.strftime("%Y-%m-%dT%H:%M:%SZ", time.gmtime())
-            nat_blob, nat_beg, nat_end = hip.upstream_call_batch(blob, nb_, ne_, kinds, now_iso)
+            nat_blob, nat_beg, nat_end = await asyncio.to_thread(
+                hip.upstream_call_batch, blob, nb_, ne_, kinds, now_iso)
 
         # --- python dispatch (non-native upstreams / rewritten args) ---
         py_results: List[Optional[bytes]] = []
@@ -896,6 +918,7 @@ class GpuPluginPipeline:
         post_flag = np.zeros(n_all, dtype=bool)
         toon_meta = np.zeros(n_all, dtype=bool)
         if n_all:
+            await self._gpu_lock.acquire()
             self._pin_reset()
             data3 = self._upload(res_blob)
             b3 = self._upload(res_beg.astype(np.int32))
@@ -908,7 +931,10 @@ class GpuPluginPipeline:
                     s_.wait_stream(main3)
                     with torch.cuda.stream(s_):
                         masks3[b], _ = hip.scan(data3, b3, e3, self.banks[b])
-            await asyncio.to_thread(torch.cuda.synchronize)
+            try:
+                await asyncio.to_thread(torch.cuda.synchronize)
+            finally:
+                self._gpu_lock.release()
             for b, t in masks3.items():
                 h = t.cpu().numpy().view(np.uint32) != 0
                 if b == "postmeta":
@@ -927,7 +953,7 @@ class GpuPluginPipeline:
         t_f = self._tic()
         t_fc = self._tic()
         now = time.monotonic()
-        arena2, rb2, re2, is_err, cacheable = hip.finalize(
+        arena2, rb2, re2, is_err, cacheable = await asyncio.to_thread(hip.finalize,
             blob, id_b, id_e, args_b, args_e, tool_idx,
             np.ascontiguousarray(all_js_np.astype(np.int32)), res_blob, res_beg, res_end,
             np.ascontiguousarray(needs_host.astype(np.uint8)),
@@ -978,7 +1004,7 @@ class GpuPluginPipeline:
             if ins.size:
                 ins_js = all_js_np[ins]
                 slots = self.semcache.assign_slots(int(ins.size))
-                self.semcache.insert_features(feats, ins_js, slots, th_arr[ins_js])
+                self.semcache.insert_features(feats, ins_js, slots, th_arr[ins_js], sketch=feats_sk)
                 if self._slot_store:
                     hip.store_put_batch(self._slot_store, slots, res_blob,
                                         np.ascontiguousarray(res_beg[ins]),

@@ -54,7 +54,7 @@ class GpuSemanticCache:
         else:
             self.proj_t = None
             self.keys_sk = None
-        self._last_sk: Optional[torch.Tensor] = None  # sketches of the last lookup batch
+
         # host-side metadata mirrors (slot-indexed)
         self.tool_hashes = np.zeros(capacity, dtype=np.int64)
         self.timestamps = np.zeros(capacity, dtype=np.float64)
@@ -64,7 +64,7 @@ class GpuSemanticCache:
         self.hits = 0
         self.misses = 0
 
-    def lookup(self, feats_bf16: torch.Tensor) -> Tuple[torch.Tensor, torch.Tensor]:
+    def lookup(self, feats_bf16: torch.Tensor) -> Tuple[torch.Tensor, torch.Tensor, Optional[torch.Tensor]]:
         """feats [Bpad, dim] bf16 → (best_val fp32 [Bpad], best_idx i32 [Bpad]) on device.
 
         Chunked sweep over the key matrix; scores for padded/invalid slots are
@@ -73,28 +73,31 @@ class GpuSemanticCache:
         bpad = feats_bf16.shape[0]
         best_val = torch.full((bpad,), -1e30, dtype=torch.float32, device=self.device)
         best_idx = torch.full((bpad,), -1, dtype=torch.int32, device=self.device)
+        sk = None
         if self.sketch_dim:
-            # always compute sketches (reused by insert_features this batch)
-            self._last_sk = hip.gemm_bt(feats_bf16, self.proj_t).to(torch.bfloat16)
+            # sketches returned to the caller, which passes them back into
+            # insert_features (no shared slot — batches may be in flight
+            # concurrently)
+            sk = hip.gemm_bt(feats_bf16, self.proj_t).to(torch.bfloat16)
         if self.size == 0:
-            return best_val, best_idx
+            return best_val, best_idx, sk
         active = min(self.size, self.capacity)
         active_pad = ((active + self.chunk - 1) // self.chunk) * self.chunk
         if self.sketch_dim:
             # stage 1: approximate ranking over the sketch matrix
             for c0 in range(0, active_pad, self.chunk):
                 nc = min(self.chunk, self.capacity - c0)
-                scores = hip.gemm_bt(self._last_sk, self.keys_sk[c0:c0 + nc])
+                scores = hip.gemm_bt(sk, self.keys_sk[c0:c0 + nc])
                 hip.rows_argmax_merge(scores, best_val, best_idx, idx_base=c0,
                                       valid=self.valid[c0:c0 + nc])
             # stage 2: exact full-dim dot of each row's top candidate
             best_val = hip.verify_dot(feats_bf16, self.keys, best_idx)
-            return best_val, best_idx
+            return best_val, best_idx, sk
         for c0 in range(0, active_pad, self.chunk):
             nc = min(self.chunk, self.capacity - c0)
             scores = hip.gemm_bt(feats_bf16, self.keys[c0:c0 + nc])
             hip.rows_argmax_merge(scores, best_val, best_idx, idx_base=c0, valid=self.valid[c0:c0 + nc])
-        return best_val, best_idx
+        return best_val, best_idx, sk
 
     def resolve_hits_np(self, best_val: np.ndarray, best_idx: np.ndarray,
                         tool_hashes: np.ndarray) -> np.ndarray:
@@ -143,14 +146,14 @@ class GpuSemanticCache:
         return slots.astype(np.int32)
 
     def insert_features(self, feats_bf16: torch.Tensor, rows: np.ndarray, slots: np.ndarray,
-                        tool_hashes: np.ndarray) -> None:
+                        tool_hashes: np.ndarray, sketch: Optional[torch.Tensor] = None) -> None:
         """Fused gather+scatter of feature rows into pre-assigned slots
         (one kernel; results are stored separately — C++ slot store)."""
         rows_t = torch.from_numpy(rows.astype(np.int32)).to(self.device, non_blocking=True)
         slot_t = torch.from_numpy(slots.astype(np.int32)).to(self.device, non_blocking=True)
         hip.rows_gather_scatter_bf16(feats_bf16, rows_t, slot_t, self.keys, self.valid)
         if self.sketch_dim:
-            sk = self._last_sk
+            sk = sketch
             if sk is None or sk.shape[0] != feats_bf16.shape[0]:
                 sk = hip.gemm_bt(feats_bf16, self.proj_t).to(torch.bfloat16)
             hip.rows_gather_scatter_bf16(sk, rows_t, slot_t, self.keys_sk, self.valid)

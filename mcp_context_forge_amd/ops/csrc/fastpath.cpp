@@ -24,6 +24,7 @@
 #include <stdint.h>
 #include <string.h>
 #include <stdio.h>
+#include <mutex>
 #include <string>
 #include <thread>
 #include <unordered_map>
@@ -35,6 +36,7 @@ namespace {
 
 struct SlotStore {
     std::vector<std::string> slots;
+    std::mutex mu;   // concurrent batches: put_batch vs decide-side get
 };
 
 struct ExactEntry {
@@ -44,11 +46,14 @@ struct ExactEntry {
 };
 
 struct ExactCache {
-    // sharded by hash so finalize's insert phase can run one thread per
-    // shard without locks (decide's lookups are single-threaded)
+    // sharded by hash: finalize's insert phase runs one thread per shard,
+    // and with two batches in flight decide(A) can overlap finalize(B) —
+    // each shard carries its own mutex
     static constexpr int SHARDS = 8;
     std::unordered_map<uint64_t, ExactEntry> shards[SHARDS];
+    std::mutex shard_mu[SHARDS];
     double ttl = 300.0;
+    int shard_of(uint64_t h) const { return (int)(h & (SHARDS - 1)); }
     std::unordered_map<uint64_t, ExactEntry>& map_for(uint64_t h) {
         return shards[h & (SHARDS - 1)];
     }
@@ -124,6 +129,7 @@ void* forge_store_new(int capacity) {
 }
 void forge_store_put(void* store, int slot, const uint8_t* data, int64_t n) {
     auto* s = (SlotStore*)store;
+    std::lock_guard<std::mutex> g(s->mu);
     s->slots[slot].assign((const char*)data, (size_t)n);
 }
 void forge_store_free(void* store) { delete (SlotStore*)store; }
@@ -275,9 +281,16 @@ int64_t forge_decide(
             }
         }
         if ((fl & TF_CACHE) && hit[i] && store != nullptr) {
-            const std::string& res = store->slots[hit_slot[i]];
-            if (!res.empty()) {
-                emit_result(a, i, blob, idb, ide, res.data(), res.size());
+            bool answered_from_store = false;
+            {
+                std::lock_guard<std::mutex> g(store->mu);
+                const std::string& res = store->slots[hit_slot[i]];
+                if (!res.empty()) {
+                    emit_result(a, i, blob, idb, ide, res.data(), res.size());
+                    answered_from_store = true;
+                }
+            }
+            if (answered_from_store) {
                 state[i] = ST_ANSWERED;
                 reason_out[i] = 6;
                 continue;
@@ -287,10 +300,18 @@ int64_t forge_decide(
             uint64_t h = fnv64(tname_blob + tname_beg[ti], (size_t)(tname_end[ti] - tname_beg[ti]));
             h = fnv64((const uint8_t*)"\x00", 1, h);
             h = fnv64(blob + args_beg[i], (size_t)(args_end[i] - args_beg[i]), h);
-            auto& shard = ec->map_for(h);
-            auto it = shard.find(h);
-            if (it != shard.end() && it->second.expires > now) {
-                emit_result(a, i, blob, idb, ide, it->second.value.data(), it->second.value.size());
+            int sh = ec->shard_of(h);
+            bool answered_from_exact = false;
+            {
+                std::lock_guard<std::mutex> g(ec->shard_mu[sh]);
+                auto& shard = ec->shards[sh];
+                auto it = shard.find(h);
+                if (it != shard.end() && it->second.expires > now) {
+                    emit_result(a, i, blob, idb, ide, it->second.value.data(), it->second.value.size());
+                    answered_from_exact = true;
+                }
+            }
+            if (answered_from_exact) {
                 state[i] = ST_ANSWERED;
                 reason_out[i] = 7;
                 continue;
@@ -390,6 +411,7 @@ int64_t forge_finalize(
     if (ec != nullptr) {
         auto insert_shard = [&](int s) {
             double exp = now + ec->ttl;
+            std::lock_guard<std::mutex> g(ec->shard_mu[s]);
             for (auto& vec : inserts)
                 for (auto& kv : vec) {
                     if ((int)(kv.first & (ExactCache::SHARDS - 1)) != s) continue;
@@ -454,6 +476,7 @@ void forge_toolmap_resolve(void* map, const uint8_t* blob,
 void forge_store_put_batch(void* store, const int32_t* slots, int n,
                            const uint8_t* blob, const int64_t* beg, const int64_t* end) {
     auto* s = (SlotStore*)store;
+    std::lock_guard<std::mutex> g(s->mu);
     for (int i = 0; i < n; ++i) {
         if (slots[i] >= 0 && slots[i] < (int32_t)s->slots.size())
             s->slots[slots[i]].assign((const char*)blob + beg[i], (size_t)(end[i] - beg[i]));
@@ -463,9 +486,14 @@ void forge_store_put_batch(void* store, const int32_t* slots, int n,
 }  // extern "C"
 
 
-extern "C" int64_t forge_store_get(void* store, int slot, const uint8_t** ptr) {
+// copy-out under the lock: a concurrent put may reassign the slot string,
+// so returning an interior pointer would dangle with two batches in flight
+extern "C" int64_t forge_store_get(void* store, int slot, uint8_t* out, int64_t cap) {
     auto* s = (SlotStore*)store;
+    std::lock_guard<std::mutex> g(s->mu);
     if (slot < 0 || slot >= (int)s->slots.size()) return -1;
-    *ptr = (const uint8_t*)s->slots[slot].data();
-    return (int64_t)s->slots[slot].size();
+    const std::string& v = s->slots[slot];
+    if ((int64_t)v.size() <= cap && !v.empty())
+        memcpy(out, v.data(), v.size());
+    return (int64_t)v.size();
 }

@@ -272,7 +272,7 @@ def _fastpath_protos() -> ctypes.CDLL:
         [ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int] * 3 + \
         [ctypes.c_void_p, ctypes.c_void_p, ctypes.c_double] + \
         [ctypes.c_void_p] * 4 + [ctypes.c_int64] + [ctypes.c_void_p] * 2
-    lib.forge_store_get.argtypes = [ctypes.c_void_p, ctypes.c_int, ctypes.c_void_p]
+    lib.forge_store_get.argtypes = [ctypes.c_void_p, ctypes.c_int, ctypes.c_void_p, ctypes.c_int64]
     lib.forge_store_get.restype = ctypes.c_int64
     lib.forge_decide.restype = ctypes.c_int64
     lib.forge_finalize.argtypes = [ctypes.c_void_p] * 6 + [ctypes.c_int] + [ctypes.c_void_p, ctypes.c_int] + \
@@ -397,8 +397,11 @@ ST_DISPATCH_NATIVE, ST_ANSWERED, ST_REWRITE, ST_HOST_SCHEMA, ST_DISPATCH_PY = 0,
 
 def store_get(store: int, slot: int) -> Optional[bytes]:
     lib = _fastpath_protos()
-    ptr = ctypes.c_void_p(0)
-    n = lib.forge_store_get(ctypes.c_void_p(store), slot, ctypes.byref(ptr))
-    if n < 0 or not ptr.value:
+    buf = ctypes.create_string_buffer(4096)
+    n = lib.forge_store_get(ctypes.c_void_p(store), slot, buf, 4096)
+    if n < 0 or n == 0:
         return None
-    return ctypes.string_at(ptr.value, int(n))
+    if n > 4096:
+        buf = ctypes.create_string_buffer(int(n))
+        n = lib.forge_store_get(ctypes.c_void_p(store), slot, buf, int(n))
+    return buf.raw[:int(n)]

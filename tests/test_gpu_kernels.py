@@ -207,7 +207,7 @@ def test_semcache_hit_miss_ttl(hip):
 
     fp = pad_rows(feats, 128)
     cache.insert_batch(fp, [0], [tool_hash("t")], [{"cached": True}])
-    bv, bi = cache.lookup(fp)
+    bv, bi, _sk = cache.lookup(fp)
     torch.cuda.synchronize()
     hits = cache.resolve_hits(bv[:2].cpu().numpy(), bi[:2].cpu().numpy(),
                               np.array([tool_hash("t"), tool_hash("t")]))
@@ -317,7 +317,7 @@ def test_semcache_sketch_two_stage(hip):
     q = torch.nn.functional.normalize(
         torch.cat([keys[5:6], keys[50:51], torch.randn(1, 1024)]), dim=1)
     qp = pad_rows(q.bfloat16().cuda().contiguous(), 128)
-    bv, bi = cache.lookup(qp)
+    bv, bi, _sk = cache.lookup(qp)
     torch.cuda.synchronize()
     bi_h, bv_h = bi[:3].cpu().numpy(), bv[:3].cpu().numpy()
     assert bi_h[0] == 5 and bi_h[1] == 50
