@@ -142,9 +142,6 @@ class GpuPluginPipeline:
         # host/C++ sections interleave. A short GIL switch interval keeps
         # to_thread C calls from stalling behind long Python stretches.
         self._gpu_lock = asyncio.Lock()
-        import sys as _sys
-
-        _sys.setswitchinterval(0.001)
         # pinned-host staging arena (bump-allocated per pass; reset after each sync)
         self._pin = torch.empty(16 << 20, dtype=torch.uint8, pin_memory=True)
         self._pin_off = 0
@@ -437,7 +434,7 @@ class GpuPluginPipeline:
         np.cumsum(np.fromiter(map(len, raws), dtype=np.int64, count=n), out=offsets[1:])
         joined = b"".join(raws)
         blob = np.frombuffer(joined, dtype=np.uint8) if joined else np.zeros(1, dtype=np.uint8)
-        env = await asyncio.to_thread(hip.parse_envelopes, blob, offsets)
+        env = hip.parse_envelopes(blob, offsets)
         self._toc("pack_envelope", t0)
         kind = env["kind"]
 
@@ -598,7 +595,7 @@ class GpuPluginPipeline:
                 flags[open_tis] |= hip.TF_BREAKER_OPEN
 
         t0 = time.monotonic()
-        state, nk, reason, arena, rb, re_, n_arena = await asyncio.to_thread(hip.decide,
+        state, nk, reason, arena, rb, re_, n_arena = hip.decide(
             blob, id_b, id_e, args_b, args_e, tool_idx, nb, ne,
             deny_m, harm_m, pii_m, regex_m, norm_m, schema_m,
             mod_block, mod_cat, mod_score, hit, hit_slot,
@@ -828,8 +825,7 @@ class GpuPluginPipeline:
             ne_ = np.ascontiguousarray(args_e[njs])
             nb_ = np.where(ne_ > nb_, nb_, -1).astype(np.int32)
             now_iso = time.strftime("%Y-%m-%dT%H:%M:%SZ", time.gmtime())
-            nat_blob, nat_beg, nat_end = await asyncio.to_thread(
-                hip.upstream_call_batch, blob, nb_, ne_, kinds, now_iso)
+            nat_blob, nat_beg, nat_end = hip.upstream_call_batch(blob, nb_, ne_, kinds, now_iso)
 
         # --- python dispatch (non-native upstreams / rewritten args) ---
         py_results: List[Optional[bytes]] = []
@@ -953,7 +949,7 @@ class GpuPluginPipeline:
         t_f = self._tic()
         t_fc = self._tic()
         now = time.monotonic()
-        arena2, rb2, re2, is_err, cacheable = await asyncio.to_thread(hip.finalize,
+        arena2, rb2, re2, is_err, cacheable = hip.finalize(
             blob, id_b, id_e, args_b, args_e, tool_idx,
             np.ascontiguousarray(all_js_np.astype(np.int32)), res_blob, res_beg, res_end,
             np.ascontiguousarray(needs_host.astype(np.uint8)),
