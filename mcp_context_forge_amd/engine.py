@@ -71,6 +71,9 @@ class GatewayEngine:
         self.completion_service = CompletionService(self.registry)
         self.root_service = RootService()
         self.llm_proxy = LLMProxyService()
+        from .services.chat_service import McpChatService
+
+        self.chat = McpChatService(self)
         from .services.diagnostics import PerformanceService, SupportBundle, ToolOps
 
         self.support_bundle = SupportBundle(self)

@@ -32,6 +32,7 @@ from ..protocol import jsonrpc
 from ..protocol.mcp import PROTOCOL_VERSION, initialize_result
 from ..registry.registry import ConflictError, NotFoundError
 from ..services.gateway_service import GatewayConnectionError
+from ..services.llm_proxy import LLMProxyError
 from ..utils import TokenBucket
 
 # ---------------------------------------------------------------------------
@@ -630,6 +631,22 @@ def build_app(engine: GatewayEngine, auth: Optional[AuthService] = None) -> Fast
                 stream = engine.llm_proxy.chat_completions_stream(body)
                 return StreamingResponse(stream, media_type="text/event-stream")
             return await engine.llm_proxy.chat_completions(body)
+        except LLMProxyError as exc:
+            raise HTTPException(exc.status, str(exc)) from exc
+
+    @app.post("/chat")
+    async def llm_chat(request: Request, ctx: AuthContext = Depends(require("tools.invoke"))):
+        """Built-in LLM chat with tool-calling over the registry
+        (reference: routers/llmchat_router.py + mcp_client_chat_service)."""
+        body = await request.json()
+        messages = body.get("messages")
+        if not isinstance(messages, list) or not messages:
+            raise HTTPException(422, "missing messages")
+        try:
+            return await engine.chat.chat(messages, model=body.get("model"),
+                                          provider=body.get("provider"), user=ctx.user,
+                                          server_id=ctx.server_id,
+                                          max_rounds=body.get("max_rounds"))
         except LLMProxyError as exc:
             raise HTTPException(exc.status, str(exc)) from exc
 

@@ -208,3 +208,54 @@ def test_sampling_without_provider(run, bare_engine):
         assert json.loads(out)["error"]["code"] == -32601
 
     run(go())
+
+
+def test_chat_service_tool_loop():
+    """Built-in chat with tool-calling (reference: mcp_client_chat_service):
+    the fake provider first requests a tool call, then answers using the
+    tool result; the service round-trips through invoke_tool (plugins on)."""
+    fake = FastAPI()
+    calls = {"n": 0}
+
+    @fake.post("/v1/chat/completions")
+    async def completions(request: Request):
+        body = await request.json()
+        calls["n"] += 1
+        if calls["n"] == 1:
+            assert any(t["function"]["name"] == "adder" for t in body["tools"])
+            return {"id": "c1", "model": "fake-model",
+                    "choices": [{"index": 0, "message": {
+                        "role": "assistant", "content": None,
+                        "tool_calls": [{"id": "tc1", "type": "function",
+                                        "function": {"name": "adder",
+                                                     "arguments": json.dumps({"a": 2, "b": 3})}}]},
+                        "finish_reason": "tool_calls"}]}
+        tool_msg = [m for m in body["messages"] if m.get("role") == "tool"][0]
+        val = json.loads(tool_msg["content"])["sum"]
+        return {"id": "c2", "model": "fake-model",
+                "choices": [{"index": 0, "message": {"role": "assistant",
+                                                     "content": f"the sum is {val}"},
+                             "finish_reason": "stop"}]}
+
+    async def go():
+        port = _free_port()
+        server, task = await _serve(fake, port)
+        engine = GatewayEngine(Settings(database_url="sqlite://", federation_enabled=False, auth_required=False))
+        try:
+            engine.llm_proxy.registry.register("fake", f"http://127.0.0.1:{port}/v1",
+                                               models=["fake-model"], default_model="fake-model")
+
+            async def adder(args):
+                return {"sum": args["a"] + args["b"]}
+
+            engine.tool_service.register_local_tool("adder", adder, "Add two numbers")
+            out = await engine.chat.chat([{"role": "user", "content": "add 2 and 3"}])
+            assert out["message"]["content"] == "the sum is 5"
+            assert out["tool_calls"][0]["tool"] == "adder" and out["tool_calls"][0]["ok"]
+            assert out["rounds"] == 1
+        finally:
+            await engine.shutdown()
+            server.should_exit = True
+            await asyncio.wait_for(task, timeout=10)
+
+    asyncio.run(go())
