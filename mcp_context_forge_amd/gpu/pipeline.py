@@ -430,9 +430,8 @@ class GpuPluginPipeline:
         responses: List[Optional[bytes]] = [None] * n
 
         t0 = self._tic()
-        offsets = np.zeros(n + 1, dtype=np.int64)
-        np.cumsum(np.fromiter(map(len, raws), dtype=np.int64, count=n), out=offsets[1:])
-        joined = b"".join(raws)
+        joined, offs_raw = self._pb.concat_with_offsets(raws)
+        offsets = np.frombuffer(offs_raw, dtype=np.int64)
         blob = np.frombuffer(joined, dtype=np.uint8) if joined else np.zeros(1, dtype=np.uint8)
         env = hip.parse_envelopes(blob, offsets)
         self._toc("pack_envelope", t0)

@@ -104,11 +104,45 @@ fail:
     return NULL;
 }
 
+/* concat_with_offsets(list_of_bytes) -> (bytes, int64 offsets bytearray)
+ * One C pass: total size, offsets array (n+1 int64 little-endian in a
+ * bytes object the caller views as numpy), and the joined blob. */
+static PyObject *concat_with_offsets(PyObject *self, PyObject *args) {
+    PyObject *lst;
+    if (!PyArg_ParseTuple(args, "O!", &PyList_Type, &lst))
+        return NULL;
+    Py_ssize_t n = PyList_GET_SIZE(lst);
+    PyObject *offs_obj = PyBytes_FromStringAndSize(NULL, (n + 1) * (Py_ssize_t)sizeof(int64_t));
+    if (!offs_obj) return NULL;
+    int64_t *offs = (int64_t *)PyBytes_AS_STRING(offs_obj);
+    int64_t total = 0;
+    offs[0] = 0;
+    for (Py_ssize_t i = 0; i < n; i++) {
+        PyObject *it = PyList_GET_ITEM(lst, i);
+        Py_ssize_t sz;
+        if (PyBytes_Check(it)) sz = PyBytes_GET_SIZE(it);
+        else { PyErr_SetString(PyExc_TypeError, "expected bytes"); Py_DECREF(offs_obj); return NULL; }
+        total += sz;
+        offs[i + 1] = total;
+    }
+    PyObject *blob = PyBytes_FromStringAndSize(NULL, total);
+    if (!blob) { Py_DECREF(offs_obj); return NULL; }
+    char *w = PyBytes_AS_STRING(blob);
+    for (Py_ssize_t i = 0; i < n; i++) {
+        PyObject *it = PyList_GET_ITEM(lst, i);
+        Py_ssize_t sz = PyBytes_GET_SIZE(it);
+        memcpy(w + offs[i], PyBytes_AS_STRING(it), (size_t)sz);
+    }
+    return Py_BuildValue("NN", blob, offs_obj);
+}
+
 static PyMethodDef Methods[] = {
     {"scatter_slices", scatter_slices, METH_VARARGS,
      "scatter arena spans into a responses list as bytes"},
     {"slices_list", slices_list, METH_VARARGS,
      "extract arena spans into a new list of bytes/None"},
+    {"concat_with_offsets", concat_with_offsets, METH_VARARGS,
+     "join a list of bytes into one blob + int64 offsets"},
     {NULL, NULL, 0, NULL},
 };
 
