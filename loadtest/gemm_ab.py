@@ -68,7 +68,7 @@ def main():
         "v1_tf": round(flop / min(t1, t1b) / 1e12, 1),
         "v2_tf": round(flop / min(t2, t2b) / 1e12, 1),
     }
-    for var in (1, 2, 3):
+    for var in (1, 2, 3, 4, 5):
         fv = mkvar(var)
         rv = fv()
         torch.cuda.synchronize()

@@ -31,6 +31,16 @@ __device__ __forceinline__ uint32_t swz2(uint32_t L) {
     return L ^ (((L >> 6) & 3u) << 4);
 }
 
+// swizzle ablation (PMC: VAR2 still shows ~25M LDS bank conflicts/call):
+//   VAR 4: the guide's m201 st_16x32 formula (row&8 flips col-byte bit 5)
+//   VAR 5: two-bit spread (row&8 -> bit5, row&4 -> bit4)
+template <int VAR>
+__device__ __forceinline__ uint32_t swz_sel(uint32_t L) {
+    if constexpr (VAR == 4) return L ^ (((L >> 9) & 1u) << 5);
+    else if constexpr (VAR == 5) return L ^ (((L >> 9) & 1u) << 5) ^ (((L >> 8) & 1u) << 4);
+    else return L ^ (((L >> 6) & 3u) << 4);
+}
+
 template <int ACT, bool OUT_BF16, bool HAS_BIAS, int VAR = 0>
 __global__ __launch_bounds__(512, 1) void gemm_bt_v2_kernel(
     const short* __restrict__ A,   // [M,K] bf16
@@ -79,7 +89,7 @@ __global__ __launch_bounds__(512, 1) void gemm_bt_v2_kernel(
     // inverse-swizzled source, swizzled read).
     auto stage_a = [&](int t, int i) {
         uint32_t P = (uint32_t)(i * 512 + tid) * 16u;
-        uint32_t L = swz2(P);
+        uint32_t L = swz_sel<VAR>(P);
         uint32_t row = L >> 6, colb = L & 63u;
         const short* g = A + ((size_t)(tile_m + row) * K + t * V2_BK) + (colb >> 1);
         __builtin_amdgcn_global_load_lds(
@@ -90,7 +100,7 @@ __global__ __launch_bounds__(512, 1) void gemm_bt_v2_kernel(
     };
     auto stage_b = [&](int t, int i) {
         uint32_t P = (uint32_t)(i * 512 + tid) * 16u;
-        uint32_t L = swz2(P);
+        uint32_t L = swz_sel<VAR>(P);
         uint32_t row = L >> 6, colb = L & 63u;
         const short* g = BT + ((size_t)(tile_n + row) * K + t * V2_BK) + (colb >> 1);
         __builtin_amdgcn_global_load_lds(
@@ -132,15 +142,15 @@ __global__ __launch_bounds__(512, 1) void gemm_bt_v2_kernel(
                 #pragma unroll
                 for (int fc = 0; fc < 4; ++fc) {
                     uint32_t row = b_row_base + fc * 16;
-                    b_frag[fc] = *(const bf16x8_v2*)(slot_b + swz2(row * V2_ROWB + a_col));
+                    b_frag[fc] = *(const bf16x8_v2*)(slot_b + swz_sel<VAR>(row * V2_ROWB + a_col));
                 }
             }
             bf16x8_v2 a0, a1;
             {
                 uint32_t r0 = a_row_base + (2 * q) * 16;
                 uint32_t r1 = a_row_base + (2 * q + 1) * 16;
-                a0 = *(const bf16x8_v2*)(slot_a + swz2(r0 * V2_ROWB + a_col));
-                a1 = *(const bf16x8_v2*)(slot_a + swz2(r1 * V2_ROWB + a_col));
+                a0 = *(const bf16x8_v2*)(slot_a + swz_sel<VAR>(r0 * V2_ROWB + a_col));
+                a1 = *(const bf16x8_v2*)(slot_a + swz_sel<VAR>(r1 * V2_ROWB + a_col));
             }
             if constexpr (VAR < 2) __builtin_amdgcn_s_setprio(1);
             #pragma unroll
@@ -196,6 +206,8 @@ extern "C" int forge_gemm_bt_v2_var(
         case 1: hipLaunchKernelGGL((gemm_bt_v2_kernel<0, false, false, 1>), grid, block, 0, s, (const short*)A, (const short*)BT, nullptr, C, M, N, K); break;
         case 2: hipLaunchKernelGGL((gemm_bt_v2_kernel<0, false, false, 2>), grid, block, 0, s, (const short*)A, (const short*)BT, nullptr, C, M, N, K); break;
         case 3: hipLaunchKernelGGL((gemm_bt_v2_kernel<0, false, false, 3>), dim3((M / V2_BM) * (N / V2_BN)), block, 0, s, (const short*)A, (const short*)BT, nullptr, C, M, N, K); break;
+        case 4: hipLaunchKernelGGL((gemm_bt_v2_kernel<0, false, false, 4>), grid, block, 0, s, (const short*)A, (const short*)BT, nullptr, C, M, N, K); break;
+        case 5: hipLaunchKernelGGL((gemm_bt_v2_kernel<0, false, false, 5>), grid, block, 0, s, (const short*)A, (const short*)BT, nullptr, C, M, N, K); break;
         default: return 9002;
     }
     return (int)hipGetLastError();
