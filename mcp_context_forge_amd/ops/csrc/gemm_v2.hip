@@ -222,11 +222,12 @@ extern "C" int forge_gemm_bt_v2(
     dim3 grid(M / V2_BM, N / V2_BN);
     dim3 block(512);
     bool hb = bias != nullptr;
-    // production configuration = VAR 2 (free-running ring + counted vmcnt;
-    // phase barriers/setprio measured NEGATIVE at this occupancy — see
-    // profiles/README.md ablation table)
+    // production configuration = VAR 4 (free-running ring + counted vmcnt +
+    // m201 st_16x32 swizzle — ablation table in profiles/README.md: 1133 TF
+    // vs 1094 for the old row&3 spread; phase barriers/setprio measured
+    // NEGATIVE at this occupancy)
     #define DISPATCH2(A_, O_, B_)                                                                \
-        hipLaunchKernelGGL((gemm_bt_v2_kernel<A_, O_, B_, 2>), grid, block, 0, s,                \
+        hipLaunchKernelGGL((gemm_bt_v2_kernel<A_, O_, B_, 4>), grid, block, 0, s,                \
                            (const short*)A, (const short*)BT, (const float*)bias, C, M, N, K)
     switch (act * 4 + (out_bf16 ? 2 : 0) + (hb ? 1 : 0)) {
         case 0: DISPATCH2(0, false, false); break;
