@@ -12,8 +12,11 @@ to feed the GPU pipeline (uvicorn+asyncio ≈ 7-8k RPS/process), so:
     to the owner (batched per event-loop tick), everything else is proxied
     to the owner's private port.
 
-Wire framing (worker→owner): [u32 n] n×{[u64 req_id][u32 len][bytes]}
-(owner→worker): same with len = 0xFFFFFFFF meaning "no response" (202).
+Wire framing (both directions): [u32 payload_bytes][u32 n] then payload
+n×{[u64 req_id][u32 len][bytes]} (len = 0xFFFFFFFF means "no response",
+202). The leading byte count lets each side read a whole frame in two
+reads and parse it with struct.unpack_from — the first protocol did one
+readexactly per field (3 awaits per request) and capped the owner loop.
 """
 
 from __future__ import annotations
@@ -75,28 +78,36 @@ class GpuOwnerServer:
         try:
             while True:
                 try:
-                    (n,) = _U32.unpack(await _read_exact(reader, 4))
+                    nbytes, n = struct.unpack("<II", await _read_exact(reader, 8))
+                    payload = await _read_exact(reader, nbytes)
                 except (asyncio.IncompleteReadError, ConnectionResetError):
                     return
                 self.frames += 1
-                items: List[tuple] = []
+                ids: List[int] = []
+                bodies: List[bytes] = []
+                off = 0
                 for _ in range(n):
-                    (req_id,) = _U64.unpack(await _read_exact(reader, 8))
-                    (ln,) = _U32.unpack(await _read_exact(reader, 4))
-                    body = await _read_exact(reader, ln)
-                    items.append((req_id, body))
-                self.requests += len(items)
+                    req_id, ln = struct.unpack_from("<QI", payload, off)
+                    off += 12
+                    bodies.append(payload[off:off + ln])
+                    off += ln
+                    ids.append(req_id)
+                self.requests += n
 
-                async def run_frame(items=items):
-                    outs = await self._process_frame([b for (_i, b) in items])
-                    parts = [_U32.pack(len(items))]
-                    for (req_id, _b), out in zip(items, outs):
+                async def run_frame(ids=ids, bodies=bodies):
+                    outs = await self._process_frame(bodies)
+                    parts = [b"", _U32.pack(len(ids))]
+                    total = 0
+                    for req_id, out in zip(ids, outs):
                         parts.append(_U64.pack(req_id))
                         if out is None:
                             parts.append(_U32.pack(NO_RESPONSE))
+                            total += 12
                         else:
                             parts.append(_U32.pack(len(out)))
                             parts.append(bytes(out))
+                            total += 12 + len(out)
+                    parts[0] = _U32.pack(total)
                     async with write_lock:
                         writer.write(b"".join(parts))
                         await writer.drain()
@@ -131,11 +142,17 @@ class OwnerClient:
     async def _read_loop(self) -> None:
         try:
             while True:
-                (n,) = _U32.unpack(await _read_exact(self.reader, 4))
+                nbytes, n = struct.unpack("<II", await _read_exact(self.reader, 8))
+                payload = await _read_exact(self.reader, nbytes) if nbytes else b""
+                off = 0
                 for _ in range(n):
-                    (req_id,) = _U64.unpack(await _read_exact(self.reader, 8))
-                    (ln,) = _U32.unpack(await _read_exact(self.reader, 4))
-                    body = None if ln == NO_RESPONSE else await _read_exact(self.reader, ln)
+                    req_id, ln = struct.unpack_from("<QI", payload, off)
+                    off += 12
+                    if ln == NO_RESPONSE:
+                        body = None
+                    else:
+                        body = payload[off:off + ln]
+                        off += ln
                     fut = self._futures.pop(req_id, None)
                     if fut is not None and not fut.done():
                         fut.set_result(body)
@@ -150,11 +167,14 @@ class OwnerClient:
         if not self._pending or self.writer is None:
             return
         items, self._pending = self._pending, []
-        parts = [_U32.pack(len(items))]
+        parts = [b"", _U32.pack(len(items))]
+        total = 0
         for req_id, body in items:
             parts.append(_U64.pack(req_id))
             parts.append(_U32.pack(len(body)))
             parts.append(body)
+            total += 12 + len(body)
+        parts[0] = _U32.pack(total)
         self.writer.write(b"".join(parts))
 
     async def submit(self, raw: bytes) -> Optional[bytes]:
