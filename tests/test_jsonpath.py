@@ -72,3 +72,11 @@ def test_errors():
         evaluate(DOC, "$.store[")
     with pytest.raises(JSONPathError):
         evaluate(DOC, "$.store.book[?(!!)]")
+
+
+def test_trailing_recursive_descent_is_error():
+    # mutation-tier finding: `$..` at end of expression must raise
+    # JSONPathError, not IndexError
+    with pytest.raises(JSONPathError):
+        evaluate(DOC, "$..")
+    assert jsonpath_filter(DOC, "$..") is None  # filter maps parse errors to None

@@ -284,3 +284,16 @@ def test_binding_reenables_globally_disabled_plugin(run):
     mgr.set_bindings({"t": {"deny_filter": {"mode": "enforce"}}})
     with pytest.raises(PluginViolationError):
         run(mgr.invoke_hook(HookType.TOOL_PRE_INVOKE, ctx({"q": "classified"}, name="t")))
+
+
+def test_toon_scalar_quoting_exact():
+    # mutation-tier finding: pin the exact quoting rules of TOON scalars
+    from mcp_context_forge_amd.plugins.toon import _scalar
+
+    assert _scalar("plain") == "plain"          # simple strings stay bare
+    assert _scalar("") == '""'                  # empty must quote
+    assert _scalar("true") == '"true"'          # literal-lookalikes must quote
+    assert _scalar("12.5") == '"12.5"'          # numeric-lookalikes must quote
+    assert _scalar(True) == "true"
+    assert _scalar(False) == "false"
+    assert _scalar(None) == "null"
