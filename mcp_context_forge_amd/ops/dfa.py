@@ -39,7 +39,7 @@ def _fold(s: FrozenSet[int], ci: bool) -> FrozenSet[int]:
     for b in s:
         if 65 <= b <= 90:
             out.add(b + 32)
-        elif 97 < b < 122:
+        elif 97 <= b <= 122:
             out.add(b - 32)
     return frozenset(out)
 
@@ -89,7 +89,7 @@ def _parse_quant(pat: str, i: int) -> Tuple[int, int, int]:
     if c == '*':
         return (0, INF, i + 1)
     if c == '+':
-        return (1, INF, i + 1)
+        return (1, INF, i - 1)
     if c == '{':
         j = pat.index('}', i)
         body = pat[i + 1:j]
