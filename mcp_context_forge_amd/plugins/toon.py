@@ -178,7 +178,8 @@ def decode(text: str) -> Any:
                 is_array = True
                 pos += 1
                 if body == "-":
-                    items.append(parse_block(level + 2))
+                    # block item: contents are one level deeper than the dash
+                    items.append(parse_block(ind + 1))
                 else:
                     items.append(_parse_scalar(body[2:]))
                 continue

@@ -107,3 +107,42 @@ def test_agreement_with_re_on_random_text():
         mask = dfa.match_mask_reference(t, s.encode())
         for pid, r in enumerate(rx):
             assert bool(mask & (1 << pid)) == bool(r.search(s)), (s, pats[pid])
+
+
+# -- mutation-tier findings: pin boundary behavior ------------------------
+
+def test_case_fold_boundaries():
+    # 'A'/'Z' and 'a'/'z' are inside the fold ranges (0x41<=b<=0x5A etc.)
+    t = dfa.compile_patterns(["AZ"], case_insensitive=True)
+    assert dfa.scan_reference(t, b"az")
+    assert dfa.scan_reference(t, b"AZ")
+    t2 = dfa.compile_patterns(["az"], case_insensitive=True)
+    assert dfa.scan_reference(t2, b"AZ")
+
+
+def test_negated_class():
+    t = dfa.compile_patterns(["x[^0-9]y"], case_insensitive=False)
+    assert dfa.scan_reference(t, b"xay")
+    assert not dfa.scan_reference(t, b"x5y")
+
+
+def test_class_range_boundaries_inclusive():
+    t = dfa.compile_patterns(["[x-z]q"], case_insensitive=False)
+    assert dfa.scan_reference(t, b"xq")
+    assert dfa.scan_reference(t, b"zq")   # upper bound inclusive
+    assert not dfa.scan_reference(t, b"wq")
+
+
+def test_star_quantifier_zero_occurrences():
+    t = dfa.compile_patterns(["ab*c"], case_insensitive=False)
+    assert dfa.scan_reference(t, b"ac")    # zero b's
+    assert dfa.scan_reference(t, b"abbbc")
+    assert not dfa.scan_reference(t, b"axc")
+
+
+def test_literal_compile_escapes_metachars():
+    # compile_literals must treat '.' and '[' as literals, not regex atoms
+    t = dfa.compile_literals(["a.c", "x[1]"], case_insensitive=False)
+    assert dfa.scan_reference(t, b"a.c")
+    assert not dfa.scan_reference(t, b"abc")
+    assert dfa.scan_reference(t, b"x[1]")

@@ -80,3 +80,16 @@ def test_trailing_recursive_descent_is_error():
     with pytest.raises(JSONPathError):
         evaluate(DOC, "$..")
     assert jsonpath_filter(DOC, "$..") is None  # filter maps parse errors to None
+
+
+def test_recursive_wildcard():
+    assert evaluate({"a": {"b": 1}}, "$..*") == [{"b": 1}, 1]
+
+
+def test_filter_numeric_boundaries():
+    # operator table must keep strict vs inclusive comparisons distinct
+    docs = [{"p": 8.95}, {"p": 10}, {"p": 12}]
+    assert evaluate(docs, "$[?(@.p < 10)].p") == [8.95]
+    assert evaluate(docs, "$[?(@.p <= 10)].p") == [8.95, 10]
+    assert evaluate(docs, "$[?(@.p > 10)].p") == [12]
+    assert evaluate(docs, "$[?(@.p >= 10)].p") == [10, 12]

@@ -297,3 +297,15 @@ def test_toon_scalar_quoting_exact():
     assert _scalar(True) == "true"
     assert _scalar(False) == "false"
     assert _scalar(None) == "null"
+
+
+def test_toon_decode_escapes_and_single_pair():
+    # mutation-tier: quoted-string escape handling + single-line forms
+    from mcp_context_forge_amd.plugins import toon as toon_codec
+
+    assert toon_codec.decode('k: "a\\"b"') == {"k": 'a"b'}
+    assert toon_codec.decode("k: v") == {"k": "v"}
+    assert toon_codec.decode("plain") == "plain"
+    # nested list-of-dicts roundtrip (indent arithmetic under '-' items)
+    obj = {"rows": [{"a": 1}, {"a": 2, "b": [1, 2]}]}
+    assert toon_codec.decode(toon_codec.encode(obj)) == obj
