@@ -146,3 +146,16 @@ def test_literal_compile_escapes_metachars():
     assert dfa.scan_reference(t, b"a.c")
     assert not dfa.scan_reference(t, b"abc")
     assert dfa.scan_reference(t, b"x[1]")
+
+
+def test_class_trailing_dash_and_optional():
+    # '[a-]' : dash at class end is a literal
+    t = dfa.compile_patterns(["x[a-]y"], case_insensitive=False)
+    assert dfa.scan_reference(t, b"xay")
+    assert dfa.scan_reference(t, b"x-y")
+    assert not dfa.scan_reference(t, b"xby")
+    # optional quantifier
+    t2 = dfa.compile_patterns(["ab?c"], case_insensitive=False)
+    assert dfa.scan_reference(t2, b"ac")
+    assert dfa.scan_reference(t2, b"abc")
+    assert not dfa.scan_reference(t2, b"abbc")

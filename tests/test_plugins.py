@@ -309,3 +309,11 @@ def test_toon_decode_escapes_and_single_pair():
     # nested list-of-dicts roundtrip (indent arithmetic under '-' items)
     obj = {"rows": [{"a": 1}, {"a": 2, "b": [1, 2]}]}
     assert toon_codec.decode(toon_codec.encode(obj)) == obj
+
+
+def test_toon_tabular_quoted_commas_and_escapes():
+    from mcp_context_forge_amd.plugins import toon as toon_codec
+
+    obj = {"rows": [{"a": 'x,y', "b": 1}, {"a": 'q"z', "b": 2}]}
+    enc = toon_codec.encode(obj)
+    assert toon_codec.decode(enc) == obj
