@@ -65,6 +65,7 @@ class Settings(BaseModel):
 
     # --- transports / sessions ---
     sse_keepalive_interval: int = 30
+    session_persistence: bool = False  # DB-backed session continuity (reference: database backend)
     session_ttl: int = 3600
     message_ttl: int = 600
     event_store_max_events: int = 512

@@ -90,7 +90,8 @@ class GatewayEngine:
         self.audit = AuditTrail(self.db)
         self.siem = SiemExporter(self.db)
         self.compliance = ComplianceService(self)
-        self.sessions = SessionRegistry(self.settings.session_ttl, rank, world_size=world_size)
+        self.sessions = SessionRegistry(self.settings.session_ttl, rank, world_size=world_size,
+                                        db=self.db if self.settings.session_persistence else None)
         self.cancellations = CancellationService()
         self.elicitation = ElicitationService(self.sessions)
         self.sync_plugin_bindings()

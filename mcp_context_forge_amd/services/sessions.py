@@ -77,15 +77,59 @@ class EventStore:
 
 
 class SessionRegistry:
-    """Memory-backend session registry (reference: cache/session_registry.py:105-215)."""
+    """Session registry: memory backend with optional DB durability
+    (reference: cache/session_registry.py:105-215 memory/database backends;
+    the redis backend's cross-worker role is rank-ownership here).
+
+    With a `db` attached, create/remove/touch mirror into mcp_sessions so a
+    restarted gateway can tell a resumed session id (client may replay with
+    Last-Event-ID) from an unknown one."""
 
     def __init__(self, ttl_s: float = 3600.0, rank: int = 0, event_store: Optional[EventStore] = None,
-                 world_size: int = 1):
+                 world_size: int = 1, db=None):
         self.ttl_s = ttl_s
         self.rank = rank
         self.world_size = max(1, world_size)
         self._sessions: Dict[str, Session] = {}
         self.event_store = event_store or EventStore()
+        self.db = db
+        self.resumable: Dict[str, dict] = {}
+        if db is not None:
+            self._load_persisted()
+
+    def _load_persisted(self) -> None:
+        import datetime as _dt
+
+        from sqlalchemy import delete, select
+
+        from ..db.models import DbSessionRecord, utcnow
+
+        cutoff = utcnow() - _dt.timedelta(seconds=self.ttl_s)
+        with self.db.session() as s:
+            s.execute(delete(DbSessionRecord).where(DbSessionRecord.last_accessed < cutoff))
+            for r in s.execute(select(DbSessionRecord)).scalars():
+                self.resumable[r.session_id] = {"transport": r.transport,
+                                                "server_id": r.server_id, "user": r.user_email}
+
+    def _persist(self, sess: "Session") -> None:
+        if self.db is None:
+            return
+        from ..db.models import DbSessionRecord
+
+        with self.db.session() as s:
+            s.merge(DbSessionRecord(session_id=sess.session_id, transport=sess.transport,
+                                    owner_rank=sess.owner_rank, server_id=sess.server_id,
+                                    user_email=sess.user))
+
+    def _unpersist(self, session_id: str) -> None:
+        if self.db is None:
+            return
+        from sqlalchemy import delete
+
+        from ..db.models import DbSessionRecord
+
+        with self.db.session() as s:
+            s.execute(delete(DbSessionRecord).where(DbSessionRecord.session_id == session_id))
 
     def create(self, transport: str = "streamablehttp", server_id: Optional[str] = None,
                user: Optional[str] = None, session_id: Optional[str] = None) -> Session:
@@ -102,6 +146,7 @@ class SessionRegistry:
                     break
         sess = Session(session_id=sid, transport=transport, server_id=server_id, user=user, owner_rank=self.rank)
         self._sessions[sid] = sess
+        self._persist(sess)
         return sess
 
     def get(self, session_id: str) -> Optional[Session]:
@@ -112,7 +157,9 @@ class SessionRegistry:
 
     def remove(self, session_id: str) -> None:
         self._sessions.pop(session_id, None)
+        self.resumable.pop(session_id, None)
         self.event_store.drop(session_id)
+        self._unpersist(session_id)
 
     def count(self) -> int:
         return len(self._sessions)
@@ -150,6 +197,21 @@ class SessionRegistry:
         for sid in stale:
             self.remove(sid)
         return len(stale)
+
+    def resume(self, session_id: str, transport: str = "streamablehttp") -> Optional["Session"]:
+        """Re-materialize a persisted session after a restart (reference:
+        database session backend — the client resumes with its old id and
+        replays from Last-Event-ID; server-side queue starts fresh)."""
+        meta = self.resumable.pop(session_id, None)
+        if meta is None:
+            return None
+        sess = Session(session_id=session_id, transport=meta.get("transport") or transport,
+                       server_id=meta.get("server_id"), user=meta.get("user"),
+                       owner_rank=self.rank)
+        sess.initialized = True
+        self._sessions[session_id] = sess
+        self._persist(sess)
+        return sess
 
 
 class CancellationService:

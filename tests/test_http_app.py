@@ -496,3 +496,32 @@ def test_admin_runtime_state(client_engine, run):
             assert r.status_code == 200
 
     run(go())
+
+
+def test_session_persistence_across_restart(tmp_path, run):
+    """DB session backend (reference: cache/session_registry.py database
+    backend): a session survives an engine restart as resumable metadata."""
+    from mcp_context_forge_amd.config import Settings
+    from mcp_context_forge_amd.engine import GatewayEngine
+
+    async def go():
+        url = f"sqlite:///{tmp_path}/sess.db"
+        e1 = GatewayEngine(Settings(database_url=url, federation_enabled=False,
+                                    auth_required=False, session_persistence=True))
+        sess = e1.sessions.create(transport="streamablehttp", user="alice")
+        sid = sess.session_id
+        gone = e1.sessions.create()
+        e1.sessions.remove(gone.session_id)
+        await e1.shutdown()
+
+        e2 = GatewayEngine(Settings(database_url=url, federation_enabled=False,
+                                    auth_required=False, session_persistence=True))
+        assert sid in e2.sessions.resumable
+        assert gone.session_id not in e2.sessions.resumable
+        resumed = e2.sessions.resume(sid)
+        assert resumed is not None and resumed.user == "alice" and resumed.initialized
+        assert e2.sessions.get(sid) is resumed
+        assert e2.sessions.resume("unknown-id") is None
+        await e2.shutdown()
+
+    run(go())
