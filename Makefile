@@ -26,3 +26,6 @@ smoke:
 	$(PY) __graft_entry__.py && $(PY) __graft_entry__.py smoke
 
 .PHONY: test test-gpu build serve bench bench-cpu load smoke
+
+mutation:        ## mutation-testing sweep over core modules (writes profiles/mutation_report.txt)
+	python tools/mutation_check.py --stride 5
