@@ -857,6 +857,27 @@ def build_app(engine: GatewayEngine, auth: Optional[AuthService] = None) -> Fast
     async def compliance_report(ctx: AuthContext = Depends(require("admin.read"))):
         return engine.compliance.report()
 
+    @app.get("/admin/logs")
+    async def admin_logs(level: str = "", logger_name: str = "", q: str = "",
+                         limit: int = 200, ctx: AuthContext = Depends(require("admin.read"))):
+        """Structured-log search (reference: routers/log_search.py +
+        log_storage_service): filter persisted logs by level/logger/text."""
+        from sqlalchemy import select
+
+        from ..db.models import DbStructuredLog
+
+        stmt = select(DbStructuredLog).order_by(DbStructuredLog.id.desc()).limit(min(limit, 2000))
+        if level:
+            stmt = stmt.where(DbStructuredLog.level == level.upper())
+        if logger_name:
+            stmt = stmt.where(DbStructuredLog.logger.like(f"%{logger_name}%"))
+        if q:
+            stmt = stmt.where(DbStructuredLog.message.like(f"%{q}%"))
+        with engine.db.session() as s_:
+            return [{"timestamp": r.timestamp.isoformat(), "level": r.level,
+                     "logger": r.logger, "message": r.message, "context": r.context}
+                    for r in s_.execute(stmt).scalars()]
+
     @app.get("/admin/audit")
     async def admin_audit(limit: int = 100, ctx: AuthContext = Depends(require("admin.read"))):
         return engine.audit.query(limit)

@@ -111,3 +111,43 @@ def test_siem_export_and_compliance(run):
         await e.shutdown()
 
     run(go())
+
+
+def test_admin_log_search_endpoint(run):
+    """Log search (reference: routers/log_search.py)."""
+    import base64
+
+    import httpx
+
+    from mcp_context_forge_amd.config import Settings
+    from mcp_context_forge_amd.engine import GatewayEngine
+    from mcp_context_forge_amd.services.observability import DbLogHandler
+    from mcp_context_forge_amd.transports.http_app import build_app
+
+    async def go():
+        e = GatewayEngine(Settings(database_url="sqlite://", federation_enabled=False,
+                                   auth_required=True))
+        h = DbLogHandler(e.db)
+        import logging as _l
+
+        rec = _l.LogRecord("forge.test", _l.WARNING, __file__, 1, "upstream flaked badly", (), None)
+        h.emit(rec)
+        rec2 = _l.LogRecord("other.mod", _l.ERROR, __file__, 2, "boom", (), None)
+        h.emit(rec2)
+        h.flush_to_db()
+        app = build_app(e)
+        admin = {"Authorization": "Basic " + base64.b64encode(b"admin:changeme").decode()}
+        async with app.router.lifespan_context(app):
+            async with httpx.AsyncClient(transport=httpx.ASGITransport(app=app),
+                                         base_url="http://gw") as c:
+                r = await c.get("/admin/logs", headers=admin)
+                assert r.status_code == 200 and len(r.json()) == 2
+                r = await c.get("/admin/logs", headers=admin, params={"level": "error"})
+                assert [x["message"] for x in r.json()] == ["boom"]
+                r = await c.get("/admin/logs", headers=admin, params={"q": "flaked"})
+                assert len(r.json()) == 1
+                r = await c.get("/admin/logs", headers=admin, params={"logger_name": "forge"})
+                assert len(r.json()) == 1
+        await e.shutdown()
+
+    run(go())
