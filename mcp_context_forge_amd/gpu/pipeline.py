@@ -658,7 +658,8 @@ class GpuPluginPipeline:
         if rewrite_js:
             self.slow_path += len(rewrite_js)
             rewrite_dispatch = await self._rewrite_pass(
-                blob, env, rows, args_b, args_e, rewrite_js, tool_idx, hit, hit_slot, responses)
+                blob, env, rows, args_b, args_e, rewrite_js, tool_idx, hit, hit_slot, responses,
+                pii_m=pii_m, regex_m=regex_m, norm_m=norm_m)
 
         self._toc("answer_assign", t_a)
         py_items = [(int(j), None) for j in np.nonzero(state == hip.ST_DISPATCH_PY)[0]] + \
@@ -707,14 +708,19 @@ class GpuPluginPipeline:
         return res if isinstance(res, bytes) else json.dumps(res, separators=(",", ":")).encode()
 
     # ------------------------------------------------------------------
-    def _apply_rewrites(self, name: str, args: Any) -> Tuple[str, Any]:
+    def _apply_rewrites(self, name: str, args: Any, do_norm: bool = True,
+                        do_regex: bool = True, do_pii: bool = True) -> Tuple[str, Any]:
         """Host rewrites in CPU-chain priority order: normalizer(15) →
-        regex(20) → pii(30), using the plugins' own functions."""
-        if self.normalizer is not None and self._applies(self.normalizer, name):
+        regex(20) → pii(30), using the plugins' own functions. The do_*
+        flags come from the per-bank GPU masks: a bank that did not fire on
+        this row means its plugin is an identity transform here (the DFA
+        banks are conservative supersets of the host matchers), so it can
+        be skipped without changing the outcome."""
+        if do_norm and self.normalizer is not None and self._applies(self.normalizer, name):
             args = _walk_strings(args, self.normalizer.norm)
-        if self.regex is not None and self._applies(self.regex, name):
+        if do_regex and self.regex is not None and self._applies(self.regex, name):
             args = _walk_strings(args, self.regex.apply_rules)
-        if self.pii is not None and self._applies(self.pii, name):
+        if do_pii and self.pii is not None and self._applies(self.pii, name):
             found: List[str] = []
 
             def fn(s: str) -> str:
@@ -731,7 +737,8 @@ class GpuPluginPipeline:
 
 
     async def _rewrite_pass(self, blob, env, rows, args_b, args_e, rewrite_js, tool_idx,
-                            hit, hit_slot, responses) -> List[Tuple[int, Any]]:
+                            hit, hit_slot, responses,
+                            pii_m=None, regex_m=None, norm_m=None) -> List[Tuple[int, Any]]:
         ok_items: List[Tuple[int, Any]] = []
         for j in rewrite_js:
             mt = self._meta_list[tool_idx[j]]
@@ -742,7 +749,11 @@ class GpuPluginPipeline:
             except Exception:
                 responses[r] = self._splice_error(idb, jsonrpc.INVALID_PARAMS, "invalid arguments")
                 continue
-            status, payload = self._apply_rewrites(mt.name, args)
+            status, payload = self._apply_rewrites(
+                mt.name, args,
+                do_norm=bool(norm_m[j]) if norm_m is not None else True,
+                do_regex=bool(regex_m[j]) if regex_m is not None else True,
+                do_pii=bool(pii_m[j]) if pii_m is not None else True)
             if status == "__block__":
                 responses[r] = self._splice_error(idb, jsonrpc.POLICY_DENIED, payload)
                 self.blocked += 1
