@@ -39,6 +39,10 @@ class GatewayService:
         self.world_size = world_size
         self._health_task: Optional[asyncio.Task] = None
         self._stop = asyncio.Event()
+        from ..auth.crypto import EncryptionService
+
+        # credential material is sealed at rest (reference: EncryptedText db.py:277)
+        self.crypto = EncryptionService(self.settings.jwt_secret_key)
 
     # -- client construction ----------------------------------------------------
     def _make_client(self, gateway: Dict[str, Any]) -> UpstreamClient:
@@ -46,18 +50,19 @@ class GatewayService:
 
         headers: Dict[str, str] = {}
         token_provider = None
-        if gateway.get("auth_type") == "bearer" and gateway.get("auth_value"):
-            headers["authorization"] = f"Bearer {gateway['auth_value']}"
-        elif gateway.get("auth_type") == "headers" and gateway.get("auth_value"):
+        auth_value = self.crypto.open_(gateway.get("auth_value"))
+        if gateway.get("auth_type") == "bearer" and auth_value:
+            headers["authorization"] = f"Bearer {auth_value}"
+        elif gateway.get("auth_type") == "headers" and auth_value:
             try:
-                headers.update(_json.loads(gateway["auth_value"]))
+                headers.update(_json.loads(auth_value))
             except Exception:
                 pass
-        elif gateway.get("auth_type") == "oauth" and gateway.get("auth_value"):
+        elif gateway.get("auth_type") == "oauth" and auth_value:
             # client-credentials upstream auth (reference: oauth_manager.py)
             from ..auth.oauth import provider_from_auth_value
 
-            cfg = gateway["auth_value"]
+            cfg = auth_value
             token_provider = provider_from_auth_value(_json.loads(cfg) if isinstance(cfg, str) else cfg)
         return HttpUpstreamClient(gateway["url"], headers=headers,
                                   timeout=self.settings.federation_timeout,
@@ -81,7 +86,7 @@ class GatewayService:
             transport=transport,
             description=description,
             auth_type=auth_type,
-            auth_value=auth_value,
+            auth_value=self.crypto.seal(auth_value),
             tags=tags or [],
             status="pending",
             owner_rank=owner_rank if owner_rank is not None else (hash(name) % self.world_size),

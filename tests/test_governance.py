@@ -154,3 +154,51 @@ def test_toolops_schema_synthesis(run, bare_engine):
         assert report["total"] == 2 and report["passed"] == 2
 
     run(go())
+
+
+def test_encryption_service_roundtrip_and_integrity():
+    """Credential sealing at rest (reference: EncryptedText db.py:277)."""
+    from mcp_context_forge_amd.auth.crypto import EncryptionService
+
+    svc = EncryptionService("topsecret", iterations=1000)
+    blob = svc.seal("Bearer abc123")
+    assert blob.startswith("enc1:") and "abc123" not in blob
+    assert svc.open_(blob) == "Bearer abc123"
+    assert svc.seal(None) is None
+    # legacy plaintext passthrough
+    assert svc.open_("plain-old-token") == "plain-old-token"
+    # integrity: tampering is detected
+    import pytest as _pt
+
+    tampered = blob[:-4] + ("AAAA" if not blob.endswith("AAAA") else "BBBB")
+    with _pt.raises(ValueError):
+        svc.open_(tampered)
+    # wrong key fails the MAC
+    other = EncryptionService("other", iterations=1000)
+    with _pt.raises(ValueError):
+        other.open_(blob)
+    # distinct nonces: same plaintext, different blobs
+    assert svc.seal("x") != svc.seal("x")
+
+
+def test_gateway_auth_value_sealed_at_rest(run):
+    from mcp_context_forge_amd.config import Settings
+    from mcp_context_forge_amd.engine import GatewayEngine
+    from mcp_context_forge_amd.services.upstream import InProcUpstream, make_fake_time_upstream
+
+    async def go():
+        e = GatewayEngine(Settings(database_url="sqlite://", federation_enabled=False,
+                                   auth_required=False))
+        await e.gateway_service.register_gateway(
+            name="sealed-up", url="inproc://x", auth_type="bearer",
+            auth_value="supersecret-token", client=make_fake_time_upstream())
+        row = e.registry.find("gateway", "sealed-up")
+        assert row["auth_value"].startswith("enc1:")
+        assert "supersecret-token" not in row["auth_value"]
+        # client construction decrypts transparently
+        client = e.gateway_service._make_client(row)
+        assert client.base_headers["authorization"] == "Bearer supersecret-token"
+        await client.aclose()
+        await e.shutdown()
+
+    run(go())
