@@ -84,6 +84,9 @@ class GatewayEngine:
         from .services.governance import TagService
 
         self.tags = TagService(self.registry)
+        from .services.governance import ServerClassificationService
+
+        self.classification = ServerClassificationService(self.registry)
         self.password_policy = PasswordPolicy()
         self.token_blocklist = TokenBlocklist()
         self.content_security = ContentSecurity()

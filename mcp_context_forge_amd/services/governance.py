@@ -190,3 +190,49 @@ class TagService:
             if rec["name"] == tag:
                 return rec["entities"]
         return []
+
+
+class ServerClassificationService:
+    """Heuristic server/tool classification (reference:
+    services/server_classification.py): bucket registered entities into
+    capability categories from names, descriptions and tags — feeds the
+    admin catalog view and policy defaults."""
+
+    CATEGORIES = {
+        "time": ("time", "clock", "timezone", "date"),
+        "data": ("sql", "database", "query", "table", "csv", "dataset"),
+        "devops": ("deploy", "kubernetes", "docker", "ci", "build", "git"),
+        "communication": ("mail", "slack", "message", "notify", "chat"),
+        "search": ("search", "lookup", "find", "index"),
+        "ai": ("llm", "embed", "summar", "classif", "generate"),
+        "files": ("file", "read", "write", "fs", "storage"),
+    }
+
+    def __init__(self, registry):
+        self.registry = registry
+
+    def classify_text(self, text: str) -> str:
+        t = text.lower()
+        best, hits = "other", 0
+        for cat, kws in self.CATEGORIES.items():
+            n = sum(1 for k in kws if k in t)
+            if n > hits:
+                best, hits = cat, n
+        return best
+
+    def classify_all(self) -> Dict[str, Any]:
+        out: Dict[str, Any] = {"servers": [], "gateways": [], "by_category": {}}
+        for kind, dest in (("server", "servers"), ("gateway", "gateways")):
+            for ent in self.registry.list(kind):
+                tools = [t for t in self.registry.list("tool")
+                         if t.get("gateway_id") == ent.get("id")] if kind == "gateway" else []
+                text = " ".join([ent.get("name", ""), ent.get("description") or "",
+                                 " ".join(ent.get("tags") or [])]
+                                + [t.get("name", "") + " " + (t.get("description") or "")
+                                   for t in tools])
+                cat = self.classify_text(text)
+                out[dest].append({"id": ent.get("id"), "name": ent.get("name"),
+                                  "category": cat, "tools": len(tools)})
+                out["by_category"].setdefault(cat, 0)
+                out["by_category"][cat] += 1
+        return out

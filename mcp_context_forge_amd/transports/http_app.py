@@ -853,6 +853,11 @@ def build_app(engine: GatewayEngine, auth: Optional[AuthService] = None) -> Fast
         n = await engine.siem.push_webhook(url, limit=int(body.get("limit", 1000)))
         return {"pushed": n}
 
+    @app.get("/admin/classification")
+    async def admin_classification(ctx: AuthContext = Depends(require("admin.read"))):
+        """Server/gateway capability classification (reference: server_classification)."""
+        return engine.classification.classify_all()
+
     @app.get("/admin/compliance/report")
     async def compliance_report(ctx: AuthContext = Depends(require("admin.read"))):
         return engine.compliance.report()

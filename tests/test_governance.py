@@ -202,3 +202,25 @@ def test_gateway_auth_value_sealed_at_rest(run):
         await e.shutdown()
 
     run(go())
+
+
+def test_server_classification(run):
+    from mcp_context_forge_amd.config import Settings
+    from mcp_context_forge_amd.engine import GatewayEngine
+    from mcp_context_forge_amd.services.upstream import make_fake_time_upstream
+
+    async def go():
+        e = GatewayEngine(Settings(database_url="sqlite://", federation_enabled=False,
+                                   auth_required=False))
+        await e.gateway_service.register_gateway(name="timezone-tools", url="inproc://t",
+                                                 client=make_fake_time_upstream())
+        e.registry.create("server", name="pg-helper", description="SQL query runner",
+                          tags=["database"])
+        out = e.classification.classify_all()
+        cats = {g["name"]: g["category"] for g in out["gateways"] + out["servers"]}
+        assert cats["timezone-tools"] == "time"
+        assert cats["pg-helper"] == "data"
+        assert out["by_category"]["time"] >= 1
+        await e.shutdown()
+
+    run(go())
