@@ -33,15 +33,27 @@ def render_admin_page(engine) -> str:
     sections.append(f"<h2>A2A agents ({len(ents['a2a_agent'])})</h2>" +
                     table(ents["a2a_agent"], ["name", "endpoint_url", "agent_type", "enabled"]))
     sections.append(f"<h2>Plugins</h2>" + table(plugins, ["name", "mode", "priority"]))
+    bindings = engine.registry.list("plugin_binding")
+    if bindings:
+        sections.append(f"<h2>Plugin bindings ({len(bindings)})</h2>" +
+                        table(bindings, ["tool_name", "plugin_name", "mode", "enabled"]))
+    tags = engine.tags.list_tags()
+    if tags:
+        sections.append(f"<h2>Tags ({len(tags)})</h2>" + table(tags, ["name", "count"]))
     if gpu:
         sections.append("<h2>GPU pipeline</h2><pre>" + html.escape(json.dumps(gpu, indent=2)) + "</pre>")
     metrics = engine.metrics.snapshot()
     sections.append("<h2>Metrics</h2><pre>" + html.escape(json.dumps(metrics, indent=2, default=str)) + "</pre>")
+    api_links = ["stats", "traces", "audit", "logs", "metrics/rollups", "siem/export",
+                 "compliance/report", "classification", "runtime", "performance",
+                 "support-bundle", "plugins"]
+    nav = " · ".join(f'<a href="/admin/{p}">{p}</a>' for p in api_links)
     return (
         "<!doctype html><html><head><title>MCP Context Forge AMD — Admin</title>"
         "<style>body{font-family:sans-serif;margin:2em}table{border-collapse:collapse;font-size:13px}"
         "th{background:#eee;text-align:left}</style></head><body>"
         "<h1>MCP Context Forge AMD</h1>"
         f"<p>MI355X-native gateway · sessions: {engine.sessions.count()}</p>"
+        f"<p>API: {nav}</p>"
         + "".join(sections) + "</body></html>"
     )
