@@ -709,7 +709,8 @@ class GpuPluginPipeline:
 
     # ------------------------------------------------------------------
     def _apply_rewrites(self, name: str, args: Any, do_norm: bool = True,
-                        do_regex: bool = True, do_pii: bool = True) -> Tuple[str, Any]:
+                        do_regex: bool = True, do_pii: bool = True,
+                        pii_bits: int = -1) -> Tuple[str, Any]:
         """Host rewrites in CPU-chain priority order: normalizer(15) →
         regex(20) → pii(30), using the plugins' own functions. The do_*
         flags come from the per-bank GPU masks: a bank that did not fire on
@@ -724,7 +725,10 @@ class GpuPluginPipeline:
             found: List[str] = []
 
             def fn(s: str) -> str:
-                masked, f = self.pii.mask_text(s)
+                if pii_bits >= 0:
+                    masked, f = self.pii.mask_text_subset(s, pii_bits)
+                else:
+                    masked, f = self.pii.mask_text(s)
                 found.extend(f)
                 return masked
 
@@ -753,7 +757,8 @@ class GpuPluginPipeline:
                 mt.name, args,
                 do_norm=bool(norm_m[j]) if norm_m is not None else True,
                 do_regex=bool(regex_m[j]) if regex_m is not None else True,
-                do_pii=bool(pii_m[j]) if pii_m is not None else True)
+                do_pii=bool(pii_m[j]) if pii_m is not None else True,
+                pii_bits=int(pii_m[j]) if pii_m is not None else -1)
             if status == "__block__":
                 responses[r] = self._splice_error(idb, jsonrpc.POLICY_DENIED, payload)
                 self.blocked += 1

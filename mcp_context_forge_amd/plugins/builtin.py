@@ -176,6 +176,19 @@ class PIIFilterPlugin(Plugin):
                 text = rx.sub(f"[{name.upper()}_REDACTED]", text)
         return text, found
 
+    def mask_text_subset(self, text: str, pattern_bits: int) -> Tuple[str, List[str]]:
+        """Masked-pattern fast path: bit i of `pattern_bits` is the GPU scan
+        bank's accept bit for self.active[i] (same compile order) — only the
+        patterns the DFA flagged need their Python regex run."""
+        found: List[str] = []
+        for i, (name, _d, rx) in enumerate(self.active):
+            if not (pattern_bits >> i) & 1:
+                continue
+            if rx.search(text):
+                found.append(name)
+                text = rx.sub(f"[{name.upper()}_REDACTED]", text)
+        return text, found
+
     async def _apply(self, ctx: PluginContext) -> PluginResult:
         found_all: List[str] = []
 

@@ -317,3 +317,17 @@ def test_toon_tabular_quoted_commas_and_escapes():
     obj = {"rows": [{"a": 'x,y', "b": 1}, {"a": 'q"z', "b": 2}]}
     enc = toon_codec.encode(obj)
     assert toon_codec.decode(enc) == obj
+
+
+def test_pii_mask_subset_matches_full():
+    p = PIIFilterPlugin({})
+    text = "ssn 123-45-6789 mail a@b.co ip 10.0.0.1"
+    full, found_full = p.mask_text(text)
+    # all-bits subset == full
+    sub, found_sub = p.mask_text_subset(text, (1 << len(p.active)) - 1)
+    assert sub == full and found_sub == found_full
+    # only the ssn bit: email/ip untouched
+    ssn_bit = 1 << [n for n, _, _ in ((a, b, c) for a, b, c in [(x[0], 0, 0) for x in p.active])].index("ssn")
+    only_ssn, found = p.mask_text_subset(text, ssn_bit)
+    assert "[SSN_REDACTED]" in only_ssn and "a@b.co" in only_ssn
+    assert found == ["ssn"]
