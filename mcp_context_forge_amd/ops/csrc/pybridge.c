@@ -136,6 +136,57 @@ static PyObject *concat_with_offsets(PyObject *self, PyObject *args) {
     return Py_BuildValue("NN", blob, offs_obj);
 }
 
+/* pack_frame(ids_int64_buffer, list_of_bytes_or_None) -> bytes
+ * Owner->worker edge frame: [u32 payload_bytes][u32 n] payload
+ * n x {[u64 id][u32 len][bytes]}; None -> len = 0xFFFFFFFF, no body. */
+static PyObject *pack_frame(PyObject *self, PyObject *args) {
+    Py_buffer ids;
+    PyObject *outs;
+    if (!PyArg_ParseTuple(args, "y*O!", &ids, &PyList_Type, &outs))
+        return NULL;
+    Py_ssize_t n = (Py_ssize_t)(ids.len / (Py_ssize_t)sizeof(int64_t));
+    if (PyList_GET_SIZE(outs) != n) {
+        PyBuffer_Release(&ids);
+        PyErr_SetString(PyExc_ValueError, "ids/outs length mismatch");
+        return NULL;
+    }
+    const int64_t *idp = (const int64_t *)ids.buf;
+    int64_t payload = 0;
+    for (Py_ssize_t i = 0; i < n; i++) {
+        PyObject *o = PyList_GET_ITEM(outs, i);
+        payload += 12;
+        if (o != Py_None) {
+            if (!PyBytes_Check(o)) {
+                PyBuffer_Release(&ids);
+                PyErr_SetString(PyExc_TypeError, "outs must be bytes or None");
+                return NULL;
+            }
+            payload += PyBytes_GET_SIZE(o);
+        }
+    }
+    PyObject *frame = PyBytes_FromStringAndSize(NULL, 8 + payload);
+    if (!frame) { PyBuffer_Release(&ids); return NULL; }
+    unsigned char *w = (unsigned char *)PyBytes_AS_STRING(frame);
+    uint32_t pb = (uint32_t)payload, nn = (uint32_t)n;
+    memcpy(w, &pb, 4); memcpy(w + 4, &nn, 4);
+    w += 8;
+    for (Py_ssize_t i = 0; i < n; i++) {
+        PyObject *o = PyList_GET_ITEM(outs, i);
+        uint64_t id = (uint64_t)idp[i];
+        memcpy(w, &id, 8); w += 8;
+        if (o == Py_None) {
+            uint32_t nr = 0xFFFFFFFFu;
+            memcpy(w, &nr, 4); w += 4;
+        } else {
+            uint32_t ln = (uint32_t)PyBytes_GET_SIZE(o);
+            memcpy(w, &ln, 4); w += 4;
+            memcpy(w, PyBytes_AS_STRING(o), ln); w += ln;
+        }
+    }
+    PyBuffer_Release(&ids);
+    return frame;
+}
+
 static PyMethodDef Methods[] = {
     {"scatter_slices", scatter_slices, METH_VARARGS,
      "scatter arena spans into a responses list as bytes"},
@@ -143,6 +194,8 @@ static PyMethodDef Methods[] = {
      "extract arena spans into a new list of bytes/None"},
     {"concat_with_offsets", concat_with_offsets, METH_VARARGS,
      "join a list of bytes into one blob + int64 offsets"},
+    {"pack_frame", pack_frame, METH_VARARGS,
+     "pack an owner->worker edge response frame"},
     {NULL, NULL, 0, NULL},
 };
 

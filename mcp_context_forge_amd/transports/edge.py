@@ -47,8 +47,11 @@ class GpuOwnerServer:
     (through the collector when attached, so cross-worker coalescing works)."""
 
     def __init__(self, engine, collector=None, path: str = "/tmp/forge-edge.sock"):
+        from ..ops.pybridge import get as _pb_get
+
         self.engine = engine
         self.collector = collector
+        self._pb = _pb_get()   # C frame packer (ops/csrc/pybridge.c)
         self.path = path
         self._server: Optional[asyncio.AbstractServer] = None
         self.frames = 0
@@ -96,20 +99,12 @@ class GpuOwnerServer:
 
                 async def run_frame(ids=ids, bodies=bodies):
                     outs = await self._process_frame(bodies)
-                    parts = [b"", _U32.pack(len(ids))]
-                    total = 0
-                    for req_id, out in zip(ids, outs):
-                        parts.append(_U64.pack(req_id))
-                        if out is None:
-                            parts.append(_U32.pack(NO_RESPONSE))
-                            total += 12
-                        else:
-                            parts.append(_U32.pack(len(out)))
-                            parts.append(bytes(out))
-                            total += 12 + len(out)
-                    parts[0] = _U32.pack(total)
+                    outs = [o if (o is None or isinstance(o, bytes)) else bytes(o) for o in outs]
+                    import numpy as _np
+
+                    frame = self._pb.pack_frame(_np.asarray(ids, dtype=_np.int64), outs)
                     async with write_lock:
-                        writer.write(b"".join(parts))
+                        writer.write(frame)
                         await writer.drain()
 
                 asyncio.ensure_future(run_frame())
