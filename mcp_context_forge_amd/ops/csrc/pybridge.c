@@ -187,6 +187,39 @@ static PyObject *pack_frame(PyObject *self, PyObject *args) {
     return frame;
 }
 
+/* unpack_frame(payload, n) -> (bytes ids_int64, list_of_bytes)
+ * Worker->owner edge frame payload: n x {[u64 id][u32 len][bytes]}. */
+static PyObject *unpack_frame(PyObject *self, PyObject *args) {
+    Py_buffer pay;
+    Py_ssize_t n;
+    if (!PyArg_ParseTuple(args, "y*n", &pay, &n))
+        return NULL;
+    const unsigned char *p = (const unsigned char *)pay.buf;
+    Py_ssize_t remain = pay.len;
+    PyObject *ids = PyBytes_FromStringAndSize(NULL, n * (Py_ssize_t)sizeof(int64_t));
+    PyObject *lst = PyList_New(n);
+    if (!ids || !lst) goto fail;
+    int64_t *idp = (int64_t *)PyBytes_AS_STRING(ids);
+    for (Py_ssize_t i = 0; i < n; i++) {
+        if (remain < 12) { PyErr_SetString(PyExc_ValueError, "truncated frame"); goto fail; }
+        uint64_t id; uint32_t ln;
+        memcpy(&id, p, 8); memcpy(&ln, p + 8, 4);
+        p += 12; remain -= 12;
+        if ((Py_ssize_t)ln > remain) { PyErr_SetString(PyExc_ValueError, "truncated body"); goto fail; }
+        idp[i] = (int64_t)id;
+        PyObject *b = PyBytes_FromStringAndSize((const char *)p, (Py_ssize_t)ln);
+        if (!b) goto fail;
+        PyList_SET_ITEM(lst, i, b);
+        p += ln; remain -= ln;
+    }
+    PyBuffer_Release(&pay);
+    return Py_BuildValue("NN", ids, lst);
+fail:
+    Py_XDECREF(ids); Py_XDECREF(lst);
+    PyBuffer_Release(&pay);
+    return NULL;
+}
+
 static PyMethodDef Methods[] = {
     {"scatter_slices", scatter_slices, METH_VARARGS,
      "scatter arena spans into a responses list as bytes"},
@@ -196,6 +229,8 @@ static PyMethodDef Methods[] = {
      "join a list of bytes into one blob + int64 offsets"},
     {"pack_frame", pack_frame, METH_VARARGS,
      "pack an owner->worker edge response frame"},
+    {"unpack_frame", unpack_frame, METH_VARARGS,
+     "unpack a worker->owner edge request frame payload"},
     {NULL, NULL, 0, NULL},
 };
 

@@ -86,23 +86,13 @@ class GpuOwnerServer:
                 except (asyncio.IncompleteReadError, ConnectionResetError):
                     return
                 self.frames += 1
-                ids: List[int] = []
-                bodies: List[bytes] = []
-                off = 0
-                for _ in range(n):
-                    req_id, ln = struct.unpack_from("<QI", payload, off)
-                    off += 12
-                    bodies.append(payload[off:off + ln])
-                    off += ln
-                    ids.append(req_id)
+                ids_raw, bodies = self._pb.unpack_frame(payload, n)
                 self.requests += n
 
-                async def run_frame(ids=ids, bodies=bodies):
+                async def run_frame(ids_raw=ids_raw, bodies=bodies):
                     outs = await self._process_frame(bodies)
                     outs = [o if (o is None or isinstance(o, bytes)) else bytes(o) for o in outs]
-                    import numpy as _np
-
-                    frame = self._pb.pack_frame(_np.asarray(ids, dtype=_np.int64), outs)
+                    frame = self._pb.pack_frame(ids_raw, outs)
                     async with write_lock:
                         writer.write(frame)
                         await writer.drain()
