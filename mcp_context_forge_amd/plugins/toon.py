@@ -14,8 +14,8 @@ import json
 import re
 from typing import Any, List, Optional, Tuple
 
-_SIMPLE = re.compile(r"^[A-Za-z0-9_.@+\-]+$")
-_NUMERIC = re.compile(r"^-?\d+(\.\d+)?([eE][+-]?\d+)?$")
+_SIMPLE = re.compile(r"^[A-Za-z0-9_.@+\-]+\Z")  # \Z: $ would match before a trailing newline
+_NUMERIC = re.compile(r"^-?\d+(\.\d+)?([eE][+-]?\d+)?\Z")
 
 
 def _scalar(v: Any) -> str:

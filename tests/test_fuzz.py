@@ -137,3 +137,60 @@ def test_toon_fuzz_roundtrip():
     for _ in range(200):
         v = rand_val()
         assert toon.decode(toon.encode(v)) == v, v
+
+
+# -- property-based fuzz for round-1 additions (hypothesis) ----------------
+
+from hypothesis import given, settings as hsettings, strategies as st
+
+_json_scalars = st.one_of(st.none(), st.booleans(), st.integers(-10**9, 10**9),
+                          st.floats(allow_nan=False, allow_infinity=False, width=32),
+                          st.text(max_size=40))
+_json_values = st.recursive(
+    _json_scalars,
+    lambda children: st.one_of(
+        st.lists(children, max_size=5),
+        st.dictionaries(st.text(min_size=1, max_size=12), children, max_size=5)),
+    max_leaves=20)
+
+
+@hsettings(max_examples=150, deadline=None)
+@given(_json_values)
+def test_toon_roundtrip_property(value):
+    """encode→decode is identity for JSON-shaped values (TOON's contract)."""
+    from mcp_context_forge_amd.plugins import toon
+
+    enc = toon.encode(value)
+    out = toon.decode(enc)
+    if isinstance(value, float):
+        assert out == __import__("json").loads(__import__("json").dumps(value))
+    else:
+        assert out == value, (value, enc, out)
+
+
+@hsettings(max_examples=200, deadline=None)
+@given(st.text(max_size=30))
+def test_jsonpath_never_crashes_unexpectedly(expr):
+    """Arbitrary expressions either evaluate or raise JSONPathError —
+    never IndexError/KeyError/etc. (jsonpath_filter maps errors to None)."""
+    from mcp_context_forge_amd.utils.jsonpath import JSONPathError, evaluate, jsonpath_filter
+
+    doc = {"a": [1, {"b": "x"}], "c": {"d": None}}
+    try:
+        evaluate(doc, expr)
+    except JSONPathError:
+        pass
+    jsonpath_filter(doc, expr)  # must never raise
+
+
+@hsettings(max_examples=100, deadline=None)
+@given(st.binary(max_size=200))
+def test_envelope_oracle_never_crashes(raw):
+    """The Python JSON-RPC parser (CPU oracle of envelope.cpp) returns a
+    clean protocol error for arbitrary bytes."""
+    from mcp_context_forge_amd.protocol import jsonrpc
+
+    try:
+        jsonrpc.parse_request_bytes(raw)
+    except jsonrpc.JSONRPCError:
+        pass
