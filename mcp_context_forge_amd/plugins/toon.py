@@ -43,6 +43,10 @@ def _uniform_object_array(arr: List[Any]) -> Optional[List[str]]:
     if not arr or not all(isinstance(x, dict) and x for x in arr):
         return None
     keys = sorted(arr[0].keys())
+    # field names ride inside the `{a,b}` header — anything that needs
+    # quoting (commas, quotes, newlines) forces the block-item form
+    if not all(isinstance(k, str) and _SIMPLE.match(k) for k in keys):
+        return None
     for x in arr:
         if sorted(x.keys()) != keys:
             return None
