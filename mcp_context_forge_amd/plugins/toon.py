@@ -123,7 +123,10 @@ def _parse_scalar(tok: str) -> Any:
     if tok == "false":
         return False
     if tok.startswith('"'):
-        return json.loads(tok)
+        try:
+            return json.loads(tok)
+        except Exception:
+            return tok  # malformed quote: treat as raw text (defensive)
     if _NUMERIC.match(tok):
         return json.loads(tok)
     return tok
@@ -155,9 +158,17 @@ _HEAD = re.compile(r"^(?P<key>\"(?:[^\"\\]|\\.)*\"|[^:\[{]+)?(?:\[(?P<n>\d+)\](?
 
 def decode(text: str) -> Any:
     lines = [ln for ln in text.splitlines() if ln.strip()]
-    # bare top-level scalar
-    if len(lines) == 1 and not _HEAD.match(lines[0].strip()):
-        return _parse_scalar(lines[0].strip())
+    # bare top-level scalar; a whole-line JSON string wins over the k:v
+    # reading (the encoder quotes scalars like ":" that contain separators)
+    if len(lines) == 1:
+        t = lines[0].strip()
+        if t.startswith('"'):
+            try:
+                return json.loads(t)
+            except Exception:
+                pass
+        if not _HEAD.match(t):
+            return _parse_scalar(t)
     if len(lines) == 1:
         m = _HEAD.match(lines[0].strip())
         if m and m.group("n") is None and m.group("key") is not None and ":" not in lines[0]:

@@ -154,7 +154,7 @@ _json_values = st.recursive(
     max_leaves=20)
 
 
-@hsettings(max_examples=150, deadline=None)
+@hsettings(max_examples=150, deadline=None, derandomize=True)
 @given(_json_values)
 def test_toon_roundtrip_property(value):
     """encode→decode is identity for JSON-shaped values (TOON's contract)."""
@@ -168,7 +168,7 @@ def test_toon_roundtrip_property(value):
         assert out == value, (value, enc, out)
 
 
-@hsettings(max_examples=200, deadline=None)
+@hsettings(max_examples=200, deadline=None, derandomize=True)
 @given(st.text(max_size=30))
 def test_jsonpath_never_crashes_unexpectedly(expr):
     """Arbitrary expressions either evaluate or raise JSONPathError —
@@ -183,7 +183,7 @@ def test_jsonpath_never_crashes_unexpectedly(expr):
     jsonpath_filter(doc, expr)  # must never raise
 
 
-@hsettings(max_examples=100, deadline=None)
+@hsettings(max_examples=100, deadline=None, derandomize=True)
 @given(st.binary(max_size=200))
 def test_envelope_oracle_never_crashes(raw):
     """The Python JSON-RPC parser (CPU oracle of envelope.cpp) returns a
