@@ -329,16 +329,19 @@ def build_app(engine: GatewayEngine, auth: Optional[AuthService] = None) -> Fast
             sess = engine.sessions.create(transport="streamablehttp", server_id=server_id, user=ctx.user)
             headers["mcp-session-id"] = sess.session_id
         elif sess is None and session_id:
-            # multi-rank session affinity: route to the owner rank over the bus
-            # (reference: session_affinity.forward_request_to_owner :747)
-            if engine.bus is not None and engine.forward_rpc is not None:
-                owner = engine.bus.owner_of(session_id)
-                if owner != engine.rank:
-                    out = await engine.forward_rpc(owner, raw, user=ctx.user, server_id=server_id)
-                    if out is None:
-                        return Response(status_code=202)
-                    return Response(content=out, media_type="application/json")
-            return JSONResponse({"detail": "session not found"}, status_code=404)
+            # restart continuity: a DB-persisted session resumes in place
+            # (reference: database session backend; client replays from
+            # Last-Event-ID on its GET stream)
+            sess = engine.sessions.resume(session_id)
+            if sess is None:
+                if engine.bus is not None and engine.forward_rpc is not None:
+                    owner = engine.bus.owner_of(session_id)
+                    if owner != engine.rank:
+                        out = await engine.forward_rpc(owner, raw, user=ctx.user, server_id=server_id)
+                        if out is None:
+                            return Response(status_code=202)
+                        return Response(content=out, media_type="application/json")
+                return JSONResponse({"detail": "session not found"}, status_code=404)
         resp = await engine.handle_rpc(req, user=ctx.user, server_id=server_id or ctx.server_id, session=sess)
         if resp is None:
             return Response(status_code=202, headers=headers)

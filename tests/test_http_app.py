@@ -525,3 +525,46 @@ def test_session_persistence_across_restart(tmp_path, run):
         await e2.shutdown()
 
     run(go())
+
+
+def test_mcp_resume_after_restart(tmp_path, run):
+    """POST /mcp with a persisted session id after an engine restart resumes
+    in place instead of 404 (database session backend)."""
+    from contextlib import asynccontextmanager
+
+    from mcp_context_forge_amd.config import Settings
+    from mcp_context_forge_amd.engine import GatewayEngine
+
+    async def go():
+        url = f"sqlite:///{tmp_path}/mcp.db"
+
+        def mk():
+            e = GatewayEngine(Settings(database_url=url, federation_enabled=False,
+                                       auth_required=True, session_persistence=True))
+            app = build_app(e)
+            return e, app
+
+        e1, app1 = mk()
+        async with app1.router.lifespan_context(app1):
+            async with httpx.AsyncClient(transport=httpx.ASGITransport(app=app1),
+                                         base_url="http://gw") as c:
+                r = await c.post("/mcp", headers=ADMIN,
+                                 json={"jsonrpc": "2.0", "id": 1, "method": "initialize",
+                                       "params": {"protocolVersion": "2025-11-25"}})
+                sid = r.headers["mcp-session-id"]
+        await e1.shutdown()
+
+        e2, app2 = mk()
+        async with app2.router.lifespan_context(app2):
+            async with httpx.AsyncClient(transport=httpx.ASGITransport(app=app2),
+                                         base_url="http://gw") as c:
+                r = await c.post("/mcp", headers={**ADMIN, "mcp-session-id": sid},
+                                 json={"jsonrpc": "2.0", "id": 2, "method": "ping"})
+                assert r.status_code == 200 and r.json()["result"] == {}
+                # unknown ids still 404
+                r = await c.post("/mcp", headers={**ADMIN, "mcp-session-id": "nope"},
+                                 json={"jsonrpc": "2.0", "id": 3, "method": "ping"})
+                assert r.status_code == 404
+        await e2.shutdown()
+
+    run(go())
