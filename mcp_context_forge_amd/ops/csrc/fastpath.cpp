@@ -196,14 +196,22 @@ int64_t forge_decide(
     int32_t* state, int32_t* native_kind_out, int8_t* reason_out,
     uint8_t* arena, int64_t arena_cap, int64_t* resp_beg, int64_t* resp_end)
 {
-    Arena a;
-    a.buf.reserve((size_t)m * 48);
-    a.beg = resp_beg;
-    a.end = resp_end;
     auto* store = (SlotStore*)slot_store;
     auto* ec = (ExactCache*)exact_cache;
+    // rows are independent (stores are mutex/shard-guarded): chunk across
+    // threads with per-chunk arenas, stitch + rebase at the end — same
+    // pattern as forge_finalize
+    int nthreads = m >= 2048 ? 8 : (m >= 256 ? 4 : 1);
+    int chunk_sz = (m + nthreads - 1) / nthreads;
+    std::vector<std::string> bufs((size_t)nthreads);
 
-    for (int i = 0; i < m; ++i) {
+    auto run_rows = [&](int t) {
+    Arena a;
+    a.buf.reserve((size_t)chunk_sz * 48);
+    a.beg = resp_beg;
+    a.end = resp_end;
+    int i0 = t * chunk_sz, i1 = i0 + chunk_sz < m ? i0 + chunk_sz : m;
+    for (int i = i0; i < i1; ++i) {
         resp_beg[i] = -1;
         resp_end[i] = -1;
         native_kind_out[i] = -1;
@@ -331,9 +339,27 @@ int64_t forge_decide(
             state[i] = ST_DISPATCH_PY;
         }
     }
-    if ((int64_t)a.buf.size() > arena_cap) return -(int64_t)a.buf.size();
-    memcpy(arena, a.buf.data(), a.buf.size());
-    return (int64_t)a.buf.size();
+    bufs[(size_t)t] = std::move(a.buf);
+    };
+    if (nthreads == 1) {
+        run_rows(0);
+    } else {
+        std::vector<std::thread> threads;
+        for (int t = 0; t < nthreads; ++t) threads.emplace_back(run_rows, t);
+        for (auto& th : threads) th.join();
+    }
+    int64_t total = 0;
+    std::vector<int64_t> base((size_t)nthreads, 0);
+    for (int t = 0; t < nthreads; ++t) { base[(size_t)t] = total; total += (int64_t)bufs[(size_t)t].size(); }
+    if (total > arena_cap) return -total;
+    for (int t = 0; t < nthreads; ++t) {
+        if (!bufs[(size_t)t].empty())
+            memcpy(arena + base[(size_t)t], bufs[(size_t)t].data(), bufs[(size_t)t].size());
+        int i0 = t * chunk_sz, i1 = i0 + chunk_sz < m ? i0 + chunk_sz : m;
+        for (int i = i0; i < i1; ++i)
+            if (resp_beg[i] >= 0) { resp_beg[i] += base[(size_t)t]; resp_end[i] += base[(size_t)t]; }
+    }
+    return total;
 }
 
 // Finalize dispatched rows: splice responses, detect isError, exact-cache
