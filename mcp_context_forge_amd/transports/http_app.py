@@ -527,6 +527,29 @@ def build_app(engine: GatewayEngine, auth: Optional[AuthService] = None) -> Fast
     for plural, kind in ENTITY_ROUTES.items():
         _register_crud(plural, kind)
 
+    @app.get("/search")
+    async def global_search(q: str, kinds: str = "", limit: int = 50,
+                            ctx: AuthContext = Depends(require("tools.read"))):
+        """Global entity search (reference: routers/search.py): substring
+        match over name/description/tags across registry kinds."""
+        klist = [k.strip() for k in kinds.split(",") if k.strip()] or             ["tool", "gateway", "server", "resource", "prompt", "a2a_agent"]
+        ql = q.lower()
+        out = []
+        for kind in klist:
+            if kind not in ("tool", "gateway", "server", "resource", "prompt", "a2a_agent"):
+                continue
+            for ent in engine.registry.list(kind):
+                hay = " ".join([str(ent.get("name") or ent.get("uri") or ""),
+                                str(ent.get("description") or ""),
+                                " ".join(ent.get("tags") or [])]).lower()
+                if ql in hay:
+                    out.append({"kind": kind, "id": ent.get("id"),
+                                "name": ent.get("name") or ent.get("uri"),
+                                "description": (ent.get("description") or "")[:200]})
+                    if len(out) >= limit:
+                        return out
+        return out
+
     # -- per-tool plugin bindings (reference: routers/tool_plugin_bindings.py) --
     @app.get("/tools/{tool_name}/plugin-bindings")
     async def list_plugin_bindings(tool_name: str, ctx: AuthContext = Depends(require("tools.read"))):
@@ -553,6 +576,13 @@ def build_app(engine: GatewayEngine, auth: Optional[AuthService] = None) -> Fast
         except NotFoundError as exc:
             raise HTTPException(404, str(exc)) from exc
         return Response(status_code=204)
+
+    # agent-hook bindings share the same keyed map (ctx.name = agent name)
+    # (reference: routers/a2a_agent_plugin_bindings.py)
+    app.add_api_route("/a2a/{tool_name}/plugin-bindings", list_plugin_bindings, methods=["GET"])
+    app.add_api_route("/a2a/{tool_name}/plugin-bindings/{plugin_name}", set_plugin_binding, methods=["PUT"])
+    app.add_api_route("/a2a/{tool_name}/plugin-bindings/{plugin_name}", delete_plugin_binding,
+                      methods=["DELETE"], status_code=204)
 
     @app.post("/gateways/{gateway_id}/refresh")
     async def refresh_gateway(gateway_id: str, ctx: AuthContext = Depends(require("gateways.update"))):

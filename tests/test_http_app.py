@@ -568,3 +568,32 @@ def test_mcp_resume_after_restart(tmp_path, run):
         await e2.shutdown()
 
     run(go())
+
+
+def test_global_search(client_engine, run):
+    """Global search (reference: routers/search.py)."""
+    client_factory, engine, app = client_engine
+
+    async def go():
+        engine.registry.create("tool", name="weather-lookup", original_name="weather-lookup",
+                               integration_type="LOCAL", description="Forecast fetcher",
+                               tags=["meteo"])
+        engine.registry.create("server", name="ops-box", description="weather dashboards")
+        async with client_factory() as c:
+            r = await c.get("/search", headers=ADMIN, params={"q": "weather"})
+            assert r.status_code == 200
+            kinds = {(x["kind"], x["name"]) for x in r.json()}
+            assert ("tool", "weather-lookup") in kinds and ("server", "ops-box") in kinds
+            r = await c.get("/search", headers=ADMIN, params={"q": "meteo", "kinds": "tool"})
+            assert [x["name"] for x in r.json()] == ["weather-lookup"]
+            r = await c.get("/search", headers=ADMIN, params={"q": "zzz-nothing"})
+            assert r.json() == []
+            # a2a binding alias routes share the tool binding map
+            r = await c.put("/a2a/helper-agent/plugin-bindings/deny_filter", headers=ADMIN,
+                            json={"mode": "permissive"})
+            assert r.status_code == 200
+            assert engine.plugins.bindings["helper-agent"]["deny_filter"]["mode"] == "permissive"
+            r = await c.delete("/a2a/helper-agent/plugin-bindings/deny_filter", headers=ADMIN)
+            assert r.status_code == 204
+
+    run(go())
