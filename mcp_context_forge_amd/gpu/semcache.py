@@ -78,7 +78,7 @@ class GpuSemanticCache:
             # sketches returned to the caller, which passes them back into
             # insert_features (no shared slot — batches may be in flight
             # concurrently)
-            sk = hip.gemm_bt(feats_bf16, self.proj_t).to(torch.bfloat16)
+            sk = hip.gemm_bt(feats_bf16, self.proj_t, out_bf16=True)
         if self.size == 0:
             return best_val, best_idx, sk
         active = min(self.size, self.capacity)
@@ -155,7 +155,7 @@ class GpuSemanticCache:
         if self.sketch_dim:
             sk = sketch
             if sk is None or sk.shape[0] != feats_bf16.shape[0]:
-                sk = hip.gemm_bt(feats_bf16, self.proj_t).to(torch.bfloat16)
+                sk = hip.gemm_bt(feats_bf16, self.proj_t, out_bf16=True)
             hip.rows_gather_scatter_bf16(sk, rows_t, slot_t, self.keys_sk, self.valid)
         now = time.monotonic()
         self.tool_hashes[slots] = tool_hashes

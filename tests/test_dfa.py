@@ -159,3 +159,31 @@ def test_class_trailing_dash_and_optional():
     assert dfa.scan_reference(t2, b"ac")
     assert dfa.scan_reference(t2, b"abc")
     assert not dfa.scan_reference(t2, b"abbc")
+
+
+def test_class_edge_cases_and_dot():
+    # leading ']' is a literal member; escapes work inside classes;
+    # unterminated classes are rejected; '.' matches all but newline
+    t = dfa.compile_patterns(["[]]x"], case_insensitive=False)
+    assert dfa.scan_reference(t, b"]x")
+    t2 = dfa.compile_patterns([r"[\d]q"], case_insensitive=False)
+    assert dfa.scan_reference(t2, b"5q") and not dfa.scan_reference(t2, b"aq")
+    with pytest.raises(ValueError):
+        dfa.compile_patterns(["[ab"], case_insensitive=False)
+    t3 = dfa.compile_patterns(["a.b"], case_insensitive=False)
+    assert dfa.scan_reference(t3, b"axb") and not dfa.scan_reference(t3, b"a\nb")
+
+
+def test_plus_quantifier_unbounded():
+    t = dfa.compile_patterns(["ab+c"], case_insensitive=False)
+    assert dfa.scan_reference(t, b"abc")
+    assert dfa.scan_reference(t, b"ab" + b"b" * 40 + b"c")  # truly unbounded loop
+    assert not dfa.scan_reference(t, b"ac")
+
+
+def test_compile_literals_case_insensitive_by_default():
+    # deny words match regardless of case unless told otherwise
+    t = dfa.compile_literals(["Word"])
+    assert dfa.scan_reference(t, b"word") and dfa.scan_reference(t, b"WORD")
+    t2 = dfa.compile_literals(["Word"], case_insensitive=False)
+    assert not dfa.scan_reference(t2, b"word")

@@ -93,3 +93,10 @@ def test_filter_numeric_boundaries():
     assert evaluate(docs, "$[?(@.p <= 10)].p") == [8.95, 10]
     assert evaluate(docs, "$[?(@.p > 10)].p") == [12]
     assert evaluate(docs, "$[?(@.p >= 10)].p") == [10, 12]
+
+
+def test_truncated_expressions_raise():
+    for expr in ("$.", "$..", "$['a'", "$[?(@.x"):
+        with pytest.raises(JSONPathError):
+            evaluate(DOC, expr)
+        assert jsonpath_filter(DOC, expr) is None
