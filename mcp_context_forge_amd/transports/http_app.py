@@ -967,6 +967,10 @@ def build_app(engine: GatewayEngine, auth: Optional[AuthService] = None) -> Fast
             col.window_s = max(0, int(body["window_us"])) / 1e6
         if "max_batch" in body and col is not None:
             col.max_batch = max(1, int(body["max_batch"]))
+        if "timing" in body and engine.gpu_pipeline is not None:
+            engine.gpu_pipeline.timing_enabled = bool(body["timing"])
+            if not body["timing"]:
+                engine.gpu_pipeline.timing.clear()
         if body.get("mode") == "cpu" and engine.gpu_pipeline is not None:
             app.state.collector = None     # per-request CPU chain
         elif body.get("mode") == "gpu-batched" and engine.gpu_pipeline is not None and col is None:
