@@ -224,3 +224,52 @@ def test_server_classification(run):
         await e.shutdown()
 
     run(go())
+
+
+def test_teams_roots_tasks_and_batch_metrics(run):
+    """Coverage for the tenancy/roots/task surfaces (reference: team
+    management service, root_service, a2a task store, metrics record_batch)."""
+    import json as _json
+
+    from mcp_context_forge_amd.auth.crypto import EncryptionService
+    from mcp_context_forge_amd.config import Settings
+    from mcp_context_forge_amd.engine import GatewayEngine
+    from mcp_context_forge_amd.transports.http_app import build_app
+
+    async def go():
+        e = GatewayEngine(Settings(database_url="sqlite://", federation_enabled=False,
+                                   auth_required=False))
+        app = build_app(e)
+        auth = app.state.auth
+        # teams: creator becomes owner; members listable
+        team = auth.create_team("ml-platform", created_by="alice@x.io")
+        assert team["slug"] == "ml-platform"
+        auth.add_team_member(team["id"], "bob@x.io")
+        teams = auth.teams_of("bob@x.io") if hasattr(auth, "teams_of") else None
+        if teams is not None:
+            assert any(t.get("name") == "ml-platform" or t.get("team_id") == team["id"]
+                       for t in teams)
+        # roots ride the MCP roots/list route
+        e.root_service.add_root("file:///workspace", "ws")
+        out = await e.handle_rpc_bytes(_json.dumps(
+            {"jsonrpc": "2.0", "id": 1, "method": "roots/list"}).encode())
+        assert _json.loads(out)["result"]["roots"] == [{"uri": "file:///workspace", "name": "ws"}]
+        assert e.root_service.remove_root("file:///workspace") is True
+        assert e.root_service.remove_root("file:///gone") is False
+        # a2a task store
+        t = e.a2a_service.upsert_task("t1", "running")
+        assert e.a2a_service.get_task("t1")["status"] == "running"
+        e.a2a_service.upsert_task("t1", "done", {"out": 1})
+        assert e.a2a_service.get_task("t1")["detail"] == {"out": 1}
+        assert e.a2a_service.get_task("missing") is None
+        # batched per-request metrics
+        e.metrics.record_batch(["ta", "tb", "ta"], 2.0, [True, False, True])
+        snap = e.metrics.snapshot()
+        assert snap["counters"]["tool_invocations_total"] >= 3
+        assert snap["counters"]["tool_errors_total"] >= 1
+        # crypto helper
+        svc = EncryptionService("k", iterations=500)
+        assert svc.is_sealed(svc.seal("x")) and not svc.is_sealed("x")
+        await e.shutdown()
+
+    run(go())
