@@ -91,6 +91,22 @@ class ToolService:
         return tool
 
     # -- dispatch (the upstream leg; shared with the GPU pipeline) --------------
+    def _apply_jsonpath(self, tool: Dict[str, Any], result: Dict[str, Any]) -> Dict[str, Any]:
+        """Tool-level result filter (reference: main.py:1281 jsonpath_modifier
+        applied via the tool row's jsonpath_filter)."""
+        jp = tool.get("jsonpath_filter")
+        if not jp or not isinstance(result, dict) or result.get("isError"):
+            return result
+        payload = result.get("structuredContent")
+        if payload is None:
+            return result
+        filtered = jsonpath_filter(payload, jp)
+        return {**result,
+                "structuredContent": filtered,
+                "content": [{"type": "text",
+                             "text": filtered if isinstance(filtered, str)
+                             else json.dumps(filtered, default=str)}]}
+
     async def dispatch(self, tool: Dict[str, Any], arguments: Dict[str, Any],
                        headers: Optional[Dict[str, str]] = None) -> Dict[str, Any]:
         itype = tool.get("integration_type", "MCP")
@@ -100,20 +116,21 @@ class ToolService:
                 raise ToolInvocationError(f"no local handler for {tool['name']}")
             value = await handler(arguments or {})
             if isinstance(value, dict) and "content" in value:
-                return value
-            return {
+                return self._apply_jsonpath(tool, value)
+            return self._apply_jsonpath(tool, {
                 "content": [{"type": "text", "text": value if isinstance(value, str) else json.dumps(value, default=str)}],
                 "structuredContent": value if isinstance(value, (dict, list)) else None,
                 "isError": False,
-            }
+            })
         if itype == "MCP":
             client = self._upstreams.get(tool.get("gateway_id") or "")
             if client is None:
                 raise ToolInvocationError(f"no upstream for tool {tool['name']}", code=jsonrpc.SERVER_UNAVAILABLE)
-            return await self.retry.run(
+            out = await self.retry.run(
                 lambda: client.call_tool(tool["original_name"], arguments or {}, headers),
                 retry_on=(UpstreamError,),
             )
+            return self._apply_jsonpath(tool, out)
         if itype == "REST":
             return await self._dispatch_rest(tool, arguments, headers)
         if itype == "A2A":

@@ -201,3 +201,24 @@ def test_metrics_recorded(run, bare_engine):
     call(run, bare_engine, "tools/call", {"name": "echo", "arguments": {}})
     snap = bare_engine.metrics.snapshot()
     assert snap["counters"]["tool_invocations_total"] == 1
+
+
+def test_tool_jsonpath_filter_applies_to_results(run, engine):
+    """jsonpath_filter on a tool row filters its results (reference:
+    main.py jsonpath_modifier) — on LOCAL/MCP dispatch, not just REST."""
+    import json as _json
+
+    async def go():
+        async def lister(args):
+            return {"items": [{"name": "a", "secret": 1}, {"name": "b", "secret": 2}]}
+
+        engine.tool_service.register_local_tool(
+            "lister", lister, jsonpath_filter="$.items[*].name")
+        out = await engine.handle_rpc_bytes(_json.dumps(
+            {"jsonrpc": "2.0", "id": 1, "method": "tools/call",
+             "params": {"name": "lister", "arguments": {}}}).encode())
+        res = _json.loads(out)["result"]
+        assert res["structuredContent"] == ["a", "b"]
+        assert "secret" not in res["content"][0]["text"]
+
+    run(go())
