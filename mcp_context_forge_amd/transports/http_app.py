@@ -23,6 +23,7 @@ from fastapi import Depends, FastAPI, HTTPException, Query, Request, Response, W
 from fastapi.responses import HTMLResponse, JSONResponse, PlainTextResponse, StreamingResponse
 from starlette.middleware.base import BaseHTTPMiddleware
 from starlette.middleware.cors import CORSMiddleware
+from starlette.middleware.gzip import GZipMiddleware
 
 from ..auth.service import AuthContext, AuthError, AuthService, PermissionError_
 from ..config import Settings
@@ -83,6 +84,20 @@ class RateLimitMiddleware(BaseHTTPMiddleware):
         if not bucket.allow():
             return JSONResponse({"detail": "rate limit exceeded"}, status_code=429)
         return await call_next(request)
+
+
+class SseAwareGZipMiddleware(GZipMiddleware):
+    """Compression that leaves streaming endpoints alone (reference:
+    SSE-aware compression layer in the middleware stack §2.4)."""
+
+    _STREAM_PREFIXES = ("/servers", "/mcp", "/reverse-proxy", "/ws")
+
+    async def __call__(self, scope, receive, send):
+        if scope["type"] == "http" and any(
+                scope["path"].startswith(p) for p in self._STREAM_PREFIXES):
+            await self.app(scope, receive, send)
+            return
+        await super().__call__(scope, receive, send)
 
 
 class CorrelationIDMiddleware(BaseHTTPMiddleware):
@@ -218,6 +233,7 @@ def build_app(engine: GatewayEngine, auth: Optional[AuthService] = None) -> Fast
 
     # middleware registration — innermost added first (starlette wraps outward);
     # effective order matches reference §2.4 (CORS outermost → ... → handlers)
+    app.add_middleware(SseAwareGZipMiddleware, minimum_size=1024)
     app.add_middleware(CorrelationIDMiddleware)
     app.add_middleware(BodyLimitMiddleware, max_bytes=settings.max_request_body_bytes)
     if settings.rate_limit_enabled:

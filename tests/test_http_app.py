@@ -597,3 +597,26 @@ def test_global_search(client_engine, run):
             assert r.status_code == 204
 
     run(go())
+
+
+def test_gzip_compression_non_streaming_only(client_engine, run):
+    """SSE-aware compression: big JSON responses gzip, /rpc fast lane and
+    streaming endpoints stay uncompressed."""
+    client_factory, engine, app = client_engine
+
+    async def go():
+        for i in range(40):
+            engine.registry.create("tool", name=f"pad-{i}", original_name=f"pad-{i}",
+                                   integration_type="LOCAL", description="x" * 200)
+        async with client_factory() as c:
+            r = await c.get("/tools", headers={**ADMIN, "accept-encoding": "gzip"})
+            assert r.status_code == 200
+            assert r.headers.get("content-encoding") == "gzip"
+            assert len(r.json()) >= 40  # httpx transparently decompresses
+            # the pure-ASGI /rpc lane bypasses middleware: no gzip
+            r = await c.post("/rpc", headers={**ADMIN, "accept-encoding": "gzip"},
+                             json={"jsonrpc": "2.0", "id": 1, "method": "tools/list"})
+            assert r.status_code == 200
+            assert r.headers.get("content-encoding") is None
+
+    run(go())
