@@ -84,6 +84,9 @@ class GatewayEngine:
         from .services.governance import TagService
 
         self.tags = TagService(self.registry)
+        from .services.mcp_apps import McpAppsService
+
+        self.mcp_apps = McpAppsService(self)
         from .services.governance import ServerClassificationService
 
         self.classification = ServerClassificationService(self.registry)
@@ -182,6 +185,9 @@ class GatewayEngine:
                     session.initialized = True
                     session.protocol_version = params.get("protocolVersion")
                 result = initialize_result(params.get("protocolVersion"))
+                ext = self.mcp_apps.capabilities_extension()
+                if ext:
+                    result["capabilities"]["extensions"] = ext
             elif method == "notifications/cancelled":
                 # reference: cancellation_router — cancel the in-flight task
                 if session is not None and params.get("requestId") is not None:
