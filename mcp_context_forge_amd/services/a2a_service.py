@@ -104,9 +104,20 @@ class A2AService:
         self.registry = registry
         self.plugins = plugins or PluginManager([])
         self.allow_private = allow_private_urls
+        self._crypto = None   # lazy (PBKDF2 key stretch is ~50 ms)
         self._client: Optional[httpx.AsyncClient] = None
         self._tasks: Dict[str, dict] = {}  # task store (reference: upsert_task :3248)
         self._local_handlers: Dict[str, Any] = {}
+
+    @property
+    def crypto(self):
+        """Agent credentials sealed at rest like gateway auth material."""
+        if self._crypto is None:
+            from ..auth.crypto import EncryptionService
+            from ..config import get_settings
+
+            self._crypto = EncryptionService(get_settings().jwt_secret_key)
+        return self._crypto
 
     def register_local_agent(self, name: str, handler, description: str = "", **fields) -> dict:
         """In-proc agent for tests/bench (endpoint_url inproc://)."""
@@ -128,7 +139,8 @@ class A2AService:
         return self.registry.create(
             "a2a_agent", name=name, slug=slugify(name), endpoint_url=endpoint_url,
             agent_type=agent_type, protocol_version=protocol_version, description=description,
-            auth_type=auth_type, auth_value=auth_value, tags=tags or [], config=config or {})
+            auth_type=auth_type, auth_value=self.crypto.seal(auth_value),
+            tags=tags or [], config=config or {})
 
     async def invoke_agent(self, name: str, message: str, user: Optional[str] = None,
                            context: Optional[dict] = None, hop_count: int = 0,
@@ -184,7 +196,7 @@ class A2AService:
         payload = build_payload(agent.get("protocol_version", "1.0"), message, context)
         headers = {"content-type": "application/json", HOP_HEADER: str(hop_count + 1), UAID_HEADER: uaid}
         if agent.get("auth_type") == "bearer" and agent.get("auth_value"):
-            headers["authorization"] = f"Bearer {agent['auth_value']}"
+            headers["authorization"] = f"Bearer {self.crypto.open_(agent['auth_value'])}"
         try:
             resp = await self._client.post(agent["endpoint_url"], json=payload, headers=headers)
         except httpx.HTTPError as exc:

@@ -259,3 +259,17 @@ def test_chat_service_tool_loop():
             await asyncio.wait_for(task, timeout=10)
 
     asyncio.run(go())
+
+
+def test_a2a_agent_credentials_sealed(run):
+    async def go():
+        engine = GatewayEngine(Settings(database_url="sqlite://", federation_enabled=False,
+                                        auth_required=False))
+        await engine.a2a_service.register_agent(
+            "authy", "http://127.0.0.1:1/agent", auth_type="bearer", auth_value="agent-secret")
+        row = engine.registry.find("a2a_agent", "authy")
+        assert row["auth_value"].startswith("enc1:") and "agent-secret" not in row["auth_value"]
+        assert engine.a2a_service.crypto.open_(row["auth_value"]) == "agent-secret"
+        await engine.shutdown()
+
+    run(go())
