@@ -618,7 +618,10 @@ def build_app(engine: GatewayEngine, auth: Optional[AuthService] = None) -> Fast
     async def import_config(request: Request, conflict_strategy: str = "update",
                             ctx: AuthContext = Depends(require("admin.import"))):
         body = await request.json()
-        return engine.registry.import_configuration(body, conflict_strategy)
+        out = engine.registry.import_configuration(body, conflict_strategy)
+        engine.sync_plugin_bindings()   # imported bindings reach the manager
+        engine.invalidate_peers("import")
+        return out
 
     # -- auth endpoints (reference: routers/auth.py, routers/tokens.py) ------------
     @app.post("/auth/login")

@@ -620,3 +620,29 @@ def test_gzip_compression_non_streaming_only(client_engine, run):
             assert r.headers.get("content-encoding") is None
 
     run(go())
+
+
+def test_export_import_roundtrip_includes_bindings(client_engine, run):
+    """Config export/import carries plugin bindings and re-syncs the
+    plugin manager on import (reference: export/import services)."""
+    client_factory, engine, app = client_engine
+
+    async def go():
+        engine.set_plugin_binding("echo", "deny_filter", mode="disabled")
+        async with client_factory() as c:
+            r = await c.get("/export", headers=ADMIN)
+            dump = r.json()
+            assert any(b["tool_name"] == "echo" for b in dump["entities"]["plugin_binding"])
+            # wipe the binding, then import the dump back
+            engine.delete_plugin_binding("echo", "deny_filter")
+            assert engine.plugins.bindings == {}
+            r = await c.post("/import", headers=ADMIN, json=dump)
+            assert r.status_code == 200 and r.json()["created"] >= 1
+            assert engine.plugins.bindings["echo"]["deny_filter"]["mode"] == "disabled"
+            # and the chain honors it immediately
+            r = await c.post("/rpc", headers=ADMIN,
+                             json={"jsonrpc": "2.0", "id": 1, "method": "tools/call",
+                                   "params": {"name": "echo", "arguments": {"q": "forbidden"}}})
+            assert "result" in r.json()
+
+    run(go())
