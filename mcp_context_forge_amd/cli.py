@@ -208,6 +208,7 @@ def main(argv=None) -> int:
                 open(args.file, "w").write(payload)
         else:
             counts = engine.registry.import_configuration(json.load(open(args.file)))
+            engine.sync_plugin_bindings()
             print(json.dumps(counts))
         asyncio.run(engine.shutdown())
         return 0
