@@ -232,3 +232,37 @@ def test_summarizer_uses_llm(run):
                        args={"content": [{"type": "text", "text": "tiny"}]})
     out2 = run(PluginManager([p]).invoke_hook(HookType.TOOL_POST_INVOKE, c2))
     assert out2.args["content"][0]["text"] == "tiny"
+
+
+# -- prompt template security (reference: prompt_service Jinja sandbox) ----
+
+def test_prompt_template_sandbox_blocks_escapes(run):
+    """Template injection attempts must not reach Python internals."""
+    from mcp_context_forge_amd.config import Settings
+    from mcp_context_forge_amd.engine import GatewayEngine
+
+    async def go():
+        e = GatewayEngine(Settings(database_url="sqlite://", federation_enabled=False,
+                                   auth_required=False, plugins_enabled=False))
+        e.registry.create("prompt", name="leaky",
+                          template="{{ args.__class__.__mro__ }}")
+        try:
+            await e.prompt_service.get_prompt("leaky", {"args": "x"})
+            leaked = True
+        except Exception:
+            leaked = False
+        assert not leaked, "sandbox allowed dunder access"
+        # benign templates still render with strict-undefined semantics
+        e.registry.create("prompt", name="greet", template="hello {{ name }}")
+        out = await e.prompt_service.get_prompt("greet", {"name": "ada"})
+        text = str(out)
+        assert "hello ada" in text
+        try:
+            await e.prompt_service.get_prompt("greet", {})
+            missing_ok = True
+        except Exception:
+            missing_ok = False
+        assert not missing_ok, "StrictUndefined should reject missing args"
+        await e.shutdown()
+
+    run(go())
