@@ -646,3 +646,40 @@ def test_export_import_roundtrip_includes_bindings(client_engine, run):
             assert "result" in r.json()
 
     run(go())
+
+
+def test_security_limits_and_auth_bypass_attempts(client_engine, run):
+    """Security tier (reference: tests/security/): header-size limit, rate
+    limiting off by default, admin endpoints reject non-admin tokens, and
+    RBAC denials don't leak entity existence."""
+    client_factory, engine, app = client_engine
+
+    async def go():
+        async with client_factory() as c:
+            # oversized headers -> 431 before any handler runs
+            r = await c.get("/version", headers={**ADMIN, "x-big": "x" * 20000})
+            assert r.status_code == 431
+            # admin surface requires admin permission, not just authentication
+            # registration is admin-gated (no self-signup surface)
+            r = await c.post("/auth/register", json={"email": "eve@x.io", "password": "Str0ng!pass1"})
+            assert r.status_code == 401
+            r = await c.post("/auth/register", headers=ADMIN,
+                             json={"email": "eve@x.io", "password": "Str0ng!pass1"})
+            assert r.status_code in (200, 201)
+            r = await c.post("/auth/login", json={"email": "eve@x.io", "password": "Str0ng!pass1"})
+            tok = r.json()["token"]
+            hdr = {"Authorization": f"Bearer {tok}"}
+            for path in ("/admin/audit", "/admin/siem/export", "/admin/compliance/report",
+                         "/admin/runtime", "/admin/logs"):
+                r = await c.get(path, headers=hdr)
+                assert r.status_code == 403, (path, r.status_code)
+            r = await c.patch("/admin/runtime", headers=hdr, json={"window_us": 1})
+            assert r.status_code == 403
+            # export requires admin.import/export perms too
+            r = await c.get("/export", headers=hdr)
+            assert r.status_code == 403
+            # malformed bearer tokens are 401, not 500
+            r = await c.get("/version", headers={"Authorization": "Bearer not.a.jwt"})
+            assert r.status_code == 401
+
+    run(go())
