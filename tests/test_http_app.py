@@ -99,7 +99,7 @@ def test_api_token_lifecycle(client_engine, run):
     async def go():
         async with client_factory() as c:
             r = await c.post("/tokens", json={"name": "ci"}, headers=ADMIN)
-            tok = r.json()["token"]
+            tok = r.json()["access_token"]
             assert tok.startswith("mcpg_")
             r = await c.get("/tools", headers={"Authorization": f"Bearer {tok}"})
             assert r.status_code == 200
@@ -667,7 +667,7 @@ def test_security_limits_and_auth_bypass_attempts(client_engine, run):
                              json={"email": "eve@x.io", "password": "Str0ng!pass1"})
             assert r.status_code in (200, 201)
             r = await c.post("/auth/login", json={"email": "eve@x.io", "password": "Str0ng!pass1"})
-            tok = r.json()["token"]
+            tok = r.json()["access_token"]
             hdr = {"Authorization": f"Bearer {tok}"}
             for path in ("/admin/audit", "/admin/siem/export", "/admin/compliance/report",
                          "/admin/runtime", "/admin/logs"):
