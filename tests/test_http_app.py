@@ -99,7 +99,7 @@ def test_api_token_lifecycle(client_engine, run):
     async def go():
         async with client_factory() as c:
             r = await c.post("/tokens", json={"name": "ci"}, headers=ADMIN)
-            tok = r.json()["access_token"]
+            tok = r.json()["token"]
             assert tok.startswith("mcpg_")
             r = await c.get("/tools", headers={"Authorization": f"Bearer {tok}"})
             assert r.status_code == 200
