@@ -29,6 +29,15 @@ for src in range(world):
     assert [o["from"] for o in got[src]] == [src] * 3
     assert all(o["to"] == rank for o in got[src])
 
+# empty buckets: zero-byte exchanges must not wedge the collective
+empty = collectives.all_to_all_bytes([b"", b""])
+assert empty == [b"", b""], empty
+one_way = collectives.all_to_all_bytes([b"x" * 5 if rank == 0 else b"", b""])
+if rank == 1:
+    assert one_way[0] == b"x" * 5 and one_way[1] == b""
+else:
+    assert one_way == [b"", b""], one_way
+
 # broadcast (registry invalidation analog)
 cfg = {"plugins_enabled": True, "gen": 42} if rank == 0 else None
 out = collectives.broadcast_object(cfg, src=0)
