@@ -32,9 +32,10 @@ for src in range(world):
 # empty buckets: zero-byte exchanges must not wedge the collective
 empty = collectives.all_to_all_bytes([b"", b""])
 assert empty == [b"", b""], empty
+# rank 0 sends 5 bytes to itself only; rank 1 sends nothing
 one_way = collectives.all_to_all_bytes([b"x" * 5 if rank == 0 else b"", b""])
-if rank == 1:
-    assert one_way[0] == b"x" * 5 and one_way[1] == b""
+if rank == 0:
+    assert one_way == [b"x" * 5, b""], one_way
 else:
     assert one_way == [b"", b""], one_way
 
