@@ -67,10 +67,17 @@ def decode_token(token: str, secret: str, audience: Optional[str] = None,
         raise JWTError("signature mismatch")
     if verify_exp and "exp" in claims and time.time() > claims["exp"]:
         raise JWTError("token expired")
-    if audience is not None and "aud" in claims:
+    if audience is not None:
+        # a configured audience REQUIRES the claim (reference: PyJWT
+        # verify_aud rejects tokens missing a required aud)
+        if "aud" not in claims:
+            raise JWTError("audience claim missing")
         auds = claims["aud"] if isinstance(claims["aud"], list) else [claims["aud"]]
         if audience not in auds:
             raise JWTError("audience mismatch")
-    if issuer is not None and "iss" in claims and claims["iss"] != issuer:
-        raise JWTError("issuer mismatch")
+    if issuer is not None:
+        if "iss" not in claims:
+            raise JWTError("issuer claim missing")
+        if claims["iss"] != issuer:
+            raise JWTError("issuer mismatch")
     return claims

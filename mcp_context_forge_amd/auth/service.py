@@ -79,6 +79,15 @@ class AuthService:
         self.settings = settings
         self.token_blocklist = token_blocklist  # reference: services/token_blocklist.py
         self._perm_cache: dict = {}
+        self._revocations = 0  # bumped on API-token revocation
+
+    @property
+    def revocation_epoch(self) -> int:
+        """Monotone counter that changes on ANY token revocation; auth-context
+        caches (RpcFastPath, edge workers) key on it so revoked tokens stop
+        working immediately instead of riding out a TTL."""
+        bl = self.token_blocklist.version if self.token_blocklist is not None else 0
+        return self._revocations + bl
 
     # -- bootstrap -------------------------------------------------------------
     def bootstrap_admin(self) -> None:
@@ -144,6 +153,7 @@ class AuthService:
             if t is None:
                 return False
             t.revoked = True
+            self._revocations += 1
             return True
 
     def list_api_tokens(self, user_email: str) -> List[dict]:

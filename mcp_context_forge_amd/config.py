@@ -36,6 +36,10 @@ class Settings(BaseModel):
     basic_auth_password: str = "changeme"
     auth_required: bool = True
     jwt_secret_key: str = "my-test-key"
+    # credential-at-rest sealing key, DISTINCT from the JWT signing key so a
+    # leaked signing secret cannot decrypt stored upstream credentials
+    # (reference: config auth_encryption_secret, separate from jwt_secret_key)
+    auth_encryption_secret: str = "my-test-salt"
     jwt_algorithm: str = "HS256"
     jwt_audience: str = "mcpgateway-api"
     jwt_issuer: str = "mcpgateway"

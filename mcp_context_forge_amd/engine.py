@@ -330,12 +330,15 @@ class GatewayEngine:
 
     # -- batched entry point (GPU hot path; bench + micro-batching transports) ----
     async def process_rpc_batch(self, raws: List[bytes], user: Optional[str] = None,
-                                server_id: Optional[str] = None) -> List[Optional[bytes]]:
+                                server_id: Optional[str] = None,
+                                users: Optional[List[Optional[str]]] = None) -> List[Optional[bytes]]:
         """Process a batch of raw JSON-RPC requests.
 
         With a GPU pipeline attached, tools/call requests take the staged
         HIP plugin chain (parse → scan/mask/validate/classify on device →
         fan-out → post chain); everything else falls through per-request.
+        `users` (per-request identities from the micro-batch collector)
+        overrides the scalar `user` and scopes cache tenancy per row.
         """
         if self.gpu_pipeline is not None:
             # measured: splitting into two concurrent half-batches LOSES ~9%
@@ -343,7 +346,10 @@ class GatewayEngine:
             # the GPU-sync overlap buys) — one batch at a time is fastest.
             # The C++ stores stay mutex-protected so concurrent callers of
             # process_batch (e.g. collector + direct) remain safe.
-            return await self.gpu_pipeline.process_batch(raws, user=user, server_id=server_id)
+            return await self.gpu_pipeline.process_batch(raws, user=user, server_id=server_id, users=users)
+        if users is not None:
+            return list(await asyncio.gather(
+                *(self.handle_rpc_bytes(r, user=u, server_id=server_id) for r, u in zip(raws, users))))
         return list(await asyncio.gather(*(self.handle_rpc_bytes(r, user=user, server_id=server_id) for r in raws)))
 
     def invalidate_peers(self, what: str = "registry") -> None:

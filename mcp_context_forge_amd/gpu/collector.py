@@ -31,15 +31,16 @@ class BatchCollector:
             self._wakeup = asyncio.Event()
             self._worker = asyncio.create_task(self._run_loop())
 
-    async def submit(self, raw: bytes) -> Optional[bytes]:
+    async def submit(self, raw: bytes, user: Optional[str] = None) -> Optional[bytes]:
         loop = asyncio.get_running_loop()
         fut: asyncio.Future = loop.create_future()
         self._ensure_worker()
-        self._pending.append((raw, fut))
+        self._pending.append((raw, fut, user))
         self._wakeup.set()
         return await fut
 
-    async def submit_many(self, raws: List[bytes]) -> List[Optional[bytes]]:
+    async def submit_many(self, raws: List[bytes],
+                          users: Optional[List[Optional[str]]] = None) -> List[Optional[bytes]]:
         """Submit a pre-formed group (e.g. one edge frame) with ONE future for
         the whole group — the owner loop pays O(frames), not O(requests)."""
         if not raws:
@@ -47,7 +48,7 @@ class BatchCollector:
         loop = asyncio.get_running_loop()
         fut: asyncio.Future = loop.create_future()
         self._ensure_worker()
-        self._pending.append((raws, fut))
+        self._pending.append((raws, fut, users))
         self._wakeup.set()
         return await fut
 
@@ -63,19 +64,24 @@ class BatchCollector:
             if not batch:
                 continue
             self.batches += 1
-            # flatten: an entry is a single bytes or a list (one frame)
+            # flatten: an entry is a single bytes or a list (one frame);
+            # per-request user identity rides along so plugin context,
+            # metrics attribution and cache tenancy survive micro-batching
             raws: List[bytes] = []
+            users: List[Optional[str]] = []
             spans: List[tuple] = []
-            for item, fut in batch:
+            for item, fut, u in batch:
                 if isinstance(item, list):
                     spans.append((len(raws), len(item), fut, True))
                     raws.extend(item)
+                    users.extend(u if isinstance(u, list) else [u] * len(item))
                 else:
                     spans.append((len(raws), 1, fut, False))
                     raws.append(item)
+                    users.append(u)
             self.max_seen = max(self.max_seen, len(raws))
             try:
-                outs = await self.process(raws)
+                outs = await self.process(raws, users=users)
             except Exception as exc:
                 for _, _n, fut, _f in spans:
                     if not fut.done():

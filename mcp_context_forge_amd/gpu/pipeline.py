@@ -122,7 +122,11 @@ class GpuPluginPipeline:
             self.feat_dim = self.moderation.dim
 
         self.semcache: Optional[GpuSemanticCache] = None
-        if self.semcache_plugin is not None:
+        # the HBM cache only exists when the plugin has an EXPLICIT per-tool
+        # allowlist (empty by default, like the reference's cacheable_tools):
+        # with no allowlisted tool the cosine sweep would be pure overhead
+        # and a hit could never legally substitute a result anyway
+        if self.semcache_plugin is not None and self.semcache_plugin.cacheable_tools:
             self.semcache = GpuSemanticCache(
                 capacity=max(128, (s.gpu_semcache_capacity // 128) * 128),
                 dim=self.feat_dim,
@@ -371,12 +375,16 @@ class GpuPluginPipeline:
                     f |= hip.TF_SCHEMA_FAST
                 elif m.schema_mode == "host":
                     f |= hip.TF_SCHEMA_HOST
-            if self.semcache is not None:
+            if self.semcache is not None and self.semcache_plugin.cacheable(m.name):
                 f |= hip.TF_CACHE
             if self.exact_cache is not None:
                 f |= hip.TF_EXACT
             flags[i] = f
         self._t_flags = flags
+        # semcache insert allowlist (tool-level; lookups are gated by TF_CACHE)
+        self._t_semallow = np.array(
+            [bool(self.semcache is not None and self.semcache_plugin.cacheable(m.name)) for m in metas],
+            dtype=bool) if nt else np.zeros(1, dtype=bool)
         self._t_hostbound = hostbound if hostbound.any() else None
         self._meta_gen = self.engine.registry.generation
         self._plugins_ver = getattr(self.engine.plugins, "version", 0)
@@ -423,11 +431,15 @@ class GpuPluginPipeline:
             self.timing[key] = self.timing.get(key, 0.0) + (time.monotonic() - t0)
 
     async def process_batch(self, raws: List[bytes], user: Optional[str] = None,
-                            server_id: Optional[str] = None) -> List[Optional[bytes]]:
+                            server_id: Optional[str] = None,
+                            users: Optional[List[Optional[str]]] = None) -> List[Optional[bytes]]:
         self.batches += 1
         n = len(raws)
         self.requests += n
         responses: List[Optional[bytes]] = [None] * n
+
+        def row_user(i: int) -> Optional[str]:
+            return users[i] if users is not None else user
 
         t0 = self._tic()
         joined, offs_raw = self._pb.concat_with_offsets(raws)
@@ -441,14 +453,14 @@ class GpuPluginPipeline:
         if other_rows.size:
             self.py_fallback += int(other_rows.size)
             outs = await asyncio.gather(
-                *(self.engine.handle_rpc_bytes(raws[int(i)], user=user, server_id=server_id)
+                *(self.engine.handle_rpc_bytes(raws[int(i)], user=row_user(int(i)), server_id=server_id)
                   for i in other_rows))
             for i, out in zip(other_rows, outs):
                 responses[int(i)] = out
 
         fast_rows = np.nonzero(kind == hip.ENV_TOOLS_CALL)[0]
         if fast_rows.size:
-            await self._fast_toolcalls(raws, blob, env, fast_rows, responses, user, server_id)
+            await self._fast_toolcalls(raws, blob, env, fast_rows, responses, user, server_id, users)
         return responses
 
     # ------------------------------------------------------------------
@@ -469,9 +481,25 @@ class GpuPluginPipeline:
 
     async def _fast_toolcalls(self, raws: List[bytes], blob: np.ndarray, env: dict,
                               rows: np.ndarray, responses: List[Optional[bytes]],
-                              user: Optional[str], server_id: Optional[str]) -> None:
+                              user: Optional[str], server_id: Optional[str],
+                              users: Optional[List[Optional[str]]] = None) -> None:
         self._meta()  # refresh tool tables on registry/plugin change
         m = rows.shape[0]
+        # per-row tenant hash: cache identities (semantic + exact) are
+        # (tool, user)-scoped so results never cross users (63-bit so the
+        # XOR with tool_hash stays in int64 domain)
+        if users is not None:
+            memo: Dict[Optional[str], int] = {}
+            uh_all = np.empty(len(users), dtype=np.int64)
+            for _i, _u in enumerate(users):
+                _h = memo.get(_u)
+                if _h is None:
+                    _h = memo[_u] = tool_hash(_u or "")
+                uh_all[_i] = _h
+        else:
+            uh_all = None
+        _uh_scalar = tool_hash(user or "")
+        uh = uh_all[rows] if uh_all is not None else np.full(m, _uh_scalar, dtype=np.int64)
         nb = np.ascontiguousarray(env["name_beg"][rows])
         ne = np.ascontiguousarray(env["name_end"][rows])
         id_b = np.ascontiguousarray(env["id_beg"][rows])
@@ -497,7 +525,10 @@ class GpuPluginPipeline:
                 hb_rows = rows[hb_mask]
                 self.host_bound += int(hb_rows.size)
                 outs = await asyncio.gather(
-                    *(self.engine.handle_rpc_bytes(raws[int(i)], user=user, server_id=server_id)
+                    *(self.engine.handle_rpc_bytes(
+                        raws[int(i)],
+                        user=(users[int(i)] if users is not None else user),
+                        server_id=server_id)
                       for i in hb_rows))
                 for i, out in zip(hb_rows, outs):
                     responses[int(i)] = out
@@ -506,6 +537,7 @@ class GpuPluginPipeline:
                 m = rows.shape[0]
                 if m == 0:
                     return
+                uh = np.ascontiguousarray(uh[keep])
                 tool_idx = np.ascontiguousarray(tool_idx[keep])
                 nb = np.ascontiguousarray(nb[keep])
                 ne = np.ascontiguousarray(ne[keep])
@@ -575,7 +607,8 @@ class GpuPluginPipeline:
             mod_cat = np.ascontiguousarray(sc.argmax(axis=1).astype(np.int32))
             mod_block = (mod_score >= self.moderation.threshold).astype(np.uint8)
 
-        th_arr = np.where(tool_idx >= 0, self._t_thash[np.clip(tool_idx, 0, None)], 0)
+        # semcache identity = tool_hash XOR user_hash (tenant-scoped)
+        th_arr = np.where(tool_idx >= 0, self._t_thash[np.clip(tool_idx, 0, None)] ^ uh, 0)
         hit = np.zeros(m, dtype=np.uint8)
         hit_slot = np.full(m, -1, dtype=np.int32)
         if cache_val_t is not None:
@@ -597,7 +630,7 @@ class GpuPluginPipeline:
         state, nk, reason, arena, rb, re_, n_arena = hip.decide(
             blob, id_b, id_e, args_b, args_e, tool_idx, nb, ne,
             deny_m, harm_m, pii_m, regex_m, norm_m, schema_m,
-            mod_block, mod_cat, mod_score, hit, hit_slot,
+            mod_block, mod_cat, mod_score, hit, hit_slot, np.ascontiguousarray(uh),
             flags, self._t_required, self._t_typed,
             self._t_name_beg, self._t_name_end, self._t_name_blob, self._t_native_kind,
             self._nest_bits,
@@ -668,7 +701,7 @@ class GpuPluginPipeline:
         if native_js or py_items:
             await self._dispatch_and_post(blob, env, rows, id_b, id_e, args_b, args_e,
                                           tool_idx, nk, feats, th_arr, native_js, py_items,
-                                          responses, t0, feats_sk)
+                                          responses, t0, feats_sk, uh)
 
     def _side_stream(self, i: int) -> "torch.cuda.Stream":
         ss = getattr(self, "_streams", None)
@@ -828,7 +861,7 @@ class GpuPluginPipeline:
                                  tool_idx, nk, feats, th_arr,
                                  native_js: List[int], py_items: List[Tuple[int, Any]],
                                  responses: List[Optional[bytes]], t0: float,
-                                 feats_sk=None) -> None:
+                                 feats_sk=None, uh=None) -> None:
         # --- native upstream batch (C++) ---
         t_u = self._tic()
         nat_blob = np.zeros(0, dtype=np.uint8)
@@ -964,8 +997,10 @@ class GpuPluginPipeline:
         t_f = self._tic()
         t_fc = self._tic()
         now = time.monotonic()
+        if uh is None:
+            uh = np.zeros(id_b.shape[0], dtype=np.int64)
         arena2, rb2, re2, is_err, cacheable = hip.finalize(
-            blob, id_b, id_e, args_b, args_e, tool_idx,
+            blob, id_b, id_e, args_b, args_e, tool_idx, np.ascontiguousarray(uh),
             np.ascontiguousarray(all_js_np.astype(np.int32)), res_blob, res_beg, res_end,
             np.ascontiguousarray(needs_host.astype(np.uint8)),
             self._t_name_beg, self._t_name_end, self._t_name_blob, self._t_flags,
@@ -1011,7 +1046,11 @@ class GpuPluginPipeline:
         t_fs = self._tic()
         # --- bookkeeping: semcache insert, breaker, metrics ---
         if self.semcache is not None and n_all:
-            ins = np.nonzero((cacheable == 1))[0]
+            # only allowlisted tools are inserted (cacheable_tools gate);
+            # that also gates lookups implicitly — a non-allowlisted tool's
+            # (tool,user) identity never exists in the cache
+            ins = np.nonzero((cacheable == 1) &
+                             self._t_semallow[np.clip(tool_idx[all_js_np], 0, None)])[0]
             if ins.size:
                 ins_js = all_js_np[ins]
                 slots = self.semcache.assign_slots(int(ins.size))

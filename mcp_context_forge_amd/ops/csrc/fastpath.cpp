@@ -179,6 +179,7 @@ int64_t forge_decide(
     const uint32_t* regex_m, const uint32_t* norm_m, const uint32_t* schema_m,
     const uint8_t* mod_block, const int32_t* mod_cat, const float* mod_score,
     const uint8_t* hit, const int32_t* hit_slot,
+    const uint64_t* user_hash,                          // [m] tenant scope for cache keys
     int m,
     // unique-tool table (nt tools)
     const uint32_t* tool_flags, const uint32_t* tool_required_bits,
@@ -307,6 +308,8 @@ int64_t forge_decide(
         if ((fl & TF_EXACT) && ec != nullptr) {
             uint64_t h = fnv64(tname_blob + tname_beg[ti], (size_t)(tname_end[ti] - tname_beg[ti]));
             h = fnv64((const uint8_t*)"\x00", 1, h);
+            uint64_t uh = user_hash ? user_hash[i] : 0;
+            h = fnv64((const uint8_t*)&uh, 8, h);  // tenant scope
             h = fnv64(blob + args_beg[i], (size_t)(args_end[i] - args_beg[i]), h);
             int sh = ec->shard_of(h);
             bool answered_from_exact = false;
@@ -369,7 +372,9 @@ int64_t forge_finalize(
     const uint8_t* blob,
     const int32_t* id_beg, const int32_t* id_end,
     const int32_t* args_beg, const int32_t* args_end,
-    const int32_t* tool_idx, int m,
+    const int32_t* tool_idx,
+    const uint64_t* user_hash,                     // [m] tenant scope for cache keys
+    int m,
     const int32_t* rows, int n_rows,               // row indices being finalized
     const uint8_t* res_blob, const int64_t* res_beg, const int64_t* res_end,  // [n_rows]
     const uint8_t* needs_host,                      // [n_rows]
@@ -415,6 +420,8 @@ int64_t forge_finalize(
                 if (ec != nullptr && ti >= 0 && (tool_flags[ti] & TF_EXACT)) {
                     uint64_t h = fnv64(tname_blob + tname_beg[ti], (size_t)(tname_end[ti] - tname_beg[ti]));
                     h = fnv64((const uint8_t*)"\x00", 1, h);
+                    uint64_t uh = user_hash ? user_hash[i] : 0;
+                    h = fnv64((const uint8_t*)&uh, 8, h);  // tenant scope (must match decide)
                     h = fnv64(blob + args_beg[i], (size_t)(args_end[i] - args_beg[i]), h);
                     ins.emplace_back(h, std::make_pair(res_beg[j], res_end[j]));
                 }

@@ -267,7 +267,7 @@ def _fastpath_protos() -> ctypes.CDLL:
     lib.forge_toolmap_free.restype = None
     lib.forge_toolmap_resolve.argtypes = [ctypes.c_void_p] * 4 + [ctypes.c_int, ctypes.c_void_p]
     lib.forge_toolmap_resolve.restype = None
-    lib.forge_decide.argtypes = [ctypes.c_void_p] * 19 + [ctypes.c_int] + [ctypes.c_void_p] * 7 + \
+    lib.forge_decide.argtypes = [ctypes.c_void_p] * 20 + [ctypes.c_int] + [ctypes.c_void_p] * 7 + \
         [ctypes.c_uint32] + \
         [ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int] * 3 + \
         [ctypes.c_void_p, ctypes.c_void_p, ctypes.c_double] + \
@@ -275,7 +275,7 @@ def _fastpath_protos() -> ctypes.CDLL:
     lib.forge_store_get.argtypes = [ctypes.c_void_p, ctypes.c_int, ctypes.c_void_p, ctypes.c_int64]
     lib.forge_store_get.restype = ctypes.c_int64
     lib.forge_decide.restype = ctypes.c_int64
-    lib.forge_finalize.argtypes = [ctypes.c_void_p] * 6 + [ctypes.c_int] + [ctypes.c_void_p, ctypes.c_int] + \
+    lib.forge_finalize.argtypes = [ctypes.c_void_p] * 7 + [ctypes.c_int] + [ctypes.c_void_p, ctypes.c_int] + \
         [ctypes.c_void_p] * 4 + [ctypes.c_void_p] * 4 + \
         [ctypes.c_void_p, ctypes.c_double, ctypes.c_double] + \
         [ctypes.c_void_p, ctypes.c_int64] + [ctypes.c_void_p] * 4
@@ -327,7 +327,7 @@ def toolmap_resolve(tm: int, blob: np.ndarray, name_beg: np.ndarray, name_end: n
 
 def decide(blob, id_beg, id_end, args_beg, args_end, tool_idx, name_beg, name_end,
            deny_m, harm_m, pii_m, regex_m, norm_m, schema_m,
-           mod_block, mod_cat, mod_score, hit, hit_slot,
+           mod_block, mod_cat, mod_score, hit, hit_slot, user_hash,
            tool_flags, tool_required_bits, tool_typed_pairs,
            tname_beg, tname_end, tname_blob, tool_native_kind, nest_bits,
            deny_words, deny_off, harm_cats, harm_off, mod_cats, mod_off,
@@ -348,6 +348,7 @@ def decide(blob, id_beg, id_end, args_beg, args_end, tool_idx, name_beg, name_en
             _np_ptr(tool_idx), _np_ptr(name_beg), _np_ptr(name_end),
             _np_ptr(deny_m), _np_ptr(harm_m), _np_ptr(pii_m), _np_ptr(regex_m), _np_ptr(norm_m), _np_ptr(schema_m),
             _np_ptr(mod_block), _np_ptr(mod_cat), _np_ptr(mod_score), _np_ptr(hit), _np_ptr(hit_slot),
+            _np_ptr(user_hash),
             m,
             _np_ptr(tool_flags), _np_ptr(tool_required_bits), _np_ptr(tool_typed_pairs),
             _np_ptr(tname_beg), _np_ptr(tname_end), _np_ptr(tname_blob), _np_ptr(tool_native_kind),
@@ -362,7 +363,7 @@ def decide(blob, id_beg, id_end, args_beg, args_end, tool_idx, name_beg, name_en
         cap = -int(n) + 4096
 
 
-def finalize(blob, id_beg, id_end, args_beg, args_end, tool_idx,
+def finalize(blob, id_beg, id_end, args_beg, args_end, tool_idx, user_hash,
              rows, res_blob, res_beg, res_end, needs_host,
              tname_beg, tname_end, tname_blob, tool_flags,
              exact_cache: int, now: float, exact_ttl: float):
@@ -378,7 +379,7 @@ def finalize(blob, id_beg, id_end, args_beg, args_end, tool_idx,
         arena = np.empty(cap, dtype=np.uint8)
         n = lib.forge_finalize(
             _np_ptr(blob), _np_ptr(id_beg), _np_ptr(id_end), _np_ptr(args_beg), _np_ptr(args_end),
-            _np_ptr(tool_idx), id_beg.shape[0],
+            _np_ptr(tool_idx), _np_ptr(user_hash), id_beg.shape[0],
             _np_ptr(rows), n_rows,
             _np_ptr(res_blob), _np_ptr(res_beg), _np_ptr(res_end), _np_ptr(needs_host),
             _np_ptr(tname_beg), _np_ptr(tname_end), _np_ptr(tname_blob), _np_ptr(tool_flags),

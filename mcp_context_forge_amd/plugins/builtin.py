@@ -392,15 +392,24 @@ class ResponseCacheByPromptPlugin(Plugin):
         self.threshold = float(self.config.get("threshold", 0.92))
         self.capacity = int(self.config.get("capacity", 4096))
         self.ttl = float(self.config.get("ttl", 600.0))
+        # result substitution is gated on an EXPLICIT per-tool allowlist
+        # (reference default: empty cacheable_tools — a near-miss cosine hit
+        # on an unlisted tool must never replace a live invocation), and
+        # entries are tenant-scoped: the identity compared on hit is
+        # (tool, user), so one user's result is never served to another.
+        self.cacheable_tools = frozenset(self.config.get("cacheable_tools") or [])
         # preallocated ring (matrix never reallocates — mirrors the GPU cache)
         self.vectors = np.zeros((self.capacity, self.dim), dtype=np.float32)
-        self.entries: List[Optional[Tuple[str, float, Any]]] = [None] * self.capacity
+        self.entries: List[Optional[Tuple[Tuple[str, str], float, Any]]] = [None] * self.capacity
         self.size = 0
         self.write_ptr = 0
         self.hits = 0
         self.misses = 0
 
-    def lookup(self, tool: str, text: str) -> Optional[Any]:
+    def cacheable(self, tool: str) -> bool:
+        return tool in self.cacheable_tools
+
+    def lookup(self, tool: str, text: str, user: Optional[str] = None) -> Optional[Any]:
         if self.size == 0:
             self.misses += 1
             return None
@@ -410,31 +419,33 @@ class ResponseCacheByPromptPlugin(Plugin):
         now = time.monotonic()
         ent = self.entries[best]
         if ent is not None:
-            ent_tool, ts, result = ent
-            if sims[best] >= self.threshold and ent_tool == tool and now - ts <= self.ttl:
+            ent_key, ts, result = ent
+            if sims[best] >= self.threshold and ent_key == (tool, user or "") and now - ts <= self.ttl:
                 self.hits += 1
                 return result
         self.misses += 1
         return None
 
-    def insert(self, tool: str, text: str, result: Any) -> None:
+    def insert(self, tool: str, text: str, result: Any, user: Optional[str] = None) -> None:
         slot = self.write_ptr
         self.vectors[slot] = featurize(text, self.dim)
-        self.entries[slot] = (tool, time.monotonic(), result)
+        self.entries[slot] = ((tool, user or ""), time.monotonic(), result)
         self.write_ptr = (self.write_ptr + 1) % self.capacity
         self.size = min(self.size + 1, self.capacity)
 
     async def tool_pre_invoke(self, ctx: PluginContext) -> PluginResult:
-        hit = self.lookup(ctx.name, _text_of(ctx.args))
+        if not self.cacheable(ctx.name):
+            return PluginResult.ok()
+        hit = self.lookup(ctx.name, _text_of(ctx.args), user=ctx.user)
         if hit is not None:
             ctx.state["cache_hit"] = hit
         return PluginResult.ok()
 
     async def tool_post_invoke(self, ctx: PluginContext) -> PluginResult:
-        if "cache_hit" not in ctx.state and ctx.args is not None:
+        if self.cacheable(ctx.name) and "cache_hit" not in ctx.state and ctx.args is not None:
             req_text = ctx.state.get("request_text", "")
             if req_text:
-                self.insert(ctx.name, req_text, ctx.args)
+                self.insert(ctx.name, req_text, ctx.args, user=ctx.user)
         return PluginResult.ok()
 
 
@@ -450,11 +461,12 @@ class CachedToolResultPlugin(Plugin):
         self.ttl = float(self.config.get("ttl", 300.0))
         self.store: Dict[str, Tuple[float, Any]] = {}
 
-    def _key(self, name: str, args: Any) -> str:
-        return hashlib.sha256((name + _text_of(args)).encode()).hexdigest()
+    def _key(self, name: str, args: Any, user: Optional[str] = None) -> str:
+        # user is part of the key: exact-match results never cross tenants
+        return hashlib.sha256((name + "\x00" + (user or "") + "\x00" + _text_of(args)).encode()).hexdigest()
 
     async def tool_pre_invoke(self, ctx: PluginContext) -> PluginResult:
-        k = self._key(ctx.name, ctx.args)
+        k = self._key(ctx.name, ctx.args, user=ctx.user)
         ent = self.store.get(k)
         if ent and time.monotonic() - ent[0] <= self.ttl:
             ctx.state["cache_hit"] = ent[1]

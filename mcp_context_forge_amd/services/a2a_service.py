@@ -116,7 +116,7 @@ class A2AService:
             from ..auth.crypto import EncryptionService
             from ..config import get_settings
 
-            self._crypto = EncryptionService(get_settings().jwt_secret_key)
+            self._crypto = EncryptionService(get_settings().auth_encryption_secret)
         return self._crypto
 
     def register_local_agent(self, name: str, handler, description: str = "", **fields) -> dict:

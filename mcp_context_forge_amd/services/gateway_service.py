@@ -42,7 +42,8 @@ class GatewayService:
         from ..auth.crypto import EncryptionService
 
         # credential material is sealed at rest (reference: EncryptedText db.py:277)
-        self.crypto = EncryptionService(self.settings.jwt_secret_key)
+        # keyed by auth_encryption_secret, NOT the JWT signing key
+        self.crypto = EncryptionService(self.settings.auth_encryption_secret)
 
     # -- client construction ----------------------------------------------------
     def _make_client(self, gateway: Dict[str, Any]) -> UpstreamClient:

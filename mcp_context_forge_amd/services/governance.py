@@ -101,9 +101,11 @@ class TokenBlocklist:
 
     def __init__(self):
         self._blocked: Dict[str, float] = {}  # jti -> expiry epoch (0 = forever)
+        self.version = 0  # bumped on every revocation; auth caches key on it
 
     def block(self, jti: str, expires_at: float = 0.0) -> None:
         self._blocked[jti] = expires_at
+        self.version += 1
 
     def is_blocked(self, jti: Optional[str]) -> bool:
         if not jti:

@@ -135,16 +135,24 @@ class RpcFastPath:
         self.app = app
         self.fastapi = fastapi_app
         self._auth_cache: Dict[bytes, tuple] = {}
+        self._auth_epoch = -1
 
     def _auth(self, authz: Optional[bytes]):
         import time as _t
 
+        auth = self.fastapi.state.auth
+        epoch = auth.revocation_epoch
+        if epoch != self._auth_epoch:
+            # a token was revoked somewhere: drop every memoized context so
+            # revoked credentials stop working immediately, not after the TTL
+            self._auth_cache.clear()
+            self._auth_epoch = epoch
         ent = self._auth_cache.get(authz)
         now = _t.monotonic()
         if ent is not None and ent[1] > now:
             return ent[0]
         try:
-            ctx = self.fastapi.state.auth.authenticate(authz.decode() if authz else None)
+            ctx = auth.authenticate(authz.decode() if authz else None)
         except AuthError:
             return None
         if len(self._auth_cache) > 4096:
@@ -169,15 +177,20 @@ class RpcFastPath:
         if ctx is None:
             await self._respond(send, 401, b'{"detail":"Not authenticated"}')
             return
+        max_body = st.engine.settings.max_request_body_bytes
         body = b""
         while True:
             msg = await receive()
             body += msg.get("body", b"")
+            if len(body) > max_body:
+                # chunked uploads (no Content-Length) must hit the same guard
+                await self._respond(send, 413, b'{"detail":"request body too large"}')
+                return
             if not msg.get("more_body", False):
                 break
         collector = st.collector
         if collector is not None and ctx.server_id is None:
-            out = await collector.submit(body)
+            out = await collector.submit(body, user=ctx.user)
         else:
             out = await st.engine.handle_rpc_bytes(body, user=ctx.user, server_id=ctx.server_id)
         if out is None:
@@ -270,7 +283,7 @@ def build_app(engine: GatewayEngine, auth: Optional[AuthService] = None) -> Fast
     async def rpc_bytes(raw: bytes, ctx: AuthContext, server_id: Optional[str] = None) -> Optional[bytes]:
         sid = server_id or ctx.server_id  # token-scoped server (reference: token_scoping.py)
         if app.state.collector is not None and server_id is None and ctx.server_id is None:
-            return await app.state.collector.submit(raw)
+            return await app.state.collector.submit(raw, user=ctx.user)
         return await engine.handle_rpc_bytes(raw, user=ctx.user, server_id=sid)
 
     # -- health / version / metrics (reference: main.py /health /version, prometheus) --
