@@ -120,6 +120,10 @@ class Settings(BaseModel):
     gpu_streams: int = 2              # compute + copy overlap
     gpu_dtype: str = "bf16"
 
+    # --- native HTTP edge (C++ epoll owner loop, transports/native_edge.py) ---
+    native_edge_port: int = 0         # 0 = disabled
+    native_edge_threads: int = 0      # 0 = auto (min(8, cpus // 2))
+
     # --- multi-GPU scale-out (reference analog: session_affinity over Redis; here RCCL) ---
     world_size: int = 1
     rank: int = 0

@@ -23,6 +23,8 @@ SOURCES = ["scan.hip", "featurize.hip", "json_guard.hip", "gemm_bf16.hip", "runt
 
 
 PYBRIDGE = Path(__file__).parent / "forge_pybridge.so"
+EDGE = Path(__file__).parent / "forge_edge.so"
+HEY = Path(__file__).parent / "forge_hey"
 
 
 def needs_build() -> bool:
@@ -52,8 +54,40 @@ def build_pybridge(force: bool = False, verbose: bool = True) -> Path:
     return PYBRIDGE
 
 
+def build_edge(force: bool = False, verbose: bool = True) -> Path:
+    """Native epoll HTTP edge (CPython extension, plain g++ — no HIP):
+    forge_edge.so in-tree so the gpurun snapshot ships it."""
+    src = CSRC / "edge.cpp"
+    if not force and EDGE.exists() and EDGE.stat().st_mtime > src.stat().st_mtime:
+        return EDGE
+    import sysconfig
+
+    cxx = os.environ.get("CXX", "g++")
+    cmd = [cxx, "-O2", "-std=c++17", "-fPIC", "-shared", "-pthread",
+           f"-I{sysconfig.get_paths()['include']}", "-o", str(EDGE), str(src)]
+    if verbose:
+        print("[forge-edge]", " ".join(cmd), file=sys.stderr)
+    subprocess.run(cmd, check=True)
+    return EDGE
+
+
+def build_hey(force: bool = False, verbose: bool = True) -> Path:
+    """Native closed-loop HTTP load generator (hey analog) used by bench.py."""
+    src = CSRC / "forge_hey.cpp"
+    if not force and HEY.exists() and HEY.stat().st_mtime > src.stat().st_mtime:
+        return HEY
+    cxx = os.environ.get("CXX", "g++")
+    cmd = [cxx, "-O2", "-std=c++17", "-pthread", "-o", str(HEY), str(src)]
+    if verbose:
+        print("[forge-hey]", " ".join(cmd), file=sys.stderr)
+    subprocess.run(cmd, check=True)
+    return HEY
+
+
 def build(force: bool = False, verbose: bool = True) -> Path:
     build_pybridge(force=force, verbose=verbose)
+    build_edge(force=force, verbose=verbose)
+    build_hey(force=force, verbose=verbose)
     if not force and not needs_build():
         return LIB
     cmd = [

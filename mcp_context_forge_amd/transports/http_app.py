@@ -232,7 +232,23 @@ def build_app(engine: GatewayEngine, auth: Optional[AuthService] = None) -> Fast
             owner_server = GpuOwnerServer(engine, app_.state.collector, settings.edge_socket)
             await owner_server.start()
             app_.state.owner_server = owner_server
+        native_edge = None
+        if settings.native_edge_port:
+            # C++ epoll front door: hot /rpc batches + cold control-plane
+            # passthrough to this app (transports/native_edge.py)
+            import os as _os
+
+            from .native_edge import NativeEdge
+
+            threads = settings.native_edge_threads or min(8, max(2, (_os.cpu_count() or 8) // 2))
+            native_edge = NativeEdge(engine, app=app_, auth=auth,
+                                     port=settings.native_edge_port, threads=threads,
+                                     control_url=f"http://{settings.host}:{settings.port}")
+            await native_edge.start()
+            app_.state.native_edge = native_edge
         yield
+        if native_edge is not None:
+            await native_edge.stop()
         if owner_server is not None:
             await owner_server.stop()
         await engine.shutdown()
