@@ -105,6 +105,11 @@ class GatewayEngine:
         self.gpu_pipeline = None  # attached lazily by gpu.pipeline when available
         self.bus = None           # RcclBus when running multi-rank (parallel/runtime.py)
         self.forward_rpc = None   # set by DistributedGateway: (dest, raw, ...) -> bytes
+        self.forward_rpc_batch = None  # (dest, raws, users) -> List[bytes]
+        # cross-rank tool ownership (reference analog: every gunicorn worker
+        # sees every tool via the shared DB; here each rank shards upstreams
+        # and peers learn name -> owner_rank via sync_tool_ownership)
+        self.foreign_tools: Dict[str, int] = {}
         self._log_level = "info"
         self._maintenance_task = None
         self._maintenance_stop = None
@@ -325,6 +330,14 @@ class GatewayEngine:
                 self.elicitation.resolve(obj["id"], obj.get("result"), obj.get("error"))
                 return None
             return jsonrpc.JSONRPCResponse(id=None, error=exc).to_bytes()
+        # cross-rank routing: a tools/call for a tool owned by a peer rank
+        # rides the RCCL bus to its owner (reference: session_affinity
+        # forward_request_to_owner :747 over Redis)
+        if self.foreign_tools and self.forward_rpc is not None and req.method == "tools/call" \
+                and isinstance(req.params, dict):
+            dest = self.foreign_tools.get(req.params.get("name") or "")
+            if dest is not None and dest != self.rank:
+                return await self.forward_rpc(dest, raw, user, server_id)
         resp = await self.handle_rpc(req, user=user, server_id=server_id, session=session, headers=headers)
         return resp.to_bytes() if resp is not None else None
 

@@ -513,6 +513,48 @@ class GpuPluginPipeline:
             else np.full(m, -1, dtype=np.int32)
         self._toc("toolmap", t_r)
 
+        # --- cross-rank routing: rows whose tool is owned by a peer rank
+        # ride the RCCL bus as per-destination batches (BASELINE config 3:
+        # federation shard fan-out); they overlap the local GPU pass and are
+        # spliced into `responses` at the end ---
+        fw_state = None
+        eng = self.engine
+        if eng.foreign_tools and eng.forward_rpc_batch is not None and (tool_idx < 0).any():
+            groups: Dict[int, List[int]] = {}
+            for j in np.nonzero(tool_idx < 0)[0]:
+                j = int(j)
+                name = blob[nb[j]:ne[j]].tobytes().decode("utf-8", "replace")
+                dest = eng.foreign_tools.get(name)
+                if dest is not None and dest != eng.rank:
+                    groups.setdefault(dest, []).append(j)
+            if groups:
+                tasks, fw_rows = [], []
+                fw_mask = np.zeros(m, dtype=bool)
+                for dest, js in groups.items():
+                    rws = [raws[int(rows[j])] for j in js]
+                    if users is not None:
+                        us = [users[int(rows[j])] for j in js]
+                    else:
+                        us = [user] * len(js) if user is not None else None
+                    tasks.append(eng.forward_rpc_batch(dest, rws, us))
+                    fw_rows.append([int(rows[j]) for j in js])
+                    fw_mask[js] = True
+                fw_state = (asyncio.gather(*tasks), fw_rows)
+                keep = ~fw_mask
+                rows = rows[keep]
+                m = rows.shape[0]
+                if m == 0:
+                    await self._finish_forward(fw_state, responses)
+                    return
+                uh = np.ascontiguousarray(uh[keep])
+                tool_idx = np.ascontiguousarray(tool_idx[keep])
+                nb = np.ascontiguousarray(nb[keep])
+                ne = np.ascontiguousarray(ne[keep])
+                id_b = np.ascontiguousarray(id_b[keep])
+                id_e = np.ascontiguousarray(id_e[keep])
+                args_b = np.ascontiguousarray(args_b[keep])
+                args_e = np.ascontiguousarray(args_e[keep])
+
         # tools with semantics-altering plugin bindings run the full CPU
         # chain (the binding-aware invoke_hook path); everything else stays
         # on the GPU fast path
@@ -536,6 +578,7 @@ class GpuPluginPipeline:
                 rows = rows[keep]
                 m = rows.shape[0]
                 if m == 0:
+                    await self._finish_forward(fw_state, responses)
                     return
                 uh = np.ascontiguousarray(uh[keep])
                 tool_idx = np.ascontiguousarray(tool_idx[keep])
@@ -702,6 +745,17 @@ class GpuPluginPipeline:
             await self._dispatch_and_post(blob, env, rows, id_b, id_e, args_b, args_e,
                                           tool_idx, nk, feats, th_arr, native_js, py_items,
                                           responses, t0, feats_sk, uh)
+        await self._finish_forward(fw_state, responses)
+
+    async def _finish_forward(self, fw_state, responses: List[Optional[bytes]]) -> None:
+        """Splice cross-rank (bus-forwarded) responses into the batch output."""
+        if fw_state is None:
+            return
+        gathered, fw_rows = fw_state
+        outs_groups = await gathered
+        for grp_rows, outs in zip(fw_rows, outs_groups):
+            for r, o in zip(grp_rows, outs):
+                responses[r] = o if (o is None or isinstance(o, bytes)) else bytes(o)
 
     def _side_stream(self, i: int) -> "torch.cuda.Stream":
         ss = getattr(self, "_streams", None)

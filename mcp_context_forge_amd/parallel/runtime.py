@@ -37,6 +37,7 @@ class DistributedGateway:
             self.bus.on_publish = self._on_publish
             engine.bus = self.bus
             engine.forward_rpc = self.forward_rpc
+            engine.forward_rpc_batch = self.forward_rpc_batch
 
     async def start(self) -> None:
         await self.engine.startup()
@@ -57,6 +58,9 @@ class DistributedGateway:
                                                      user=payload.get("user"),
                                                      server_id=payload.get("server_id"))
             return {"raw": out}
+        if isinstance(payload, dict) and payload.get("kind") == "rpc_batch":
+            outs = await self.engine.process_rpc_batch(payload["raws"], users=payload.get("users"))
+            return {"raws": outs}
         return {"error": "unknown payload kind"}
 
     async def forward_rpc(self, dest_rank: int, raw: bytes, user: Optional[str] = None,
@@ -66,6 +70,43 @@ class DistributedGateway:
         out = await self.bus.submit(dest_rank, {"kind": "rpc", "raw": raw, "user": user,
                                                 "server_id": server_id})
         return out.get("raw")
+
+    async def forward_rpc_batch(self, dest_rank: int, raws: list,
+                                users: Optional[list] = None) -> list:
+        """Batch variant: one bus message carries a whole group of requests
+        destined for `dest_rank` (the pipeline ships foreign-tool rows this
+        way so a micro-batch costs O(dest ranks) bus messages, not O(rows))."""
+        assert self.bus is not None
+        out = await self.bus.submit(dest_rank, {"kind": "rpc_batch", "raws": raws, "users": users})
+        if "raws" not in out:
+            raise RuntimeError(f"bus forward failed: {out}")
+        return out["raws"]
+
+    async def sync_tool_ownership(self) -> None:
+        """COLLECTIVE: every rank publishes the tool names it owns locally;
+        peers record name -> owner_rank so requests route over the bus.
+        Call after registering upstreams (and after registry changes, from
+        all ranks together). Reference analog: the shared registry DB that
+        lets any gunicorn worker resolve any tool."""
+        if self.world <= 1:
+            return
+        local = [t["name"] for t in self.engine.registry.list("tool", include_disabled=False)]
+
+        def gather():
+            import torch.distributed as dist
+
+            box = [None] * self.world
+            dist.all_gather_object(box, local)
+            return box
+
+        box = await asyncio.to_thread(gather)
+        fmap = {}
+        for r, names in enumerate(box):
+            if r == self.rank:
+                continue
+            for n in names:
+                fmap[n] = r
+        self.engine.foreign_tools = fmap
 
     def owner_of_tool(self, tool_name: str) -> int:
         tool = self.engine.registry.lookup_tool(tool_name)

@@ -73,3 +73,75 @@ def test_distributed_runtime_world2(tmp_path):
         capture_output=True, text=True, timeout=240, env=env, cwd="/root/repo")
     assert out.returncode == 0, out.stdout[-3000:] + out.stderr[-3000:]
     assert "DISTRIBUTED_OK" in out.stdout
+
+
+ROUTER_WORKER = r"""
+import asyncio, json, os, sys
+import torch.distributed as dist
+from mcp_context_forge_amd.config import Settings
+from mcp_context_forge_amd.engine import GatewayEngine
+from mcp_context_forge_amd.parallel import collectives
+from mcp_context_forge_amd.parallel.runtime import DistributedGateway
+from mcp_context_forge_amd.services.upstream import NativeInProcUpstream
+
+rank, world = collectives.init_from_env()
+
+async def main():
+    # each rank has its OWN registry (sharded upstreams, BASELINE config 3)
+    engine = GatewayEngine(Settings(database_url="sqlite://",
+                                    federation_enabled=False, auth_required=False,
+                                    plugins_enabled=True),
+                           rank=rank, world_size=world)
+    dg = DistributedGateway(engine, cadence_us=2000)
+    up = NativeInProcUpstream(name=f"up{rank}")
+    await engine.gateway_service.register_gateway(
+        name=f"up{rank}", url=f"inproc://up{rank}", client=up, owner_rank=rank)
+    # ownership sync is a collective on the default group: run it BEFORE the
+    # bus pump starts so the two never interleave
+    await dg.sync_tool_ownership()
+    await dg.start()
+    try:
+        other = 1 - rank
+        assert f"up{other}-echo" in engine.foreign_tools, engine.foreign_tools
+        assert engine.foreign_tools[f"up{other}-echo"] == other
+        # a batch mixing local + foreign tools routes transparently
+        raws = []
+        for i in range(8):
+            target = rank if i % 2 == 0 else other
+            raws.append(json.dumps({
+                "jsonrpc": "2.0", "id": i, "method": "tools/call",
+                "params": {"name": f"up{target}-echo", "arguments": {"msg": f"r{rank}-i{i}"}},
+            }).encode())
+        outs = await engine.process_rpc_batch(raws, users=[f"user{rank}"] * len(raws))
+        assert len(outs) == 8
+        for i, o in enumerate(outs):
+            res = json.loads(o)
+            assert res["id"] == i, res
+            assert "result" in res, res
+            text = res["result"]["content"][0]["text"]
+            assert f"r{rank}-i{i}" in text, (i, res)
+        await asyncio.to_thread(collectives.barrier)
+    finally:
+        await dg.stop()
+    if rank == 0:
+        print("ROUTING_OK")
+
+asyncio.run(main())
+dist.destroy_process_group()
+"""
+
+
+def test_cross_rank_tool_routing_world2(tmp_path):
+    """Foreign-tool rows in a batch ride the bus to their owner rank and the
+    responses splice back in order (the pipeline's fw_state path on GPU; the
+    per-request handle_rpc_bytes path on CPU — this exercises the latter plus
+    the rpc_batch bus handler both directions)."""
+    script = tmp_path / "router_worker.py"
+    script.write_text(ROUTER_WORKER)
+    env = dict(os.environ, PYTHONPATH="/root/repo")
+    out = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1", "--nproc-per-node", "2",
+         "--master-addr", "127.0.0.1", "--master-port", "29673", str(script)],
+        capture_output=True, text=True, timeout=240, env=env, cwd="/root/repo")
+    assert out.returncode == 0, out.stdout[-3000:] + out.stderr[-3000:]
+    assert "ROUTING_OK" in out.stdout
