@@ -179,7 +179,8 @@ async def run_http(args, rank: int, world: int, use_gpu: bool, R: int):
 
     port = args.port_base + rank
     threads = args.edge_threads or min(8, max(2, (os.cpu_count() or 16) // (2 * max(world, 1))))
-    edge = NativeEdge(engine, app=None, auth=auth, port=port, threads=threads)
+    edge = NativeEdge(engine, app=None, auth=auth, port=port, threads=threads,
+                      depth=args.edge_depth)
     await edge.start()
 
     # payload corpus: P distinct requests, rotated by the generator; targets
@@ -360,6 +361,8 @@ async def main() -> None:
     ap.add_argument("--payloads", type=int, default=2048, help="distinct payloads in the corpus")
     ap.add_argument("--port-base", type=int, default=18400)
     ap.add_argument("--edge-threads", type=int, default=0)
+    ap.add_argument("--edge-depth", type=int, default=2,
+                    help="batches in flight in the edge loop (host/GPU overlap)")
     ap.add_argument("--upstreams", type=int, default=64)
     ap.add_argument("--flagged-frac", type=float, default=0.02)
     ap.add_argument("--semcache", action="store_true",

@@ -47,13 +47,18 @@ _STREAM_GET_SUFFIXES = ("/mcp", "/sse", "/events")
 
 class NativeEdge:
     def __init__(self, engine, app=None, auth=None, port: int = 4446,
-                 threads: int = 4, control_url: Optional[str] = None):
+                 threads: int = 4, control_url: Optional[str] = None,
+                 depth: int = 2):
         self.engine = engine
         self.app = app            # FastAPI app for the cold path (optional)
         self.auth = auth          # AuthService; falls back to app.state.auth
         self.port = port
         self.threads = threads
         self.control_url = control_url
+        # batches processed concurrently: with 2 in flight, batch B's C++
+        # envelope/decide host stages overlap batch A's GPU sync (the
+        # pipeline's _gpu_lock serializes only the device sections)
+        self.depth = max(1, depth)
         self.edge = get_edge_module()
         self.handle = 0
         self._task: Optional[asyncio.Task] = None
@@ -139,22 +144,35 @@ class NativeEdge:
         wait_us = 50_000
         linger_us = s.gpu_batch_window_us
         max_n = s.gpu_batch_max_requests
+        sem = asyncio.Semaphore(self.depth)
+
+        async def run_one(batch):
+            try:
+                await self._handle_batch(*batch)
+            except Exception:
+                logger.exception("native edge batch error")
+            finally:
+                sem.release()
+
         while self._running:
             try:
+                await sem.acquire()
                 ids, kinds, bodies, users, authzs, meta = await asyncio.to_thread(
                     self.edge.poll, self.handle, wait_us, linger_us, max_n)
                 n = len(bodies)
                 if n == 0:
+                    sem.release()
                     continue
                 self._check_revocations()
                 self.batches += 1
                 self.requests += n
                 self.max_batch = max(self.max_batch, n)
-                await self._handle_batch(ids, kinds, bodies, users, authzs, meta)
+                asyncio.ensure_future(run_one((ids, kinds, bodies, users, authzs, meta)))
             except asyncio.CancelledError:
                 return
             except Exception:
                 logger.exception("native edge loop error")
+                sem.release()
                 await asyncio.sleep(0.05)
 
     async def _handle_batch(self, ids: bytes, kinds: bytes, bodies, users, authzs, meta) -> None:
