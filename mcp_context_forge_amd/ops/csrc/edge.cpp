@@ -161,6 +161,17 @@ struct EdgeThread {
         ssize_t rc = ::write(evfd, &one, 8);
         (void)rc;
     }
+
+    void wake_with_many(std::vector<uint32_t>&& slots) {
+        if (slots.empty()) return;
+        {
+            std::lock_guard<std::mutex> g(todo_mu);
+            todo.insert(todo.end(), slots.begin(), slots.end());
+        }
+        uint64_t one = 1;
+        ssize_t rc = ::write(evfd, &one, 8);
+        (void)rc;
+    }
 };
 
 uint64_t make_id(int tid, uint32_t slot, uint32_t gen) {
@@ -583,8 +594,15 @@ Edge* edge_start_impl(int port, int nthreads, size_t max_body, bool auth_require
     return e;
 }
 
-// complete one response: returns false if the connection is gone
-bool complete_one(Edge* e, uint64_t id, int status, const char* body, size_t n, bool raw) {
+// complete one response: returns false if the connection is gone.
+// When `deferred_wake` is non-null the response is only appended to the
+// connection's output buffer and the slot recorded per thread — the owning
+// epoll threads then flush everything with ONE eventfd wake per thread,
+// parallelizing the per-connection write() syscalls across the edge
+// threads instead of serializing them on the completer thread (measured:
+// the serial flush was ~10 ms of a 2.6k-row batch at saturation).
+bool complete_one(Edge* e, uint64_t id, int status, const char* body, size_t n, bool raw,
+                  std::vector<std::vector<uint32_t>>* deferred_wake = nullptr) {
     int tid = (int)(id >> 56);
     uint32_t gen = (uint32_t)((id >> 32) & 0xFFFFFF);
     uint32_t slot = (uint32_t)(id & 0xFFFFFFFF);
@@ -600,11 +618,15 @@ bool complete_one(Edge* e, uint64_t id, int status, const char* body, size_t n, 
         else format_response(c->out, status, body, n, c->keep_alive);
         if (!c->keep_alive) c->want_close = true;
         c->inflight--;
-        flush_locked(e, t, c);
-        // epoll thread must take over when: more buffered input to parse,
-        // unflushed output pending, or the conn should close now
-        need_wake = (!c->in.empty() && c->inflight == 0) || c->epollout ||
-                    (c->want_close && c->inflight == 0);
+        if (deferred_wake != nullptr) {
+            (*deferred_wake)[(size_t)tid].push_back(slot);
+        } else {
+            flush_locked(e, t, c);
+            // epoll thread must take over when: more buffered input to
+            // parse, unflushed output pending, or the conn should close
+            need_wake = (!c->in.empty() && c->inflight == 0) || c->epollout ||
+                        (c->want_close && c->inflight == 0);
+        }
     }
     e->responses++;
     if (need_wake) t->wake_with(slot);
@@ -728,8 +750,14 @@ PyObject* py_edge_complete(PyObject*, PyObject* args) {
     }
     uint64_t done = 0;
     Py_BEGIN_ALLOW_THREADS;
-    for (auto& it : items)
-        if (complete_one(e, it.id, it.status, it.body.data(), it.body.size(), false)) done++;
+    {
+        std::vector<std::vector<uint32_t>> wake(e->threads.size());
+        for (auto& it : items)
+            if (complete_one(e, it.id, it.status, it.body.data(), it.body.size(), false, &wake))
+                done++;
+        for (size_t t = 0; t < wake.size(); ++t)
+            e->threads[t]->wake_with_many(std::move(wake[t]));
+    }
     Py_END_ALLOW_THREADS;
     return PyLong_FromUnsignedLongLong(done);
 }
