@@ -62,6 +62,11 @@ class Settings(BaseModel):
     health_check_interval: int = 60
     health_check_timeout: int = 10
     unhealthy_threshold: int = 3
+    # async registration lifecycle (reference: gateway_service.py:4077-4362)
+    gateway_max_retries: int = 8
+    gateway_retry_base_s: float = 2.0
+    gateway_retry_cap_s: float = 300.0
+    gateway_lifecycle_tick_s: float = 1.0
     max_tool_retries: int = 3
     retry_base_delay_ms: int = 100
     retry_max_delay_ms: int = 5000

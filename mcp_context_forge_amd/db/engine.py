@@ -76,11 +76,30 @@ def _rev_0003_plugin_bindings(conn: Connection) -> None:
     Base.metadata.tables["plugin_bindings"].create(conn, checkfirst=True)
 
 
+def _rev_0004_gateway_lifecycle(conn: Connection) -> None:
+    """Async registration lifecycle columns on gateways (retry backoff +
+    failure classification — reference: gateway_service.py:4077-4362,:7469).
+    No-op for fresh DBs (0001 create_all builds the current model)."""
+    if conn.engine.dialect.name == "sqlite":
+        cols = {r[1] for r in conn.exec_driver_sql("PRAGMA table_info(gateways)")}
+        ddl = [("retry_count", "INTEGER DEFAULT 0"), ("next_retry_at", "FLOAT"),
+               ("last_error", "TEXT"), ("failure_class", "VARCHAR(32)")]
+        for name, typ in ddl:
+            if name not in cols:
+                conn.exec_driver_sql(f"ALTER TABLE gateways ADD COLUMN {name} {typ}")
+    else:  # pragma: no cover - postgres path
+        conn.exec_driver_sql("ALTER TABLE gateways ADD COLUMN IF NOT EXISTS retry_count INTEGER DEFAULT 0")
+        conn.exec_driver_sql("ALTER TABLE gateways ADD COLUMN IF NOT EXISTS next_retry_at FLOAT")
+        conn.exec_driver_sql("ALTER TABLE gateways ADD COLUMN IF NOT EXISTS last_error TEXT")
+        conn.exec_driver_sql("ALTER TABLE gateways ADD COLUMN IF NOT EXISTS failure_class VARCHAR(32)")
+
+
 # Linear chain: (revision_id, apply_fn). Append-only.
 MIGRATIONS: List[Tuple[str, Callable[[Connection], None]]] = [
     ("0001_initial_registry", _rev_0001_initial),
     ("0002_tool_metrics_count", _rev_0002_tool_metrics_count),
     ("0003_plugin_bindings", _rev_0003_plugin_bindings),
+    ("0004_gateway_lifecycle", _rev_0004_gateway_lifecycle),
 ]
 
 

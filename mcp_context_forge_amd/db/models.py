@@ -55,6 +55,12 @@ class DbGateway(Base):
     created_at: Mapped[datetime.datetime] = mapped_column(DateTime, default=utcnow)
     updated_at: Mapped[datetime.datetime] = mapped_column(DateTime, default=utcnow, onupdate=utcnow)
     owner_rank: Mapped[int] = mapped_column(Integer, default=0)  # MI355X: GPU rank that owns this upstream shard
+    # async lifecycle state machine (reference: _process_gateway_lifecycle_*
+    # db rows claimed by the loop, gateway_service.py:4077-4362)
+    retry_count: Mapped[int] = mapped_column(Integer, default=0)
+    next_retry_at: Mapped[Optional[float]] = mapped_column(Float, nullable=True)  # epoch seconds
+    last_error: Mapped[Optional[str]] = mapped_column(Text, nullable=True)
+    failure_class: Mapped[Optional[str]] = mapped_column(String(32), nullable=True)
 
 
 class DbTool(Base):

@@ -118,6 +118,9 @@ class GatewayEngine:
     async def startup(self) -> None:
         if self.settings.federation_enabled and self.settings.health_check_interval > 0:
             await self.gateway_service.start_health_loop()
+        if self.settings.federation_enabled:
+            # resume pending/deleting lifecycle rows left from a previous run
+            self.gateway_service.ensure_lifecycle_loop()
         # periodic maintenance (reference: lifespan background tasks — metrics
         # buffer flush :60s, session reaper, span persistence)
         if self._maintenance_task is None:

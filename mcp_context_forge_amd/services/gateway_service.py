@@ -20,7 +20,7 @@ from ..config import Settings
 from ..registry.registry import ConflictError, NotFoundError, Registry
 from ..utils import slugify
 from .tool_service import ToolService
-from .upstream import HttpUpstreamClient, InProcUpstream, UpstreamClient, UpstreamError
+from .upstream import HttpUpstreamClient, InProcUpstream, SseUpstreamClient, UpstreamClient, UpstreamError
 
 logger = logging.getLogger(__name__)
 
@@ -38,6 +38,8 @@ class GatewayService:
         self.rank = rank
         self.world_size = world_size
         self._health_task: Optional[asyncio.Task] = None
+        self._lifecycle_task: Optional[asyncio.Task] = None
+        self._injected_clients: Dict[str, UpstreamClient] = {}  # deferred test/bench clients
         self._stop = asyncio.Event()
         from ..auth.crypto import EncryptionService
 
@@ -65,17 +67,56 @@ class GatewayService:
 
             cfg = auth_value
             token_provider = provider_from_auth_value(_json.loads(cfg) if isinstance(cfg, str) else cfg)
-        return HttpUpstreamClient(gateway["url"], headers=headers,
-                                  timeout=self.settings.federation_timeout,
-                                  token_provider=token_provider)
+        cls = SseUpstreamClient if gateway.get("transport") == "sse" else HttpUpstreamClient
+        return cls(gateway["url"], headers=headers,
+                   timeout=self.settings.federation_timeout,
+                   token_provider=token_provider)
+
+    # -- failure classification (reference: handshake classifier :7469) --------
+    @staticmethod
+    def classify_failure(exc: BaseException) -> str:
+        if isinstance(exc, asyncio.TimeoutError):
+            return "timeout"
+        msg = str(exc).lower()
+        if isinstance(exc, UpstreamError) or "upstream" in msg:
+            if "http 401" in msg or "http 403" in msg:
+                return "auth_error"
+            if "http 4" in msg or "http 5" in msg:
+                return "http_error"
+            if "unreachable" in msg or "connect" in msg:
+                return "connect_error"
+            return "protocol_error"
+        if "connect" in msg or "refused" in msg or "resolve" in msg:
+            return "connect_error"
+        return "internal_error"
+
+    def _retry_backoff(self, retry_count: int) -> float:
+        """Exponential backoff with jitter for pending-gateway retries
+        (reference: _calculate_gateway_retry_backoff :4332)."""
+        import random
+
+        base = self.settings.gateway_retry_base_s
+        cap = self.settings.gateway_retry_cap_s
+        d = min(cap, base * (2 ** max(0, retry_count - 1)))
+        return d * (1.0 + self.settings.retry_jitter * random.random())
 
     # -- registration (reference: register_gateway :1636) -----------------------
     async def register_gateway(self, name: str, url: str, transport: str = "streamablehttp",
                                description: str = "", auth_type: Optional[str] = None,
                                auth_value: Optional[str] = None, tags: Optional[List[str]] = None,
                                client: Optional[UpstreamClient] = None,
-                               owner_rank: Optional[int] = None) -> Dict[str, Any]:
-        """Synchronous-path registration: connect, initialize, sync capabilities.
+                               owner_rank: Optional[int] = None,
+                               defer: bool = False) -> Dict[str, Any]:
+        """Gateway registration — sync or async lifecycle.
+
+        Sync (default): connect, initialize, sync capabilities before
+        returning; a failure raises and marks the row unreachable.
+
+        `defer=True` (reference: _register_gateway_pending :1564): the row
+        is created in status `pending` and returned immediately; the
+        lifecycle loop claims it, attempts initialization with exponential
+        backoff + failure classification, and flips it to `active` when the
+        peer becomes reachable (or `failed` after gateway_max_retries).
 
         `client` lets tests/bench inject an InProcUpstream (fake upstream
         harness) — the sync + health machinery is identical either way.
@@ -90,12 +131,20 @@ class GatewayService:
             auth_value=self.crypto.seal(auth_value),
             tags=tags or [],
             status="pending",
+            retry_count=0,
+            next_retry_at=time.time(),
             owner_rank=owner_rank if owner_rank is not None else (hash(name) % self.world_size),
         )
+        if defer:
+            if client is not None:
+                self._injected_clients[gateway["id"]] = client
+            self.ensure_lifecycle_loop()
+            return gateway
         try:
             await self._initialize_gateway(gateway, client)
         except Exception as exc:
-            self.registry.update("gateway", gateway["id"], status="unreachable", reachable=False)
+            self.registry.update("gateway", gateway["id"], status="unreachable", reachable=False,
+                                 last_error=str(exc)[:500], failure_class=self.classify_failure(exc))
             raise GatewayConnectionError(f"failed to initialize gateway {name}: {exc}") from exc
         return self.registry.get("gateway", gateway["id"])
 
@@ -189,6 +238,71 @@ class GatewayService:
         self.registry.update("gateway", gw["id"], **fields)
         return False
 
+    # -- async lifecycle loop (reference: _run_gateway_lifecycle_loop :4154) --
+    def ensure_lifecycle_loop(self) -> None:
+        if self._lifecycle_task is None or self._lifecycle_task.done():
+            self._stop.clear()
+
+            async def loop() -> None:
+                tick = max(0.05, self.settings.gateway_lifecycle_tick_s)
+                while not self._stop.is_set():
+                    try:
+                        await self.lifecycle_tick()
+                    except Exception:  # pragma: no cover - defensive
+                        logger.exception("gateway lifecycle tick error")
+                    try:
+                        await asyncio.wait_for(self._stop.wait(), timeout=tick)
+                        return
+                    except asyncio.TimeoutError:
+                        pass
+
+            self._lifecycle_task = asyncio.create_task(loop())
+
+    async def lifecycle_tick(self) -> Dict[str, int]:
+        """One pass over claimable rows (reference: _claim_due_gateway_
+        lifecycle_ids :4111 + _process_pending_gateway :4250). Rows are
+        sharded by owner_rank, so claims cannot race across ranks; within a
+        rank the pending→initializing flip guards re-entry."""
+        now = time.time()
+        counts = {"activated": 0, "retried": 0, "failed": 0, "deleted": 0}
+        for gw in list(self.registry.list("gateway", include_disabled=True)):
+            if gw.get("owner_rank", 0) % self.world_size != self.rank:
+                continue
+            status = gw.get("status")
+            if status == "pending" and (gw.get("next_retry_at") or 0) <= now:
+                self.registry.update("gateway", gw["id"], status="initializing")
+                client = self._injected_clients.get(gw["id"])
+                try:
+                    await self._initialize_gateway(gw, client)
+                    self.registry.update("gateway", gw["id"], retry_count=0,
+                                         last_error=None, failure_class=None, next_retry_at=None)
+                    counts["activated"] += 1
+                except Exception as exc:
+                    rc = int(gw.get("retry_count") or 0) + 1
+                    cls = self.classify_failure(exc)
+                    fields = dict(retry_count=rc, last_error=str(exc)[:500], failure_class=cls)
+                    if rc >= self.settings.gateway_max_retries:
+                        # terminal until a manual refresh (reference: failed
+                        # lifecycle rows need operator action)
+                        fields.update(status="failed", reachable=False, next_retry_at=None)
+                        counts["failed"] += 1
+                    else:
+                        fields.update(status="pending", reachable=False,
+                                      next_retry_at=now + self._retry_backoff(rc))
+                        counts["retried"] += 1
+                    self.registry.update("gateway", gw["id"], **fields)
+            elif status == "deleting":
+                await self._finish_delete(gw["id"])
+                counts["deleted"] += 1
+        return counts
+
+    async def retry_failed_gateway(self, gateway_id: str) -> Dict[str, Any]:
+        """Operator action: put a failed row back into the pending lifecycle."""
+        self.registry.update("gateway", gateway_id, status="pending", retry_count=0,
+                             next_retry_at=time.time(), last_error=None, failure_class=None)
+        self.ensure_lifecycle_loop()
+        return self.registry.get("gateway", gateway_id)
+
     async def start_health_loop(self) -> None:
         if self._health_task is not None:
             return
@@ -209,13 +323,15 @@ class GatewayService:
 
     async def stop(self) -> None:
         self._stop.set()
-        if self._health_task:
-            self._health_task.cancel()
-            try:
-                await self._health_task
-            except (asyncio.CancelledError, Exception):
-                pass
-            self._health_task = None
+        for attr in ("_health_task", "_lifecycle_task"):
+            task = getattr(self, attr)
+            if task:
+                task.cancel()
+                try:
+                    await task
+                except (asyncio.CancelledError, Exception):
+                    pass
+                setattr(self, attr, None)
 
     async def refresh_gateway(self, gateway_id: str) -> Dict[str, Any]:
         """Manual refresh (reference: refresh_gateway_manually :6548)."""
@@ -223,13 +339,26 @@ class GatewayService:
         await self._initialize_gateway(gw, self.tools.upstream_for(gateway_id))
         return self.registry.get("gateway", gateway_id)
 
-    async def delete_gateway(self, gateway_id: str) -> None:
-        gw = self.registry.get("gateway", gateway_id)
+    async def delete_gateway(self, gateway_id: str, defer: bool = False) -> None:
+        """Delete a gateway. `defer=True` (reference: deleting lifecycle
+        rows :4077) marks the row `deleting` and lets the lifecycle loop
+        finish the teardown; callers observing the registry see the state."""
+        self.registry.get("gateway", gateway_id)  # raises NotFoundError
         self.registry.update("gateway", gateway_id, status="deleting")
+        if defer:
+            self.ensure_lifecycle_loop()
+            return
+        await self._finish_delete(gateway_id)
+
+    async def _finish_delete(self, gateway_id: str) -> None:
         for t in self.registry.tools_for_gateway(gateway_id):
             self.registry.delete("tool", t["id"])
         client = self.tools.upstream_for(gateway_id)
         if client:
             await client.aclose()
             self.tools._upstreams.pop(gateway_id, None)
-        self.registry.delete("gateway", gateway_id)
+        self._injected_clients.pop(gateway_id, None)
+        try:
+            self.registry.delete("gateway", gateway_id)
+        except NotFoundError:  # pragma: no cover - already gone
+            pass

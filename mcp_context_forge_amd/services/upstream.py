@@ -179,6 +179,142 @@ class HttpUpstreamClient(UpstreamClient):
             await self._client.aclose()
 
 
+class SseUpstreamClient(HttpUpstreamClient):
+    """Legacy SSE transport client (reference: connect_to_sse_server :6900).
+
+    Protocol: GET {url} opens a long-lived text/event-stream; the server's
+    first frame is `event: endpoint` whose data is the message-POST URL;
+    JSON-RPC requests are POSTed there (the POST returns 202) and responses
+    arrive as `event: message` frames on the GET stream, correlated by id.
+    Matches this gateway's own SSE pair (/servers/{id}/sse + /message), so
+    two forges federate over SSE end-to-end.
+    """
+
+    def __init__(self, url: str, headers: Optional[Dict[str, str]] = None, timeout: float = 30.0,
+                 client: Optional[httpx.AsyncClient] = None, token_provider=None):
+        super().__init__(url, headers=headers, timeout=timeout, client=client,
+                         token_provider=token_provider)
+        self._endpoint: Optional[str] = None
+        self._endpoint_ready: Optional[asyncio.Future] = None
+        self._pending: Dict[Any, asyncio.Future] = {}
+        self._reader_task: Optional[asyncio.Task] = None
+        self._connect_lock = asyncio.Lock()
+
+    async def _connect(self) -> None:
+        async with self._connect_lock:
+            if self._reader_task is not None and not self._reader_task.done():
+                return
+            loop = asyncio.get_running_loop()
+            self._endpoint_ready = loop.create_future()
+            self._reader_task = asyncio.create_task(self._read_stream())
+            try:
+                await asyncio.wait_for(asyncio.shield(self._endpoint_ready), timeout=self.timeout)
+            except asyncio.TimeoutError as exc:
+                raise UpstreamError(f"upstream {self.url}: no SSE endpoint event") from exc
+
+    async def _read_stream(self) -> None:
+        headers = {"accept": "text/event-stream"}
+        headers.update(self.base_headers)
+        if self.token_provider is not None:
+            headers["authorization"] = f"Bearer {await self.token_provider.get_token(self._client)}"
+        try:
+            async with self._client.stream("GET", self.url, headers=headers, timeout=None) as resp:
+                if resp.status_code >= 400:
+                    raise UpstreamError(f"upstream {self.url} HTTP {resp.status_code}")
+                event, data_lines = "", []
+                async for line in resp.aiter_lines():
+                    if line.startswith("event:"):
+                        event = line[6:].strip()
+                    elif line.startswith("data:"):
+                        data_lines.append(line[5:].strip())
+                    elif line == "":
+                        if data_lines:
+                            self._dispatch_frame(event, "\n".join(data_lines))
+                        event, data_lines = "", []
+        except Exception as exc:
+            if self._endpoint_ready is not None and not self._endpoint_ready.done():
+                self._endpoint_ready.set_exception(
+                    UpstreamError(f"upstream {self.url} SSE stream failed: {exc}"))
+            for fut in self._pending.values():
+                if not fut.done():
+                    fut.set_exception(UpstreamError(f"upstream {self.url} SSE stream lost: {exc}"))
+            self._pending.clear()
+
+    def _dispatch_frame(self, event: str, data: str) -> None:
+        if event == "endpoint":
+            # relative or absolute message-POST URL
+            self._endpoint = data if "://" in data else str(httpx.URL(self.url).join(data))
+            if self._endpoint_ready is not None and not self._endpoint_ready.done():
+                self._endpoint_ready.set_result(True)
+            return
+        if event in ("message", ""):
+            try:
+                obj = json.loads(data)
+            except Exception:
+                return
+            fut = self._pending.pop(obj.get("id"), None)
+            if fut is not None and not fut.done():
+                fut.set_result(obj)
+
+    async def _rpc(self, method: str, params: Any = None, notification: bool = False,
+                   extra_headers: Optional[Dict[str, str]] = None) -> Any:
+        await self._connect()
+        body: Dict[str, Any] = {"jsonrpc": "2.0", "method": method}
+        if params is not None:
+            body["params"] = params
+        rid = None
+        fut: Optional[asyncio.Future] = None
+        if not notification:
+            rid = next(self._ids)
+            body["id"] = rid
+            fut = asyncio.get_running_loop().create_future()
+            self._pending[rid] = fut
+        headers = {"content-type": "application/json"}
+        headers.update(self.base_headers)
+        if extra_headers:
+            headers.update(extra_headers)
+        if self.token_provider is not None:
+            headers["authorization"] = f"Bearer {await self.token_provider.get_token(self._client)}"
+        try:
+            resp = await self._client.post(self._endpoint, content=json.dumps(body).encode(),
+                                           headers=headers)
+        except httpx.HTTPError as exc:
+            self._pending.pop(rid, None)
+            raise UpstreamError(f"upstream {self.url} unreachable: {exc}") from exc
+        if resp.status_code >= 400:
+            self._pending.pop(rid, None)
+            raise UpstreamError(f"upstream {self.url} HTTP {resp.status_code}")
+        if notification:
+            return None
+        try:
+            data = await asyncio.wait_for(fut, timeout=self.timeout)
+        except asyncio.TimeoutError as exc:
+            self._pending.pop(rid, None)
+            raise UpstreamError(f"upstream {self.url}: SSE response timeout") from exc
+        if "error" in data:
+            err = data["error"]
+            raise UpstreamError(f"upstream error {err.get('code')}: {err.get('message')}",
+                                code=err.get("code", jsonrpc.SERVER_ERROR))
+        return data.get("result")
+
+    async def ping(self) -> bool:
+        try:
+            await self._rpc("ping")
+            return True
+        except UpstreamError:
+            return False
+
+    async def aclose(self) -> None:
+        if self._reader_task is not None:
+            self._reader_task.cancel()
+            try:
+                await self._reader_task
+            except (asyncio.CancelledError, Exception):
+                pass
+            self._reader_task = None
+        await super().aclose()
+
+
 ToolHandler = Callable[[Dict[str, Any]], Awaitable[Any]]
 
 

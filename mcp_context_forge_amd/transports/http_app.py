@@ -515,12 +515,15 @@ def build_app(engine: GatewayEngine, auth: Optional[AuthService] = None) -> Fast
             body = await request.json()
             try:
                 if kind == "gateway":
+                    # defer=true → async lifecycle: 201 with a `pending` row;
+                    # the lifecycle loop initializes it with backoff
+                    # (reference: _register_gateway_pending :1564)
                     return await engine.gateway_service.register_gateway(
                         name=body["name"], url=body.get("url", ""),
                         transport=body.get("transport", "streamablehttp"),
                         description=body.get("description", ""),
                         auth_type=body.get("auth_type"), auth_value=body.get("auth_value"),
-                        tags=body.get("tags"))
+                        tags=body.get("tags"), defer=bool(body.get("defer", False)))
                 if kind == "tool":
                     body.setdefault("original_name", body.get("name"))
                     body.setdefault("integration_type", "REST" if body.get("url") else "LOCAL")
@@ -637,6 +640,14 @@ def build_app(engine: GatewayEngine, auth: Optional[AuthService] = None) -> Fast
             raise HTTPException(404, str(exc)) from exc
         except Exception as exc:
             raise HTTPException(502, str(exc)) from exc
+
+    @app.post("/gateways/{gateway_id}/retry")
+    async def retry_gateway(gateway_id: str, ctx: AuthContext = Depends(require("gateways.update"))):
+        """Put a `failed` lifecycle row back into `pending` (operator action)."""
+        try:
+            return await engine.gateway_service.retry_failed_gateway(gateway_id)
+        except NotFoundError as exc:
+            raise HTTPException(404, str(exc)) from exc
 
     # -- export / import (reference: services/export_service.py:268) --------------
     @app.get("/export")
