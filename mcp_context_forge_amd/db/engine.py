@@ -82,6 +82,9 @@ def _rev_0004_gateway_lifecycle(conn: Connection) -> None:
     No-op for fresh DBs (0001 create_all builds the current model)."""
     if conn.engine.dialect.name == "sqlite":
         cols = {r[1] for r in conn.exec_driver_sql("PRAGMA table_info(gateways)")}
+        if not cols:  # legacy DB without the table: build it from the model
+            Base.metadata.tables["gateways"].create(conn, checkfirst=True)
+            return
         ddl = [("retry_count", "INTEGER DEFAULT 0"), ("next_retry_at", "FLOAT"),
                ("last_error", "TEXT"), ("failure_class", "VARCHAR(32)")]
         for name, typ in ddl:
