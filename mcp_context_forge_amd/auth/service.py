@@ -80,6 +80,26 @@ class AuthService:
         self.token_blocklist = token_blocklist  # reference: services/token_blocklist.py
         self._perm_cache: dict = {}
         self._revocations = 0  # bumped on API-token revocation
+        self._jwks = None      # lazy JWKSet when RS256 is accepted
+
+    def jwks(self):
+        """JWKSet from settings (inline JSON / file / URL), lazily built."""
+        if self._jwks is None:
+            from . import rsa as rsa_mod
+
+            s = self.settings
+            if s.jwks_inline:
+                self._jwks = rsa_mod.JWKSet.from_json(s.jwks_inline)
+            elif s.jwks_file:
+                from pathlib import Path
+
+                self._jwks = rsa_mod.JWKSet.from_json(Path(s.jwks_file).read_text())
+            elif s.jwks_url:
+                self._jwks = rsa_mod.JWKSet(url=s.jwks_url)
+                self._jwks.refresh_from_url()
+            else:
+                self._jwks = rsa_mod.JWKSet()
+        return self._jwks
 
     @property
     def revocation_epoch(self) -> int:
@@ -178,10 +198,13 @@ class AuthService:
                 if ctx is None:
                     raise AuthError("Invalid or revoked API token")
                 return ctx
+            algs = tuple(self.settings.jwt_accepted_algorithms or ["HS256"])
             try:
                 claims = jwt_mod.decode_token(value, self.settings.jwt_secret_key,
                                               audience=self.settings.jwt_audience,
-                                              issuer=self.settings.jwt_issuer)
+                                              issuer=self.settings.jwt_issuer,
+                                              jwks=self.jwks() if "RS256" in algs else None,
+                                              algorithms=algs)
             except jwt_mod.JWTError as exc:
                 raise AuthError(f"Invalid token: {exc}") from exc
             if self.token_blocklist is not None and self.token_blocklist.is_blocked(claims.get("jti")):

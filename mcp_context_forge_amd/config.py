@@ -36,6 +36,12 @@ class Settings(BaseModel):
     basic_auth_password: str = "changeme"
     auth_required: bool = True
     jwt_secret_key: str = "my-test-key"
+    # accepted verification algorithms (allowlist; reference: auth.py RS*/HS*)
+    jwt_accepted_algorithms: List[str] = Field(default_factory=lambda: ["HS256"])
+    # RS256 key material: a JWKS from inline JSON, a file path, or a URL
+    jwks_inline: str = ""
+    jwks_file: str = ""
+    jwks_url: str = ""
     # credential-at-rest sealing key, DISTINCT from the JWT signing key so a
     # leaked signing secret cannot decrypt stored upstream credentials
     # (reference: config auth_encryption_secret, separate from jwt_secret_key)

@@ -97,12 +97,18 @@ def _rev_0004_gateway_lifecycle(conn: Connection) -> None:
         conn.exec_driver_sql("ALTER TABLE gateways ADD COLUMN IF NOT EXISTS failure_class VARCHAR(32)")
 
 
+def _rev_0005_oauth_tokens(conn: Connection) -> None:
+    """Upstream OAuth token storage (reference: token_storage_service.py)."""
+    Base.metadata.tables["oauth_tokens"].create(conn, checkfirst=True)
+
+
 # Linear chain: (revision_id, apply_fn). Append-only.
 MIGRATIONS: List[Tuple[str, Callable[[Connection], None]]] = [
     ("0001_initial_registry", _rev_0001_initial),
     ("0002_tool_metrics_count", _rev_0002_tool_metrics_count),
     ("0003_plugin_bindings", _rev_0003_plugin_bindings),
     ("0004_gateway_lifecycle", _rev_0004_gateway_lifecycle),
+    ("0005_oauth_tokens", _rev_0005_oauth_tokens),
 ]
 
 

@@ -244,6 +244,24 @@ class DbApiToken(Base):
     last_used_at: Mapped[Optional[datetime.datetime]] = mapped_column(DateTime, nullable=True)
 
 
+class DbOAuthToken(Base):
+    """Stored upstream OAuth tokens (reference: token_storage_service.py +
+    db.py:5338 OAuth tables). Token material is sealed with the
+    auth_encryption_secret before it reaches this row."""
+
+    __tablename__ = "oauth_tokens"
+
+    id: Mapped[str] = mapped_column(String(36), primary_key=True, default=_uuid)
+    storage_key: Mapped[str] = mapped_column(String(255), unique=True, index=True)
+    access_token: Mapped[str] = mapped_column(Text)           # sealed
+    refresh_token: Mapped[Optional[str]] = mapped_column(Text, nullable=True)  # sealed
+    token_type: Mapped[str] = mapped_column(String(32), default="Bearer")
+    scopes: Mapped[list] = mapped_column(JSON, default=list)
+    expires_at: Mapped[Optional[float]] = mapped_column(Float, nullable=True)  # epoch seconds
+    created_at: Mapped[datetime.datetime] = mapped_column(DateTime, default=utcnow)
+    updated_at: Mapped[datetime.datetime] = mapped_column(DateTime, default=utcnow, onupdate=utcnow)
+
+
 class DbSessionRecord(Base):
     """Transport session (reference: db.py:5311) — DB backend of the session registry."""
 

@@ -42,10 +42,14 @@ class GatewayService:
         self._injected_clients: Dict[str, UpstreamClient] = {}  # deferred test/bench clients
         self._stop = asyncio.Event()
         from ..auth.crypto import EncryptionService
+        from ..auth.oauth import TokenStorage
 
         # credential material is sealed at rest (reference: EncryptedText db.py:277)
         # keyed by auth_encryption_secret, NOT the JWT signing key
         self.crypto = EncryptionService(self.settings.auth_encryption_secret)
+        # upstream OAuth tokens (auth-code flow) persist sealed in the DB
+        self.token_storage = TokenStorage(self.registry.db, self.crypto)
+        self._oauth_providers: Dict[str, Any] = {}  # gateway_id -> provider
 
     # -- client construction ----------------------------------------------------
     def _make_client(self, gateway: Dict[str, Any]) -> UpstreamClient:
@@ -62,11 +66,20 @@ class GatewayService:
             except Exception:
                 pass
         elif gateway.get("auth_type") == "oauth" and auth_value:
-            # client-credentials upstream auth (reference: oauth_manager.py)
+            # client-credentials OR authorization-code upstream auth
+            # (reference: oauth_manager.py both flows)
             from ..auth.oauth import provider_from_auth_value
 
             cfg = auth_value
-            token_provider = provider_from_auth_value(_json.loads(cfg) if isinstance(cfg, str) else cfg)
+            gid = gateway.get("id", gateway.get("name", ""))
+            token_provider = self._oauth_providers.get(gid)
+            if token_provider is None:
+                token_provider = provider_from_auth_value(
+                    _json.loads(cfg) if isinstance(cfg, str) else cfg,
+                    storage=self.token_storage,
+                    storage_key=f"gateway:{gid}",
+                    state_secret=self.settings.jwt_secret_key)
+                self._oauth_providers[gid] = token_provider
         cls = SseUpstreamClient if gateway.get("transport") == "sse" else HttpUpstreamClient
         return cls(gateway["url"], headers=headers,
                    timeout=self.settings.federation_timeout,
@@ -295,6 +308,44 @@ class GatewayService:
                 await self._finish_delete(gw["id"])
                 counts["deleted"] += 1
         return counts
+
+    # -- upstream OAuth authorization-code flow (reference: oauth_router +
+    # oauth_manager.py token exchange for upstreams) ------------------------
+    def oauth_provider_for(self, gateway_id: str):
+        from ..auth.oauth import AuthorizationCodeProvider, OAuthError, provider_from_auth_value
+
+        prov = self._oauth_providers.get(gateway_id)
+        if prov is None:
+            import json as _json
+
+            gw = self.registry.get("gateway", gateway_id)
+            if gw.get("auth_type") != "oauth":
+                raise OAuthError(f"gateway {gateway_id} does not use OAuth auth")
+            cfg = self.crypto.open_(gw.get("auth_value"))
+            prov = provider_from_auth_value(
+                _json.loads(cfg) if isinstance(cfg, str) else (cfg or {}),
+                storage=self.token_storage, storage_key=f"gateway:{gateway_id}",
+                state_secret=self.settings.jwt_secret_key)
+            self._oauth_providers[gateway_id] = prov
+        if not isinstance(prov, AuthorizationCodeProvider):
+            raise OAuthError("gateway uses client-credentials (no browser flow needed)")
+        return prov
+
+    def begin_upstream_authorization(self, gateway_id: str,
+                                     redirect_uri: Optional[str] = None) -> Dict[str, str]:
+        return self.oauth_provider_for(gateway_id).begin_authorization(redirect_uri)
+
+    async def complete_upstream_authorization(self, code: str, state: str,
+                                              redirect_uri: Optional[str] = None) -> str:
+        """Callback half: state carries the storage key `gateway:{id}`."""
+        from ..auth.oauth import verify_state
+
+        payload = verify_state(state, self.settings.jwt_secret_key)
+        key = payload.get("k", "")
+        gateway_id = key.split(":", 1)[1] if key.startswith("gateway:") else key
+        prov = self.oauth_provider_for(gateway_id)
+        await prov.complete_authorization(code, state, redirect_uri=redirect_uri)
+        return gateway_id
 
     async def retry_failed_gateway(self, gateway_id: str) -> Dict[str, Any]:
         """Operator action: put a failed row back into the pending lifecycle."""

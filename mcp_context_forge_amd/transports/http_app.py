@@ -641,6 +641,38 @@ def build_app(engine: GatewayEngine, auth: Optional[AuthService] = None) -> Fast
         except Exception as exc:
             raise HTTPException(502, str(exc)) from exc
 
+    @app.post("/gateways/{gateway_id}/oauth/authorize")
+    async def gateway_oauth_authorize(gateway_id: str, request: Request,
+                                      ctx: AuthContext = Depends(require("gateways.update"))):
+        """Start the upstream authorization-code flow: returns the provider
+        authorize URL + signed state for the operator's browser
+        (reference: oauth_router initiate)."""
+        from ..auth.oauth import OAuthError
+
+        body = {}
+        if await request.body():
+            body = await request.json()
+        try:
+            return engine.gateway_service.begin_upstream_authorization(
+                gateway_id, redirect_uri=body.get("redirect_uri"))
+        except NotFoundError as exc:
+            raise HTTPException(404, str(exc)) from exc
+        except OAuthError as exc:
+            raise HTTPException(422, str(exc)) from exc
+
+    @app.get("/oauth/upstream/callback")
+    async def gateway_oauth_callback(code: str, state: str):
+        """Provider redirect target: exchanges the code, stores tokens
+        sealed (reference: oauth_router callback). Unauthenticated by
+        design — the signed state is the proof of initiation."""
+        from ..auth.oauth import OAuthError
+
+        try:
+            gid = await engine.gateway_service.complete_upstream_authorization(code, state)
+        except OAuthError as exc:
+            raise HTTPException(400, str(exc)) from exc
+        return {"status": "authorized", "gateway_id": gid}
+
     @app.post("/gateways/{gateway_id}/retry")
     async def retry_gateway(gateway_id: str, ctx: AuthContext = Depends(require("gateways.update"))):
         """Put a `failed` lifecycle row back into `pending` (operator action)."""
