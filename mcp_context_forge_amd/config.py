@@ -73,6 +73,10 @@ class Settings(BaseModel):
     gateway_retry_base_s: float = 2.0
     gateway_retry_cap_s: float = 300.0
     gateway_lifecycle_tick_s: float = 1.0
+    # DB leader lease for shared-DB multi-process deployments (services/leader.py);
+    # collective (torchrun) worlds use rank 0 by construction instead
+    leader_election_enabled: bool = False
+    leader_lease_ttl_s: float = 15.0
     max_tool_retries: int = 3
     retry_base_delay_ms: int = 100
     retry_max_delay_ms: int = 5000

@@ -102,6 +102,14 @@ def _rev_0005_oauth_tokens(conn: Connection) -> None:
     Base.metadata.tables["oauth_tokens"].create(conn, checkfirst=True)
 
 
+def _rev_0006_leader_leases(conn: Connection) -> None:
+    """DB leader lease for shared-DB multi-process deployments
+    (reference: Redis SET NX leader, gateway_service.py:1254/:5272)."""
+    conn.exec_driver_sql(
+        "CREATE TABLE IF NOT EXISTS leader_leases ("
+        "name VARCHAR(128) PRIMARY KEY, holder VARCHAR(255), expires_at FLOAT)")
+
+
 # Linear chain: (revision_id, apply_fn). Append-only.
 MIGRATIONS: List[Tuple[str, Callable[[Connection], None]]] = [
     ("0001_initial_registry", _rev_0001_initial),
@@ -109,6 +117,7 @@ MIGRATIONS: List[Tuple[str, Callable[[Connection], None]]] = [
     ("0003_plugin_bindings", _rev_0003_plugin_bindings),
     ("0004_gateway_lifecycle", _rev_0004_gateway_lifecycle),
     ("0005_oauth_tokens", _rev_0005_oauth_tokens),
+    ("0006_leader_leases", _rev_0006_leader_leases),
 ]
 
 

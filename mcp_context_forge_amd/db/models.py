@@ -6,8 +6,9 @@ A2AAgent :4901, metrics :2564-2851, SessionRecord :5311, EmailUser :1465,
 Role :1154, tokens :5338-5697, StructuredLogEntry :6141, audit :6624) on
 SQLAlchemy 2.0, trimmed to the columns the runtime actually consults.
 Durability lives here (SQLite/Postgres); the *hot* lookup structures are
-mirrored into HBM-resident tensors by the GPU registry
-(:mod:`mcp_context_forge_amd.registry.hbm_mirror`).
+the in-memory registry caches plus the GPU pipeline's native toolmap and
+per-tool flag tables (gpu/pipeline.py:_rebuild_tool_meta); HBM holds the
+scan banks, classifier weights and semantic-cache key matrix.
 """
 
 from __future__ import annotations

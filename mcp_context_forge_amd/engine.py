@@ -113,9 +113,19 @@ class GatewayEngine:
         self._log_level = "info"
         self._maintenance_task = None
         self._maintenance_stop = None
+        self.leader_elector = None  # DbLeaderElector when leader_election_enabled
 
     # -- lifecycle ---------------------------------------------------------------
     async def startup(self) -> None:
+        if self.settings.leader_election_enabled:
+            # shared-DB multi-process deployment: background singletons
+            # (health + lifecycle loops) run only on the lease holder
+            # (reference: Redis leader election :1254; services/leader.py)
+            from .services.leader import DbLeaderElector
+
+            self.leader_elector = DbLeaderElector(self.db, ttl_s=self.settings.leader_lease_ttl_s)
+            await self.leader_elector.start()
+            self.gateway_service.leader_check = lambda: self.leader_elector.is_leader
         if self.settings.federation_enabled and self.settings.health_check_interval > 0:
             await self.gateway_service.start_health_loop()
         if self.settings.federation_enabled:
@@ -152,6 +162,9 @@ class GatewayEngine:
             except (asyncio.CancelledError, Exception):
                 pass
             self._maintenance_task = None
+        if self.leader_elector is not None:
+            await self.leader_elector.stop()
+            self.leader_elector = None
         await self.gateway_service.stop()
         await self.a2a_service.aclose()
         await self.llm_proxy.aclose()
@@ -422,9 +435,7 @@ class GatewayEngine:
                   "prompt": "notifications/prompts/list_changed"}.get(kind)
         if method is None:
             return
-        msg = {"jsonrpc": "2.0", "method": method}
-        for sid in list(self.sessions._sessions.keys()):
-            await self.sessions.broadcast(sid, msg)
+        await self.sessions.broadcast_all({"jsonrpc": "2.0", "method": method})
 
     # -- health/version ------------------------------------------------------------
     def health(self) -> Dict[str, Any]:

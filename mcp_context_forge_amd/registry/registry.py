@@ -3,9 +3,14 @@
 Reference analogs: mcpgateway/services/{tool,resource,prompt,server}_service
 CRUD + mcpgateway/cache/{tool_lookup_cache,registry_cache}. The reference
 caches lookups in Redis; here the hot name→tool mapping is an in-process
-dict mirrored into HBM by the GPU pipeline (registry.hbm_mirror), and the
-DB is durability only — the MI355X design from BASELINE.json ("tool
-registry ... live in HBM instead of Redis").
+dict plus, when the GPU pipeline is attached, a native C++ toolmap and
+flat per-tool flag tables rebuilt on every registry generation bump
+(gpu/pipeline.py:_rebuild_tool_meta) — name resolution for a whole
+micro-batch is one C call, never a DB or network hop. The DB is
+durability only. (What lives in HBM proper: the DFA scan banks, the
+classifier weights and the semantic-cache key matrix — the decision
+tables stay host-side C++ because the per-batch decision pass is
+latency-, not bandwidth-, bound.)
 
 All methods are synchronous and thread-safe; the asyncio layer calls them
 directly (they only touch memory + short sqlite transactions).

@@ -39,6 +39,10 @@ class GatewayService:
         self.world_size = world_size
         self._health_task: Optional[asyncio.Task] = None
         self._lifecycle_task: Optional[asyncio.Task] = None
+        # shared-DB multi-process mode: background singletons gate on the
+        # DB lease (services/leader.py); None = this process is the leader
+        # (single process, or rank 0 of a collective world by construction)
+        self.leader_check: Optional[Any] = None
         self._injected_clients: Dict[str, UpstreamClient] = {}  # deferred test/bench clients
         self._stop = asyncio.Event()
         from ..auth.crypto import EncryptionService
@@ -210,6 +214,8 @@ class GatewayService:
 
     # -- health loop (reference: check_health_of_gateways :4412) ----------------
     async def check_health_once(self, concurrency: int = 16) -> Dict[str, bool]:
+        if self.leader_check is not None and not self.leader_check():
+            return {}  # follower: the lease holder runs the checks
         sem = asyncio.Semaphore(concurrency)
         results: Dict[str, bool] = {}
 
@@ -278,6 +284,8 @@ class GatewayService:
         rank the pending→initializing flip guards re-entry."""
         now = time.time()
         counts = {"activated": 0, "retried": 0, "failed": 0, "deleted": 0}
+        if self.leader_check is not None and not self.leader_check():
+            return counts  # follower: lifecycle rows belong to the leader
         for gw in list(self.registry.list("gateway", include_disabled=True)):
             if gw.get("owner_rank", 0) % self.world_size != self.rank:
                 continue

@@ -164,6 +164,20 @@ class SessionRegistry:
     def count(self) -> int:
         return len(self._sessions)
 
+    def session_ids(self) -> list:
+        """Public snapshot of live session ids (stable across backends —
+        callers must not reach into the internal map)."""
+        return list(self._sessions.keys())
+
+    async def broadcast_all(self, message: dict) -> int:
+        """Deliver a message to every live session; returns delivery count
+        (reference: notification fan-out to all listeners)."""
+        n = 0
+        for sid in self.session_ids():
+            if await self.broadcast(sid, message):
+                n += 1
+        return n
+
     async def broadcast(self, session_id: str, message: dict) -> bool:
         """Deliver a message to the transport holding `session_id`
         (reference: session_registry.broadcast :1271)."""
