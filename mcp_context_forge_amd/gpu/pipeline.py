@@ -21,11 +21,25 @@ exact per-request semantics (the plugins' own functions), so the parity
 gate (tests/test_gpu_parity.py, reference analog
 tests/live_gateway/mcp/test_mcp_plugin_parity.py) holds by construction.
 
-Scanning operates on the request's RAW argument bytes (and the CPU oracle
-on the canonical sorted-JSON text): the two representations contain the
-same string values and the same token multiset, so detection agrees except
-for matches spanning value boundaries — rewrite-class flags only route to
-the exact host path, and parity traffic uses compact payloads.
+Raw-bytes-vs-decoded-text containment (tested invariant —
+tests/test_parity_fuzz.py runs adversarial generators against the CPU
+chain):
+
+  * literal deny/harm words: a RAW match implies a DECODED match (words
+    are alphanumeric; no word can span JSON structural separators), so
+    raw-scan blocks are sound;
+  * content that can decode DIFFERENTLY from its raw bytes (JSON escapes,
+    non-ASCII) always trips the escape-trigger bank, which routes the row
+    to the exact host path where deny (pre-rewrite, CPU order), rewrites,
+    harm, moderation and schema run over decoded text;
+  * schema-shape fast patterns cannot false-positive from key lookalikes
+    inside string values (embedded quotes are escaped, which breaks the
+    byte pattern) and any nesting defers to exact host validation;
+  * the one intentional approximation: the moderation classifier featurizes
+    raw bytes for rows with NO escape triggers — wire forms that differ
+    from canonical text only by whitespace/key order can score differently
+    near the threshold. Rows with triggers are re-scored over canonical
+    text on the host path.
 """
 
 from __future__ import annotations
@@ -109,7 +123,14 @@ class GpuPluginPipeline:
             tables = p.scan_tables() if p is not None and hasattr(p, "scan_tables") else None
             if tables is not None:
                 self.banks[name] = hip.DeviceScanTables(tables, device)
-        if self.normalizer is not None:
+        # the escape/normalize trigger bank routes rows whose RAW bytes may
+        # decode to different text (JSON escapes, non-ASCII) onto the exact
+        # host path. It must exist whenever ANY content bank scans raw bytes
+        # — not only when the normalizer plugin is enabled — otherwise an
+        # escape-hidden deny/harm word (e.g. "forbidden") would slip
+        # past the raw-byte scans with no host recheck (parity-fuzz finding)
+        if any(p is not None for p in (self.normalizer, self.deny, self.harm,
+                                       self.pii, self.regex, self.moderation)):
             self.banks["normalize"] = hip.DeviceScanTables(
                 dfa.compile_patterns(
                     ["".join(ch if ch.isalnum() else "\\" + ch for ch in t) if not t.startswith("[") else t
@@ -364,7 +385,14 @@ class GpuPluginPipeline:
                 f |= hip.TF_PII
             if self._active(self.regex, m.name, block_class=False):
                 f |= hip.TF_REGEX
-            if self._active(self.normalizer, m.name, block_class=False):
+            # TF_NORM = "escape triggers route this tool to the host path":
+            # needed when the normalizer applies OR any raw-byte content
+            # bank is active (their exact semantics are over DECODED text)
+            if self._active(self.normalizer, m.name, block_class=False) or \
+               self._active(self.deny, m.name, block_class=True) or \
+               self._active(self.harm, m.name, block_class=True) or \
+               self._active(self.pii, m.name, block_class=False) or \
+               self._active(self.moderation, m.name, block_class=True):
                 f |= hip.TF_NORM
             if self._active(self.moderation, m.name, block_class=True):
                 f |= hip.TF_MOD
@@ -804,11 +832,11 @@ class GpuPluginPipeline:
         this row means its plugin is an identity transform here (the DFA
         banks are conservative supersets of the host matchers), so it can
         be skipped without changing the outcome."""
-        if do_norm and self.normalizer is not None and self._applies(self.normalizer, name):
+        if do_norm and self._active(self.normalizer, name, block_class=False):
             args = _walk_strings(args, self.normalizer.norm)
-        if do_regex and self.regex is not None and self._applies(self.regex, name):
+        if do_regex and self._active(self.regex, name, block_class=False):
             args = _walk_strings(args, self.regex.apply_rules)
-        if do_pii and self.pii is not None and self._applies(self.pii, name):
+        if do_pii and self._active(self.pii, name, block_class=False):
             found: List[str] = []
 
             def fn(s: str) -> str:
@@ -830,6 +858,8 @@ class GpuPluginPipeline:
     async def _rewrite_pass(self, blob, env, rows, args_b, args_e, rewrite_js, tool_idx,
                             hit, hit_slot, responses,
                             pii_m=None, regex_m=None, norm_m=None) -> List[Tuple[int, Any]]:
+        from ..plugins.builtin import _text_of
+
         ok_items: List[Tuple[int, Any]] = []
         for j in rewrite_js:
             mt = self._meta_list[tool_idx[j]]
@@ -840,6 +870,20 @@ class GpuPluginPipeline:
             except Exception:
                 responses[r] = self._splice_error(idb, jsonrpc.INVALID_PARAMS, "invalid arguments")
                 continue
+            # deny recheck over DECODED text, pre-rewrite (CPU chain order:
+            # deny@10 before normalizer@15) — the raw-byte scan misses
+            # escape-hidden words (forbidden), which is exactly what
+            # routed this row here (parity-fuzz finding)
+            if self.deny is not None and self._active(self.deny, mt.name, block_class=True):
+                hay = _text_of(args)
+                hay = hay.lower() if self.deny.case_insensitive else hay
+                word = next((w for w in self.deny.words
+                             if (w.lower() if self.deny.case_insensitive else w) in hay), None)
+                if word is not None:
+                    responses[r] = self._splice_error(
+                        idb, jsonrpc.POLICY_DENIED, f"deny_filter: deny word {word!r} present")
+                    self.blocked += 1
+                    continue
             status, payload = self._apply_rewrites(
                 mt.name, args,
                 do_norm=bool(norm_m[j]) if norm_m is not None else True,
@@ -928,6 +972,18 @@ class GpuPluginPipeline:
             nb_ = np.where(ne_ > nb_, nb_, -1).astype(np.int32)
             now_iso = time.strftime("%Y-%m-%dT%H:%M:%SZ", time.gmtime())
             nat_blob, nat_beg, nat_end = hip.upstream_call_batch(blob, nb_, ne_, kinds, now_iso)
+            # empty spans = C++ punted a non-canonical args span (escapes,
+            # floats, whitespace formatting); answer those rows on the exact
+            # Python path so results match the CPU reference byte-for-byte
+            punted = np.nonzero((nat_end == nat_beg) & (nb_ >= 0))[0]
+            if punted.size:
+                pm = set(int(x) for x in punted)
+                py_items = py_items + [(native_js[i], None) for i in pm]
+                keep = np.ones(len(native_js), dtype=bool)
+                keep[list(pm)] = False
+                native_js = [j for i, j in enumerate(native_js) if i not in pm]
+                nat_beg = np.ascontiguousarray(nat_beg[keep])
+                nat_end = np.ascontiguousarray(nat_end[keep])
 
         # --- python dispatch (non-native upstreams / rewritten args) ---
         py_results: List[Optional[bytes]] = []
@@ -977,7 +1033,11 @@ class GpuPluginPipeline:
                     kinds2, now_iso2)
                 outs2 = self._pb.slices_list(ob2, ob2b, ob2e)
                 for (idx, _a, _k), rbytes in zip(nat2, outs2):
-                    py_results[idx] = rbytes
+                    if rbytes:
+                        py_results[idx] = rbytes
+                    else:
+                        # canonical-gate punt on the redispatch too → exact path
+                        seq.append((idx, py_items[idx][0], py_items[idx][1]))
             for idx, j, args2 in seq:
                 await one(idx, j, args2)
             if conc:
@@ -1157,9 +1217,9 @@ class GpuPluginPipeline:
         except Exception:
             return rb, True, None
         name = mt.name
-        if self.regex is not None and self._applies(self.regex, name):
+        if self._active(self.regex, name, block_class=False):
             result = _walk_strings(result, self.regex.apply_rules)
-        if self.pii is not None and self._applies(self.pii, name):
+        if self._active(self.pii, name, block_class=False):
             found: List[str] = []
 
             def fn(s: str) -> str:

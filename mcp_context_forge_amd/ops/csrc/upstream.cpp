@@ -100,6 +100,42 @@ void append_span(std::string& out, const uint8_t* b, const uint8_t* e) {
     out.append((const char*)b, (size_t)(e - b));
 }
 
+// Strict-canonical gate for the raw-span fast path. Handlers copy argument
+// VALUE bytes verbatim into results; that is only byte-identical to the
+// CPU reference (json.loads -> json.dumps, compact, ensure_ascii) when the
+// span already IS that canonical form. Anything else — whitespace outside
+// strings, any backslash escape, non-ASCII bytes, non-integer numbers —
+// returns false and the row is answered on the exact host path instead
+// (parity-fuzz finding: pretty-printed args echoed verbatim diverged).
+bool is_canonical_span(const uint8_t* b, const uint8_t* e) {
+    const uint8_t* p = b;
+    bool in_str = false;
+    while (p < e) {
+        uint8_t c = *p;
+        if (c == '\\') return false;     // escapes decode differently
+        if (c >= 0x80) return false;     // ensure_ascii would re-escape
+        if (in_str) {
+            if (c == '"') in_str = false;
+            ++p;
+            continue;
+        }
+        if (c == '"') { in_str = true; ++p; continue; }
+        if (c == ' ' || c == '\t' || c == '\n' || c == '\r') return false;
+        if (c == '-' || (c >= '0' && c <= '9')) {  // number: integers only
+            if (c == '-') ++p;
+            while (p < e && *p >= '0' && *p <= '9') ++p;
+            if (p < e && (*p == '.' || *p == 'e' || *p == 'E')) return false;
+            continue;
+        }
+        if (c == 't') { if (e - p < 4 || memcmp(p, "true", 4) != 0) return false; p += 4; continue; }
+        if (c == 'f') { if (e - p < 5 || memcmp(p, "false", 5) != 0) return false; p += 5; continue; }
+        if (c == 'n') { if (e - p < 4 || memcmp(p, "null", 4) != 0) return false; p += 4; continue; }
+        if (c == '{' || c == '}' || c == '[' || c == ']' || c == ':' || c == ',') { ++p; continue; }
+        return false;
+    }
+    return !in_str;
+}
+
 }  // namespace
 
 // Batch upstream call.
@@ -119,6 +155,10 @@ static void upstream_rows(
         const uint8_t* ab = args_beg[r] >= 0 ? data + args_beg[r] : EMPTY;
         const uint8_t* ae = args_beg[r] >= 0 ? data + args_end[r] : EMPTY + 2;
         int k = kinds[r];
+        if (!is_canonical_span(ab, ae)) {
+            res_end[r] = res_beg[r];  // empty span = punt to the host path
+            continue;
+        }
         if (k == 0) {  // convert_time
             const uint8_t *tb, *te, *sb, *se, *gb, *ge;
             bool ht = find_key(ab, ae, "time", &tb, &te);

@@ -464,7 +464,28 @@ class NativeInProcUpstream(UpstreamClient):
             raise UpstreamError(f"tool {name} not found on upstream {self.name}", code=jsonrpc.METHOD_NOT_FOUND)
         now = time.strftime("%Y-%m-%dT%H:%M:%SZ", time.gmtime())
         raw = json.dumps(arguments or {}, separators=(",", ":")).encode()
-        return json.loads(self.call_tool_raw(kind, raw, now))
+        out = self.call_tool_raw(kind, raw, now)
+        if not out:
+            # non-canonical span (escapes/floats/non-ASCII): the C++ fast
+            # path punts — compute the identical result in Python
+            return self._python_result(kind, arguments or {}, now)
+        return json.loads(out)
+
+    def _python_result(self, kind: int, args: Dict[str, Any], now_iso: str) -> Dict[str, Any]:
+        """Exact Python mirror of the C++ handlers for punted spans."""
+        if kind == 0:
+            sc = {"time": args.get("time", "1970-01-01T00:00:00Z"),
+                  "source_timezone": args.get("source_timezone", "UTC"),
+                  "target_timezone": args.get("target_timezone", "UTC"),
+                  "converted": True}
+            return {"content": [{"type": "text", "text": "converted"}],
+                    "structuredContent": sc, "isError": False}
+        if kind == 1:
+            return {"content": [{"type": "text", "text": now_iso}],
+                    "structuredContent": {"time": now_iso, "timezone": args.get("timezone", "UTC")},
+                    "isError": False}
+        return {"content": [{"type": "text", "text": json.dumps(args, separators=(",", ":"))}],
+                "structuredContent": args, "isError": False}
 
     async def ping(self) -> bool:
         return True
