@@ -1102,6 +1102,60 @@ def build_app(engine: GatewayEngine, auth: Optional[AuthService] = None) -> Fast
 
         return render_admin_page(engine)
 
+    # -- admin UI partials (reference: admin.py HTMX partial endpoints) -----
+    @app.get("/admin/ui/{partial}", response_class=HTMLResponse)
+    async def admin_ui_partial(partial: str, q: str = "", level: str = "",
+                               ctx: AuthContext = Depends(require("admin.read"))):
+        from ..admin import ui as adm
+
+        kind_map = {v: k for k, v in adm._ENTITY_PLURAL.items()}
+        if partial == "dashboard":
+            return adm.partial_dashboard(engine)
+        if partial in kind_map:
+            return adm.partial_entities(engine, kind_map[partial])
+        if partial == "plugins":
+            return adm.partial_plugins(engine)
+        if partial == "bindings":
+            return adm.partial_bindings(engine)
+        if partial == "metrics":
+            return adm.partial_metrics(engine)
+        if partial == "logs":
+            return adm.partial_logs(engine, q=q, level=level)
+        if partial == "traces":
+            return adm.partial_traces(engine)
+        if partial == "audit":
+            return adm.partial_audit(engine)
+        if partial == "tokens":
+            return adm.partial_tokens(engine, auth)
+        if partial == "runtime":
+            ne = getattr(app.state, "native_edge", None)
+            return adm.partial_runtime(engine, collector=app.state.collector,
+                                       edge_stats=ne.stats() if ne else None)
+        raise HTTPException(404, f"unknown partial {partial!r}")
+
+    @app.post("/admin/ui/bind")
+    async def admin_ui_bind(request: Request, ctx: AuthContext = Depends(require("tools.update"))):
+        body = await request.json()
+        if engine.plugins.get(body.get("plugin_name", "")) is None:
+            raise HTTPException(404, f"plugin {body.get('plugin_name')!r} not registered")
+        engine.set_plugin_binding(body["tool_name"], body["plugin_name"],
+                                  mode=body.get("mode") or None)
+        engine.audit.record(ctx.user, "bind", "plugin_binding",
+                            f"{body['tool_name']}::{body['plugin_name']}")
+        return {"status": "bound"}
+
+    @app.post("/admin/ui/runtime")
+    async def admin_ui_runtime(request: Request, ctx: AuthContext = Depends(require("admin.write"))):
+        body = await request.json()
+        col = app.state.collector
+        if col is not None:
+            if body.get("window_us"):
+                col.window_s = max(0, int(body["window_us"])) / 1e6
+            if body.get("max_batch"):
+                col.max_batch = max(1, int(body["max_batch"]))
+        engine.audit.record(ctx.user, "update", "runtime", None, detail=body)
+        return {"status": "applied"}
+
     return app
 
 
