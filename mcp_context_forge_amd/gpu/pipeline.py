@@ -1016,7 +1016,11 @@ class GpuPluginPipeline:
                     # upstream instead of per-request Python dispatch
                     nat2.append((idx, json.dumps(args2, separators=(",", ":"), default=str).encode(),
                                  mt.native_kind))
-                elif mt.itype == "LOCAL" or isinstance(mt.client, InProcUpstream) or mt.native_kind >= 0:
+                elif (mt.itype == "LOCAL" and not (mt.tool.get("annotations") or {}).get("io")) \
+                        or isinstance(mt.client, InProcUpstream) or mt.native_kind >= 0:
+                    # pure in-proc handlers: sequential await beats gather
+                    # overhead; IO-backed LOCAL tools (annotations.io, e.g.
+                    # gRPC-translated methods) need real concurrency
                     seq.append((idx, j, args2))
                 else:
                     conc.append((idx, j, args2))

@@ -295,7 +295,8 @@ class GrpcToMcpTranslator:
                 return await self.call_tool(_s, _m, args)
 
             out.append(tool_service.register_local_tool(
-                td["name"], handler, td["description"], input_schema=td["inputSchema"]))
+                td["name"], handler, td["description"], input_schema=td["inputSchema"],
+                annotations={"io": True}))  # real network hop → batch dispatch concurrently
         return out
 
     def close(self) -> None:
