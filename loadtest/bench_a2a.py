@@ -67,14 +67,15 @@ async def main() -> None:
     for a in range(args.agents):
         model = f"forge-model-{a}"
 
-        async def handler(payload, _m=model):
-            msg = payload.get("message", "") if isinstance(payload, dict) else str(payload)
-            return {
+        async def handler(message, context, _m=model):
+            # OpenAI-compat completion payload, serialized (the agent reply
+            # is text per the A2A message contract)
+            return json.dumps({
                 "id": "chatcmpl-bench", "object": "chat.completion", "model": _m,
                 "choices": [{"index": 0, "finish_reason": "stop",
                              "message": {"role": "assistant",
-                                         "content": f"[{_m}] routed reply to: {msg[:96]}"}}],
-            }
+                                         "content": f"[{_m}] routed reply to: {message[:96]}"}}],
+            }, separators=(",", ":"))
 
         engine.a2a_service.register_local_agent(f"agent{a}", handler,
                                                 description="OpenAI-compat router",

@@ -80,7 +80,7 @@ _TYPED_PAT = {
 class _ToolMeta:
     __slots__ = ("tool", "name", "tid", "itype", "thash", "native_kind", "native_client",
                  "client", "handler", "schema_mode", "required_bits", "typed_pairs",
-                 "has_output_schema", "original_name", "reachable", "host_chain")
+                 "has_output_schema", "original_name", "reachable", "host_chain", "a2a_fast")
 
     def __init__(self):
         self.schema_mode = "host"   # "trivial" | "fast" | "host"
@@ -92,6 +92,7 @@ class _ToolMeta:
         self.handler = None
         self.has_output_schema = False
         self.host_chain = False     # per-tool plugin binding forces the CPU chain
+        self.a2a_fast = False       # in-proc agent, hooks fully GPU-covered
 
 
 class GpuPluginPipeline:
@@ -301,6 +302,33 @@ class GpuPluginPipeline:
             m.reachable = tool.get("reachable", True)
             if m.itype == "LOCAL":
                 m.handler = ts._local_handlers.get(m.name)
+            elif m.itype == "A2A":
+                # A2A fast lane (BASELINE config 4): an in-proc agent whose
+                # agent-hook plugins are ALL GPU-covered banks with no
+                # conditions/bindings can skip the per-row Python hook chain
+                # — the decide pass applied the same deny/harm/moderation
+                # blocks over the same text (args dict incl. the message
+                # wrapper), and pass 3 + the host post chain cover the
+                # response-side pii/regex/harm hooks
+                a2a = getattr(self.engine, "a2a_service", None)
+                agent = self.engine.registry.find("a2a_agent", m.original_name)
+                handler = a2a._local_handlers.get(m.original_name) if a2a else None
+                if agent is not None and agent.get("enabled", True) and handler is not None \
+                        and str(agent.get("endpoint_url", "")).startswith("inproc://"):
+                    from ..plugins.framework import HookType as _HT
+
+                    bank_set = {p.name for p in (self.deny, self.pii, self.regex, self.normalizer,
+                                                 self.moderation, self.harm) if p is not None}
+                    agent_hooked = {p.name for p in self.engine.plugins.plugins
+                                    if p.mode != PluginMode.DISABLED and not p.conditions
+                                    and (_HT.AGENT_PRE_INVOKE in p.hooks or _HT.AGENT_POST_INVOKE in p.hooks)}
+                    conditioned = any(p.conditions and
+                                      (_HT.AGENT_PRE_INVOKE in p.hooks or _HT.AGENT_POST_INVOKE in p.hooks)
+                                      for p in self.engine.plugins.plugins if p.mode != PluginMode.DISABLED)
+                    if agent_hooked <= bank_set and not conditioned \
+                            and not self.engine.plugins.bindings_for_tool(m.original_name):
+                        m.a2a_fast = True
+                        m.handler = handler
             elif m.itype == "MCP":
                 m.client = ts._upstreams.get(tool.get("gateway_id") or "")
                 if isinstance(m.client, NativeInProcUpstream):
@@ -785,6 +813,23 @@ class GpuPluginPipeline:
             for r, o in zip(grp_rows, outs):
                 responses[r] = o if (o is None or isinstance(o, bytes)) else bytes(o)
 
+    async def _a2a_fast_invoke(self, mt: _ToolMeta, args: Any) -> Dict[str, Any]:
+        """A2A fast lane: handler call + the exact result shape of
+        tool_service A2A dispatch wrapping invoke_agent, minus the per-row
+        Python hook chain (GPU banks + pass 3 + host post cover it)."""
+        import uuid as _uuid
+
+        message = args.get("message") if isinstance(args, dict) else None
+        if not isinstance(message, str):
+            message = json.dumps(args or {}, default=str)
+        t0 = time.monotonic()
+        reply_text = await mt.handler(message, {})
+        sc = {"agent": mt.original_name, "response": reply_text,
+              "latency_ms": round((time.monotonic() - t0) * 1000, 2),
+              "uaid": _uuid.uuid4().hex}
+        return {"content": [{"type": "text", "text": reply_text}],
+                "structuredContent": sc, "isError": False}
+
     def _side_stream(self, i: int) -> "torch.cuda.Stream":
         ss = getattr(self, "_streams", None)
         if ss is None:
@@ -1002,7 +1047,10 @@ class GpuPluginPipeline:
                         py_errors[idx] = jsonrpc.JSONRPCError(jsonrpc.INVALID_PARAMS, "invalid arguments")
                         return
                 try:
-                    result = await ts.dispatch(mt.tool, args)
+                    if mt.a2a_fast:
+                        result = await self._a2a_fast_invoke(mt, args)
+                    else:
+                        result = await ts.dispatch(mt.tool, args)
                     py_results[idx] = json.dumps(result, separators=(",", ":"), default=str).encode()
                 except Exception as exc:
                     py_errors[idx] = exc
@@ -1017,6 +1065,7 @@ class GpuPluginPipeline:
                     nat2.append((idx, json.dumps(args2, separators=(",", ":"), default=str).encode(),
                                  mt.native_kind))
                 elif (mt.itype == "LOCAL" and not (mt.tool.get("annotations") or {}).get("io")) \
+                        or mt.a2a_fast \
                         or isinstance(mt.client, InProcUpstream) or mt.native_kind >= 0:
                     # pure in-proc handlers: sequential await beats gather
                     # overhead; IO-backed LOCAL tools (annotations.io, e.g.
