@@ -29,7 +29,11 @@ def init_from_env(device: Optional[str] = None) -> tuple[int, int]:
     world = int(os.environ.get("WORLD_SIZE", "1"))
     if world <= 1:
         return 0, 1
-    backend = "nccl" if torch.cuda.is_available() else "gloo"
+    # FORGE_DIST_BACKEND forces gloo/nccl (e.g. world-2 on a single-GPU box:
+    # RCCL refuses two ranks on one device, gloo keeps the GPU pipeline
+    # testable with the bus pump live)
+    backend = os.environ.get("FORGE_DIST_BACKEND") or \
+        ("nccl" if torch.cuda.is_available() else "gloo")
     os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
     os.environ.setdefault("MASTER_PORT", "29517")
     dist.init_process_group(backend=backend)
