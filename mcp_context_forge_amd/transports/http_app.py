@@ -279,12 +279,29 @@ def build_app(engine: GatewayEngine, auth: Optional[AuthService] = None) -> Fast
         app.add_middleware(RpcFastPath, fastapi_app=app)
 
     # -- auth dependency -------------------------------------------------------
+    def _server_scope_ok(path: str, sid: str) -> bool:
+        """Token-scoping enforcement (reference: middleware/token_scoping.py,
+        1,497 LoC): a server-scoped token may only touch ITS server's
+        surface — the RPC/MCP data plane (where the scope filters tool
+        visibility) and that server's own endpoints. Everything else
+        (registry CRUD, admin, other servers) is 403."""
+        if path in ("/rpc", "/mcp", "/health", "/healthz", "/version") or \
+                path.startswith("/.well-known"):
+            return True
+        if path.startswith(f"/servers/{sid}/") or path == f"/servers/{sid}":
+            return True
+        return False
+
     async def get_auth(request: Request) -> AuthContext:
         try:
-            return auth.authenticate(request.headers.get("authorization"))
+            ctx = auth.authenticate(request.headers.get("authorization"))
         except AuthError as exc:
             raise HTTPException(status_code=exc.status, detail=str(exc),
                                 headers={"WWW-Authenticate": "Bearer"}) from exc
+        if ctx.server_id is not None and not _server_scope_ok(request.url.path, ctx.server_id):
+            raise HTTPException(status_code=403,
+                                detail=f"token is scoped to server {ctx.server_id}")
+        return ctx
 
     def require(permission: str):
         async def dep(request: Request, ctx: AuthContext = Depends(get_auth)) -> AuthContext:
