@@ -119,6 +119,16 @@ class GpuPluginPipeline:
         self.exact_cache = plug("cached_tool_result")
         self.breaker = plug("circuit_breaker")
 
+        # plugins the fast path MODELS (as GPU banks, post-chain stages or
+        # decision flags). Any OTHER tool-hooked plugin in the chain
+        # (header_injector, an external-process plugin, a user plugin)
+        # routes its tools to the exact host chain — correctness by
+        # construction, never a silently skipped hook.
+        self._modeled = {"deny_filter", "harmful_content_detector", "pii_filter", "regex_filter",
+                         "argument_normalizer", "content_moderation", "response_cache_by_prompt",
+                         "schema_guard", "toon_encoder", "output_length_guard",
+                         "cached_tool_result", "circuit_breaker"}
+
         self.banks: Dict[str, hip.DeviceScanTables] = {}
         for name, p in (("deny", self.deny), ("harm", self.harm), ("pii", self.pii), ("regex", self.regex)):
             tables = p.scan_tables() if p is not None and hasattr(p, "scan_tables") else None
@@ -397,12 +407,21 @@ class GpuPluginPipeline:
         mgr = self.engine.plugins
         bank_names = {p.name for p in (self.deny, self.pii, self.regex, self.normalizer,
                                        self.moderation, self.harm, self.schema_guard) if p is not None}
+        from ..plugins.framework import HookType as _HT
+
+        extra_chain = [p for p in mgr.plugins
+                       if p.mode != PluginMode.DISABLED and p.name not in self._modeled
+                       and (_HT.TOOL_PRE_INVOKE in p.hooks or _HT.TOOL_POST_INVOKE in p.hooks)]
         flags = np.zeros(max(nt, 1), dtype=np.uint32)
         hostbound = np.zeros(max(nt, 1), dtype=bool)
         for i, m in enumerate(metas):
             bmap = mgr.bindings_for_tool(m.name)
             m.host_chain = any(b.get("config") or (pname not in bank_names)
                                for pname, b in bmap.items())
+            # unmodeled tool-hooked plugins (external-process, custom) that
+            # apply to this tool force the exact host chain for it
+            if not m.host_chain and extra_chain:
+                m.host_chain = any(self._applies(p, m.name) for p in extra_chain)
             hostbound[i] = m.host_chain
             f = 0
             if m.reachable:

@@ -40,6 +40,11 @@ def _resolve_class(name: str, kind: str):
         if cls is None:
             raise KeyError(f"unknown builtin plugin {name!r}")
         return cls
+    if kind == "external":
+        # external-process plugin service (reference: plugins/external/*)
+        from .external import ExternalServicePlugin
+
+        return ExternalServicePlugin
     mod_name, _, cls_name = kind.rpartition(".")
     mod = importlib.import_module(mod_name)
     return getattr(mod, cls_name)
@@ -56,6 +61,8 @@ def build_plugin(spec: Dict[str, Any]) -> Plugin:
     if "conditions" in spec:
         cfg["conditions"] = spec["conditions"]
     plugin = cls(cfg)
+    if spec.get("kind") == "external":
+        plugin.name = name  # external instances are named by their spec
     if spec.get("hooks"):
         plugin.hooks = tuple(HookType(h) for h in spec["hooks"])
     return plugin
