@@ -115,6 +115,11 @@ class Settings(BaseModel):
     metrics_buffer_flush_interval: float = 60.0
     metrics_buffer_max_size: int = 1000
     otel_enable_observability: bool = False
+    # OTLP/HTTP export (services/otel_export.py): collector base URL, e.g.
+    # http://otel-collector:4318 — spans ship to {endpoint}/v1/traces
+    otel_endpoint: str = ""
+    otel_headers: str = ""            # JSON object of extra headers (auth etc.)
+    otel_service_name: str = "mcp-context-forge-amd"
     log_level: str = "INFO"
 
     # --- admin UI / APIs ---

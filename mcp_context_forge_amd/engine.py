@@ -80,6 +80,19 @@ class GatewayEngine:
         self.performance = PerformanceService(self)
         self.toolops = ToolOps(self)
         self.observability = ObservabilityService(self.db)
+        if self.settings.otel_endpoint:
+            # OTLP/HTTP export to a collector (reference: init_telemetry :970)
+            from .services.otel_export import OtlpHttpExporter
+
+            hdrs = {}
+            if self.settings.otel_headers:
+                try:
+                    hdrs = json.loads(self.settings.otel_headers)
+                except ValueError:
+                    logger.warning("otel_headers is not valid JSON; ignoring")
+            self.observability.exporter = OtlpHttpExporter(
+                self.settings.otel_endpoint, headers=hdrs,
+                service_name=self.settings.otel_service_name)
         self.catalog = CatalogService()
         from .services.governance import TagService
 

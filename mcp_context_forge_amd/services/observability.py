@@ -60,6 +60,9 @@ class ObservabilityService:
         self.max_buffer = max_buffer
         self._spans: List[Span] = []
         self._lock = threading.Lock()
+        # optional OTLP/HTTP shipper (services/otel_export.py); set by the
+        # engine when Settings.otel_endpoint is configured
+        self.exporter = None
 
     @contextmanager
     def span(self, name: str, **attrs: Any):
@@ -95,6 +98,8 @@ class ObservabilityService:
     def flush(self) -> int:
         with self._lock:
             spans, self._spans = self._spans, []
+        if spans and self.exporter is not None:
+            self.exporter.export(spans)  # fail-open, background thread
         if not spans or self.db is None:
             return len(spans)
         with self.db.session() as s:
