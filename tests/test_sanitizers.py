@@ -1,0 +1,43 @@
+"""ASan + TSan tier for the C++ host data plane (the
+reference needs no sanitizers — pure-Python hot path; this build's
+decision plane is native and threaded, so it gets real ones).
+
+Builds csrc/{envelope,upstream,fastpath,san_driver}.cpp with
+-fsanitize=address and -fsanitize=thread (plain g++ — these translation
+units are pure host C++) and runs the driver, which overlaps
+decide/finalize/upstream/store traffic from several threads against
+shared stores — the exact shape the depth-2 edge produces. A sanitizer
+report makes the binary exit non-zero."""
+
+import subprocess
+from pathlib import Path
+
+import pytest
+
+CSRC = Path("/root/repo/mcp_context_forge_amd/ops/csrc")
+BUILD = Path("/root/repo/mcp_context_forge_amd/ops")
+SOURCES = ["envelope.cpp", "upstream.cpp", "fastpath.cpp", "san_driver.cpp"]
+
+
+def _build(flavor: str) -> Path:
+    out = BUILD / f"san_driver_{flavor}"
+    srcs = [str(CSRC / s) for s in SOURCES]
+    newest = max((CSRC / s).stat().st_mtime for s in SOURCES)
+    if out.exists() and out.stat().st_mtime > newest:
+        return out
+    cmd = ["g++", "-O1", "-g", "-std=c++17", "-pthread", f"-fsanitize={flavor}",
+           "-fno-omit-frame-pointer", "-o", str(out)] + srcs
+    subprocess.run(cmd, check=True, capture_output=True, text=True)
+    return out
+
+
+@pytest.mark.parametrize("flavor", ["address", "thread"])
+def test_host_data_plane_under_sanitizer(flavor):
+    binary = _build(flavor)
+    env = {"ASAN_OPTIONS": "detect_leaks=1:abort_on_error=0",
+           "TSAN_OPTIONS": "halt_on_error=1"}
+    iters, threads, rows = ("4", "4", "384") if flavor == "thread" else ("6", "4", "512")
+    r = subprocess.run([str(binary), iters, threads, rows], capture_output=True,
+                       text=True, timeout=600, env=env)
+    assert r.returncode == 0, f"{flavor} sanitizer:\n{r.stdout[-2000:]}\n{r.stderr[-4000:]}"
+    assert "san driver ok" in r.stdout
