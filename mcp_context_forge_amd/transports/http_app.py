@@ -124,6 +124,35 @@ class BodyLimitMiddleware(BaseHTTPMiddleware):
         return await call_next(request)
 
 
+class ApiVersionAlias:
+    """Versioned API facade (reference: api/v1/__init__.py:355 + the legacy
+    alias shim wired at main.py:12811): every route is also reachable under
+    /v1/… — one implementation surface, a stable versioned path for
+    clients. GET /v1 returns the version index."""
+
+    PREFIX = "/v1"
+
+    def __init__(self, app):
+        self.app = app
+
+    async def __call__(self, scope, receive, send):
+        if scope["type"] == "http":
+            p = scope.get("path", "")
+            if p in ("/v1", "/v1/"):
+                body = (b'{"version":"v1","status":"stable",'
+                        b'"note":"all unversioned routes are aliased under /v1"}')
+                await send({"type": "http.response.start", "status": 200,
+                            "headers": [(b"content-type", b"application/json"),
+                                        (b"content-length", str(len(body)).encode())]})
+                await send({"type": "http.response.body", "body": body})
+                return
+            if p.startswith("/v1/"):
+                scope = dict(scope)
+                scope["path"] = p[len(self.PREFIX):]
+                scope["raw_path"] = scope["path"].encode()
+        await self.app(scope, receive, send)
+
+
 class RpcFastPath:
     """Pure-ASGI fast lane for POST /rpc — bypasses FastAPI routing/DI and
     the BaseHTTPMiddleware stack for the hot endpoint, exactly the role the
@@ -277,6 +306,8 @@ def build_app(engine: GatewayEngine, auth: Optional[AuthService] = None) -> Fast
     if not settings.rate_limit_enabled:
         # hot-lane /rpc (outermost; skipped when rate limiting must apply)
         app.add_middleware(RpcFastPath, fastapi_app=app)
+    # /v1 alias outermost so versioned paths hit the fast lane too
+    app.add_middleware(ApiVersionAlias)
 
     # -- auth dependency -------------------------------------------------------
     def _server_scope_ok(path: str, sid: str) -> bool:
@@ -530,6 +561,16 @@ def build_app(engine: GatewayEngine, auth: Optional[AuthService] = None) -> Fast
         @app.post(f"/{plural}", status_code=201, name=f"create_{plural}")
         async def create_entity(request: Request, ctx: AuthContext = Depends(require(f"{perm}.create"))):
             body = await request.json()
+            # typed request validation (reference: schemas.py models) —
+            # field-level 422s before anything touches the registry
+            from pydantic import ValidationError
+
+            from ..protocol.schemas import validate_create
+
+            try:
+                body = validate_create(kind, body if isinstance(body, dict) else {})
+            except ValidationError as exc:
+                raise HTTPException(422, detail=json.loads(exc.json())) from exc
             try:
                 if kind == "gateway":
                     # defer=true → async lifecycle: 201 with a `pending` row;
