@@ -186,6 +186,13 @@ class GpuPluginPipeline:
         self.banks["postmeta"] = hip.DeviceScanTables(
             dfa.compile_literals(['"structuredContent"'], case_insensitive=False), device)
 
+        # fused multi-bank scans: pass 3's bank set is static; pass 1's
+        # includes the per-registry schema bank (rebuilt in _rebuild_tool_meta)
+        self._bankset3 = hip.ScanBankSet(
+            [(n, self.banks[n]) for n in ("pii", "regex", "harm", "postmeta") if n in self.banks],
+            device)
+        self._bankset1: Optional[hip.ScanBankSet] = None
+
         # native decision-plane stores + string tables (fastpath.cpp)
         self._slot_store = hip.store_new(self.semcache.capacity) if self.semcache is not None else 0
         self._exact_native = hip.cache_new(self.exact_cache.ttl) if self.exact_cache is not None else 0
@@ -362,6 +369,13 @@ class GpuPluginPipeline:
                         m.schema_mode = "host"
         else:
             self._schema_bank = None
+
+        # pass-1 fused bank set (depends on the schema bank just built)
+        b1 = [(n, self.banks[n]) for n in ("deny", "harm", "pii", "regex", "normalize")
+              if n in self.banks]
+        if self._schema_bank is not None:
+            b1.append(("schema", self._schema_bank))
+        self._bankset1 = hip.ScanBankSet(b1, self.device)
 
         # ---- flat arrays for the native decision plane (fastpath.cpp) ----
         metas = list(self._tool_meta.values())
@@ -673,20 +687,10 @@ class GpuPluginPipeline:
         end_t = self._upload(args_e)
         self._toc("gp1_upload", t_g)
         t_l = self._tic()
-        # the scan banks are latency-bound and underfill the chip (one lane
-        # per request ⇒ ~128 waves over 256 CUs) — launch each bank on its
-        # own HIP stream so the six DFA sweeps and the MFMA chain overlap
-        out: Dict[str, torch.Tensor] = {}
-        scan_banks = [(n, self.banks[n]) for n in ("deny", "harm", "pii", "regex", "normalize")
-                      if n in self.banks]
-        if self._schema_bank is not None:
-            scan_banks.append(("schema", self._schema_bank))
-        main = torch.cuda.current_stream()
-        for i, (bname, bank) in enumerate(scan_banks):
-            s_ = self._side_stream(i)
-            s_.wait_stream(main)
-            with torch.cuda.stream(s_):
-                out[bname], _ = hip.scan(data_gpu, beg_t, end_t, bank)
+        # fused multi-bank scan: all DFA banks in ONE launch (grid.y=bank;
+        # fills the chip bank-parallel where per-bank launches ran ~128
+        # waves each) writing one [n_banks, m] mask matrix — one D2H later
+        out_multi = hip.scan_multi(data_gpu, beg_t, end_t, self._bankset1)
         feats = None
         if self.classifier is not None or self.semcache is not None:
             feats_b, _ = hip.featurize(data_gpu, beg_t, end_t, self.feat_dim)
@@ -709,9 +713,11 @@ class GpuPluginPipeline:
         self._toc("gpu_pass1", t_g)
         t_d = self._tic()
 
+        mm = out_multi.cpu().numpy().view(np.uint32)  # one D2H for every bank
+
         def mask(name):
-            t = out.get(name)
-            return t.cpu().numpy().view(np.uint32) if t is not None else np.zeros(m, dtype=np.uint32)
+            i = self._bankset1.index.get(name)
+            return mm[i] if i is not None else np.zeros(m, dtype=np.uint32)
 
         deny_m, harm_m, pii_m = mask("deny"), mask("harm"), mask("pii")
         regex_m, norm_m, schema_m = mask("regex"), mask("normalize"), mask("schema")
@@ -1153,20 +1159,14 @@ class GpuPluginPipeline:
             data3 = self._upload(res_blob)
             b3 = self._upload(res_beg.astype(np.int32))
             e3 = self._upload(res_end.astype(np.int32))
-            masks3 = {}
-            main3 = torch.cuda.current_stream()
-            for i, b in enumerate(("pii", "regex", "harm", "postmeta")):
-                if b in self.banks:
-                    s_ = self._side_stream(i)
-                    s_.wait_stream(main3)
-                    with torch.cuda.stream(s_):
-                        masks3[b], _ = hip.scan(data3, b3, e3, self.banks[b])
+            out3 = hip.scan_multi(data3, b3, e3, self._bankset3)  # fused, one launch
             try:
                 await asyncio.to_thread(torch.cuda.synchronize)
             finally:
                 self._gpu_lock.release()
-            for b, t in masks3.items():
-                h = t.cpu().numpy().view(np.uint32) != 0
+            mm3 = out3.cpu().numpy().view(np.uint32)
+            for b, i in self._bankset3.index.items():
+                h = mm3[i] != 0
                 if b == "postmeta":
                     toon_meta = h
                 else:

@@ -63,6 +63,80 @@ __global__ __launch_bounds__(256) void scan_kernel(
     if (out_first_end) out_first_end[r] = first_end;
 }
 
+// ---------------------------------------------------------------------------
+// Fused multi-bank scan: ALL DFA banks in ONE launch (grid.y = bank).
+//
+// Measured motivation (profiles/README_r02.md stage budget): at HTTP batch
+// sizes (~1-3k rows) the per-bank launches + per-bank D2H copies dominate
+// the scan stage — six launches of ~128 waves each also underfill 256 CUs.
+// One launch with grid (rows/256, n_banks) fills the chip bank-parallel,
+// writes one [n_banks, B] mask matrix, and the host does ONE D2H.
+//
+// Bank descriptors live in device memory (built once per bank-set); each
+// block stages ITS bank's table into LDS when it fits the per-block
+// allocation (banks over the limit read through L2, same as scan_kernel).
+
+struct ScanBankDesc {
+    const uint16_t* next;
+    const uint8_t* klass;
+    const uint32_t* accept;
+    int n_states;
+    int n_classes;
+    int use_lds;   // table fits the dynamic LDS allocation
+    int _pad[2];
+};
+
+__global__ __launch_bounds__(256) void scan_multi_kernel(
+    const uint8_t* __restrict__ data,
+    const int32_t* __restrict__ beg,
+    const int32_t* __restrict__ end_,
+    int batch,
+    const ScanBankDesc* __restrict__ descs,
+    uint32_t* __restrict__ out_mask)      // [n_banks, batch]
+{
+    extern __shared__ uint8_t lds_raw[];
+    const ScanBankDesc d = descs[blockIdx.y];
+    const uint16_t* tnext = d.next;
+    const uint8_t* tklass = d.klass;
+    const uint32_t* taccept = d.accept;
+    if (d.use_lds) {
+        uint16_t* lnext = (uint16_t*)lds_raw;
+        uint8_t* lklass = (uint8_t*)(lds_raw + (size_t)d.n_states * d.n_classes * 2);
+        uint32_t* laccept = (uint32_t*)(lklass + 256);
+        int total16 = d.n_states * d.n_classes;
+        for (int i = threadIdx.x; i < total16; i += blockDim.x) lnext[i] = d.next[i];
+        for (int i = threadIdx.x; i < 256; i += blockDim.x) lklass[i] = d.klass[i];
+        for (int i = threadIdx.x; i < d.n_states; i += blockDim.x) laccept[i] = d.accept[i];
+        __syncthreads();
+        tnext = lnext; tklass = lklass; taccept = laccept;
+    }
+    int r = blockIdx.x * blockDim.x + threadIdx.x;
+    if (r >= batch) return;
+    int32_t rbeg = beg[r], rend = end_[r];
+    uint32_t mask = 0;
+    uint32_t state = 0;
+    int nc = d.n_classes;
+    for (int32_t p = rbeg; p < rend; ++p) {
+        state = tnext[state * nc + tklass[data[p]]];
+        mask |= taccept[state];
+    }
+    out_mask[(size_t)blockIdx.y * batch + r] = mask;
+}
+
+extern "C" int forge_scan_multi(
+    const void* data, const void* beg, const void* end_, int batch,
+    const void* descs_dev, int n_banks, int lds_bytes,
+    void* out_mask, void* stream)
+{
+    hipStream_t s = (hipStream_t)stream;
+    int block = 256;
+    dim3 grid((unsigned)ceil_div(batch, block), (unsigned)n_banks);
+    hipLaunchKernelGGL(scan_multi_kernel, grid, dim3(block), (size_t)lds_bytes, s,
+                       (const uint8_t*)data, (const int32_t*)beg, (const int32_t*)end_, batch,
+                       (const ScanBankDesc*)descs_dev, (uint32_t*)out_mask);
+    return (int)hipGetLastError();
+}
+
 extern "C" int forge_scan(
     const void* data, const void* beg, const void* end_, int batch,
     const void* next, const void* klass, const void* accept,

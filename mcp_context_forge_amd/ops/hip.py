@@ -10,7 +10,7 @@ from __future__ import annotations
 
 import ctypes
 from functools import lru_cache
-from typing import Optional, Tuple
+from typing import List, Optional, Tuple
 
 import numpy as np
 import torch
@@ -31,6 +31,8 @@ def _load() -> ctypes.CDLL:
     lib = ctypes.CDLL(str(LIB))
     protos = {
         "forge_scan": [ctypes.c_void_p] * 3 + [ctypes.c_int] + [ctypes.c_void_p] * 3 + [ctypes.c_int] * 2 + [ctypes.c_void_p] * 3,
+        "forge_scan_multi": [ctypes.c_void_p] * 3 + [ctypes.c_int, ctypes.c_void_p,
+                                                    ctypes.c_int, ctypes.c_int] + [ctypes.c_void_p] * 2,
         "forge_featurize": [ctypes.c_void_p] * 3 + [ctypes.c_int, ctypes.c_int] + [ctypes.c_void_p] * 3,
         "forge_json_guard": [ctypes.c_void_p] * 3 + [ctypes.c_int] * 3 + [ctypes.c_void_p] * 3,
         "forge_gemm_bt": [ctypes.c_void_p] * 4 + [ctypes.c_int] * 5 + [ctypes.c_void_p],
@@ -95,6 +97,48 @@ def scan(data: torch.Tensor, beg: torch.Tensor, end: torch.Tensor, tables: Devic
         tables.n_states, tables.n_classes,
         _ptr(out_mask), _ptr(out_end), _stream()))
     return out_mask, out_end
+
+
+class ScanBankSet:
+    """Device descriptor table for the fused multi-bank scan
+    (scan_multi_kernel): all banks in ONE launch, one [n_banks, B] output.
+    Built once per bank combination; holds references to the bank tensors
+    so their device memory outlives the descriptor pointers."""
+
+    LDS_LIMIT = 64 * 1024
+
+    def __init__(self, banks: List[Tuple[str, "DeviceScanTables"]], device: str = "cuda"):
+        self.names = [n for n, _ in banks]
+        self.banks = [b for _, b in banks]
+        self.n = len(self.banks)
+        self.index = {n: i for i, n in enumerate(self.names)}
+        self.lds_bytes = 0
+        recs = np.zeros((max(self.n, 1), 6), dtype=np.int64)
+        for i, b in enumerate(self.banks):
+            tbytes = b.n_states * b.n_classes * 2 + 256 + b.n_states * 4
+            use_lds = 1 if tbytes <= self.LDS_LIMIT else 0
+            if use_lds:
+                self.lds_bytes = max(self.lds_bytes, tbytes)
+            recs[i, 0] = b.next.data_ptr()
+            recs[i, 1] = b.klass.data_ptr()
+            recs[i, 2] = b.accept.data_ptr()
+            recs[i, 3] = (b.n_states & 0xFFFFFFFF) | (b.n_classes << 32)
+            recs[i, 4] = use_lds
+        self.descs = torch.from_numpy(recs).to(device)
+
+
+def scan_multi(data: torch.Tensor, beg: torch.Tensor, end: torch.Tensor,
+               bankset: ScanBankSet) -> torch.Tensor:
+    """→ int32 [n_banks, B] mask matrix (bit-identical to per-bank scan)."""
+    batch = beg.numel()
+    out = torch.zeros((bankset.n, batch), dtype=torch.int32, device=data.device)
+    if bankset.n == 0 or batch == 0:
+        return out
+    _check("forge_scan_multi", _load().forge_scan_multi(
+        _ptr(data), _ptr(beg), _ptr(end), batch,
+        _ptr(bankset.descs), bankset.n, bankset.lds_bytes,
+        _ptr(out), _stream()))
+    return out
 
 
 def featurize(data: torch.Tensor, beg: torch.Tensor, end: torch.Tensor, dim: int,

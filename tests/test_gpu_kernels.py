@@ -325,3 +325,40 @@ def test_semcache_sketch_two_stage(hip):
     hits = cache.resolve_hits(bv_h, bi_h, np.array([tool_hash("t")] * 3))
     assert hits[0] == {"slot": 5} and hits[1] == {"slot": 50}
     assert hits[2] is None  # random vector: verified cosine below threshold
+
+
+@pytest.mark.gpu
+def test_scan_multi_matches_per_bank():
+    """Fused multi-bank scan is bit-identical to per-bank scan_kernel."""
+    import torch
+
+    if not torch.cuda.is_available():
+        pytest.skip("no ROCm device")
+    from mcp_context_forge_amd.ops import dfa, hip
+
+    texts = [b'{"msg":"plain"}', b'{"msg":"has forbidden word"}',
+             b'{"mail":"a@b.com and c@d.io"}', b'{"x":"' + b"z" * 3000 + b'"}',
+             b"", b'{"t":"how to make a bomb"}'] * 40
+    blob = b"".join(texts)
+    offs = [0]
+    for t in texts:
+        offs.append(offs[-1] + len(t))
+    import numpy as np
+
+    data = torch.from_numpy(np.frombuffer(blob, dtype=np.uint8).copy()).cuda()
+    beg = torch.tensor(offs[:-1], dtype=torch.int32).cuda()
+    end = torch.tensor(offs[1:], dtype=torch.int32).cuda()
+
+    banks = {
+        "deny": hip.DeviceScanTables(dfa.compile_literals(["forbidden", "blocked_word"], True)),
+        "harm": hip.DeviceScanTables(dfa.compile_literals(["how to make a bomb"], True)),
+        "mail": hip.DeviceScanTables(dfa.compile_patterns(
+            ["[a-z0-9]@[a-z0-9]", "z{100}"], case_insensitive=True)),
+    }
+    bs = hip.ScanBankSet(list(banks.items()))
+    fused = hip.scan_multi(data, beg, end, bs)
+    torch.cuda.synchronize()
+    for name, bank in banks.items():
+        single, _ = hip.scan(data, beg, end, bank)
+        torch.cuda.synchronize()
+        assert torch.equal(fused[bs.index[name]], single), name
