@@ -585,6 +585,10 @@ def build_app(engine: GatewayEngine, auth: Optional[AuthService] = None) -> Fast
                 if kind == "tool":
                     body.setdefault("original_name", body.get("name"))
                     body.setdefault("integration_type", "REST" if body.get("url") else "LOCAL")
+                if kind == "a2a_agent":
+                    from ..utils import slugify
+
+                    body.setdefault("slug", slugify(body["name"]))
                 ent = engine.registry.create(kind, **body)
                 engine.audit.record(ctx.user, "create", kind, ent.get("id"))
                 await engine.notify_list_changed(kind)
