@@ -190,3 +190,55 @@ def test_edge_under_forge_hey(edge_app, run):
             assert res["p50_ms"] > 0
 
     run(go())
+
+
+@pytest.mark.gpu
+def test_edge_with_gpu_pipeline_under_load(run):
+    """GPU tier: the C++ edge feeding the GPU batch pipeline end to end,
+    driven by the native load generator — the flagship serving shape."""
+    import torch
+
+    if not torch.cuda.is_available():
+        pytest.skip("no ROCm device")
+    from mcp_context_forge_amd.ops.build import HEY, build_hey
+
+    build_hey(verbose=False)
+    port = free_port()
+    s = Settings(database_url="sqlite://", federation_enabled=False, auth_required=True,
+                 plugins_enabled=True, gpu_enabled=True, native_edge_port=port,
+                 native_edge_threads=4)
+    engine = GatewayEngine(s)
+
+    async def go():
+        from mcp_context_forge_amd.services.upstream import NativeInProcUpstream
+
+        await engine.gateway_service.register_gateway(
+            name="nt", url="inproc://nt", client=NativeInProcUpstream(), owner_rank=0)
+        app = build_app(engine)
+        async with app.router.lifespan_context(app):
+            assert engine.gpu_pipeline is not None, "GPU pipeline must attach"
+            tok = app.state.auth.create_api_token("admin@example.com", "edge-gpu")
+            payload = json.dumps({"jsonrpc": "2.0", "id": 1, "method": "tools/call",
+                                  "params": {"name": "nt-convert_time",
+                                             "arguments": {"time": "2026-01-01T00:00:00Z",
+                                                           "source_timezone": "UTC",
+                                                           "target_timezone": "UTC"}}})
+            proc = await asyncio.create_subprocess_exec(
+                str(HEY), "--host", "127.0.0.1", "--port", str(port), "--path", "/rpc",
+                "--connections", "64", "--threads", "2",
+                "--requests-per-step", "2000", "--warmup", "1", "--steps", "2",
+                "--payload", payload, "--auth", f"Bearer {tok}",
+                stdin=asyncio.subprocess.PIPE, stdout=asyncio.subprocess.PIPE)
+            line = await asyncio.wait_for(proc.stdout.readline(), timeout=60)
+            assert line.strip() == b"WARM"
+            proc.stdin.write(b"GO\n")
+            await proc.stdin.drain()
+            out = await asyncio.wait_for(proc.stdout.readline(), timeout=120)
+            await proc.wait()
+            res = json.loads(out)
+            assert res["errors"] == 0 and res["non200"] == 0, res
+            assert res["requests"] == 4000
+            st = engine.gpu_pipeline.stats()
+            assert st["requests"] >= 4000  # batches actually rode the GPU path
+
+    run(go())
