@@ -29,6 +29,10 @@ def main(argv=None) -> int:
     p.add_argument("--host", default=None)
     p.add_argument("--port", type=int, default=None)
     p.add_argument("--workers", type=int, default=1)
+    p.add_argument("--native-edge-port", type=int, default=None,
+                   help="serve POST /rpc on this port with the C++ epoll edge "
+                        "(the flagship hot lane; control plane stays on --port)")
+    p.add_argument("--native-edge-threads", type=int, default=None)
 
     sub.add_parser("stdio", help="run the gateway engine as a stdio MCP server")
 
@@ -77,6 +81,10 @@ def main(argv=None) -> int:
             settings.host = args.host
         if args.port:
             settings.port = args.port
+        if args.native_edge_port:
+            settings.native_edge_port = args.native_edge_port
+        if args.native_edge_threads:
+            settings.native_edge_threads = args.native_edge_threads
         import uvicorn
 
         if args.workers and args.workers > 1:
