@@ -250,3 +250,64 @@ def test_make_client_selects_sse_transport():
                                 "auth_type": None, "auth_value": None})
     assert isinstance(client2, HttpUpstreamClient)
     assert not isinstance(client2, SseUpstreamClient)
+
+
+def test_health_loop_failure_and_enabled_filter(run):
+    """Mutation survivors pinned: a failing ping marks failures (the
+    except-path must set ok=False) and disabled gateways are skipped."""
+
+    async def go():
+        e = _engine()
+        up = FlakyUpstream(fail_times=0)
+        gw = await e.gateway_service.register_gateway(
+            name="hc", url="inproc://hc", client=up, owner_rank=0)
+
+        async def bad_ping():
+            raise RuntimeError("down")
+
+        up.ping = bad_ping
+        res = await e.gateway_service.check_health_once()
+        assert res.get(gw["id"]) is False
+        assert e.registry.get("gateway", gw["id"])["consecutive_failures"] >= 1
+        # disabled gateways are not checked at all
+        e.registry.set_enabled("gateway", gw["id"], False)
+        res2 = await e.gateway_service.check_health_once()
+        assert gw["id"] not in res2
+        await e.shutdown()
+
+    run(go())
+
+
+def test_backoff_schedules_future_retry(run):
+    """next_retry_at must land in the FUTURE after a failed attempt."""
+    import time as _t
+
+    async def go():
+        e = _engine()
+        up = FlakyUpstream(fail_times=99)
+        gw = await e.gateway_service.register_gateway(
+            name="fut", url="inproc://fut", client=up, owner_rank=0, defer=True)
+        for _ in range(200):
+            gw = e.registry.get("gateway", gw["id"])
+            if (gw.get("retry_count") or 0) >= 1 and gw["status"] == "pending":
+                break
+            await asyncio.sleep(0.01)
+        assert gw["next_retry_at"] is not None and gw["next_retry_at"] > _t.time() - 0.001
+        # strictly scheduled AFTER the failure that produced it
+        assert gw["next_retry_at"] > float(gw["updated_at"].timestamp()) - 60  # sanity
+        await e.shutdown()
+
+    run(go())
+
+
+def test_make_client_bearer_header(run):
+    async def go():
+        e = _engine()
+        gws = e.gateway_service
+        sealed = gws.crypto.seal("tok-123")
+        client = gws._make_client({"url": "http://x/mcp", "transport": "streamablehttp",
+                                   "auth_type": "bearer", "auth_value": sealed})
+        assert client.base_headers.get("authorization") == "Bearer tok-123"
+        await e.shutdown()
+
+    run(go())

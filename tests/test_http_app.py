@@ -683,3 +683,31 @@ def test_security_limits_and_auth_bypass_attempts(client_engine, run):
             assert r.status_code == 401
 
     run(go())
+
+
+def test_respond_stream_delivers_broadcast(run):
+    """Mutation survivor pinned: the respond loop actually LOOPS — a
+    broadcast lands on a live respond_stream consumer."""
+    import asyncio as _a
+
+    from mcp_context_forge_amd.services.sessions import SessionRegistry
+
+    async def go():
+        reg = SessionRegistry()
+        sess = reg.create(transport="sse")
+        got = []
+
+        async def consume():
+            async for eid, msg in reg.respond_stream(sess.session_id, keepalive_s=5.0):
+                got.append(msg)
+                if len(got) >= 2:
+                    return
+
+        task = _a.ensure_future(consume())
+        await _a.sleep(0.05)
+        assert await reg.broadcast(sess.session_id, {"n": 1})
+        assert await reg.broadcast(sess.session_id, {"n": 2})
+        await _a.wait_for(task, timeout=5)
+        assert [m["n"] for m in got] == [1, 2]
+
+    run(go())

@@ -83,3 +83,16 @@ def test_heartbeat_loop_and_engine_gating(tmp_path, run):
         pre.close()
 
     run(go())
+
+
+def test_mutation_kills_initial_and_release_state(tmp_path):
+    """Mutation-sweep survivors pinned down: a fresh elector is NOT leader
+    until it acquires, and release actually relinquishes."""
+    db = _db(tmp_path)
+    a = DbLeaderElector(db, ttl_s=5.0, holder_id="A")
+    assert a.is_leader is False           # pre-acquire state
+    assert a.try_acquire()
+    assert a.is_leader is True
+    a.release()
+    assert a.is_leader is False           # release flips local state too
+    db.close()
