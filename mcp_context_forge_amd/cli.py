@@ -58,6 +58,14 @@ def main(argv=None) -> int:
     p.add_argument("--name", default=None)
     p.add_argument("--token", default=None)
 
+    p = sub.add_parser("plugin-scaffold", help="generate a plugin skeleton "
+                       "(reference: cforge plugin bootstrap, tools/cli.py)")
+    p.add_argument("name", help="plugin name, e.g. my_policy")
+    p.add_argument("--dir", default="plugins", help="output directory")
+    p.add_argument("--hooks", default="tool_pre_invoke,tool_post_invoke")
+    p.add_argument("--external", action="store_true",
+                   help="scaffold an external-process HTTP service instead of an in-proc plugin")
+
     p = sub.add_parser("token", help="mint an HS256 JWT for the gateway")
     p.add_argument("--user", default="admin@example.com")
     p.add_argument("--admin", action="store_true")
@@ -191,6 +199,65 @@ def main(argv=None) -> int:
 
         asyncio.run(run())
         return 0
+
+    if args.cmd == "plugin-scaffold":
+        from pathlib import Path
+
+        name = args.name.replace("-", "_")
+        hooks = [h.strip() for h in args.hooks.split(",") if h.strip()]
+        out = Path(args.dir)
+        out.mkdir(parents=True, exist_ok=True)
+        if args.external:
+            (out / f"{name}_service.py").write_text(f'''"""External plugin service `{name}` — runs as its own process.
+
+Start:  uvicorn {name}_service:app --port 9901
+Wire into plugins/config.yaml:
+    - name: {name}
+      kind: external
+      mode: enforce
+      config: {{url: "http://127.0.0.1:9901", hooks: {hooks!r}}}
+"""
+
+from mcp_context_forge_amd.plugins.external import build_external_service_app
+
+
+def policy(req: dict) -> dict:
+    """req: {{hook, plugin, name, args, user, server_id}} ->
+    {{action: allow|block|transform, reason?, payload?, metadata?}}"""
+    # TODO: implement your policy
+    return {{"action": "allow"}}
+
+
+app = build_external_service_app(policy, name="{name}")
+''')
+            print(out / f"{name}_service.py")
+        else:
+            hook_methods = "\n\n".join(
+                f"    async def {h}(self, ctx: PluginContext) -> PluginResult:\n"
+                f"        # TODO: inspect ctx.args / ctx.name / ctx.user\n"
+                f"        return PluginResult.ok()" for h in hooks)
+            (out / f"{name}.py").write_text(f'''"""Plugin `{name}` (scaffolded).
+
+Wire into plugins/config.yaml:
+    - name: {name}
+      kind: {name}.{name.title().replace("_", "")}Plugin
+      mode: enforce
+      priority: 100
+"""
+
+from mcp_context_forge_amd.plugins.framework import (HookType, Plugin, PluginContext,
+                                                     PluginResult)
+
+
+class {name.title().replace("_", "")}Plugin(Plugin):
+    name = "{name}"
+    hooks = ({", ".join(f"HookType.{h.upper()}" for h in hooks)},)
+    priority = 100
+
+{hook_methods}
+''')
+            print(out / f"{name}.py")
+        return
 
     if args.cmd == "token":
         from .auth import jwt as jwt_mod

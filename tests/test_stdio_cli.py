@@ -85,3 +85,31 @@ def test_cli_token_mint():
 
     claims = jwt_mod.decode_token(token, Settings().jwt_secret_key)
     assert claims["sub"] == "x@y.com" and claims["admin"] is True
+
+
+def test_plugin_scaffold_cli(tmp_path):
+    """cforge-analog plugin bootstrap (reference: tools/cli.py): generated
+    skeletons import cleanly and register as plugins."""
+    import importlib.util
+    import subprocess
+    import sys
+
+    env = {"PYTHONPATH": "/root/repo", "PATH": "/usr/bin:/bin"}
+    r = subprocess.run([sys.executable, "-m", "mcp_context_forge_amd", "plugin-scaffold",
+                        "my_guard", "--dir", str(tmp_path)],
+                       capture_output=True, text=True, env=env, timeout=120)
+    assert r.returncode == 0, r.stderr
+    spec = importlib.util.spec_from_file_location("my_guard", tmp_path / "my_guard.py")
+    m = importlib.util.module_from_spec(spec)
+    spec.loader.exec_module(m)
+    p = m.MyGuardPlugin()
+    assert p.name == "my_guard" and len(p.hooks) == 2
+
+    r2 = subprocess.run([sys.executable, "-m", "mcp_context_forge_amd", "plugin-scaffold",
+                         "av", "--dir", str(tmp_path), "--external"],
+                        capture_output=True, text=True, env=env, timeout=120)
+    assert r2.returncode == 0, r2.stderr
+    spec2 = importlib.util.spec_from_file_location("av_service", tmp_path / "av_service.py")
+    m2 = importlib.util.module_from_spec(spec2)
+    spec2.loader.exec_module(m2)
+    assert callable(m2.app)
