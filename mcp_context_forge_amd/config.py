@@ -129,7 +129,10 @@ class Settings(BaseModel):
     # --- GPU pipeline (MI355X-native; replaces cache_type=redis in the reference) ---
     gpu_enabled: bool = True          # auto-falls back to CPU reference path when no HIP device
     gpu_batch_max_requests: int = 8192
-    gpu_batch_window_us: int = 500    # adaptive micro-batch window
+    # micro-batch linger: with two batches in flight the pipeline fills
+    # itself; measured at the 1000-user knee: 100 µs beats 500 µs by ~12%
+    # RPS and ~0.5 ms p50 (profiles/README_r02.md linger sweep)
+    gpu_batch_window_us: int = 100
     gpu_feature_dim: int = 4096       # hashed count-vector dim for classifiers/semantic cache
     gpu_classifier_hidden: int = 1024
     gpu_classifier_classes: int = 8
