@@ -35,9 +35,29 @@ PROVIDER_PRESETS: Dict[str, Dict[str, str]] = {
         "email_field": "email",
         "scopes": "openid email profile",
     },
-    "okta": {"email_field": "email", "scopes": "openid email profile"},
-    "keycloak": {"email_field": "email", "scopes": "openid email profile"},
-    "entra": {"email_field": "mail", "scopes": "openid email profile"},
+    # templated presets: placeholders fill from register_provider kwargs
+    # (reference: sso_service.py Entra/Okta/Keycloak specifics)
+    "okta": {
+        "authorize_url": "https://{domain}/oauth2/v1/authorize",
+        "token_url": "https://{domain}/oauth2/v1/token",
+        "userinfo_url": "https://{domain}/oauth2/v1/userinfo",
+        "email_field": "email",
+        "scopes": "openid email profile",
+    },
+    "keycloak": {
+        "authorize_url": "{base}/realms/{realm}/protocol/openid-connect/auth",
+        "token_url": "{base}/realms/{realm}/protocol/openid-connect/token",
+        "userinfo_url": "{base}/realms/{realm}/protocol/openid-connect/userinfo",
+        "email_field": "email",
+        "scopes": "openid email profile",
+    },
+    "entra": {
+        "authorize_url": "https://login.microsoftonline.com/{tenant}/oauth2/v2.0/authorize",
+        "token_url": "https://login.microsoftonline.com/{tenant}/oauth2/v2.0/token",
+        "userinfo_url": "https://graph.microsoft.com/oidc/userinfo",
+        "email_field": "mail",
+        "scopes": "openid email profile",
+    },
     "oidc": {"email_field": "email", "scopes": "openid email profile"},
 }
 
@@ -62,6 +82,13 @@ class SSOService:
         for req in ("authorize_url", "token_url", "userinfo_url"):
             if req not in cfg:
                 raise SSOError(f"provider {name}: missing {req}")
+            if "{" in cfg[req]:
+                # templated preset (entra {tenant}, okta {domain},
+                # keycloak {base}/{realm}) fills from the kwargs
+                try:
+                    cfg[req] = cfg[req].format(**overrides)
+                except KeyError as exc:
+                    raise SSOError(f"provider {name}: preset {preset!r} needs {exc.args[0]}=") from exc
         cfg.update({"client_id": client_id, "client_secret": client_secret, "name": name})
         self.providers[name] = cfg
         return {k: v for k, v in cfg.items() if k != "client_secret"}

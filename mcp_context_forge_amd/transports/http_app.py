@@ -409,6 +409,17 @@ def build_app(engine: GatewayEngine, auth: Optional[AuthService] = None) -> Fast
 
     # -- streamable HTTP /mcp (reference: transports/streamablehttp_transport.py) --
     async def mcp_post(request: Request, ctx: AuthContext, server_id: Optional[str] = None):
+        # protocol-version negotiation guard (reference:
+        # middleware/protocol_version.py MCPProtocolVersionMiddleware):
+        # a client-declared MCP-Protocol-Version we do not speak is a 400
+        from ..protocol.mcp import SUPPORTED_PROTOCOL_VERSIONS
+
+        pv = request.headers.get("mcp-protocol-version")
+        if pv and pv not in SUPPORTED_PROTOCOL_VERSIONS:
+            return JSONResponse(
+                {"detail": f"unsupported MCP protocol version {pv!r}; "
+                           f"supported: {list(SUPPORTED_PROTOCOL_VERSIONS)}"},
+                status_code=400)
         raw = await request.body()
         session_id = request.headers.get("mcp-session-id")
         sess = engine.sessions.get(session_id) if session_id else None

@@ -711,3 +711,24 @@ def test_respond_stream_delivers_broadcast(run):
         assert [m["n"] for m in got] == [1, 2]
 
     run(go())
+
+
+def test_mcp_protocol_version_guard(client_engine, run):
+    """reference: MCPProtocolVersionMiddleware — unsupported declared
+    versions are rejected before dispatch; supported ones pass."""
+    client_factory, engine, app = client_engine
+
+    async def go():
+        async with client_factory() as c:
+            body = {"jsonrpc": "2.0", "id": 1, "method": "ping"}
+            r = await c.post("/mcp", json=body, headers={**ADMIN,
+                             "MCP-Protocol-Version": "1999-01-01"})
+            assert r.status_code == 400
+            assert "unsupported" in r.text
+            r2 = await c.post("/mcp", json=body, headers={**ADMIN,
+                              "MCP-Protocol-Version": "2025-06-18"})
+            assert r2.status_code == 200
+            r3 = await c.post("/mcp", json=body, headers=ADMIN)  # absent = fine
+            assert r3.status_code == 200
+
+    run(go())

@@ -126,3 +126,31 @@ def test_dcr_registration():
             await asyncio.wait_for(idp_task, timeout=10)
 
     asyncio.run(go())
+
+
+def test_templated_provider_presets():
+    """Entra/Okta/Keycloak presets build their real endpoint URLs from
+    tenant/domain/realm parameters (reference: sso_service specifics)."""
+    from mcp_context_forge_amd.auth.service import AuthService
+    from mcp_context_forge_amd.auth.sso import SSOError, SSOService
+    from mcp_context_forge_amd.config import Settings
+    from mcp_context_forge_amd.db.engine import Database
+
+    db = Database("sqlite://")
+    db.migrate()
+    svc = SSOService(AuthService(db, Settings(database_url="sqlite://")), Settings())
+    e = svc.register_provider("corp-entra", "cid", "cs", preset="entra", tenant="contoso-id")
+    assert e["authorize_url"] == \
+        "https://login.microsoftonline.com/contoso-id/oauth2/v2.0/authorize"
+    assert e["userinfo_url"] == "https://graph.microsoft.com/oidc/userinfo"
+    o = svc.register_provider("corp-okta", "cid", "cs", preset="okta", domain="corp.okta.com")
+    assert o["token_url"] == "https://corp.okta.com/oauth2/v1/token"
+    k = svc.register_provider("kc", "cid", "cs", preset="keycloak",
+                              base="https://kc.corp", realm="main")
+    assert k["authorize_url"] == "https://kc.corp/realms/main/protocol/openid-connect/auth"
+    # missing template parameter → clear error
+    import pytest as _pytest
+
+    with _pytest.raises(SSOError, match="needs tenant"):
+        svc.register_provider("bad", "cid", "cs", preset="entra")
+    db.close()
