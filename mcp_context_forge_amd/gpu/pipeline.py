@@ -814,9 +814,11 @@ class GpuPluginPipeline:
         rewrite_dispatch: List[Tuple[int, Any]] = []
         if rewrite_js:
             self.slow_path += len(rewrite_js)
+            t_rw = self._tic()
             rewrite_dispatch = await self._rewrite_pass(
                 blob, env, rows, args_b, args_e, rewrite_js, tool_idx, hit, hit_slot, responses,
                 pii_m=pii_m, regex_m=regex_m, norm_m=norm_m)
+            self._toc("rewrite_pass", t_rw)
 
         self._toc("answer_assign", t_a)
         py_items = [(int(j), None) for j in np.nonzero(state == hip.ST_DISPATCH_PY)[0]] + \
