@@ -39,7 +39,7 @@ def _esc_variants(word: str, rng: random.Random):
 def gen_adversarial(rng: random.Random, tool_names):
     """One adversarial tools/call request as RAW bytes (hand-assembled so
     formatting is NOT canonical: random whitespace, unsorted keys)."""
-    kind = rng.randrange(10)
+    kind = rng.randrange(12)
     name = rng.choice(tool_names)
     args = {}
     if kind == 0:    # escape-hidden deny word
@@ -72,6 +72,14 @@ def gen_adversarial(rng: random.Random, tool_names):
     elif kind == 8:  # long strings + odd keys
         args = {"k" * rng.randrange(1, 30): "z" * rng.randrange(100, 2000),
                 "under_score": rng.randrange(1000)}
+    elif kind == 10:  # duplicate keys: banned word in the DISCARDED value
+        word = rng.choice(DENY_WORDS + HARM_PHRASES)
+        return _assemble(name, {}, rng,
+                         raw_args='{"msg":"%s","msg":"benign wins"}' % word)
+    elif kind == 11:  # duplicate keys: banned word in the KEPT value
+        word = rng.choice(DENY_WORDS + HARM_PHRASES)
+        return _assemble(name, {}, rng,
+                         raw_args='{"msg":"benign loses","msg":"has %s here"}' % word)
     else:            # benign mixed types
         args = {"n": rng.randrange(100), "f": rng.random(), "b": rng.random() < 0.5,
                 "s": "".join(rng.choice(string.printable[:80]) for _ in range(rng.randrange(1, 40)))}
@@ -205,3 +213,26 @@ def test_adversarial_parity(moderation, n_cases, seed):
         await gpu.shutdown()
 
     asyncio.run(go())
+
+
+def test_cpu_duplicate_key_semantics(run):
+    """json.loads keeps the LAST duplicate key — the CPU oracle dispatches
+    when the banned word lives only in the discarded value, and blocks when
+    it lives in the kept one. (The GPU path confirms raw deny/harm hits on
+    decoded text via the host lane for exactly this reason.)"""
+
+    async def go():
+        e = await build_pair(gpu_mod=False, moderation=False)
+        raw_discarded = ('{"jsonrpc":"2.0","id":1,"method":"tools/call","params":'
+                         '{"name":"fast-time-echo","arguments":'
+                         '{"msg":"forbidden","msg":"benign wins"}}}').encode()
+        raw_kept = ('{"jsonrpc":"2.0","id":2,"method":"tools/call","params":'
+                    '{"name":"fast-time-echo","arguments":'
+                    '{"msg":"benign loses","msg":"has forbidden here"}}}').encode()
+        o1, o2 = await e.process_rpc_batch([raw_discarded, raw_kept])
+        assert "result" in json.loads(o1), o1      # discarded value → no block
+        r2 = json.loads(o2)
+        assert "error" in r2 and r2["error"]["code"] == -32003, o2
+        await e.shutdown()
+
+    run(go())
