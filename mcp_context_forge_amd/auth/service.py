@@ -53,6 +53,7 @@ class AuthContext:
     teams: List[str] = field(default_factory=list)
     scopes: List[str] = field(default_factory=list)
     server_id: Optional[str] = None  # token scoped to one virtual server
+    credential: Optional[str] = None  # usage-accounting key (token id / method)
 
 
 class AuthError(Exception):
@@ -165,7 +166,7 @@ class AuthService:
             u = s.get(DbUser, t.user_email)
             return AuthContext(user=t.user_email, is_admin=bool(u and u.is_admin), auth_method="api_token",
                                scopes=list(t.scopes or []), server_id=t.server_id,
-                               teams=self._teams(s, t.user_email))
+                               teams=self._teams(s, t.user_email), credential=f"token:{t.id}")
 
     def revoke_api_token(self, token_id: str) -> bool:
         with self.db.session() as s:
@@ -213,7 +214,8 @@ class AuthService:
             return AuthContext(user=user, is_admin=bool(claims.get("admin") or user == self.settings.platform_admin_email
                                                         or user == self.settings.basic_auth_user),
                                auth_method="jwt", scopes=claims.get("scopes") or [],
-                               server_id=(claims.get("server_id")))
+                               server_id=(claims.get("server_id")),
+                               credential=f"jwt:{claims.get('jti') or user}")
         if scheme == "basic" and basic_ok and value:
             try:
                 decoded = base64.b64decode(value).decode()

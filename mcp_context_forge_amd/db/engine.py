@@ -110,6 +110,11 @@ def _rev_0006_leader_leases(conn: Connection) -> None:
         "name VARCHAR(128) PRIMARY KEY, holder VARCHAR(255), expires_at FLOAT)")
 
 
+def _rev_0007_token_usage(conn: Connection) -> None:
+    """Hourly-bucketed credential usage (reference: TokenUsageLog)."""
+    Base.metadata.tables["token_usage"].create(conn, checkfirst=True)
+
+
 # Linear chain: (revision_id, apply_fn). Append-only.
 MIGRATIONS: List[Tuple[str, Callable[[Connection], None]]] = [
     ("0001_initial_registry", _rev_0001_initial),
@@ -118,6 +123,7 @@ MIGRATIONS: List[Tuple[str, Callable[[Connection], None]]] = [
     ("0004_gateway_lifecycle", _rev_0004_gateway_lifecycle),
     ("0005_oauth_tokens", _rev_0005_oauth_tokens),
     ("0006_leader_leases", _rev_0006_leader_leases),
+    ("0007_token_usage", _rev_0007_token_usage),
 ]
 
 

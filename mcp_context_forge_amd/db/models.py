@@ -263,6 +263,20 @@ class DbOAuthToken(Base):
     updated_at: Mapped[datetime.datetime] = mapped_column(DateTime, default=utcnow, onupdate=utcnow)
 
 
+class DbTokenUsage(Base):
+    """Hourly-bucketed per-credential usage (reference: TokenUsageLog
+    db.py:5584 + TokenUsageMiddleware — here aggregated in memory and
+    flushed, same shape as the metrics buffer)."""
+
+    __tablename__ = "token_usage"
+
+    id: Mapped[str] = mapped_column(String(36), primary_key=True, default=_uuid)
+    bucket: Mapped[datetime.datetime] = mapped_column(DateTime, index=True)
+    credential: Mapped[str] = mapped_column(String(255), index=True)  # token id or auth method
+    user_email: Mapped[Optional[str]] = mapped_column(String(255), nullable=True)
+    requests: Mapped[int] = mapped_column(Integer, default=0)
+
+
 class DbSessionRecord(Base):
     """Transport session (reference: db.py:5311) — DB backend of the session registry."""
 

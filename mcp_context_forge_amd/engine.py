@@ -61,6 +61,9 @@ class GatewayEngine:
             self.plugins = PluginManager([], enabled=False)
         self.metrics = MetricsBuffer(self.db, self.settings.metrics_buffer_flush_interval,
                                      self.settings.metrics_buffer_max_size)
+        from .services.metrics import TokenUsageTracker
+
+        self.token_usage = TokenUsageTracker(self.db)
         self.tool_service = ToolService(self.registry, self.plugins, self.metrics,
                                         max_retries=self.settings.max_tool_retries)
         self.gateway_service = GatewayService(self.registry, self.tool_service, self.settings, rank, world_size)
@@ -160,6 +163,7 @@ class GatewayEngine:
                 pass
             try:
                 self.metrics.maybe_flush()
+                self.token_usage.flush()
                 self.sessions.cleanup_expired()
                 self.observability.flush()
                 self.token_blocklist.purge_expired()

@@ -182,3 +182,58 @@ def rollup_hourly(db: Database, entity_type: str = "tool") -> int:
             created += 1
         s.execute(delete(DbToolMetric))
     return created
+
+
+class TokenUsageTracker:
+    """Per-credential request accounting, hourly-bucketed (reference:
+    TokenUsageLog db.py:5584 + TokenUsageMiddleware). record() is a dict
+    increment under a lock — safe on the hot paths; flush() upserts the
+    current hour's rows (same buffered shape as MetricsBuffer)."""
+
+    def __init__(self, db: Optional[Database] = None):
+        self.db = db
+        self._counts: Dict[tuple, int] = {}
+        self._lock = threading.Lock()
+
+    def record(self, credential: str, user: Optional[str] = None, n: int = 1) -> None:
+        key = (credential or "anonymous", user)
+        with self._lock:
+            self._counts[key] = self._counts.get(key, 0) + n
+
+    def flush(self) -> int:
+        with self._lock:
+            counts, self._counts = self._counts, {}
+        if not counts or self.db is None:
+            return 0
+        from sqlalchemy import select
+
+        from ..db.models import DbTokenUsage, utcnow
+
+        bucket = utcnow().replace(minute=0, second=0, microsecond=0)
+        with self.db.session() as s:
+            for (credential, user), n in counts.items():
+                row = s.execute(select(DbTokenUsage).where(
+                    DbTokenUsage.bucket == bucket,
+                    DbTokenUsage.credential == credential)).scalar_one_or_none()
+                if row is None:
+                    s.add(DbTokenUsage(bucket=bucket, credential=credential,
+                                       user_email=user, requests=n))
+                else:
+                    row.requests += n
+        return len(counts)
+
+    def query(self, limit: int = 200) -> List[Dict[str, Any]]:
+        self.flush()
+        if self.db is None:
+            return []
+        from sqlalchemy import select
+
+        from ..db.models import DbTokenUsage
+
+        with self.db.session() as s:
+            rows = s.execute(select(DbTokenUsage)
+                             .order_by(DbTokenUsage.bucket.desc(),
+                                       DbTokenUsage.requests.desc())
+                             .limit(limit)).scalars().all()
+            return [{"bucket": str(r.bucket), "credential": r.credential,
+                     "user": r.user_email, "requests": r.requests} for r in rows]

@@ -217,6 +217,7 @@ class RpcFastPath:
                 return
             if not msg.get("more_body", False):
                 break
+        st.engine.token_usage.record(ctx.credential or ctx.auth_method, ctx.user)
         collector = st.collector
         if collector is not None and ctx.server_id is None:
             out = await collector.submit(body, user=ctx.user)
@@ -332,6 +333,7 @@ def build_app(engine: GatewayEngine, auth: Optional[AuthService] = None) -> Fast
         if ctx.server_id is not None and not _server_scope_ok(request.url.path, ctx.server_id):
             raise HTTPException(status_code=403,
                                 detail=f"token is scoped to server {ctx.server_id}")
+        engine.token_usage.record(ctx.credential or ctx.auth_method, ctx.user)
         return ctx
 
     def require(permission: str):
@@ -1083,6 +1085,13 @@ def build_app(engine: GatewayEngine, auth: Optional[AuthService] = None) -> Fast
     @app.get("/admin/audit")
     async def admin_audit(limit: int = 100, ctx: AuthContext = Depends(require("admin.read"))):
         return engine.audit.query(limit)
+
+    @app.get("/admin/token-usage")
+    async def admin_token_usage(limit: int = 200,
+                                ctx: AuthContext = Depends(require("admin.read"))):
+        """Per-credential hourly usage (reference: TokenUsageLog +
+        TokenUsageMiddleware)."""
+        return engine.token_usage.query(limit)
 
     @app.get("/admin/metrics/rollups")
     async def admin_rollups(ctx: AuthContext = Depends(require("admin.read"))):

@@ -203,6 +203,11 @@ class NativeEdge:
                 cold.append(i)
 
         if hot_idx:
+            # usage accounting for the C++-authenticated hot lane (one
+            # increment per request; credential granularity is the user —
+            # the native cache intentionally does not carry token ids)
+            for u in hot_users:
+                self.engine.token_usage.record(f"user:{u}" if u else "anonymous", u)
             raws = [bodies[i] for i in hot_idx]
             results = await self.engine.process_rpc_batch(raws, users=hot_users)
             for i, r in zip(hot_idx, results):

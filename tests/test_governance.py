@@ -273,3 +273,40 @@ def test_teams_roots_tasks_and_batch_metrics(run):
         await e.shutdown()
 
     run(go())
+
+
+def test_token_usage_accounting(run):
+    """reference: TokenUsageLog + TokenUsageMiddleware — per-credential
+    hourly usage flows from the auth paths into /admin/token-usage."""
+    import base64
+
+    import httpx
+
+    from mcp_context_forge_amd.config import Settings
+    from mcp_context_forge_amd.engine import GatewayEngine
+    from mcp_context_forge_amd.transports.http_app import build_app
+
+    ADMIN_H = {"Authorization": "Basic " + base64.b64encode(b"admin:changeme").decode()}
+
+    async def go():
+        e = GatewayEngine(Settings(database_url="sqlite://", federation_enabled=False,
+                                   auth_required=True, plugins_enabled=False, gpu_enabled=False))
+        app = build_app(e)
+        async with app.router.lifespan_context(app):
+            async with httpx.AsyncClient(transport=httpx.ASGITransport(app=app),
+                                         base_url="http://gw") as c:
+                raw = app.state.auth.create_api_token("admin@example.com", "usage-tok")
+                for i in range(5):
+                    r = await c.post("/rpc", headers={"Authorization": f"Bearer {raw}"},
+                                     json={"jsonrpc": "2.0", "id": i, "method": "ping"})
+                    assert r.status_code == 200
+                await c.get("/version", headers=ADMIN_H)
+                rows = (await c.get("/admin/token-usage", headers=ADMIN_H)).json()
+                by_cred = {r["credential"]: r for r in rows}
+                tok_rows = [r for r in rows if r["credential"].startswith("token:")]
+                assert tok_rows and tok_rows[0]["requests"] >= 5
+                assert tok_rows[0]["user"] == "admin@example.com"
+                assert any(c0.startswith("basic") or c0.startswith("jwt") or c0 == "basic"
+                           for c0 in by_cred), by_cred
+
+    run(go())
