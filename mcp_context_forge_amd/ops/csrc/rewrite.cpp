@@ -1,0 +1,583 @@
+// Native rewrite pass: the flagged-row exact semantics in C++.
+//
+// Reference behavior being replicated (the CPU chain order, plugins/
+// builtin.py): argument_normalizer@15 (whitespace collapse + strip; NFC is
+// identity on ASCII) → pii_filter@30 (six fixed patterns, re.sub global
+// replace, patterns applied sequentially on the already-substituted text)
+// over every STRING value of the parsed JSON arguments, then canonical
+// re-serialization (json.dumps sort_keys=True, separators=(",",":"),
+// ensure_ascii=True).
+//
+// ENVELOPE: a row is handled natively ONLY when byte-exact equivalence
+// with the Python path is provable:
+//   * pure-ASCII span (NFC identity; ensure_ascii re-escaping identity)
+//   * JSON parses, depth <= 32, numbers are plain integers (float repr
+//     round-trips are Python-specific)
+//   * no regex_filter work requested (user-configured Python regexes)
+//   * string escapes limited to \" \\ \/ \b \f \n \r \t and \u00XX (ASCII)
+// Anything else returns PUNT and the row takes the existing Python path.
+// tests/test_rewrite_native.py cross-validates against the Python
+// implementation over randomized corpora; the GPU parity fuzzer covers
+// the full pipeline.
+
+#ifndef _GNU_SOURCE
+#define _GNU_SOURCE
+#endif
+#include <stdint.h>
+#include <stdio.h>
+#include <stdlib.h>
+#include <string.h>
+
+#include <algorithm>
+#include <string>
+#include <vector>
+
+namespace {
+
+constexpr int MAX_DEPTH = 32;
+
+inline bool is_word(uint8_t c) {  // Python re \w (ASCII)
+    return (c >= 'A' && c <= 'Z') || (c >= 'a' && c <= 'z') || (c >= '0' && c <= '9') || c == '_';
+}
+inline bool is_digit(uint8_t c) { return c >= '0' && c <= '9'; }
+inline bool is_alpha(uint8_t c) { return (c >= 'A' && c <= 'Z') || (c >= 'a' && c <= 'z'); }
+
+// ---------------------------------------------------------------- PII
+
+// word boundary \b at position i of s (between i-1 and i)
+inline bool wb(const std::string& s, size_t i) {
+    bool l = i > 0 && is_word((uint8_t)s[i - 1]);
+    bool r = i < s.size() && is_word((uint8_t)s[i]);
+    return l != r;
+}
+
+inline bool klass_email_local(uint8_t c) {
+    return is_word(c) || c == '.' || c == '%' || c == '+' || c == '-';
+}
+inline bool klass_email_domain(uint8_t c) {
+    return is_alpha(c) || is_digit(c) || c == '.' || c == '-';
+}
+inline bool is_sep_sd(uint8_t c) { return c == ' ' || c == '-'; }          // [ \-]
+inline bool is_sep_pd(uint8_t c) { return c == ' ' || c == '.' || c == '-'; }  // [ .\-]
+
+// Each matcher: try a match STARTING at i; on success set len. Mirrors the
+// exact python regex (incl. \b positions).
+// ssn: \b\d{3}-\d{2}-\d{4}\b
+bool m_ssn(const std::string& s, size_t i, size_t& len) {
+    if (!wb(s, i)) return false;
+    if (i + 11 > s.size()) return false;
+    const char* p = s.data() + i;
+    for (int k = 0; k < 3; ++k) if (!is_digit(p[k])) return false;
+    if (p[3] != '-') return false;
+    for (int k = 4; k < 6; ++k) if (!is_digit(p[k])) return false;
+    if (p[6] != '-') return false;
+    for (int k = 7; k < 11; ++k) if (!is_digit(p[k])) return false;
+    if (!wb(s, i + 11)) return false;
+    len = 11;
+    return true;
+}
+
+// email: [\w.%+\-]+@[A-Za-z0-9.\-]+\.[A-Za-z]{2,}   (no \b)
+bool m_email(const std::string& s, size_t i, size_t& len) {
+    size_t j = i;
+    while (j < s.size() && klass_email_local((uint8_t)s[j])) ++j;
+    if (j == i || j >= s.size() || s[j] != '@') return false;
+    size_t d0 = j + 1, d = d0;
+    while (d < s.size() && klass_email_domain((uint8_t)s[d])) ++d;
+    if (d == d0) return false;
+    // backtrack: rightmost '.' inside [d0, d) with >=2 alphas after it
+    // (alphas counted within the maximal domain run — greedy {2,} then
+    // nothing follows, so the match ends at the end of the alpha run)
+    for (size_t dot = d; dot-- > d0 + 1;) {  // dot position candidate; needs chars before it
+        if (s[dot] != '.') continue;
+        size_t a = dot + 1, e = a;
+        while (e < d && is_alpha((uint8_t)s[e])) ++e;
+        if (e - a >= 2) {
+            len = e - i;
+            return true;
+        }
+    }
+    return false;
+}
+
+// credit_card: \b\d{4}[ \-]\d{4}[ \-]\d{4}[ \-]\d{4}\b
+bool m_cc(const std::string& s, size_t i, size_t& len) {
+    if (!wb(s, i)) return false;
+    if (i + 19 > s.size()) return false;
+    const char* p = s.data() + i;
+    for (int g = 0; g < 4; ++g) {
+        for (int k = 0; k < 4; ++k)
+            if (!is_digit(p[g * 5 + k])) return false;
+        if (g < 3 && !is_sep_sd((uint8_t)p[g * 5 + 4])) return false;
+    }
+    if (!wb(s, i + 19)) return false;
+    len = 19;
+    return true;
+}
+
+// phone: \(?\b\d{3}\)?[ .\-]\d{3}[ .\-]\d{4}\b
+bool m_phone(const std::string& s, size_t i, size_t& len) {
+    size_t j = i;
+    if (j < s.size() && s[j] == '(') ++j;          // \(? greedy
+    if (!wb(s, j)) {
+        // backtrack \(? to zero-width only if '(' consumed
+        if (j == i) return false;
+        j = i;
+        if (!wb(s, j)) return false;
+    }
+    if (j + 3 > s.size()) return false;
+    for (int k = 0; k < 3; ++k) if (!is_digit((uint8_t)s[j + k])) return false;
+    size_t k2 = j + 3;
+    if (k2 < s.size() && s[k2] == ')') ++k2;       // \)? greedy (no backtrack needed:
+    if (k2 >= s.size() || !is_sep_pd((uint8_t)s[k2])) {
+        // backtrack \)? — try without consuming ')'
+        if (k2 > j + 3) {
+            k2 = j + 3;
+            if (k2 >= s.size() || !is_sep_pd((uint8_t)s[k2])) return false;
+        } else {
+            return false;
+        }
+    }
+    ++k2;
+    if (k2 + 3 > s.size()) return false;
+    for (int k = 0; k < 3; ++k) if (!is_digit((uint8_t)s[k2 + k])) return false;
+    k2 += 3;
+    if (k2 >= s.size() || !is_sep_pd((uint8_t)s[k2])) return false;
+    ++k2;
+    if (k2 + 4 > s.size()) return false;
+    for (int k = 0; k < 4; ++k) if (!is_digit((uint8_t)s[k2 + k])) return false;
+    k2 += 4;
+    if (!wb(s, k2)) return false;
+    len = k2 - i;
+    return true;
+}
+
+// ipv4: \b\d{1,3}\.\d{1,3}\.\d{1,3}\.\d{1,3}\b  (greedy {1,3} with backtrack)
+bool m_ipv4(const std::string& s, size_t i, size_t& len) {
+    if (!wb(s, i)) return false;
+    size_t j = i;
+    // octets 1..3: digits{1..3} then '.'; greedy semantics: take max digits
+    // (<=3) such that the next char is '.'; since digits can't be '.', the
+    // run length is fixed by the input — just bound it
+    for (int oct = 0; oct < 3; ++oct) {
+        size_t d = 0;
+        while (j + d < s.size() && is_digit((uint8_t)s[j + d]) && d < 3) ++d;
+        if (d == 0) return false;
+        if (j + d >= s.size() || s[j + d] != '.') return false;
+        // python would fail if a 4th digit precedes '.', because {1,3} can't
+        // cover it and '.' won't match a digit — detect: next char after the
+        // run of <=3 must be '.', but if the digit run continues past 3 the
+        // char at j+3 is a digit, not '.', and backtracking to shorter runs
+        // still faces a digit → no match at this i
+        j += d + 1;
+    }
+    size_t d = 0;
+    while (j + d < s.size() && is_digit((uint8_t)s[j + d]) && d < 3) ++d;
+    if (d == 0) return false;
+    // \b after: if a 4th digit follows, {1,3} backtracks: \b between digit
+    // and digit fails for d=3,2,1 → no match
+    if (j + d < s.size() && is_digit((uint8_t)s[j + d])) return false;
+    j += d;
+    if (!wb(s, j)) return false;
+    len = j - i;
+    return true;
+}
+
+// aws_key: \bAKIA[0-9A-Z]{16}\b
+bool m_aws(const std::string& s, size_t i, size_t& len) {
+    if (!wb(s, i)) return false;
+    if (i + 20 > s.size()) return false;
+    if (memcmp(s.data() + i, "AKIA", 4) != 0) return false;
+    for (int k = 4; k < 20; ++k) {
+        uint8_t c = (uint8_t)s[i + k];
+        if (!((c >= '0' && c <= '9') || (c >= 'A' && c <= 'Z'))) return false;
+    }
+    if (!wb(s, i + 20)) return false;
+    len = 20;
+    return true;
+}
+
+typedef bool (*Matcher)(const std::string&, size_t, size_t&);
+struct PiiDef { const char* name; const char* repl; Matcher fn; };
+const PiiDef PII[6] = {
+    {"ssn", "[SSN_REDACTED]", m_ssn},
+    {"email", "[EMAIL_REDACTED]", m_email},
+    {"credit_card", "[CREDIT_CARD_REDACTED]", m_cc},
+    {"phone", "[PHONE_REDACTED]", m_phone},
+    {"ipv4", "[IPV4_REDACTED]", m_ipv4},
+    {"aws_key", "[AWS_KEY_REDACTED]", m_aws},
+};
+
+// re.sub semantics: left-to-right non-overlapping global replace
+bool pii_sub_one(std::string& s, const PiiDef& p) {
+    std::string out;
+    bool any = false;
+    size_t i = 0;
+    while (i < s.size()) {
+        size_t len = 0;
+        if (p.fn(s, i, len)) {
+            out += p.repl;
+            i += len;
+            any = true;
+        } else {
+            out += s[i++];
+        }
+    }
+    if (any) s.swap(out);
+    return any;
+}
+
+// patterns applied sequentially over the already-substituted text, in the
+// fixed order, gated by active_mask (config categories) and want_mask
+// (the GPU scan's per-row accept bits — mask_text_subset semantics)
+uint32_t pii_mask_text(std::string& s, uint32_t active_mask, uint32_t want_mask) {
+    uint32_t found = 0;
+    for (int k = 0; k < 6; ++k) {
+        if (!((active_mask >> k) & 1) || !((want_mask >> k) & 1)) continue;
+        if (pii_sub_one(s, PII[k])) found |= 1u << k;
+    }
+    return found;
+}
+
+// ---------------------------------------------------------------- normalize
+
+// argument_normalizer.norm for ASCII: collapse [ \t\f\v]+ → " ", then strip()
+// (python str.strip removes all whitespace incl. \n)
+void norm_ascii(std::string& s, bool collapse, bool strip) {
+    if (collapse) {
+        std::string out;
+        out.reserve(s.size());
+        for (size_t i = 0; i < s.size();) {
+            char c = s[i];
+            if (c == ' ' || c == '\t' || c == '\f' || c == '\v') {
+                out += ' ';
+                while (i < s.size() && (s[i] == ' ' || s[i] == '\t' || s[i] == '\f' || s[i] == '\v')) ++i;
+            } else {
+                out += c;
+                ++i;
+            }
+        }
+        s.swap(out);
+    }
+    if (strip) {
+        size_t a = 0, b = s.size();
+        auto is_ws = [](char c) {
+            return c == ' ' || c == '\t' || c == '\n' || c == '\r' || c == '\f' || c == '\v';
+        };
+        while (a < b && is_ws(s[a])) ++a;
+        while (b > a && is_ws(s[b - 1])) --b;
+        s = s.substr(a, b - a);
+    }
+}
+
+// ---------------------------------------------------------------- JSON
+
+struct Val;
+struct Member { std::string key; size_t vidx; };
+struct Val {
+    enum Kind { OBJ, ARR, STR, INT, TRUE_, FALSE_, NULL_ } kind;
+    std::string str;                 // STR: decoded text; INT: literal digits
+    std::vector<Member> members;     // OBJ
+    std::vector<size_t> items;       // ARR
+};
+
+struct Parser {
+    const uint8_t* p;
+    const uint8_t* e;
+    std::vector<Val>& pool;
+    bool ok = true;
+
+    void skip_ws() {
+        while (p < e && (*p == ' ' || *p == '\t' || *p == '\n' || *p == '\r')) ++p;
+    }
+
+    bool decode_string(std::string& out) {
+        // at opening quote
+        if (p >= e || *p != '"') return false;
+        ++p;
+        while (p < e) {
+            uint8_t c = *p;
+            if (c == '"') { ++p; return true; }
+            if (c >= 0x80) return false;   // non-ASCII → punt
+            if (c == '\\') {
+                if (p + 1 >= e) return false;
+                uint8_t n = p[1];
+                p += 2;
+                switch (n) {
+                    case '"': out += '"'; break;
+                    case '\\': out += '\\'; break;
+                    case '/': out += '/'; break;
+                    case 'b': out += '\b'; break;
+                    case 'f': out += '\f'; break;
+                    case 'n': out += '\n'; break;
+                    case 'r': out += '\r'; break;
+                    case 't': out += '\t'; break;
+                    case 'u': {
+                        if (p + 4 > e) return false;
+                        unsigned v = 0;
+                        for (int k = 0; k < 4; ++k) {
+                            uint8_t h = p[k];
+                            v <<= 4;
+                            if (h >= '0' && h <= '9') v |= h - '0';
+                            else if (h >= 'a' && h <= 'f') v |= h - 'a' + 10;
+                            else if (h >= 'A' && h <= 'F') v |= h - 'A' + 10;
+                            else return false;
+                        }
+                        if (v > 0x7F) return false;  // decodes to non-ASCII → punt
+                        out += (char)v;
+                        p += 4;
+                        break;
+                    }
+                    default: return false;
+                }
+            } else {
+                out += (char)c;
+                ++p;
+            }
+        }
+        return false;
+    }
+
+    size_t parse_value(int depth) {
+        if (depth > MAX_DEPTH) { ok = false; return 0; }
+        skip_ws();
+        if (p >= e) { ok = false; return 0; }
+        size_t idx = pool.size();
+        pool.emplace_back();
+        uint8_t c = *p;
+        if (c == '{') {
+            pool[idx].kind = Val::OBJ;
+            ++p;
+            skip_ws();
+            if (p < e && *p == '}') { ++p; return idx; }
+            while (ok) {
+                skip_ws();
+                std::string key;
+                if (!decode_string(key)) { ok = false; return 0; }
+                skip_ws();
+                if (p >= e || *p != ':') { ok = false; return 0; }
+                ++p;
+                size_t v = parse_value(depth + 1);
+                if (!ok) return 0;
+                // json.loads: last duplicate key wins
+                bool dup = false;
+                for (auto& m : pool[idx].members)
+                    if (m.key == key) { m.vidx = v; dup = true; break; }
+                if (!dup) pool[idx].members.push_back({std::move(key), v});
+                skip_ws();
+                if (p < e && *p == ',') { ++p; continue; }
+                if (p < e && *p == '}') { ++p; return idx; }
+                ok = false;
+                return 0;
+            }
+            return 0;
+        }
+        if (c == '[') {
+            pool[idx].kind = Val::ARR;
+            ++p;
+            skip_ws();
+            if (p < e && *p == ']') { ++p; return idx; }
+            while (ok) {
+                size_t v = parse_value(depth + 1);
+                if (!ok) return 0;
+                pool[idx].items.push_back(v);
+                skip_ws();
+                if (p < e && *p == ',') { ++p; continue; }
+                if (p < e && *p == ']') { ++p; return idx; }
+                ok = false;
+                return 0;
+            }
+            return 0;
+        }
+        if (c == '"') {
+            pool[idx].kind = Val::STR;
+            if (!decode_string(pool[idx].str)) { ok = false; return 0; }
+            return idx;
+        }
+        if (c == 't') {
+            if (e - p < 4 || memcmp(p, "true", 4)) { ok = false; return 0; }
+            pool[idx].kind = Val::TRUE_;
+            p += 4;
+            return idx;
+        }
+        if (c == 'f') {
+            if (e - p < 5 || memcmp(p, "false", 5)) { ok = false; return 0; }
+            pool[idx].kind = Val::FALSE_;
+            p += 5;
+            return idx;
+        }
+        if (c == 'n') {
+            if (e - p < 4 || memcmp(p, "null", 4)) { ok = false; return 0; }
+            pool[idx].kind = Val::NULL_;
+            p += 4;
+            return idx;
+        }
+        if (c == '-' || is_digit(c)) {
+            pool[idx].kind = Val::INT;
+            const uint8_t* s0 = p;
+            if (*p == '-') ++p;
+            if (p >= e || !is_digit(*p)) { ok = false; return 0; }
+            while (p < e && is_digit(*p)) ++p;
+            if (p < e && (*p == '.' || *p == 'e' || *p == 'E')) { ok = false; return 0; }  // float → punt
+            // leading zeros: json.loads rejects 0123 → punt (stay exact)
+            size_t dn = (size_t)(p - s0) - (s0[0] == '-' ? 1 : 0);
+            const uint8_t* dp = s0 + (s0[0] == '-' ? 1 : 0);
+            if (dn > 1 && dp[0] == '0') { ok = false; return 0; }
+            // int canonical form == literal digits (python int() round-trip)
+            // except "-0" → "0"
+            if (dn == 1 && dp[0] == '0' && s0[0] == '-')
+                pool[idx].str = "0";
+            else
+                pool[idx].str.assign((const char*)s0, (size_t)(p - s0));
+            return idx;
+        }
+        ok = false;
+        return 0;
+    }
+};
+
+void append_escaped(std::string& out, const std::string& s) {
+    // json.dumps ensure_ascii on ASCII input: escape " \ and control chars
+    for (char ch : s) {
+        uint8_t c = (uint8_t)ch;
+        switch (c) {
+            case '"': out += "\\\""; break;
+            case '\\': out += "\\\\"; break;
+            case '\b': out += "\\b"; break;
+            case '\f': out += "\\f"; break;
+            case '\n': out += "\\n"; break;
+            case '\r': out += "\\r"; break;
+            case '\t': out += "\\t"; break;
+            default:
+                if (c < 0x20) {
+                    char b[8];
+                    snprintf(b, sizeof(b), "\\u%04x", c);
+                    out += b;
+                } else {
+                    out += (char)c;
+                }
+        }
+    }
+}
+
+void serialize(const std::vector<Val>& pool, size_t idx, std::string& out) {
+    const Val& v = pool[idx];
+    switch (v.kind) {
+        case Val::OBJ: {
+            // sort_keys=True: ASCII keys sort identically by bytes
+            std::vector<const Member*> ms;
+            ms.reserve(v.members.size());
+            for (auto& m : v.members) ms.push_back(&m);
+            std::sort(ms.begin(), ms.end(),
+                      [](const Member* a, const Member* b) { return a->key < b->key; });
+            out += '{';
+            for (size_t k = 0; k < ms.size(); ++k) {
+                if (k) out += ',';
+                out += '"';
+                append_escaped(out, ms[k]->key);
+                out += "\":";
+                serialize(pool, ms[k]->vidx, out);
+            }
+            out += '}';
+            break;
+        }
+        case Val::ARR:
+            out += '[';
+            for (size_t k = 0; k < v.items.size(); ++k) {
+                if (k) out += ',';
+                serialize(pool, v.items[k], out);
+            }
+            out += ']';
+            break;
+        case Val::STR:
+            out += '"';
+            append_escaped(out, v.str);
+            out += '"';
+            break;
+        case Val::INT: out += v.str; break;
+        case Val::TRUE_: out += "true"; break;
+        case Val::FALSE_: out += "false"; break;
+        case Val::NULL_: out += "null"; break;
+    }
+}
+
+}  // namespace
+
+// row status
+enum : int32_t { RW_DONE = 0, RW_PUNT = 1, RW_BLOCKED = 2, RW_BADJSON = 3 };
+
+// Rewrite a batch of flagged rows. Per row:
+//   do_flags bit0 = apply normalizer, bit1 = apply pii  (regex rows must
+//   not be passed here — caller punts them)
+//   pii_want  = per-row GPU accept bits (mask_text_subset gating)
+// Outputs: status[], found_bits[] (pii categories found), canonical
+// rewritten args in the arena (grow-retry contract like forge_decide).
+extern "C" int64_t forge_rewrite_rows(
+    const uint8_t* blob, const int32_t* args_beg, const int32_t* args_end, int n,
+    const uint8_t* do_flags, const uint32_t* pii_want,
+    uint32_t pii_active_mask, int pii_mode /*0 mask, 1 block, 2 audit*/,
+    int norm_collapse, int norm_strip,
+    int32_t* status, uint32_t* found_bits,
+    uint8_t* arena, int64_t arena_cap, int64_t* out_beg, int64_t* out_end)
+{
+    std::string buf;
+    buf.reserve((size_t)n * 64);
+    for (int i = 0; i < n; ++i) {
+        status[i] = RW_PUNT;
+        found_bits[i] = 0;
+        out_beg[i] = out_end[i] = -1;
+        const uint8_t* b = blob + args_beg[i];
+        const uint8_t* e = blob + args_end[i];
+        if (e < b) continue;
+        // pure-ASCII precheck (NFC identity + ensure_ascii identity)
+        bool ascii = true;
+        for (const uint8_t* q = b; q < e; ++q)
+            if (*q >= 0x80) { ascii = false; break; }
+        if (!ascii) continue;  // PUNT
+        std::vector<Val> pool;
+        pool.reserve(32);
+        Parser ps{b, e, pool};
+        ps.skip_ws();
+        if (ps.p >= ps.e) { status[i] = RW_BADJSON; continue; }
+        size_t root = ps.parse_value(0);
+        if (ps.ok) {
+            ps.skip_ws();
+            if (ps.p != ps.e) ps.ok = false;
+        }
+        if (!ps.ok) continue;  // PUNT (floats, deep nesting, bad escapes…)
+        // walk strings in insertion order (json.loads dict order == wire
+        // order; _walk_strings visits values in that order — ordering only
+        // matters for found-category accumulation, which is a set)
+        uint8_t fl = do_flags[i];
+        uint32_t found = 0;
+        for (auto& v : pool) {
+            if (v.kind != Val::STR) continue;
+            if (fl & 1) norm_ascii(v.str, norm_collapse != 0, norm_strip != 0);
+            if (fl & 2) {
+                if (pii_mode == 0) {
+                    found |= pii_mask_text(v.str, pii_active_mask, pii_want[i]);
+                } else {
+                    // block/audit: python computes the substitution then
+                    // discards it — found must reflect substitution-order
+                    // matching, the text must stay unchanged
+                    std::string tmp = v.str;
+                    found |= pii_mask_text(tmp, pii_active_mask, pii_want[i]);
+                }
+            }
+        }
+        found_bits[i] = found;
+        if (found && pii_mode == 1) {
+            status[i] = RW_BLOCKED;
+            continue;
+        }
+        std::string out;
+        serialize(pool, root, out);
+        out_beg[i] = (int64_t)buf.size();
+        buf += out;
+        out_end[i] = (int64_t)buf.size();
+        status[i] = RW_DONE;
+    }
+    if ((int64_t)buf.size() > arena_cap) return -(int64_t)buf.size();
+    if (!buf.empty()) memcpy(arena, buf.data(), buf.size());
+    return (int64_t)buf.size();
+}

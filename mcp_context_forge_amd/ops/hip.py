@@ -44,11 +44,16 @@ def _load() -> ctypes.CDLL:
         "forge_synchronize": [ctypes.c_void_p],
         "forge_parse_envelopes": [ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int] + [ctypes.c_void_p] * 7,
         "forge_upstream_call_batch": [ctypes.c_void_p] * 4 + [ctypes.c_int, ctypes.c_char_p, ctypes.c_void_p, ctypes.c_int64, ctypes.c_void_p, ctypes.c_void_p],
+        "forge_rewrite_rows": [ctypes.c_void_p] * 3 + [ctypes.c_int] + [ctypes.c_void_p] * 2 +
+                              [ctypes.c_uint32] + [ctypes.c_int] * 3 +
+                              [ctypes.c_void_p] * 2 +
+                              [ctypes.c_void_p, ctypes.c_int64, ctypes.c_void_p, ctypes.c_void_p],
     }
     for name, argtypes in protos.items():
         fn = getattr(lib, name)
         fn.argtypes = argtypes
-        fn.restype = ctypes.c_int64 if name == "forge_upstream_call_batch" else ctypes.c_int
+        fn.restype = ctypes.c_int64 if name in ("forge_upstream_call_batch",
+                                                "forge_rewrite_rows") else ctypes.c_int
     return lib
 
 
@@ -433,6 +438,38 @@ def finalize(blob, id_beg, id_end, args_beg, args_end, tool_idx, user_hash,
         if n >= 0:
             return arena, resp_beg2, resp_end2, is_err, cacheable
         cap = -int(n) + 4096
+
+
+RW_DONE, RW_PUNT, RW_BLOCKED, RW_BADJSON = 0, 1, 2, 3
+
+
+def rewrite_rows(blob: np.ndarray, args_beg: np.ndarray, args_end: np.ndarray,
+                 do_flags: np.ndarray, pii_want: np.ndarray,
+                 pii_active_mask: int, pii_mode: int,
+                 norm_collapse: bool, norm_strip: bool):
+    """Native rewrite pass (rewrite.cpp): normalizer + PII over the flagged
+    rows that fit the provable-equivalence envelope; everything else gets
+    RW_PUNT and takes the Python path.
+    → (status i32[n], found u32[n], arena u8, out_beg i64[n], out_end i64[n])."""
+    n = args_beg.shape[0]
+    lib = _load()
+    status = np.empty(n, dtype=np.int32)
+    found = np.empty(n, dtype=np.uint32)
+    out_beg = np.empty(n, dtype=np.int64)
+    out_end = np.empty(n, dtype=np.int64)
+    cap = int((args_end - args_beg).sum()) + n * 32 + 4096
+    while True:
+        arena = np.empty(cap, dtype=np.uint8)
+        rc = lib.forge_rewrite_rows(
+            _np_ptr(blob), _np_ptr(args_beg), _np_ptr(args_end), n,
+            _np_ptr(do_flags), _np_ptr(pii_want),
+            ctypes.c_uint32(pii_active_mask), pii_mode,
+            1 if norm_collapse else 0, 1 if norm_strip else 0,
+            _np_ptr(status), _np_ptr(found),
+            _np_ptr(arena), cap, _np_ptr(out_beg), _np_ptr(out_end))
+        if rc >= 0:
+            return status, found, arena, out_beg, out_end
+        cap = -int(rc) + 4096
 
 
 TF_REACHABLE, TF_DENY, TF_PII, TF_REGEX, TF_NORM, TF_MOD, TF_HARM = (1 << i for i in range(7))
