@@ -170,6 +170,18 @@ class GpuPluginPipeline:
 
         self.max_depth = s.max_json_depth
         self.max_string = s.max_string_length
+
+        # native rewrite lane (rewrite.cpp): fixed-index PII maps for the
+        # C matchers (the plugin's active list may be a config subset)
+        from ..plugins.builtin import PII_PATTERNS as _PIIP
+
+        self._pii_fixed: List[int] = []
+        self._pii_active_mask = 0
+        self._pii_names = [n for (n, _, _) in _PIIP]
+        if self.pii is not None:
+            fixed_index = {n: k for k, (n, _, _) in enumerate(_PIIP)}
+            self._pii_fixed = [fixed_index[a[0]] for a in self.pii.active]
+            self._pii_active_mask = sum(1 << i for i in self._pii_fixed)
         from ..ops.pybridge import get as _pb_get
 
         self._pb = _pb_get()  # C response-assembly loops (fails loudly if missing)
@@ -830,7 +842,7 @@ class GpuPluginPipeline:
             t_rw = self._tic()
             rewrite_dispatch = await self._rewrite_pass(
                 blob, env, rows, args_b, args_e, rewrite_js, tool_idx, hit, hit_slot, responses,
-                pii_m=pii_m, regex_m=regex_m, norm_m=norm_m)
+                pii_m=pii_m, regex_m=regex_m, norm_m=norm_m, deny_cand=cand)
             self._toc("rewrite_pass", t_rw)
 
         self._toc("answer_assign", t_a)
@@ -942,11 +954,76 @@ class GpuPluginPipeline:
 
     async def _rewrite_pass(self, blob, env, rows, args_b, args_e, rewrite_js, tool_idx,
                             hit, hit_slot, responses,
-                            pii_m=None, regex_m=None, norm_m=None) -> List[Tuple[int, Any]]:
+                            pii_m=None, regex_m=None, norm_m=None,
+                            deny_cand=None) -> List[Tuple[int, Any]]:
         from ..plugins.builtin import _text_of
 
         ok_items: List[Tuple[int, Any]] = []
-        for j in rewrite_js:
+        scan_bytes: Dict[int, bytes] = {}  # native lane: sorted-keys scan form
+
+        # --- native lane (rewrite.cpp): rows with no deny candidacy, no
+        # user regexes, and provably-equivalent content (the C side punts
+        # anything outside its envelope back here) ---
+        py_js = rewrite_js
+        if rewrite_js:
+            nat_idx: List[int] = []
+            flags_l: List[int] = []
+            want_l: List[int] = []
+            for j in rewrite_js:
+                if deny_cand is not None and deny_cand[j]:
+                    continue  # deny recheck over decoded text → Python lane
+                mt = self._meta_list[tool_idx[j]]
+                if regex_m is not None and regex_m[j] and \
+                        self._active(self.regex, mt.name, block_class=False):
+                    continue  # user-configured Python regexes → Python lane
+                do_norm = (norm_m is None or bool(norm_m[j])) and \
+                    self._active(self.normalizer, mt.name, block_class=False)
+                do_pii = (pii_m is None or bool(pii_m[j])) and \
+                    self._active(self.pii, mt.name, block_class=False)
+                fl = (1 if do_norm else 0) | (2 if do_pii else 0)
+                bits = int(pii_m[j]) if (pii_m is not None and do_pii) else -1
+                if bits < 0:
+                    want = self._pii_active_mask
+                else:
+                    want = 0
+                    for k, fixed in enumerate(self._pii_fixed):
+                        if (bits >> k) & 1:
+                            want |= 1 << fixed
+                nat_idx.append(j)
+                flags_l.append(fl)
+                want_l.append(want)
+            if nat_idx:
+                pii_mode = 0
+                if self.pii is not None and self.pii.action == "block" and self._enforcing(self.pii):
+                    pii_mode = 1
+                elif self.pii is not None and self.pii.action not in ("mask",):
+                    pii_mode = 2
+                njs = np.asarray(nat_idx, dtype=np.int64)
+                st, found, rw_arena, rb, re_, sb, se = hip.rewrite_rows(
+                    blob, np.ascontiguousarray(args_b[njs]), np.ascontiguousarray(args_e[njs]),
+                    np.asarray(flags_l, dtype=np.uint8), np.asarray(want_l, dtype=np.uint32),
+                    self._pii_active_mask, pii_mode,
+                    bool(self.normalizer and self.normalizer.collapse_ws),
+                    bool(self.normalizer and self.normalizer.strip))
+                punted: List[int] = []
+                for k, j in enumerate(nat_idx):
+                    if st[k] == hip.RW_DONE:
+                        ok_items.append((j, rw_arena[rb[k]:re_[k]].tobytes()))
+                        scan_bytes[j] = rw_arena[sb[k]:se[k]].tobytes()
+                    elif st[k] == hip.RW_BLOCKED:
+                        r = int(rows[j])
+                        idb = self._id_bytes(blob, env, r)
+                        names = sorted(self._pii_names[i] for i in range(len(self._pii_names))
+                                       if (int(found[k]) >> i) & 1)
+                        responses[r] = self._splice_error(
+                            idb, jsonrpc.POLICY_DENIED, f"pii_filter: PII detected: {names}")
+                        self.blocked += 1
+                    else:
+                        punted.append(j)
+                py_js = [j for j in rewrite_js if j not in set(nat_idx)] + punted
+                py_js.sort()
+
+        for j in py_js:
             mt = self._meta_list[tool_idx[j]]
             r = int(rows[j])
             idb = self._id_bytes(blob, env, r)
@@ -983,8 +1060,11 @@ class GpuPluginPipeline:
         if not ok_items:
             return []
 
-        texts2 = [json.dumps(a, separators=(",", ":"), sort_keys=True, default=str).encode()
-                  for (_j, a) in ok_items]
+        # scan texts are the sorted-keys form (_text_of): native-lane rows
+        # have it precomputed; python-lane entries serialize here
+        texts2 = [scan_bytes[j] if j in scan_bytes
+                  else json.dumps(a, separators=(",", ":"), sort_keys=True, default=str).encode()
+                  for (j, a) in ok_items]
         data2, beg2, end2 = pack_texts(texts2, self.device)
         harm2_t = None
         if "harm" in self.banks:
@@ -1024,6 +1104,8 @@ class GpuPluginPipeline:
             if self._enforcing(self.schema_guard) and mt.schema_mode != "trivial":
                 from ..utils.jsonschema import validate as _validate
 
+                if isinstance(args2, bytes):
+                    args2 = json.loads(args2)
                 errs = _validate(args2 or {}, mt.tool.get("input_schema") or {})
                 if errs:
                     responses[r] = self._splice_error(idb, jsonrpc.POLICY_DENIED,
@@ -1086,6 +1168,8 @@ class GpuPluginPipeline:
                     except Exception:
                         py_errors[idx] = jsonrpc.JSONRPCError(jsonrpc.INVALID_PARAMS, "invalid arguments")
                         return
+                elif isinstance(args, bytes):  # native rewrite-lane output
+                    args = json.loads(args)
                 try:
                     if mt.a2a_fast:
                         result = await self._a2a_fast_invoke(mt, args)
@@ -1101,8 +1185,11 @@ class GpuPluginPipeline:
                 mt = self._meta_list[tool_idx[j]]
                 if mt.native_kind >= 0 and args2 is not None:
                     # rewritten args for a native tool: batch through the C++
-                    # upstream instead of per-request Python dispatch
-                    nat2.append((idx, json.dumps(args2, separators=(",", ":"), default=str).encode(),
+                    # upstream instead of per-request Python dispatch (the
+                    # native rewrite lane already produced canonical bytes)
+                    nat2.append((idx,
+                                 args2 if isinstance(args2, bytes)
+                                 else json.dumps(args2, separators=(",", ":"), default=str).encode(),
                                  mt.native_kind))
                 elif (mt.itype == "LOCAL" and not (mt.tool.get("annotations") or {}).get("io")) \
                         or mt.a2a_fast \

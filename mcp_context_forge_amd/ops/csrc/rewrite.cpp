@@ -460,23 +460,26 @@ void append_escaped(std::string& out, const std::string& s) {
     }
 }
 
-void serialize(const std::vector<Val>& pool, size_t idx, std::string& out) {
+// sorted=true → json.dumps(..., sort_keys=True) (the _text_of/scan form);
+// sorted=false → json.dumps(...) wire/insertion order (the dispatch form,
+// matching the CPU chain's serialization of the parsed dict)
+void serialize(const std::vector<Val>& pool, size_t idx, std::string& out, bool sorted) {
     const Val& v = pool[idx];
     switch (v.kind) {
         case Val::OBJ: {
-            // sort_keys=True: ASCII keys sort identically by bytes
             std::vector<const Member*> ms;
             ms.reserve(v.members.size());
             for (auto& m : v.members) ms.push_back(&m);
-            std::sort(ms.begin(), ms.end(),
-                      [](const Member* a, const Member* b) { return a->key < b->key; });
+            if (sorted)  // ASCII keys sort identically by bytes
+                std::sort(ms.begin(), ms.end(),
+                          [](const Member* a, const Member* b) { return a->key < b->key; });
             out += '{';
             for (size_t k = 0; k < ms.size(); ++k) {
                 if (k) out += ',';
                 out += '"';
                 append_escaped(out, ms[k]->key);
                 out += "\":";
-                serialize(pool, ms[k]->vidx, out);
+                serialize(pool, ms[k]->vidx, out, sorted);
             }
             out += '}';
             break;
@@ -485,7 +488,7 @@ void serialize(const std::vector<Val>& pool, size_t idx, std::string& out) {
             out += '[';
             for (size_t k = 0; k < v.items.size(); ++k) {
                 if (k) out += ',';
-                serialize(pool, v.items[k], out);
+                serialize(pool, v.items[k], out, sorted);
             }
             out += ']';
             break;
@@ -518,7 +521,9 @@ extern "C" int64_t forge_rewrite_rows(
     uint32_t pii_active_mask, int pii_mode /*0 mask, 1 block, 2 audit*/,
     int norm_collapse, int norm_strip,
     int32_t* status, uint32_t* found_bits,
-    uint8_t* arena, int64_t arena_cap, int64_t* out_beg, int64_t* out_end)
+    uint8_t* arena, int64_t arena_cap,
+    int64_t* out_beg, int64_t* out_end,        // dispatch form (wire key order)
+    int64_t* scan_beg, int64_t* scan_end)      // scan form (sorted keys)
 {
     std::string buf;
     buf.reserve((size_t)n * 64);
@@ -526,6 +531,7 @@ extern "C" int64_t forge_rewrite_rows(
         status[i] = RW_PUNT;
         found_bits[i] = 0;
         out_beg[i] = out_end[i] = -1;
+        scan_beg[i] = scan_end[i] = -1;
         const uint8_t* b = blob + args_beg[i];
         const uint8_t* e = blob + args_end[i];
         if (e < b) continue;
@@ -570,11 +576,20 @@ extern "C" int64_t forge_rewrite_rows(
             status[i] = RW_BLOCKED;
             continue;
         }
-        std::string out;
-        serialize(pool, root, out);
+        std::string wire, sorted_s;
+        serialize(pool, root, wire, /*sorted=*/false);
+        serialize(pool, root, sorted_s, /*sorted=*/true);
         out_beg[i] = (int64_t)buf.size();
-        buf += out;
+        buf += wire;
         out_end[i] = (int64_t)buf.size();
+        if (sorted_s == wire) {  // common: already sorted / no objects
+            scan_beg[i] = out_beg[i];
+            scan_end[i] = out_end[i];
+        } else {
+            scan_beg[i] = (int64_t)buf.size();
+            buf += sorted_s;
+            scan_end[i] = (int64_t)buf.size();
+        }
         status[i] = RW_DONE;
     }
     if ((int64_t)buf.size() > arena_cap) return -(int64_t)buf.size();

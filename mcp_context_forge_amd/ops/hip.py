@@ -47,7 +47,7 @@ def _load() -> ctypes.CDLL:
         "forge_rewrite_rows": [ctypes.c_void_p] * 3 + [ctypes.c_int] + [ctypes.c_void_p] * 2 +
                               [ctypes.c_uint32] + [ctypes.c_int] * 3 +
                               [ctypes.c_void_p] * 2 +
-                              [ctypes.c_void_p, ctypes.c_int64, ctypes.c_void_p, ctypes.c_void_p],
+                              [ctypes.c_void_p, ctypes.c_int64] + [ctypes.c_void_p] * 4,
     }
     for name, argtypes in protos.items():
         fn = getattr(lib, name)
@@ -450,14 +450,18 @@ def rewrite_rows(blob: np.ndarray, args_beg: np.ndarray, args_end: np.ndarray,
     """Native rewrite pass (rewrite.cpp): normalizer + PII over the flagged
     rows that fit the provable-equivalence envelope; everything else gets
     RW_PUNT and takes the Python path.
-    → (status i32[n], found u32[n], arena u8, out_beg i64[n], out_end i64[n])."""
+    → (status i32[n], found u32[n], arena u8,
+       out_beg/out_end i64[n] (dispatch form, wire key order),
+       scan_beg/scan_end i64[n] (sorted-keys scan form))."""
     n = args_beg.shape[0]
     lib = _load()
     status = np.empty(n, dtype=np.int32)
     found = np.empty(n, dtype=np.uint32)
     out_beg = np.empty(n, dtype=np.int64)
     out_end = np.empty(n, dtype=np.int64)
-    cap = int((args_end - args_beg).sum()) + n * 32 + 4096
+    scan_beg = np.empty(n, dtype=np.int64)
+    scan_end = np.empty(n, dtype=np.int64)
+    cap = int((args_end - args_beg).sum()) * 2 + n * 32 + 4096
     while True:
         arena = np.empty(cap, dtype=np.uint8)
         rc = lib.forge_rewrite_rows(
@@ -466,9 +470,10 @@ def rewrite_rows(blob: np.ndarray, args_beg: np.ndarray, args_end: np.ndarray,
             ctypes.c_uint32(pii_active_mask), pii_mode,
             1 if norm_collapse else 0, 1 if norm_strip else 0,
             _np_ptr(status), _np_ptr(found),
-            _np_ptr(arena), cap, _np_ptr(out_beg), _np_ptr(out_end))
+            _np_ptr(arena), cap, _np_ptr(out_beg), _np_ptr(out_end),
+            _np_ptr(scan_beg), _np_ptr(scan_end))
         if rc >= 0:
-            return status, found, arena, out_beg, out_end
+            return status, found, arena, out_beg, out_end, scan_beg, scan_end
         cap = -int(rc) + 4096
 
 

@@ -52,7 +52,9 @@ def py_reference(raw, do_norm=True, do_pii=True):
             return masked
 
         args = _walk_strings(args, fn)
-    return json.dumps(args, separators=(",", ":"), sort_keys=True).encode(), sorted(set(found))
+    sorted_form = json.dumps(args, separators=(",", ":"), sort_keys=True).encode()
+    wire_form = json.dumps(args, separators=(",", ":")).encode()
+    return sorted_form, wire_form, sorted(set(found))
 
 
 FIXED_NAMES = [n for n, _, _ in PII_PATTERNS]
@@ -87,23 +89,24 @@ WS_SAMPLES = [
 
 def test_pii_masking_matches_python_re():
     raws = [json.dumps({"msg": s}, separators=(",", ":")).encode() for s in PII_SAMPLES]
-    status, found, arena, ob, oe = run_c(raws, do_flags=[2] * len(raws))
+    status, found, arena, ob, oe, sb, se = run_c(raws, do_flags=[2] * len(raws))
     for i, raw in enumerate(raws):
-        expected, names = py_reference(raw, do_norm=False, do_pii=True)
+        sorted_exp, wire_exp, names = py_reference(raw, do_norm=False, do_pii=True)
         assert status[i] == hip.RW_DONE, (i, status[i])
-        got = arena[ob[i]:oe[i]].tobytes()
-        assert got == expected, (i, PII_SAMPLES[i], got, expected)
+        assert arena[sb[i]:se[i]].tobytes() == sorted_exp, (i, PII_SAMPLES[i])
+        assert arena[ob[i]:oe[i]].tobytes() == wire_exp, (i, PII_SAMPLES[i])
         assert names_of(int(found[i])) == names, (i, PII_SAMPLES[i])
 
 
 def test_normalizer_matches_python():
     raws = [json.dumps({"m": s, "k": [s, {"d": s}]}, separators=(",", ":")).encode()
             for s in WS_SAMPLES]
-    status, found, arena, ob, oe = run_c(raws, do_flags=[1] * len(raws))
+    status, found, arena, ob, oe, sb, se = run_c(raws, do_flags=[1] * len(raws))
     for i, raw in enumerate(raws):
-        expected, _ = py_reference(raw, do_norm=True, do_pii=False)
+        sorted_exp, wire_exp, _ = py_reference(raw, do_norm=True, do_pii=False)
         assert status[i] == hip.RW_DONE
-        assert arena[ob[i]:oe[i]].tobytes() == expected, (i, WS_SAMPLES[i])
+        assert arena[sb[i]:se[i]].tobytes() == sorted_exp, (i, WS_SAMPLES[i])
+        assert arena[ob[i]:oe[i]].tobytes() == wire_exp, (i, WS_SAMPLES[i])
 
 
 def test_canonicalization_matches_json_dumps():
@@ -115,11 +118,13 @@ def test_canonicalization_matches_json_dumps():
         b'{"neg":-5,"zero":0,"nzero":-0,"big":123456789012345678901234567890}',
         b'[]', b'{}', b'"bare string"', b'42', b'true', b'null',
     ]
-    status, found, arena, ob, oe = run_c(cases, do_flags=[0] * len(cases))
+    status, found, arena, ob, oe, sb, se = run_c(cases, do_flags=[0] * len(cases))
     for i, raw in enumerate(cases):
-        expected = json.dumps(json.loads(raw), separators=(",", ":"), sort_keys=True).encode()
+        sorted_exp = json.dumps(json.loads(raw), separators=(",", ":"), sort_keys=True).encode()
+        wire_exp = json.dumps(json.loads(raw), separators=(",", ":")).encode()
         assert status[i] == hip.RW_DONE, (i, raw, status[i])
-        assert arena[ob[i]:oe[i]].tobytes() == expected, (i, raw)
+        assert arena[sb[i]:se[i]].tobytes() == sorted_exp, (i, raw)
+        assert arena[ob[i]:oe[i]].tobytes() == wire_exp, (i, raw)
 
 
 def test_punt_envelope():
@@ -133,7 +138,7 @@ def test_punt_envelope():
         (("[" * 40) + ("]" * 40), hip.RW_PUNT),         # depth > 32
     ]
     raws = [c if isinstance(c, bytes) else c.encode() for c, _ in cases]
-    status, *_ = run_c(raws)
+    status = run_c(raws)[0]
     for i, (_, want) in enumerate(cases):
         assert status[i] != hip.RW_DONE, (i, cases[i])
 
@@ -142,10 +147,10 @@ def test_block_and_audit_modes():
     raw = json.dumps({"m": "mail a@b.co now", "n": "  sp  aced "},
                      separators=(",", ":")).encode()
     # block: found reported, no output
-    status, found, arena, ob, oe = run_c([raw], do_flags=[3], pii_mode=1)
+    status, found, arena, ob, oe, sb, se = run_c([raw], do_flags=[3], pii_mode=1)
     assert status[0] == hip.RW_BLOCKED and names_of(int(found[0])) == ["email"]
     # audit: found reported, text NOT substituted (but normalizer applies)
-    status, found, arena, ob, oe = run_c([raw], do_flags=[3], pii_mode=2)
+    status, found, arena, ob, oe, sb, se = run_c([raw], do_flags=[3], pii_mode=2)
     assert status[0] == hip.RW_DONE and names_of(int(found[0])) == ["email"]
     out = json.loads(arena[ob[0]:oe[0]].tobytes())
     assert "a@b.co" in out["m"]
@@ -177,15 +182,15 @@ def test_randomized_cross_validation():
             else:
                 obj[key] = {"inner": rnd_str(), "n": None}
         raws.append(json.dumps(obj, separators=(",", ":")).encode())
-    status, found, arena, ob, oe = run_c(raws)
+    status, found, arena, ob, oe, sb, se = run_c(raws)
     done = punt = 0
     for i, raw in enumerate(raws):
         if status[i] != hip.RW_DONE:
             punt += 1
             continue
         done += 1
-        expected, names = py_reference(raw)
-        got = arena[ob[i]:oe[i]].tobytes()
-        assert got == expected, (i, raw, got, expected)
+        sorted_exp, wire_exp, names = py_reference(raw)
+        assert arena[sb[i]:se[i]].tobytes() == sorted_exp, (i, raw)
+        assert arena[ob[i]:oe[i]].tobytes() == wire_exp, (i, raw)
         assert names_of(int(found[i])) == names, (i, raw)
     assert done >= n_cases * 0.9, (done, punt)  # envelope covers the corpus
