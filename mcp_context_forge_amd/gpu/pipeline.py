@@ -182,6 +182,23 @@ class GpuPluginPipeline:
             fixed_index = {n: k for k, (n, _, _) in enumerate(_PIIP)}
             self._pii_fixed = [fixed_index[a[0]] for a in self.pii.active]
             self._pii_active_mask = sum(1 << i for i in self._pii_fixed)
+        # deny word table for the native lane's pre-rewrite recheck
+        # (lowercased when case-insensitive; non-ASCII words disable the
+        # lane for deny-active tools — those rows stay on the Python path)
+        self._deny_lane_ok = False
+        self._deny_lane_blob = None
+        self._deny_lane_off = None
+        if self.deny is not None and self.deny.words:
+            words = [w.lower() if self.deny.case_insensitive else w for w in self.deny.words]
+            if all(w.isascii() for w in words):
+                blob_d = b"".join(w.encode() for w in words)
+                off_d = np.zeros(len(words) + 1, dtype=np.int32)
+                for i, w in enumerate(words):
+                    off_d[i + 1] = off_d[i] + len(w.encode())
+                self._deny_lane_blob = np.frombuffer(blob_d, dtype=np.uint8).copy() \
+                    if blob_d else np.zeros(1, dtype=np.uint8)
+                self._deny_lane_off = off_d
+                self._deny_lane_ok = True
         from ..ops.pybridge import get as _pb_get
 
         self._pb = _pb_get()  # C response-assembly loops (fails loudly if missing)
@@ -970,17 +987,18 @@ class GpuPluginPipeline:
             flags_l: List[int] = []
             want_l: List[int] = []
             for j in rewrite_js:
-                if deny_cand is not None and deny_cand[j]:
-                    continue  # deny recheck over decoded text → Python lane
                 mt = self._meta_list[tool_idx[j]]
                 if regex_m is not None and regex_m[j] and \
                         self._active(self.regex, mt.name, block_class=False):
                     continue  # user-configured Python regexes → Python lane
+                deny_needed = self._active(self.deny, mt.name, block_class=True)
+                if deny_needed and not self._deny_lane_ok:
+                    continue  # non-ASCII deny words → Python lane recheck
                 do_norm = (norm_m is None or bool(norm_m[j])) and \
                     self._active(self.normalizer, mt.name, block_class=False)
                 do_pii = (pii_m is None or bool(pii_m[j])) and \
                     self._active(self.pii, mt.name, block_class=False)
-                fl = (1 if do_norm else 0) | (2 if do_pii else 0)
+                fl = (1 if do_norm else 0) | (2 if do_pii else 0) | (4 if deny_needed else 0)
                 bits = int(pii_m[j]) if (pii_m is not None and do_pii) else -1
                 if bits < 0:
                     want = self._pii_active_mask
@@ -999,17 +1017,26 @@ class GpuPluginPipeline:
                 elif self.pii is not None and self.pii.action not in ("mask",):
                     pii_mode = 2
                 njs = np.asarray(nat_idx, dtype=np.int64)
-                st, found, rw_arena, rb, re_, sb, se = hip.rewrite_rows(
+                st, found, deny_hit, rw_arena, rb, re_, sb, se = hip.rewrite_rows(
                     blob, np.ascontiguousarray(args_b[njs]), np.ascontiguousarray(args_e[njs]),
                     np.asarray(flags_l, dtype=np.uint8), np.asarray(want_l, dtype=np.uint32),
                     self._pii_active_mask, pii_mode,
                     bool(self.normalizer and self.normalizer.collapse_ws),
-                    bool(self.normalizer and self.normalizer.strip))
+                    bool(self.normalizer and self.normalizer.strip),
+                    deny_blob=self._deny_lane_blob, deny_off=self._deny_lane_off,
+                    deny_ci=bool(self.deny and self.deny.case_insensitive))
                 punted: List[int] = []
                 for k, j in enumerate(nat_idx):
                     if st[k] == hip.RW_DONE:
                         ok_items.append((j, rw_arena[rb[k]:re_[k]].tobytes()))
                         scan_bytes[j] = rw_arena[sb[k]:se[k]].tobytes()
+                    elif st[k] == hip.RW_DENY:
+                        r = int(rows[j])
+                        idb = self._id_bytes(blob, env, r)
+                        word = self.deny.words[int(deny_hit[k])]
+                        responses[r] = self._splice_error(
+                            idb, jsonrpc.POLICY_DENIED, f"deny_filter: deny word {word!r} present")
+                        self.blocked += 1
                     elif st[k] == hip.RW_BLOCKED:
                         r = int(rows[j])
                         idb = self._id_bytes(blob, env, r)

@@ -46,7 +46,8 @@ def _load() -> ctypes.CDLL:
         "forge_upstream_call_batch": [ctypes.c_void_p] * 4 + [ctypes.c_int, ctypes.c_char_p, ctypes.c_void_p, ctypes.c_int64, ctypes.c_void_p, ctypes.c_void_p],
         "forge_rewrite_rows": [ctypes.c_void_p] * 3 + [ctypes.c_int] + [ctypes.c_void_p] * 2 +
                               [ctypes.c_uint32] + [ctypes.c_int] * 3 +
-                              [ctypes.c_void_p] * 2 +
+                              [ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int, ctypes.c_int] +
+                              [ctypes.c_void_p] * 3 +
                               [ctypes.c_void_p, ctypes.c_int64] + [ctypes.c_void_p] * 4,
     }
     for name, argtypes in protos.items():
@@ -440,13 +441,16 @@ def finalize(blob, id_beg, id_end, args_beg, args_end, tool_idx, user_hash,
         cap = -int(n) + 4096
 
 
-RW_DONE, RW_PUNT, RW_BLOCKED, RW_BADJSON = 0, 1, 2, 3
+RW_DONE, RW_PUNT, RW_BLOCKED, RW_BADJSON, RW_DENY = 0, 1, 2, 3, 4
 
 
 def rewrite_rows(blob: np.ndarray, args_beg: np.ndarray, args_end: np.ndarray,
                  do_flags: np.ndarray, pii_want: np.ndarray,
                  pii_active_mask: int, pii_mode: int,
-                 norm_collapse: bool, norm_strip: bool):
+                 norm_collapse: bool, norm_strip: bool,
+                 deny_blob: Optional[np.ndarray] = None,
+                 deny_off: Optional[np.ndarray] = None,
+                 deny_ci: bool = True):
     """Native rewrite pass (rewrite.cpp): normalizer + PII over the flagged
     rows that fit the provable-equivalence envelope; everything else gets
     RW_PUNT and takes the Python path.
@@ -461,6 +465,8 @@ def rewrite_rows(blob: np.ndarray, args_beg: np.ndarray, args_end: np.ndarray,
     out_end = np.empty(n, dtype=np.int64)
     scan_beg = np.empty(n, dtype=np.int64)
     scan_end = np.empty(n, dtype=np.int64)
+    deny_hit = np.empty(n, dtype=np.int32)
+    n_deny = (deny_off.shape[0] - 1) if deny_off is not None else 0
     cap = int((args_end - args_beg).sum()) * 2 + n * 32 + 4096
     while True:
         arena = np.empty(cap, dtype=np.uint8)
@@ -469,11 +475,12 @@ def rewrite_rows(blob: np.ndarray, args_beg: np.ndarray, args_end: np.ndarray,
             _np_ptr(do_flags), _np_ptr(pii_want),
             ctypes.c_uint32(pii_active_mask), pii_mode,
             1 if norm_collapse else 0, 1 if norm_strip else 0,
-            _np_ptr(status), _np_ptr(found),
+            _np_ptr(deny_blob), _np_ptr(deny_off), n_deny, 1 if deny_ci else 0,
+            _np_ptr(status), _np_ptr(found), _np_ptr(deny_hit),
             _np_ptr(arena), cap, _np_ptr(out_beg), _np_ptr(out_end),
             _np_ptr(scan_beg), _np_ptr(scan_end))
         if rc >= 0:
-            return status, found, arena, out_beg, out_end, scan_beg, scan_end
+            return status, found, deny_hit, arena, out_beg, out_end, scan_beg, scan_end
         cap = -int(rc) + 4096
 
 

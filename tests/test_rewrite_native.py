@@ -89,7 +89,7 @@ WS_SAMPLES = [
 
 def test_pii_masking_matches_python_re():
     raws = [json.dumps({"msg": s}, separators=(",", ":")).encode() for s in PII_SAMPLES]
-    status, found, arena, ob, oe, sb, se = run_c(raws, do_flags=[2] * len(raws))
+    status, found, deny_hit, arena, ob, oe, sb, se = run_c(raws, do_flags=[2] * len(raws))
     for i, raw in enumerate(raws):
         sorted_exp, wire_exp, names = py_reference(raw, do_norm=False, do_pii=True)
         assert status[i] == hip.RW_DONE, (i, status[i])
@@ -101,7 +101,7 @@ def test_pii_masking_matches_python_re():
 def test_normalizer_matches_python():
     raws = [json.dumps({"m": s, "k": [s, {"d": s}]}, separators=(",", ":")).encode()
             for s in WS_SAMPLES]
-    status, found, arena, ob, oe, sb, se = run_c(raws, do_flags=[1] * len(raws))
+    status, found, deny_hit, arena, ob, oe, sb, se = run_c(raws, do_flags=[1] * len(raws))
     for i, raw in enumerate(raws):
         sorted_exp, wire_exp, _ = py_reference(raw, do_norm=True, do_pii=False)
         assert status[i] == hip.RW_DONE
@@ -118,7 +118,7 @@ def test_canonicalization_matches_json_dumps():
         b'{"neg":-5,"zero":0,"nzero":-0,"big":123456789012345678901234567890}',
         b'[]', b'{}', b'"bare string"', b'42', b'true', b'null',
     ]
-    status, found, arena, ob, oe, sb, se = run_c(cases, do_flags=[0] * len(cases))
+    status, found, deny_hit, arena, ob, oe, sb, se = run_c(cases, do_flags=[0] * len(cases))
     for i, raw in enumerate(cases):
         sorted_exp = json.dumps(json.loads(raw), separators=(",", ":"), sort_keys=True).encode()
         wire_exp = json.dumps(json.loads(raw), separators=(",", ":")).encode()
@@ -147,10 +147,10 @@ def test_block_and_audit_modes():
     raw = json.dumps({"m": "mail a@b.co now", "n": "  sp  aced "},
                      separators=(",", ":")).encode()
     # block: found reported, no output
-    status, found, arena, ob, oe, sb, se = run_c([raw], do_flags=[3], pii_mode=1)
+    status, found, deny_hit, arena, ob, oe, sb, se = run_c([raw], do_flags=[3], pii_mode=1)
     assert status[0] == hip.RW_BLOCKED and names_of(int(found[0])) == ["email"]
     # audit: found reported, text NOT substituted (but normalizer applies)
-    status, found, arena, ob, oe, sb, se = run_c([raw], do_flags=[3], pii_mode=2)
+    status, found, deny_hit, arena, ob, oe, sb, se = run_c([raw], do_flags=[3], pii_mode=2)
     assert status[0] == hip.RW_DONE and names_of(int(found[0])) == ["email"]
     out = json.loads(arena[ob[0]:oe[0]].tobytes())
     assert "a@b.co" in out["m"]
@@ -182,7 +182,7 @@ def test_randomized_cross_validation():
             else:
                 obj[key] = {"inner": rnd_str(), "n": None}
         raws.append(json.dumps(obj, separators=(",", ":")).encode())
-    status, found, arena, ob, oe, sb, se = run_c(raws)
+    status, found, deny_hit, arena, ob, oe, sb, se = run_c(raws)
     done = punt = 0
     for i, raw in enumerate(raws):
         if status[i] != hip.RW_DONE:
@@ -194,3 +194,45 @@ def test_randomized_cross_validation():
         assert arena[ob[i]:oe[i]].tobytes() == wire_exp, (i, raw)
         assert names_of(int(found[i])) == names, (i, raw)
     assert done >= n_cases * 0.9, (done, punt)  # envelope covers the corpus
+
+
+def test_deny_check_matches_python():
+    """do_flags bit2: the C deny recheck replicates DenyFilterPlugin over
+    the sorted pre-rewrite text — incl. escape-hidden words and duplicate
+    keys (the decoded-text semantics the raw scan cannot provide)."""
+    words = ["forbidden", "blocked_word"]
+    blob_d = b"".join(w.encode() for w in words)
+    off_d = np.zeros(len(words) + 1, dtype=np.int32)
+    for i, w in enumerate(words):
+        off_d[i + 1] = off_d[i] + len(w)
+    cases = [
+        (b'{"msg":"has forbi\\u0064den word"}', 0),          # escape-hidden → deny idx 0
+        (b'{"msg":"FORBIDDEN caps"}', 0),                    # case-insensitive
+        (b'{"msg":"blocked_word here"}', 1),
+        (b'{"msg":"forbidden","msg":"clean"}', None),        # dup key: discarded value
+        (b'{"msg":"clean","msg":"forbidden"}', 0),           # dup key: kept value
+        (b'{"msg":"benign"}', None),
+        (b'{"a":"forbi","b":"dden"}', None),                 # boundary-spanning no-match
+    ]
+    raws = [c for c, _ in cases]
+    blob = b"".join(raws)
+    beg, end = [], []
+    off = 0
+    for r in raws:
+        beg.append(off)
+        off += len(r)
+        end.append(off)
+    status, found, deny_hit, arena, ob, oe, sb, se = hip.rewrite_rows(
+        np.frombuffer(blob, dtype=np.uint8).copy(),
+        np.asarray(beg, dtype=np.int32), np.asarray(end, dtype=np.int32),
+        np.asarray([4] * len(raws), dtype=np.uint8),
+        np.asarray([0] * len(raws), dtype=np.uint32),
+        0, 0, True, True,
+        deny_blob=np.frombuffer(blob_d, dtype=np.uint8).copy(), deny_off=off_d, deny_ci=True)
+    for i, (raw, want) in enumerate(cases):
+        if want is None:
+            assert status[i] == hip.RW_DONE, (i, raw, status[i])
+            assert deny_hit[i] == -1
+        else:
+            assert status[i] == hip.RW_DENY, (i, raw, status[i])
+            assert deny_hit[i] == want, (i, raw, deny_hit[i])
