@@ -128,13 +128,19 @@ TOOLS = ["fast-time-echo", "fast-time-convert_time", "fast-time-get_system_time"
 
 
 def _outcome(resp):
-    """Outcome class of a response: (kind, code-or-result)."""
+    """Outcome class of a response: (kind, code-or-result). Wall-clock
+    timestamps (get_system_time) are normalized — the CPU and GPU batches
+    run ~a second apart and their 'current time' may differ."""
+    import re as _re
+
     if resp is None:
         return ("none", None)
     o = json.loads(resp)
     if "error" in o:
         return ("error", o["error"]["code"])
-    return ("result", json.dumps(o["result"], sort_keys=True))
+    text = json.dumps(o["result"], sort_keys=True)
+    text = _re.sub(r"\d{4}-\d{2}-\d{2}T\d{2}:\d{2}:\d{2}Z", "TS", text)
+    return ("result", text)
 
 
 # ---------------------------------------------------------------- tier 1 (CPU)
