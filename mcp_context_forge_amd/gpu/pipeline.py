@@ -199,6 +199,23 @@ class GpuPluginPipeline:
                     if blob_d else np.zeros(1, dtype=np.uint8)
                 self._deny_lane_off = off_d
                 self._deny_lane_ok = True
+        # harm phrase table for the native POST lane (_host_post's C path):
+        # pre-lowercased, plugin order; non-ASCII phrases disable the lane
+        # for harm-active tools (those rows stay on the Python path)
+        self._harm_lane_ok = self.harm is None or not getattr(self.harm, "phrases", None)
+        self._harm_lane_blob = None
+        self._harm_lane_off = None
+        if self.harm is not None and self.harm.phrases:
+            ph = [p.lower() for p in self.harm.phrases]
+            if all(p.isascii() for p in ph):
+                blob_h = b"".join(p.encode() for p in ph)
+                off_h = np.zeros(len(ph) + 1, dtype=np.int32)
+                for i, p in enumerate(ph):
+                    off_h[i + 1] = off_h[i] + len(p.encode())
+                self._harm_lane_blob = np.frombuffer(blob_h, dtype=np.uint8).copy() \
+                    if blob_h else np.zeros(1, dtype=np.uint8)
+                self._harm_lane_off = off_h
+                self._harm_lane_ok = True
         from ..ops.pybridge import get as _pb_get
 
         self._pb = _pb_get()  # C response-assembly loops (fails loudly if missing)
@@ -457,6 +474,7 @@ class GpuPluginPipeline:
                        and (_HT.TOOL_PRE_INVOKE in p.hooks or _HT.TOOL_POST_INVOKE in p.hooks)]
         flags = np.zeros(max(nt, 1), dtype=np.uint32)
         hostbound = np.zeros(max(nt, 1), dtype=bool)
+        postlane = np.full(max(nt, 1), -1, dtype=np.int8)
         for i, m in enumerate(metas):
             bmap = mgr.bindings_for_tool(m.name)
             m.host_chain = any(b.get("config") or (pname not in bank_names)
@@ -498,7 +516,30 @@ class GpuPluginPipeline:
             if self.exact_cache is not None:
                 f |= hip.TF_EXACT
             flags[i] = f
+            # native POST lane eligibility (forge_post_rows): -1 = this
+            # tool's flagged results must run the exact Python _host_post
+            # (user regexes, output schemas); otherwise bit0 pii, bit1
+            # harm, bit2 toon — mirroring _host_post's own gates
+            pf = 0
+            if m.host_chain or m.has_output_schema or \
+               self._active(self.regex, m.name, block_class=False):
+                pf = -1
+            if pf >= 0 and self._active(self.pii, m.name, block_class=False):
+                pf |= 1
+            if pf >= 0 and self.harm is not None and self._enforcing(self.harm) and \
+               self._applies(self.harm, m.name):
+                pf = (pf | 2) if self._harm_lane_ok else -1
+            if pf >= 0 and self.toon is not None and self._applies(self.toon, m.name):
+                # guard soundness: with positive min_savings the TOON text is
+                # strictly shorter than the JSON it replaces, so a row under
+                # the guard bound stays under it; otherwise punt to Python
+                if self.out_guard is not None and self.toon.min_savings <= 0:
+                    pf = -1
+                else:
+                    pf |= 4
+            postlane[i] = pf
         self._t_flags = flags
+        self._t_postlane = postlane
         # semcache insert allowlist (tool-level; lookups are gated by TF_CACHE)
         self._t_semallow = np.array(
             [bool(self.semcache is not None and self.semcache_plugin.cacheable(m.name)) for m in metas],
@@ -1330,8 +1371,72 @@ class GpuPluginPipeline:
         # --- host post chain for flagged rows ---
         host_ks = np.nonzero(needs_host)[0]
         host_cached: List[Tuple[int, bytes]] = []  # (k, result bytes) for cache insert
+
+        # native POST lane (rewrite.cpp forge_post_rows): pii → harm → toon
+        # → serialize in C for eligible rows; punts fall through to the
+        # exact Python _host_post below
+        c_handled: set = set()
+        if host_ks.size:
+            nat_ks = [int(k) for k in host_ks
+                      if not err_rows[k]
+                      and self._t_postlane[tool_idx[all_js[int(k)]]] >= 0
+                      and res_len[int(k)] <= guard_max]
+            if nat_ks:
+                pii_mode = 0
+                if self.pii is not None and self.pii.action == "block" and self._enforcing(self.pii):
+                    pii_mode = 1
+                elif self.pii is not None and self.pii.action not in ("mask",):
+                    pii_mode = 2
+                nk_np = np.asarray(nat_ks, dtype=np.int64)
+                st_p, found_p, harm_p, iserr_p, p_arena, pb, pe = hip.post_rows(
+                    res_blob,
+                    np.ascontiguousarray(res_beg[nk_np].astype(np.int64)),
+                    np.ascontiguousarray(res_end[nk_np].astype(np.int64)),
+                    np.ascontiguousarray(self._t_postlane[tool_idx[all_js_np[nk_np]]].astype(np.uint8)),
+                    self._pii_active_mask, pii_mode,
+                    harm_blob=self._harm_lane_blob, harm_off=self._harm_lane_off,
+                    toon_min_size=self.toon.min_size if self.toon is not None else (1 << 60),
+                    toon_min_savings=self.toon.min_savings if self.toon is not None else 1.0)
+                for q, k in enumerate(nat_ks):
+                    j = all_js[k]
+                    r = int(rows[j])
+                    idb = self._id_bytes(blob, env, r)
+                    if st_p[q] == hip.RW_DONE:
+                        self.post_rewrites += 1
+                        rbytes = p_arena[pb[q]:pe[q]].tobytes()
+                        if idb is not None:
+                            responses[r] = self._splice_result(idb, rbytes)
+                        if iserr_p[q]:
+                            is_err[k] = 1
+                        else:
+                            is_err[k] = 0
+                            cacheable[k] = 1
+                            host_cached.append((k, rbytes))
+                        c_handled.add(k)
+                    elif st_p[q] == hip.RW_BLOCKED:
+                        self.post_rewrites += 1
+                        names = sorted(self._pii_names[i2] for i2 in range(len(self._pii_names))
+                                       if (int(found_p[q]) >> i2) & 1)
+                        responses[r] = self._splice_error(
+                            idb, jsonrpc.POLICY_DENIED, f"pii_filter: PII detected: {names}")
+                        self.blocked += 1
+                        is_err[k] = 1
+                        c_handled.add(k)
+                    elif st_p[q] == hip.RW_DENY:
+                        self.post_rewrites += 1
+                        cat = self.harm.cats[int(harm_p[q])]
+                        responses[r] = self._splice_error(
+                            idb, jsonrpc.POLICY_DENIED,
+                            f"harmful_content_detector: harmful content ({cat})")
+                        self.blocked += 1
+                        is_err[k] = 1
+                        c_handled.add(k)
+                    # PUNT/BADJSON → python loop below
+
         for k in host_ks:
             k = int(k)
+            if k in c_handled:
+                continue
             j = all_js[k]
             mt = self._meta_list[tool_idx[j]]
             r = int(rows[j])

@@ -275,7 +275,9 @@ void norm_ascii(std::string& s, bool collapse, bool strip) {
 struct Val;
 struct Member { std::string key; size_t vidx; };
 struct Val {
-    enum Kind { OBJ, ARR, STR, INT, TRUE_, FALSE_, NULL_ } kind;
+    // FLT is never produced by the parser (floats punt) — only by the TOON
+    // meta builder, with str holding the exact Python repr text
+    enum Kind { OBJ, ARR, STR, INT, TRUE_, FALSE_, NULL_, FLT } kind;
     std::string str;                 // STR: decoded text; INT: literal digits
     std::vector<Member> members;     // OBJ
     std::vector<size_t> items;       // ARR
@@ -498,10 +500,220 @@ void serialize(const std::vector<Val>& pool, size_t idx, std::string& out, bool 
             out += '"';
             break;
         case Val::INT: out += v.str; break;
+        case Val::FLT: out += v.str; break;
         case Val::TRUE_: out += "true"; break;
         case Val::FALSE_: out += "false"; break;
         case Val::NULL_: out += "null"; break;
     }
+}
+
+// ---------------------------------------------------------------- TOON
+// Exact port of plugins/toon.py encode(): unquoted simple tokens,
+// `field[N]: a,b,c` primitive arrays, columnar `[N]{f1,f2}:` blocks for
+// uniform object arrays, `- ` items for mixed arrays, 2-space indents.
+
+inline bool toon_simple(const std::string& s) {  // _SIMPLE: ^[A-Za-z0-9_.@+\-]+\Z
+    if (s.empty()) return false;
+    for (char ch : s) {
+        uint8_t c = (uint8_t)ch;
+        if (!(is_alpha(c) || is_digit(c) || c == '_' || c == '.' || c == '@' || c == '+' || c == '-'))
+            return false;
+    }
+    return true;
+}
+
+inline bool toon_numeric(const std::string& s) {  // _NUMERIC: ^-?\d+(\.\d+)?([eE][+-]?\d+)?\Z
+    size_t i = 0, n = s.size();
+    if (i < n && s[i] == '-') ++i;
+    size_t d0 = i;
+    while (i < n && is_digit((uint8_t)s[i])) ++i;
+    if (i == d0) return false;
+    if (i < n && s[i] == '.') {
+        ++i;
+        size_t f0 = i;
+        while (i < n && is_digit((uint8_t)s[i])) ++i;
+        if (i == f0) return false;
+    }
+    if (i < n && (s[i] == 'e' || s[i] == 'E')) {
+        ++i;
+        if (i < n && (s[i] == '+' || s[i] == '-')) ++i;
+        size_t e0 = i;
+        while (i < n && is_digit((uint8_t)s[i])) ++i;
+        if (i == e0) return false;
+    }
+    return i == n;
+}
+
+// _scalar for a python str: quote (json.dumps) when ambiguous
+void toon_scalar_str(const std::string& s, std::string& out) {
+    if (s.empty() || !toon_simple(s) || s == "null" || s == "true" || s == "false" ||
+        toon_numeric(s)) {
+        out += '"';
+        append_escaped(out, s);
+        out += '"';
+    } else {
+        out += s;
+    }
+}
+
+inline bool toon_is_scalar(const Val& v) { return v.kind != Val::OBJ && v.kind != Val::ARR; }
+
+void toon_scalar(const std::vector<Val>& pool, size_t vi, std::string& out) {
+    const Val& v = pool[vi];
+    switch (v.kind) {
+        case Val::NULL_: out += "null"; return;
+        case Val::TRUE_: out += "true"; return;
+        case Val::FALSE_: out += "false"; return;
+        case Val::INT: out += v.str; return;   // json.dumps(int) == literal digits
+        case Val::FLT: out += v.str; return;
+        case Val::STR: toon_scalar_str(v.str, out); return;
+        default: return;  // containers handled by the caller
+    }
+}
+
+// fields if arr is a non-empty list of flat dicts with identical scalar
+// keys, all _SIMPLE (they ride unquoted inside the `{a,b}` header)
+bool toon_uniform(const std::vector<Val>& pool, const Val& arr, std::vector<std::string>& keys) {
+    if (arr.items.empty()) return false;
+    for (size_t it : arr.items) {
+        const Val& x = pool[it];
+        if (x.kind != Val::OBJ || x.members.empty()) return false;
+    }
+    keys.clear();
+    for (auto& m : pool[arr.items[0]].members) keys.push_back(m.key);
+    std::sort(keys.begin(), keys.end());
+    for (auto& k : keys)
+        if (!toon_simple(k)) return false;
+    std::vector<std::string> ks;
+    for (size_t it : arr.items) {
+        const Val& x = pool[it];
+        ks.clear();
+        for (auto& m : x.members) {
+            ks.push_back(m.key);
+            if (!toon_is_scalar(pool[m.vidx])) return false;
+        }
+        std::sort(ks.begin(), ks.end());
+        if (ks != keys) return false;
+    }
+    return true;
+}
+
+void toon_encode_value(const std::vector<Val>& pool, const std::string* key, size_t vi,
+                       int indent, std::vector<std::string>& out);
+
+void toon_encode_container(const std::vector<Val>& pool, size_t vi, int indent,
+                           std::vector<std::string>& out) {
+    const Val& v = pool[vi];
+    if (v.kind == Val::OBJ) {
+        for (auto& m : v.members) toon_encode_value(pool, &m.key, m.vidx, indent, out);
+    } else {
+        toon_encode_value(pool, nullptr, vi, indent, out);
+    }
+}
+
+void toon_encode_value(const std::vector<Val>& pool, const std::string* key, size_t vi,
+                       int indent, std::vector<std::string>& out) {
+    std::string pad((size_t)indent * 2, ' ');
+    std::string label;
+    if (key) toon_scalar_str(*key, label);
+    const Val& v = pool[vi];
+
+    if (toon_is_scalar(v)) {
+        std::string line = pad;
+        if (key) { line += label; line += ": "; }
+        toon_scalar(pool, vi, line);
+        out.push_back(std::move(line));
+        return;
+    }
+
+    if (v.kind == Val::ARR) {
+        std::vector<std::string> fields;
+        if (toon_uniform(pool, v, fields)) {
+            std::string head = pad;
+            if (key) head += label;
+            head += "[" + std::to_string(v.items.size()) + "]{";
+            for (size_t k = 0; k < fields.size(); ++k) {
+                if (k) head += ',';
+                head += fields[k];
+            }
+            head += "}:";
+            out.push_back(std::move(head));
+            for (size_t it : v.items) {
+                std::string line = pad + "  ";
+                const Val& x = pool[it];
+                for (size_t k = 0; k < fields.size(); ++k) {
+                    if (k) line += ',';
+                    for (auto& m : x.members)
+                        if (m.key == fields[k]) { toon_scalar(pool, m.vidx, line); break; }
+                }
+                out.push_back(std::move(line));
+            }
+            return;
+        }
+        bool all_sc = true;
+        for (size_t it : v.items)
+            if (!toon_is_scalar(pool[it])) { all_sc = false; break; }
+        if (all_sc) {  // includes the empty array (trailing space like python)
+            std::string line = pad;
+            if (key) line += label;
+            line += "[" + std::to_string(v.items.size()) + "]: ";
+            for (size_t k = 0; k < v.items.size(); ++k) {
+                if (k) line += ',';
+                toon_scalar(pool, v.items[k], line);
+            }
+            out.push_back(std::move(line));
+            return;
+        }
+        std::string head = pad;
+        if (key) head += label;
+        head += "[" + std::to_string(v.items.size()) + "]:";
+        out.push_back(std::move(head));
+        for (size_t it : v.items) {
+            if (toon_is_scalar(pool[it])) {
+                std::string line = pad + "  - ";
+                toon_scalar(pool, it, line);
+                out.push_back(std::move(line));
+            } else {
+                out.push_back(pad + "  -");
+                toon_encode_container(pool, it, indent + 2, out);
+            }
+        }
+        return;
+    }
+
+    // OBJ
+    if (key) {
+        out.push_back(pad + label + ":");
+        toon_encode_container(pool, vi, indent + 1, out);
+    } else {
+        toon_encode_container(pool, vi, indent, out);
+    }
+}
+
+std::string toon_encode(const std::vector<Val>& pool, size_t root) {
+    std::vector<std::string> lines;
+    toon_encode_value(pool, nullptr, root, 0, lines);
+    std::string s;
+    for (size_t i = 0; i < lines.size(); ++i) {
+        if (i) s += '\n';
+        s += lines[i];
+    }
+    return s;
+}
+
+// repr(round(frac, 4)) for the savings field. glibc %.4f rounds the exact
+// binary value to nearest (ties cannot occur: k.00005 decimals have no
+// finite binary expansion), matching python round(); stripping trailing
+// zeros (keeping one fractional digit) reproduces float.__repr__'s
+// shortest round-trip form for 4-decimal values.
+std::string py_float4(double frac) {
+    char b[64];
+    snprintf(b, sizeof(b), "%.4f", frac);
+    std::string s(b);
+    size_t dot = s.find('.');
+    size_t last = s.size();
+    while (last - 1 > dot + 1 && s[last - 1] == '0') --last;
+    return s.substr(0, last);
 }
 
 }  // namespace
@@ -618,6 +830,185 @@ extern "C" int64_t forge_rewrite_rows(
             buf += sorted_s;
             scan_end[i] = (int64_t)buf.size();
         }
+        status[i] = RW_DONE;
+    }
+    if ((int64_t)buf.size() > arena_cap) return -(int64_t)buf.size();
+    if (!buf.empty()) memcpy(arena, buf.data(), buf.size());
+    return (int64_t)buf.size();
+}
+
+// Native result post chain: the _host_post hot path (gpu/pipeline.py) for
+// rows inside the same provable-equivalence envelope as the rewrite lane.
+// Per row, in the exact CPU-chain order: pii_filter@30 over every string
+// of the parsed result → harmful_content@60 (case-insensitive substring
+// over the SORTED compact serialization of the post-PII result) →
+// toon_encoder@900 (structuredContent compression incl. the _meta.toon
+// record) → wire-order re-serialization (json.dumps insertion order).
+// Rows with user regexes, output schemas, or guard-length overflow must
+// not be passed here (caller punts them); non-ASCII / float / bad-JSON
+// results return RW_PUNT and take the Python path.
+//   do_flags: bit0 pii, bit1 harm, bit2 toon
+//   harm_blob/off: phrases, pre-lowercased ASCII, plugin order
+//   is_err: truthiness of result["isError"] on the FINAL result
+extern "C" int64_t forge_post_rows(
+    const uint8_t* blob, const int64_t* res_beg, const int64_t* res_end, int n,
+    const uint8_t* do_flags,
+    uint32_t pii_active_mask, int pii_mode /*0 mask, 1 block, 2 audit*/,
+    const uint8_t* harm_blob, const int32_t* harm_off, int n_harm,
+    int64_t toon_min_size, double toon_min_savings,
+    int32_t* status, uint32_t* found_bits, int32_t* harm_hit, uint8_t* is_err,
+    uint8_t* arena, int64_t arena_cap, int64_t* out_beg, int64_t* out_end)
+{
+    std::string buf;
+    buf.reserve((size_t)n * 128);
+    for (int i = 0; i < n; ++i) {
+        status[i] = RW_PUNT;
+        found_bits[i] = 0;
+        harm_hit[i] = -1;
+        is_err[i] = 0;
+        out_beg[i] = out_end[i] = -1;
+        const uint8_t* b = blob + res_beg[i];
+        const uint8_t* e = blob + res_end[i];
+        if (e < b) continue;
+        bool ascii = true;
+        for (const uint8_t* q = b; q < e; ++q)
+            if (*q >= 0x80) { ascii = false; break; }
+        if (!ascii) continue;  // PUNT
+        std::vector<Val> pool;
+        pool.reserve(64);
+        Parser ps{b, e, pool};
+        ps.skip_ws();
+        if (ps.p >= ps.e) { status[i] = RW_BADJSON; continue; }
+        size_t root = ps.parse_value(0);
+        if (ps.ok) {
+            ps.skip_ws();
+            if (ps.p != ps.e) ps.ok = false;
+        }
+        if (!ps.ok) continue;  // PUNT (floats, depth, escapes…)
+        uint8_t fl = do_flags[i];
+
+        // --- pii over every string of the result tree ---
+        uint32_t found = 0;
+        if (fl & 1) {
+            for (auto& v : pool) {
+                if (v.kind != Val::STR) continue;
+                if (pii_mode == 0) {
+                    found |= pii_mask_text(v.str, pii_active_mask, 0xFFFFFFFFu);
+                } else {
+                    std::string tmp = v.str;
+                    found |= pii_mask_text(tmp, pii_active_mask, 0xFFFFFFFFu);
+                }
+            }
+        }
+        found_bits[i] = found;
+        if (found && pii_mode == 1) { status[i] = RW_BLOCKED; continue; }
+
+        // --- harm phrases over sorted serialization of the post-pii tree ---
+        if ((fl & 2) && n_harm > 0) {
+            std::string hay;
+            serialize(pool, root, hay, /*sorted=*/true);
+            for (auto& ch : hay)
+                if (ch >= 'A' && ch <= 'Z') ch += 32;
+            for (int w = 0; w < n_harm; ++w) {
+                size_t wn = (size_t)(harm_off[w + 1] - harm_off[w]);
+                if (wn == 0) continue;
+                if (memmem(hay.data(), hay.size(), harm_blob + harm_off[w], wn) != nullptr) {
+                    harm_hit[i] = w;
+                    break;
+                }
+            }
+            if (harm_hit[i] >= 0) { status[i] = RW_DENY; continue; }
+        }
+
+        // --- toon: structuredContent compression + _meta.toon record ---
+        bool punt_row = false;
+        if ((fl & 4) && pool[root].kind == Val::OBJ) {
+            size_t sc = SIZE_MAX;
+            for (auto& m : pool[root].members)
+                if (m.key == "structuredContent") { sc = m.vidx; break; }
+            if (sc != SIZE_MAX && pool[sc].kind != Val::NULL_) {
+                std::string scj;
+                serialize(pool, sc, scj, /*sorted=*/false);
+                int64_t j = (int64_t)scj.size();
+                if (j >= toon_min_size) {
+                    std::string enc = toon_encode(pool, sc);
+                    int64_t t = (int64_t)enc.size();
+                    double frac = j ? 1.0 - (double)t / (double)j : 0.0;
+                    if (frac >= toon_min_savings) {
+                        // existing non-dict "_meta" → python (setdefault semantics)
+                        size_t meta_i = SIZE_MAX;
+                        for (auto& m : pool[root].members)
+                            if (m.key == "_meta") {
+                                meta_i = m.vidx;
+                                if (pool[m.vidx].kind != Val::OBJ) punt_row = true;
+                                break;
+                            }
+                        if (!punt_row) {
+                            // NOTE: pool grows below — take indices, never refs
+                            auto mk = [&pool](Val::Kind k) {
+                                size_t ix = pool.size();
+                                pool.emplace_back();
+                                pool[ix].kind = k;
+                                return ix;
+                            };
+                            size_t s_type = mk(Val::STR); pool[s_type].str = "text";
+                            size_t s_text = mk(Val::STR); pool[s_text].str = std::move(enc);
+                            size_t item = mk(Val::OBJ);
+                            pool[item].members.push_back({"type", s_type});
+                            pool[item].members.push_back({"text", s_text});
+                            size_t carr = mk(Val::ARR);
+                            pool[carr].items.push_back(item);
+                            bool had = false;
+                            for (auto& m : pool[root].members)
+                                if (m.key == "content") { m.vidx = carr; had = true; break; }
+                            if (!had) pool[root].members.push_back({"content", carr});
+                            size_t jb = mk(Val::INT); pool[jb].str = std::to_string(j);
+                            size_t tb = mk(Val::INT); pool[tb].str = std::to_string(t);
+                            size_t sv = mk(Val::FLT); pool[sv].str = py_float4(frac);
+                            size_t toonobj = mk(Val::OBJ);
+                            pool[toonobj].members.push_back({"json_bytes", jb});
+                            pool[toonobj].members.push_back({"toon_bytes", tb});
+                            pool[toonobj].members.push_back({"savings", sv});
+                            if (meta_i == SIZE_MAX) {
+                                size_t mo = mk(Val::OBJ);
+                                pool[mo].members.push_back({"toon", toonobj});
+                                pool[root].members.push_back({"_meta", mo});
+                            } else {
+                                bool hadt = false;
+                                for (auto& m : pool[meta_i].members)
+                                    if (m.key == "toon") { m.vidx = toonobj; hadt = true; break; }
+                                if (!hadt) pool[meta_i].members.push_back({"toon", toonobj});
+                            }
+                        }
+                    }
+                }
+            }
+        }
+        if (punt_row) continue;  // PUNT
+
+        // --- is_err + final wire serialization ---
+        if (pool[root].kind == Val::OBJ) {
+            for (auto& m : pool[root].members)
+                if (m.key == "isError") {
+                    const Val& x = pool[m.vidx];
+                    bool err = false;
+                    switch (x.kind) {
+                        case Val::TRUE_: err = true; break;
+                        case Val::INT: err = x.str != "0"; break;
+                        case Val::STR: err = !x.str.empty(); break;
+                        case Val::OBJ: err = !x.members.empty(); break;
+                        case Val::ARR: err = !x.items.empty(); break;
+                        default: err = false;
+                    }
+                    is_err[i] = err ? 1 : 0;
+                    break;
+                }
+        }
+        std::string wire;
+        serialize(pool, root, wire, /*sorted=*/false);
+        out_beg[i] = (int64_t)buf.size();
+        buf += wire;
+        out_end[i] = (int64_t)buf.size();
         status[i] = RW_DONE;
     }
     if ((int64_t)buf.size() > arena_cap) return -(int64_t)buf.size();

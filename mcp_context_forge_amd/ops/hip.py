@@ -49,12 +49,19 @@ def _load() -> ctypes.CDLL:
                               [ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int, ctypes.c_int] +
                               [ctypes.c_void_p] * 3 +
                               [ctypes.c_void_p, ctypes.c_int64] + [ctypes.c_void_p] * 4,
+        "forge_post_rows": [ctypes.c_void_p] * 3 + [ctypes.c_int, ctypes.c_void_p] +
+                           [ctypes.c_uint32, ctypes.c_int] +
+                           [ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int] +
+                           [ctypes.c_int64, ctypes.c_double] +
+                           [ctypes.c_void_p] * 4 +
+                           [ctypes.c_void_p, ctypes.c_int64] + [ctypes.c_void_p] * 2,
     }
     for name, argtypes in protos.items():
         fn = getattr(lib, name)
         fn.argtypes = argtypes
         fn.restype = ctypes.c_int64 if name in ("forge_upstream_call_batch",
-                                                "forge_rewrite_rows") else ctypes.c_int
+                                                "forge_rewrite_rows",
+                                                "forge_post_rows") else ctypes.c_int
     return lib
 
 
@@ -481,6 +488,41 @@ def rewrite_rows(blob: np.ndarray, args_beg: np.ndarray, args_end: np.ndarray,
             _np_ptr(scan_beg), _np_ptr(scan_end))
         if rc >= 0:
             return status, found, deny_hit, arena, out_beg, out_end, scan_beg, scan_end
+        cap = -int(rc) + 4096
+
+
+def post_rows(res_blob: np.ndarray, res_beg: np.ndarray, res_end: np.ndarray,
+              do_flags: np.ndarray, pii_active_mask: int, pii_mode: int,
+              harm_blob: Optional[np.ndarray] = None,
+              harm_off: Optional[np.ndarray] = None,
+              toon_min_size: int = 1 << 60, toon_min_savings: float = 1.0):
+    """Native result post chain (rewrite.cpp forge_post_rows): the
+    _host_post hot path — pii → harm → toon → wire serialization — for
+    eligible flagged results; RW_PUNT rows take the Python path.
+    → (status i32[n], found u32[n], harm_hit i32[n], is_err u8[n],
+       arena u8, out_beg/out_end i64[n])."""
+    n = res_beg.shape[0]
+    lib = _load()
+    status = np.empty(n, dtype=np.int32)
+    found = np.empty(n, dtype=np.uint32)
+    harm_hit = np.empty(n, dtype=np.int32)
+    is_err = np.empty(n, dtype=np.uint8)
+    out_beg = np.empty(n, dtype=np.int64)
+    out_end = np.empty(n, dtype=np.int64)
+    n_harm = (harm_off.shape[0] - 1) if harm_off is not None else 0
+    cap = int((res_end - res_beg).sum()) * 2 + n * 64 + 4096
+    while True:
+        arena = np.empty(cap, dtype=np.uint8)
+        rc = lib.forge_post_rows(
+            _np_ptr(res_blob), _np_ptr(res_beg), _np_ptr(res_end), n,
+            _np_ptr(do_flags),
+            ctypes.c_uint32(pii_active_mask), pii_mode,
+            _np_ptr(harm_blob), _np_ptr(harm_off), n_harm,
+            ctypes.c_int64(toon_min_size), ctypes.c_double(toon_min_savings),
+            _np_ptr(status), _np_ptr(found), _np_ptr(harm_hit), _np_ptr(is_err),
+            _np_ptr(arena), cap, _np_ptr(out_beg), _np_ptr(out_end))
+        if rc >= 0:
+            return status, found, harm_hit, is_err, arena, out_beg, out_end
         cap = -int(rc) + 4096
 
 

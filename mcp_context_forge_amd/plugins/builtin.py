@@ -276,7 +276,11 @@ class ToonEncoderPlugin(Plugin):
             return None
         new = dict(result)
         new["content"] = [{"type": "text", "text": encoded}]
-        new.setdefault("_meta", {})["toon"] = {"json_bytes": j, "toon_bytes": t, "savings": round(frac, 4)}
+        meta = new.get("_meta")
+        if "_meta" not in new:
+            meta = new["_meta"] = {}
+        if isinstance(meta, dict):  # a non-dict _meta from the upstream is left alone
+            meta["toon"] = {"json_bytes": j, "toon_bytes": t, "savings": round(frac, 4)}
         return new
 
     async def tool_post_invoke(self, ctx: PluginContext) -> PluginResult:
