@@ -274,6 +274,8 @@ class GpuPluginPipeline:
         self.blocked = 0
         self.cache_hits = 0
         self.post_rewrites = 0
+        self.post_c = 0       # flagged results fully handled by forge_post_rows
+        self.post_c_punt = 0  # rows the C post lane punted back to Python
         self.py_fallback = 0
         self.host_bound = 0   # requests routed to the CPU chain by plugin bindings
         # optional per-stage wall-clock accounting (FORGE_PIPELINE_TIMING=1)
@@ -521,9 +523,14 @@ class GpuPluginPipeline:
             # (user regexes, output schemas); otherwise bit0 pii, bit1
             # harm, bit2 toon — mirroring _host_post's own gates
             pf = 0
-            if m.host_chain or m.has_output_schema or \
-               self._active(self.regex, m.name, block_class=False):
+            if m.host_chain or m.has_output_schema:
                 pf = -1
+            if pf >= 0 and self._active(self.regex, m.name, block_class=False):
+                # regex rules run in C-lane rows ONLY when the pass-3 regex
+                # bank scan proves identity (no hit): rows WITH a bank hit
+                # punt (bit3). No bank at all (un-DFA-able rules) → no
+                # identity proof → the whole tool stays on the Python path.
+                pf = (pf | 8) if "regex" in self.banks else -1
             if pf >= 0 and self._active(self.pii, m.name, block_class=False):
                 pf |= 1
             if pf >= 0 and self.harm is not None and self._enforcing(self.harm) and \
@@ -1323,6 +1330,7 @@ class GpuPluginPipeline:
         all_js_np = np.asarray(all_js, dtype=np.int64)
         post_flag = np.zeros(n_all, dtype=bool)
         toon_meta = np.zeros(n_all, dtype=bool)
+        regex3 = np.zeros(n_all, dtype=bool)
         if n_all:
             await self._gpu_lock.acquire()
             self._pin_reset()
@@ -1340,6 +1348,8 @@ class GpuPluginPipeline:
                 if b == "postmeta":
                     toon_meta = h
                 else:
+                    if b == "regex":
+                        regex3 = h
                     post_flag |= h
 
         res_len = res_end - res_beg
@@ -1377,9 +1387,11 @@ class GpuPluginPipeline:
         # exact Python _host_post below
         c_handled: set = set()
         if host_ks.size:
+            pl = self._t_postlane
             nat_ks = [int(k) for k in host_ks
                       if not err_rows[k]
-                      and self._t_postlane[tool_idx[all_js[int(k)]]] >= 0
+                      and pl[tool_idx[all_js[int(k)]]] >= 0
+                      and not (pl[tool_idx[all_js[int(k)]]] & 8 and regex3[int(k)])
                       and res_len[int(k)] <= guard_max]
             if nat_ks:
                 pii_mode = 0
@@ -1392,7 +1404,7 @@ class GpuPluginPipeline:
                     res_blob,
                     np.ascontiguousarray(res_beg[nk_np].astype(np.int64)),
                     np.ascontiguousarray(res_end[nk_np].astype(np.int64)),
-                    np.ascontiguousarray(self._t_postlane[tool_idx[all_js_np[nk_np]]].astype(np.uint8)),
+                    np.ascontiguousarray((self._t_postlane[tool_idx[all_js_np[nk_np]]] & 7).astype(np.uint8)),
                     self._pii_active_mask, pii_mode,
                     harm_blob=self._harm_lane_blob, harm_off=self._harm_lane_off,
                     toon_min_size=self.toon.min_size if self.toon is not None else (1 << 60),
@@ -1432,6 +1444,8 @@ class GpuPluginPipeline:
                         is_err[k] = 1
                         c_handled.add(k)
                     # PUNT/BADJSON → python loop below
+                self.post_c += len(c_handled)
+                self.post_c_punt += len(nat_ks) - len(c_handled)
 
         for k in host_ks:
             k = int(k)
@@ -1583,7 +1597,8 @@ class GpuPluginPipeline:
             "batches": self.batches, "requests": self.requests, "fast_path": self.fast_path,
             "slow_path": self.slow_path, "blocked": self.blocked, "cache_hits": self.cache_hits,
             "host_bound": self.host_bound,
-            "post_rewrites": self.post_rewrites, "py_fallback": self.py_fallback,
+            "post_rewrites": self.post_rewrites, "post_c": self.post_c,
+            "post_c_punt": self.post_c_punt, "py_fallback": self.py_fallback,
             "banks": {k: {"states": v.n_states, "classes": v.n_classes} for k, v in self.banks.items()},
         }
         if self.semcache is not None:
