@@ -477,6 +477,14 @@ class GpuPluginPipeline:
         flags = np.zeros(max(nt, 1), dtype=np.uint32)
         hostbound = np.zeros(max(nt, 1), dtype=bool)
         postlane = np.full(max(nt, 1), -1, dtype=np.int8)
+        # rewrite/verdict lane per-tool activation (hoists the per-row
+        # _active/_applies python calls out of the hot loops)
+        rw_norm = np.zeros(max(nt, 1), dtype=bool)
+        rw_regex = np.zeros(max(nt, 1), dtype=bool)
+        rw_deny = np.zeros(max(nt, 1), dtype=bool)
+        rw_pii = np.zeros(max(nt, 1), dtype=bool)
+        rw_harm = np.zeros(max(nt, 1), dtype=bool)
+        mod_app = np.zeros(max(nt, 1), dtype=bool)
         for i, m in enumerate(metas):
             bmap = mgr.bindings_for_tool(m.name)
             m.host_chain = any(b.get("config") or (pname not in bank_names)
@@ -545,8 +553,17 @@ class GpuPluginPipeline:
                 else:
                     pf |= 4
             postlane[i] = pf
+            rw_norm[i] = self._active(self.normalizer, m.name, block_class=False)
+            rw_regex[i] = self._active(self.regex, m.name, block_class=False)
+            rw_deny[i] = self._active(self.deny, m.name, block_class=True)
+            rw_pii[i] = self._active(self.pii, m.name, block_class=False)
+            rw_harm[i] = bool(self.harm is not None and self._enforcing(self.harm) and
+                              self._applies(self.harm, m.name))
+            mod_app[i] = bool(self.moderation is not None and self._applies(self.moderation, m.name))
         self._t_flags = flags
         self._t_postlane = postlane
+        self._t_rw_norm, self._t_rw_regex, self._t_rw_deny = rw_norm, rw_regex, rw_deny
+        self._t_rw_pii, self._t_rw_harm, self._t_mod_app = rw_pii, rw_harm, mod_app
         # semcache insert allowlist (tool-level; lookups are gated by TF_CACHE)
         self._t_semallow = np.array(
             [bool(self.semcache is not None and self.semcache_plugin.cacheable(m.name)) for m in metas],
@@ -1025,28 +1042,31 @@ class GpuPluginPipeline:
 
         ok_items: List[Tuple[int, Any]] = []
         scan_bytes: Dict[int, bytes] = {}  # native lane: sorted-keys scan form
+        harm_c: Dict[int, int] = {}        # row → harm phrase idx (post-rewrite text)
 
         # --- native lane (rewrite.cpp): rows with no deny candidacy, no
         # user regexes, and provably-equivalent content (the C side punts
         # anything outside its envelope back here) ---
         py_js = rewrite_js
+        t_sub = self._tic()
         if rewrite_js:
             nat_idx: List[int] = []
             flags_l: List[int] = []
             want_l: List[int] = []
             for j in rewrite_js:
-                mt = self._meta_list[tool_idx[j]]
-                if regex_m is not None and regex_m[j] and \
-                        self._active(self.regex, mt.name, block_class=False):
+                ti = tool_idx[j]
+                if regex_m is not None and regex_m[j] and self._t_rw_regex[ti]:
                     continue  # user-configured Python regexes → Python lane
-                deny_needed = self._active(self.deny, mt.name, block_class=True)
+                deny_needed = self._t_rw_deny[ti]
                 if deny_needed and not self._deny_lane_ok:
                     continue  # non-ASCII deny words → Python lane recheck
-                do_norm = (norm_m is None or bool(norm_m[j])) and \
-                    self._active(self.normalizer, mt.name, block_class=False)
-                do_pii = (pii_m is None or bool(pii_m[j])) and \
-                    self._active(self.pii, mt.name, block_class=False)
-                fl = (1 if do_norm else 0) | (2 if do_pii else 0) | (4 if deny_needed else 0)
+                harm_needed = self._t_rw_harm[ti]
+                if harm_needed and not self._harm_lane_ok:
+                    continue  # non-ASCII harm phrases → Python lane
+                do_norm = (norm_m is None or bool(norm_m[j])) and self._t_rw_norm[ti]
+                do_pii = (pii_m is None or bool(pii_m[j])) and self._t_rw_pii[ti]
+                fl = (1 if do_norm else 0) | (2 if do_pii else 0) | (4 if deny_needed else 0) | \
+                    (8 if harm_needed else 0)
                 bits = int(pii_m[j]) if (pii_m is not None and do_pii) else -1
                 if bits < 0:
                     want = self._pii_active_mask
@@ -1058,6 +1078,8 @@ class GpuPluginPipeline:
                 nat_idx.append(j)
                 flags_l.append(fl)
                 want_l.append(want)
+            self._toc("rw_elig", t_sub)
+            t_sub = self._tic()
             if nat_idx:
                 pii_mode = 0
                 if self.pii is not None and self.pii.action == "block" and self._enforcing(self.pii):
@@ -1065,19 +1087,22 @@ class GpuPluginPipeline:
                 elif self.pii is not None and self.pii.action not in ("mask",):
                     pii_mode = 2
                 njs = np.asarray(nat_idx, dtype=np.int64)
-                st, found, deny_hit, rw_arena, rb, re_, sb, se = hip.rewrite_rows(
+                st, found, deny_hit, harm_hit, rw_arena, rb, re_, sb, se = hip.rewrite_rows(
                     blob, np.ascontiguousarray(args_b[njs]), np.ascontiguousarray(args_e[njs]),
                     np.asarray(flags_l, dtype=np.uint8), np.asarray(want_l, dtype=np.uint32),
                     self._pii_active_mask, pii_mode,
                     bool(self.normalizer and self.normalizer.collapse_ws),
                     bool(self.normalizer and self.normalizer.strip),
                     deny_blob=self._deny_lane_blob, deny_off=self._deny_lane_off,
-                    deny_ci=bool(self.deny and self.deny.case_insensitive))
+                    deny_ci=bool(self.deny and self.deny.case_insensitive),
+                    harm_blob=self._harm_lane_blob, harm_off=self._harm_lane_off)
                 punted: List[int] = []
                 for k, j in enumerate(nat_idx):
                     if st[k] == hip.RW_DONE:
                         ok_items.append((j, rw_arena[rb[k]:re_[k]].tobytes()))
                         scan_bytes[j] = rw_arena[sb[k]:se[k]].tobytes()
+                        if harm_hit[k] >= 0:
+                            harm_c[j] = int(harm_hit[k])
                     elif st[k] == hip.RW_DENY:
                         r = int(rows[j])
                         idb = self._id_bytes(blob, env, r)
@@ -1097,6 +1122,8 @@ class GpuPluginPipeline:
                         punted.append(j)
                 py_js = [j for j in rewrite_js if j not in set(nat_idx)] + punted
                 py_js.sort()
+        self._toc("rw_c", t_sub)
+        t_sub = self._tic()
 
         for j in py_js:
             mt = self._meta_list[tool_idx[j]]
@@ -1111,7 +1138,7 @@ class GpuPluginPipeline:
             # deny@10 before normalizer@15) — the raw-byte scan misses
             # escape-hidden words (forbidden), which is exactly what
             # routed this row here (parity-fuzz finding)
-            if self.deny is not None and self._active(self.deny, mt.name, block_class=True):
+            if self.deny is not None and self._t_rw_deny[tool_idx[j]]:
                 hay = _text_of(args)
                 hay = hay.lower() if self.deny.case_insensitive else hay
                 word = next((w for w in self.deny.words
@@ -1131,85 +1158,111 @@ class GpuPluginPipeline:
                 responses[r] = self._splice_error(idb, jsonrpc.POLICY_DENIED, payload)
                 self.blocked += 1
                 continue
+            # harmful_content@60 over the rewritten text (the CPU plugin's
+            # own check; verdict ordering vs moderation happens in
+            # _finish_rewrite exactly like the chain)
+            if self.harm is not None and self._t_rw_harm[tool_idx[j]]:
+                hay2 = _text_of(payload).lower()
+                for w, ph in enumerate(self.harm.phrases):
+                    if ph.lower() in hay2:
+                        harm_c[j] = w
+                        break
             ok_items.append((j, payload))
+        self._toc("rw_pyloop", t_sub)
         if not ok_items:
             return None
+        t_sub = self._tic()
 
         # scan texts are the sorted-keys form (_text_of): native-lane rows
         # have it precomputed; python-lane entries serialize here
         texts2 = [scan_bytes[j] if j in scan_bytes
                   else json.dumps(a, separators=(",", ":"), sort_keys=True, default=str).encode()
                   for (j, a) in ok_items]
-        data2, beg2, end2 = pack_texts(texts2, self.device)
-        harm2_t = None
-        if "harm" in self.banks:
-            harm2_t, _ = hip.scan(data2, beg2, end2, self.banks["harm"])
         scores2_t = None
         if self.classifier is not None:
+            data2, beg2, end2 = pack_texts(texts2, self.device)
             f2, _ = hip.featurize(data2, beg2, end2, self.feat_dim)
             scores2_t = self.classifier.forward(pad_rows(f2, 128))[: len(ok_items)]
 
+        self._toc("rw_launch", t_sub)
         # the rescan kernels are IN FLIGHT — the sync happens inside the
         # returned closure, which the caller awaits AFTER the big native
         # upstream batch so the GPU rescan overlaps that C++ work
         async def finish() -> List[Tuple[int, Any]]:
-            return await self._finish_rewrite(ok_items, harm2_t, scores2_t,
+            return await self._finish_rewrite(ok_items, harm_c, scores2_t,
                                               blob, env, rows, tool_idx, hit, hit_slot, responses)
 
         return finish
 
-    async def _finish_rewrite(self, ok_items, harm2_t, scores2_t,
+    async def _finish_rewrite(self, ok_items, harm_c, scores2_t,
                               blob, env, rows, tool_idx, hit, hit_slot,
                               responses: List[Optional[bytes]]) -> List[Tuple[int, Any]]:
         """Phase B of the rewrite pass: await the rescan verdicts of the
         rewritten texts (harm scan + moderation classifier) and emit the
         surviving (row, args) dispatch items."""
-        await asyncio.to_thread(torch.cuda.synchronize)
-        harm2 = harm2_t.cpu().numpy() if harm2_t is not None else np.zeros(len(ok_items), dtype=np.int64)
+        t_sub = self._tic()
+        if scores2_t is not None:
+            await asyncio.to_thread(torch.cuda.synchronize)
+        self._toc("rw_sync", t_sub)
+        t_sub = self._tic()
         scores2 = scores2_t.cpu().numpy() if scores2_t is not None else None
+
+        # hoisted gating: vectorized score reduction, per-tool applies from
+        # the precomputed arrays, id bytes only for rows that need a splice
+        mod_on = scores2 is not None and self._enforcing(self.moderation)
+        sc_max = scores2.max(axis=1) if mod_on else None
+        sc_arg = scores2.argmax(axis=1) if mod_on else None
+        mod_thr = self.moderation.threshold if mod_on else 0.0
+        sg_on = self._enforcing(self.schema_guard)
+        if mod_on:
+            from ..models.classifier import category_names
+            cat_names = category_names(scores2.shape[1])
+        if sg_on:
+            from ..utils.jsonschema import validate as _validate
+        mod_app, harm_cats = self._t_mod_app, (self.harm.cats if self.harm else [])
 
         out: List[Tuple[int, Any]] = []
         for jj, (j, args2) in enumerate(ok_items):
-            mt = self._meta_list[tool_idx[j]]
-            name = mt.name
-            r = int(rows[j])
-            idb = self._id_bytes(blob, env, r)
-            if scores2 is not None and self._enforcing(self.moderation) and self._applies(self.moderation, name):
-                row_sc = scores2[jj]
-                if float(row_sc.max()) >= self.moderation.threshold:
-                    from ..models.classifier import category_names
-
-                    cat = category_names(len(row_sc))[int(row_sc.argmax())]
-                    responses[r] = self._splice_error(
-                        idb, jsonrpc.POLICY_DENIED,
-                        f"content_moderation: moderation: category {cat} score {float(row_sc.max()):.3f}")
-                    self.blocked += 1
-                    continue
-            if harm2[jj] and self._enforcing(self.harm) and self._applies(self.harm, name):
-                pid = int(harm2[jj]).bit_length() - 1
-                cat = self.harm.cats[pid] if pid < len(self.harm.cats) else "?"
-                responses[r] = self._splice_error(idb, jsonrpc.POLICY_DENIED,
-                                                  f"harmful_content_detector: harmful content ({cat})")
+            ti = tool_idx[j]
+            if mod_on and mod_app[ti] and float(sc_max[jj]) >= mod_thr:
+                r = int(rows[j])
+                cat = cat_names[int(sc_arg[jj])]
+                responses[r] = self._splice_error(
+                    self._id_bytes(blob, env, r), jsonrpc.POLICY_DENIED,
+                    f"content_moderation: moderation: category {cat} score {float(sc_max[jj]):.3f}")
                 self.blocked += 1
                 continue
-            if self._enforcing(self.schema_guard) and mt.schema_mode != "trivial":
-                from ..utils.jsonschema import validate as _validate
-
+            hj = harm_c.get(j, -1)
+            if hj >= 0:   # gating applied where harm_c was computed
+                r = int(rows[j])
+                cat = harm_cats[hj] if hj < len(harm_cats) else "?"
+                responses[r] = self._splice_error(
+                    self._id_bytes(blob, env, r), jsonrpc.POLICY_DENIED,
+                    f"harmful_content_detector: harmful content ({cat})")
+                self.blocked += 1
+                continue
+            mt = self._meta_list[ti]
+            if sg_on and mt.schema_mode != "trivial":
                 if isinstance(args2, bytes):
                     args2 = json.loads(args2)
                 errs = _validate(args2 or {}, mt.tool.get("input_schema") or {})
                 if errs:
-                    responses[r] = self._splice_error(idb, jsonrpc.POLICY_DENIED,
-                                                      "schema_guard: schema violation: " + "; ".join(errs[:5]))
+                    r = int(rows[j])
+                    responses[r] = self._splice_error(
+                        self._id_bytes(blob, env, r), jsonrpc.POLICY_DENIED,
+                        "schema_guard: schema violation: " + "; ".join(errs[:5]))
                     self.blocked += 1
                     continue
             if hit[j]:
+                r = int(rows[j])
                 res = self._cache_result_bytes(int(hit_slot[j]))
+                idb = self._id_bytes(blob, env, r)
                 if idb is not None and res is not None:
                     responses[r] = self._splice_result(idb, res)
                 self.cache_hits += 1
                 continue
             out.append((j, args2))
+        self._toc("rw_verdict", t_sub)
         return out
 
     # ------------------------------------------------------------------

@@ -741,7 +741,13 @@ extern "C" int64_t forge_rewrite_rows(
     int32_t* status, uint32_t* found_bits, int32_t* deny_hit,
     uint8_t* arena, int64_t arena_cap,
     int64_t* out_beg, int64_t* out_end,        // dispatch form (wire key order)
-    int64_t* scan_beg, int64_t* scan_end)      // scan form (sorted keys)
+    int64_t* scan_beg, int64_t* scan_end,      // scan form (sorted keys)
+    // do_flags bit3: harmful_content@60 over the POST-rewrite sorted text
+    // (phrase.lower() in _text_of(args).lower()). REPORTED, not a status —
+    // the caller orders it against the moderation verdict exactly as the
+    // CPU chain does (moderation first).
+    const uint8_t* harm_blob, const int32_t* harm_off, int n_harm,
+    int32_t* harm_out)
 {
     std::string buf;
     buf.reserve((size_t)n * 64);
@@ -749,6 +755,7 @@ extern "C" int64_t forge_rewrite_rows(
         status[i] = RW_PUNT;
         found_bits[i] = 0;
         deny_hit[i] = -1;
+        if (harm_out) harm_out[i] = -1;
         out_beg[i] = out_end[i] = -1;
         scan_beg[i] = scan_end[i] = -1;
         const uint8_t* b = blob + args_beg[i];
@@ -819,6 +826,19 @@ extern "C" int64_t forge_rewrite_rows(
         std::string wire, sorted_s;
         serialize(pool, root, wire, /*sorted=*/false);
         serialize(pool, root, sorted_s, /*sorted=*/true);
+        if ((fl & 8) && n_harm > 0 && harm_out) {
+            std::string hay2 = sorted_s;
+            for (auto& ch : hay2)
+                if (ch >= 'A' && ch <= 'Z') ch += 32;
+            for (int w = 0; w < n_harm; ++w) {
+                size_t wn = (size_t)(harm_off[w + 1] - harm_off[w]);
+                if (wn == 0) continue;
+                if (memmem(hay2.data(), hay2.size(), harm_blob + harm_off[w], wn) != nullptr) {
+                    harm_out[i] = w;
+                    break;
+                }
+            }
+        }
         out_beg[i] = (int64_t)buf.size();
         buf += wire;
         out_end[i] = (int64_t)buf.size();

@@ -48,7 +48,8 @@ def _load() -> ctypes.CDLL:
                               [ctypes.c_uint32] + [ctypes.c_int] * 3 +
                               [ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int, ctypes.c_int] +
                               [ctypes.c_void_p] * 3 +
-                              [ctypes.c_void_p, ctypes.c_int64] + [ctypes.c_void_p] * 4,
+                              [ctypes.c_void_p, ctypes.c_int64] + [ctypes.c_void_p] * 4 +
+                              [ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int, ctypes.c_void_p],
         "forge_post_rows": [ctypes.c_void_p] * 3 + [ctypes.c_int, ctypes.c_void_p] +
                            [ctypes.c_uint32, ctypes.c_int] +
                            [ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int] +
@@ -457,12 +458,14 @@ def rewrite_rows(blob: np.ndarray, args_beg: np.ndarray, args_end: np.ndarray,
                  norm_collapse: bool, norm_strip: bool,
                  deny_blob: Optional[np.ndarray] = None,
                  deny_off: Optional[np.ndarray] = None,
-                 deny_ci: bool = True):
+                 deny_ci: bool = True,
+                 harm_blob: Optional[np.ndarray] = None,
+                 harm_off: Optional[np.ndarray] = None):
     """Native rewrite pass (rewrite.cpp): normalizer + PII over the flagged
     rows that fit the provable-equivalence envelope; everything else gets
     RW_PUNT and takes the Python path.
-    → (status i32[n], found u32[n], arena u8,
-       out_beg/out_end i64[n] (dispatch form, wire key order),
+    → (status i32[n], found u32[n], deny_hit i32[n], harm_hit i32[n],
+       arena u8, out_beg/out_end i64[n] (dispatch form, wire key order),
        scan_beg/scan_end i64[n] (sorted-keys scan form))."""
     n = args_beg.shape[0]
     lib = _load()
@@ -473,7 +476,9 @@ def rewrite_rows(blob: np.ndarray, args_beg: np.ndarray, args_end: np.ndarray,
     scan_beg = np.empty(n, dtype=np.int64)
     scan_end = np.empty(n, dtype=np.int64)
     deny_hit = np.empty(n, dtype=np.int32)
+    harm_hit = np.empty(n, dtype=np.int32)
     n_deny = (deny_off.shape[0] - 1) if deny_off is not None else 0
+    n_harm = (harm_off.shape[0] - 1) if harm_off is not None else 0
     cap = int((args_end - args_beg).sum()) * 2 + n * 32 + 4096
     while True:
         arena = np.empty(cap, dtype=np.uint8)
@@ -485,9 +490,10 @@ def rewrite_rows(blob: np.ndarray, args_beg: np.ndarray, args_end: np.ndarray,
             _np_ptr(deny_blob), _np_ptr(deny_off), n_deny, 1 if deny_ci else 0,
             _np_ptr(status), _np_ptr(found), _np_ptr(deny_hit),
             _np_ptr(arena), cap, _np_ptr(out_beg), _np_ptr(out_end),
-            _np_ptr(scan_beg), _np_ptr(scan_end))
+            _np_ptr(scan_beg), _np_ptr(scan_end),
+            _np_ptr(harm_blob), _np_ptr(harm_off), n_harm, _np_ptr(harm_hit))
         if rc >= 0:
-            return status, found, deny_hit, arena, out_beg, out_end, scan_beg, scan_end
+            return status, found, deny_hit, harm_hit, arena, out_beg, out_end, scan_beg, scan_end
         cap = -int(rc) + 4096
 
 
