@@ -55,14 +55,25 @@ def parse_batch(raws: List[bytes]) -> List[ParsedRequest]:
     return out
 
 
+def _pow2_upload(arr: np.ndarray, device: str) -> torch.Tensor:
+    """H2D into a pow2-bucketed device alloc (recycles under the caching
+    allocator even when per-call sizes drift — exact-size allocs degrade
+    into real hipMalloc on hot paths)."""
+    n = arr.nbytes
+    cap = 1 << max(12, (n - 1).bit_length()) if n else 1
+    dev = torch.empty(cap, dtype=torch.uint8, device=device)[:n]
+    dev.copy_(torch.from_numpy(arr.view(np.uint8).reshape(-1)), non_blocking=True)
+    return dev
+
+
 def pack_texts(texts: List[bytes], device: str = "cuda") -> Tuple[torch.Tensor, torch.Tensor, torch.Tensor]:
     """Pack byte strings into (data u8 [total], beg i32 [B], end i32 [B]) on device."""
     offsets = np.zeros(len(texts) + 1, dtype=np.int32)
     np.cumsum(np.fromiter(map(len, texts), dtype=np.int32, count=len(texts)), out=offsets[1:])
     blob = b"".join(texts)
-    data_np = np.frombuffer(blob, dtype=np.uint8) if blob else np.zeros(1, dtype=np.uint8)
-    data = torch.from_numpy(data_np.copy()).to(device, non_blocking=True)
-    offs = torch.from_numpy(offsets).to(device, non_blocking=True)
+    data_np = np.frombuffer(blob, dtype=np.uint8).copy() if blob else np.zeros(1, dtype=np.uint8)
+    data = _pow2_upload(data_np, device)
+    offs = _pow2_upload(offsets, device).view(torch.int32)
     return data, offs[:-1], offs[1:]
 
 

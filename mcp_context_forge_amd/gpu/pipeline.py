@@ -787,8 +787,7 @@ class GpuPluginPipeline:
         out_multi = hip.scan_multi(data_gpu, beg_t, end_t, self._bankset1)
         feats = None
         if self.classifier is not None or self.semcache is not None:
-            feats_b, _ = hip.featurize(data_gpu, beg_t, end_t, self.feat_dim)
-            feats = pad_rows(feats_b, 128)
+            feats, _ = hip.featurize(data_gpu, beg_t, end_t, self.feat_dim, pad_to=128)
         scores_t = self.classifier.forward(feats)[:m] if self.classifier is not None else None
         cache_val_t = cache_idx_t = None
         feats_sk = None
@@ -986,7 +985,13 @@ class GpuPluginPipeline:
         self._pin_off = off + n
         view = self._pin[off:off + n]
         view.numpy()[:] = raw
-        dev = view.to(self.device, non_blocking=True)
+        # pow2-bucketed device alloc: per-batch byte counts vary by a few
+        # KB, and exact-size allocs make the caching allocator split blocks
+        # and fall into real hipMalloc on the hot path (measured ~1ms/batch
+        # of launch-side stalls) — pow2 buckets recycle perfectly
+        cap = 1 << max(16, (n - 1).bit_length()) if n else 1
+        dev = torch.empty(cap, dtype=torch.uint8, device=self.device)[:n]
+        dev.copy_(view, non_blocking=True)
         tdt = {np.dtype(np.uint8): torch.uint8, np.dtype(np.int32): torch.int32,
                np.dtype(np.int64): torch.int64, np.dtype(np.float32): torch.float32}[np.dtype(arr.dtype)]
         return dev.view(tdt).reshape(arr.shape)
@@ -1181,8 +1186,8 @@ class GpuPluginPipeline:
         scores2_t = None
         if self.classifier is not None:
             data2, beg2, end2 = pack_texts(texts2, self.device)
-            f2, _ = hip.featurize(data2, beg2, end2, self.feat_dim)
-            scores2_t = self.classifier.forward(pad_rows(f2, 128))[: len(ok_items)]
+            f2, _ = hip.featurize(data2, beg2, end2, self.feat_dim, pad_to=128)
+            scores2_t = self.classifier.forward(f2)[: len(ok_items)]
 
         self._toc("rw_launch", t_sub)
         # the rescan kernels are IN FLIGHT — the sync happens inside the

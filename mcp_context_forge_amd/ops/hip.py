@@ -156,10 +156,18 @@ def scan_multi(data: torch.Tensor, beg: torch.Tensor, end: torch.Tensor,
 
 
 def featurize(data: torch.Tensor, beg: torch.Tensor, end: torch.Tensor, dim: int,
-              want_f32: bool = False) -> Tuple[torch.Tensor, Optional[torch.Tensor]]:
+              want_f32: bool = False, pad_to: int = 0) -> Tuple[torch.Tensor, Optional[torch.Tensor]]:
+    """pad_to > 0: allocate (and zero) dim0 rounded up to that multiple and
+    return the FULL padded tensor (kernel fills the first `batch` rows) —
+    stable alloc shapes for the caching allocator and MFMA-ready M."""
     batch = beg.numel()
-    out_bf16 = torch.empty((batch, dim), dtype=torch.bfloat16, device=data.device)
-    out_f32 = torch.empty((batch, dim), dtype=torch.float32, device=data.device) if want_f32 else None
+    rows = ((batch + pad_to - 1) // pad_to) * pad_to if pad_to else batch
+    out_bf16 = torch.empty((rows, dim), dtype=torch.bfloat16, device=data.device)
+    out_f32 = torch.empty((rows, dim), dtype=torch.float32, device=data.device) if want_f32 else None
+    if rows > batch:
+        out_bf16[batch:].zero_()
+        if out_f32 is not None:
+            out_f32[batch:].zero_()
     _check("forge_featurize", _load().forge_featurize(
         _ptr(data), _ptr(beg), _ptr(end), batch, dim, _ptr(out_bf16), _ptr(out_f32), _stream()))
     return out_bf16, out_f32
