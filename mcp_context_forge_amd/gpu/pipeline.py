@@ -799,7 +799,7 @@ class GpuPluginPipeline:
         # off-loop sync: a blocking synchronize would stall the event loop and
         # starve the micro-batch collector under live HTTP load
         try:
-            await asyncio.to_thread(torch.cuda.synchronize)
+            await self._await_gpu()
         finally:
             self._gpu_lock.release()
         self._toc("gp1_sync", t_s)
@@ -971,6 +971,22 @@ class GpuPluginPipeline:
 
     def _pin_reset(self) -> None:
         self._pin_off = 0
+
+    async def _await_gpu(self) -> None:
+        """Await completion of the current stream WITHOUT
+        hipDeviceSynchronize: a device-wide sync holds HIP runtime locks
+        that serialize the OTHER in-flight batch's kernel launches
+        (measured ~1ms/batch of launch-side stalls). An event record +
+        non-blocking hipEventQuery poll yields the event loop instead."""
+        ev = torch.cuda.Event()
+        ev.record()
+        if ev.query():
+            return
+        for _ in range(200000):
+            await asyncio.sleep(0)
+            if ev.query():
+                return
+        await asyncio.to_thread(ev.synchronize)
 
     def _upload(self, arr: np.ndarray) -> torch.Tensor:
         """Stage a host array through the pinned arena → async H2D. Safe to
@@ -1207,7 +1223,7 @@ class GpuPluginPipeline:
         surviving (row, args) dispatch items."""
         t_sub = self._tic()
         if scores2_t is not None:
-            await asyncio.to_thread(torch.cuda.synchronize)
+            await self._await_gpu()
         self._toc("rw_sync", t_sub)
         t_sub = self._tic()
         scores2 = scores2_t.cpu().numpy() if scores2_t is not None else None
@@ -1422,7 +1438,7 @@ class GpuPluginPipeline:
             e3 = self._upload(res_end.astype(np.int32))
             out3 = hip.scan_multi(data3, b3, e3, self._bankset3)  # fused, one launch
             try:
-                await asyncio.to_thread(torch.cuda.synchronize)
+                await self._await_gpu()
             finally:
                 self._gpu_lock.release()
             mm3 = out3.cpu().numpy().view(np.uint32)
