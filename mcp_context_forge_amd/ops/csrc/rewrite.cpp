@@ -30,6 +30,7 @@
 
 #include <algorithm>
 #include <string>
+#include <thread>
 #include <vector>
 
 namespace {
@@ -732,26 +733,22 @@ enum : int32_t { RW_DONE = 0, RW_PUNT = 1, RW_BLOCKED = 2, RW_BADJSON = 3, RW_DE
 // Outputs: status[], found_bits[] (pii categories), deny_hit[] (first
 // matching deny-word index, -1 none), canonical rewritten args in the
 // arena (grow-retry contract like forge_decide).
-extern "C" int64_t forge_rewrite_rows(
-    const uint8_t* blob, const int32_t* args_beg, const int32_t* args_end, int n,
+static void rewrite_rows_range(
+    const uint8_t* blob, const int32_t* args_beg, const int32_t* args_end,
+    int r0, int r1,
     const uint8_t* do_flags, const uint32_t* pii_want,
-    uint32_t pii_active_mask, int pii_mode /*0 mask, 1 block, 2 audit*/,
+    uint32_t pii_active_mask, int pii_mode,
     int norm_collapse, int norm_strip,
     const uint8_t* deny_blob, const int32_t* deny_off, int n_deny, int deny_ci,
     int32_t* status, uint32_t* found_bits, int32_t* deny_hit,
-    uint8_t* arena, int64_t arena_cap,
-    int64_t* out_beg, int64_t* out_end,        // dispatch form (wire key order)
-    int64_t* scan_beg, int64_t* scan_end,      // scan form (sorted keys)
-    // do_flags bit3: harmful_content@60 over the POST-rewrite sorted text
-    // (phrase.lower() in _text_of(args).lower()). REPORTED, not a status —
-    // the caller orders it against the moderation verdict exactly as the
-    // CPU chain does (moderation first).
+    std::string& buf,
+    int64_t* out_beg, int64_t* out_end,
+    int64_t* scan_beg, int64_t* scan_end,
     const uint8_t* harm_blob, const int32_t* harm_off, int n_harm,
     int32_t* harm_out)
 {
-    std::string buf;
-    buf.reserve((size_t)n * 64);
-    for (int i = 0; i < n; ++i) {
+    buf.reserve((size_t)(r1 - r0) * 64);
+    for (int i = r0; i < r1; ++i) {
         status[i] = RW_PUNT;
         found_bits[i] = 0;
         deny_hit[i] = -1;
@@ -852,9 +849,63 @@ extern "C" int64_t forge_rewrite_rows(
         }
         status[i] = RW_DONE;
     }
-    if ((int64_t)buf.size() > arena_cap) return -(int64_t)buf.size();
-    if (!buf.empty()) memcpy(arena, buf.data(), buf.size());
-    return (int64_t)buf.size();
+}
+
+// Rewrite a batch of flagged rows — threaded over row ranges (rows are
+// independent; per-thread arenas are stitched and spans rebased).
+extern "C" int64_t forge_rewrite_rows(
+    const uint8_t* blob, const int32_t* args_beg, const int32_t* args_end, int n,
+    const uint8_t* do_flags, const uint32_t* pii_want,
+    uint32_t pii_active_mask, int pii_mode /*0 mask, 1 block, 2 audit*/,
+    int norm_collapse, int norm_strip,
+    const uint8_t* deny_blob, const int32_t* deny_off, int n_deny, int deny_ci,
+    int32_t* status, uint32_t* found_bits, int32_t* deny_hit,
+    uint8_t* arena, int64_t arena_cap,
+    int64_t* out_beg, int64_t* out_end,        // dispatch form (wire key order)
+    int64_t* scan_beg, int64_t* scan_end,      // scan form (sorted keys)
+    // do_flags bit3: harmful_content@60 over the POST-rewrite sorted text
+    // (phrase.lower() in _text_of(args).lower()). REPORTED, not a status —
+    // the caller orders it against the moderation verdict exactly as the
+    // CPU chain does (moderation first).
+    const uint8_t* harm_blob, const int32_t* harm_off, int n_harm,
+    int32_t* harm_out)
+{
+    int nthreads = n >= 512 ? 8 : (n >= 64 ? 4 : 1);
+    std::vector<std::string> bufs((size_t)nthreads);
+    int chunk = (n + nthreads - 1) / nthreads;
+    auto run = [&](int t) {
+        int r0 = t * chunk, r1 = r0 + chunk < n ? r0 + chunk : n;
+        if (r0 >= r1) return;
+        rewrite_rows_range(blob, args_beg, args_end, r0, r1, do_flags, pii_want,
+                           pii_active_mask, pii_mode, norm_collapse, norm_strip,
+                           deny_blob, deny_off, n_deny, deny_ci,
+                           status, found_bits, deny_hit, bufs[(size_t)t],
+                           out_beg, out_end, scan_beg, scan_end,
+                           harm_blob, harm_off, n_harm, harm_out);
+    };
+    if (nthreads == 1) {
+        run(0);
+    } else {
+        std::vector<std::thread> threads;
+        for (int t = 0; t < nthreads; ++t) threads.emplace_back(run, t);
+        for (auto& th : threads) th.join();
+    }
+    int64_t total = 0;
+    std::vector<int64_t> base((size_t)nthreads, 0);
+    for (int t = 0; t < nthreads; ++t) { base[(size_t)t] = total; total += (int64_t)bufs[(size_t)t].size(); }
+    if (total > arena_cap) return -total;
+    for (int t = 0; t < nthreads; ++t) {
+        if (!bufs[(size_t)t].empty()) memcpy(arena + base[(size_t)t], bufs[(size_t)t].data(), bufs[(size_t)t].size());
+        int r0 = t * chunk, r1 = r0 + chunk < n ? r0 + chunk : n;
+        for (int r = r0; r < r1; ++r)
+            if (out_beg[r] >= 0) {
+                out_beg[r] += base[(size_t)t];
+                out_end[r] += base[(size_t)t];
+                scan_beg[r] += base[(size_t)t];
+                scan_end[r] += base[(size_t)t];
+            }
+    }
+    return total;
 }
 
 // Native result post chain: the _host_post hot path (gpu/pipeline.py) for
@@ -870,18 +921,18 @@ extern "C" int64_t forge_rewrite_rows(
 //   do_flags: bit0 pii, bit1 harm, bit2 toon
 //   harm_blob/off: phrases, pre-lowercased ASCII, plugin order
 //   is_err: truthiness of result["isError"] on the FINAL result
-extern "C" int64_t forge_post_rows(
-    const uint8_t* blob, const int64_t* res_beg, const int64_t* res_end, int n,
+static void post_rows_range(
+    const uint8_t* blob, const int64_t* res_beg, const int64_t* res_end,
+    int r0, int r1,
     const uint8_t* do_flags,
-    uint32_t pii_active_mask, int pii_mode /*0 mask, 1 block, 2 audit*/,
+    uint32_t pii_active_mask, int pii_mode,
     const uint8_t* harm_blob, const int32_t* harm_off, int n_harm,
     int64_t toon_min_size, double toon_min_savings,
     int32_t* status, uint32_t* found_bits, int32_t* harm_hit, uint8_t* is_err,
-    uint8_t* arena, int64_t arena_cap, int64_t* out_beg, int64_t* out_end)
+    std::string& buf, int64_t* out_beg, int64_t* out_end)
 {
-    std::string buf;
-    buf.reserve((size_t)n * 128);
-    for (int i = 0; i < n; ++i) {
+    buf.reserve((size_t)(r1 - r0) * 128);
+    for (int i = r0; i < r1; ++i) {
         status[i] = RW_PUNT;
         found_bits[i] = 0;
         harm_hit[i] = -1;
@@ -1031,7 +1082,48 @@ extern "C" int64_t forge_post_rows(
         out_end[i] = (int64_t)buf.size();
         status[i] = RW_DONE;
     }
-    if ((int64_t)buf.size() > arena_cap) return -(int64_t)buf.size();
-    if (!buf.empty()) memcpy(arena, buf.data(), buf.size());
-    return (int64_t)buf.size();
+}
+
+extern "C" int64_t forge_post_rows(
+    const uint8_t* blob, const int64_t* res_beg, const int64_t* res_end, int n,
+    const uint8_t* do_flags,
+    uint32_t pii_active_mask, int pii_mode /*0 mask, 1 block, 2 audit*/,
+    const uint8_t* harm_blob, const int32_t* harm_off, int n_harm,
+    int64_t toon_min_size, double toon_min_savings,
+    int32_t* status, uint32_t* found_bits, int32_t* harm_hit, uint8_t* is_err,
+    uint8_t* arena, int64_t arena_cap, int64_t* out_beg, int64_t* out_end)
+{
+    int nthreads = n >= 512 ? 8 : (n >= 64 ? 4 : 1);
+    std::vector<std::string> bufs((size_t)nthreads);
+    int chunk = (n + nthreads - 1) / nthreads;
+    auto run = [&](int t) {
+        int r0 = t * chunk, r1 = r0 + chunk < n ? r0 + chunk : n;
+        if (r0 >= r1) return;
+        post_rows_range(blob, res_beg, res_end, r0, r1, do_flags,
+                        pii_active_mask, pii_mode, harm_blob, harm_off, n_harm,
+                        toon_min_size, toon_min_savings,
+                        status, found_bits, harm_hit, is_err,
+                        bufs[(size_t)t], out_beg, out_end);
+    };
+    if (nthreads == 1) {
+        run(0);
+    } else {
+        std::vector<std::thread> threads;
+        for (int t = 0; t < nthreads; ++t) threads.emplace_back(run, t);
+        for (auto& th : threads) th.join();
+    }
+    int64_t total = 0;
+    std::vector<int64_t> base((size_t)nthreads, 0);
+    for (int t = 0; t < nthreads; ++t) { base[(size_t)t] = total; total += (int64_t)bufs[(size_t)t].size(); }
+    if (total > arena_cap) return -total;
+    for (int t = 0; t < nthreads; ++t) {
+        if (!bufs[(size_t)t].empty()) memcpy(arena + base[(size_t)t], bufs[(size_t)t].data(), bufs[(size_t)t].size());
+        int r0 = t * chunk, r1 = r0 + chunk < n ? r0 + chunk : n;
+        for (int r = r0; r < r1; ++r)
+            if (out_beg[r] >= 0) {
+                out_beg[r] += base[(size_t)t];
+                out_end[r] += base[(size_t)t];
+            }
+    }
+    return total;
 }

@@ -200,3 +200,22 @@ def test_toon_codec_roundtrip_of_c_output():
     text = out["content"][0]["text"]
     assert text == toon_codec.encode(sc)
     assert toon_codec.decode(text) == sc
+
+
+def test_threaded_range_stitching():
+    """n >= 512 engages the 8-thread row split in forge_post_rows."""
+    raws = []
+    for i in range(700):
+        raws.append(json.dumps(
+            {"content": [{"type": "text", "text": f"r{i} a@b{i % 5}.co"}],
+             "structuredContent": {"rows": [{"a": i, "b": f"v{i}"}] * 12},
+             "isError": i % 13 == 0}, separators=(",", ":")).encode())
+    st, found, hh, ie, arena, ob, oe = run_c(raws, fl=5, toon_min_size=64)
+    n_done = 0
+    for i, raw in enumerate(raws):
+        want_b, blocked, want_err = py_reference(raw, fl=5, toon_min_size=64)
+        assert blocked is None and st[i] == hip.RW_DONE, (i, st[i])
+        assert arena[ob[i]:oe[i]].tobytes() == want_b, i
+        assert bool(ie[i]) == want_err, i
+        n_done += 1
+    assert n_done == 700

@@ -269,3 +269,21 @@ def test_deny_check_matches_python():
         else:
             assert status[i] == hip.RW_DENY, (i, raw, status[i])
             assert deny_hit[i] == want, (i, raw, deny_hit[i])
+
+
+def test_threaded_range_stitching():
+    """n >= 512 engages the 8-thread row split — spans must rebase into one
+    arena with byte-exact content (same outputs as the single-row calls)."""
+    rng = random.Random(42)
+    raws = []
+    for i in range(700):
+        obj = {"msg": f"row {i} mail a@b{i % 7}.co data", "n": i,
+               "pad": "x" * rng.randrange(0, 60)}
+        raws.append(json.dumps(obj, separators=(",", ":")).encode())
+    status, found, deny_hit, harm_hit, arena, ob, oe, sb, se = run_c(raws)
+    for i, raw in enumerate(raws):
+        sorted_exp, wire_exp, names = py_reference(raw)
+        assert status[i] == hip.RW_DONE, (i, status[i])
+        assert arena[ob[i]:oe[i]].tobytes() == wire_exp, i
+        assert arena[sb[i]:se[i]].tobytes() == sorted_exp, i
+        assert names_of(int(found[i])) == names, i
