@@ -122,9 +122,14 @@ bool is_canonical_span(const uint8_t* b, const uint8_t* e) {
         if (c == '"') { in_str = true; ++p; continue; }
         if (c == ' ' || c == '\t' || c == '\n' || c == '\r') return false;
         if (c == '-' || (c >= '0' && c <= '9')) {  // number: integers only
-            if (c == '-') ++p;
+            bool neg = c == '-';
+            if (neg) ++p;
+            const uint8_t* d0 = p;
             while (p < e && *p >= '0' && *p <= '9') ++p;
+            if (p == d0) return false;                       // bare '-'
             if (p < e && (*p == '.' || *p == 'e' || *p == 'E')) return false;
+            if (p - d0 > 1 && *d0 == '0') return false;      // leading zeros
+            if (neg && p - d0 == 1 && *d0 == '0') return false;  // -0 → python re-dumps as 0
             continue;
         }
         if (c == 't') { if (e - p < 4 || memcmp(p, "true", 4) != 0) return false; p += 4; continue; }

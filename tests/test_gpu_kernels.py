@@ -362,3 +362,26 @@ def test_scan_multi_matches_per_bank():
         single, _ = hip.scan(data, beg, end, bank)
         torch.cuda.synchronize()
         assert torch.equal(fused[bs.index[name]], single), name
+
+
+def test_upstream_canonical_gate_number_forms():
+    """Parity-fuzz find (round 2, widened generators): the C++ echo upstream
+    must PUNT '-0' spans — python json round-trips -0 → 0, so echoing the
+    raw span verbatim diverges. Leading zeros and bare '-' punt too."""
+    import numpy as np
+
+    from mcp_context_forge_amd.ops import hip
+
+    def spans(raw):
+        blob = np.frombuffer(raw, dtype=np.uint8)
+        _, rb, re_ = hip.upstream_call_batch(
+            blob, np.array([0], np.int32), np.array([len(raw)], np.int32),
+            np.array([2], np.int32), "2026-01-01T00:00:00Z")
+        return int(rb[0]), int(re_[0])
+
+    for bad in (b'{"n":-0}', b'{"n":0123}', b'{"n":-}', b'{"a":[-0]}'):
+        rb, re_ = spans(bad)
+        assert rb == re_, bad  # punt → host path computes the exact result
+    for ok in (b'{"n":0}', b'{"n":-5}', b'{"n":123}', b'{"a":[0,1]}'):
+        rb, re_ = spans(ok)
+        assert re_ > rb, ok
