@@ -31,11 +31,11 @@ TARGETS = {
     "mcp_context_forge_amd/protocol/jsonrpc.py": ["tests/test_jsonrpc.py"],
     "mcp_context_forge_amd/plugins/toon.py": ["tests/test_plugins.py"],
     "mcp_context_forge_amd/auth/jwt.py": ["tests/test_security_fixes.py", "tests/test_auth_breadth.py"],
-    "mcp_context_forge_amd/auth/rsa.py": ["tests/test_auth_breadth.py"],
-    "mcp_context_forge_amd/auth/oauth.py": ["tests/test_auth_breadth.py"],
+    "mcp_context_forge_amd/auth/rsa.py": ["tests/test_auth_breadth.py", "tests/test_mutation_kills.py"],
+    "mcp_context_forge_amd/auth/oauth.py": ["tests/test_auth_breadth.py", "tests/test_mutation_kills.py"],
     "mcp_context_forge_amd/plugins/framework.py": ["tests/test_plugins.py"],
     "mcp_context_forge_amd/plugins/external.py": ["tests/test_external_plugins.py"],
-    "mcp_context_forge_amd/services/leader.py": ["tests/test_leader_election.py"],
+    "mcp_context_forge_amd/services/leader.py": ["tests/test_leader_election.py", "tests/test_mutation_kills.py"],
     "mcp_context_forge_amd/services/gateway_service.py": ["tests/test_federation_lifecycle.py",
                                                           "tests/test_federation.py"],
     "mcp_context_forge_amd/services/sessions.py": ["tests/test_http_app.py"],
