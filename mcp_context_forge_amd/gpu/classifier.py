@@ -30,3 +30,9 @@ class GpuClassifier:
         """feats [Bpad, D] bf16 (Bpad%128==0) → probabilities fp32 [Bpad, C]."""
         h = hip.gemm_bt(feats_bf16, self.w1t, self.b1, act=hip.ACT_GELU, out_bf16=True)
         return hip.gemv_head(h, self.w2t, self.b2, act=hip.ACT_SIGMOID)
+
+    def forward_into(self, feats_bf16: torch.Tensor, h_buf: torch.Tensor,
+                     scores_buf: torch.Tensor) -> None:
+        """Capture-safe forward: writes into preallocated buffers."""
+        hip.gemm_bt(feats_bf16, self.w1t, self.b1, act=hip.ACT_GELU, out_bf16=True, out=h_buf)
+        hip.gemv_head(h_buf, self.w2t, self.b2, act=hip.ACT_SIGMOID, out=scores_buf)

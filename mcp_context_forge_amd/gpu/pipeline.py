@@ -238,6 +238,7 @@ class GpuPluginPipeline:
             [(n, self.banks[n]) for n in ("pii", "regex", "harm", "postmeta") if n in self.banks],
             device)
         self._bankset1: Optional[hip.ScanBankSet] = None
+        self._graphs = None  # Pass1Graphs; rebuilt when _bankset1 changes
 
         # native decision-plane stores + string tables (fastpath.cpp)
         self._slot_store = hip.store_new(self.semcache.capacity) if self.semcache is not None else 0
@@ -424,6 +425,7 @@ class GpuPluginPipeline:
         if self._schema_bank is not None:
             b1.append(("schema", self._schema_bank))
         self._bankset1 = hip.ScanBankSet(b1, self.device)
+        self._graphs = None  # captured graphs bake the old bank pointers
 
         # ---- flat arrays for the native decision plane (fastpath.cpp) ----
         metas = list(self._tool_meta.values())
@@ -775,38 +777,69 @@ class GpuPluginPipeline:
         # --- GPU pass 1 over raw argument spans ---
         t_g = self._tic()
         await self._gpu_lock.acquire()
-        self._pin_reset()
-        data_gpu = self._upload(blob)
-        beg_t = self._upload(args_b)
-        end_t = self._upload(args_e)
-        self._toc("gp1_upload", t_g)
-        t_l = self._tic()
-        # fused multi-bank scan: all DFA banks in ONE launch (grid.y=bank;
-        # fills the chip bank-parallel where per-bank launches ran ~128
-        # waves each) writing one [n_banks, m] mask matrix — one D2H later
-        out_multi = hip.scan_multi(data_gpu, beg_t, end_t, self._bankset1)
+        # hipGraph fast path (gpu/graphs.py): the whole pass — H2D stages,
+        # fused scan, featurize, classifier, D2H into pinned — is ONE
+        # hipGraphLaunch when the batch fits a bucket and the semcache is
+        # off (the cache probe stays eager)
+        gbucket = None
+        if self.semcache is None:
+            if self._graphs is None:
+                from .graphs import Pass1Graphs
+                self._graphs = Pass1Graphs(self)
+            gbucket = self._graphs.get(m, int(blob.nbytes))
         feats = None
-        if self.classifier is not None or self.semcache is not None:
-            feats, _ = hip.featurize(data_gpu, beg_t, end_t, self.feat_dim, pad_to=128)
-        scores_t = self.classifier.forward(feats)[:m] if self.classifier is not None else None
+        scores_t = None
+        sc_np = None
         cache_val_t = cache_idx_t = None
         feats_sk = None
-        if self.semcache is not None:
-            bv, bi, feats_sk = self.semcache.lookup(feats)
-            cache_val_t, cache_idx_t = bv[:m], bi[:m]
-        self._toc("gp1_launch", t_l)
-        t_s = self._tic()
-        # off-loop sync: a blocking synchronize would stall the event loop and
-        # starve the micro-batch collector under live HTTP load
-        try:
-            await self._await_gpu()
-        finally:
-            self._gpu_lock.release()
-        self._toc("gp1_sync", t_s)
+        if gbucket is not None:
+            gbucket.stage(blob, args_b, args_e, m)
+            self._toc("gp1_upload", t_g)
+            t_l = self._tic()
+            gbucket.replay()
+            self._graphs.replays += 1
+            self._toc("gp1_launch", t_l)
+            t_s = self._tic()
+            try:
+                await self._await_gpu()
+                mm = gbucket.read_multi(m)
+                sc_np = gbucket.read_scores(m) if self.classifier is not None else None
+            finally:
+                self._gpu_lock.release()
+            self._toc("gp1_sync", t_s)
+        else:
+            self._pin_reset()
+            data_gpu = self._upload(blob)
+            beg_t = self._upload(args_b)
+            end_t = self._upload(args_e)
+            self._toc("gp1_upload", t_g)
+            t_l = self._tic()
+            # fused multi-bank scan: all DFA banks in ONE launch (grid.y=bank;
+            # fills the chip bank-parallel where per-bank launches ran ~128
+            # waves each) writing one [n_banks, m] mask matrix — one D2H later
+            out_multi = hip.scan_multi(data_gpu, beg_t, end_t, self._bankset1)
+            if self.classifier is not None or self.semcache is not None:
+                feats, _ = hip.featurize(data_gpu, beg_t, end_t, self.feat_dim, pad_to=128)
+            scores_t = self.classifier.forward(feats)[:m] if self.classifier is not None else None
+            if self.semcache is not None:
+                bv, bi, feats_sk = self.semcache.lookup(feats)
+                cache_val_t, cache_idx_t = bv[:m], bi[:m]
+            self._toc("gp1_launch", t_l)
+            t_s = self._tic()
+            # off-loop sync: a blocking synchronize would stall the event loop
+            # and starve the micro-batch collector under live HTTP load
+            try:
+                await self._await_gpu()
+            finally:
+                self._gpu_lock.release()
+            self._toc("gp1_sync", t_s)
         self._toc("gpu_pass1", t_g)
         t_d = self._tic()
 
-        mm = out_multi.cpu().numpy().view(np.uint32)  # one D2H for every bank
+        if gbucket is None:
+            mm = out_multi.cpu().numpy().view(np.uint32)  # one D2H for every bank
+        if scores_t is not None:
+            sc_np = scores_t.cpu().numpy()
 
         def mask(name):
             i = self._bankset1.index.get(name)
@@ -831,10 +864,9 @@ class GpuPluginPipeline:
         mod_block = np.zeros(m, dtype=np.uint8)
         mod_cat = np.zeros(m, dtype=np.int32)
         mod_score = np.zeros(m, dtype=np.float32)
-        if scores_t is not None:
-            sc = scores_t.cpu().numpy()
-            mod_score = np.ascontiguousarray(sc.max(axis=1).astype(np.float32))
-            mod_cat = np.ascontiguousarray(sc.argmax(axis=1).astype(np.int32))
+        if sc_np is not None:
+            mod_score = np.ascontiguousarray(sc_np.max(axis=1).astype(np.float32))
+            mod_cat = np.ascontiguousarray(sc_np.argmax(axis=1).astype(np.int32))
             mod_block = (mod_score >= self.moderation.threshold).astype(np.uint8)
 
         # semcache identity = tool_hash XOR user_hash (tenant-scoped)

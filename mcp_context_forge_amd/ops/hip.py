@@ -142,10 +142,14 @@ class ScanBankSet:
 
 
 def scan_multi(data: torch.Tensor, beg: torch.Tensor, end: torch.Tensor,
-               bankset: ScanBankSet) -> torch.Tensor:
-    """→ int32 [n_banks, B] mask matrix (bit-identical to per-bank scan)."""
+               bankset: ScanBankSet, out: Optional[torch.Tensor] = None) -> torch.Tensor:
+    """→ int32 [n_banks, B] mask matrix (bit-identical to per-bank scan).
+    `out` (preallocated [n_banks, B]) makes the call graph-capture safe."""
     batch = beg.numel()
-    out = torch.zeros((bankset.n, batch), dtype=torch.int32, device=data.device)
+    if out is None:
+        out = torch.zeros((bankset.n, batch), dtype=torch.int32, device=data.device)
+    else:
+        out.zero_()
     if bankset.n == 0 or batch == 0:
         return out
     _check("forge_scan_multi", _load().forge_scan_multi(
@@ -156,11 +160,17 @@ def scan_multi(data: torch.Tensor, beg: torch.Tensor, end: torch.Tensor,
 
 
 def featurize(data: torch.Tensor, beg: torch.Tensor, end: torch.Tensor, dim: int,
-              want_f32: bool = False, pad_to: int = 0) -> Tuple[torch.Tensor, Optional[torch.Tensor]]:
+              want_f32: bool = False, pad_to: int = 0,
+              out: Optional[torch.Tensor] = None) -> Tuple[torch.Tensor, Optional[torch.Tensor]]:
     """pad_to > 0: allocate (and zero) dim0 rounded up to that multiple and
     return the FULL padded tensor (kernel fills the first `batch` rows) —
-    stable alloc shapes for the caching allocator and MFMA-ready M."""
+    stable alloc shapes for the caching allocator and MFMA-ready M.
+    `out` (preallocated bf16 [batch, dim]) makes the call capture-safe."""
     batch = beg.numel()
+    if out is not None:
+        _check("forge_featurize", _load().forge_featurize(
+            _ptr(data), _ptr(beg), _ptr(end), batch, dim, _ptr(out), _ptr(None), _stream()))
+        return out, None
     rows = ((batch + pad_to - 1) // pad_to) * pad_to if pad_to else batch
     out_bf16 = torch.empty((rows, dim), dtype=torch.bfloat16, device=data.device)
     out_f32 = torch.empty((rows, dim), dtype=torch.float32, device=data.device) if want_f32 else None
@@ -187,14 +197,16 @@ ACT_NONE, ACT_GELU, ACT_SIGMOID = 0, 1, 2
 
 
 def gemm_bt(a: torch.Tensor, bt: torch.Tensor, bias: Optional[torch.Tensor] = None,
-            act: int = ACT_NONE, out_bf16: bool = False) -> torch.Tensor:
+            act: int = ACT_NONE, out_bf16: bool = False,
+            out: Optional[torch.Tensor] = None) -> torch.Tensor:
     """C[M,N] = act(A[M,K] @ BT[N,K]^T + bias). M%128==0, N%128==0, K%64==0."""
     assert a.dtype == torch.bfloat16 and bt.dtype == torch.bfloat16
     assert a.is_contiguous() and bt.is_contiguous()
     m, k = a.shape
     n, k2 = bt.shape
     assert k == k2, (a.shape, bt.shape)
-    out = torch.empty((m, n), dtype=torch.bfloat16 if out_bf16 else torch.float32, device=a.device)
+    if out is None:
+        out = torch.empty((m, n), dtype=torch.bfloat16 if out_bf16 else torch.float32, device=a.device)
     # v2 (256² tile, 4-phase ring, counted vmcnt) when shapes allow; v1 fallback
     import os
 
@@ -210,11 +222,12 @@ def gemm_bt(a: torch.Tensor, bt: torch.Tensor, bias: Optional[torch.Tensor] = No
 
 
 def gemv_head(a: torch.Tensor, wt: torch.Tensor, bias: Optional[torch.Tensor] = None,
-              act: int = ACT_NONE) -> torch.Tensor:
+              act: int = ACT_NONE, out: Optional[torch.Tensor] = None) -> torch.Tensor:
     m, k = a.shape
     c, k2 = wt.shape
     assert k == k2 and c <= 32
-    out = torch.empty((m, c), dtype=torch.float32, device=a.device)
+    if out is None:
+        out = torch.empty((m, c), dtype=torch.float32, device=a.device)
     _check("forge_gemv_head", _load().forge_gemv_head(_ptr(a), _ptr(wt), _ptr(bias), _ptr(out), m, c, k, act, _stream()))
     return out
 

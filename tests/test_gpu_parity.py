@@ -221,3 +221,44 @@ def test_a2a_fast_lane_parity():
         await gpu.shutdown()
 
     asyncio.run(run_())
+
+
+@pytest.mark.gpu
+def test_pass1_graph_matches_eager():
+    """The hipGraph pass-1 fast path (gpu/graphs.py) must produce
+    byte-identical responses to the eager launch sequence on the same
+    payloads (echo/convert only — no wall-clock tools)."""
+    import torch
+
+    if not torch.cuda.is_available():
+        pytest.skip("no ROCm device")
+
+    async def go():
+        from mcp_context_forge_amd.config import Settings
+        from mcp_context_forge_amd.engine import GatewayEngine
+        from mcp_context_forge_amd.services.upstream import make_fake_time_upstream
+
+        e = GatewayEngine(Settings(database_url="sqlite://", federation_enabled=False,
+                                   auth_required=False, gpu_enabled=True))
+        await e.gateway_service.register_gateway(name="t", url="inproc://t",
+                                                 client=make_fake_time_upstream())
+        assert e.enable_gpu()
+        raws = []
+        for i in range(300):
+            if i % 3 == 0:
+                args = '{"time":"2026-01-02T03:04:05Z","source_timezone":"UTC","target_timezone":"UTC"}'
+                raws.append((f'{{"jsonrpc":"2.0","id":{i},"method":"tools/call","params":'
+                             f'{{"name":"t-convert_time","arguments":{args}}}}}').encode())
+            else:
+                raws.append((f'{{"jsonrpc":"2.0","id":{i},"method":"tools/call","params":'
+                             f'{{"name":"t-echo","arguments":{{"m":"row {i} a@b.co","n":{i}}}}}}}').encode())
+        out_graph = await e.process_rpc_batch(list(raws))
+        pipe = e.gpu_pipeline
+        assert pipe._graphs is not None and pipe._graphs.replays >= 1, \
+            "graph path did not engage"
+        pipe._graphs.disabled = True
+        out_eager = await e.process_rpc_batch(list(raws))
+        assert out_graph == out_eager
+        await e.shutdown()
+
+    asyncio.run(go())
