@@ -54,6 +54,17 @@ int64_t forge_decide(
 int64_t forge_upstream_call_batch(const uint8_t*, const int32_t*, const int32_t*,
                                   const int32_t*, int, const char*,
                                   uint8_t*, int64_t, int64_t*, int64_t*);
+int64_t forge_rewrite_rows(const uint8_t*, const int32_t*, const int32_t*, int,
+                           const uint8_t*, const uint32_t*, uint32_t, int, int, int,
+                           const uint8_t*, const int32_t*, int, int,
+                           int32_t*, uint32_t*, int32_t*, uint8_t*, int64_t,
+                           int64_t*, int64_t*, int64_t*, int64_t*,
+                           const uint8_t*, const int32_t*, int, int32_t*);
+int64_t forge_post_rows(const uint8_t*, const int64_t*, const int64_t*, int,
+                        const uint8_t*, uint32_t, int,
+                        const uint8_t*, const int32_t*, int, int64_t, double,
+                        int32_t*, uint32_t*, int32_t*, uint8_t*,
+                        uint8_t*, int64_t, int64_t*, int64_t*);
 int64_t forge_finalize(
     const uint8_t*, const int32_t*, const int32_t*, const int32_t*, const int32_t*,
     const int32_t*, const uint64_t*, int, const int32_t*, int,
@@ -193,6 +204,62 @@ int run_pipeline_once(void* toolmap, void* store, void* cache, int n, unsigned s
 
 }  // namespace
 
+// the threaded rewrite + post lanes (rewrite.cpp): internal 4/8-way row
+// split with per-thread arena stitching — run them ALSO from several
+// outer threads to model two batches in flight
+static int run_rewrite_lanes_once(int n, unsigned seed) {
+    std::string blob;
+    std::vector<int32_t> beg((size_t)n), end_((size_t)n);
+    for (int i = 0; i < n; ++i) {
+        beg[(size_t)i] = (int32_t)blob.size();
+        char row[256];
+        snprintf(row, sizeof(row),
+                 "{\"msg\":\"row %u mail a%u@ex%u.co and 123-45-6789\",\"n\":%d}",
+                 seed + (unsigned)i, seed % 97, (unsigned)i % 89, i);
+        blob += row;
+        end_[(size_t)i] = (int32_t)blob.size();
+    }
+    std::vector<uint8_t> fl((size_t)n, 3 | 4 | 8);
+    std::vector<uint32_t> want((size_t)n, 0x3Fu);
+    const char* deny = "forbiddenblocked";
+    int32_t deny_off[3] = {0, 9, 16};
+    std::vector<int32_t> st((size_t)n), dh((size_t)n), hh((size_t)n);
+    std::vector<uint32_t> fb((size_t)n);
+    std::vector<int64_t> ob((size_t)n), oe((size_t)n), sb((size_t)n), se((size_t)n);
+    int64_t cap = (int64_t)blob.size() * 3 + n * 64 + 4096;
+    std::vector<uint8_t> arena((size_t)cap);
+    int64_t rc = forge_rewrite_rows((const uint8_t*)blob.data(), beg.data(), end_.data(), n,
+                                    fl.data(), want.data(), 0x3Fu, 0, 1, 1,
+                                    (const uint8_t*)deny, deny_off, 2, 1,
+                                    st.data(), fb.data(), dh.data(), arena.data(), cap,
+                                    ob.data(), oe.data(), sb.data(), se.data(),
+                                    (const uint8_t*)deny, deny_off, 2, hh.data());
+    if (rc < 0) return 1;
+    // feed the rewritten rows (as results) through the post lane
+    std::string rblob;
+    std::vector<int64_t> rb2((size_t)n), re2((size_t)n);
+    for (int i = 0; i < n; ++i) {
+        rb2[(size_t)i] = (int64_t)rblob.size();
+        if (st[(size_t)i] == 0)
+            rblob.append((const char*)arena.data() + ob[(size_t)i],
+                         (size_t)(oe[(size_t)i] - ob[(size_t)i]));
+        re2[(size_t)i] = (int64_t)rblob.size();
+    }
+    std::vector<uint8_t> pfl((size_t)n, 7);
+    std::vector<int32_t> pst((size_t)n), phh((size_t)n);
+    std::vector<uint32_t> pfb((size_t)n);
+    std::vector<uint8_t> pie((size_t)n);
+    std::vector<int64_t> pob((size_t)n), poe((size_t)n);
+    int64_t pcap = (int64_t)rblob.size() * 3 + n * 128 + 4096;
+    std::vector<uint8_t> parena((size_t)pcap);
+    int64_t prc = forge_post_rows((const uint8_t*)rblob.data(), rb2.data(), re2.data(), n,
+                                  pfl.data(), 0x3Fu, 0,
+                                  (const uint8_t*)deny, deny_off, 2, 64, 0.01,
+                                  pst.data(), pfb.data(), phh.data(), pie.data(),
+                                  parena.data(), pcap, pob.data(), poe.data());
+    return prc < 0 ? 1 : 0;
+}
+
 int main(int argc, char** argv) {
     int iters = argc > 1 ? atoi(argv[1]) : 6;
     int nthreads = argc > 2 ? atoi(argv[2]) : 4;
@@ -212,7 +279,8 @@ int main(int argc, char** argv) {
         std::vector<int> rc((size_t)nthreads, 0);
         for (int t = 0; t < nthreads; ++t)
             ts.emplace_back([&, t] { rc[(size_t)t] = run_pipeline_once(toolmap, store, cache, n,
-                                                                       (unsigned)(it * 131 + t)); });
+                                                                       (unsigned)(it * 131 + t))
+                                                    + run_rewrite_lanes_once(n, (unsigned)(it * 17 + t)); });
         for (auto& th : ts) th.join();
         for (int r : rc) fails += r;
     }
