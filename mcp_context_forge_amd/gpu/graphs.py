@@ -36,13 +36,15 @@ BYTES_PER_ROW = 384  # pinned blob capacity per bucket row (HTTP payloads ~150 B
 
 
 class _Bucket:
-    def __init__(self, pipeline, rows: int):
+    def __init__(self, pipeline, rows: int, bankset, with_classifier: bool,
+                 bytes_per_row: int):
         self.rows = rows
-        self.byte_cap = rows * BYTES_PER_ROW
+        self.byte_cap = rows * bytes_per_row
+        self.bankset = bankset
         dev = pipeline.device
-        nb = pipeline._bankset1.n
+        nb = bankset.n
         dim = pipeline.feat_dim
-        clf = pipeline.classifier
+        clf = pipeline.classifier if with_classifier else None
 
         self.pin_blob = torch.empty(self.byte_cap, dtype=torch.uint8, pin_memory=True)
         self.pin_beg = torch.empty(rows, dtype=torch.int32, pin_memory=True)
@@ -83,7 +85,7 @@ class _Bucket:
         self.d_blob.copy_(self.pin_blob, non_blocking=True)
         self.d_beg.copy_(self.pin_beg, non_blocking=True)
         self.d_end.copy_(self.pin_end, non_blocking=True)
-        hip.scan_multi(self.d_blob, self.d_beg, self.d_end, p._bankset1, out=self.out_multi)
+        hip.scan_multi(self.d_blob, self.d_beg, self.d_end, self.bankset, out=self.out_multi)
         if self.feats is not None:
             hip.featurize(self.d_blob, self.d_beg, self.d_end, p.feat_dim, out=self.feats)
             p.classifier.forward_into(self.feats, self.h, self.scores)
@@ -114,11 +116,15 @@ class _Bucket:
         return np.array(self.pin_scores.numpy()[:m])
 
 
-class Pass1Graphs:
-    """Lazy per-bucket hipGraph cache for one pipeline generation."""
+class GraphCache:
+    """Lazy per-bucket hipGraph cache for one (bank set, shape) family."""
 
-    def __init__(self, pipeline):
+    def __init__(self, pipeline, bankset, with_classifier: bool,
+                 bytes_per_row: int = BYTES_PER_ROW):
         self._pipeline = pipeline
+        self._bankset = bankset
+        self._with_classifier = with_classifier
+        self._bytes_per_row = bytes_per_row
         self._buckets: Dict[int, _Bucket] = {}
         self.disabled = os.environ.get("FORGE_PASS1_GRAPH", "1") == "0"
         self.replays = 0
@@ -129,17 +135,29 @@ class Pass1Graphs:
             return None
         for rows in ROW_BUCKETS:
             if m <= rows:
-                if nbytes > rows * BYTES_PER_ROW:
+                if nbytes > rows * self._bytes_per_row:
                     continue  # oversized payloads → next bucket or eager
                 b = self._buckets.get(rows)
                 if b is None:
                     try:
-                        b = _Bucket(self._pipeline, rows)
+                        b = _Bucket(self._pipeline, rows, self._bankset,
+                                    self._with_classifier, self._bytes_per_row)
                         self.captures += 1
                     except Exception:
-                        logger.exception("pass-1 graph capture failed; eager fallback")
+                        logger.exception("pass graph capture failed; eager fallback")
                         self.disabled = True
                         return None
                     self._buckets[rows] = b
                 return b
         return None
+
+
+def Pass1Graphs(pipeline):
+    return GraphCache(pipeline, pipeline._bankset1,
+                      with_classifier=pipeline.classifier is not None)
+
+
+def Pass3Graphs(pipeline):
+    # results are larger than requests (content + structuredContent echo)
+    return GraphCache(pipeline, pipeline._bankset3, with_classifier=False,
+                      bytes_per_row=640)

@@ -238,7 +238,8 @@ class GpuPluginPipeline:
             [(n, self.banks[n]) for n in ("pii", "regex", "harm", "postmeta") if n in self.banks],
             device)
         self._bankset1: Optional[hip.ScanBankSet] = None
-        self._graphs = None  # Pass1Graphs; rebuilt when _bankset1 changes
+        self._graphs = None   # pass-1 GraphCache; rebuilt when _bankset1 changes
+        self._graphs3 = None  # pass-3 GraphCache (static _bankset3)
 
         # native decision-plane stores + string tables (fastpath.cpp)
         self._slot_store = hip.store_new(self.semcache.capacity) if self.semcache is not None else 0
@@ -777,6 +778,7 @@ class GpuPluginPipeline:
         # --- GPU pass 1 over raw argument spans ---
         t_g = self._tic()
         await self._gpu_lock.acquire()
+        self._toc("gp1_lockwait", t_g)
         # hipGraph fast path (gpu/graphs.py): the whole pass — H2D stages,
         # fused scan, featurize, classifier, D2H into pinned — is ONE
         # hipGraphLaunch when the batch fits a bucket and the semcache is
@@ -793,7 +795,9 @@ class GpuPluginPipeline:
         cache_val_t = cache_idx_t = None
         feats_sk = None
         if gbucket is not None:
+            t_st = self._tic()
             gbucket.stage(blob, args_b, args_e, m)
+            self._toc("gp1_stage", t_st)
             self._toc("gp1_upload", t_g)
             t_l = self._tic()
             gbucket.replay()
@@ -1464,16 +1468,31 @@ class GpuPluginPipeline:
         regex3 = np.zeros(n_all, dtype=bool)
         if n_all:
             await self._gpu_lock.acquire()
-            self._pin_reset()
-            data3 = self._upload(res_blob)
-            b3 = self._upload(res_beg.astype(np.int32))
-            e3 = self._upload(res_end.astype(np.int32))
-            out3 = hip.scan_multi(data3, b3, e3, self._bankset3)  # fused, one launch
-            try:
-                await self._await_gpu()
-            finally:
-                self._gpu_lock.release()
-            mm3 = out3.cpu().numpy().view(np.uint32)
+            g3 = None
+            if self._graphs3 is None:
+                from .graphs import Pass3Graphs
+                self._graphs3 = Pass3Graphs(self)
+            g3 = self._graphs3.get(n_all, int(res_blob.nbytes))
+            if g3 is not None:
+                g3.stage(res_blob, res_beg.astype(np.int32), res_end.astype(np.int32), n_all)
+                g3.replay()
+                self._graphs3.replays += 1
+                try:
+                    await self._await_gpu()
+                    mm3 = g3.read_multi(n_all)
+                finally:
+                    self._gpu_lock.release()
+            else:
+                self._pin_reset()
+                data3 = self._upload(res_blob)
+                b3 = self._upload(res_beg.astype(np.int32))
+                e3 = self._upload(res_end.astype(np.int32))
+                out3 = hip.scan_multi(data3, b3, e3, self._bankset3)  # fused, one launch
+                try:
+                    await self._await_gpu()
+                finally:
+                    self._gpu_lock.release()
+                mm3 = out3.cpu().numpy().view(np.uint32)
             for b, i in self._bankset3.index.items():
                 h = mm3[i] != 0
                 if b == "postmeta":
