@@ -127,3 +127,53 @@ def test_schema_depth_guard():
     pool.Add(fdp)
     schema = _schema_from_descriptor(pool.FindMessageTypeByName("rec.Node"))
     assert schema["type"] == "object"  # terminated
+
+
+def test_invoke_error_status_propagates(grpc_server, run):
+    """A server-side gRPC error surfaces as a tool error through the engine
+    (not a crash): INVALID_ARGUMENT from the handler → isError response."""
+    import asyncio
+
+    from mcp_context_forge_amd.config import Settings
+    from mcp_context_forge_amd.engine import GatewayEngine
+
+    async def go():
+        e = GatewayEngine(Settings(database_url="sqlite://", federation_enabled=False,
+                                   auth_required=False))
+        tr = GrpcToMcpTranslator(grpc_server, prefix="g")
+        tr.register_into(e.tool_service)
+        await e.startup()
+        # count must be an int32; a string that can't coerce raises in
+        # ParseDict client-side → JSON-RPC error, engine stays alive
+        raw = json.dumps({"jsonrpc": "2.0", "id": 1, "method": "tools/call",
+                          "params": {"name": "g-test-v1-Greeter-SayHello",
+                                     "arguments": {"name": "x", "count": "NaN"}}}).encode()
+        out = json.loads((await e.process_rpc_batch([raw]))[0])
+        assert "error" in out or out.get("result", {}).get("isError"), out
+        # and a good call on the same engine still works
+        raw2 = json.dumps({"jsonrpc": "2.0", "id": 2, "method": "tools/call",
+                           "params": {"name": "g-test-v1-Greeter-SayHello",
+                                      "arguments": {"name": "ok", "count": 2}}}).encode()
+        out2 = json.loads((await e.process_rpc_batch([raw2]))[0])
+        assert "result" in out2, out2
+        sc = out2["result"]["structuredContent"]
+        assert sc["message"] == "Hello ok!" and sc["echoes"] == ["ok", "ok"]
+        await e.shutdown()
+
+    run(go())
+
+
+def test_unknown_fields_ignored_and_schema_types(grpc_server):
+    """ParseDict(ignore_unknown_fields=True) mirrors the reference's lax
+    argument handling; the synthesized schema types match the proto."""
+    tr = GrpcToMcpTranslator(grpc_server)
+    tools = tr.discover_tools()
+    assert len(tools) == 1
+    schema = tools[0]["inputSchema"]
+    assert schema["properties"]["name"] == {"type": "string"}
+    assert schema["properties"]["count"] == {"type": "integer"}
+    # unknown argument key is dropped, not an error
+    v = tr.endpoint.invoke("test.v1.Greeter", "SayHello",
+                           {"name": "n", "count": 1, "bogus_key": "zzz"})
+    assert v["message"] == "Hello n!"
+    tr.endpoint.close()

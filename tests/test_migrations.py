@@ -63,3 +63,89 @@ def test_migration_0002_adds_count_to_legacy_table(tmp_path):
     with eng2.connect() as c:
         cols = {r[1] for r in c.exec_driver_sql("PRAGMA table_info(tool_metrics)")}
     assert "count" in cols
+
+
+def test_upgrade_across_release_boundary_preserves_data(tmp_path):
+    """A deployed DB at revision 0003 (pre-lifecycle) upgrades through
+    0004-0007 with its gateway rows intact and the new columns usable."""
+    url = f"sqlite:///{tmp_path}/rel.db"
+    eng = build_engine(url)
+    run_migrations(eng)
+    with eng.begin() as c:
+        # roll back to "release with 0003": drop the 0004+ version rows and
+        # the columns/tables they added (sqlite: rebuild gateways narrow),
+        # then seed a legacy row in the OLD shape
+        for rev in ("0004%", "0005%", "0006%", "0007%"):
+            c.exec_driver_sql(f"DELETE FROM forge_schema_version WHERE revision LIKE '{rev}'")
+        c.exec_driver_sql("DROP TABLE gateways")
+        c.exec_driver_sql(
+            "CREATE TABLE gateways (id VARCHAR PRIMARY KEY, name VARCHAR, url VARCHAR, "
+            "transport VARCHAR, enabled BOOLEAN, reachable BOOLEAN)")
+        c.exec_driver_sql(
+            "INSERT INTO gateways VALUES ('g1', 'legacy', 'http://u', 'streamablehttp', 1, 1)")
+        c.exec_driver_sql("DROP TABLE IF EXISTS oauth_tokens")
+        c.exec_driver_sql("DROP TABLE IF EXISTS leader_leases")
+        c.exec_driver_sql("DROP TABLE IF EXISTS token_usage")
+    eng.dispose()
+
+    eng2 = build_engine(url)
+    applied = run_migrations(eng2)
+    assert applied == ["0004_gateway_lifecycle", "0005_oauth_tokens",
+                       "0006_leader_leases", "0007_token_usage"]
+    with eng2.connect() as c:
+        row = c.exec_driver_sql(
+            "SELECT name, retry_count FROM gateways WHERE id='g1'").first()
+        assert row[0] == "legacy"          # data preserved
+        # new columns/tables usable
+        c.exec_driver_sql("SELECT next_retry_at, failure_class FROM gateways")
+        for t in ("oauth_tokens", "leader_leases", "token_usage"):
+            c.exec_driver_sql(f"SELECT * FROM {t} LIMIT 1")
+
+
+def test_migration_0004_builds_gateways_on_legacy_db(tmp_path):
+    """A legacy DB that never had a gateways table at all (pre-federation
+    deployment) gets it built from the current model."""
+    url = f"sqlite:///{tmp_path}/nogw.db"
+    eng = build_engine(url)
+    run_migrations(eng)
+    with eng.begin() as c:
+        for rev in ("0004%", "0005%", "0006%", "0007%"):
+            c.exec_driver_sql(f"DELETE FROM forge_schema_version WHERE revision LIKE '{rev}'")
+        c.exec_driver_sql("DROP TABLE gateways")
+        c.exec_driver_sql("DROP TABLE IF EXISTS oauth_tokens")
+        c.exec_driver_sql("DROP TABLE IF EXISTS leader_leases")
+        c.exec_driver_sql("DROP TABLE IF EXISTS token_usage")
+    eng.dispose()
+    eng2 = build_engine(url)
+    assert "0004_gateway_lifecycle" in run_migrations(eng2)
+    with eng2.connect() as c:
+        cols = {r[1] for r in c.exec_driver_sql("PRAGMA table_info(gateways)")}
+    assert {"id", "name", "url", "retry_count", "failure_class"} <= cols
+
+
+def test_concurrent_migrations_single_winner(tmp_path):
+    """Two processes racing run_migrations on a shared DB: the chain is
+    applied exactly once (the retry path absorbs sqlite lock errors)."""
+    import threading
+
+    url = f"sqlite:///{tmp_path}/race.db"
+    results = {}
+
+    def runner(tag):
+        eng = build_engine(url)
+        try:
+            results[tag] = run_migrations(eng)
+        finally:
+            eng.dispose()
+
+    ts = [threading.Thread(target=runner, args=(i,)) for i in range(2)]
+    for t in ts:
+        t.start()
+    for t in ts:
+        t.join()
+    all_applied = sum(results.values(), [])
+    assert sorted(all_applied) == sorted(rev for rev, _ in MIGRATIONS), results
+    eng = build_engine(url)
+    with eng.connect() as c:
+        rows = [r[0] for r in c.exec_driver_sql("SELECT revision FROM forge_schema_version")]
+    assert sorted(rows) == sorted(rev for rev, _ in MIGRATIONS)
