@@ -273,3 +273,48 @@ def test_a2a_agent_credentials_sealed(run):
         await engine.shutdown()
 
     run(go())
+
+
+def test_chat_service_round_cap_and_tool_errors():
+    """A provider that ALWAYS tool-calls terminates at max_rounds (no
+    infinite loop); a failing tool lands in the transcript as ok=False and
+    its error text is fed back to the model."""
+    fake = FastAPI()
+
+    @fake.post("/v1/chat/completions")
+    async def completions(request: Request):
+        body = await request.json()
+        # always request another (broken) tool call
+        return {"id": "c", "model": "fake-model",
+                "choices": [{"index": 0, "message": {
+                    "role": "assistant", "content": None,
+                    "tool_calls": [{"id": "t", "type": "function",
+                                    "function": {"name": "boom",
+                                                 "arguments": "{not json"}}]},
+                    "finish_reason": "tool_calls"}]}
+
+    async def go():
+        port = _free_port()
+        server, task = await _serve(fake, port)
+        engine = GatewayEngine(Settings(database_url="sqlite://", federation_enabled=False,
+                                        auth_required=False))
+        try:
+            engine.llm_proxy.registry.register("fake", f"http://127.0.0.1:{port}/v1",
+                                               models=["fake-model"], default_model="fake-model")
+
+            async def boom(args):
+                raise RuntimeError("tool exploded")
+
+            engine.tool_service.register_local_tool("boom", boom, "always fails")
+            out = await engine.chat.chat([{"role": "user", "content": "go"}], max_rounds=3)
+            assert out["rounds"] == 3                       # capped, no hang
+            assert all(not tc["ok"] for tc in out["tool_calls"])
+            assert "tool error" in out["tool_calls"][0]["result"]
+            # malformed arguments JSON degraded to {} rather than raising
+            assert out["tool_calls"][0]["arguments"] == {}
+        finally:
+            await engine.shutdown()
+            server.should_exit = True
+            await asyncio.wait_for(task, timeout=10)
+
+    asyncio.run(go())
