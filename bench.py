@@ -298,6 +298,12 @@ async def run_engine_only(args, rank: int, world: int, use_gpu: bool, R: int):
         assert len(out) == len(raws)
     await sync(collectives, use_gpu)
 
+    prof = None
+    if os.environ.get("FORGE_CPROFILE"):  # timed-loop-only python profile
+        import cProfile
+
+        prof = cProfile.Profile()
+        prof.enable()
     step_times: List[float] = []
     t_start = time.monotonic()
     for s in range(args.steps):
@@ -306,6 +312,9 @@ async def run_engine_only(args, rank: int, world: int, use_gpu: bool, R: int):
         if use_gpu:
             torch.cuda.synchronize()
         step_times.append(time.monotonic() - t0)
+    if prof is not None:
+        prof.disable()
+        prof.dump_stats(os.environ["FORGE_CPROFILE"])
     await sync(collectives, use_gpu)
     elapsed = max_over_ranks(time.monotonic() - t_start, world, use_gpu)
 
