@@ -567,6 +567,52 @@ class GpuPluginPipeline:
         self._t_postlane = postlane
         self._t_rw_norm, self._t_rw_regex, self._t_rw_deny = rw_norm, rw_regex, rw_deny
         self._t_rw_pii, self._t_rw_harm, self._t_mod_app = rw_pii, rw_harm, mod_app
+        # FAST-mode schema descriptors for the C rewrite lane (presence +
+        # type checks only — richer schemas never compile to "fast"); a C
+        # pass means the exact validator is skipped, a C fail reruns it
+        # for the byte-exact error message
+        TYPE_CODE = {"string": 1, "integer": 2, "boolean": 3, "array": 4,
+                     "object": 5, "null": 6}
+        skb = bytearray()
+        sk_beg: List[int] = []
+        sk_end: List[int] = []
+        sk_ty: List[int] = []
+        sk_rq: List[int] = []
+        sk_range = np.full((max(nt, 1), 2), -1, dtype=np.int32)
+        for i, m in enumerate(metas):
+            if m.schema_mode != "fast":
+                continue
+            schema = m.tool.get("input_schema") or {}
+            props = schema.get("properties") or {}
+            required = set(schema.get("required") or [])
+            lo = len(sk_beg)
+            seen = set()
+            for k, sub in props.items():
+                ty = sub.get("type") if isinstance(sub, dict) else None
+                kb = k.encode()
+                sk_beg.append(len(skb))
+                skb += kb
+                sk_end.append(len(skb))
+                sk_ty.append(TYPE_CODE.get(ty, 0))
+                sk_rq.append(1 if k in required else 0)
+                seen.add(k)
+            for k in required:
+                if k not in seen:
+                    kb = k.encode()
+                    sk_beg.append(len(skb))
+                    skb += kb
+                    sk_end.append(len(skb))
+                    sk_ty.append(0)
+                    sk_rq.append(1)
+            sk_range[i] = (lo, len(sk_beg))
+        if sk_beg:
+            self._sk_tables = (
+                np.frombuffer(bytes(skb), dtype=np.uint8).copy() if skb else np.zeros(1, dtype=np.uint8),
+                np.asarray(sk_beg, dtype=np.int32), np.asarray(sk_end, dtype=np.int32),
+                np.asarray(sk_ty, dtype=np.int8), np.asarray(sk_rq, dtype=np.uint8))
+        else:
+            self._sk_tables = None
+        self._sk_range = sk_range
         # semcache insert allowlist (tool-level; lookups are gated by TF_CACHE)
         self._t_semallow = np.array(
             [bool(self.semcache is not None and self.semcache_plugin.cacheable(m.name)) for m in metas],
@@ -1105,6 +1151,8 @@ class GpuPluginPipeline:
         # user regexes, and provably-equivalent content (the C side punts
         # anything outside its envelope back here) ---
         py_js = rewrite_js
+        sg_en = self._enforcing(self.schema_guard)
+        schema_c: Dict[int, int] = {}
         t_sub = self._tic()
         if rewrite_js:
             nat_idx: List[int] = []
@@ -1124,6 +1172,8 @@ class GpuPluginPipeline:
                 do_pii = (pii_m is None or bool(pii_m[j])) and self._t_rw_pii[ti]
                 fl = (1 if do_norm else 0) | (2 if do_pii else 0) | (4 if deny_needed else 0) | \
                     (8 if harm_needed else 0)
+                if sg_en and self._sk_tables is not None and self._sk_range[ti, 0] >= 0:
+                    fl |= 16
                 bits = int(pii_m[j]) if (pii_m is not None and do_pii) else -1
                 if bits < 0:
                     want = self._pii_active_mask
@@ -1144,7 +1194,8 @@ class GpuPluginPipeline:
                 elif self.pii is not None and self.pii.action not in ("mask",):
                     pii_mode = 2
                 njs = np.asarray(nat_idx, dtype=np.int64)
-                st, found, deny_hit, harm_hit, rw_arena, rb, re_, sb, se = hip.rewrite_rows(
+                tis = tool_idx[njs]
+                st, found, deny_hit, harm_hit, schema_ok, rw_arena, rb, re_, sb, se = hip.rewrite_rows(
                     blob, np.ascontiguousarray(args_b[njs]), np.ascontiguousarray(args_e[njs]),
                     np.asarray(flags_l, dtype=np.uint8), np.asarray(want_l, dtype=np.uint32),
                     self._pii_active_mask, pii_mode,
@@ -1152,7 +1203,10 @@ class GpuPluginPipeline:
                     bool(self.normalizer and self.normalizer.strip),
                     deny_blob=self._deny_lane_blob, deny_off=self._deny_lane_off,
                     deny_ci=bool(self.deny and self.deny.case_insensitive),
-                    harm_blob=self._harm_lane_blob, harm_off=self._harm_lane_off)
+                    harm_blob=self._harm_lane_blob, harm_off=self._harm_lane_off,
+                    sk_tables=self._sk_tables,
+                    sk_lo=np.ascontiguousarray(self._sk_range[tis, 0]) if self._sk_tables is not None else None,
+                    sk_hi=np.ascontiguousarray(self._sk_range[tis, 1]) if self._sk_tables is not None else None)
                 punted: List[int] = []
                 for k, j in enumerate(nat_idx):
                     if st[k] == hip.RW_DONE:
@@ -1160,6 +1214,8 @@ class GpuPluginPipeline:
                         scan_bytes[j] = rw_arena[sb[k]:se[k]].tobytes()
                         if harm_hit[k] >= 0:
                             harm_c[j] = int(harm_hit[k])
+                        if schema_ok[k] != 2:
+                            schema_c[j] = int(schema_ok[k])
                     elif st[k] == hip.RW_DENY:
                         r = int(rows[j])
                         idb = self._id_bytes(blob, env, r)
@@ -1246,12 +1302,12 @@ class GpuPluginPipeline:
         # returned closure, which the caller awaits AFTER the big native
         # upstream batch so the GPU rescan overlaps that C++ work
         async def finish() -> List[Tuple[int, Any]]:
-            return await self._finish_rewrite(ok_items, harm_c, scores2_t,
+            return await self._finish_rewrite(ok_items, harm_c, schema_c, scores2_t,
                                               blob, env, rows, tool_idx, hit, hit_slot, responses)
 
         return finish
 
-    async def _finish_rewrite(self, ok_items, harm_c, scores2_t,
+    async def _finish_rewrite(self, ok_items, harm_c, schema_c, scores2_t,
                               blob, env, rows, tool_idx, hit, hit_slot,
                               responses: List[Optional[bytes]]) -> List[Tuple[int, Any]]:
         """Phase B of the rewrite pass: await the rescan verdicts of the
@@ -1298,8 +1354,10 @@ class GpuPluginPipeline:
                     f"harmful_content_detector: harmful content ({cat})")
                 self.blocked += 1
                 continue
-            mt = self._meta_list[ti]
-            if sg_on and mt.schema_mode != "trivial":
+            if sg_on and schema_c.get(j, 2) == 1:
+                pass  # C fast-schema check passed on the rewritten tree
+            elif sg_on and self._meta_list[ti].schema_mode != "trivial":
+                mt = self._meta_list[ti]
                 if isinstance(args2, bytes):
                     args2 = json.loads(args2)
                 errs = _validate(args2 or {}, mt.tool.get("input_schema") or {})

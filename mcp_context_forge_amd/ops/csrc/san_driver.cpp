@@ -59,7 +59,10 @@ int64_t forge_rewrite_rows(const uint8_t*, const int32_t*, const int32_t*, int,
                            const uint8_t*, const int32_t*, int, int,
                            int32_t*, uint32_t*, int32_t*, uint8_t*, int64_t,
                            int64_t*, int64_t*, int64_t*, int64_t*,
-                           const uint8_t*, const int32_t*, int, int32_t*);
+                           const uint8_t*, const int32_t*, int, int32_t*,
+                           const uint8_t*, const int32_t*, const int32_t*,
+                           const int8_t*, const uint8_t*,
+                           const int32_t*, const int32_t*, uint8_t*);
 int64_t forge_post_rows(const uint8_t*, const int64_t*, const int64_t*, int,
                         const uint8_t*, uint32_t, int,
                         const uint8_t*, const int32_t*, int, int64_t, double,
@@ -233,7 +236,9 @@ static int run_rewrite_lanes_once(int n, unsigned seed) {
                                     (const uint8_t*)deny, deny_off, 2, 1,
                                     st.data(), fb.data(), dh.data(), arena.data(), cap,
                                     ob.data(), oe.data(), sb.data(), se.data(),
-                                    (const uint8_t*)deny, deny_off, 2, hh.data());
+                                    (const uint8_t*)deny, deny_off, 2, hh.data(),
+                                    nullptr, nullptr, nullptr, nullptr, nullptr,
+                                    nullptr, nullptr, nullptr);
     if (rc < 0) return 1;
     // feed the rewritten rows (as results) through the post lane
     std::string rblob;

@@ -49,7 +49,8 @@ def _load() -> ctypes.CDLL:
                               [ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int, ctypes.c_int] +
                               [ctypes.c_void_p] * 3 +
                               [ctypes.c_void_p, ctypes.c_int64] + [ctypes.c_void_p] * 4 +
-                              [ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int, ctypes.c_void_p],
+                              [ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int, ctypes.c_void_p] +
+                              [ctypes.c_void_p] * 8,
         "forge_post_rows": [ctypes.c_void_p] * 3 + [ctypes.c_int, ctypes.c_void_p] +
                            [ctypes.c_uint32, ctypes.c_int] +
                            [ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int] +
@@ -481,11 +482,15 @@ def rewrite_rows(blob: np.ndarray, args_beg: np.ndarray, args_end: np.ndarray,
                  deny_off: Optional[np.ndarray] = None,
                  deny_ci: bool = True,
                  harm_blob: Optional[np.ndarray] = None,
-                 harm_off: Optional[np.ndarray] = None):
+                 harm_off: Optional[np.ndarray] = None,
+                 sk_tables: Optional[tuple] = None,
+                 sk_lo: Optional[np.ndarray] = None,
+                 sk_hi: Optional[np.ndarray] = None):
     """Native rewrite pass (rewrite.cpp): normalizer + PII over the flagged
     rows that fit the provable-equivalence envelope; everything else gets
     RW_PUNT and takes the Python path.
     → (status i32[n], found u32[n], deny_hit i32[n], harm_hit i32[n],
+       schema_ok u8[n] (1 pass / 0 fail / 2 unchecked),
        arena u8, out_beg/out_end i64[n] (dispatch form, wire key order),
        scan_beg/scan_end i64[n] (sorted-keys scan form))."""
     n = args_beg.shape[0]
@@ -498,8 +503,12 @@ def rewrite_rows(blob: np.ndarray, args_beg: np.ndarray, args_end: np.ndarray,
     scan_end = np.empty(n, dtype=np.int64)
     deny_hit = np.empty(n, dtype=np.int32)
     harm_hit = np.empty(n, dtype=np.int32)
+    schema_ok = np.full(n, 2, dtype=np.uint8)
     n_deny = (deny_off.shape[0] - 1) if deny_off is not None else 0
     n_harm = (harm_off.shape[0] - 1) if harm_off is not None else 0
+    sk_blob = sk_beg = sk_end = sk_type = sk_req = None
+    if sk_tables is not None:
+        sk_blob, sk_beg, sk_end, sk_type, sk_req = sk_tables
     cap = int((args_end - args_beg).sum()) * 2 + n * 32 + 4096
     while True:
         arena = np.empty(cap, dtype=np.uint8)
@@ -512,9 +521,12 @@ def rewrite_rows(blob: np.ndarray, args_beg: np.ndarray, args_end: np.ndarray,
             _np_ptr(status), _np_ptr(found), _np_ptr(deny_hit),
             _np_ptr(arena), cap, _np_ptr(out_beg), _np_ptr(out_end),
             _np_ptr(scan_beg), _np_ptr(scan_end),
-            _np_ptr(harm_blob), _np_ptr(harm_off), n_harm, _np_ptr(harm_hit))
+            _np_ptr(harm_blob), _np_ptr(harm_off), n_harm, _np_ptr(harm_hit),
+            _np_ptr(sk_blob), _np_ptr(sk_beg), _np_ptr(sk_end),
+            _np_ptr(sk_type), _np_ptr(sk_req),
+            _np_ptr(sk_lo), _np_ptr(sk_hi), _np_ptr(schema_ok))
         if rc >= 0:
-            return status, found, deny_hit, harm_hit, arena, out_beg, out_end, scan_beg, scan_end
+            return status, found, deny_hit, harm_hit, schema_ok, arena, out_beg, out_end, scan_beg, scan_end
         cap = -int(rc) + 4096
 
 
