@@ -304,11 +304,17 @@ async def run_engine_only(args, rank: int, world: int, use_gpu: bool, R: int):
 
         prof = cProfile.Profile()
         prof.enable()
+    conc = int(os.environ.get("FORGE_ENGINE_CONC", "1"))
     step_times: List[float] = []
     t_start = time.monotonic()
     for s in range(args.steps):
         t0 = time.monotonic()
-        await engine.process_rpc_batch(step_data[s])
+        if conc > 1:
+            k = (len(step_data[s]) + conc - 1) // conc
+            parts = [step_data[s][i:i + k] for i in range(0, len(step_data[s]), k)]
+            await asyncio.gather(*(engine.process_rpc_batch(p) for p in parts))
+        else:
+            await engine.process_rpc_batch(step_data[s])
         if use_gpu:
             torch.cuda.synchronize()
         step_times.append(time.monotonic() - t0)
