@@ -19,7 +19,7 @@ ARCH = os.environ.get("PYTORCH_ROCM_ARCH", "gfx950")
 HIPCC = os.environ.get("HIPCC", "/opt/rocm/bin/hipcc")
 
 SOURCES = ["scan.hip", "featurize.hip", "json_guard.hip", "gemm_bf16.hip", "runtime.hip",
-           "gemm_v2.hip", "envelope.cpp", "upstream.cpp", "fastpath.cpp", "rewrite.cpp"]
+           "gemm_v2.hip", "envelope.cpp", "upstream.cpp", "fastpath.cpp", "rewrite.cpp", "pool.cpp"]
 
 
 PYBRIDGE = Path(__file__).parent / "forge_pybridge.so"

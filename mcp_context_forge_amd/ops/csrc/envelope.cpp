@@ -14,7 +14,17 @@
 #include <stdint.h>
 #include <string.h>
 #include <thread>
+#include <functional>
 #include <vector>
+
+extern "C" void forge_parallel_for(int n, void (*fn)(int, void*), void* ctx);
+namespace {
+inline void run_parallel(int n, const std::function<void(int)>& f) {
+    forge_parallel_for(
+        n, [](int i, void* c) { (*static_cast<const std::function<void(int)>*>(c))(i); },
+        (void*)&f);
+}
+}  // namespace
 
 namespace {
 
@@ -245,13 +255,11 @@ extern "C" int forge_parse_envelopes(
         return 0;
     }
     int chunk = (n + nthreads - 1) / nthreads;
-    std::vector<std::thread> threads;
-    for (int t = 0; t < nthreads; ++t) {
+    run_parallel(nthreads, [&](int t) {
         int r0 = t * chunk, r1 = r0 + chunk < n ? r0 + chunk : n;
-        if (r0 >= r1) break;
-        threads.emplace_back(parse_rows, data, offsets, r0, r1,
-                             kind, id_beg, id_end, name_beg, name_end, args_beg, args_end);
-    }
-    for (auto& th : threads) th.join();
+        if (r0 >= r1) return;
+        parse_rows(data, offsets, r0, r1,
+                   kind, id_beg, id_end, name_beg, name_end, args_beg, args_end);
+    });
     return 0;
 }

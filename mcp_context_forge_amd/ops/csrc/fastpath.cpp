@@ -24,11 +24,21 @@
 #include <stdint.h>
 #include <string.h>
 #include <stdio.h>
+#include <functional>
 #include <mutex>
 #include <string>
 #include <thread>
 #include <unordered_map>
 #include <vector>
+
+extern "C" void forge_parallel_for(int n, void (*fn)(int, void*), void* ctx);
+namespace {
+inline void run_parallel(int n, const std::function<void(int)>& f) {
+    forge_parallel_for(
+        n, [](int i, void* c) { (*static_cast<const std::function<void(int)>*>(c))(i); },
+        (void*)&f);
+}
+}  // namespace
 
 namespace {
 
@@ -347,9 +357,7 @@ int64_t forge_decide(
     if (nthreads == 1) {
         run_rows(0);
     } else {
-        std::vector<std::thread> threads;
-        for (int t = 0; t < nthreads; ++t) threads.emplace_back(run_rows, t);
-        for (auto& th : threads) th.join();
+        run_parallel(nthreads, run_rows);
     }
     int64_t total = 0;
     std::vector<int64_t> base((size_t)nthreads, 0);
@@ -432,9 +440,7 @@ int64_t forge_finalize(
     if (nthreads == 1) {
         run_chunk(0);
     } else {
-        std::vector<std::thread> threads;
-        for (int t = 0; t < nthreads; ++t) threads.emplace_back(run_chunk, t);
-        for (auto& th : threads) th.join();
+        run_parallel(nthreads, run_chunk);
     }
     int64_t total = 0;
     std::vector<int64_t> base((size_t)nthreads, 0);
@@ -457,9 +463,7 @@ int64_t forge_finalize(
         bool any = false;
         for (auto& vec : inserts) if (!vec.empty()) { any = true; break; }
         if (any && nthreads > 1) {
-            std::vector<std::thread> threads;
-            for (int s = 0; s < ExactCache::SHARDS; ++s) threads.emplace_back(insert_shard, s);
-            for (auto& th : threads) th.join();
+            run_parallel(ExactCache::SHARDS, insert_shard);
         } else if (any) {
             for (int s = 0; s < ExactCache::SHARDS; ++s) insert_shard(s);
         }

@@ -29,9 +29,19 @@
 #include <string.h>
 
 #include <algorithm>
+#include <functional>
 #include <string>
 #include <thread>
 #include <vector>
+
+extern "C" void forge_parallel_for(int n, void (*fn)(int, void*), void* ctx);
+namespace {
+inline void run_parallel(int n, const std::function<void(int)>& f) {
+    forge_parallel_for(
+        n, [](int i, void* c) { (*static_cast<const std::function<void(int)>*>(c))(i); },
+        (void*)&f);
+}
+}  // namespace
 
 namespace {
 
@@ -934,9 +944,7 @@ extern "C" int64_t forge_rewrite_rows(
     if (nthreads == 1) {
         run(0);
     } else {
-        std::vector<std::thread> threads;
-        for (int t = 0; t < nthreads; ++t) threads.emplace_back(run, t);
-        for (auto& th : threads) th.join();
+        run_parallel(nthreads, run);
     }
     int64_t total = 0;
     std::vector<int64_t> base((size_t)nthreads, 0);
@@ -1156,9 +1164,7 @@ extern "C" int64_t forge_post_rows(
     if (nthreads == 1) {
         run(0);
     } else {
-        std::vector<std::thread> threads;
-        for (int t = 0; t < nthreads; ++t) threads.emplace_back(run, t);
-        for (auto& th : threads) th.join();
+        run_parallel(nthreads, run);
     }
     int64_t total = 0;
     std::vector<int64_t> base((size_t)nthreads, 0);

@@ -13,9 +13,19 @@
 #include <stdint.h>
 #include <stdlib.h>
 #include <string.h>
+#include <functional>
 #include <string>
 #include <thread>
 #include <vector>
+
+extern "C" void forge_parallel_for(int n, void (*fn)(int, void*), void* ctx);
+namespace {
+inline void run_parallel(int n, const std::function<void(int)>& f) {
+    forge_parallel_for(
+        n, [](int i, void* c) { (*static_cast<const std::function<void(int)>*>(c))(i); },
+        (void*)&f);
+}
+}  // namespace
 
 namespace {
 
@@ -215,17 +225,13 @@ extern "C" int64_t forge_upstream_call_batch(
         return (int64_t)buf.size();
     }
     std::vector<std::string> bufs(nthreads);
-    std::vector<std::thread> threads;
     int chunk = (n + nthreads - 1) / nthreads;
-    for (int t = 0; t < nthreads; ++t) {
+    run_parallel(nthreads, [&](int t) {
         int r0 = t * chunk, r1 = r0 + chunk < n ? r0 + chunk : n;
-        if (r0 >= r1) break;
-        threads.emplace_back([&, t, r0, r1] {
-            upstream_rows(data, args_beg, args_end, kinds, r0, r1, now_iso,
-                          bufs[t], res_beg, res_end);
-        });
-    }
-    for (auto& th : threads) th.join();
+        if (r0 >= r1) return;
+        upstream_rows(data, args_beg, args_end, kinds, r0, r1, now_iso,
+                      bufs[(size_t)t], res_beg, res_end);
+    });
     int64_t total = 0;
     std::vector<int64_t> base(nthreads, 0);
     for (int t = 0; t < nthreads; ++t) { base[t] = total; total += (int64_t)bufs[t].size(); }

@@ -16,7 +16,7 @@ import pytest
 
 CSRC = Path("/root/repo/mcp_context_forge_amd/ops/csrc")
 BUILD = Path("/root/repo/mcp_context_forge_amd/ops")
-SOURCES = ["envelope.cpp", "upstream.cpp", "fastpath.cpp", "rewrite.cpp", "san_driver.cpp"]
+SOURCES = ["envelope.cpp", "upstream.cpp", "fastpath.cpp", "rewrite.cpp", "pool.cpp", "san_driver.cpp"]
 
 
 def _build(flavor: str) -> Path:
