@@ -132,7 +132,10 @@ class Settings(BaseModel):
     # micro-batch linger: with two batches in flight the pipeline fills
     # itself; measured at the 1000-user knee: 100 µs beats 500 µs by ~12%
     # RPS and ~0.5 ms p50 (profiles/README_r02.md linger sweep)
-    gpu_batch_window_us: int = 100
+    # 30 µs measured optimal at the 1000-conn knee after the round-2 host
+    # optimizations (the cycle shortened ~2.3×, so less lingering pays;
+    # sweeps: 100 µs was best pre-pool, 30 µs wins 248k→261k RPS after)
+    gpu_batch_window_us: int = 30
     gpu_feature_dim: int = 4096       # hashed count-vector dim for classifiers/semantic cache
     gpu_classifier_hidden: int = 1024
     gpu_classifier_classes: int = 8
