@@ -465,6 +465,8 @@ class GpuPluginPipeline:
         self._t_outschema = np.array([m.has_output_schema for m in metas], dtype=bool) if nt else np.zeros(1, dtype=bool)
         self._t_thash = np.array([m.thash for m in metas], dtype=np.int64) if nt else np.zeros(1, dtype=np.int64)
         self._t_index = {m.name: i for i, m in enumerate(metas)}
+        self._t_name_list = [m.name for m in metas]
+        self._t_tid_list = [m.tid for m in metas]
 
         # per-tool plugin bindings: mode flips for the bank plugins map onto
         # the flag bits; config overrides or bindings on non-bank plugins
@@ -1711,18 +1713,22 @@ class GpuPluginPipeline:
         if n_all:
             ms = (time.monotonic() - t0) * 1000.0
             tis = tool_idx[all_js_np]
-            uniq, inv = np.unique(tis, return_inverse=True)
-            counts = np.bincount(inv, minlength=uniq.size)
-            errs = np.bincount(inv, weights=is_err.astype(np.float64), minlength=uniq.size).astype(np.int64)
-            names = [self._meta_list[int(ti)].name for ti in uniq]
+            nt = len(self._meta_list)
+            counts_all = np.bincount(tis, minlength=nt)
+            errs_all = np.bincount(tis, weights=is_err.astype(np.float64),
+                                   minlength=nt).astype(np.int64)
+            uniq = np.nonzero(counts_all)[0]
+            counts = counts_all[uniq]
+            errs = errs_all[uniq]
             if self.breaker is not None:
                 # breaker bookkeeping only where it can change state: tools with
                 # errors this batch, or tools that already have a window/state
-                for k, name in enumerate(names):
+                for k, ti in enumerate(uniq):
+                    name = self._t_name_list[int(ti)]
                     if errs[k] or name in self.breaker.state:
                         self._breaker_record_bulk(name, int(counts[k]), int(errs[k]))
             self.engine.metrics.record_aggregate_many(
-                [self._meta_list[int(ti)].tid for ti in uniq], counts, errs, ms)
+                [self._t_tid_list[int(ti)] for ti in uniq], counts, errs, ms)
         self._toc("fin_metrics", t_fm)
         self._toc("finalize", t_f)
 
