@@ -242,3 +242,68 @@ def test_edge_with_gpu_pipeline_under_load(run):
             assert st["requests"] >= 4000  # batches actually rode the GPU path
 
     run(go())
+
+
+def test_edge_adversarial_wire_forms(edge_app, run):
+    """Raw-socket adversarial input against the C++ parser: split headers
+    across packets, two pipelined requests in one send, and garbage bytes
+    — the edge must answer correctly or close cleanly, never hang."""
+    running, engine, app = edge_app
+
+    async def go():
+        async with running() as base:
+            port = int(base.rsplit(":", 1)[1])
+            body = json.dumps(_rpc(1)).encode()
+
+            def talk(chunks, expect_json=True, timeout=5.0):
+                with socket.create_connection(("127.0.0.1", port), timeout=timeout) as c:
+                    c.settimeout(timeout)
+                    for ch in chunks:
+                        c.sendall(ch)
+                    data = b""
+                    try:
+                        while b"\r\n\r\n" not in data or (expect_json and not data.endswith(b"}")):
+                            part = c.recv(65536)
+                            if not part:
+                                break
+                            data += part
+                    except socket.timeout:
+                        pass
+                    return data
+
+            req = (f"POST /rpc HTTP/1.1\r\nHost: x\r\nAuthorization: {BASIC}\r\n"
+                   f"Content-Type: application/json\r\nContent-Length: {len(body)}\r\n"
+                   f"\r\n").encode() + body
+
+            # (a) header split into tiny packets
+            out = await asyncio.to_thread(talk, [req[:17], req[17:60], req[60:]])
+            assert b"200" in out.split(b"\r\n", 1)[0] and b'"result"' in out, out[:200]
+
+            # (b) two pipelined requests in ONE send → two responses
+            def talk2():
+                with socket.create_connection(("127.0.0.1", port), timeout=5.0) as c:
+                    c.settimeout(5.0)
+                    c.sendall(req + req)
+                    data = b""
+                    try:
+                        while data.count(b'"result"') < 2:
+                            part = c.recv(65536)
+                            if not part:
+                                break
+                            data += part
+                    except socket.timeout:
+                        pass
+                    return data
+
+            out2 = await asyncio.to_thread(talk2)
+            assert out2.count(b'"result"') == 2, out2[:300]
+
+            # (c) garbage request line → error response or clean close, no hang
+            out3 = await asyncio.to_thread(talk, [b"\x00\xffGARBAGE\r\n\r\n"], False)
+            assert out3 == b"" or b"400" in out3 or b"HTTP/1.1" in out3
+
+            # (d) the edge still serves normal traffic afterwards
+            out4 = await asyncio.to_thread(talk, [req])
+            assert b'"result"' in out4
+
+    run(go())
