@@ -221,6 +221,8 @@ def test_semcache_hit_miss_ttl(hip):
 @requires_gpu
 @pytest.mark.parametrize("m,n,k", [(256, 256, 32), (256, 256, 64), (256, 256, 96),
                                    (512, 256, 4096), (256, 512, 128)])
+@pytest.mark.skipif(__import__("os").environ.get("FORGE_GEMM_V1") == "1",
+                    reason="v1 forced by env; these shapes are v2-only")
 def test_gemm_v2_matches_torch(hip, m, n, k):
     """256²-tile 4-phase ring GEMM (gemm_v2.hip) vs fp32 torch — asymmetric
     random inputs (guide G9: transpose-detecting refcheck)."""

@@ -228,10 +228,14 @@ def test_pass1_graph_matches_eager():
     """The hipGraph pass-1 fast path (gpu/graphs.py) must produce
     byte-identical responses to the eager launch sequence on the same
     payloads (echo/convert only — no wall-clock tools)."""
+    import os
+
     import torch
 
     if not torch.cuda.is_available():
         pytest.skip("no ROCm device")
+    if os.environ.get("FORGE_PASS1_GRAPH", "1") == "0":
+        pytest.skip("graphs disabled by env — nothing to compare")
 
     async def go():
         from mcp_context_forge_amd.config import Settings
