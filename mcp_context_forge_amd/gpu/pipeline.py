@@ -1210,14 +1210,26 @@ class GpuPluginPipeline:
                     sk_lo=np.ascontiguousarray(self._sk_range[tis, 0]) if self._sk_tables is not None else None,
                     sk_hi=np.ascontiguousarray(self._sk_range[tis, 1]) if self._sk_tables is not None else None)
                 punted: List[int] = []
-                for k, j in enumerate(nat_idx):
-                    if st[k] == hip.RW_DONE:
-                        ok_items.append((j, rw_arena[rb[k]:re_[k]].tobytes()))
-                        scan_bytes[j] = rw_arena[sb[k]:se[k]].tobytes()
+                done_m = st == hip.RW_DONE
+                done_ks = np.nonzero(done_m)[0]
+                if done_ks.size:
+                    # C-loop slicing for the common path: both byte forms in
+                    # two bridge calls instead of 2 python slices per row
+                    wires = self._pb.slices_list(rw_arena, np.ascontiguousarray(rb[done_ks]),
+                                                 np.ascontiguousarray(re_[done_ks]))
+                    scans = self._pb.slices_list(rw_arena, np.ascontiguousarray(sb[done_ks]),
+                                                 np.ascontiguousarray(se[done_ks]))
+                    for q, k in enumerate(done_ks):
+                        j = nat_idx[int(k)]
+                        ok_items.append((j, wires[q]))
+                        scan_bytes[j] = scans[q]
                         if harm_hit[k] >= 0:
                             harm_c[j] = int(harm_hit[k])
                         if schema_ok[k] != 2:
                             schema_c[j] = int(schema_ok[k])
+                for k, j in enumerate(nat_idx):
+                    if done_m[k]:
+                        pass
                     elif st[k] == hip.RW_DENY:
                         r = int(rows[j])
                         idb = self._id_bytes(blob, env, r)
