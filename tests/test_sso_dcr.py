@@ -154,3 +154,56 @@ def test_templated_provider_presets():
     with _pytest.raises(SSOError, match="needs tenant"):
         svc.register_provider("bad", "cid", "cs", preset="entra")
     db.close()
+
+
+def test_sso_redirect_dance_live_sockets():
+    """The full browser-style redirect dance over REAL sockets on both
+    sides: gateway (uvicorn) 302 → IdP /authorize 302 back with
+    code+state → gateway callback mints the JWT — one client following
+    redirects like a user agent, no hand-simulated hops."""
+
+    async def go():
+        from urllib.parse import parse_qs, urlparse
+
+        from fastapi.responses import RedirectResponse
+
+        idp = _fake_idp()
+
+        @idp.get("/authorize")
+        async def authorize(request: Request):
+            q = request.query_params
+            assert q["client_id"] == "gw-client" and q["response_type"] == "code"
+            return RedirectResponse(
+                f"{q['redirect_uri']}?code=good-code&state={q['state']}", status_code=302)
+
+        idp_port = _free_port()
+        gw_port = _free_port()
+        idp_server, idp_task = await _serve(idp, idp_port)
+        engine = GatewayEngine(Settings(database_url="sqlite://", federation_enabled=False,
+                                        auth_required=True))
+        app = build_app(engine)
+        gw_server, gw_task = await _serve(app, gw_port)
+        try:
+            async with httpx.AsyncClient() as c:
+                base = f"http://127.0.0.1:{gw_port}"
+                r = await c.post(f"{base}/auth/sso/providers", headers=ADMIN, json={
+                    "name": "corp", "client_id": "gw-client", "client_secret": "gw-secret",
+                    "preset": "oidc",
+                    "authorize_url": f"http://127.0.0.1:{idp_port}/authorize",
+                    "token_url": f"http://127.0.0.1:{idp_port}/token",
+                    "userinfo_url": f"http://127.0.0.1:{idp_port}/userinfo"})
+                assert r.status_code == 201
+                r = await c.get(f"{base}/auth/sso/corp/login", follow_redirects=True)
+                assert r.status_code == 200, r.text
+                body = r.json()
+                assert body["email"] == "sso-user@corp.com"
+                tok = body["access_token"]
+                r = await c.get(f"{base}/tools", headers={"Authorization": f"Bearer {tok}"})
+                assert r.status_code == 200
+        finally:
+            gw_server.should_exit = True
+            idp_server.should_exit = True
+            await asyncio.wait_for(gw_task, timeout=10)
+            await asyncio.wait_for(idp_task, timeout=10)
+
+    asyncio.run(go())
