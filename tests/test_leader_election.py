@@ -96,3 +96,43 @@ def test_mutation_kills_initial_and_release_state(tmp_path):
     a.release()
     assert a.is_leader is False           # release flips local state too
     db.close()
+
+
+def test_leader_lease_across_processes(tmp_path):
+    """Two real PROCESSES share one sqlite file: the subprocess holds the
+    lease, this process is denied, and after the holder dies + TTL expiry
+    this process takes over — the deployment shape (shared-DB workers)."""
+    import subprocess
+    import sys
+    import textwrap
+
+    db_path = tmp_path / "xproc.db"
+    script = textwrap.dedent(f"""
+        import sys, time
+        sys.path.insert(0, {repr(str(__import__('pathlib').Path(__file__).resolve().parents[1]))})
+        from mcp_context_forge_amd.db.engine import Database
+        from mcp_context_forge_amd.services.leader import DbLeaderElector
+        db = Database("sqlite:///{db_path}")
+        db.migrate()
+        el = DbLeaderElector(db, ttl_s=1.0, holder_id="SUBPROC")
+        assert el.try_acquire()
+        print("HELD", flush=True)
+        time.sleep(30)   # hold until killed
+    """)
+    proc = subprocess.Popen([sys.executable, "-c", script],
+                            stdout=subprocess.PIPE, text=True)
+    try:
+        line = proc.stdout.readline().strip()
+        assert line == "HELD", line
+        db2 = Database(f"sqlite:///{db_path}")
+        db2.migrate()
+        me = DbLeaderElector(db2, ttl_s=1.0, holder_id="MAIN")
+        assert me.try_acquire() is False, "other process holds the lease"
+    finally:
+        proc.kill()
+        proc.wait(timeout=10)
+    # holder crashed without releasing: lease expires after TTL
+    time.sleep(1.2)
+    assert me.try_acquire() is True
+    assert me.is_leader
+    db2.close()
