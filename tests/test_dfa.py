@@ -187,3 +187,47 @@ def test_compile_literals_case_insensitive_by_default():
     assert dfa.scan_reference(t, b"word") and dfa.scan_reference(t, b"WORD")
     t2 = dfa.compile_literals(["Word"], case_insensitive=False)
     assert not dfa.scan_reference(t2, b"word")
+
+
+def test_quantifier_required_vs_optional():
+    """Mutation survivors dfa.py ~:211 (`lo > 0` ladder): `+` requires at
+    least one atom, `*` requires none — both with unbounded tails."""
+    plus = dfa.compile_patterns([r"ab+c"])
+    assert dfa.match_mask_reference(plus, b"abc") == 1
+    assert dfa.match_mask_reference(plus, b"abbbbc") == 1
+    assert dfa.match_mask_reference(plus, b"ac") == 0, "b+ must require one b"
+    star = dfa.compile_patterns([r"ab*c"])
+    assert dfa.match_mask_reference(star, b"ac") == 1, "b* allows zero b"
+    assert dfa.match_mask_reference(star, b"abbc") == 1
+    assert dfa.match_mask_reference(star, b"axc") == 0
+
+
+def test_trailing_backslash_is_literal():
+    """dfa.py ~:127 (`i + 1 < len(pat)` escape bound): a pattern ENDING in
+    a single backslash treats it as a literal byte, not an escape."""
+    t = dfa.compile_patterns(["ab" + "\\"])
+    assert dfa.match_mask_reference(t, b"xab\\y") == 1
+    assert dfa.match_mask_reference(t, b"xaby") == 0
+
+
+def test_max_len_metadata_bounded_vs_unbounded():
+    """dfa.py ~:186 (`hi == INF` in the max-length fold): bounded patterns
+    report their exact max length; any unbounded atom zeroes it."""
+    t = dfa.compile_patterns([r"\d{3}-\d{2}", r"x+y"])
+    assert t.max_len[0] == 6       # 3 digits + dash + 2 digits
+    assert t.max_len[1] == 0       # unbounded → sentinel 0
+    t2 = dfa.compile_patterns([r"a{2,4}b"])
+    assert t2.max_len[0] == 5
+
+
+def test_byte_class_partition_count():
+    """dfa.py ~:246 (`n_classes > 255` guard + class partition): the class
+    table must map equivalent bytes to one class and distinct sets apart —
+    matching behavior is preserved under the compression."""
+    t = dfa.compile_patterns([r"[ab]z"])
+    assert dfa.match_mask_reference(t, b"az") == 1
+    assert dfa.match_mask_reference(t, b"bz") == 1
+    assert dfa.match_mask_reference(t, b"cz") == 0
+    # a and b must share a class; z distinct; others background
+    assert t.klass[ord("a")] == t.klass[ord("b")]
+    assert t.klass[ord("z")] != t.klass[ord("a")]
