@@ -1197,6 +1197,7 @@ class GpuPluginPipeline:
                     pii_mode = 2
                 njs = np.asarray(nat_idx, dtype=np.int64)
                 tis = tool_idx[njs]
+                t_call = self._tic()
                 st, found, deny_hit, harm_hit, schema_ok, rw_arena, rb, re_, sb, se = hip.rewrite_rows(
                     blob, np.ascontiguousarray(args_b[njs]), np.ascontiguousarray(args_e[njs]),
                     np.asarray(flags_l, dtype=np.uint8), np.asarray(want_l, dtype=np.uint32),
@@ -1209,6 +1210,7 @@ class GpuPluginPipeline:
                     sk_tables=self._sk_tables,
                     sk_lo=np.ascontiguousarray(self._sk_range[tis, 0]) if self._sk_tables is not None else None,
                     sk_hi=np.ascontiguousarray(self._sk_range[tis, 1]) if self._sk_tables is not None else None)
+                self._toc("rw_call", t_call)
                 punted: List[int] = []
                 done_m = st == hip.RW_DONE
                 done_ks = np.nonzero(done_m)[0]
