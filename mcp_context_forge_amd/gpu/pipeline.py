@@ -1249,7 +1249,9 @@ class GpuPluginPipeline:
                         self.blocked += 1
                     else:
                         punted.append(j)
-                py_js = [j for j in rewrite_js if j not in set(nat_idx)] + punted
+                nat_set = set(nat_idx)  # hoisted: inside the comprehension this
+                # was rebuilt per element — O(n²), 31 ms/batch at 20% flagged
+                py_js = [j for j in rewrite_js if j not in nat_set] + punted
                 py_js.sort()
         self._toc("rw_c", t_sub)
         t_sub = self._tic()
