@@ -42,7 +42,8 @@ from typing import List, Optional
 REFERENCE_BEST_RPS = 10454.16  # MCP tools-only 60s/1000u, Rust full mode
 
 
-def make_request(rng: random.Random, tool_names: List[str], rid: int, flagged_frac: float) -> bytes:
+def make_request(rng: random.Random, tool_names: List[str], rid: int, flagged_frac: float,
+                 pad_bytes: int = 0) -> bytes:
     name = tool_names[rng.randrange(len(tool_names))]
     r = rng.random()
     if name.endswith("convert_time"):
@@ -57,6 +58,8 @@ def make_request(rng: random.Random, tool_names: List[str], rid: int, flagged_fr
         args = {"msg": f"payload {rng.randrange(1 << 30)} lorem ipsum dolor sit amet", "n": rng.randrange(100)}
     if r < flagged_frac * 0.5:
         args["note"] = f"contact me at user{rng.randrange(1000)}@example.com"  # PII slow path
+    if pad_bytes:
+        args["payload"] = "x" * pad_bytes  # size probe: larger real-world bodies
     elif r < flagged_frac:
         args["note"] = "this is   spaced\ttext"  # normalizer slow path
     return json.dumps(
@@ -190,7 +193,7 @@ async def run_http(args, rank: int, world: int, use_gpu: bool, R: int):
     with tempfile.NamedTemporaryFile("wb", suffix=".jsonl", delete=False) as f:
         payload_file = f.name
         for i in range(args.payloads):
-            f.write(make_request(rng, all_names, i, args.flagged_frac) + b"\n")
+            f.write(make_request(rng, all_names, i, args.flagged_frac, args.payload_bytes) + b"\n")
 
     hey_threads = max(2, min(6, (os.cpu_count() or 16) // (2 * max(world, 1))))
     proc = await asyncio.create_subprocess_exec(
@@ -287,7 +290,7 @@ async def run_engine_only(args, rank: int, world: int, use_gpu: bool, R: int):
         raws = []
         for i in range(R):
             d = rng.randrange(world)
-            raws.append(make_request(rng, all_names[d], step * R + i, args.flagged_frac))
+            raws.append(make_request(rng, all_names[d], step * R + i, args.flagged_frac, args.payload_bytes))
         return raws
 
     warm_data = [gen_step(s) for s in range(args.warmup)]
@@ -380,6 +383,8 @@ async def main() -> None:
                     help="batches in flight in the edge loop (host/GPU overlap)")
     ap.add_argument("--upstreams", type=int, default=64)
     ap.add_argument("--flagged-frac", type=float, default=0.02)
+    ap.add_argument("--payload-bytes", type=int, default=0,
+                    help="pad every request's arguments by N bytes (size probe)")
     ap.add_argument("--semcache", action="store_true",
                     help="allowlist the bench tools in the semantic cache (labeled in output)")
     ap.add_argument("--no-gpu", action="store_true")
