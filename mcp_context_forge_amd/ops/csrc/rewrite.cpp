@@ -224,12 +224,22 @@ bool pii_sub_one(std::string& s, const PiiDef& p) {
     std::string out;
     bool any = false;
     size_t i = 0;
+    const bool email = p.fn == m_email;
     while (i < s.size()) {
         size_t len = 0;
         if (p.fn(s, i, len)) {
             out += p.repl;
             i += len;
             any = true;
+        } else if (email && klass_email_local((uint8_t)s[i])) {
+            // every start inside one local-part run shares the same '@'
+            // position and domain, so they all fail identically — skip the
+            // run. Kills the O(L²) cost on long word runs (python re pays
+            // it; outputs are unchanged, only the scan order collapses).
+            size_t j = i + 1;
+            while (j < s.size() && klass_email_local((uint8_t)s[j])) ++j;
+            out.append(s, i, j - i);
+            i = j;
         } else {
             out += s[i++];
         }
