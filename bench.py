@@ -43,8 +43,11 @@ REFERENCE_BEST_RPS = 10454.16  # MCP tools-only 60s/1000u, Rust ful
 
 
 def make_request(rng: random.Random, tool_names: List[str], rid: int, flagged_frac: float,
-                 pad_bytes: int = 0) -> bytes:
+                 pad_bytes: int = 0, unknown_frac: float = 0.0,
+                 nonascii_frac: float = 0.0) -> bytes:
     name = tool_names[rng.randrange(len(tool_names))]
+    if unknown_frac and rng.random() < unknown_frac:
+        name = f"no-such-tool-{rng.randrange(64)}"
     r = rng.random()
     if name.endswith("convert_time"):
         args = {
@@ -60,6 +63,8 @@ def make_request(rng: random.Random, tool_names: List[str], rid: int, flagged_fr
         args["note"] = f"contact me at user{rng.randrange(1000)}@example.com"  # PII slow path
     if pad_bytes:
         args["payload"] = "x" * pad_bytes  # size probe: larger real-world bodies
+    if nonascii_frac and rng.random() < nonascii_frac:
+        args["intl"] = "café ünïcode 日本語 №" + str(rid % 97)  # punt-path pressure
     elif r < flagged_frac:
         args["note"] = "this is   spaced\ttext"  # normalizer slow path
     return json.dumps(
@@ -193,7 +198,8 @@ async def run_http(args, rank: int, world: int, use_gpu: bool, R: int):
     with tempfile.NamedTemporaryFile("wb", suffix=".jsonl", delete=False) as f:
         payload_file = f.name
         for i in range(args.payloads):
-            f.write(make_request(rng, all_names, i, args.flagged_frac, args.payload_bytes) + b"\n")
+            f.write(make_request(rng, all_names, i, args.flagged_frac, args.payload_bytes,
+                                 args.unknown_frac, args.nonascii_frac) + b"\n")
 
     hey_threads = max(2, min(6, (os.cpu_count() or 16) // (2 * max(world, 1))))
     proc = await asyncio.create_subprocess_exec(
@@ -290,7 +296,8 @@ async def run_engine_only(args, rank: int, world: int, use_gpu: bool, R: int):
         raws = []
         for i in range(R):
             d = rng.randrange(world)
-            raws.append(make_request(rng, all_names[d], step * R + i, args.flagged_frac, args.payload_bytes))
+            raws.append(make_request(rng, all_names[d], step * R + i, args.flagged_frac,
+                                      args.payload_bytes, args.unknown_frac, args.nonascii_frac))
         return raws
 
     warm_data = [gen_step(s) for s in range(args.warmup)]
@@ -385,6 +392,10 @@ async def main() -> None:
     ap.add_argument("--flagged-frac", type=float, default=0.02)
     ap.add_argument("--payload-bytes", type=int, default=0,
                     help="pad every request's arguments by N bytes (size probe)")
+    ap.add_argument("--unknown-frac", type=float, default=0.0,
+                    help="fraction of requests targeting unknown tools (error-path probe)")
+    ap.add_argument("--nonascii-frac", type=float, default=0.0,
+                    help="fraction of requests with non-ASCII args (punt-path probe)")
     ap.add_argument("--semcache", action="store_true",
                     help="allowlist the bench tools in the semantic cache (labeled in output)")
     ap.add_argument("--no-gpu", action="store_true")
