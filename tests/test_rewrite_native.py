@@ -339,3 +339,25 @@ def test_fast_schema_check_on_rewritten_tree():
         want = 1 if not validate(json.loads(raw), schema) else 0
         assert status[i] == hip.RW_DONE
         assert int(schema_ok[i]) == want, (i, raw, schema_ok[i])
+
+
+def test_email_long_run_skip_exactness():
+    """Regression for the O(L²) email-matcher fix: the run-skip must not
+    change outputs on long word runs, runs ending in '@', adjacent
+    emails, or runs that ARE the local part of a match."""
+    cases = [
+        "x" * 3000,                                  # pure run, no @
+        "x" * 2000 + "@example.com",                 # run IS the local part
+        "x" * 500 + "@x tail",                       # @ present, domain invalid
+        "a@b.co" + "y" * 1000 + "c@d.io",            # match, run, match
+        "@" * 50 + "a@b.cc",                         # @ runs
+        "." * 100 + "u@v.ww" + "." * 100,            # dots are local-class chars
+        "p%q+r@s-t.uv mixed " + "_" * 800,
+    ]
+    raws = [json.dumps({"m": c}, separators=(",", ":")).encode() for c in cases]
+    status, found, dh, hh, sok, arena, ob, oe, sb, se = run_c(raws, do_flags=[2] * len(raws))
+    for i, raw in enumerate(raws):
+        sorted_exp, wire_exp, names = py_reference(raw, do_norm=False, do_pii=True)
+        assert status[i] == hip.RW_DONE, (i, status[i])
+        assert arena[ob[i]:oe[i]].tobytes() == wire_exp, (i, cases[i][:40])
+        assert names_of(int(found[i])) == names, (i, cases[i][:40])
